@@ -1,0 +1,200 @@
+/* gx_executor.h — C-ABI drop-in boundary for the TiDB analytical hot path.
+ *
+ * This header declares the operator surface a cgo shim in TiDB would bind in
+ * place of the CPU executors built by `executorBuilder.build`
+ * (reference: pkg/executor/builder.go:283-311 dispatch on
+ * PhysicalSelection/PhysicalProjection/PhysicalHashAgg/PhysicalTopN/...).
+ *
+ * The operator contract mirrors `exec.Executor`
+ * (reference: pkg/executor/internal/exec/executor.go:224-250):
+ *   Open() -> repeated Next(chunk) filling <= 1024 rows, 0 rows = EOF -> Close().
+ * gx_next is single-threaded per executor handle; the engine may use internal
+ * streams/threads (reference threading contract, executor.go:238).
+ *
+ * Chunk layout matches pkg/util/chunk exactly
+ * (reference: pkg/util/chunk/column.go:74-82, chunk.go:35-54):
+ *   - fixed-width columns: data = length*elem_size raw little-endian bytes
+ *     (8 B int64/uint64/float64/Time, 40 B MyDecimal);
+ *   - null bitmap: 1 bit per row, LSB-first within each byte, 1 = NOT NULL
+ *     (column.go:225-267);
+ *   - var-len columns: row i spans data[offsets[i] .. offsets[i+1]).
+ *
+ * Two shared libraries implement this ABI:
+ *   - liboracle.so   (oracle/…):  CPU restatement of the reference executor —
+ *                                 TEST INFRASTRUCTURE ONLY (parity anchor).
+ *   - libgxexec.so   (tidb_amd/csrc/…): the MI355X product engine (HIP/CDNA4).
+ */
+#ifndef GX_EXECUTOR_H
+#define GX_EXECUTOR_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- status codes (mirror reference error classes) ---- */
+enum {
+  GX_OK = 0,
+  GX_ERR_TRUNCATED = 1, /* types.ErrTruncated  */
+  GX_ERR_OVERFLOW = 2,  /* types.ErrOverflow   */
+  GX_ERR_DIV_ZERO = 3,  /* types.ErrDivByZero  */
+  GX_ERR_BAD_NUMBER = 4,/* types.ErrBadNumber  */
+  GX_ERR_INVALID = -1,  /* bad plan / usage    */
+  GX_ERR_NO_GPU = -2,   /* product engine built without / cannot reach a GPU */
+  GX_ERR_INTERNAL = -3
+};
+
+/* ---- column element types (subset of mysql types on the hot path) ---- */
+enum {
+  GX_TYPE_I64 = 0,     /* TypeLonglong, 8 B */
+  GX_TYPE_F64 = 1,     /* TypeDouble, 8 B */
+  GX_TYPE_DECIMAL = 2, /* TypeNewDecimal, 40 B MyDecimal struct (mydecimal.go:236-248) */
+  GX_TYPE_TIME = 3,    /* TypeDate/Datetime, 8 B packed CoreTime (core_time.go:25, time.go:233-265) */
+  GX_TYPE_STRING = 4   /* var-len (char/varchar), offsets + bytes */
+};
+
+/* ---- chunk ---- */
+typedef struct gx_col {
+  void*    data;        /* fixed: elem data; varlen: byte payload */
+  uint8_t* null_bitmap; /* ceil(length/8) bytes; LSB-first; 1 = NOT NULL; may be NULL => all not-null */
+  int64_t* offsets;     /* varlen only: length+1 entries, offsets[0] = 0; NULL for fixed */
+  int32_t  length;      /* number of rows */
+  int32_t  elem_size;   /* fixed elem bytes; -1 for varlen */
+  int64_t  data_cap;    /* capacity of data in bytes (output chunks: caller-provided) */
+  int64_t  offsets_cap; /* capacity of offsets in entries (output chunks) */
+} gx_col;
+
+typedef struct gx_chunk {
+  gx_col* cols;
+  int32_t n_cols;
+  int32_t n_rows;
+} gx_chunk;
+
+/* ---- scalar function codes (vectorized builtins on the hot path) ----
+ * compare family: reference pkg/expression/builtin_compare_vec_generated.go:64,138-170
+ * decimal arithmetic: builtin_arithmetic_vec.go:529 (Mul), :328 (Minus), :973 (Plus), :67 (Div)
+ */
+enum {
+  GX_F_LT = 0, GX_F_LE = 1, GX_F_GT = 2, GX_F_GE = 3, GX_F_EQ = 4, GX_F_NE = 5,
+  GX_F_PLUS = 16, GX_F_MINUS = 17, GX_F_MUL = 18, GX_F_DIV = 19
+};
+
+/* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */
+enum {
+  GX_AGG_COUNT = 0,    /* arg expr = -1 => count(*) */
+  GX_AGG_SUM = 1,      /* sum4Decimal / sum float (func_sum.go:44,224) */
+  GX_AGG_AVG = 2,      /* avgOriginal4Decimal: state {sum, count} (func_avg.go:69-135) */
+  GX_AGG_MIN = 3,
+  GX_AGG_MAX = 4,
+  GX_AGG_FIRSTROW = 5
+};
+
+/* ---- hash-agg execution mode (AggFuncDesc partial/final split,
+ * reference pkg/executor/builder.go:2239-2258, aggfuncs.go:224-263) ---- */
+enum {
+  GX_AGG_MODE_COMPLETE = 0, /* raw rows in, final values out */
+  GX_AGG_MODE_PARTIAL = 1,  /* raw rows in, canonical partial-state chunk out */
+  GX_AGG_MODE_FINAL = 2     /* partial-state chunks in, final values out (MergePartialResult) */
+};
+
+/* ---- synthetic TPC-H tables (MockDataSource analog,
+ * reference pkg/executor/internal/testutil/testutil.go:45-348) ---- */
+enum { GX_TPCH_LINEITEM = 0, GX_TPCH_ORDERS = 1, GX_TPCH_CUSTOMER = 2 };
+
+/* ---- plan builder ---- */
+typedef struct gx_pb gx_pb;
+typedef struct gx_exec gx_exec;
+
+gx_pb* gx_pb_new(void);
+void   gx_pb_free(gx_pb* pb);
+
+/* expressions: return expr id (>=0) or negative status */
+int32_t gx_pb_colref(gx_pb* pb, int32_t col_idx, int32_t type, int32_t frac);
+int32_t gx_pb_const_i64(gx_pb* pb, int64_t v);
+int32_t gx_pb_const_f64(gx_pb* pb, double v);
+int32_t gx_pb_const_time(gx_pb* pb, uint64_t packed_core_time);
+int32_t gx_pb_const_dec(gx_pb* pb, const uint8_t dec40[40]);
+int32_t gx_pb_const_str(gx_pb* pb, const char* s, int32_t len);
+/* ret_frac: decimal result fraction metadata (field-type frac); -1 if N/A */
+int32_t gx_pb_call(gx_pb* pb, int32_t func, int32_t ret_type, int32_t ret_frac,
+                   const int32_t* args, int32_t n_args);
+
+/* plan nodes: return node id (>=0) or negative status */
+int32_t gx_pb_source(gx_pb* pb, const int32_t* col_types, const int32_t* col_fracs,
+                     int32_t n_cols);
+int32_t gx_pb_selection(gx_pb* pb, int32_t child, const int32_t* conds, int32_t n_conds);
+int32_t gx_pb_projection(gx_pb* pb, int32_t child, const int32_t* exprs, int32_t n_exprs);
+int32_t gx_pb_hashagg(gx_pb* pb, int32_t child,
+                      const int32_t* group_exprs, int32_t n_group,
+                      const int32_t* agg_funcs, const int32_t* agg_args,
+                      const int32_t* agg_fracs, int32_t n_aggs, int32_t mode);
+int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
+                   const uint8_t* key_desc, int32_t n_keys,
+                   int64_t limit, int64_t offset);
+/* join_type: 0 = inner (HashJoinV2 equivalent, pkg/executor/join/hash_join_v2.go) */
+int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
+                       const int32_t* build_keys, const int32_t* probe_keys,
+                       int32_t n_keys, int32_t join_type);
+
+/* build: device = -1 for the library's default device (oracle: CPU; product:
+ * current HIP device). The PRODUCT library fails loudly (GX_ERR_NO_GPU) if no
+ * MI355X is reachable — it has no CPU fallback. */
+gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device);
+
+/* bind a source node to caller-owned host chunks (uploaded once at Open) */
+int32_t gx_bind_chunks(gx_exec* ex, int32_t source_node,
+                       const gx_chunk* chunks, int32_t n_chunks);
+/* bind a source node to the deterministic synthetic TPC-H generator
+ * (seeded; row_offset for multi-GPU row-range sharding — the region-shard
+ * analog of store/copr/coprocessor.go:525) */
+int32_t gx_bind_tpch(gx_exec* ex, int32_t source_node, int32_t table,
+                     int64_t n_rows, uint64_t seed, int64_t row_offset);
+
+int32_t gx_open(gx_exec* ex);
+/* fills out-chunk (caller-allocated buffers, data_cap/offsets_cap honored),
+ * rows_out = rows appended; 0 = EOF. */
+int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out);
+int32_t gx_close(gx_exec* ex);
+void    gx_exec_free(gx_exec* ex);
+const char* gx_last_error(gx_exec* ex);
+
+/* engine info: returns 1 if this library executes on GPU, 0 if CPU oracle */
+int32_t gx_engine_is_gpu(void);
+const char* gx_engine_name(void);
+
+/* ---- decimal helpers (each library's own implementation; used by golden
+ * vector tests against pkg/types/mydecimal_test.go answers) ---- */
+int32_t gx_dec_from_string(const char* s, int32_t len, uint8_t out40[40]);
+/* ToString (no rounding), returns length written (mydecimal.go:328) */
+int32_t gx_dec_to_string(const uint8_t dec40[40], char* buf, int32_t buf_len);
+/* String() = Round(resultFrac, HalfUp) + ToString (mydecimal.go:277) */
+int32_t gx_dec_display_string(const uint8_t dec40[40], char* buf, int32_t buf_len);
+int32_t gx_dec_add(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]);
+int32_t gx_dec_sub(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]);
+int32_t gx_dec_mul(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]);
+int32_t gx_dec_div(const uint8_t a[40], const uint8_t b[40], uint8_t out[40], int32_t frac_incr);
+/* round_mode: 5 = ModeHalfUp, 10 = ModeTruncate, 0 = ModeCeiling */
+int32_t gx_dec_round(const uint8_t in[40], int32_t frac, int32_t round_mode, uint8_t out[40]);
+int32_t gx_dec_compare(const uint8_t a[40], const uint8_t b[40]);
+int32_t gx_dec_to_bin(const uint8_t in[40], int32_t precision, int32_t frac,
+                      uint8_t* out, int32_t* out_len);
+int32_t gx_dec_from_bin(const uint8_t* bin, int32_t bin_len, int32_t precision,
+                        int32_t frac, uint8_t out[40]);
+int32_t gx_dec_to_hash_key(const uint8_t in[40], uint8_t* out, int32_t* out_len);
+int32_t gx_dec_from_i64(int64_t v, uint8_t out[40]);
+int32_t gx_dec_shift(const uint8_t in[40], int32_t shift, uint8_t out[40]);
+int32_t gx_dec_result_frac(const uint8_t dec40[40]);
+
+/* ---- time helpers (packed CoreTime, core_time.go:25 / time.go:233-265) ---- */
+uint64_t gx_time_from_date(int32_t year, int32_t month, int32_t day);
+uint64_t gx_time_from_datetime(int32_t year, int32_t month, int32_t day,
+                               int32_t hour, int32_t minute, int32_t second,
+                               int32_t microsecond, int32_t type_and_fsp);
+int32_t  gx_time_compare(uint64_t a, uint64_t b); /* compareTime, core_time.go:256 */
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* GX_EXECUTOR_H */

@@ -1,0 +1,115 @@
+// oracle/capi_helpers.cpp — C-ABI decimal/time helper exports (oracle flavor).
+// ORACLE / TEST INFRASTRUCTURE ONLY.
+#include <cstring>
+
+#include "../include/gx_executor.h"
+#include "core_time.h"
+#include "mydecimal.h"
+
+using oracle::MyDecimal;
+
+static inline const MyDecimal* D(const uint8_t* p) {
+  return reinterpret_cast<const MyDecimal*>(p);
+}
+static inline MyDecimal* D(uint8_t* p) { return reinterpret_cast<MyDecimal*>(p); }
+
+extern "C" {
+
+int32_t gx_dec_from_string(const char* s, int32_t len, uint8_t out40[40]) {
+  MyDecimal d;
+  int32_t err = d.FromString(s, len);
+  std::memcpy(out40, &d, 40);
+  return err;
+}
+
+int32_t gx_dec_to_string(const uint8_t dec40[40], char* buf, int32_t buf_len) {
+  std::string s = D(dec40)->ToString();
+  if ((int32_t)s.size() + 1 > buf_len) return GX_ERR_INVALID;
+  std::memcpy(buf, s.data(), s.size());
+  buf[s.size()] = 0;
+  return (int32_t)s.size();
+}
+
+int32_t gx_dec_display_string(const uint8_t dec40[40], char* buf, int32_t buf_len) {
+  std::string s = D(dec40)->DisplayString();
+  if ((int32_t)s.size() + 1 > buf_len) return GX_ERR_INVALID;
+  std::memcpy(buf, s.data(), s.size());
+  buf[s.size()] = 0;
+  return (int32_t)s.size();
+}
+
+int32_t gx_dec_add(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]) {
+  return oracle::DecimalAdd(D(a), D(b), D(out));
+}
+int32_t gx_dec_sub(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]) {
+  return oracle::DecimalSub(D(a), D(b), D(out));
+}
+int32_t gx_dec_mul(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]) {
+  return oracle::DecimalMul(D(a), D(b), D(out));
+}
+int32_t gx_dec_div(const uint8_t a[40], const uint8_t b[40], uint8_t out[40],
+                   int32_t frac_incr) {
+  return oracle::DecimalDiv(D(a), D(b), D(out), frac_incr);
+}
+int32_t gx_dec_round(const uint8_t in[40], int32_t frac, int32_t round_mode,
+                     uint8_t out[40]) {
+  MyDecimal tmp = *D(in);
+  int32_t err = tmp.Round(&tmp, frac, (oracle::RoundMode)round_mode);
+  std::memcpy(out, &tmp, 40);
+  return err;
+}
+int32_t gx_dec_compare(const uint8_t a[40], const uint8_t b[40]) {
+  return D(a)->Compare(*D(b));
+}
+int32_t gx_dec_to_bin(const uint8_t in[40], int32_t precision, int32_t frac,
+                      uint8_t* out, int32_t* out_len) {
+  int written = 0;
+  int32_t err = D(in)->WriteBin(precision, frac, out, &written);
+  *out_len = written;
+  return err;
+}
+int32_t gx_dec_from_bin(const uint8_t* bin, int32_t bin_len, int32_t precision,
+                        int32_t frac, uint8_t out[40]) {
+  MyDecimal d;
+  int binSize = 0;
+  int32_t err = d.FromBin(bin, bin_len, precision, frac, &binSize);
+  std::memcpy(out, &d, 40);
+  return err;
+}
+int32_t gx_dec_to_hash_key(const uint8_t in[40], uint8_t* out, int32_t* out_len) {
+  int written = 0;
+  int32_t err = D(in)->ToHashKey(out, &written);
+  *out_len = written;
+  return err;
+}
+int32_t gx_dec_from_i64(int64_t v, uint8_t out[40]) {
+  MyDecimal d;
+  d.FromInt(v);
+  std::memcpy(out, &d, 40);
+  return GX_OK;
+}
+int32_t gx_dec_shift(const uint8_t in[40], int32_t shift, uint8_t out[40]) {
+  MyDecimal tmp = *D(in);
+  int32_t err = tmp.Shift(shift);
+  std::memcpy(out, &tmp, 40);
+  return err;
+}
+int32_t gx_dec_result_frac(const uint8_t dec40[40]) {
+  return (int32_t)D(dec40)->resultFrac;
+}
+
+uint64_t gx_time_from_date(int32_t year, int32_t month, int32_t day) {
+  return oracle::TimeFromDate(year, month, day);
+}
+uint64_t gx_time_from_datetime(int32_t year, int32_t month, int32_t day,
+                               int32_t hour, int32_t minute, int32_t second,
+                               int32_t microsecond, int32_t type_and_fsp) {
+  return oracle::TimeFromDatetime(year, month, day, hour, minute, second,
+                                  microsecond, type_and_fsp);
+}
+int32_t gx_time_compare(uint64_t a, uint64_t b) { return oracle::CompareTime(a, b); }
+
+int32_t gx_engine_is_gpu(void) { return 0; }
+const char* gx_engine_name(void) { return "oracle-cpu"; }
+
+}  // extern "C"

@@ -151,6 +151,11 @@ int32_t gx_bind_chunks(gx_exec* ex, int32_t source_node,
  * analog of store/copr/coprocessor.go:525) */
 int32_t gx_bind_tpch(gx_exec* ex, int32_t source_node, int32_t table,
                      int64_t n_rows, uint64_t seed, int64_t row_offset);
+/* like gx_bind_tpch but with the TOTAL table row count (for cross-table key
+ * ranges under row-range sharding); gx_bind_tpch == total_rows = n_rows. */
+int32_t gx_bind_tpch_sharded(gx_exec* ex, int32_t source_node, int32_t table,
+                             int64_t n_rows, uint64_t seed, int64_t row_offset,
+                             int64_t total_rows);
 
 int32_t gx_open(gx_exec* ex);
 /* fills out-chunk (caller-allocated buffers, data_cap/offsets_cap honored),

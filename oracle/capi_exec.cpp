@@ -1,28 +1,305 @@
-// oracle/capi_exec.cpp — executor C-ABI (oracle flavor). Implementation lands
-// with exec.cpp; stubs below are replaced incrementally.
+// oracle/capi_exec.cpp — executor C-ABI implementation (oracle flavor).
+// ORACLE / TEST INFRASTRUCTURE ONLY.
+#include <cstring>
+#include <map>
+#include <memory>
+#include <string>
+
 #include "../include/gx_executor.h"
+#include "exec.h"
+
+using namespace oracle;
+
+struct gx_pb {
+  Plan plan;
+  std::string err;
+};
+
+struct gx_exec {
+  Plan plan;
+  int root = -1;
+  std::map<int, SourceBinding> bindings;
+  std::unique_ptr<Exec> exec;
+  std::string err;
+  bool opened = false;
+};
+
 extern "C" {
-gx_pb* gx_pb_new(void) { return nullptr; }
-void gx_pb_free(gx_pb*) {}
-int32_t gx_pb_colref(gx_pb*, int32_t, int32_t, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_const_i64(gx_pb*, int64_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_const_f64(gx_pb*, double) { return GX_ERR_INVALID; }
-int32_t gx_pb_const_time(gx_pb*, uint64_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_const_dec(gx_pb*, const uint8_t*) { return GX_ERR_INVALID; }
-int32_t gx_pb_const_str(gx_pb*, const char*, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_call(gx_pb*, int32_t, int32_t, int32_t, const int32_t*, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_source(gx_pb*, const int32_t*, const int32_t*, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_selection(gx_pb*, int32_t, const int32_t*, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_projection(gx_pb*, int32_t, const int32_t*, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_hashagg(gx_pb*, int32_t, const int32_t*, int32_t, const int32_t*, const int32_t*, const int32_t*, int32_t, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_topn(gx_pb*, int32_t, const int32_t*, const uint8_t*, int32_t, int64_t, int64_t) { return GX_ERR_INVALID; }
-int32_t gx_pb_hashjoin(gx_pb*, int32_t, int32_t, const int32_t*, const int32_t*, int32_t, int32_t) { return GX_ERR_INVALID; }
-gx_exec* gx_build(gx_pb*, int32_t, int32_t) { return nullptr; }
-int32_t gx_bind_chunks(gx_exec*, int32_t, const gx_chunk*, int32_t) { return GX_ERR_INVALID; }
-int32_t gx_bind_tpch(gx_exec*, int32_t, int32_t, int64_t, uint64_t, int64_t) { return GX_ERR_INVALID; }
-int32_t gx_open(gx_exec*) { return GX_ERR_INVALID; }
-int32_t gx_next(gx_exec*, gx_chunk*, int32_t*) { return GX_ERR_INVALID; }
-int32_t gx_close(gx_exec*) { return GX_ERR_INVALID; }
-void gx_exec_free(gx_exec*) {}
-const char* gx_last_error(gx_exec*) { return "not implemented"; }
+
+gx_pb* gx_pb_new(void) { return new gx_pb(); }
+void gx_pb_free(gx_pb* pb) { delete pb; }
+
+static int32_t addExpr(gx_pb* pb, Expr e) {
+  pb->plan.exprs.push_back(std::move(e));
+  return (int32_t)pb->plan.exprs.size() - 1;
 }
+
+int32_t gx_pb_colref(gx_pb* pb, int32_t col_idx, int32_t type, int32_t frac) {
+  Expr e;
+  e.kind = EK_COLREF;
+  e.colIdx = col_idx;
+  e.retType = type;
+  e.retFrac = frac;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_i64(gx_pb* pb, int64_t v) {
+  Expr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_I64;
+  e.constI64 = v;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_f64(gx_pb* pb, double v) {
+  Expr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_F64;
+  e.constF64 = v;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_time(gx_pb* pb, uint64_t v) {
+  Expr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_TIME;
+  e.constTime = v;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_dec(gx_pb* pb, const uint8_t dec40[40]) {
+  Expr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_DECIMAL;
+  std::memcpy(&e.constDec, dec40, 40);
+  e.retFrac = e.constDec.resultFrac;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_str(gx_pb* pb, const char* s, int32_t len) {
+  Expr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_STRING;
+  e.constStr.assign(s, len);
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_call(gx_pb* pb, int32_t func, int32_t ret_type, int32_t ret_frac,
+                   const int32_t* args, int32_t n_args) {
+  Expr e;
+  e.kind = EK_CALL;
+  e.func = func;
+  e.retType = ret_type;
+  e.retFrac = ret_frac < 0 ? 0 : ret_frac;
+  for (int i = 0; i < n_args; i++) {
+    if (args[i] < 0 || args[i] >= (int32_t)pb->plan.exprs.size()) return GX_ERR_INVALID;
+    e.args.push_back(args[i]);
+  }
+  return addExpr(pb, std::move(e));
+}
+
+static int32_t addNode(gx_pb* pb, PlanNode n) {
+  pb->plan.nodes.push_back(std::move(n));
+  return (int32_t)pb->plan.nodes.size() - 1;
+}
+
+int32_t gx_pb_source(gx_pb* pb, const int32_t* col_types, const int32_t* col_fracs,
+                     int32_t n_cols) {
+  PlanNode n;
+  n.kind = PK_SOURCE;
+  for (int i = 0; i < n_cols; i++) {
+    n.colTypes.push_back(col_types[i]);
+    n.colFracs.push_back(col_fracs ? col_fracs[i] : 0);
+  }
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_selection(gx_pb* pb, int32_t child, const int32_t* conds, int32_t n_conds) {
+  PlanNode n;
+  n.kind = PK_SELECTION;
+  n.child = child;
+  for (int i = 0; i < n_conds; i++) n.exprs.push_back(conds[i]);
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_projection(gx_pb* pb, int32_t child, const int32_t* exprs, int32_t n_exprs) {
+  PlanNode n;
+  n.kind = PK_PROJECTION;
+  n.child = child;
+  for (int i = 0; i < n_exprs; i++) n.exprs.push_back(exprs[i]);
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_hashagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
+                      int32_t n_group, const int32_t* agg_funcs,
+                      const int32_t* agg_args, const int32_t* agg_fracs,
+                      int32_t n_aggs, int32_t mode) {
+  PlanNode n;
+  n.kind = PK_HASHAGG;
+  n.child = child;
+  n.aggMode = mode;
+  for (int i = 0; i < n_group; i++) n.exprs.push_back(group_exprs[i]);
+  for (int i = 0; i < n_aggs; i++) {
+    n.aggFuncs.push_back(agg_funcs[i]);
+    n.aggArgs.push_back(agg_args[i]);
+    n.aggFracs.push_back(agg_fracs ? agg_fracs[i] : 0);
+  }
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
+                   const uint8_t* key_desc, int32_t n_keys, int64_t limit,
+                   int64_t offset) {
+  PlanNode n;
+  n.kind = limit < 0 ? PK_SORT : PK_TOPN;
+  n.child = child;
+  for (int i = 0; i < n_keys; i++) {
+    n.exprs.push_back(key_exprs[i]);
+    n.keyDesc.push_back(key_desc ? key_desc[i] : 0);
+  }
+  n.limit = limit;
+  n.offset = offset;
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
+                       const int32_t* build_keys, const int32_t* probe_keys,
+                       int32_t n_keys, int32_t join_type) {
+  PlanNode n;
+  n.kind = PK_HASHJOIN;
+  n.child = build_child;
+  n.child2 = probe_child;
+  n.joinType = join_type;
+  for (int i = 0; i < n_keys; i++) {
+    n.buildKeys.push_back(build_keys[i]);
+    n.probeKeys.push_back(probe_keys[i]);
+  }
+  return addNode(pb, std::move(n));
+}
+
+gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
+  (void)device;  // oracle is CPU-only
+  if (!pb || root < 0 || root >= (int32_t)pb->plan.nodes.size()) return nullptr;
+  auto* ex = new gx_exec();
+  ex->plan = pb->plan;
+  ex->root = root;
+  return ex;
+}
+
+int32_t gx_bind_chunks(gx_exec* ex, int32_t source_node, const gx_chunk* chunks,
+                       int32_t n_chunks) {
+  if (!ex || source_node < 0 || source_node >= (int32_t)ex->plan.nodes.size())
+    return GX_ERR_INVALID;
+  const PlanNode& node = ex->plan.nodes[source_node];
+  if (node.kind != PK_SOURCE) return GX_ERR_INVALID;
+  SourceBinding b;
+  b.haveChunks = true;
+  for (int i = 0; i < n_chunks; i++) {
+    Chunk c;
+    if (chunks[i].n_cols != (int32_t)node.colTypes.size()) return GX_ERR_INVALID;
+    for (int j = 0; j < chunks[i].n_cols; j++)
+      c.cols.push_back(Column::fromGx(chunks[i].cols[j], node.colTypes[j], node.colFracs[j]));
+    b.chunks.push_back(std::move(c));
+  }
+  ex->bindings[source_node] = std::move(b);
+  return GX_OK;
+}
+
+int32_t gx_bind_tpch_sharded(gx_exec* ex, int32_t source_node, int32_t table,
+                             int64_t n_rows, uint64_t seed, int64_t row_offset,
+                             int64_t total_rows) {
+  if (!ex || source_node < 0 || source_node >= (int32_t)ex->plan.nodes.size())
+    return GX_ERR_INVALID;
+  if (ex->plan.nodes[source_node].kind != PK_SOURCE) return GX_ERR_INVALID;
+  SourceBinding b;
+  b.tpchTable = table;
+  b.tpchRows = n_rows;
+  b.tpchSeed = seed;
+  b.tpchRowOffset = row_offset;
+  // stash total rows in tpchRows semantics: generator needs it for key ranges
+  b.tpchTotalRows = total_rows;
+  ex->bindings[source_node] = std::move(b);
+  return GX_OK;
+}
+
+int32_t gx_bind_tpch(gx_exec* ex, int32_t source_node, int32_t table,
+                     int64_t n_rows, uint64_t seed, int64_t row_offset) {
+  return gx_bind_tpch_sharded(ex, source_node, table, n_rows, seed, row_offset,
+                              n_rows);
+}
+
+int32_t gx_open(gx_exec* ex) {
+  if (!ex) return GX_ERR_INVALID;
+  ex->exec = BuildExec(ex->plan, ex->root, &ex->bindings, &ex->err);
+  if (!ex->exec) return GX_ERR_INVALID;
+  int32_t ec = ex->exec->open();
+  if (ec) {
+    ex->err = ex->exec->err;
+    return ec;
+  }
+  ex->opened = true;
+  return GX_OK;
+}
+
+// copy an oracle Chunk into the caller's gx_chunk buffers
+static int32_t chunkToGx(const Chunk& in, gx_chunk* out, std::string* err) {
+  int n = in.numRows();
+  if (in.cols.empty() || n == 0) {  // EOF (0 rows)
+    out->n_rows = 0;
+    return GX_OK;
+  }
+  if (out->n_cols != (int32_t)in.cols.size()) {
+    *err = "output chunk column count mismatch";
+    return GX_ERR_INVALID;
+  }
+  for (size_t c = 0; c < in.cols.size(); c++) {
+    const Column& col = in.cols[c];
+    gx_col* g = &out->cols[c];
+    int nb = (n + 7) / 8;
+    if (col.isVarlen()) {
+      if (g->offsets_cap < n + 1 || g->data_cap < (int64_t)col.data.size()) {
+        *err = "output buffer too small";
+        return GX_ERR_INVALID;
+      }
+      std::memcpy(g->offsets, col.offsets.data(), (n + 1) * 8);
+      std::memcpy(g->data, col.data.data(), col.data.size());
+    } else {
+      int64_t bytes = (int64_t)n * col.elemSize();
+      if (g->data_cap < bytes) {
+        *err = "output buffer too small";
+        return GX_ERR_INVALID;
+      }
+      std::memcpy(g->data, col.data.data(), bytes);
+    }
+    if (g->null_bitmap) {
+      std::memset(g->null_bitmap, 0, nb);
+      std::memcpy(g->null_bitmap, col.nullBitmap.data(),
+                  std::min((size_t)nb, col.nullBitmap.size()));
+    }
+    g->length = n;
+    g->elem_size = col.elemSize();
+  }
+  out->n_rows = n;
+  return GX_OK;
+}
+
+int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
+  if (!ex || !ex->opened) return GX_ERR_INVALID;
+  Chunk c;
+  int32_t ec = ex->exec->next(c);
+  if (ec) {
+    ex->err = ex->exec->err;
+    *rows_out = 0;
+    return ec;
+  }
+  ec = chunkToGx(c, out, &ex->err);
+  if (ec) {
+    *rows_out = 0;
+    return ec;
+  }
+  *rows_out = c.numRows();
+  return GX_OK;
+}
+
+int32_t gx_close(gx_exec* ex) {
+  if (!ex || !ex->exec) return GX_ERR_INVALID;
+  return ex->exec->close();
+}
+
+void gx_exec_free(gx_exec* ex) { delete ex; }
+
+const char* gx_last_error(gx_exec* ex) {
+  if (!ex) return "null exec";
+  return ex->err.c_str();
+}
+
+}  // extern "C"

@@ -1,0 +1,1113 @@
+// oracle/exec.cpp — CPU restatement of the reference chunk executors.
+// ORACLE / TEST INFRASTRUCTURE ONLY.
+//
+// Follows: pkg/executor/select.go:750 (Selection), projection.go (Projection),
+// expression/chunk_executor.go:99-124 + expression.go:420-504 (vectorized
+// eval / VectorizedFilter), aggregate/agg_hash_executor.go + aggfuncs
+// (HashAgg), sortexec/topn.go + topn_chunk_heap.go (TopN), sortexec/sort.go
+// (Sort).
+#include "exec.h"
+
+#include <algorithm>
+#include <cassert>
+#include <cstring>
+#include <functional>
+
+#include "codec.h"
+#include "core_time.h"
+
+namespace oracle {
+
+namespace {
+
+// ---------- expression evaluation ----------
+// Mirrors ScalarFunction.VecEval* (builtin_compare_vec_generated.go,
+// builtin_arithmetic_vec.go): children evaluated to full columns, nulls
+// merged, per-row op.
+
+struct EvalCtx {
+  const Plan* plan;
+  std::string* err;
+};
+
+int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out);
+
+void fillConst(const Expr& e, int n, Column& out) {
+  out.type = e.retType;
+  out.frac = e.retFrac;
+  out.reset();
+  for (int i = 0; i < n; i++) {
+    switch (e.retType) {
+      case GX_TYPE_I64: out.appendI64(e.constI64); break;
+      case GX_TYPE_F64: out.appendF64(e.constF64); break;
+      case GX_TYPE_TIME: out.appendU64(e.constTime); break;
+      case GX_TYPE_DECIMAL: out.appendDecimal(e.constDec); break;
+      case GX_TYPE_STRING: out.appendBytes(e.constStr.data(), e.constStr.size()); break;
+    }
+  }
+}
+
+inline bool bothNotNull(const Column& a, const Column& b, int i) {
+  return !a.isNull(i) && !b.isNull(i);
+}
+
+int cmpString(const Column& a, const Column& b, int i) {
+  int la, lb;
+  const uint8_t* pa = a.getBytes(i, &la);
+  const uint8_t* pb = b.getBytes(i, &lb);
+  std::string ka = BinCollatorKey(pa, la);
+  std::string kb = BinCollatorKey(pb, lb);
+  int c = ka.compare(kb);
+  return c < 0 ? -1 : (c > 0 ? 1 : 0);
+}
+
+int32_t evalCompare(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
+  Column a, b;
+  int32_t err = evalVec(ctx, e.args[0], in, a);
+  if (err) return err;
+  err = evalVec(ctx, e.args[1], in, b);
+  if (err) return err;
+  int n = in.numRows();
+  out.type = GX_TYPE_I64;
+  out.reset();
+  for (int i = 0; i < n; i++) {
+    if (!bothNotNull(a, b, i)) { out.appendNull(); continue; }
+    int c = 0;
+    switch (a.type) {
+      case GX_TYPE_I64: {
+        int64_t x = a.getI64(i), y = b.getI64(i);
+        c = x < y ? -1 : (x > y ? 1 : 0);
+        break;
+      }
+      case GX_TYPE_F64: {
+        double x = a.getF64(i), y = b.getF64(i);
+        c = x < y ? -1 : (x > y ? 1 : 0);
+        break;
+      }
+      case GX_TYPE_TIME:
+        c = CompareTime(a.getU64(i), b.getU64(i));
+        break;
+      case GX_TYPE_DECIMAL:
+        c = a.getDecimal(i)->Compare(*b.getDecimal(i));
+        break;
+      case GX_TYPE_STRING:
+        c = cmpString(a, b, i);
+        break;
+    }
+    int64_t r = 0;
+    switch (e.func) {
+      case GX_F_LT: r = c < 0; break;
+      case GX_F_LE: r = c <= 0; break;
+      case GX_F_GT: r = c > 0; break;
+      case GX_F_GE: r = c >= 0; break;
+      case GX_F_EQ: r = c == 0; break;
+      case GX_F_NE: r = c != 0; break;
+    }
+    out.appendI64(r);
+  }
+  return GX_OK;
+}
+
+int32_t evalArith(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
+  Column a, b;
+  int32_t err = evalVec(ctx, e.args[0], in, a);
+  if (err) return err;
+  err = evalVec(ctx, e.args[1], in, b);
+  if (err) return err;
+  int n = in.numRows();
+  out.type = e.retType;
+  out.frac = e.retFrac;
+  out.reset();
+  if (e.retType == GX_TYPE_DECIMAL) {
+    for (int i = 0; i < n; i++) {
+      if (!bothNotNull(a, b, i)) { out.appendNull(); continue; }
+      MyDecimal to;
+      int32_t ec = GX_OK;
+      switch (e.func) {
+        case GX_F_PLUS: ec = DecimalAdd(a.getDecimal(i), b.getDecimal(i), &to); break;
+        case GX_F_MINUS: ec = DecimalSub(a.getDecimal(i), b.getDecimal(i), &to); break;
+        case GX_F_MUL: ec = DecimalMul(a.getDecimal(i), b.getDecimal(i), &to); break;
+        case GX_F_DIV: ec = DecimalDiv(a.getDecimal(i), b.getDecimal(i), &to, kDivFracIncr); break;
+      }
+      if (ec == E_DIV_ZERO) { out.appendNull(); continue; }  // div-by-0 -> NULL (MySQL)
+      if (ec != GX_OK && ec != E_TRUNCATED) {
+        *ctx.err = "decimal arithmetic error";
+        return ec;
+      }
+      out.appendDecimal(to);
+    }
+    return GX_OK;
+  }
+  if (e.retType == GX_TYPE_I64) {
+    for (int i = 0; i < n; i++) {
+      if (!bothNotNull(a, b, i)) { out.appendNull(); continue; }
+      int64_t x = a.getI64(i), y = b.getI64(i), r = 0;
+      bool ovf = false;
+      switch (e.func) {
+        case GX_F_PLUS: ovf = __builtin_add_overflow(x, y, &r); break;
+        case GX_F_MINUS: ovf = __builtin_sub_overflow(x, y, &r); break;
+        case GX_F_MUL: ovf = __builtin_mul_overflow(x, y, &r); break;
+        case GX_F_DIV:
+          if (y == 0) { out.appendNull(); continue; }
+          r = x / y;
+          break;
+      }
+      if (ovf) { *ctx.err = "BIGINT value out of range"; return E_OVERFLOW; }
+      out.appendI64(r);
+    }
+    return GX_OK;
+  }
+  if (e.retType == GX_TYPE_F64) {
+    for (int i = 0; i < n; i++) {
+      if (!bothNotNull(a, b, i)) { out.appendNull(); continue; }
+      double x = a.getF64(i), y = b.getF64(i), r = 0;
+      switch (e.func) {
+        case GX_F_PLUS: r = x + y; break;
+        case GX_F_MINUS: r = x - y; break;
+        case GX_F_MUL: r = x * y; break;
+        case GX_F_DIV:
+          if (y == 0) { out.appendNull(); continue; }
+          r = x / y;
+          break;
+      }
+      out.appendF64(r);
+    }
+    return GX_OK;
+  }
+  *ctx.err = "unsupported arith type";
+  return GX_ERR_INVALID;
+}
+
+int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
+  const Expr& e = ctx.plan->exprs[exprId];
+  switch (e.kind) {
+    case EK_COLREF:
+      out = in.cols[e.colIdx];
+      return GX_OK;
+    case EK_CONST:
+      fillConst(e, in.numRows(), out);
+      return GX_OK;
+    case EK_CALL:
+      if (e.func <= GX_F_NE) return evalCompare(ctx, e, in, out);
+      return evalArith(ctx, e, in, out);
+  }
+  return GX_ERR_INVALID;
+}
+
+// VectorizedFilter semantics (expression.go:420-504): conjuncts narrow the
+// selection progressively; NULL or 0 rejects the row.
+int32_t vectorizedFilter(EvalCtx& ctx, const std::vector<int>& conds,
+                         const Chunk& in, std::vector<uint8_t>& selected) {
+  int n = in.numRows();
+  selected.assign(n, 1);
+  for (int cond : conds) {
+    Column r;
+    int32_t err = evalVec(ctx, cond, in, r);
+    if (err) return err;
+    for (int i = 0; i < n; i++) {
+      if (!selected[i]) continue;
+      if (r.isNull(i) || r.getI64(i) == 0) selected[i] = 0;
+    }
+  }
+  return GX_OK;
+}
+
+// ---------- operators ----------
+
+class SourceExec : public Exec {
+ public:
+  SourceExec(const PlanNode& node, SourceBinding* b) : node_(node), bind_(b) {
+    outTypes = node.colTypes;
+    outFracs = node.colFracs;
+  }
+  int32_t open() override {
+    pos_ = 0;
+    return GX_OK;
+  }
+  int32_t next(Chunk& out) override {
+    out.reset();
+    if (bind_ == nullptr) return GX_OK;  // empty source
+    if (bind_->haveChunks) {
+      if (pos_ >= (int64_t)bind_->chunks.size()) return GX_OK;
+      out = bind_->chunks[(size_t)pos_++];
+      return GX_OK;
+    }
+    if (bind_->tpchTable >= 0) {
+      int64_t remaining = bind_->tpchRows - pos_;
+      if (remaining <= 0) return GX_OK;
+      int n = (int)std::min<int64_t>(remaining, kMaxChunkSize);
+      TpchGenChunk(bind_->tpchTable, bind_->tpchRowOffset + pos_, n,
+                   bind_->tpchSeed,
+                   bind_->tpchTotalRows > 0 ? bind_->tpchTotalRows : bind_->tpchRows,
+                   out);
+      pos_ += n;
+      return GX_OK;
+    }
+    return GX_OK;
+  }
+  int32_t close() override { return GX_OK; }
+
+ private:
+  const PlanNode& node_;
+  SourceBinding* bind_;
+  int64_t pos_ = 0;
+};
+
+// select.go:750-785: filter child chunks, copy surviving rows.
+class SelectionExec : public Exec {
+ public:
+  SelectionExec(const Plan& plan, const PlanNode& node, std::unique_ptr<Exec> child)
+      : plan_(plan), node_(node), child_(std::move(child)) {
+    outTypes = child_->outTypes;
+    outFracs = child_->outFracs;
+  }
+  int32_t open() override { return child_->open(); }
+  int32_t next(Chunk& out) override {
+    out.cols.resize(outTypes.size());
+    for (size_t c = 0; c < out.cols.size(); c++) {
+      out.cols[c].type = outTypes[c];
+      out.cols[c].frac = outFracs[c];
+    }
+    out.reset();
+    EvalCtx ctx{&plan_, &err};
+    for (;;) {
+      Chunk in;
+      int32_t ec = child_->next(in);
+      if (ec) { err = child_->err; return ec; }
+      int n = in.numRows();
+      if (n == 0) return GX_OK;  // EOF
+      std::vector<uint8_t> selected;
+      ec = vectorizedFilter(ctx, node_.exprs, in, selected);
+      if (ec) return ec;
+      for (int i = 0; i < n; i++) {
+        if (!selected[i]) continue;
+        for (size_t c = 0; c < out.cols.size(); c++)
+          out.cols[c].appendFrom(in.cols[c], i);
+      }
+      if (out.numRows() > 0) return GX_OK;
+    }
+  }
+  int32_t close() override { return child_->close(); }
+
+ private:
+  const Plan& plan_;
+  const PlanNode& node_;
+  std::unique_ptr<Exec> child_;
+};
+
+// projection.go + chunk_executor.go:99-124
+class ProjectionExec : public Exec {
+ public:
+  ProjectionExec(const Plan& plan, const PlanNode& node, std::unique_ptr<Exec> child)
+      : plan_(plan), node_(node), child_(std::move(child)) {
+    for (int eid : node.exprs) {
+      outTypes.push_back(plan.exprs[eid].retType);
+      outFracs.push_back(plan.exprs[eid].retFrac);
+    }
+  }
+  int32_t open() override { return child_->open(); }
+  int32_t next(Chunk& out) override {
+    out.cols.resize(node_.exprs.size());
+    Chunk in;
+    int32_t ec = child_->next(in);
+    if (ec) { err = child_->err; return ec; }
+    EvalCtx ctx{&plan_, &err};
+    if (in.numRows() == 0) {
+      for (size_t c = 0; c < out.cols.size(); c++) {
+        out.cols[c].type = outTypes[c];
+        out.cols[c].frac = outFracs[c];
+        out.cols[c].reset();
+      }
+      return GX_OK;
+    }
+    for (size_t c = 0; c < node_.exprs.size(); c++) {
+      ec = evalVec(ctx, node_.exprs[c], in, out.cols[c]);
+      if (ec) return ec;
+      out.cols[c].frac = outFracs[c];
+    }
+    return GX_OK;
+  }
+  int32_t close() override { return child_->close(); }
+
+ private:
+  const Plan& plan_;
+  const PlanNode& node_;
+  std::unique_ptr<Exec> child_;
+};
+
+// ---------- hash aggregation ----------
+// agg_hash_executor.go semantics with aggfuncs update/merge/finalize rules.
+// Complete mode == partial-then-final in one process; the canonical
+// partial-state chunk layout (for GX_AGG_MODE_PARTIAL output / FINAL input):
+//   [group cols...] then per agg:
+//     COUNT           -> I64 count
+//     SUM (decimal)   -> DECIMAL val, I64 notNullRowCount
+//     SUM (f64)       -> F64 val, I64 notNullRowCount
+//     AVG (decimal)   -> DECIMAL sum, I64 count
+//     AVG (f64)       -> F64 sum, I64 count
+//     MIN/MAX         -> value col (null = empty)
+//     FIRSTROW        -> value col
+struct AggState {
+  MyDecimal dec;       // sum/avg accumulator or min/max/firstrow decimal
+  double f64 = 0;
+  int64_t i64 = 0;     // count / notNullRowCount
+  int64_t aux = 0;     // generic value for int64 min/max/firstrow
+  uint64_t u64 = 0;    // time value
+  std::string str;
+  bool hasValue = false;
+};
+
+struct Group {
+  std::vector<AggState> states;
+  Chunk keyRow;  // 1-row chunk holding the group-by column values (firstrow)
+};
+
+class HashAggExec : public Exec {
+ public:
+  HashAggExec(const Plan& plan, const PlanNode& node, std::unique_ptr<Exec> child)
+      : plan_(plan), node_(node), child_(std::move(child)) {
+    // output schema
+    if (node.aggMode == GX_AGG_MODE_PARTIAL) {
+      appendGroupSchema();
+      for (size_t a = 0; a < node.aggFuncs.size(); a++) {
+        int f = node.aggFuncs[a];
+        int vt = argType(a);
+        switch (f) {
+          case GX_AGG_COUNT: outTypes.push_back(GX_TYPE_I64); outFracs.push_back(0); break;
+          case GX_AGG_SUM:
+          case GX_AGG_AVG:
+            outTypes.push_back(vt); outFracs.push_back(node.aggFracs[a]);
+            outTypes.push_back(GX_TYPE_I64); outFracs.push_back(0);
+            break;
+          default:
+            outTypes.push_back(vt); outFracs.push_back(node.aggFracs[a]);
+        }
+      }
+    } else {
+      appendGroupSchema();
+      for (size_t a = 0; a < node.aggFuncs.size(); a++) {
+        int f = node.aggFuncs[a];
+        switch (f) {
+          case GX_AGG_COUNT: outTypes.push_back(GX_TYPE_I64); outFracs.push_back(0); break;
+          default:
+            outTypes.push_back(argType(a));
+            outFracs.push_back(node.aggFracs[a]);
+        }
+      }
+    }
+  }
+
+  int32_t open() override {
+    done_ = false;
+    emitPos_ = 0;
+    order_.clear();
+    groups_.clear();
+    return child_->open();
+  }
+
+  int32_t next(Chunk& out) override {
+    shapeOut(out);
+    if (!done_) {
+      int32_t ec = (node_.aggMode == GX_AGG_MODE_FINAL) ? drainFinal() : drainRaw();
+      if (ec) return ec;
+      done_ = true;
+      // no-group-by + zero rows still yields one row (e.g. count(*) = 0,
+      // sum = NULL) — aggregate semantics for empty input
+      if (node_.exprs.empty() && order_.empty() && node_.aggMode != GX_AGG_MODE_PARTIAL) {
+        Group g;
+        g.states.resize(node_.aggFuncs.size());
+        groups_[""] = g;
+        order_.push_back("");
+      }
+    }
+    int emitted = 0;
+    while (emitPos_ < order_.size() && emitted < kMaxChunkSize) {
+      Group& g = groups_[order_[emitPos_]];
+      int32_t ec = emitGroup(g, out);
+      if (ec) return ec;
+      emitPos_++;
+      emitted++;
+    }
+    return GX_OK;
+  }
+
+  int32_t close() override { return child_->close(); }
+
+ private:
+  void appendGroupSchema() {
+    for (int eid : node_.exprs) {
+      outTypes.push_back(plan_.exprs[eid].retType);
+      outFracs.push_back(plan_.exprs[eid].retFrac);
+    }
+  }
+  int argType(size_t a) const {
+    int ae = node_.aggArgs[a];
+    if (ae < 0) return GX_TYPE_I64;
+    return plan_.exprs[ae].retType;
+  }
+  void shapeOut(Chunk& out) {
+    out.cols.resize(outTypes.size());
+    for (size_t c = 0; c < out.cols.size(); c++) {
+      out.cols[c].type = outTypes[c];
+      out.cols[c].frac = outFracs[c];
+    }
+    out.reset();
+  }
+
+  // GetGroupKey (agg_util.go:106-158): eval group exprs, HashGroupKey per col
+  int32_t groupKeys(EvalCtx& ctx, const Chunk& in,
+                    std::vector<std::string>& keys,
+                    std::vector<Column>& groupCols) {
+    int n = in.numRows();
+    keys.assign(n, std::string());
+    groupCols.clear();
+    for (int eid : node_.exprs) {
+      Column c;
+      int32_t ec = evalVec(ctx, eid, in, c);
+      if (ec) return ec;
+      ec = HashGroupKeyCol(c, keys);
+      if (ec) return ec;
+      groupCols.push_back(std::move(c));
+    }
+    return GX_OK;
+  }
+
+  Group& getGroup(const std::string& key, const std::vector<Column>& groupCols, int row) {
+    auto it = groups_.find(key);
+    if (it != groups_.end()) return it->second;
+    Group g;
+    g.states.resize(node_.aggFuncs.size());
+    g.keyRow.cols.resize(groupCols.size());
+    for (size_t c = 0; c < groupCols.size(); c++) {
+      g.keyRow.cols[c].type = groupCols[c].type;
+      g.keyRow.cols[c].frac = groupCols[c].frac;
+      if (g.keyRow.cols[c].isVarlen()) g.keyRow.cols[c].offsets.assign(1, 0);
+      g.keyRow.cols[c].appendFrom(groupCols[c], row);
+    }
+    order_.push_back(key);
+    return groups_.emplace(key, std::move(g)).first->second;
+  }
+
+  // UpdatePartialResult per agg func (func_sum.go:224, func_avg.go:110, ...)
+  int32_t updateState(AggState& s, int func, const Column* argCol, int row, int valueType) {
+    bool isNull = argCol ? argCol->isNull(row) : false;
+    switch (func) {
+      case GX_AGG_COUNT:
+        if (!isNull) s.i64++;
+        break;
+      case GX_AGG_SUM:
+        if (isNull) break;
+        if (valueType == GX_TYPE_DECIMAL) {
+          if (s.i64 == 0) {
+            s.dec = *argCol->getDecimal(row);
+            s.i64 = 1;
+          } else {
+            MyDecimal tmp;
+            int32_t ec = DecimalAdd(&s.dec, argCol->getDecimal(row), &tmp);
+            if (ec != E_OK && ec != E_TRUNCATED) return ec;
+            s.dec = tmp;
+            s.i64++;
+          }
+        } else if (valueType == GX_TYPE_F64) {
+          s.f64 += argCol->getF64(row);
+          s.i64++;
+        } else {
+          return GX_ERR_INVALID;  // sum over int is decimal in MySQL: builder
+                                  // must cast; reject here
+        }
+        break;
+      case GX_AGG_AVG:
+        if (isNull) break;
+        if (valueType == GX_TYPE_DECIMAL) {
+          MyDecimal tmp;
+          int32_t ec = DecimalAdd(&s.dec, argCol->getDecimal(row), &tmp);
+          if (ec != E_OK && ec != E_TRUNCATED) return ec;
+          s.dec = tmp;
+          s.i64++;
+        } else if (valueType == GX_TYPE_F64) {
+          s.f64 += argCol->getF64(row);
+          s.i64++;
+        } else {
+          return GX_ERR_INVALID;
+        }
+        break;
+      case GX_AGG_MIN:
+      case GX_AGG_MAX: {
+        if (isNull) break;
+        bool greater = func == GX_AGG_MAX;
+        if (!s.hasValue) {
+          storeValue(s, *argCol, row, valueType);
+          s.hasValue = true;
+          break;
+        }
+        int c = cmpValue(s, *argCol, row, valueType);
+        if ((greater && c < 0) || (!greater && c > 0)) storeValue(s, *argCol, row, valueType);
+        break;
+      }
+      case GX_AGG_FIRSTROW:
+        if (!s.hasValue) {
+          s.hasValue = true;
+          s.aux = isNull ? 1 : 0;  // aux=1 => the first row was NULL
+          if (!isNull) storeValue(s, *argCol, row, valueType);
+        }
+        break;
+    }
+    return GX_OK;
+  }
+
+  void storeValue(AggState& s, const Column& col, int row, int vt) {
+    switch (vt) {
+      case GX_TYPE_I64: s.aux = col.getI64(row); s.i64 = col.getI64(row); break;
+      case GX_TYPE_F64: s.f64 = col.getF64(row); break;
+      case GX_TYPE_TIME: s.u64 = col.getU64(row); break;
+      case GX_TYPE_DECIMAL: s.dec = *col.getDecimal(row); break;
+      case GX_TYPE_STRING: s.str = col.getStr(row); break;
+    }
+  }
+  int cmpValue(const AggState& s, const Column& col, int row, int vt) {
+    switch (vt) {
+      case GX_TYPE_I64: {
+        int64_t x = s.i64, y = col.getI64(row);
+        return x < y ? -1 : (x > y ? 1 : 0);
+      }
+      case GX_TYPE_F64: {
+        double x = s.f64, y = col.getF64(row);
+        return x < y ? -1 : (x > y ? 1 : 0);
+      }
+      case GX_TYPE_TIME: return CompareTime(s.u64, col.getU64(row));
+      case GX_TYPE_DECIMAL: return s.dec.Compare(*col.getDecimal(row));
+      case GX_TYPE_STRING: {
+        int n;
+        const uint8_t* p = col.getBytes(row, &n);
+        std::string k = BinCollatorKey(p, n);
+        std::string mk = BinCollatorKey((const uint8_t*)s.str.data(), (int)s.str.size());
+        int c = mk.compare(k);
+        return c < 0 ? -1 : (c > 0 ? 1 : 0);
+      }
+    }
+    return 0;
+  }
+
+  int32_t drainRaw() {
+    EvalCtx ctx{&plan_, &err};
+    for (;;) {
+      Chunk in;
+      int32_t ec = child_->next(in);
+      if (ec) { err = child_->err; return ec; }
+      int n = in.numRows();
+      if (n == 0) return GX_OK;
+      std::vector<std::string> keys;
+      std::vector<Column> groupCols;
+      ec = groupKeys(ctx, in, keys, groupCols);
+      if (ec) return ec;
+      // eval agg arg exprs once per chunk
+      std::vector<Column> argCols(node_.aggFuncs.size());
+      std::vector<const Column*> argPtr(node_.aggFuncs.size(), nullptr);
+      for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
+        if (node_.aggArgs[a] >= 0) {
+          ec = evalVec(ctx, node_.aggArgs[a], in, argCols[a]);
+          if (ec) return ec;
+          argPtr[a] = &argCols[a];
+        }
+      }
+      for (int i = 0; i < n; i++) {
+        Group& g = getGroup(keys[i], groupCols, i);
+        for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
+          ec = updateState(g.states[a], node_.aggFuncs[a], argPtr[a], i, argType(a));
+          if (ec) return ec;
+        }
+      }
+    }
+  }
+
+  // FINAL mode: child emits canonical partial-state chunks
+  int32_t drainFinal() {
+    EvalCtx ctx{&plan_, &err};
+    size_t nGroupCols = node_.exprs.size();
+    for (;;) {
+      Chunk in;
+      int32_t ec = child_->next(in);
+      if (ec) { err = child_->err; return ec; }
+      int n = in.numRows();
+      if (n == 0) return GX_OK;
+      // group cols are the first columns of the partial chunk
+      std::vector<std::string> keys(n);
+      std::vector<Column> groupCols;
+      for (size_t c = 0; c < nGroupCols; c++) {
+        ec = HashGroupKeyCol(in.cols[c], keys);
+        if (ec) return ec;
+        groupCols.push_back(in.cols[c]);
+      }
+      for (int i = 0; i < n; i++) {
+        Group& g = getGroup(keys[i], groupCols, i);
+        size_t col = nGroupCols;
+        for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
+          AggState& s = g.states[a];
+          int f = node_.aggFuncs[a];
+          int vt = argType(a);
+          switch (f) {
+            case GX_AGG_COUNT:
+              s.i64 += in.cols[col].getI64(i);
+              col += 1;
+              break;
+            case GX_AGG_SUM:
+            case GX_AGG_AVG: {
+              int64_t srcCount = in.cols[col + 1].getI64(i);
+              if (srcCount > 0) {
+                if (vt == GX_TYPE_DECIMAL) {
+                  MyDecimal tmp;
+                  int32_t e2 = DecimalAdd(&s.dec, in.cols[col].getDecimal(i), &tmp);
+                  if (e2 != E_OK && e2 != E_TRUNCATED) return e2;
+                  s.dec = tmp;
+                } else {
+                  s.f64 += in.cols[col].getF64(i);
+                }
+                s.i64 += srcCount;
+              }
+              col += 2;
+              break;
+            }
+            case GX_AGG_MIN:
+            case GX_AGG_MAX: {
+              if (!in.cols[col].isNull(i)) {
+                bool greater = f == GX_AGG_MAX;
+                if (!s.hasValue) {
+                  storeValue(s, in.cols[col], i, vt);
+                  s.hasValue = true;
+                } else {
+                  int c2 = cmpValue(s, in.cols[col], i, vt);
+                  if ((greater && c2 < 0) || (!greater && c2 > 0))
+                    storeValue(s, in.cols[col], i, vt);
+                }
+              }
+              col += 1;
+              break;
+            }
+            case GX_AGG_FIRSTROW:
+              if (!s.hasValue) {
+                s.hasValue = true;
+                s.aux = in.cols[col].isNull(i) ? 1 : 0;
+                if (!in.cols[col].isNull(i)) storeValue(s, in.cols[col], i, vt);
+              }
+              col += 1;
+              break;
+          }
+        }
+      }
+    }
+  }
+
+  void appendValue(const AggState& s, int vt, Column& out) {
+    switch (vt) {
+      case GX_TYPE_I64: out.appendI64(s.aux); break;
+      case GX_TYPE_F64: out.appendF64(s.f64); break;
+      case GX_TYPE_TIME: out.appendU64(s.u64); break;
+      case GX_TYPE_DECIMAL: out.appendDecimal(s.dec); break;
+      case GX_TYPE_STRING: out.appendBytes(s.str.data(), s.str.size()); break;
+    }
+  }
+
+  int32_t emitGroup(Group& g, Chunk& out) {
+    size_t col = 0;
+    for (size_t c = 0; c < node_.exprs.size(); c++, col++)
+      out.cols[col].appendFrom(g.keyRow.cols[c], 0);
+    bool partial = node_.aggMode == GX_AGG_MODE_PARTIAL;
+    for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
+      AggState& s = g.states[a];
+      int f = node_.aggFuncs[a];
+      int vt = argType(a);
+      switch (f) {
+        case GX_AGG_COUNT:
+          out.cols[col++].appendI64(s.i64);
+          break;
+        case GX_AGG_SUM:
+          if (partial) {
+            if (s.i64 == 0) out.cols[col++].appendNull();
+            else if (vt == GX_TYPE_DECIMAL) out.cols[col++].appendDecimal(s.dec);
+            else out.cols[col++].appendF64(s.f64);
+            out.cols[col++].appendI64(s.i64);
+          } else {
+            if (s.i64 == 0) { out.cols[col++].appendNull(); break; }
+            if (vt == GX_TYPE_DECIMAL) {
+              // sum finalize rounds to ret frac (func_sum.go:203-222)
+              MyDecimal v = s.dec;
+              int32_t ec = v.Round(&v, node_.aggFracs[a], ModeHalfUp);
+              if (ec != E_OK && ec != E_TRUNCATED) return ec;
+              out.cols[col++].appendDecimal(v);
+            } else {
+              out.cols[col++].appendF64(s.f64);
+            }
+          }
+          break;
+        case GX_AGG_AVG:
+          if (partial) {
+            if (vt == GX_TYPE_DECIMAL) out.cols[col++].appendDecimal(s.dec);
+            else out.cols[col++].appendF64(s.f64);
+            out.cols[col++].appendI64(s.i64);
+          } else {
+            if (s.i64 == 0) { out.cols[col++].appendNull(); break; }
+            if (vt == GX_TYPE_DECIMAL) {
+              // func_avg.go:84-109: div by count (+DivPrecisionIncrement),
+              // round HalfUp to ret frac
+              MyDecimal cnt, res;
+              cnt.FromInt(s.i64);
+              int32_t ec = DecimalDiv(&s.dec, &cnt, &res, kDivFracIncr);
+              if (ec != E_OK && ec != E_TRUNCATED) return ec;
+              ec = res.Round(&res, node_.aggFracs[a], ModeHalfUp);
+              if (ec != E_OK && ec != E_TRUNCATED) return ec;
+              out.cols[col++].appendDecimal(res);
+            } else {
+              out.cols[col++].appendF64(s.f64 / (double)s.i64);
+            }
+          }
+          break;
+        case GX_AGG_MIN:
+        case GX_AGG_MAX:
+          if (!s.hasValue) out.cols[col++].appendNull();
+          else { appendValue(s, vt, out.cols[col]); col++; }
+          break;
+        case GX_AGG_FIRSTROW:
+          if (!s.hasValue || s.aux == 1) out.cols[col++].appendNull();
+          else { appendValue(s, vt, out.cols[col]); col++; }
+          break;
+      }
+    }
+    return GX_OK;
+  }
+
+  const Plan& plan_;
+  const PlanNode& node_;
+  std::unique_ptr<Exec> child_;
+  bool done_ = false;
+  std::unordered_map<std::string, Group> groups_;
+  std::vector<std::string> order_;  // deterministic emit order (insertion)
+  size_t emitPos_ = 0;
+};
+
+// ---------- TopN / Sort ----------
+// sortexec/topn.go:61-98 + topn_chunk_heap.go: keep limit+offset smallest rows
+// by the order keys; emit sorted, skipping offset.
+
+struct RowRef {
+  int chunkIdx;
+  int rowIdx;
+};
+
+class TopNExec : public Exec {
+ public:
+  TopNExec(const Plan& plan, const PlanNode& node, std::unique_ptr<Exec> child,
+           bool isSort)
+      : plan_(plan), node_(node), child_(std::move(child)), isSort_(isSort) {
+    outTypes = child_->outTypes;
+    outFracs = child_->outFracs;
+  }
+  int32_t open() override {
+    done_ = false;
+    emit_ = 0;
+    rows_.clear();
+    data_.clear();
+    keyCols_.clear();
+    return child_->open();
+  }
+  int32_t next(Chunk& out) override {
+    out.cols.resize(outTypes.size());
+    for (size_t c = 0; c < out.cols.size(); c++) {
+      out.cols[c].type = outTypes[c];
+      out.cols[c].frac = outFracs[c];
+    }
+    out.reset();
+    if (!done_) {
+      int32_t ec = drain();
+      if (ec) return ec;
+      done_ = true;
+      emit_ = std::min((size_t)node_.offset, rows_.size());
+    }
+    size_t end = isSort_ ? rows_.size()
+                         : std::min(rows_.size(), (size_t)(node_.offset + node_.limit));
+    while (emit_ < end && out.numRows() < kMaxChunkSize) {
+      const RowRef& r = rows_[emit_];
+      for (size_t c = 0; c < out.cols.size(); c++)
+        out.cols[c].appendFrom(data_[r.chunkIdx].cols[c], r.rowIdx);
+      emit_++;
+    }
+    return GX_OK;
+  }
+  int32_t close() override { return child_->close(); }
+
+ private:
+  int cmpRows(const RowRef& a, const RowRef& b) const {
+    for (size_t k = 0; k < node_.exprs.size(); k++) {
+      const Column& ca = keyCols_[a.chunkIdx][k];
+      const Column& cb = keyCols_[b.chunkIdx][k];
+      bool na = ca.isNull(a.rowIdx), nb = cb.isNull(b.rowIdx);
+      int c;
+      if (na && nb) c = 0;
+      else if (na) c = -1;  // NULL first ascending
+      else if (nb) c = 1;
+      else {
+        switch (ca.type) {
+          case GX_TYPE_I64: {
+            int64_t x = ca.getI64(a.rowIdx), y = cb.getI64(b.rowIdx);
+            c = x < y ? -1 : (x > y ? 1 : 0);
+            break;
+          }
+          case GX_TYPE_F64: {
+            double x = ca.getF64(a.rowIdx), y = cb.getF64(b.rowIdx);
+            c = x < y ? -1 : (x > y ? 1 : 0);
+            break;
+          }
+          case GX_TYPE_TIME: c = CompareTime(ca.getU64(a.rowIdx), cb.getU64(b.rowIdx)); break;
+          case GX_TYPE_DECIMAL: c = ca.getDecimal(a.rowIdx)->Compare(*cb.getDecimal(b.rowIdx)); break;
+          default: {
+            int la, lb;
+            const uint8_t* pa = ca.getBytes(a.rowIdx, &la);
+            const uint8_t* pb = cb.getBytes(b.rowIdx, &lb);
+            std::string ka = BinCollatorKey(pa, la), kb = BinCollatorKey(pb, lb);
+            int cc = ka.compare(kb);
+            c = cc < 0 ? -1 : (cc > 0 ? 1 : 0);
+          }
+        }
+      }
+      if (node_.keyDesc[k]) c = -c;
+      if (c != 0) return c;
+    }
+    return 0;
+  }
+
+  int32_t drain() {
+    EvalCtx ctx{&plan_, &err};
+    for (;;) {
+      Chunk in;
+      int32_t ec = child_->next(in);
+      if (ec) { err = child_->err; return ec; }
+      int n = in.numRows();
+      if (n == 0) break;
+      int ci = (int)data_.size();
+      std::vector<Column> kc;
+      for (int eid : node_.exprs) {
+        Column c;
+        ec = evalVec(ctx, eid, in, c);
+        if (ec) return ec;
+        kc.push_back(std::move(c));
+      }
+      keyCols_.push_back(std::move(kc));
+      data_.push_back(std::move(in));
+      for (int i = 0; i < n; i++) rows_.push_back({ci, i});
+    }
+    std::stable_sort(rows_.begin(), rows_.end(),
+                     [this](const RowRef& a, const RowRef& b) { return cmpRows(a, b) < 0; });
+    return GX_OK;
+  }
+
+  const Plan& plan_;
+  const PlanNode& node_;
+  std::unique_ptr<Exec> child_;
+  bool isSort_;
+  bool done_ = false;
+  std::vector<Chunk> data_;
+  std::vector<std::vector<Column>> keyCols_;
+  std::vector<RowRef> rows_;
+  size_t emit_ = 0;
+};
+
+// ---------- inner hash join ----------
+// HashJoinV2 semantics (join/hash_join_v2.go, inner_join_probe.go:27-86):
+// output columns = build side cols then probe side cols, one output row per
+// matching (build,probe) pair; NULL join keys never match.
+class HashJoinExec : public Exec {
+ public:
+  HashJoinExec(const Plan& plan, const PlanNode& node, std::unique_ptr<Exec> build,
+               std::unique_ptr<Exec> probe)
+      : plan_(plan), node_(node), build_(std::move(build)), probe_(std::move(probe)) {
+    for (size_t i = 0; i < build_->outTypes.size(); i++) {
+      outTypes.push_back(build_->outTypes[i]);
+      outFracs.push_back(build_->outFracs[i]);
+    }
+    for (size_t i = 0; i < probe_->outTypes.size(); i++) {
+      outTypes.push_back(probe_->outTypes[i]);
+      outFracs.push_back(probe_->outFracs[i]);
+    }
+  }
+  int32_t open() override {
+    built_ = false;
+    int32_t ec = build_->open();
+    if (ec) return ec;
+    return probe_->open();
+  }
+  int32_t next(Chunk& out) override {
+    out.cols.resize(outTypes.size());
+    for (size_t c = 0; c < out.cols.size(); c++) {
+      out.cols[c].type = outTypes[c];
+      out.cols[c].frac = outFracs[c];
+    }
+    out.reset();
+    EvalCtx ctx{&plan_, &err};
+    if (!built_) {
+      int32_t ec = buildTable(ctx);
+      if (ec) return ec;
+      built_ = true;
+    }
+    size_t nb = build_->outTypes.size();
+    for (;;) {
+      Chunk in;
+      int32_t ec = probe_->next(in);
+      if (ec) { err = probe_->err; return ec; }
+      int n = in.numRows();
+      if (n == 0) return GX_OK;
+      std::vector<std::string> keys(n);
+      std::vector<uint8_t> hasNullKey(n, 0);
+      ec = serializeJoinKeys(ctx, node_.probeKeys, in, keys, hasNullKey);
+      if (ec) return ec;
+      for (int i = 0; i < n; i++) {
+        if (hasNullKey[i]) continue;
+        auto it = table_.find(keys[i]);
+        if (it == table_.end()) continue;
+        for (const RowRef& br : it->second) {
+          for (size_t c = 0; c < nb; c++)
+            out.cols[c].appendFrom(buildData_[br.chunkIdx].cols[c], br.rowIdx);
+          for (size_t c = 0; c < probe_->outTypes.size(); c++)
+            out.cols[nb + c].appendFrom(in.cols[c], i);
+        }
+      }
+      if (out.numRows() > 0) return GX_OK;
+    }
+  }
+  int32_t close() override {
+    build_->close();
+    return probe_->close();
+  }
+
+ private:
+  int32_t serializeJoinKeys(EvalCtx& ctx, const std::vector<int>& keyExprs,
+                            const Chunk& in, std::vector<std::string>& keys,
+                            std::vector<uint8_t>& hasNull) {
+    for (int eid : keyExprs) {
+      Column c;
+      int32_t ec = evalVec(ctx, eid, in, c);
+      if (ec) return ec;
+      for (int i = 0; i < (int)keys.size(); i++) {
+        if (c.isNull(i)) { hasNull[i] = 1; continue; }
+        switch (c.type) {
+          case GX_TYPE_I64: {
+            int64_t v = c.getI64(i);
+            keys[i].append((const char*)&v, 8);
+            break;
+          }
+          case GX_TYPE_TIME: {
+            uint64_t v = c.getU64(i);
+            keys[i].append((const char*)&v, 8);
+            break;
+          }
+          case GX_TYPE_F64: {
+            double v = c.getF64(i);
+            keys[i].append((const char*)&v, 8);
+            break;
+          }
+          case GX_TYPE_DECIMAL: {
+            // join keys use the value-normalized hash key (ToHashKey) so that
+            // 1.10 == 1.1 matches (codec.go:852-910 SerializeKeys semantics)
+            uint8_t buf[48];
+            int blen = 0;
+            int32_t e2 = c.getDecimal(i)->ToHashKey(buf, &blen);
+            if (e2 != E_OK) return e2;
+            keys[i].append((const char*)buf, blen);
+            break;
+          }
+          case GX_TYPE_STRING: {
+            int len;
+            const uint8_t* p = c.getBytes(i, &len);
+            std::string k = BinCollatorKey(p, len);
+            uint32_t l = (uint32_t)k.size();
+            keys[i].append((const char*)&l, 4);
+            keys[i].append(k);
+            break;
+          }
+        }
+      }
+    }
+    return GX_OK;
+  }
+
+  int32_t buildTable(EvalCtx& ctx) {
+    for (;;) {
+      Chunk in;
+      int32_t ec = build_->next(in);
+      if (ec) { err = build_->err; return ec; }
+      int n = in.numRows();
+      if (n == 0) return GX_OK;
+      int ci = (int)buildData_.size();
+      std::vector<std::string> keys(n);
+      std::vector<uint8_t> hasNull(n, 0);
+      ec = serializeJoinKeys(ctx, node_.buildKeys, in, keys, hasNull);
+      if (ec) return ec;
+      for (int i = 0; i < n; i++) {
+        if (hasNull[i]) continue;
+        table_[keys[i]].push_back({ci, i});
+      }
+      buildData_.push_back(std::move(in));
+    }
+  }
+
+  const Plan& plan_;
+  const PlanNode& node_;
+  std::unique_ptr<Exec> build_, probe_;
+  bool built_ = false;
+  std::vector<Chunk> buildData_;
+  std::unordered_map<std::string, std::vector<RowRef>> table_;
+};
+
+}  // namespace
+
+std::unique_ptr<Exec> BuildExec(const Plan& plan, int root,
+                                std::map<int, SourceBinding>* bindings,
+                                std::string* err) {
+  if (root < 0 || root >= (int)plan.nodes.size()) {
+    *err = "bad plan root";
+    return nullptr;
+  }
+  const PlanNode& node = plan.nodes[root];
+  switch (node.kind) {
+    case PK_SOURCE: {
+      SourceBinding* b = nullptr;
+      auto it = bindings->find(root);
+      if (it != bindings->end()) b = &it->second;
+      return std::make_unique<SourceExec>(node, b);
+    }
+    case PK_SELECTION: {
+      auto child = BuildExec(plan, node.child, bindings, err);
+      if (!child) return nullptr;
+      return std::make_unique<SelectionExec>(plan, node, std::move(child));
+    }
+    case PK_PROJECTION: {
+      auto child = BuildExec(plan, node.child, bindings, err);
+      if (!child) return nullptr;
+      return std::make_unique<ProjectionExec>(plan, node, std::move(child));
+    }
+    case PK_HASHAGG: {
+      auto child = BuildExec(plan, node.child, bindings, err);
+      if (!child) return nullptr;
+      return std::make_unique<HashAggExec>(plan, node, std::move(child));
+    }
+    case PK_TOPN: {
+      auto child = BuildExec(plan, node.child, bindings, err);
+      if (!child) return nullptr;
+      return std::make_unique<TopNExec>(plan, node, std::move(child), false);
+    }
+    case PK_SORT: {
+      auto child = BuildExec(plan, node.child, bindings, err);
+      if (!child) return nullptr;
+      return std::make_unique<TopNExec>(plan, node, std::move(child), true);
+    }
+    case PK_HASHJOIN: {
+      auto build = BuildExec(plan, node.child, bindings, err);
+      if (!build) return nullptr;
+      auto probe = BuildExec(plan, node.child2, bindings, err);
+      if (!probe) return nullptr;
+      return std::make_unique<HashJoinExec>(plan, node, std::move(build), std::move(probe));
+    }
+  }
+  *err = "unknown plan kind";
+  return nullptr;
+}
+
+}  // namespace oracle

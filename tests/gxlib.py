@@ -100,6 +100,13 @@ def _decl(lib):
                                    ctypes.POINTER(GxChunk), ctypes.c_int32]
     lib.gx_bind_tpch.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
                                  ctypes.c_int64, ctypes.c_uint64, ctypes.c_int64]
+    lib.gx_bind_tpch_sharded.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                         ctypes.c_int32, ctypes.c_int64,
+                                         ctypes.c_uint64, ctypes.c_int64,
+                                         ctypes.c_int64]
+    lib.gx_pb_const_dec.argtypes = [ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8)]
+    lib.gx_dec_shift.argtypes = [ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32,
+                                 ctypes.POINTER(ctypes.c_uint8)]
     lib.gx_open.argtypes = [ctypes.c_void_p]
     lib.gx_next.argtypes = [ctypes.c_void_p, ctypes.POINTER(GxChunk), i32p]
     lib.gx_close.argtypes = [ctypes.c_void_p]

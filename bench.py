@@ -1,0 +1,208 @@
+#!/usr/bin/env python3
+"""bench.py — TPC-H Q1 on the MI355X fused executor (BASELINE.json metric).
+
+One step = one full Q1 pass (scan+filter+project+hash-agg) over a synthetic
+SF10-class lineitem shard resident in HBM (59,986,052 rows per GPU by
+default; data generated on-device once, outside the timed region).
+
+N>1 (launched by torch.distributed.run, one rank per GPU over RCCL): weak
+scaling — each rank owns a row-range shard of an N x SF10 table (the region-
+shard analog, store/copr/coprocessor.go:525); per step each rank runs the
+fused kernel in PARTIAL mode and the tiny canonical partial states (~6 groups
+x 8 aggs) are all_gather'ed and merged with MergePartialResult semantics
+(aggfuncs.go:250-255) — payload is KBs, latency-bound.
+
+Prints ONE JSON line from rank 0 (driver contract).
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+SF10_ROWS = 59_986_052
+# algorithmic bytes per row, generated reference layout:
+# shipdate 8 + 4 decimals x 40 + 2 char(1) cols x (8 offsets + 1 data);
+# synthetic tables carry no null bitmaps (no NULLs) — stated, not 187.
+BYTES_PER_ROW = 8 + 4 * 40 + 2 * 9
+HBM_PEAK_GBS = 8000.0  # spec peak (MI355X_MICROARCH.md; measured ceiling ~6290)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def run_cpu_baseline(rows=8_000_000):
+    """Oracle (CPU restatement) Q1 over pre-generated host-resident chunks,
+    single thread — the reported baseline, not the target."""
+    from tests.gxlib import load_oracle
+    lib = load_oracle()
+    lib.gx_oracle_bench_q1.argtypes = [
+        ctypes.c_int64, ctypes.c_uint64, ctypes.POINTER(ctypes.c_double),
+        ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_int64)]
+    g = ctypes.c_double()
+    e = ctypes.c_double()
+    n = ctypes.c_int64()
+    rc = lib.gx_oracle_bench_q1(rows, 42, ctypes.byref(g), ctypes.byref(e),
+                                ctypes.byref(n))
+    if rc != 0:
+        return None
+    return {
+        "value": rows / (e.value / 1000.0),
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"Q1 over {rows} pre-generated host-resident rows, "
+                  f"executor only, single thread ({e.value/1000:.1f}s)",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--rows", type=int, default=SF10_ROWS,
+                    help="rows per GPU (default SF10)")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = world if world > 1 else args.gpus
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        tdist.init_process_group("nccl")  # = RCCL on ROCm
+        dist = tdist
+
+    from tests.gxlib import (GX_AGG_MODE_COMPLETE, GX_AGG_MODE_PARTIAL,
+                             GX_TPCH_LINEITEM, load_product)
+    from tidb_amd import plan as P
+    from tests.test_dist_merge import merge_partials
+
+    lib = load_product()
+    mode = GX_AGG_MODE_PARTIAL if world > 1 else GX_AGG_MODE_COMPLETE
+    b, src, agg, out_types, out_fracs = P.q1_plan(lib, mode)
+    ex = b.build(agg, device=local_rank)
+    total_rows = args.rows * n_gpus
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, args.rows, seed=42,
+                 row_offset=rank * args.rows, total_rows=total_rows)
+    caps = [2048 if t == 4 else None for t in out_types]
+
+    lib.gx_last_kernel_ms.restype = ctypes.c_double
+    lib.gx_last_kernel_ms.argtypes = [ctypes.c_void_p]
+    lib.gx_last_sel_count.restype = ctypes.c_int64
+    lib.gx_last_sel_count.argtypes = [ctypes.c_void_p]
+
+    def step():
+        ex.open()
+        rows = ex.pull_all(out_types, out_fracs, data_caps=caps)
+        ex.close()
+        if world > 1:
+            gathered = [None] * world
+            dist.all_gather_object(gathered, rows)
+            all_rows = [r for part in gathered for r in part]
+            result = merge_partials(lib, all_rows)
+        else:
+            result = {(r[0], r[1]): tuple(r[2:]) for r in rows}
+        return result, lib.gx_last_kernel_ms(ex.ex)
+
+    def sync():
+        try:
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+        except Exception:
+            pass
+        if world > 1:
+            dist.barrier()
+
+    t_gen0 = time.perf_counter()
+    for i in range(args.warmup):
+        result, _ = step()
+        if i == 0:
+            log(f"[bench] first step (incl. on-device generation of "
+                f"{args.rows} rows): {time.perf_counter()-t_gen0:.1f}s, "
+                f"{len(result)} groups")
+    sync()
+    t0 = time.perf_counter()
+    kernel_ms = []
+    result = None
+    for _ in range(args.steps):
+        result, kms = step()
+        kernel_ms.append(kms)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if world > 1:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank != 0:
+        return
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = (args.rows * n_gpus) * args.steps / elapsed
+    gbs_scanned = value * BYTES_PER_ROW / 1e9
+
+    avg_kms = sum(kernel_ms) / len(kernel_ms) if kernel_ms else 0
+    roofline = None
+    if avg_kms > 0:
+        achieved_gbs = args.rows * BYTES_PER_ROW / (avg_kms / 1000.0) / 1e9
+        roofline = {
+            "bound": "hbm",
+            "achieved": achieved_gbs,
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": achieved_gbs / HBM_PEAK_GBS,
+            # PMC traffic comes from separate rocprofv3 --pmc runs
+            # (profiles/); not measured inline.
+            "traffic": None,
+        }
+
+    cpu_baseline = None
+    if n_gpus == 1 and not args.no_cpu_baseline:
+        log("[bench] timing CPU baseline (oracle, single thread)...")
+        cpu_baseline = run_cpu_baseline()
+
+    out = {
+        "metric": "tpch_q1_rows_per_sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int128",
+        "data": "synthetic",
+        "config": {
+            "workload": "tpch_q1_sf10_synthetic_lineitem",
+            "rows_per_gpu": args.rows,
+            "parallelism": f"shard-dp{n_gpus}+partial-merge",
+            "bytes_per_row": BYTES_PER_ROW,
+        },
+        "gb_per_sec_scanned": gbs_scanned,
+        "kernel_ms_avg": avg_kms,
+        "roofline": roofline,
+        "cpu_baseline": cpu_baseline,
+        "groups": len(result) if result else 0,
+    }
+    print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

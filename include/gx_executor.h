@@ -168,6 +168,11 @@ const char* gx_last_error(gx_exec* ex);
 /* engine info: returns 1 if this library executes on GPU, 0 if CPU oracle */
 int32_t gx_engine_is_gpu(void);
 const char* gx_engine_name(void);
+/* duration (ms) of the last fused compute kernel on this executor, measured
+ * with HIP events on the launch stream (0 if none ran / CPU engine), and the
+ * row count that passed the filter in the last run. */
+double  gx_last_kernel_ms(gx_exec* ex);
+int64_t gx_last_sel_count(gx_exec* ex);
 
 /* ---- decimal helpers (each library's own implementation; used by golden
  * vector tests against pkg/types/mydecimal_test.go answers) ---- */

@@ -7,6 +7,7 @@
 
 #include "../include/gx_executor.h"
 #include "exec.h"
+#include "core_time.h"
 
 using namespace oracle;
 
@@ -303,3 +304,97 @@ const char* gx_last_error(gx_exec* ex) {
 }
 
 }  // extern "C"
+
+// ---- CPU-baseline timing entry (bench.py cpu_baseline leg) ----
+// Generates `rows` lineitem rows into host-resident chunks, then times ONE
+// Q1 pass of the oracle executor over the bound chunks (single thread).
+#include <chrono>
+extern "C" int32_t gx_oracle_bench_q1(int64_t rows, uint64_t seed,
+                                      double* gen_ms, double* exec_ms,
+                                      int64_t* groups_out) {
+  using clk = std::chrono::steady_clock;
+  auto t0 = clk::now();
+  SourceBinding bind;
+  bind.haveChunks = true;
+  int64_t pos = 0;
+  while (pos < rows) {
+    int n = (int)std::min<int64_t>(rows - pos, kMaxChunkSize);
+    Chunk c;
+    TpchGenChunk(GX_TPCH_LINEITEM, pos, n, seed, rows, c);
+    bind.chunks.push_back(std::move(c));
+    pos += n;
+  }
+  auto t1 = clk::now();
+  // build the Q1 plan directly (mirrors tidb_amd/plan.py q1_plan)
+  Plan plan;
+  auto expr = [&](Expr e) { plan.exprs.push_back(e); return (int)plan.exprs.size() - 1; };
+  auto node = [&](PlanNode n) { plan.nodes.push_back(n); return (int)plan.nodes.size() - 1; };
+  PlanNode src;
+  src.kind = PK_SOURCE;
+  TpchSchema(GX_TPCH_LINEITEM, &src.colTypes, &src.colFracs);
+  int nsrc = node(src);
+  Expr shipdate; shipdate.kind = EK_COLREF; shipdate.colIdx = 7; shipdate.retType = GX_TYPE_TIME;
+  Expr cutoff; cutoff.kind = EK_CONST; cutoff.retType = GX_TYPE_TIME;
+  cutoff.constTime = TimeFromDate(1998, 9, 1);
+  Expr lt; lt.kind = EK_CALL; lt.func = GX_F_LT; lt.retType = GX_TYPE_I64;
+  lt.args = {expr(shipdate), expr(cutoff)};
+  PlanNode sel; sel.kind = PK_SELECTION; sel.child = nsrc; sel.exprs = {expr(lt)};
+  int nsel = node(sel);
+  auto colref = [&](int idx, int t, int f) {
+    Expr e; e.kind = EK_COLREF; e.colIdx = idx; e.retType = t; e.retFrac = f;
+    return expr(e);
+  };
+  MyDecimal one; one.FromString("1", 1);
+  Expr onee; onee.kind = EK_CONST; onee.retType = GX_TYPE_DECIMAL; onee.constDec = one;
+  int eOne = expr(onee);
+  int eQty = colref(1, GX_TYPE_DECIMAL, 2), ePrice = colref(2, GX_TYPE_DECIMAL, 2);
+  int eDisc = colref(3, GX_TYPE_DECIMAL, 2), eTax = colref(4, GX_TYPE_DECIMAL, 2);
+  int eRf = colref(5, GX_TYPE_STRING, 0), eLs = colref(6, GX_TYPE_STRING, 0);
+  Expr sub; sub.kind = EK_CALL; sub.func = GX_F_MINUS; sub.retType = GX_TYPE_DECIMAL;
+  sub.retFrac = 2; sub.args = {eOne, eDisc};
+  int eSub = expr(sub);
+  Expr mul1; mul1.kind = EK_CALL; mul1.func = GX_F_MUL; mul1.retType = GX_TYPE_DECIMAL;
+  mul1.retFrac = 4; mul1.args = {ePrice, eSub};
+  int eDp = expr(mul1);
+  Expr add; add.kind = EK_CALL; add.func = GX_F_PLUS; add.retType = GX_TYPE_DECIMAL;
+  add.retFrac = 2; add.args = {eOne, eTax};
+  int eAdd = expr(add);
+  Expr mul2; mul2.kind = EK_CALL; mul2.func = GX_F_MUL; mul2.retType = GX_TYPE_DECIMAL;
+  mul2.retFrac = 6; mul2.args = {eDp, eAdd};
+  int eCh = expr(mul2);
+  PlanNode proj; proj.kind = PK_PROJECTION; proj.child = nsel;
+  proj.exprs = {eRf, eLs, eQty, ePrice, eDisc, eDp, eCh};
+  int nproj = node(proj);
+  PlanNode agg; agg.kind = PK_HASHAGG; agg.child = nproj;
+  agg.aggMode = GX_AGG_MODE_COMPLETE;
+  agg.exprs = {colref(0, GX_TYPE_STRING, 0), colref(1, GX_TYPE_STRING, 0)};
+  int aQty = colref(2, GX_TYPE_DECIMAL, 2), aPrice = colref(3, GX_TYPE_DECIMAL, 2);
+  int aDisc = colref(4, GX_TYPE_DECIMAL, 2), aDp = colref(5, GX_TYPE_DECIMAL, 4);
+  int aCh = colref(6, GX_TYPE_DECIMAL, 6);
+  agg.aggFuncs = {GX_AGG_SUM, GX_AGG_SUM, GX_AGG_SUM, GX_AGG_SUM,
+                  GX_AGG_AVG, GX_AGG_AVG, GX_AGG_AVG, GX_AGG_COUNT};
+  agg.aggArgs = {aQty, aPrice, aDp, aCh, aQty, aPrice, aDisc, -1};
+  agg.aggFracs = {2, 2, 4, 6, 6, 6, 6, 0};
+  int nagg = node(agg);
+
+  std::map<int, SourceBinding> bindings;
+  bindings[nsrc] = std::move(bind);
+  std::string err;
+  auto execp = BuildExec(plan, nagg, &bindings, &err);
+  if (!execp) return GX_ERR_INTERNAL;
+  auto t2 = clk::now();
+  execp->open();
+  int64_t groups = 0;
+  for (;;) {
+    Chunk out;
+    if (execp->next(out) != GX_OK) return GX_ERR_INTERNAL;
+    if (out.numRows() == 0) break;
+    groups += out.numRows();
+  }
+  execp->close();
+  auto t3 = clk::now();
+  *gen_ms = std::chrono::duration<double, std::milli>(t1 - t0).count();
+  *exec_ms = std::chrono::duration<double, std::milli>(t3 - t2).count();
+  *groups_out = groups;
+  return GX_OK;
+}

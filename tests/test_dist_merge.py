@@ -1,0 +1,118 @@
+"""Multi-process merge semantics on CPU (gloo, world_size 2): each rank
+produces canonical Q1 partial states (oracle executor — the CPU stand-in for
+the per-GPU fused kernel), ranks all_gather the tiny partial chunks, and every
+rank runs the FINAL merge — MergePartialResult semantics (aggfuncs.go:250-255).
+The same merge path runs on the product library (host side, no GPU needed),
+which is what bench.py uses after RCCL all_gather.
+"""
+import os
+
+import pytest
+
+from tests.gxlib import GX_AGG_MODE_PARTIAL, GX_TPCH_LINEITEM, load_oracle
+from tidb_amd import plan as P
+
+
+def merge_partials(lib, partial_rows):
+    """Feed canonical partial rows through the FINAL plan of `lib`;
+    returns {group: aggs} map."""
+    from tidb_amd.chunkpy import PyChunk
+    from tidb_amd.decimals import str_to_decimal_bytes
+    b, src, agg, out_types, out_fracs, part_types, part_fracs = \
+        P.q1_final_plan(lib)
+    chunk = PyChunk(part_types, max(len(partial_rows), 1), part_fracs,
+                    data_caps=[4096] * len(part_types))
+    for r in partial_rows:
+        vals = []
+        for v, t in zip(r, part_types):
+            if t == 2 and v is not None:
+                vals.append(str_to_decimal_bytes(lib, v))
+            else:
+                vals.append(v)
+        chunk.append_row(vals)
+    ex = b.build(agg)
+    ex.bind_chunks(src, [chunk])
+    ex.open()
+    caps = [2048 if t == 4 else None for t in out_types]
+    rows = ex.pull_all(out_types, out_fracs, data_caps=caps)
+    ex.close()
+    ex.free()
+    b.free()
+    return {(r[0], r[1]): tuple(r[2:]) for r in rows}
+
+
+def _worker(rank, world, result_q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29711"
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    lib = load_oracle()
+    total = 20000
+    per = total // world
+    b, src, agg, out_types, out_fracs = P.q1_plan(lib, GX_AGG_MODE_PARTIAL)
+    ex = b.build(agg)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, per, 42, rank * per, total)
+    ex.open()
+    caps = [2048 if t == 4 else None for t in out_types]
+    rows = ex.pull_all(out_types, out_fracs, data_caps=caps)
+    ex.close()
+    ex.free()
+    b.free()
+    # all_gather the partial rows (object collective: payload is KBs)
+    gathered = [None] * world
+    dist.all_gather_object(gathered, rows)
+    all_rows = [r for part in gathered for r in part]
+    merged = merge_partials(lib, all_rows)
+    dist.destroy_process_group()
+    result_q.put((rank, merged))
+
+
+def test_gloo_two_rank_merge():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, 2, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, merged = q.get(timeout=180)
+        results[rank] = merged
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert results[0] == results[1]
+    # equals single-process COMPLETE
+    lib = load_oracle()
+    from tests.test_oracle_q1 import run_q1, q1_results_to_map
+    complete = q1_results_to_map(run_q1(lib, 20000))
+    # convert merged decimal strings to the same normal form
+    from fractions import Fraction
+    def norm(m):
+        out = {}
+        for k, v in m.items():
+            out[k] = tuple(Fraction(x) if isinstance(x, str) else x for x in v)
+        return out
+    got = norm(results[0])
+    want = {}
+    for k, v in complete.items():
+        # complete map is in integer units; reconstruct fractions
+        scales = [2, 2, 4, 6, 6, 6, 6, None]
+        want[k] = tuple(
+            Fraction(x, 10 ** s) if s is not None else x
+            for x, s in zip(v, scales))
+    assert got == want
+
+
+def test_product_final_merge_matches_oracle():
+    """The product library's host-side FINAL merge (used after the RCCL
+    gather) must equal the oracle's on identical partial chunks. Runs on CPU."""
+    from tests.gxlib import load_product
+    from tests.test_oracle_q1 import run_q1
+    oracle = load_oracle()
+    product = load_product()
+    partials = []
+    for off in (0, 7000, 14000):
+        partials.extend(run_q1(oracle, 7000, GX_AGG_MODE_PARTIAL,
+                               row_offset=off, total_rows=21000))
+    assert merge_partials(product, partials) == merge_partials(oracle, partials)

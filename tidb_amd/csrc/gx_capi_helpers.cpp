@@ -1,17 +1,16 @@
-// oracle/capi_helpers.cpp — C-ABI decimal/time helper exports (oracle flavor).
-// ORACLE / TEST INFRASTRUCTURE ONLY.
+// tidb_amd/csrc/gx_capi_helpers.cpp — product C-ABI decimal/time helpers.
+// Backed by the product's own host decimal (gx_decimal.cpp).
 #include <cstring>
+#include <string>
 
-#include "../include/gx_executor.h"
-#include "core_time.h"
-#include "mydecimal.h"
+#include "../../include/gx_executor.h"
+#include "gx_decimal.h"
 
-using oracle::MyDecimal;
+using gxp::MyDecimal;
 
 static inline const MyDecimal* D(const uint8_t* p) {
   return reinterpret_cast<const MyDecimal*>(p);
 }
-static inline MyDecimal* D(uint8_t* p) { return reinterpret_cast<MyDecimal*>(p); }
 
 extern "C" {
 
@@ -21,7 +20,6 @@ int32_t gx_dec_from_string(const char* s, int32_t len, uint8_t out40[40]) {
   std::memcpy(out40, &d, 40);
   return err;
 }
-
 int32_t gx_dec_to_string(const uint8_t dec40[40], char* buf, int32_t buf_len) {
   std::string s = D(dec40)->ToString();
   if ((int32_t)s.size() + 1 > buf_len) return GX_ERR_INVALID;
@@ -29,7 +27,6 @@ int32_t gx_dec_to_string(const uint8_t dec40[40], char* buf, int32_t buf_len) {
   buf[s.size()] = 0;
   return (int32_t)s.size();
 }
-
 int32_t gx_dec_display_string(const uint8_t dec40[40], char* buf, int32_t buf_len) {
   std::string s = D(dec40)->DisplayString();
   if ((int32_t)s.size() + 1 > buf_len) return GX_ERR_INVALID;
@@ -37,24 +34,23 @@ int32_t gx_dec_display_string(const uint8_t dec40[40], char* buf, int32_t buf_le
   buf[s.size()] = 0;
   return (int32_t)s.size();
 }
-
 int32_t gx_dec_add(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]) {
-  return oracle::DecimalAdd(D(a), D(b), D(out));
+  return gxp::DecimalAdd(D(a), D(b), (MyDecimal*)out);
 }
 int32_t gx_dec_sub(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]) {
-  return oracle::DecimalSub(D(a), D(b), D(out));
+  return gxp::DecimalSub(D(a), D(b), (MyDecimal*)out);
 }
 int32_t gx_dec_mul(const uint8_t a[40], const uint8_t b[40], uint8_t out[40]) {
-  return oracle::DecimalMul(D(a), D(b), D(out));
+  return gxp::DecimalMul(D(a), D(b), (MyDecimal*)out);
 }
 int32_t gx_dec_div(const uint8_t a[40], const uint8_t b[40], uint8_t out[40],
                    int32_t frac_incr) {
-  return oracle::DecimalDiv(D(a), D(b), D(out), frac_incr);
+  return gxp::DecimalDiv(D(a), D(b), (MyDecimal*)out, frac_incr);
 }
 int32_t gx_dec_round(const uint8_t in[40], int32_t frac, int32_t round_mode,
                      uint8_t out[40]) {
   MyDecimal tmp = *D(in);
-  int32_t err = tmp.Round(&tmp, frac, (oracle::RoundMode)round_mode);
+  int32_t err = tmp.Round(&tmp, frac, (gxp::RoundMode)round_mode);
   std::memcpy(out, &tmp, 40);
   return err;
 }
@@ -98,24 +94,21 @@ int32_t gx_dec_result_frac(const uint8_t dec40[40]) {
   return (int32_t)D(dec40)->resultFrac;
 }
 
-uint64_t gx_time_from_date(int32_t year, int32_t month, int32_t day) {
-  return oracle::TimeFromDate(year, month, day);
+// packed CoreTime (time.go:235-251,266-283)
+uint64_t gx_time_from_date(int32_t y, int32_t m, int32_t d) {
+  return ((uint64_t)y << 50) | ((uint64_t)m << 46) | ((uint64_t)d << 41) | 0xEULL;
 }
-uint64_t gx_time_from_datetime(int32_t year, int32_t month, int32_t day,
-                               int32_t hour, int32_t minute, int32_t second,
-                               int32_t microsecond, int32_t type_and_fsp) {
-  return oracle::TimeFromDatetime(year, month, day, hour, minute, second,
-                                  microsecond, type_and_fsp);
+uint64_t gx_time_from_datetime(int32_t y, int32_t m, int32_t d, int32_t hh,
+                               int32_t mm, int32_t ss, int32_t us,
+                               int32_t type_and_fsp) {
+  uint64_t v = ((uint64_t)y << 50) | ((uint64_t)m << 46) | ((uint64_t)d << 41) |
+               ((uint64_t)hh << 36) | ((uint64_t)mm << 30) | ((uint64_t)ss << 24) |
+               ((uint64_t)us << 4);
+  return (v & ~0xFULL) | ((uint64_t)type_and_fsp & 0xF);
 }
-int32_t gx_time_compare(uint64_t a, uint64_t b) { return oracle::CompareTime(a, b); }
-
-int32_t gx_engine_is_gpu(void) { return 0; }
-const char* gx_engine_name(void) { return "oracle-cpu"; }
+int32_t gx_time_compare(uint64_t a, uint64_t b) {
+  uint64_t ma = a & ~0xFULL, mb = b & ~0xFULL;
+  return ma < mb ? -1 : (ma > mb ? 1 : 0);
+}
 
 }  // extern "C"
-
-extern "C" {
-// oracle engine runs no GPU kernels
-double gx_last_kernel_ms(gx_exec* ex) { (void)ex; return 0; }
-int64_t gx_last_sel_count(gx_exec* ex) { (void)ex; return 0; }
-}

@@ -1,0 +1,141 @@
+// tidb_amd/csrc/gx_common.h — product engine: host/device shared descriptors.
+//
+// The engine compiles the C-ABI plan (include/gx_executor.h) into these
+// descriptors; gx_kernels.hip executes them in a single fused pass
+// (scan -> filter -> project -> hash-aggregate), the MI355X-native form of
+// the reference pipeline SelectionExec -> ProjectionExec -> HashAggExec
+// (pkg/executor/select.go:750, projection.go:77, aggregate/*).
+#ifndef GX_COMMON_H
+#define GX_COMMON_H
+
+#include <cstdint>
+
+namespace gxp {
+
+// ---- device column (reference chunk.Column layout resident in HBM) ----
+struct DevCol {
+  void* data = nullptr;          // fixed: N*elem; varlen: bytes
+  uint8_t* nullBitmap = nullptr; // LSB-first, 1 = NOT NULL; null => no NULLs
+  int64_t* offsets = nullptr;    // varlen: N+1
+  int32_t elemSize = 8;          // -1 varlen
+  int32_t type = 0;              // gx type
+  int32_t frac = 0;
+  uint8_t hasNulls = 0;
+};
+
+constexpr int kMaxCols = 16;
+
+struct DevTable {
+  DevCol cols[kMaxCols];
+  int32_t nCols = 0;
+  int64_t nRows = 0;
+};
+
+// ---- per-row expression VM (fixed-point decimal) ----
+// Registers are int128 values at a STATIC scale decided at plan-compile time;
+// null-ness is a per-register bit. Engine guarantees operand magnitudes fit:
+// column loads are validated (digitsInt <= 18, digitsFrac <= 9) and overflow
+// in MUL sets the kernel error flag (no silent wrap, no CPU fallback).
+enum VmOp : int32_t {
+  VM_LOAD_DEC = 0,   // dst <- decimal column a (units at its frac)
+  VM_LOAD_I64 = 1,   // dst <- int64 column a (scale 0)
+  VM_LOAD_CONST = 2, // dst <- const[a] (pre-scaled by engine)
+  VM_ADD = 3,        // dst <- a + b (same scale, engine-aligned)
+  VM_SUB = 4,
+  VM_MUL = 5,        // dst <- a * b (scale = sa + sb)
+  VM_SCALE_UP = 6,   // dst <- a * 10^b (align scales)
+};
+
+struct VmIns {
+  int32_t op;
+  int32_t dst;  // register index
+  int32_t a;    // column index / const index / register
+  int32_t b;    // register / power
+};
+
+constexpr int kMaxVmIns = 48;
+constexpr int kMaxVmRegs = 16;
+constexpr int kMaxVmConsts = 16;
+
+// ---- filter (CNF of simple predicates; Q1-class) ----
+enum PredKind : int32_t {
+  PRED_TIME_CMP_CONST = 0,  // masked-u64 compare (core_time.go:256 semantics)
+  PRED_I64_CMP_CONST = 1,
+  PRED_DEC_CMP_CONST = 2,   // units compare at engine-aligned scale
+};
+
+struct PredDesc {
+  int32_t kind;
+  int32_t col;
+  int32_t cmp;        // GX_F_LT..GX_F_NE
+  uint64_t constU64;  // time value or i64/units bits
+};
+
+constexpr int kMaxPreds = 8;
+
+// ---- aggregation ----
+// Per-group state layout (all aggs): int128 acc + int64 count per agg slot.
+// SUM/AVG: acc = exact fixed-point sum, count = notNullRowCount;
+// COUNT: count only; MIN/MAX/FIRSTROW: acc = value units, count = hasValue.
+struct AggDesc {
+  int32_t func;   // GX_AGG_*
+  int32_t srcReg; // VM register holding the arg value (-1 for count(*))
+  int32_t scale;  // scale of the accumulated units
+};
+
+constexpr int kMaxAggs = 12;
+
+// ---- group keys ----
+// Round-1 device grouping: group-by columns pack into ONE u64 key
+// (string cols <= 3 bytes each as len<<24|bytes in a 32-bit lane; Q1 uses two
+// char(1) columns). Wider keys (Q3's high-NDV int keys) arrive with the Q3
+// path. kEmptyKey is reserved.
+struct GroupKeyDesc {
+  int32_t nCols;
+  int32_t col[2];
+  int32_t kind[2];  // 0 = short string, 1 = small i64 (<2^31)
+};
+
+constexpr uint64_t kEmptyKey = ~0ULL;
+constexpr int kLdsGroups = 128;    // per-workgroup LDS table capacity
+constexpr int kGlobalGroups = 8192; // global table capacity (power of two)
+
+// group state sized for kMaxAggs
+struct GroupSlot {
+  uint64_t key;
+  uint64_t accLo[kMaxAggs];
+  int64_t accHi[kMaxAggs];
+  int64_t cnt[kMaxAggs];
+};
+
+struct FusedQueryDesc {
+  DevTable table;
+  // filter
+  PredDesc preds[kMaxPreds];
+  int32_t nPreds = 0;
+  // projection/arg VM
+  VmIns ins[kMaxVmIns];
+  int32_t nIns = 0;
+  int64_t constLo[kMaxVmConsts];
+  int64_t constHi[kMaxVmConsts];
+  int32_t nConsts = 0;
+  // aggs
+  AggDesc aggs[kMaxAggs];
+  int32_t nAggs = 0;
+  GroupKeyDesc gkey;
+  // outputs
+  GroupSlot* globalTable = nullptr;  // kGlobalGroups slots
+  uint32_t* errorFlag = nullptr;     // != 0 => abort with error
+  uint64_t* selCount = nullptr;      // rows passing the filter (stats)
+};
+
+// host-side launch wrappers (defined in gx_kernels.hip)
+int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows,
+                    uint64_t seed, int64_t totalRows, void* stream);
+int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
+                     void* stream);
+int gxLaunchMemset(void* p, int v, size_t n, void* stream);
+
+}  // namespace gxp
+
+#endif

@@ -1,0 +1,1326 @@
+// tidb_amd/csrc/gx_engine.cpp — MI355X product engine: C-ABI implementation.
+//
+// Drop-in for the reference executor tree behind exec.Executor
+// (pkg/executor/internal/exec/executor.go:224-250): the engine compiles the
+// plan (Source -> [Selection] -> [Projection] -> HashAgg) into ONE fused HIP
+// kernel pass (gx_kernels.hip) — columns resident in HBM, chunks cross
+// host<->device only at the tree fringe (source bind, final small results).
+//
+// There is NO CPU fallback for the compute path: a plan that needs the GPU
+// fails loudly (GX_ERR_NO_GPU) when no MI355X is reachable. The only host
+// compute is the O(#groups) finalize (avg DecimalDiv + Round — the reference
+// does this once per group too, func_avg.go:84-109) and the FINAL-mode merge
+// of canonical partial states (KB-sized; MergePartialResult semantics,
+// aggfuncs.go:250-255).
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cstring>
+#include <map>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "../../include/gx_executor.h"
+#include "gx_common.h"
+#include "gx_decimal.h"
+
+using gxp::MyDecimal;
+
+namespace {
+
+// ---------------- plan IR (mirrors the C-ABI builder calls) ----------------
+enum { EK_COLREF, EK_CONST, EK_CALL };
+enum { PK_SOURCE, PK_SELECTION, PK_PROJECTION, PK_HASHAGG, PK_TOPN, PK_HASHJOIN };
+
+struct PExpr {
+  int kind = EK_COLREF;
+  int retType = GX_TYPE_I64;
+  int retFrac = 0;
+  int colIdx = -1;
+  int func = -1;
+  std::vector<int> args;
+  int64_t constI64 = 0;
+  double constF64 = 0;
+  uint64_t constTime = 0;
+  MyDecimal constDec;
+  std::string constStr;
+};
+
+struct PNode {
+  int kind = PK_SOURCE;
+  int child = -1, child2 = -1;
+  std::vector<int> colTypes, colFracs;
+  std::vector<int> exprs;
+  std::vector<int> aggFuncs, aggArgs, aggFracs;
+  int aggMode = 0;
+  std::vector<uint8_t> keyDesc;
+  int64_t limit = 0, offset = 0;
+  std::vector<int> buildKeys, probeKeys;
+  int joinType = 0;
+};
+
+struct PPlan {
+  std::vector<PExpr> exprs;
+  std::vector<PNode> nodes;
+};
+
+// host copy of one bound chunk column
+struct HostCol {
+  int type, frac;
+  std::vector<uint8_t> data;
+  std::vector<uint8_t> nullBitmap;
+  std::vector<int64_t> offsets;
+  int length = 0;
+};
+
+struct Binding {
+  bool haveChunks = false;
+  std::vector<std::vector<HostCol>> chunks;
+  int tpchTable = -1;
+  int64_t tpchRows = 0, tpchRowOffset = 0, tpchTotalRows = 0;
+  uint64_t tpchSeed = 42;
+};
+
+struct OutRowVal {
+  bool isNull = false;
+  int type = GX_TYPE_I64;
+  int64_t i64 = 0;
+  uint64_t u64 = 0;
+  double f64 = 0;
+  MyDecimal dec;
+  std::string str;
+};
+
+}  // namespace
+
+struct gx_pb {
+  PPlan plan;
+};
+
+struct gx_exec {
+  PPlan plan;
+  int root = -1;
+  int device = -1;
+  std::map<int, Binding> bindings;
+  std::string err;
+  bool opened = false;
+
+  // compiled fused query
+  bool isFused = false;
+  bool isFinalHost = false;
+  bool isBareSource = false;
+  int sourceNode = -1;
+  gxp::FusedQueryDesc desc;
+  std::vector<std::pair<int, int>> projRegs;  // projection idx -> (reg, scale)
+  int vmNextReg = 0;
+  std::map<int, std::pair<int, int>> exprRegCache;  // exprId -> (reg, scale)
+  // device state
+  bool deviceReady = false;
+  std::vector<void*> devBufs;
+  gxp::GroupSlot* devTable = nullptr;
+  gxp::FusedQueryDesc* devDesc = nullptr;
+  uint32_t* devErr = nullptr;
+  uint64_t* devSel = nullptr;
+  hipStream_t stream = nullptr;
+  // results
+  bool ranQuery = false;
+  std::vector<std::vector<OutRowVal>> resultRows;
+  size_t emitPos = 0;
+  // bare-source emit state
+  int64_t srcPos = 0;
+  uint64_t lastSelCount = 0;
+  double lastKernelMs = 0;
+
+  ~gx_exec() {
+    for (void* p : devBufs) hipFree(p);
+  }
+};
+
+// ---------------- helpers ----------------
+
+static bool gpuAvailable() {
+  int n = 0;
+  return hipGetDeviceCount(&n) == hipSuccess && n > 0;
+}
+
+// gxp::MyDecimal -> int128 units at its digitsFrac (requires <= 38 digits)
+static bool decToUnits(const MyDecimal& d, __int128* out, int* scale) {
+  int wordsInt = (d.digitsInt + 8) / 9;
+  int wordsFrac = (d.digitsFrac + 8) / 9;
+  if (d.digitsFrac > 9) return false;  // round-1 device scales are <= 9
+  __int128 ip = 0;
+  for (int i = 0; i < wordsInt; i++) {
+    ip = ip * 1000000000 + d.wordBuf[i];
+    if (ip > (__int128)1e36) return false;
+  }
+  static const int64_t p10[10] = {1, 10, 100, 1000, 10000, 100000, 1000000,
+                                  10000000, 100000000, 1000000000};
+  int64_t fr = 0;
+  if (wordsFrac > 0) fr = d.wordBuf[wordsInt] / p10[9 - d.digitsFrac];
+  __int128 u = ip * p10[d.digitsFrac] + fr;
+  if (d.negative) u = -u;
+  *out = u;
+  *scale = d.digitsFrac;
+  return true;
+}
+
+// int128 units at `scale` -> canonical MyDecimal (digit-exact value)
+static MyDecimal decFromUnits(__int128 u, int scale) {
+  MyDecimal d;
+  bool neg = u < 0;
+  unsigned __int128 a = neg ? (unsigned __int128)(-u) : (unsigned __int128)u;
+  static const int64_t p10[10] = {1, 10, 100, 1000, 10000, 100000, 1000000,
+                                  10000000, 100000000, 1000000000};
+  unsigned __int128 ip = a / (unsigned)p10[scale];
+  int64_t fr = (int64_t)(a % (unsigned)p10[scale]);
+  // integer words, most significant first
+  int32_t words[6] = {0};
+  int nw = 0;
+  if (ip == 0) {
+    words[0] = 0;
+    nw = 1;
+  } else {
+    int32_t tmp[6];
+    int k = 0;
+    while (ip > 0) {
+      tmp[k++] = (int32_t)(ip % 1000000000u);
+      ip /= 1000000000u;
+    }
+    nw = k;
+    for (int i = 0; i < k; i++) words[i] = tmp[k - 1 - i];
+  }
+  int digitsInt = (nw - 1) * 9;
+  {
+    int32_t head = words[0];
+    int hd = 1;
+    while (head >= 10) {
+      head /= 10;
+      hd++;
+    }
+    digitsInt += (words[0] == 0 && nw == 1) ? 1 : hd;
+  }
+  d.digitsInt = (int8_t)digitsInt;
+  d.digitsFrac = (int8_t)scale;
+  d.resultFrac = (int8_t)scale;
+  d.negative = neg && (a != 0);
+  for (int i = 0; i < nw; i++) d.wordBuf[i] = words[i];
+  if (scale > 0) d.wordBuf[nw] = (int32_t)(fr * p10[9 - scale]);
+  return d;
+}
+
+static void setDevColMeta(gxp::DevCol* c, int type, int frac) {
+  c->type = type;
+  c->frac = frac;
+  c->elemSize = type == GX_TYPE_DECIMAL ? 40 : (type == GX_TYPE_STRING ? -1 : 8);
+}
+
+#define HIP_OK(ex, call)                                        \
+  do {                                                          \
+    hipError_t _e = (call);                                     \
+    if (_e != hipSuccess) {                                     \
+      (ex)->err = std::string("HIP error: ") + hipGetErrorString(_e); \
+      return GX_ERR_INTERNAL;                                   \
+    }                                                           \
+  } while (0)
+
+static void* devAlloc(gx_exec* ex, size_t n) {
+  void* p = nullptr;
+  if (hipMalloc(&p, n) != hipSuccess) return nullptr;
+  ex->devBufs.push_back(p);
+  return p;
+}
+
+// ---------------- plan compilation ----------------
+
+// compile expression exprId (over SOURCE columns) into the VM; returns reg or
+// -1 on error. *scaleOut = static scale of the value.
+static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
+  gxp::FusedQueryDesc& d = ex->desc;
+  const PExpr& e = ex->plan.exprs[exprId];
+  auto cached = ex->exprRegCache.find(exprId);
+  if (cached != ex->exprRegCache.end()) {
+    *scaleOut = cached->second.second;
+    return cached->second.first;
+  }
+  auto allocReg = [&]() -> int {
+    if (ex->vmNextReg >= gxp::kMaxVmRegs) return -1;
+    return ex->vmNextReg++;
+  };
+  auto emit = [&](int op, int dst, int a, int b) -> int {
+    if (dst < 0 || d.nIns >= gxp::kMaxVmIns) return -1;
+    d.ins[d.nIns++] = {op, dst, a, b};
+    return dst;
+  };
+  int reg = -1;
+  switch (e.kind) {
+    case EK_COLREF: {
+      if (e.colIdx < 0 || e.colIdx >= ex->desc.table.nCols) {
+        ex->err = "bad colref";
+        return -1;
+      }
+      int t = ex->desc.table.cols[e.colIdx].type;
+      if (t == GX_TYPE_DECIMAL) {
+        int frac = ex->desc.table.cols[e.colIdx].frac;
+        reg = emit(gxp::VM_LOAD_DEC, allocReg(), e.colIdx, frac);
+        *scaleOut = frac;
+      } else if (t == GX_TYPE_I64) {
+        reg = emit(gxp::VM_LOAD_I64, allocReg(), e.colIdx, 0);
+        *scaleOut = 0;
+      } else {
+        ex->err = "unsupported colref type in device expression";
+        return -1;
+      }
+      break;
+    }
+    case EK_CONST: {
+      __int128 u = 0;
+      int sc = 0;
+      if (e.retType == GX_TYPE_DECIMAL) {
+        if (!decToUnits(e.constDec, &u, &sc)) {
+          ex->err = "const decimal too wide for device path";
+          return -1;
+        }
+      } else if (e.retType == GX_TYPE_I64) {
+        u = e.constI64;
+        sc = 0;
+      } else {
+        ex->err = "unsupported const type in device expression";
+        return -1;
+      }
+      if (d.nConsts >= gxp::kMaxVmConsts) {
+        ex->err = "too many consts";
+        return -1;
+      }
+      int ci = d.nConsts++;
+      d.constLo[ci] = (int64_t)(uint64_t)u;
+      d.constHi[ci] = (int64_t)(u >> 64);
+      reg = emit(gxp::VM_LOAD_CONST, allocReg(), ci, 0);
+      *scaleOut = sc;
+      break;
+    }
+    case EK_CALL: {
+      if (e.args.size() != 2) {
+        ex->err = "unsupported call arity on device";
+        return -1;
+      }
+      int sa = 0, sb = 0;
+      int ra = compileExpr(ex, e.args[0], &sa);
+      if (ra < 0) return -1;
+      int rb = compileExpr(ex, e.args[1], &sb);
+      if (rb < 0) return -1;
+      int op;
+      switch (e.func) {
+        case GX_F_PLUS: op = gxp::VM_ADD; break;
+        case GX_F_MINUS: op = gxp::VM_SUB; break;
+        case GX_F_MUL: op = gxp::VM_MUL; break;
+        default:
+          ex->err = "unsupported function on device path";
+          return -1;
+      }
+      if (op == gxp::VM_MUL) {
+        reg = emit(op, allocReg(), ra, rb);
+        *scaleOut = sa + sb;
+      } else {
+        // align scales to max (MySQL add/sub result frac = max(f1,f2))
+        int target = std::max(sa, sb);
+        if (sa < target) ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa);
+        if (sb < target) rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb);
+        if (ra < 0 || rb < 0) break;
+        reg = emit(op, allocReg(), ra, rb);
+        *scaleOut = target;
+      }
+      break;
+    }
+  }
+  if (reg < 0 && ex->err.empty()) ex->err = "expression too large for device VM";
+  if (reg >= 0) ex->exprRegCache[exprId] = {reg, *scaleOut};
+  return reg;
+}
+
+// compile the fused Source->[Selection]->[Projection]->HashAgg pipeline
+static int32_t compileFused(gx_exec* ex) {
+  const PPlan& plan = ex->plan;
+  const PNode* agg = &plan.nodes[ex->root];
+  const PNode* proj = nullptr;
+  const PNode* sel = nullptr;
+  const PNode* src = nullptr;
+  const PNode* cur = &plan.nodes[agg->child];
+  if (cur->kind == PK_PROJECTION) {
+    proj = cur;
+    cur = &plan.nodes[cur->child];
+  }
+  if (cur->kind == PK_SELECTION) {
+    sel = cur;
+    cur = &plan.nodes[cur->child];
+  }
+  if (cur->kind != PK_SOURCE) {
+    ex->err = "unsupported plan shape for device execution "
+              "(want HashAgg <- [Projection] <- [Selection] <- Source)";
+    return GX_ERR_INVALID;
+  }
+  src = cur;
+  ex->sourceNode = (int)(src - plan.nodes.data());
+
+  // source schema -> desc.table metadata (pointers at open)
+  if ((int)src->colTypes.size() > gxp::kMaxCols) {
+    ex->err = "too many source columns";
+    return GX_ERR_INVALID;
+  }
+  ex->desc.table.nCols = (int)src->colTypes.size();
+  for (size_t c = 0; c < src->colTypes.size(); c++)
+    setDevColMeta(&ex->desc.table.cols[c], src->colTypes[c], src->colFracs[c]);
+
+  // selection -> PredDescs
+  if (sel) {
+    for (int condId : sel->exprs) {
+      const PExpr& e = plan.exprs[condId];
+      if (e.kind != EK_CALL || e.func > GX_F_NE || e.args.size() != 2) {
+        ex->err = "unsupported filter expression on device";
+        return GX_ERR_INVALID;
+      }
+      const PExpr* lhs = &plan.exprs[e.args[0]];
+      const PExpr* rhs = &plan.exprs[e.args[1]];
+      int cmp = e.func;
+      if (lhs->kind == EK_CONST && rhs->kind == EK_COLREF) {
+        std::swap(lhs, rhs);
+        // mirror the comparison
+        static const int mirror[6] = {GX_F_GT, GX_F_GE, GX_F_LT, GX_F_LE,
+                                      GX_F_EQ, GX_F_NE};
+        cmp = mirror[cmp];
+      }
+      if (lhs->kind != EK_COLREF || rhs->kind != EK_CONST) {
+        ex->err = "device filter must be <column> <cmp> <const>";
+        return GX_ERR_INVALID;
+      }
+      if (ex->desc.nPreds >= gxp::kMaxPreds) {
+        ex->err = "too many filter conjuncts";
+        return GX_ERR_INVALID;
+      }
+      gxp::PredDesc pd{};
+      pd.col = lhs->colIdx;
+      pd.cmp = cmp;
+      int ct = src->colTypes[lhs->colIdx];
+      if (ct == GX_TYPE_TIME && rhs->retType == GX_TYPE_TIME) {
+        pd.kind = gxp::PRED_TIME_CMP_CONST;
+        pd.constU64 = rhs->constTime;
+      } else if (ct == GX_TYPE_I64 && rhs->retType == GX_TYPE_I64) {
+        pd.kind = gxp::PRED_I64_CMP_CONST;
+        pd.constU64 = (uint64_t)rhs->constI64;
+      } else if (ct == GX_TYPE_DECIMAL && rhs->retType == GX_TYPE_DECIMAL) {
+        __int128 u;
+        int sc;
+        if (!decToUnits(rhs->constDec, &u, &sc)) {
+          ex->err = "filter const decimal too wide";
+          return GX_ERR_INVALID;
+        }
+        // align const to the column's declared frac
+        int colFrac = src->colFracs[lhs->colIdx];
+        while (sc < colFrac) {
+          u *= 10;
+          sc++;
+        }
+        if (sc != colFrac || u > INT64_MAX || u < INT64_MIN) {
+          ex->err = "filter decimal const/scale unsupported";
+          return GX_ERR_INVALID;
+        }
+        pd.kind = gxp::PRED_DEC_CMP_CONST;
+        pd.constU64 = (uint64_t)(int64_t)u;
+      } else {
+        ex->err = "unsupported filter column/const type combination";
+        return GX_ERR_INVALID;
+      }
+      ex->desc.preds[ex->desc.nPreds++] = pd;
+    }
+  }
+
+  // projection exprs -> VM registers (group-col projections stay colrefs)
+  std::vector<int> projSrcCol;  // proj idx -> source col for passthroughs
+  ex->projRegs.clear();
+  if (proj) {
+    for (int pe : proj->exprs) {
+      const PExpr& e = plan.exprs[pe];
+      if (e.kind == EK_COLREF &&
+          (src->colTypes[e.colIdx] == GX_TYPE_STRING ||
+           src->colTypes[e.colIdx] == GX_TYPE_TIME)) {
+        projSrcCol.push_back(e.colIdx);
+        ex->projRegs.push_back({-1, 0});
+        continue;
+      }
+      int sc = 0;
+      int reg = compileExpr(ex, pe, &sc);
+      if (reg < 0) return GX_ERR_INVALID;
+      projSrcCol.push_back(e.kind == EK_COLREF ? e.colIdx : -1);
+      ex->projRegs.push_back({reg, sc});
+    }
+  }
+
+  // group keys
+  gxp::GroupKeyDesc gk{};
+  gk.nCols = 0;
+  for (int ge : agg->exprs) {
+    const PExpr& e = plan.exprs[ge];
+    int srcCol = -1;
+    if (e.kind != EK_COLREF) {
+      ex->err = "group-by expression must be a column";
+      return GX_ERR_INVALID;
+    }
+    if (proj) {
+      if (e.colIdx >= (int)projSrcCol.size() || projSrcCol[e.colIdx] < 0) {
+        ex->err = "group-by must reference a passthrough column";
+        return GX_ERR_INVALID;
+      }
+      srcCol = projSrcCol[e.colIdx];
+    } else {
+      srcCol = e.colIdx;
+    }
+    if (gk.nCols >= 2) {
+      ex->err = "device grouping supports <= 2 key columns this round";
+      return GX_ERR_INVALID;
+    }
+    int t = src->colTypes[srcCol];
+    gk.col[gk.nCols] = srcCol;
+    gk.kind[gk.nCols] = t == GX_TYPE_STRING ? 0 : 1;
+    if (t != GX_TYPE_STRING && t != GX_TYPE_I64) {
+      ex->err = "device group key must be string or int64";
+      return GX_ERR_INVALID;
+    }
+    gk.nCols++;
+  }
+  ex->desc.gkey = gk;
+
+  // aggs
+  if ((int)agg->aggFuncs.size() > gxp::kMaxAggs) {
+    ex->err = "too many aggregates";
+    return GX_ERR_INVALID;
+  }
+  for (size_t a = 0; a < agg->aggFuncs.size(); a++) {
+    gxp::AggDesc ad{};
+    ad.func = agg->aggFuncs[a];
+    if (ad.func != GX_AGG_COUNT && ad.func != GX_AGG_SUM && ad.func != GX_AGG_AVG) {
+      ex->err = "device aggregation supports count/sum/avg this round";
+      return GX_ERR_INVALID;
+    }
+    int argE = agg->aggArgs[a];
+    if (argE < 0) {
+      ad.srcReg = -1;
+      ad.scale = 0;
+    } else {
+      const PExpr& e = plan.exprs[argE];
+      if (proj) {
+        if (e.kind != EK_COLREF || e.colIdx >= (int)ex->projRegs.size() ||
+            ex->projRegs[e.colIdx].first < 0) {
+          ex->err = "agg arg must reference a projected value";
+          return GX_ERR_INVALID;
+        }
+        ad.srcReg = ex->projRegs[e.colIdx].first;
+        ad.scale = ex->projRegs[e.colIdx].second;
+      } else {
+        int sc = 0;
+        int reg = compileExpr(ex, argE, &sc);
+        if (reg < 0) return GX_ERR_INVALID;
+        ad.srcReg = reg;
+        ad.scale = sc;
+      }
+    }
+    ex->desc.aggs[ex->desc.nAggs++] = ad;
+  }
+  ex->isFused = true;
+  return GX_OK;
+}
+
+// ---------------- device materialization ----------------
+
+static int32_t materializeDevice(gx_exec* ex) {
+  if (ex->deviceReady) return GX_OK;
+  if (!gpuAvailable()) {
+    ex->err = "no MI355X visible: the product engine has no CPU fallback "
+              "(GX_ERR_NO_GPU)";
+    return GX_ERR_NO_GPU;
+  }
+  if (ex->device >= 0) hipSetDevice(ex->device);
+  HIP_OK(ex, hipStreamCreate(&ex->stream));
+  auto it = ex->bindings.find(ex->sourceNode);
+  if (it == ex->bindings.end()) {
+    ex->err = "source not bound";
+    return GX_ERR_INVALID;
+  }
+  Binding& b = it->second;
+  gxp::DevTable& tab = ex->desc.table;
+  if (b.tpchTable >= 0) {
+    tab.nRows = b.tpchRows;
+    int64_t n = b.tpchRows;
+    for (int c = 0; c < tab.nCols; c++) {
+      gxp::DevCol& col = tab.cols[c];
+      size_t bytes;
+      if (col.type == GX_TYPE_STRING) {
+        // generator strings are single bytes (char(1)); offsets dense
+        col.offsets = (int64_t*)devAlloc(ex, (n + 1) * 8);
+        col.data = devAlloc(ex, std::max<int64_t>(n, 1));
+        if (!col.offsets || !col.data) {
+          ex->err = "hipMalloc failed";
+          return GX_ERR_INTERNAL;
+        }
+      } else {
+        bytes = (size_t)n * col.elemSize;
+        col.data = devAlloc(ex, std::max<size_t>(bytes, 1));
+        if (!col.data) {
+          ex->err = "hipMalloc failed";
+          return GX_ERR_INTERNAL;
+        }
+      }
+      col.nullBitmap = nullptr;  // synthetic data has no NULLs
+      col.hasNulls = 0;
+    }
+    int rc = gxp::gxLaunchTpchGen(b.tpchTable, &tab, b.tpchRowOffset, n,
+                                  b.tpchSeed,
+                                  b.tpchTotalRows > 0 ? b.tpchTotalRows : n,
+                                  ex->stream);
+    if (rc != 0) {
+      ex->err = "generator launch failed: " +
+                std::string(hipGetErrorString((hipError_t)rc));
+      return GX_ERR_INTERNAL;
+    }
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  } else if (b.haveChunks) {
+    // concatenate bound chunks into one device table
+    int nCols = tab.nCols;
+    int64_t total = 0;
+    for (auto& ch : b.chunks) total += ch.empty() ? 0 : ch[0].length;
+    tab.nRows = total;
+    for (int c = 0; c < nCols; c++) {
+      gxp::DevCol& col = tab.cols[c];
+      // host-side concat
+      std::vector<uint8_t> data;
+      std::vector<int64_t> offsets{0};
+      std::vector<uint8_t> nulls((total + 7) / 8, 0);
+      int64_t row = 0;
+      bool hasNulls = false;
+      for (auto& ch : b.chunks) {
+        const HostCol& hc = ch[c];
+        for (int i = 0; i < hc.length; i++, row++) {
+          bool notNull = (hc.nullBitmap[i / 8] >> (i % 8)) & 1;
+          if (notNull) nulls[row / 8] |= 1 << (row % 8);
+          else hasNulls = true;
+        }
+        if (col.type == GX_TYPE_STRING) {
+          int64_t base = data.size();
+          data.insert(data.end(), hc.data.begin(), hc.data.end());
+          for (int i = 1; i <= hc.length; i++)
+            offsets.push_back(base + hc.offsets[i]);
+        } else {
+          data.insert(data.end(), hc.data.begin(), hc.data.end());
+        }
+      }
+      col.hasNulls = hasNulls ? 1 : 0;
+      col.data = devAlloc(ex, std::max<size_t>(data.size(), 1));
+      if (!col.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      HIP_OK(ex, hipMemcpy(col.data, data.data(), data.size(),
+                           hipMemcpyHostToDevice));
+      if (col.type == GX_TYPE_STRING) {
+        col.offsets = (int64_t*)devAlloc(ex, offsets.size() * 8);
+        if (!col.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        HIP_OK(ex, hipMemcpy(col.offsets, offsets.data(), offsets.size() * 8,
+                             hipMemcpyHostToDevice));
+      }
+      if (hasNulls) {
+        col.nullBitmap = (uint8_t*)devAlloc(ex, nulls.size());
+        if (!col.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        HIP_OK(ex, hipMemcpy(col.nullBitmap, nulls.data(), nulls.size(),
+                             hipMemcpyHostToDevice));
+      } else {
+        col.nullBitmap = nullptr;
+      }
+    }
+  } else {
+    ex->err = "source not bound";
+    return GX_ERR_INVALID;
+  }
+  // result/error buffers
+  ex->devTable = (gxp::GroupSlot*)devAlloc(ex, sizeof(gxp::GroupSlot) * gxp::kGlobalGroups);
+  ex->devErr = (uint32_t*)devAlloc(ex, 4);
+  ex->devSel = (uint64_t*)devAlloc(ex, 8);
+  ex->devDesc = (gxp::FusedQueryDesc*)devAlloc(ex, sizeof(gxp::FusedQueryDesc));
+  if (!ex->devTable || !ex->devErr || !ex->devSel || !ex->devDesc) {
+    ex->err = "hipMalloc failed";
+    return GX_ERR_INTERNAL;
+  }
+  ex->desc.globalTable = ex->devTable;
+  ex->desc.errorFlag = ex->devErr;
+  ex->desc.selCount = ex->devSel;
+  ex->deviceReady = true;
+  return GX_OK;
+}
+
+// ---------------- query execution (fused path) ----------------
+
+static void decodeGroupLane(gx_exec* ex, uint32_t lane, int kind, int type,
+                            OutRowVal* v) {
+  v->type = type;
+  if (lane == 0xFF000000u) {
+    v->isNull = true;
+    return;
+  }
+  if (kind == 0) {
+    int len = (int)(lane >> 24);
+    v->str.clear();
+    for (int j = 0; j < len; j++) v->str.push_back((char)((lane >> (8 * j)) & 0xFF));
+  } else {
+    v->i64 = (int64_t)lane;
+  }
+}
+
+static int32_t runFused(gx_exec* ex) {
+  int32_t rc = materializeDevice(ex);
+  if (rc) return rc;
+  HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(ex->devSel, 0, 8, ex->stream));
+  HIP_OK(ex, hipMemcpyAsync(ex->devDesc, &ex->desc, sizeof(ex->desc),
+                            hipMemcpyHostToDevice, ex->stream));
+  hipEvent_t ev0, ev1;
+  HIP_OK(ex, hipEventCreate(&ev0));
+  HIP_OK(ex, hipEventCreate(&ev1));
+  HIP_OK(ex, hipEventRecord(ev0, ex->stream));
+  int lrc = gxp::gxLaunchFusedAgg(ex->desc, ex->devDesc, ex->stream);
+  if (lrc != 0) {
+    ex->err = "fused kernel launch failed: " +
+              std::string(hipGetErrorString((hipError_t)lrc));
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipEventRecord(ev1, ex->stream));
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  {
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    ex->lastKernelMs = ms;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+  }
+  uint32_t errFlag = 0;
+  HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+  if (errFlag != 0) {
+    ex->err = "device execution error flag 0x" + std::to_string(errFlag) +
+              " (unsupported data shape or overflow)";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipMemcpy(&ex->lastSelCount, ex->devSel, 8, hipMemcpyDeviceToHost));
+  std::vector<gxp::GroupSlot> table(gxp::kGlobalGroups);
+  HIP_OK(ex, hipMemcpy(table.data(), ex->devTable,
+                       sizeof(gxp::GroupSlot) * gxp::kGlobalGroups,
+                       hipMemcpyDeviceToHost));
+  // collect occupied slots, deterministic order (by key)
+  std::vector<const gxp::GroupSlot*> occ;
+  for (auto& s : table)
+    if (s.key != gxp::kEmptyKey) occ.push_back(&s);
+  std::sort(occ.begin(), occ.end(),
+            [](const gxp::GroupSlot* a, const gxp::GroupSlot* b) {
+              return a->key < b->key;
+            });
+
+  const PNode& agg = ex->plan.nodes[ex->root];
+  bool partial = agg.aggMode == GX_AGG_MODE_PARTIAL;
+  ex->resultRows.clear();
+  // zero-row, no-group-by => one row (count=0, sums NULL)
+  if (occ.empty() && agg.exprs.empty() && !partial) {
+    gxp::GroupSlot zero{};
+    zero.key = 0;
+    std::memset(zero.accLo, 0, sizeof(zero.accLo));
+    std::memset(zero.accHi, 0, sizeof(zero.accHi));
+    std::memset(zero.cnt, 0, sizeof(zero.cnt));
+    static gxp::GroupSlot zslot;
+    zslot = zero;
+    occ.push_back(&zslot);
+  }
+  for (const gxp::GroupSlot* s : occ) {
+    std::vector<OutRowVal> row;
+    for (int k = 0; k < ex->desc.gkey.nCols; k++) {
+      OutRowVal v;
+      int srcCol = ex->desc.gkey.col[k];
+      decodeGroupLane(ex, (uint32_t)(s->key >> (32 * k)), ex->desc.gkey.kind[k],
+                      ex->desc.table.cols[srcCol].type, &v);
+      row.push_back(std::move(v));
+    }
+    for (int a = 0; a < ex->desc.nAggs; a++) {
+      const gxp::AggDesc& ad = ex->desc.aggs[a];
+      __int128 acc = ((__int128)s->accHi[a] << 64) | s->accLo[a];
+      int64_t cnt = s->cnt[a];
+      if (ad.func == GX_AGG_COUNT) {
+        OutRowVal v;
+        v.type = GX_TYPE_I64;
+        v.i64 = cnt;
+        row.push_back(std::move(v));
+      } else if (partial) {
+        OutRowVal v;
+        v.type = GX_TYPE_DECIMAL;
+        if (ad.func == GX_AGG_SUM && cnt == 0) v.isNull = true;
+        else v.dec = decFromUnits(acc, ad.scale);
+        row.push_back(std::move(v));
+        OutRowVal c;
+        c.type = GX_TYPE_I64;
+        c.i64 = cnt;
+        row.push_back(std::move(c));
+      } else if (ad.func == GX_AGG_SUM) {
+        OutRowVal v;
+        v.type = GX_TYPE_DECIMAL;
+        if (cnt == 0) v.isNull = true;
+        else {
+          v.dec = decFromUnits(acc, ad.scale);
+          int aggFrac = ex->plan.nodes[ex->root].aggFracs[a];
+          v.dec.Round(&v.dec, aggFrac, gxp::ModeHalfUp);  // func_sum.go:203-222
+        }
+        row.push_back(std::move(v));
+      } else {  // AVG finalize: DecimalDiv(+4) then Round (func_avg.go:84-109)
+        OutRowVal v;
+        v.type = GX_TYPE_DECIMAL;
+        if (cnt == 0) v.isNull = true;
+        else {
+          MyDecimal sum = decFromUnits(acc, ad.scale);
+          MyDecimal den;
+          den.FromInt(cnt);
+          MyDecimal res;
+          int32_t ec = gxp::DecimalDiv(&sum, &den, &res, gxp::kDivFracIncr);
+          if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
+            ex->err = "avg finalize failed";
+            return GX_ERR_INTERNAL;
+          }
+          int aggFrac = ex->plan.nodes[ex->root].aggFracs[a];
+          res.Round(&res, aggFrac, gxp::ModeHalfUp);
+          v.dec = res;
+        }
+        row.push_back(std::move(v));
+      }
+    }
+    ex->resultRows.push_back(std::move(row));
+  }
+  return GX_OK;
+}
+
+// ---------------- FINAL-mode host merge ----------------
+
+static int32_t runFinalHost(gx_exec* ex) {
+  const PNode& agg = ex->plan.nodes[ex->root];
+  const PNode& src = ex->plan.nodes[agg.child];
+  auto it = ex->bindings.find(agg.child);
+  if (it == ex->bindings.end() || !it->second.haveChunks) {
+    ex->err = "FINAL mode needs bound partial chunks";
+    return GX_ERR_INVALID;
+  }
+  size_t nGroup = agg.exprs.size();
+  struct FState {
+    std::vector<OutRowVal> groupVals;
+    std::vector<MyDecimal> dec;
+    std::vector<int64_t> cnt;
+  };
+  std::map<std::string, FState> groups;
+  std::vector<std::string> order;
+  for (auto& ch : it->second.chunks) {
+    int n = ch.empty() ? 0 : ch[0].length;
+    for (int i = 0; i < n; i++) {
+      // group identity: type-tagged raw values (trimmed strings per
+      // utf8mb4_bin PAD SPACE)
+      std::string key;
+      std::vector<OutRowVal> gvals;
+      for (size_t g = 0; g < nGroup; g++) {
+        const HostCol& hc = ch[g];
+        OutRowVal v;
+        v.type = hc.type;
+        bool notNull = (hc.nullBitmap[i / 8] >> (i % 8)) & 1;
+        if (!notNull) {
+          key.push_back('\0');
+          v.isNull = true;
+        } else if (hc.type == GX_TYPE_STRING) {
+          int64_t s0 = hc.offsets[i], e0 = hc.offsets[i + 1];
+          while (e0 > s0 && hc.data[e0 - 1] == ' ') e0--;
+          v.str.assign((const char*)hc.data.data() + s0, e0 - s0);
+          key.push_back('\1');
+          uint32_t l = (uint32_t)v.str.size();
+          key.append((const char*)&l, 4);
+          key.append(v.str);
+        } else {
+          std::memcpy(&v.u64, hc.data.data() + i * 8, 8);
+          v.i64 = (int64_t)v.u64;
+          key.push_back('\2');
+          key.append((const char*)&v.u64, 8);
+        }
+        gvals.push_back(std::move(v));
+      }
+      auto git = groups.find(key);
+      if (git == groups.end()) {
+        FState st;
+        st.groupVals = std::move(gvals);
+        st.dec.resize(agg.aggFuncs.size());
+        st.cnt.assign(agg.aggFuncs.size(), 0);
+        git = groups.emplace(key, std::move(st)).first;
+        order.push_back(key);
+      }
+      FState& st = git->second;
+      size_t col = nGroup;
+      for (size_t a = 0; a < agg.aggFuncs.size(); a++) {
+        int f = agg.aggFuncs[a];
+        if (f == GX_AGG_COUNT) {
+          int64_t c;
+          std::memcpy(&c, ch[col].data.data() + i * 8, 8);
+          st.cnt[a] += c;
+          col += 1;
+        } else {  // SUM/AVG: decimal + count
+          int64_t c;
+          std::memcpy(&c, ch[col + 1].data.data() + i * 8, 8);
+          if (c > 0) {
+            MyDecimal v;
+            std::memcpy(&v, ch[col].data.data() + i * 40, 40);
+            MyDecimal tmp;
+            int32_t ec = gxp::DecimalAdd(&st.dec[a], &v, &tmp);
+            if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
+              ex->err = "merge add failed";
+              return GX_ERR_INTERNAL;
+            }
+            st.dec[a] = tmp;
+            st.cnt[a] += c;
+          }
+          col += 2;
+        }
+      }
+    }
+  }
+  ex->resultRows.clear();
+  for (const std::string& key : order) {
+    FState& st = groups[key];
+    std::vector<OutRowVal> row = st.groupVals;
+    for (size_t a = 0; a < agg.aggFuncs.size(); a++) {
+      int f = agg.aggFuncs[a];
+      OutRowVal v;
+      if (f == GX_AGG_COUNT) {
+        v.type = GX_TYPE_I64;
+        v.i64 = st.cnt[a];
+      } else if (f == GX_AGG_SUM) {
+        v.type = GX_TYPE_DECIMAL;
+        if (st.cnt[a] == 0) v.isNull = true;
+        else {
+          v.dec = st.dec[a];
+          v.dec.Round(&v.dec, agg.aggFracs[a], gxp::ModeHalfUp);
+        }
+      } else {  // AVG
+        v.type = GX_TYPE_DECIMAL;
+        if (st.cnt[a] == 0) v.isNull = true;
+        else {
+          MyDecimal den;
+          den.FromInt(st.cnt[a]);
+          MyDecimal res;
+          int32_t ec = gxp::DecimalDiv(&st.dec[a], &den, &res, gxp::kDivFracIncr);
+          if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
+            ex->err = "avg finalize failed";
+            return GX_ERR_INTERNAL;
+          }
+          res.Round(&res, agg.aggFracs[a], gxp::ModeHalfUp);
+          v.dec = res;
+        }
+      }
+      row.push_back(std::move(v));
+    }
+    ex->resultRows.push_back(std::move(row));
+  }
+  return GX_OK;
+}
+
+// ---------------- bare source download (generator parity) ----------------
+
+static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
+  int32_t rc = materializeDevice(ex);
+  if (rc) return rc;
+  gxp::DevTable& tab = ex->desc.table;
+  int64_t remaining = tab.nRows - ex->srcPos;
+  int n = (int)std::min<int64_t>(remaining, 1024);
+  if (n <= 0) {
+    *rows_out = 0;
+    return GX_OK;
+  }
+  if (out->n_cols != tab.nCols) {
+    ex->err = "output chunk column count mismatch";
+    return GX_ERR_INVALID;
+  }
+  for (int c = 0; c < tab.nCols; c++) {
+    gxp::DevCol& col = tab.cols[c];
+    gx_col* g = &out->cols[c];
+    if (col.type == GX_TYPE_STRING) {
+      std::vector<int64_t> offs(n + 1);
+      HIP_OK(ex, hipMemcpy(offs.data(), col.offsets + ex->srcPos, (n + 1) * 8,
+                           hipMemcpyDeviceToHost));
+      int64_t base = offs[0];
+      int64_t bytes = offs[n] - base;
+      if (g->offsets_cap < n + 1 || g->data_cap < bytes) {
+        ex->err = "output buffer too small";
+        return GX_ERR_INVALID;
+      }
+      for (int i = 0; i <= n; i++) g->offsets[i] = offs[i] - base;
+      HIP_OK(ex, hipMemcpy(g->data, (uint8_t*)col.data + base, bytes,
+                           hipMemcpyDeviceToHost));
+    } else {
+      int64_t bytes = (int64_t)n * col.elemSize;
+      if (g->data_cap < bytes) {
+        ex->err = "output buffer too small";
+        return GX_ERR_INVALID;
+      }
+      HIP_OK(ex, hipMemcpy(g->data, (uint8_t*)col.data + ex->srcPos * col.elemSize,
+                           bytes, hipMemcpyDeviceToHost));
+    }
+    if (g->null_bitmap) std::memset(g->null_bitmap, 0xFF, (n + 7) / 8);
+    g->length = n;
+  }
+  out->n_rows = n;
+  ex->srcPos += n;
+  *rows_out = n;
+  return GX_OK;
+}
+
+// ---------------- result marshalling ----------------
+
+static int32_t emitResultRows(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
+  size_t n = std::min<size_t>(ex->resultRows.size() - ex->emitPos, 1024);
+  if (n == 0) {
+    *rows_out = 0;
+    return GX_OK;
+  }
+  std::vector<int64_t> dataPos(out->n_cols, 0);
+  for (int c = 0; c < out->n_cols; c++) {
+    if (out->cols[c].null_bitmap)
+      std::memset(out->cols[c].null_bitmap, 0, ((int)n + 7) / 8);
+    if (out->cols[c].offsets) out->cols[c].offsets[0] = 0;
+  }
+  for (size_t i = 0; i < n; i++) {
+    const std::vector<OutRowVal>& row = ex->resultRows[ex->emitPos + i];
+    if ((int)row.size() != out->n_cols) {
+      ex->err = "output chunk column count mismatch";
+      return GX_ERR_INVALID;
+    }
+    for (int c = 0; c < out->n_cols; c++) {
+      const OutRowVal& v = row[c];
+      gx_col* g = &out->cols[c];
+      bool isStr = g->elem_size == -1 || v.type == GX_TYPE_STRING;
+      if (v.isNull) {
+        if (isStr) g->offsets[i + 1] = g->offsets[i];
+        else {
+          int es = v.type == GX_TYPE_DECIMAL ? 40 : 8;
+          if ((int64_t)(i + 1) * es > g->data_cap) {
+            ex->err = "output buffer too small";
+            return GX_ERR_INVALID;
+          }
+          std::memset((uint8_t*)g->data + i * es, 0, es);
+        }
+        continue;
+      }
+      if (g->null_bitmap) g->null_bitmap[i / 8] |= 1 << (i % 8);
+      if (isStr) {
+        int64_t base = g->offsets[i];
+        if (base + (int64_t)v.str.size() > g->data_cap || (int64_t)i + 1 > g->offsets_cap - 1) {
+          ex->err = "output buffer too small";
+          return GX_ERR_INVALID;
+        }
+        std::memcpy((uint8_t*)g->data + base, v.str.data(), v.str.size());
+        g->offsets[i + 1] = base + v.str.size();
+      } else if (v.type == GX_TYPE_DECIMAL) {
+        if ((int64_t)(i + 1) * 40 > g->data_cap) {
+          ex->err = "output buffer too small";
+          return GX_ERR_INVALID;
+        }
+        std::memcpy((uint8_t*)g->data + i * 40, &v.dec, 40);
+      } else {
+        if ((int64_t)(i + 1) * 8 > g->data_cap) {
+          ex->err = "output buffer too small";
+          return GX_ERR_INVALID;
+        }
+        if (v.type == GX_TYPE_F64)
+          std::memcpy((uint8_t*)g->data + i * 8, &v.f64, 8);
+        else
+          std::memcpy((uint8_t*)g->data + i * 8, &v.i64, 8);
+      }
+    }
+  }
+  for (int c = 0; c < out->n_cols; c++) out->cols[c].length = (int)n;
+  out->n_rows = (int)n;
+  ex->emitPos += n;
+  *rows_out = (int)n;
+  return GX_OK;
+}
+
+// ---------------- C ABI ----------------
+
+extern "C" {
+
+gx_pb* gx_pb_new(void) { return new gx_pb(); }
+void gx_pb_free(gx_pb* pb) { delete pb; }
+
+static int32_t addExpr(gx_pb* pb, PExpr e) {
+  pb->plan.exprs.push_back(std::move(e));
+  return (int32_t)pb->plan.exprs.size() - 1;
+}
+
+int32_t gx_pb_colref(gx_pb* pb, int32_t col_idx, int32_t type, int32_t frac) {
+  PExpr e;
+  e.kind = EK_COLREF;
+  e.colIdx = col_idx;
+  e.retType = type;
+  e.retFrac = frac;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_i64(gx_pb* pb, int64_t v) {
+  PExpr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_I64;
+  e.constI64 = v;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_f64(gx_pb* pb, double v) {
+  PExpr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_F64;
+  e.constF64 = v;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_time(gx_pb* pb, uint64_t v) {
+  PExpr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_TIME;
+  e.constTime = v;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_dec(gx_pb* pb, const uint8_t dec40[40]) {
+  PExpr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_DECIMAL;
+  std::memcpy(&e.constDec, dec40, 40);
+  e.retFrac = e.constDec.resultFrac;
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_const_str(gx_pb* pb, const char* s, int32_t len) {
+  PExpr e;
+  e.kind = EK_CONST;
+  e.retType = GX_TYPE_STRING;
+  e.constStr.assign(s, len);
+  return addExpr(pb, std::move(e));
+}
+int32_t gx_pb_call(gx_pb* pb, int32_t func, int32_t ret_type, int32_t ret_frac,
+                   const int32_t* args, int32_t n_args) {
+  PExpr e;
+  e.kind = EK_CALL;
+  e.func = func;
+  e.retType = ret_type;
+  e.retFrac = ret_frac < 0 ? 0 : ret_frac;
+  for (int i = 0; i < n_args; i++) {
+    if (args[i] < 0 || args[i] >= (int32_t)pb->plan.exprs.size())
+      return GX_ERR_INVALID;
+    e.args.push_back(args[i]);
+  }
+  return addExpr(pb, std::move(e));
+}
+
+static int32_t addNode(gx_pb* pb, PNode n) {
+  pb->plan.nodes.push_back(std::move(n));
+  return (int32_t)pb->plan.nodes.size() - 1;
+}
+
+int32_t gx_pb_source(gx_pb* pb, const int32_t* col_types, const int32_t* col_fracs,
+                     int32_t n_cols) {
+  PNode n;
+  n.kind = PK_SOURCE;
+  for (int i = 0; i < n_cols; i++) {
+    n.colTypes.push_back(col_types[i]);
+    n.colFracs.push_back(col_fracs ? col_fracs[i] : 0);
+  }
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_selection(gx_pb* pb, int32_t child, const int32_t* conds,
+                        int32_t n_conds) {
+  PNode n;
+  n.kind = PK_SELECTION;
+  n.child = child;
+  for (int i = 0; i < n_conds; i++) n.exprs.push_back(conds[i]);
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_projection(gx_pb* pb, int32_t child, const int32_t* exprs,
+                         int32_t n_exprs) {
+  PNode n;
+  n.kind = PK_PROJECTION;
+  n.child = child;
+  for (int i = 0; i < n_exprs; i++) n.exprs.push_back(exprs[i]);
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_hashagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
+                      int32_t n_group, const int32_t* agg_funcs,
+                      const int32_t* agg_args, const int32_t* agg_fracs,
+                      int32_t n_aggs, int32_t mode) {
+  PNode n;
+  n.kind = PK_HASHAGG;
+  n.child = child;
+  n.aggMode = mode;
+  for (int i = 0; i < n_group; i++) n.exprs.push_back(group_exprs[i]);
+  for (int i = 0; i < n_aggs; i++) {
+    n.aggFuncs.push_back(agg_funcs[i]);
+    n.aggArgs.push_back(agg_args[i]);
+    n.aggFracs.push_back(agg_fracs ? agg_fracs[i] : 0);
+  }
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
+                   const uint8_t* key_desc, int32_t n_keys, int64_t limit,
+                   int64_t offset) {
+  PNode n;
+  n.kind = PK_TOPN;
+  n.child = child;
+  for (int i = 0; i < n_keys; i++) {
+    n.exprs.push_back(key_exprs[i]);
+    n.keyDesc.push_back(key_desc ? key_desc[i] : 0);
+  }
+  n.limit = limit;
+  n.offset = offset;
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
+                       const int32_t* build_keys, const int32_t* probe_keys,
+                       int32_t n_keys, int32_t join_type) {
+  PNode n;
+  n.kind = PK_HASHJOIN;
+  n.child = build_child;
+  n.child2 = probe_child;
+  n.joinType = join_type;
+  for (int i = 0; i < n_keys; i++) {
+    n.buildKeys.push_back(build_keys[i]);
+    n.probeKeys.push_back(probe_keys[i]);
+  }
+  return addNode(pb, std::move(n));
+}
+
+gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
+  if (!pb || root < 0 || root >= (int32_t)pb->plan.nodes.size()) return nullptr;
+  auto* ex = new gx_exec();
+  ex->plan = pb->plan;
+  ex->root = root;
+  ex->device = device;
+  const PNode& rn = ex->plan.nodes[root];
+  if (rn.kind == PK_HASHAGG && rn.aggMode == GX_AGG_MODE_FINAL) {
+    if (ex->plan.nodes[rn.child].kind != PK_SOURCE) {
+      ex->err = "FINAL agg child must be a bound source";
+      return ex;
+    }
+    ex->isFinalHost = true;
+    ex->sourceNode = rn.child;
+  } else if (rn.kind == PK_HASHAGG) {
+    int32_t rc = compileFused(ex);
+    if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
+    (void)rc;
+  } else if (rn.kind == PK_SOURCE) {
+    ex->isBareSource = true;
+    ex->sourceNode = root;
+  } else {
+    ex->err = "unsupported root plan node for the device engine this round";
+  }
+  return ex;
+}
+
+int32_t gx_bind_chunks(gx_exec* ex, int32_t source_node, const gx_chunk* chunks,
+                       int32_t n_chunks) {
+  if (!ex || source_node < 0 || source_node >= (int32_t)ex->plan.nodes.size())
+    return GX_ERR_INVALID;
+  const PNode& node = ex->plan.nodes[source_node];
+  if (node.kind != PK_SOURCE) return GX_ERR_INVALID;
+  Binding b;
+  b.haveChunks = true;
+  for (int i = 0; i < n_chunks; i++) {
+    std::vector<HostCol> cols;
+    if (chunks[i].n_cols != (int32_t)node.colTypes.size()) return GX_ERR_INVALID;
+    for (int j = 0; j < chunks[i].n_cols; j++) {
+      const gx_col& g = chunks[i].cols[j];
+      HostCol hc;
+      hc.type = node.colTypes[j];
+      hc.frac = node.colFracs[j];
+      hc.length = g.length;
+      int nb = (g.length + 7) / 8;
+      if (g.null_bitmap) hc.nullBitmap.assign(g.null_bitmap, g.null_bitmap + nb);
+      else hc.nullBitmap.assign(nb, 0xFF);
+      if (hc.type == GX_TYPE_STRING) {
+        hc.offsets.assign(g.offsets, g.offsets + g.length + 1);
+        hc.data.assign((uint8_t*)g.data, (uint8_t*)g.data + hc.offsets.back());
+      } else {
+        int es = hc.type == GX_TYPE_DECIMAL ? 40 : 8;
+        hc.data.assign((uint8_t*)g.data, (uint8_t*)g.data + (size_t)g.length * es);
+      }
+      cols.push_back(std::move(hc));
+    }
+    b.chunks.push_back(std::move(cols));
+  }
+  ex->bindings[source_node] = std::move(b);
+  return GX_OK;
+}
+
+int32_t gx_bind_tpch_sharded(gx_exec* ex, int32_t source_node, int32_t table,
+                             int64_t n_rows, uint64_t seed, int64_t row_offset,
+                             int64_t total_rows) {
+  if (!ex || source_node < 0 || source_node >= (int32_t)ex->plan.nodes.size())
+    return GX_ERR_INVALID;
+  if (ex->plan.nodes[source_node].kind != PK_SOURCE) return GX_ERR_INVALID;
+  Binding b;
+  b.tpchTable = table;
+  b.tpchRows = n_rows;
+  b.tpchSeed = seed;
+  b.tpchRowOffset = row_offset;
+  b.tpchTotalRows = total_rows;
+  ex->bindings[source_node] = std::move(b);
+  return GX_OK;
+}
+
+int32_t gx_bind_tpch(gx_exec* ex, int32_t source_node, int32_t table,
+                     int64_t n_rows, uint64_t seed, int64_t row_offset) {
+  return gx_bind_tpch_sharded(ex, source_node, table, n_rows, seed, row_offset,
+                              n_rows);
+}
+
+int32_t gx_open(gx_exec* ex) {
+  if (!ex) return GX_ERR_INVALID;
+  if (!ex->err.empty()) return GX_ERR_INVALID;
+  if (!ex->isFused && !ex->isFinalHost && !ex->isBareSource) {
+    ex->err = "plan not executable";
+    return GX_ERR_INVALID;
+  }
+  ex->ranQuery = false;
+  ex->emitPos = 0;
+  ex->srcPos = 0;
+  ex->resultRows.clear();
+  ex->opened = true;
+  return GX_OK;
+}
+
+int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
+  if (!ex || !ex->opened) return GX_ERR_INVALID;
+  if (ex->isBareSource) return emitSourceChunk(ex, out, rows_out);
+  if (!ex->ranQuery) {
+    int32_t rc = ex->isFinalHost ? runFinalHost(ex) : runFused(ex);
+    if (rc) {
+      *rows_out = 0;
+      return rc;
+    }
+    ex->ranQuery = true;
+  }
+  return emitResultRows(ex, out, rows_out);
+}
+
+int32_t gx_close(gx_exec* ex) {
+  if (!ex) return GX_ERR_INVALID;
+  ex->opened = false;
+  return GX_OK;
+}
+
+void gx_exec_free(gx_exec* ex) { delete ex; }
+
+const char* gx_last_error(gx_exec* ex) {
+  if (!ex) return "null exec";
+  return ex->err.c_str();
+}
+
+double gx_last_kernel_ms(gx_exec* ex) { return ex ? ex->lastKernelMs : 0; }
+int64_t gx_last_sel_count(gx_exec* ex) {
+  return ex ? (int64_t)ex->lastSelCount : 0;
+}
+
+int32_t gx_engine_is_gpu(void) { return 1; }
+const char* gx_engine_name(void) { return "gxexec-mi355x"; }
+
+}  // extern "C"

@@ -1,0 +1,526 @@
+// tidb_amd/csrc/gx_kernels.hip — MI355X (gfx950/CDNA4) kernels for the
+// TiDB analytical hot path.
+//
+// Fused scan -> filter -> project -> hash-aggregate in ONE pass over the
+// columns (the MI355X-native form of SelectionExec -> ProjectionExec ->
+// HashAggExec, pkg/executor/select.go:750 / projection.go:77 /
+// aggregate/agg_hash_*.go): every consumed input column is read once from
+// HBM — the path is HBM-bandwidth-bound (no MFMA shape here).
+//
+// Grouping: per-workgroup LDS hash table (LDS-staged accumulator state,
+// atomically updated), flushed once per workgroup into a global table with
+// device-scope atomics — the device analog of the reference's partial/final
+// worker split (agg_hash_executor.go:54-92).
+//
+// Decimal arithmetic: fixed-point int64/int128 units at static scales; exact,
+// with overflow detection -> error flag (never silent). Equivalence to the
+// word-based MyDecimal arithmetic is covered by tests/golden + parity suites.
+#include <hip/hip_runtime.h>
+
+#include "gx_common.h"
+
+namespace gxp {
+
+// ------------------------------------------------------------------
+// shared PRNG/date spec — MUST generate bit-identical data to the CPU
+// restatement in oracle/tpch.cpp (parity-tested).
+// ------------------------------------------------------------------
+__host__ __device__ inline uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ULL;
+  uint64_t z = x;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+  return z ^ (z >> 31);
+}
+__host__ __device__ inline uint64_t fieldRand(uint64_t seed, int64_t row, int field) {
+  uint64_t h = splitmix64(seed ^ (0x9E3779B97F4A7C15ULL * (uint64_t)(row + 1)));
+  return splitmix64(h ^ (0xBF58476D1CE4E5B9ULL * (uint64_t)(field + 1)));
+}
+__host__ __device__ inline int64_t daysFromCivil(int y, int m, int d) {
+  y -= m <= 2;
+  int64_t era = (y >= 0 ? y : y - 399) / 400;
+  unsigned yoe = (unsigned)(y - era * 400);
+  unsigned doy = (unsigned)((153 * (m + (m > 2 ? -3 : 9)) + 2) / 5 + d - 1);
+  unsigned doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
+  return era * 146097 + (int64_t)doe - 719468;
+}
+__host__ __device__ inline void civilFromDays(int64_t z, int* yy, int* mm, int* dd) {
+  z += 719468;
+  int64_t era = (z >= 0 ? z : z - 146096) / 146097;
+  unsigned doe = (unsigned)(z - era * 146097);
+  unsigned yoe = (doe - doe / 1460 + doe / 36524 - doe / 146096) / 365;
+  int64_t y = (int64_t)yoe + era * 400;
+  unsigned doy = doe - (365 * yoe + yoe / 4 - yoe / 100);
+  unsigned mp = (5 * doy + 2) / 153;
+  unsigned d = doy - (153 * mp + 2) / 5 + 1;
+  unsigned m = mp < 10 ? mp + 3 : mp - 9;
+  *yy = (int)(y + (m <= 2));
+  *mm = (int)m;
+  *dd = (int)d;
+}
+
+// packed CoreTime DATE (time.go:235-251,266: fspTt 0b1110 marks TypeDate)
+__host__ __device__ inline uint64_t timeFromDate(int y, int m, int d) {
+  return ((uint64_t)y << 50) | ((uint64_t)m << 46) | ((uint64_t)d << 41) | 0xEULL;
+}
+
+#define GX_EPOCH_1992 728659L       // daysFromCivil(1992,1,1)
+#define GX_SHIPDATE_DAYS 2527L      // ..1998-12-01 inclusive
+#define GX_ORDERDATE_DAYS 2406L     // ..1998-08-02 inclusive
+
+// canonical cents -> 40-byte MyDecimal (scale 2), matches oracle/tpch.cpp
+__device__ inline void storeDecCents(uint8_t* p, int64_t cents) {
+  int64_t ip = cents / 100;
+  uint32_t f = (uint32_t)(cents % 100);
+  int digits = 1;
+  for (int64_t t = ip; t >= 10; t /= 10) digits++;
+  uint32_t hdr = (uint32_t)(uint8_t)digits | (2u << 8) | (2u << 16);
+  uint32_t* w = (uint32_t*)p;
+  w[0] = hdr;
+  w[1] = (uint32_t)ip;
+  w[2] = f * 10000000u;
+  w[3] = 0; w[4] = 0; w[5] = 0; w[6] = 0; w[7] = 0; w[8] = 0; w[9] = 0;
+}
+
+// ------------------------------------------------------------------
+// lineitem generator: cols 0..7 =
+//   orderkey i64, quantity dec, extendedprice dec, discount dec, tax dec,
+//   returnflag char(1), linestatus char(1), shipdate date
+// ------------------------------------------------------------------
+__global__ void genLineitemKernel(DevTable tab, int64_t rowBegin, int64_t nRows,
+                                  uint64_t seed, int64_t totalRows) {
+  int64_t nOrders = totalRows / 4;
+  if (nOrders < 1) nOrders = 1;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nRows;
+       i += stride) {
+    int64_t row = rowBegin + i;
+    ((int64_t*)tab.cols[0].data)[i] =
+        1 + (int64_t)(fieldRand(seed, row, 0) % (uint64_t)nOrders);
+    storeDecCents((uint8_t*)tab.cols[1].data + i * 40,
+                  (1 + (int64_t)(fieldRand(seed, row, 1) % 50)) * 100);
+    storeDecCents((uint8_t*)tab.cols[2].data + i * 40,
+                  90100 + (int64_t)(fieldRand(seed, row, 2) % (10495000 - 90100 + 1)));
+    storeDecCents((uint8_t*)tab.cols[3].data + i * 40,
+                  (int64_t)(fieldRand(seed, row, 3) % 11));
+    storeDecCents((uint8_t*)tab.cols[4].data + i * 40,
+                  (int64_t)(fieldRand(seed, row, 4) % 9));
+    const char rf[3] = {'A', 'N', 'R'};
+    ((uint8_t*)tab.cols[5].data)[i] = rf[fieldRand(seed, row, 5) % 3];
+    tab.cols[5].offsets[i + 1] = i + 1;
+    const char ls[2] = {'O', 'F'};
+    ((uint8_t*)tab.cols[6].data)[i] = ls[fieldRand(seed, row, 6) % 2];
+    tab.cols[6].offsets[i + 1] = i + 1;
+    int y, m, d;
+    civilFromDays(GX_EPOCH_1992 +
+                      (int64_t)(fieldRand(seed, row, 7) % (uint64_t)GX_SHIPDATE_DAYS),
+                  &y, &m, &d);
+    ((uint64_t*)tab.cols[7].data)[i] = timeFromDate(y, m, d);
+    if (i == 0) {
+      tab.cols[5].offsets[0] = 0;
+      tab.cols[6].offsets[0] = 0;
+    }
+  }
+}
+
+// ------------------------------------------------------------------
+// fused filter+project+aggregate
+// ------------------------------------------------------------------
+
+struct Int128 {
+  uint64_t lo;
+  int64_t hi;
+};
+
+__device__ inline Int128 i128FromI64(int64_t v) {
+  return {(uint64_t)v, v < 0 ? -1 : 0};
+}
+__device__ inline Int128 i128Add(Int128 a, Int128 b) {
+  uint64_t lo = a.lo + b.lo;
+  int64_t carry = lo < a.lo;
+  return {lo, a.hi + b.hi + carry};
+}
+__device__ inline Int128 i128Neg(Int128 a) {
+  uint64_t lo = ~a.lo + 1;
+  int64_t hi = ~a.hi + (lo == 0);
+  return {lo, hi};
+}
+__device__ inline Int128 i128Sub(Int128 a, Int128 b) { return i128Add(a, i128Neg(b)); }
+// a(int128) * b(int64) with overflow detection
+__device__ inline Int128 i128MulI64(Int128 a, int64_t b, bool* ovf) {
+  __int128 x = ((__int128)a.hi << 64) | a.lo;
+  bool neg = false;
+  if (x < 0) { x = -x; neg = true; }
+  if (b < 0) { b = -b; neg = !neg; }
+  unsigned __int128 ux = (unsigned __int128)x;
+  unsigned __int128 ub = (unsigned __int128)b;
+  // overflow check: ux > UMAX/ub
+  if (ub != 0 && ux > (~(unsigned __int128)0 >> 1) / ub) { *ovf = true; return {0, 0}; }
+  unsigned __int128 r = ux * ub;
+  __int128 sr = (__int128)r;
+  if (neg) sr = -sr;
+  return {(uint64_t)sr, (int64_t)(sr >> 64)};
+}
+__device__ inline Int128 i128Scale10(Int128 a, int pow, bool* ovf) {
+  static const int64_t p10[19] = {1, 10, 100, 1000, 10000, 100000, 1000000,
+                                  10000000, 100000000, 1000000000, 10000000000LL,
+                                  100000000000LL, 1000000000000LL, 10000000000000LL,
+                                  100000000000000LL, 1000000000000000LL,
+                                  10000000000000000LL, 100000000000000000LL,
+                                  1000000000000000000LL};
+  return i128MulI64(a, p10[pow], ovf);
+}
+
+// parse a 40-byte MyDecimal with digitsInt <= 18, digitsFrac <= 9 into units
+// at scale = digitsFrac. Wider decimals set the error flag.
+__device__ inline bool loadDecimalUnits(const uint8_t* p, Int128* out,
+                                        int* scale, uint32_t* err) {
+  const uint32_t* w = (const uint32_t*)p;
+  uint32_t hdr = w[0];
+  int digitsInt = (int)(int8_t)(hdr & 0xFF);
+  int digitsFrac = (int)(int8_t)((hdr >> 8) & 0xFF);
+  bool neg = ((hdr >> 24) & 0xFF) != 0;
+  if (digitsInt > 18 || digitsFrac > 9 || digitsInt < 0 || digitsFrac < 0) {
+    atomicOr(err, 1u);
+    return false;
+  }
+  int wordsInt = (digitsInt + 8) / 9;
+  if (digitsInt == 0) wordsInt = 0;
+  int64_t ip = 0;
+  if (wordsInt == 1) ip = (int32_t)w[1];
+  else if (wordsInt == 2) ip = (int64_t)(int32_t)w[1] * 1000000000 + (int32_t)w[2];
+  int64_t fr = 0;
+  if (digitsFrac > 0) {
+    static const int32_t p10s[10] = {1, 10, 100, 1000, 10000, 100000, 1000000,
+                                     10000000, 100000000, 1000000000};
+    fr = (int32_t)w[1 + wordsInt] / p10s[9 - digitsFrac];
+  }
+  static const int64_t p10l[10] = {1, 10, 100, 1000, 10000, 100000, 1000000,
+                                   10000000, 100000000, 1000000000};
+  __int128 units = (__int128)ip * p10l[digitsFrac] + fr;
+  if (neg) units = -units;
+  *out = {(uint64_t)units, (int64_t)(units >> 64)};
+  *scale = digitsFrac;
+  return true;
+}
+
+__device__ inline bool colIsNull(const DevCol& c, int64_t row) {
+  if (!c.hasNulls || c.nullBitmap == nullptr) return false;
+  return ((c.nullBitmap[row >> 3] >> (row & 7)) & 1) == 0;
+}
+
+__device__ inline int cmpResult(int c, int op) {
+  switch (op) {
+    case 0: return c < 0;   // LT
+    case 1: return c <= 0;  // LE
+    case 2: return c > 0;   // GT
+    case 3: return c >= 0;  // GE
+    case 4: return c == 0;  // EQ
+    default: return c != 0; // NE
+  }
+}
+
+// one VM register
+struct Reg {
+  Int128 v;
+  bool null;
+};
+
+// pack the group key (see GroupKeyDesc comment)
+__device__ inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
+                                    uint64_t* keyOut, uint32_t* err) {
+  uint64_t key = 0;
+  for (int k = 0; k < d.gkey.nCols; k++) {
+    const DevCol& c = d.table.cols[d.gkey.col[k]];
+    uint32_t lane;
+    if (colIsNull(c, row)) {
+      lane = 0xFF000000u;
+    } else if (d.gkey.kind[k] == 0) {
+      int64_t s = c.offsets[row], e = c.offsets[row + 1];
+      // utf8mb4_bin PAD SPACE: trim trailing spaces (collate.go:272)
+      const uint8_t* p = (const uint8_t*)c.data;
+      while (e > s && p[e - 1] == ' ') e--;
+      int64_t len = e - s;
+      if (len > 3) { atomicOr(err, 2u); return false; }
+      lane = (uint32_t)len << 24;
+      for (int64_t j = 0; j < len; j++) lane |= (uint32_t)p[s + j] << (8 * j);
+    } else {
+      int64_t v = ((const int64_t*)c.data)[row];
+      if (v < 0 || v > 0x7FFFFFFF) { atomicOr(err, 2u); return false; }
+      lane = (uint32_t)v;
+    }
+    key |= (uint64_t)lane << (32 * k);
+  }
+  if (d.gkey.nCols == 0) key = 0;
+  if (key == kEmptyKey) key = kEmptyKey - 1;  // avoid the sentinel
+  *keyOut = key;
+  return true;
+}
+
+// atomic int128 + count accumulation into a slot (LDS or global — the
+// atomicAdd overloads resolve per address space)
+template <typename SlotT>
+__device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
+  if (v.lo != 0 || v.hi != 0) {
+    uint64_t old = atomicAdd((unsigned long long*)&slot->accLo[a],
+                             (unsigned long long)v.lo);
+    uint64_t carry = (old + v.lo) < old ? 1 : 0;
+    int64_t hiAdd = v.hi + (int64_t)carry;
+    if (hiAdd != 0)
+      atomicAdd((unsigned long long*)&slot->accHi[a], (unsigned long long)hiAdd);
+  }
+  if (dc != 0)
+    atomicAdd((unsigned long long*)&slot->cnt[a], (unsigned long long)dc);
+}
+
+__launch_bounds__(256)
+__global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
+  const FusedQueryDesc& d = *dp;
+  bool failed = false;
+  __shared__ GroupSlot lds[kLdsGroups];
+  for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
+    lds[i].key = kEmptyKey;
+    for (int a = 0; a < kMaxAggs; a++) {
+      lds[i].accLo[a] = 0;
+      lds[i].accHi[a] = 0;
+      lds[i].cnt[a] = 0;
+    }
+  }
+  __syncthreads();
+
+  int64_t n = d.table.nRows;
+  int64_t per = (n + gridDim.x - 1) / gridDim.x;
+  int64_t begin = (int64_t)blockIdx.x * per;
+  int64_t end = begin + per;
+  if (end > n) end = n;
+  uint64_t mySel = 0;
+
+  for (int64_t row = begin + threadIdx.x; row < end; row += blockDim.x) {
+    // ---- filter (CNF; NULL rejects — expression.go:507 toBool) ----
+    bool pass = true;
+    for (int p = 0; p < d.nPreds && pass; p++) {
+      const PredDesc& pd = d.preds[p];
+      const DevCol& c = d.table.cols[pd.col];
+      if (colIsNull(c, row)) { pass = false; break; }
+      if (pd.kind == PRED_TIME_CMP_CONST) {
+        uint64_t v = ((const uint64_t*)c.data)[row] & ~0xFULL;
+        uint64_t k = pd.constU64 & ~0xFULL;
+        int cmp = v < k ? -1 : (v > k ? 1 : 0);
+        pass = cmpResult(cmp, pd.cmp);
+      } else if (pd.kind == PRED_I64_CMP_CONST) {
+        int64_t v = ((const int64_t*)c.data)[row];
+        int64_t k = (int64_t)pd.constU64;
+        int cmp = v < k ? -1 : (v > k ? 1 : 0);
+        pass = cmpResult(cmp, pd.cmp);
+      } else {  // PRED_DEC_CMP_CONST: engine aligned const to column scale
+        Int128 u;
+        int sc;
+        if (!loadDecimalUnits((const uint8_t*)c.data + row * 40, &u, &sc,
+                              d.errorFlag)) {
+          failed = true;
+          break;
+        }
+        __int128 v = ((__int128)u.hi << 64) | u.lo;
+        __int128 k = ((__int128)(int64_t)pd.constU64);
+        int cmp = v < k ? -1 : (v > k ? 1 : 0);
+        pass = cmpResult(cmp, pd.cmp);
+      }
+    }
+    if (failed) break;
+    if (!pass) continue;
+    mySel++;
+
+    // ---- projection / agg-arg VM ----
+    Reg regs[kMaxVmRegs];
+    bool bad = false;
+    (void)bad;
+    for (int i = 0; i < d.nIns && !bad; i++) {
+      const VmIns& ins = d.ins[i];
+      switch (ins.op) {
+        case VM_LOAD_DEC: {
+          const DevCol& c = d.table.cols[ins.a];
+          Reg r;
+          r.null = colIsNull(c, row);
+          r.v = {0, 0};
+          if (!r.null) {
+            int sc;
+            if (!loadDecimalUnits((const uint8_t*)c.data + row * 40, &r.v, &sc,
+                                  d.errorFlag)) {
+              bad = true;
+              break;
+            }
+            if (sc != ins.b) {  // engine encoded expected scale in b
+              bool ovf = false;
+              if (sc < ins.b) r.v = i128Scale10(r.v, ins.b - sc, &ovf);
+              else { atomicOr(d.errorFlag, 4u); bad = true; break; }
+              if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; break; }
+            }
+          }
+          regs[ins.dst] = r;
+          break;
+        }
+        case VM_LOAD_I64: {
+          const DevCol& c = d.table.cols[ins.a];
+          Reg r;
+          r.null = colIsNull(c, row);
+          r.v = r.null ? Int128{0, 0} : i128FromI64(((const int64_t*)c.data)[row]);
+          regs[ins.dst] = r;
+          break;
+        }
+        case VM_LOAD_CONST:
+          regs[ins.dst].v = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
+          regs[ins.dst].null = false;
+          break;
+        case VM_ADD:
+          regs[ins.dst].null = regs[ins.a].null || regs[ins.b].null;
+          regs[ins.dst].v = i128Add(regs[ins.a].v, regs[ins.b].v);
+          break;
+        case VM_SUB:
+          regs[ins.dst].null = regs[ins.a].null || regs[ins.b].null;
+          regs[ins.dst].v = i128Sub(regs[ins.a].v, regs[ins.b].v);
+          break;
+        case VM_MUL: {
+          Reg& ra = regs[ins.a];
+          Reg& rb = regs[ins.b];
+          Reg r;
+          r.null = ra.null || rb.null;
+          r.v = {0, 0};
+          if (!r.null) {
+            // rb must fit in i64 (engine orders operands)
+            bool ovf = false;
+            int64_t b64 = (int64_t)rb.v.lo;
+            bool fits = (rb.v.hi == 0 && b64 >= 0) || (rb.v.hi == -1 && b64 < 0);
+            if (!fits) { atomicOr(d.errorFlag, 8u); bad = true; break; }
+            r.v = i128MulI64(ra.v, b64, &ovf);
+            if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; break; }
+          }
+          regs[ins.dst] = r;
+          break;
+        }
+        case VM_SCALE_UP: {
+          bool ovf = false;
+          regs[ins.dst].null = regs[ins.a].null;
+          regs[ins.dst].v = i128Scale10(regs[ins.a].v, ins.b, &ovf);
+          if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; }
+          break;
+        }
+      }
+    }
+
+    if (bad) { failed = true; break; }
+
+    // ---- group lookup / insert in LDS ----
+    uint64_t key;
+    if (!makeGroupKey(d, row, &key, d.errorFlag)) { failed = true; break; }
+    uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
+    bool slotOk = true;
+    for (int probe = 0;; probe++) {
+      if (probe >= kLdsGroups) { atomicOr(d.errorFlag, 16u); slotOk = false; break; }
+      uint64_t cur = lds[slot].key;
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & (kLdsGroups - 1);
+    }
+    if (!slotOk) { failed = true; break; }
+
+    // ---- update states ----
+    for (int a = 0; a < d.nAggs; a++) {
+      const AggDesc& ad = d.aggs[a];
+      if (ad.func == 0 /*COUNT*/) {
+        bool isNull = ad.srcReg >= 0 && regs[ad.srcReg].null;
+        if (!isNull) accumInto(&lds[slot], a, Int128{0, 0}, 1);
+      } else {  // SUM / AVG
+        if (ad.srcReg >= 0 && !regs[ad.srcReg].null)
+          accumInto(&lds[slot], a, regs[ad.srcReg].v, 1);
+      }
+    }
+  }
+
+  if (d.selCount) {
+    // wave-level reduce then one atomic per wave (G12)
+    uint64_t total = mySel;
+    for (int off = 32; off > 0; off >>= 1)
+      total += __shfl_down(total, off, 64);
+    if ((threadIdx.x & 63) == 0 && total)
+      atomicAdd((unsigned long long*)d.selCount, (unsigned long long)total);
+  }
+  __syncthreads();
+
+  // ---- flush LDS table into the global table ----
+  for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
+    if (lds[i].key == kEmptyKey) continue;
+    uint64_t key = lds[i].key;
+    uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
+    bool ok = true;
+    for (int probe = 0;; probe++) {
+      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, 32u); ok = false; break; }
+      uint64_t cur = d.globalTable[slot].key;
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & (kGlobalGroups - 1);
+    }
+    if (!ok) continue;
+    for (int a = 0; a < d.nAggs; a++) {
+      Int128 v = {lds[i].accLo[a], lds[i].accHi[a]};
+      accumInto(&d.globalTable[slot], a, v, lds[i].cnt[a]);
+    }
+  }
+}
+
+__global__ void initGlobalTableKernel(GroupSlot* table, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  table[i].key = kEmptyKey;
+  for (int a = 0; a < kMaxAggs; a++) {
+    table[i].accLo[a] = 0;
+    table[i].accHi[a] = 0;
+    table[i].cnt[a] = 0;
+  }
+}
+
+// ---------------- host launch wrappers ----------------
+
+int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows,
+                    uint64_t seed, int64_t totalRows, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int block = 256;
+  int grid = (int)((nRows + block - 1) / block);
+  if (grid > 4096) grid = 4096;
+  if (grid < 1) grid = 1;
+  if (table == 0) {
+    hipLaunchKernelGGL(genLineitemKernel, dim3(grid), dim3(block), 0, s, *devTab,
+                       rowBegin, nRows, seed, totalRows);
+  } else {
+    return -1;  // orders/customer device generation lands with the Q3 path
+  }
+  return (int)hipGetLastError();
+}
+
+int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
+                     void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int grid = (int)((desc.table.nRows + 255) / 256);
+  // >> 256 workgroups to fill 256 CUs / 8 XCDs; cap and grid-stride per block
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(initGlobalTableKernel, dim3((kGlobalGroups + 255) / 256),
+                     dim3(256), 0, s, desc.globalTable, kGlobalGroups);
+  hipLaunchKernelGGL(fusedAggKernel, dim3(grid), dim3(256), 0, s, devDesc);
+  return (int)hipGetLastError();
+}
+
+int gxLaunchMemset(void* p, int v, size_t n, void* stream) {
+  return (int)hipMemsetAsync(p, v, n, (hipStream_t)stream);
+}
+
+}  // namespace gxp

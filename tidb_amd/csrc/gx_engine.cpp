@@ -13,6 +13,8 @@
 // of canonical partial states (KB-sized; MergePartialResult semantics,
 // aggfuncs.go:250-255).
 #include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
 
 #include <algorithm>
 #include <cstring>
@@ -704,6 +706,14 @@ static int32_t runFused(gx_exec* ex) {
     return GX_ERR_INTERNAL;
   }
   HIP_OK(ex, hipMemcpy(&ex->lastSelCount, ex->devSel, 8, hipMemcpyDeviceToHost));
+  if (getenv("GX_DEBUG")) {
+    fprintf(stderr,
+            "[gx] nRows=%lld nPreds=%d nIns=%d nAggs=%d gkeyCols=%d sel=%llu "
+            "kms=%.3f\n",
+            (long long)ex->desc.table.nRows, ex->desc.nPreds, ex->desc.nIns,
+            ex->desc.nAggs, ex->desc.gkey.nCols,
+            (unsigned long long)ex->lastSelCount, ex->lastKernelMs);
+  }
   std::vector<gxp::GroupSlot> table(gxp::kGlobalGroups);
   HIP_OK(ex, hipMemcpy(table.data(), ex->devTable,
                        sizeof(gxp::GroupSlot) * gxp::kGlobalGroups,
@@ -716,6 +726,8 @@ static int32_t runFused(gx_exec* ex) {
             [](const gxp::GroupSlot* a, const gxp::GroupSlot* b) {
               return a->key < b->key;
             });
+  if (getenv("GX_DEBUG"))
+    fprintf(stderr, "[gx] occupied group slots: %zu\n", occ.size());
 
   const PNode& agg = ex->plan.nodes[ex->root];
   bool partial = agg.aggMode == GX_AGG_MODE_PARTIAL;
@@ -1210,6 +1222,9 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
   } else if (rn.kind == PK_SOURCE) {
     ex->isBareSource = true;
     ex->sourceNode = root;
+    ex->desc.table.nCols = (int)rn.colTypes.size();
+    for (size_t c = 0; c < rn.colTypes.size(); c++)
+      setDevColMeta(&ex->desc.table.cols[c], rn.colTypes[c], rn.colFracs[c]);
   } else {
     ex->err = "unsupported root plan node for the device engine this round";
   }

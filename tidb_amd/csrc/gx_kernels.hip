@@ -64,9 +64,10 @@ __host__ __device__ inline uint64_t timeFromDate(int y, int m, int d) {
   return ((uint64_t)y << 50) | ((uint64_t)m << 46) | ((uint64_t)d << 41) | 0xEULL;
 }
 
-#define GX_EPOCH_1992 728659L       // daysFromCivil(1992,1,1)
-#define GX_SHIPDATE_DAYS 2527L      // ..1998-12-01 inclusive
-#define GX_ORDERDATE_DAYS 2406L     // ..1998-08-02 inclusive
+// date-range constants are computed (constant-folded), never hand-written
+#define GX_EPOCH_1992 daysFromCivil(1992, 1, 1)
+#define GX_SHIPDATE_DAYS (daysFromCivil(1998, 12, 1) - daysFromCivil(1992, 1, 1) + 1)
+#define GX_ORDERDATE_DAYS (daysFromCivil(1998, 8, 2) - daysFromCivil(1992, 1, 1) + 1)
 
 // canonical cents -> 40-byte MyDecimal (scale 2), matches oracle/tpch.cpp
 __device__ inline void storeDecCents(uint8_t* p, int64_t cents) {

@@ -54,7 +54,7 @@ struct VmIns {
 };
 
 constexpr int kMaxVmIns = 48;
-constexpr int kMaxVmRegs = 16;
+constexpr int kMaxVmRegs = 12;
 constexpr int kMaxVmConsts = 16;
 
 // ---- filter (CNF of simple predicates; Q1-class) ----

@@ -176,7 +176,11 @@ __device__ inline Int128 i128Scale10(Int128 a, int pow, bool* ovf) {
 // at scale = digitsFrac. Wider decimals set the error flag.
 __device__ inline bool loadDecimalUnits(const uint8_t* p, Int128* out,
                                         int* scale, uint32_t* err) {
-  const uint32_t* w = (const uint32_t*)p;
+  // 40-byte stride keeps rows 8-byte aligned: two dwordx2 loads cover
+  // header + wordBuf[0..2] (enough for digitsInt <= 18, digitsFrac <= 9)
+  uint2 lo = *(const uint2*)p;
+  uint2 hi = *(const uint2*)(p + 8);
+  uint32_t w[4] = {lo.x, lo.y, hi.x, hi.y};
   uint32_t hdr = w[0];
   int digitsInt = (int)(int8_t)(hdr & 0xFF);
   int digitsFrac = (int)(int8_t)((hdr >> 8) & 0xFF);
@@ -221,10 +225,31 @@ __device__ inline int cmpResult(int c, int op) {
   }
 }
 
-// one VM register
-struct Reg {
-  Int128 v;
-  bool null;
+// VM state: named registers (runtime-indexed arrays would spill to scratch —
+// the instruction stream is wave-uniform, so the switches below lower to
+// cheap scalar branches)
+struct VmState {
+  Int128 r0, r1, r2, r3, r4, r5, r6, r7, r8, r9, r10, r11;
+  uint32_t nullBits;
+  __device__ Int128 get(int i) const {
+    switch (i) {
+      case 0: return r0; case 1: return r1; case 2: return r2; case 3: return r3;
+      case 4: return r4; case 5: return r5; case 6: return r6; case 7: return r7;
+      case 8: return r8; case 9: return r9; case 10: return r10; default: return r11;
+    }
+  }
+  __device__ void set(int i, Int128 v) {
+    switch (i) {
+      case 0: r0 = v; break; case 1: r1 = v; break; case 2: r2 = v; break;
+      case 3: r3 = v; break; case 4: r4 = v; break; case 5: r5 = v; break;
+      case 6: r6 = v; break; case 7: r7 = v; break; case 8: r8 = v; break;
+      case 9: r9 = v; break; case 10: r10 = v; break; default: r11 = v; break;
+    }
+  }
+  __device__ bool isNull(int i) const { return (nullBits >> i) & 1; }
+  __device__ void setNull(int i, bool n) {
+    nullBits = (nullBits & ~(1u << i)) | ((uint32_t)n << i);
+  }
 };
 
 // pack the group key (see GroupKeyDesc comment)
@@ -332,76 +357,74 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     mySel++;
 
     // ---- projection / agg-arg VM ----
-    Reg regs[kMaxVmRegs];
+    VmState vm;
+    vm.nullBits = 0;
     bool bad = false;
-    (void)bad;
     for (int i = 0; i < d.nIns && !bad; i++) {
       const VmIns& ins = d.ins[i];
       switch (ins.op) {
         case VM_LOAD_DEC: {
           const DevCol& c = d.table.cols[ins.a];
-          Reg r;
-          r.null = colIsNull(c, row);
-          r.v = {0, 0};
-          if (!r.null) {
+          bool nul = colIsNull(c, row);
+          Int128 v = {0, 0};
+          if (!nul) {
             int sc;
-            if (!loadDecimalUnits((const uint8_t*)c.data + row * 40, &r.v, &sc,
+            if (!loadDecimalUnits((const uint8_t*)c.data + row * 40, &v, &sc,
                                   d.errorFlag)) {
               bad = true;
               break;
             }
             if (sc != ins.b) {  // engine encoded expected scale in b
               bool ovf = false;
-              if (sc < ins.b) r.v = i128Scale10(r.v, ins.b - sc, &ovf);
+              if (sc < ins.b) v = i128Scale10(v, ins.b - sc, &ovf);
               else { atomicOr(d.errorFlag, 4u); bad = true; break; }
               if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; break; }
             }
           }
-          regs[ins.dst] = r;
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
           break;
         }
         case VM_LOAD_I64: {
           const DevCol& c = d.table.cols[ins.a];
-          Reg r;
-          r.null = colIsNull(c, row);
-          r.v = r.null ? Int128{0, 0} : i128FromI64(((const int64_t*)c.data)[row]);
-          regs[ins.dst] = r;
+          bool nul = colIsNull(c, row);
+          vm.set(ins.dst, nul ? Int128{0, 0}
+                              : i128FromI64(((const int64_t*)c.data)[row]));
+          vm.setNull(ins.dst, nul);
           break;
         }
         case VM_LOAD_CONST:
-          regs[ins.dst].v = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
-          regs[ins.dst].null = false;
+          vm.set(ins.dst, {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]});
+          vm.setNull(ins.dst, false);
           break;
         case VM_ADD:
-          regs[ins.dst].null = regs[ins.a].null || regs[ins.b].null;
-          regs[ins.dst].v = i128Add(regs[ins.a].v, regs[ins.b].v);
+          vm.set(ins.dst, i128Add(vm.get(ins.a), vm.get(ins.b)));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         case VM_SUB:
-          regs[ins.dst].null = regs[ins.a].null || regs[ins.b].null;
-          regs[ins.dst].v = i128Sub(regs[ins.a].v, regs[ins.b].v);
+          vm.set(ins.dst, i128Sub(vm.get(ins.a), vm.get(ins.b)));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         case VM_MUL: {
-          Reg& ra = regs[ins.a];
-          Reg& rb = regs[ins.b];
-          Reg r;
-          r.null = ra.null || rb.null;
-          r.v = {0, 0};
-          if (!r.null) {
-            // rb must fit in i64 (engine orders operands)
+          bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
+          Int128 v = {0, 0};
+          if (!nul) {
+            Int128 rb = vm.get(ins.b);
             bool ovf = false;
-            int64_t b64 = (int64_t)rb.v.lo;
-            bool fits = (rb.v.hi == 0 && b64 >= 0) || (rb.v.hi == -1 && b64 < 0);
+            int64_t b64 = (int64_t)rb.lo;
+            bool fits = (rb.hi == 0 && b64 >= 0) || (rb.hi == -1 && b64 < 0);
             if (!fits) { atomicOr(d.errorFlag, 8u); bad = true; break; }
-            r.v = i128MulI64(ra.v, b64, &ovf);
+            v = i128MulI64(vm.get(ins.a), b64, &ovf);
             if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; break; }
           }
-          regs[ins.dst] = r;
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
           break;
         }
         case VM_SCALE_UP: {
           bool ovf = false;
-          regs[ins.dst].null = regs[ins.a].null;
-          regs[ins.dst].v = i128Scale10(regs[ins.a].v, ins.b, &ovf);
+          vm.set(ins.dst, i128Scale10(vm.get(ins.a), ins.b, &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a));
           if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; }
           break;
         }
@@ -433,11 +456,11 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     for (int a = 0; a < d.nAggs; a++) {
       const AggDesc& ad = d.aggs[a];
       if (ad.func == 0 /*COUNT*/) {
-        bool isNull = ad.srcReg >= 0 && regs[ad.srcReg].null;
+        bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
         if (!isNull) accumInto(&lds[slot], a, Int128{0, 0}, 1);
       } else {  // SUM / AVG
-        if (ad.srcReg >= 0 && !regs[ad.srcReg].null)
-          accumInto(&lds[slot], a, regs[ad.srcReg].v, 1);
+        if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
+          accumInto(&lds[slot], a, vm.get(ad.srcReg), 1);
       }
     }
   }

@@ -147,18 +147,26 @@ __device__ inline Int128 i128Neg(Int128 a) {
   return {lo, hi};
 }
 __device__ inline Int128 i128Sub(Int128 a, Int128 b) { return i128Add(a, i128Neg(b)); }
-// a(int128) * b(int64) with overflow detection
+// a(int128) * b(int64) with overflow detection. No division: the check is
+// on the 192-bit partial products (a 128-bit divide is a slow software loop
+// on gfx950 and would dominate the row cost).
 __device__ inline Int128 i128MulI64(Int128 a, int64_t b, bool* ovf) {
-  __int128 x = ((__int128)a.hi << 64) | a.lo;
   bool neg = false;
-  if (x < 0) { x = -x; neg = true; }
-  if (b < 0) { b = -b; neg = !neg; }
-  unsigned __int128 ux = (unsigned __int128)x;
-  unsigned __int128 ub = (unsigned __int128)b;
-  // overflow check: ux > UMAX/ub
-  if (ub != 0 && ux > (~(unsigned __int128)0 >> 1) / ub) { *ovf = true; return {0, 0}; }
-  unsigned __int128 r = ux * ub;
-  __int128 sr = (__int128)r;
+  __int128 x = ((__int128)a.hi << 64) | a.lo;
+  unsigned __int128 ua;
+  if (x < 0) { ua = (unsigned __int128)(-x); neg = true; }
+  else ua = (unsigned __int128)x;
+  uint64_t ub;
+  if (b < 0) { ub = (uint64_t)(-b); neg = !neg; }
+  else ub = (uint64_t)b;
+  uint64_t alo = (uint64_t)ua;
+  uint64_t ahi = (uint64_t)(ua >> 64);
+  unsigned __int128 plo = (unsigned __int128)alo * ub;
+  unsigned __int128 mid = (plo >> 64) + (unsigned __int128)ahi * ub;
+  if (mid >> 63) { *ovf = true; return {0, 0}; }  // |result| >= 2^127
+  uint64_t rlo = (uint64_t)plo;
+  uint64_t rhi = (uint64_t)mid;
+  __int128 sr = ((__int128)(int64_t)rhi << 64) | rlo;
   if (neg) sr = -sr;
   return {(uint64_t)sr, (int64_t)(sr >> 64)};
 }

@@ -127,6 +127,12 @@ struct FusedQueryDesc {
   GroupSlot* globalTable = nullptr;  // kGlobalGroups slots
   uint32_t* errorFlag = nullptr;     // != 0 => abort with error
   uint64_t* selCount = nullptr;      // rows passing the filter (stats)
+  // perf ablation (GX_ABLATE env; results are WRONG when nonzero — timing
+  // experiments only): 1 = skip LDS agg update, 2 = skip VM+agg (filter only)
+  int32_t ablate = 0;
+  // 0 = narrow int64 VM (fast path); 1 = wide int128 VM. The narrow kernel
+  // reports kErrRetryWide on overflow and the engine relaunches wide.
+  int32_t wide = 0;
 };
 
 // host-side launch wrappers (defined in gx_kernels.hip)

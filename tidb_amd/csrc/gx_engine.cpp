@@ -297,6 +297,8 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       int ci = d.nConsts++;
       d.constLo[ci] = (int64_t)(uint64_t)u;
       d.constHi[ci] = (int64_t)(u >> 64);
+      if (u > (__int128)INT64_MAX || u < (__int128)INT64_MIN)
+        ex->desc.wide = 1;  // narrow VM cannot hold this constant
       reg = emit(gxp::VM_LOAD_CONST, allocReg(), ci, 0);
       *scaleOut = sc;
       break;
@@ -675,6 +677,7 @@ static void decodeGroupLane(gx_exec* ex, uint32_t lane, int kind, int type,
 static int32_t runFused(gx_exec* ex) {
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;
+  ex->desc.ablate = getenv("GX_ABLATE") ? atoi(getenv("GX_ABLATE")) : 0;
   HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
   HIP_OK(ex, hipMemsetAsync(ex->devSel, 0, 8, ex->stream));
   HIP_OK(ex, hipMemcpyAsync(ex->devDesc, &ex->desc, sizeof(ex->desc),
@@ -700,6 +703,16 @@ static int32_t runFused(gx_exec* ex) {
   }
   uint32_t errFlag = 0;
   HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+  if (errFlag == 256u /*kErrRetryWide only*/ && !ex->desc.wide) {
+    // a value overflowed the int64 fast path: rerun with the int128 VM
+    // (sticky for this executor)
+    ex->desc.wide = 1;
+    if (getenv("GX_DEBUG")) fprintf(stderr, "[gx] narrow overflow -> wide retry\n");
+    return runFused(ex);
+  }
+  errFlag &= ~256u;
+  if (ex->desc.ablate != 0 && getenv("GX_DEBUG"))
+    fprintf(stderr, "[gx] ABLATE=%d kms=%.3f\n", ex->desc.ablate, ex->lastKernelMs);
   if (errFlag != 0) {
     ex->err = "device execution error flag 0x" + std::to_string(errFlag) +
               " (unsupported data shape or overflow)";

@@ -127,6 +127,20 @@ __global__ void genLineitemKernel(DevTable tab, int64_t rowBegin, int64_t nRows,
 // ------------------------------------------------------------------
 // fused filter+project+aggregate
 // ------------------------------------------------------------------
+//
+// Two instantiations: NARROW (int64 VM registers — covers every value with
+// <= 18 significant digits, i.e. all real decimal(15,2)-class workloads) and
+// WIDE (int128). NARROW detects overflow per operation and sets the
+// RETRY_WIDE flag; the engine then relaunches the WIDE variant — exactness
+// is never traded away.
+
+constexpr uint32_t kErrBadDecimal = 1u;
+constexpr uint32_t kErrBadKey = 2u;
+constexpr uint32_t kErrScale = 4u;
+constexpr uint32_t kErrOverflow = 8u;
+constexpr uint32_t kErrLdsFull = 16u;
+constexpr uint32_t kErrGlobalFull = 32u;
+constexpr uint32_t kErrRetryWide = 256u;  // narrow VM overflowed; not an error
 
 struct Int128 {
   uint64_t lo;
@@ -147,9 +161,8 @@ __device__ inline Int128 i128Neg(Int128 a) {
   return {lo, hi};
 }
 __device__ inline Int128 i128Sub(Int128 a, Int128 b) { return i128Add(a, i128Neg(b)); }
-// a(int128) * b(int64) with overflow detection. No division: the check is
-// on the 192-bit partial products (a 128-bit divide is a slow software loop
-// on gfx950 and would dominate the row cost).
+// a(int128) * b(int64) with multiply-only overflow detection (a 128-bit
+// divide is a slow software loop on gfx950).
 __device__ inline Int128 i128MulI64(Int128 a, int64_t b, bool* ovf) {
   bool neg = false;
   __int128 x = ((__int128)a.hi << 64) | a.lo;
@@ -170,49 +183,122 @@ __device__ inline Int128 i128MulI64(Int128 a, int64_t b, bool* ovf) {
   if (neg) sr = -sr;
   return {(uint64_t)sr, (int64_t)(sr >> 64)};
 }
-__device__ inline Int128 i128Scale10(Int128 a, int pow, bool* ovf) {
-  static const int64_t p10[19] = {1, 10, 100, 1000, 10000, 100000, 1000000,
-                                  10000000, 100000000, 1000000000, 10000000000LL,
-                                  100000000000LL, 1000000000000LL, 10000000000000LL,
-                                  100000000000000LL, 1000000000000000LL,
-                                  10000000000000000LL, 100000000000000000LL,
-                                  1000000000000000000LL};
-  return i128MulI64(a, p10[pow], ovf);
+
+__device__ __constant__ const int64_t kP10[19] = {
+    1, 10, 100, 1000, 10000, 100000, 1000000, 10000000, 100000000, 1000000000,
+    10000000000LL, 100000000000LL, 1000000000000LL, 10000000000000LL,
+    100000000000000LL, 1000000000000000LL, 10000000000000000LL,
+    100000000000000000LL, 1000000000000000000LL};
+
+// magic reciprocals for n / 10^k, exact for 0 <= n < 2^31:
+// q = (n * kMagic[k]) >> 62  with kMagic[k] = ceil(2^62 / 10^k)
+__device__ __constant__ const uint64_t kDivMagic[10] = {
+    4611686018427387904ULL,  // 10^0
+    461168601842738791ULL,   // 10^1
+    46116860184273880ULL,    // 10^2
+    4611686018427388ULL,     // 10^3
+    461168601842739ULL,      // 10^4
+    46116860184274ULL,       // 10^5
+    4611686018428ULL,        // 10^6
+    461168601843ULL,         // 10^7
+    46116860185ULL,          // 10^8
+    4611686019ULL,           // 10^9
+};
+__device__ inline int64_t divP10(uint32_t n, int k) {
+  return (int64_t)(((unsigned __int128)n * kDivMagic[k]) >> 62);
 }
 
-// parse a 40-byte MyDecimal with digitsInt <= 18, digitsFrac <= 9 into units
-// at scale = digitsFrac. Wider decimals set the error flag.
-__device__ inline bool loadDecimalUnits(const uint8_t* p, Int128* out,
+// ---- value type abstraction (NARROW = int64, WIDE = Int128) ----
+template <bool WIDE>
+struct VT;
+
+template <>
+struct VT<false> {
+  using T = int64_t;
+  static __device__ T fromI64(int64_t v, bool*) { return v; }
+  static __device__ T add(T a, T b, bool* ovf) {
+    T r;
+    *ovf |= __builtin_add_overflow(a, b, &r);
+    return r;
+  }
+  static __device__ T sub(T a, T b, bool* ovf) {
+    T r;
+    *ovf |= __builtin_sub_overflow(a, b, &r);
+    return r;
+  }
+  static __device__ T mul(T a, T b, bool* ovf) {
+    T r;
+    *ovf |= __builtin_mul_overflow(a, b, &r);
+    return r;
+  }
+  static __device__ T scale10(T a, int k, bool* ovf) { return mul(a, kP10[k], ovf); }
+  static __device__ T zero() { return 0; }
+  static __device__ Int128 toAcc(T v) { return i128FromI64(v); }
+  static __device__ int cmp(T a, T b) { return a < b ? -1 : (a > b ? 1 : 0); }
+};
+
+template <>
+struct VT<true> {
+  using T = Int128;
+  static __device__ T fromI64(int64_t v, bool*) { return i128FromI64(v); }
+  static __device__ T add(T a, T b, bool*) { return i128Add(a, b); }  // 127-bit headroom
+  static __device__ T sub(T a, T b, bool*) { return i128Sub(a, b); }
+  static __device__ T mul(T a, T b, bool* ovf) {
+    bool fits = (b.hi == 0 && (int64_t)b.lo >= 0) || (b.hi == -1 && (int64_t)b.lo < 0);
+    if (!fits) { *ovf = true; return {0, 0}; }
+    return i128MulI64(a, (int64_t)b.lo, ovf);
+  }
+  static __device__ T scale10(T a, int k, bool* ovf) { return i128MulI64(a, kP10[k], ovf); }
+  static __device__ T zero() { return {0, 0}; }
+  static __device__ Int128 toAcc(T v) { return v; }
+  static __device__ int cmp(T a, T b) {
+    __int128 x = ((__int128)a.hi << 64) | a.lo;
+    __int128 y = ((__int128)b.hi << 64) | b.lo;
+    return x < y ? -1 : (x > y ? 1 : 0);
+  }
+};
+
+// parse a 40-byte MyDecimal (digitsInt <= 18, digitsFrac <= 9) into units at
+// scale = digitsFrac. Returns false on malformed input (sets kErrBadDecimal)
+// or narrow overflow (sets kErrRetryWide).
+template <bool WIDE>
+__device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* out,
                                         int* scale, uint32_t* err) {
   // 40-byte stride keeps rows 8-byte aligned: two dwordx2 loads cover
-  // header + wordBuf[0..2] (enough for digitsInt <= 18, digitsFrac <= 9)
-  uint2 lo = *(const uint2*)p;
-  uint2 hi = *(const uint2*)(p + 8);
-  uint32_t w[4] = {lo.x, lo.y, hi.x, hi.y};
-  uint32_t hdr = w[0];
+  // header + wordBuf[0..2]
+  uint2 lo2 = *(const uint2*)p;
+  uint2 hi2 = *(const uint2*)(p + 8);
+  uint32_t hdr = lo2.x;
   int digitsInt = (int)(int8_t)(hdr & 0xFF);
   int digitsFrac = (int)(int8_t)((hdr >> 8) & 0xFF);
   bool neg = ((hdr >> 24) & 0xFF) != 0;
   if (digitsInt > 18 || digitsFrac > 9 || digitsInt < 0 || digitsFrac < 0) {
-    atomicOr(err, 1u);
+    atomicOr(err, kErrBadDecimal);
     return false;
   }
   int wordsInt = (digitsInt + 8) / 9;
   if (digitsInt == 0) wordsInt = 0;
   int64_t ip = 0;
-  if (wordsInt == 1) ip = (int32_t)w[1];
-  else if (wordsInt == 2) ip = (int64_t)(int32_t)w[1] * 1000000000 + (int32_t)w[2];
-  int64_t fr = 0;
-  if (digitsFrac > 0) {
-    static const int32_t p10s[10] = {1, 10, 100, 1000, 10000, 100000, 1000000,
-                                     10000000, 100000000, 1000000000};
-    fr = (int32_t)w[1 + wordsInt] / p10s[9 - digitsFrac];
+  if (wordsInt == 1) ip = (int32_t)lo2.y;
+  else if (wordsInt == 2) ip = (int64_t)(int32_t)lo2.y * 1000000000 + (int32_t)hi2.x;
+  uint32_t fw = wordsInt == 0 ? lo2.y : (wordsInt == 1 ? hi2.x : hi2.y);
+  int64_t fr = digitsFrac > 0 ? divP10(fw, 9 - digitsFrac) : 0;
+  if (WIDE) {
+    __int128 units = (__int128)ip * kP10[digitsFrac] + fr;
+    if (neg) units = -units;
+    Int128 u = {(uint64_t)units, (int64_t)(units >> 64)};
+    *out = *(typename VT<WIDE>::T*)&u;
+  } else {
+    int64_t units;
+    bool ovf = __builtin_mul_overflow(ip, kP10[digitsFrac], &units);
+    ovf |= __builtin_add_overflow(units, fr, &units);
+    if (ovf) {
+      atomicOr(err, kErrRetryWide);
+      return false;
+    }
+    if (neg) units = -units;
+    *out = *(typename VT<WIDE>::T*)&units;
   }
-  static const int64_t p10l[10] = {1, 10, 100, 1000, 10000, 100000, 1000000,
-                                   10000000, 100000000, 1000000000};
-  __int128 units = (__int128)ip * p10l[digitsFrac] + fr;
-  if (neg) units = -units;
-  *out = {(uint64_t)units, (int64_t)(units >> 64)};
   *scale = digitsFrac;
   return true;
 }
@@ -233,20 +319,22 @@ __device__ inline int cmpResult(int c, int op) {
   }
 }
 
-// VM state: named registers (runtime-indexed arrays would spill to scratch —
-// the instruction stream is wave-uniform, so the switches below lower to
-// cheap scalar branches)
+// VM state: named registers (runtime-indexed arrays would spill to scratch;
+// the instruction stream is wave-uniform so these switches are cheap scalar
+// branches)
+template <bool WIDE>
 struct VmState {
-  Int128 r0, r1, r2, r3, r4, r5, r6, r7, r8, r9, r10, r11;
+  using T = typename VT<WIDE>::T;
+  T r0, r1, r2, r3, r4, r5, r6, r7, r8, r9, r10, r11;
   uint32_t nullBits;
-  __device__ Int128 get(int i) const {
+  __device__ T get(int i) const {
     switch (i) {
       case 0: return r0; case 1: return r1; case 2: return r2; case 3: return r3;
       case 4: return r4; case 5: return r5; case 6: return r6; case 7: return r7;
       case 8: return r8; case 9: return r9; case 10: return r10; default: return r11;
     }
   }
-  __device__ void set(int i, Int128 v) {
+  __device__ void set(int i, T v) {
     switch (i) {
       case 0: r0 = v; break; case 1: r1 = v; break; case 2: r2 = v; break;
       case 3: r3 = v; break; case 4: r4 = v; break; case 5: r5 = v; break;
@@ -275,12 +363,12 @@ __device__ inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
       const uint8_t* p = (const uint8_t*)c.data;
       while (e > s && p[e - 1] == ' ') e--;
       int64_t len = e - s;
-      if (len > 3) { atomicOr(err, 2u); return false; }
+      if (len > 3) { atomicOr(err, kErrBadKey); return false; }
       lane = (uint32_t)len << 24;
       for (int64_t j = 0; j < len; j++) lane |= (uint32_t)p[s + j] << (8 * j);
     } else {
       int64_t v = ((const int64_t*)c.data)[row];
-      if (v < 0 || v > 0x7FFFFFFF) { atomicOr(err, 2u); return false; }
+      if (v < 0 || v > 0x7FFFFFFF) { atomicOr(err, kErrBadKey); return false; }
       lane = (uint32_t)v;
     }
     key |= (uint64_t)lane << (32 * k);
@@ -291,8 +379,7 @@ __device__ inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
   return true;
 }
 
-// atomic int128 + count accumulation into a slot (LDS or global — the
-// atomicAdd overloads resolve per address space)
+// atomic int128 + count accumulation into a slot (LDS or global)
 template <typename SlotT>
 __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
   if (v.lo != 0 || v.hi != 0) {
@@ -307,6 +394,7 @@ __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
     atomicAdd((unsigned long long*)&slot->cnt[a], (unsigned long long)dc);
 }
 
+template <bool WIDE>
 __launch_bounds__(256)
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
@@ -347,46 +435,44 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
         int cmp = v < k ? -1 : (v > k ? 1 : 0);
         pass = cmpResult(cmp, pd.cmp);
       } else {  // PRED_DEC_CMP_CONST: engine aligned const to column scale
-        Int128 u;
+        typename VT<WIDE>::T u;
         int sc;
-        if (!loadDecimalUnits((const uint8_t*)c.data + row * 40, &u, &sc,
-                              d.errorFlag)) {
+        if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &u, &sc,
+                                    d.errorFlag)) {
           failed = true;
           break;
         }
-        __int128 v = ((__int128)u.hi << 64) | u.lo;
-        __int128 k = ((__int128)(int64_t)pd.constU64);
-        int cmp = v < k ? -1 : (v > k ? 1 : 0);
+        int cmp = VT<WIDE>::cmp(u, VT<WIDE>::fromI64((int64_t)pd.constU64, nullptr));
         pass = cmpResult(cmp, pd.cmp);
       }
     }
     if (failed) break;
     if (!pass) continue;
     mySel++;
+    if (d.ablate == 2) continue;  // timing ablation: filter only
 
     // ---- projection / agg-arg VM ----
-    VmState vm;
+    VmState<WIDE> vm;
     vm.nullBits = 0;
     bool bad = false;
+    bool ovf = false;
     for (int i = 0; i < d.nIns && !bad; i++) {
       const VmIns& ins = d.ins[i];
       switch (ins.op) {
         case VM_LOAD_DEC: {
           const DevCol& c = d.table.cols[ins.a];
           bool nul = colIsNull(c, row);
-          Int128 v = {0, 0};
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
           if (!nul) {
             int sc;
-            if (!loadDecimalUnits((const uint8_t*)c.data + row * 40, &v, &sc,
-                                  d.errorFlag)) {
+            if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &v,
+                                        &sc, d.errorFlag)) {
               bad = true;
               break;
             }
             if (sc != ins.b) {  // engine encoded expected scale in b
-              bool ovf = false;
-              if (sc < ins.b) v = i128Scale10(v, ins.b - sc, &ovf);
-              else { atomicOr(d.errorFlag, 4u); bad = true; break; }
-              if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; break; }
+              if (sc < ins.b) v = VT<WIDE>::scale10(v, ins.b - sc, &ovf);
+              else { atomicOr(d.errorFlag, kErrScale); bad = true; break; }
             }
           }
           vm.set(ins.dst, v);
@@ -396,50 +482,60 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
         case VM_LOAD_I64: {
           const DevCol& c = d.table.cols[ins.a];
           bool nul = colIsNull(c, row);
-          vm.set(ins.dst, nul ? Int128{0, 0}
-                              : i128FromI64(((const int64_t*)c.data)[row]));
+          vm.set(ins.dst, nul ? VT<WIDE>::zero()
+                              : VT<WIDE>::fromI64(((const int64_t*)c.data)[row],
+                                                  &ovf));
           vm.setNull(ins.dst, nul);
           break;
         }
-        case VM_LOAD_CONST:
-          vm.set(ins.dst, {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]});
+        case VM_LOAD_CONST: {
+          if (WIDE) {
+            Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
+            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+          } else {
+            // engine guarantees narrow-mode consts fit i64 (else it forces WIDE)
+            int64_t cv = d.constLo[ins.a];
+            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+          }
           vm.setNull(ins.dst, false);
           break;
+        }
         case VM_ADD:
-          vm.set(ins.dst, i128Add(vm.get(ins.a), vm.get(ins.b)));
+          vm.set(ins.dst, VT<WIDE>::add(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         case VM_SUB:
-          vm.set(ins.dst, i128Sub(vm.get(ins.a), vm.get(ins.b)));
+          vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         case VM_MUL: {
           bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
-          Int128 v = {0, 0};
-          if (!nul) {
-            Int128 rb = vm.get(ins.b);
-            bool ovf = false;
-            int64_t b64 = (int64_t)rb.lo;
-            bool fits = (rb.hi == 0 && b64 >= 0) || (rb.hi == -1 && b64 < 0);
-            if (!fits) { atomicOr(d.errorFlag, 8u); bad = true; break; }
-            v = i128MulI64(vm.get(ins.a), b64, &ovf);
-            if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; break; }
-          }
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
+          if (!nul) v = VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf);
           vm.set(ins.dst, v);
           vm.setNull(ins.dst, nul);
           break;
         }
-        case VM_SCALE_UP: {
-          bool ovf = false;
-          vm.set(ins.dst, i128Scale10(vm.get(ins.a), ins.b, &ovf));
+        case VM_SCALE_UP:
+          vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a));
-          if (ovf) { atomicOr(d.errorFlag, 8u); bad = true; }
           break;
-        }
       }
     }
-
+    if (ovf) {
+      atomicOr(d.errorFlag, WIDE ? kErrOverflow : kErrRetryWide);
+      failed = true;
+      break;
+    }
     if (bad) { failed = true; break; }
+
+    if (d.ablate == 1) {  // timing ablation: keep VM results live, skip agg
+      uint64_t sink = 0;
+      for (int a = 0; a < d.nAggs; a++)
+        if (d.aggs[a].srcReg >= 0) sink ^= (uint64_t)VT<WIDE>::toAcc(vm.get(d.aggs[a].srcReg)).lo;
+      asm volatile("" ::"v"(sink));
+      continue;
+    }
 
     // ---- group lookup / insert in LDS ----
     uint64_t key;
@@ -447,7 +543,7 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
     bool slotOk = true;
     for (int probe = 0;; probe++) {
-      if (probe >= kLdsGroups) { atomicOr(d.errorFlag, 16u); slotOk = false; break; }
+      if (probe >= kLdsGroups) { atomicOr(d.errorFlag, kErrLdsFull); slotOk = false; break; }
       uint64_t cur = lds[slot].key;
       if (cur == key) break;
       if (cur == kEmptyKey) {
@@ -468,7 +564,7 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
         if (!isNull) accumInto(&lds[slot], a, Int128{0, 0}, 1);
       } else {  // SUM / AVG
         if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
-          accumInto(&lds[slot], a, vm.get(ad.srcReg), 1);
+          accumInto(&lds[slot], a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
       }
     }
   }
@@ -490,7 +586,7 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
     bool ok = true;
     for (int probe = 0;; probe++) {
-      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, 32u); ok = false; break; }
+      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
       uint64_t cur = d.globalTable[slot].key;
       if (cur == key) break;
       if (cur == kEmptyKey) {
@@ -547,7 +643,10 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL(initGlobalTableKernel, dim3((kGlobalGroups + 255) / 256),
                      dim3(256), 0, s, desc.globalTable, kGlobalGroups);
-  hipLaunchKernelGGL(fusedAggKernel, dim3(grid), dim3(256), 0, s, devDesc);
+  if (desc.wide)
+    hipLaunchKernelGGL(fusedAggKernel<true>, dim3(grid), dim3(256), 0, s, devDesc);
+  else
+    hipLaunchKernelGGL(fusedAggKernel<false>, dim3(grid), dim3(256), 0, s, devDesc);
   return (int)hipGetLastError();
 }
 

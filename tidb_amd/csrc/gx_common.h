@@ -51,6 +51,7 @@ struct VmIns {
   int32_t dst;  // register index
   int32_t a;    // column index / const index / register
   int32_t b;    // register / power
+  int32_t c;    // LOAD ops: raw fetch slot (see FetchDesc)
 };
 
 constexpr int kMaxVmIns = 48;
@@ -68,10 +69,28 @@ struct PredDesc {
   int32_t kind;
   int32_t col;
   int32_t cmp;        // GX_F_LT..GX_F_NE
+  int32_t slot;       // raw fetch slot (TIME/I64 preds; -1 = load at use)
   uint64_t constU64;  // time value or i64/units bits
 };
 
 constexpr int kMaxPreds = 8;
+
+// ---- batched row fetch plan ----
+// The engine lists every 8/16-byte fetch the row pipeline needs; the kernel
+// issues them for R rows back-to-back before consuming (load-latency
+// pipelining — the kernel is otherwise wait-bound on the per-row load chain).
+enum FetchKind : int32_t {
+  FETCH_8B = 0,       // 8-byte element at data + row*8
+  FETCH_DEC16 = 1,    // 16 bytes at data + row*40 (decimal header+3 words)
+  FETCH_OFFSETS = 2,  // offsets[row], offsets[row+1]
+};
+
+struct FetchDesc {
+  int32_t kind;
+  int32_t col;
+};
+
+constexpr int kMaxFetch = 8;
 
 // ---- aggregation ----
 // Per-group state layout (all aggs): int128 acc + int64 count per agg slot.
@@ -94,10 +113,11 @@ struct GroupKeyDesc {
   int32_t nCols;
   int32_t col[2];
   int32_t kind[2];  // 0 = short string, 1 = small i64 (<2^31)
+  int32_t slot[2];  // raw fetch slot (string: the offsets pair; i64: value)
 };
 
 constexpr uint64_t kEmptyKey = ~0ULL;
-constexpr int kLdsGroups = 128;    // per-workgroup LDS table capacity
+constexpr int kLdsGroups = 64;     // per-workgroup LDS table capacity (19 KiB state => 8 blocks/CU, full 32-wave occupancy)
 constexpr int kGlobalGroups = 8192; // global table capacity (power of two)
 
 // group state sized for kMaxAggs
@@ -113,9 +133,13 @@ struct FusedQueryDesc {
   // filter
   PredDesc preds[kMaxPreds];
   int32_t nPreds = 0;
-  // projection/arg VM
+  // projection/arg VM (engine orders LOAD ops first: [0, nLoadIns))
   VmIns ins[kMaxVmIns];
   int32_t nIns = 0;
+  int32_t nLoadIns = 0;
+  // batched fetch plan
+  FetchDesc fetch[kMaxFetch];
+  int32_t nFetch = 0;
   int64_t constLo[kMaxVmConsts];
   int64_t constHi[kMaxVmConsts];
   int32_t nConsts = 0;

@@ -251,7 +251,7 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
   };
   auto emit = [&](int op, int dst, int a, int b) -> int {
     if (dst < 0 || d.nIns >= gxp::kMaxVmIns) return -1;
-    d.ins[d.nIns++] = {op, dst, a, b};
+    d.ins[d.nIns++] = {op, dst, a, b, -1};
     return dst;
   };
   int reg = -1;
@@ -342,6 +342,16 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
   return reg;
 }
 
+// allocate (or reuse) a raw-fetch slot for (kind, col)
+static int fetchSlot(gx_exec* ex, int kind, int col) {
+  gxp::FusedQueryDesc& d = ex->desc;
+  for (int i = 0; i < d.nFetch; i++)
+    if (d.fetch[i].kind == kind && d.fetch[i].col == col) return i;
+  if (d.nFetch >= gxp::kMaxFetch) return -1;
+  d.fetch[d.nFetch] = {kind, col};
+  return d.nFetch++;
+}
+
 // compile the fused Source->[Selection]->[Projection]->HashAgg pipeline
 static int32_t compileFused(gx_exec* ex) {
   const PPlan& plan = ex->plan;
@@ -405,12 +415,17 @@ static int32_t compileFused(gx_exec* ex) {
       pd.col = lhs->colIdx;
       pd.cmp = cmp;
       int ct = src->colTypes[lhs->colIdx];
+      pd.slot = -1;
       if (ct == GX_TYPE_TIME && rhs->retType == GX_TYPE_TIME) {
         pd.kind = gxp::PRED_TIME_CMP_CONST;
         pd.constU64 = rhs->constTime;
+        pd.slot = fetchSlot(ex, gxp::FETCH_8B, lhs->colIdx);
+        if (pd.slot < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
       } else if (ct == GX_TYPE_I64 && rhs->retType == GX_TYPE_I64) {
         pd.kind = gxp::PRED_I64_CMP_CONST;
         pd.constU64 = (uint64_t)rhs->constI64;
+        pd.slot = fetchSlot(ex, gxp::FETCH_8B, lhs->colIdx);
+        if (pd.slot < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
       } else if (ct == GX_TYPE_DECIMAL && rhs->retType == GX_TYPE_DECIMAL) {
         __int128 u;
         int sc;
@@ -489,6 +504,9 @@ static int32_t compileFused(gx_exec* ex) {
       ex->err = "device group key must be string or int64";
       return GX_ERR_INVALID;
     }
+    gk.slot[gk.nCols] = fetchSlot(
+        ex, t == GX_TYPE_STRING ? gxp::FETCH_OFFSETS : gxp::FETCH_8B, srcCol);
+    if (gk.slot[gk.nCols] < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
     gk.nCols++;
   }
   ex->desc.gkey = gk;
@@ -528,6 +546,32 @@ static int32_t compileFused(gx_exec* ex) {
       }
     }
     ex->desc.aggs[ex->desc.nAggs++] = ad;
+  }
+
+  // assign raw-fetch slots to VM load ops, then move loads to the front of
+  // the instruction stream (each dst is written once, loads have no deps —
+  // the kernel issues [0, nLoadIns) as the batched fetch-consume pipeline)
+  {
+    gxp::FusedQueryDesc& d = ex->desc;
+    std::vector<gxp::VmIns> loads, rest;
+    for (int i = 0; i < d.nIns; i++) {
+      gxp::VmIns ins = d.ins[i];
+      if (ins.op == gxp::VM_LOAD_DEC) {
+        ins.c = fetchSlot(ex, gxp::FETCH_DEC16, ins.a);
+        if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+        loads.push_back(ins);
+      } else if (ins.op == gxp::VM_LOAD_I64) {
+        ins.c = fetchSlot(ex, gxp::FETCH_8B, ins.a);
+        if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+        loads.push_back(ins);
+      } else {
+        rest.push_back(ins);
+      }
+    }
+    d.nLoadIns = (int)loads.size();
+    int k = 0;
+    for (auto& ins : loads) d.ins[k++] = ins;
+    for (auto& ins : rest) d.ins[k++] = ins;
   }
   ex->isFused = true;
   return GX_OK;

@@ -258,16 +258,14 @@ struct VT<true> {
   }
 };
 
-// parse a 40-byte MyDecimal (digitsInt <= 18, digitsFrac <= 9) into units at
-// scale = digitsFrac. Returns false on malformed input (sets kErrBadDecimal)
-// or narrow overflow (sets kErrRetryWide).
+// parse 16 raw bytes of a 40-byte MyDecimal (digitsInt <= 18, digitsFrac <= 9)
+// into units at scale = digitsFrac. Returns false on malformed input
+// (kErrBadDecimal) or narrow overflow (kErrRetryWide).
 template <bool WIDE>
-__device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* out,
-                                        int* scale, uint32_t* err) {
-  // 40-byte stride keeps rows 8-byte aligned: two dwordx2 loads cover
-  // header + wordBuf[0..2]
-  uint2 lo2 = *(const uint2*)p;
-  uint2 hi2 = *(const uint2*)(p + 8);
+__device__ inline bool parseDecimalRaw(ulonglong2 raw, typename VT<WIDE>::T* out,
+                                       int* scale, uint32_t* err) {
+  uint2 lo2 = {(uint32_t)raw.x, (uint32_t)(raw.x >> 32)};
+  uint2 hi2 = {(uint32_t)raw.y, (uint32_t)(raw.y >> 32)};
   uint32_t hdr = lo2.x;
   int digitsInt = (int)(int8_t)(hdr & 0xFF);
   int digitsFrac = (int)(int8_t)((hdr >> 8) & 0xFF);
@@ -301,6 +299,56 @@ __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* 
   }
   *scale = digitsFrac;
   return true;
+}
+
+template <bool WIDE>
+__device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* out,
+                                        int* scale, uint32_t* err) {
+  // 40-byte stride keeps rows 8-byte aligned: two dwordx2 loads
+  ulonglong2 raw;
+  raw.x = *(const uint64_t*)p;
+  raw.y = *(const uint64_t*)(p + 8);
+  return parseDecimalRaw<WIDE>(raw, out, scale, err);
+}
+
+// raw per-row fetch buffer: setters use compile-time slot indices (phase A),
+// the getter is a wave-uniform switch (runtime-indexed arrays would spill)
+struct RawState {
+  ulonglong2 s0, s1, s2, s3, s4, s5, s6, s7;
+  __device__ ulonglong2 get(int i) const {
+    switch (i) {
+      case 0: return s0; case 1: return s1; case 2: return s2; case 3: return s3;
+      case 4: return s4; case 5: return s5; case 6: return s6; default: return s7;
+    }
+  }
+  __device__ void set(int i, ulonglong2 v) {
+    switch (i) {
+      case 0: s0 = v; break; case 1: s1 = v; break; case 2: s2 = v; break;
+      case 3: s3 = v; break; case 4: s4 = v; break; case 5: s5 = v; break;
+      case 6: s6 = v; break; default: s7 = v; break;
+    }
+  }
+};
+
+// phase A: issue every fetch for one row, no consumption (loads overlap)
+__device__ inline void fetchRow(const FusedQueryDesc& d, int64_t row, RawState& raw) {
+  for (int f = 0; f < d.nFetch; f++) {
+    const FetchDesc& fd = d.fetch[f];
+    const DevCol& c = d.table.cols[fd.col];
+    ulonglong2 v;
+    if (fd.kind == FETCH_8B) {
+      v.x = ((const uint64_t*)c.data)[row];
+      v.y = 0;
+    } else if (fd.kind == FETCH_DEC16) {
+      const uint8_t* p = (const uint8_t*)c.data + row * 40;
+      v.x = *(const uint64_t*)p;
+      v.y = *(const uint64_t*)(p + 8);
+    } else {  // FETCH_OFFSETS
+      v.x = (uint64_t)c.offsets[row];
+      v.y = (uint64_t)c.offsets[row + 1];
+    }
+    raw.set(f, v);
+  }
 }
 
 __device__ inline bool colIsNull(const DevCol& c, int64_t row) {
@@ -348,9 +396,11 @@ struct VmState {
   }
 };
 
-// pack the group key (see GroupKeyDesc comment)
+// pack the group key (see GroupKeyDesc comment); offsets/values come from the
+// batched raw fetch
 __device__ inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
-                                    uint64_t* keyOut, uint32_t* err) {
+                                    const RawState& raw, uint64_t* keyOut,
+                                    uint32_t* err) {
   uint64_t key = 0;
   for (int k = 0; k < d.gkey.nCols; k++) {
     const DevCol& c = d.table.cols[d.gkey.col[k]];
@@ -358,7 +408,8 @@ __device__ inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
     if (colIsNull(c, row)) {
       lane = 0xFF000000u;
     } else if (d.gkey.kind[k] == 0) {
-      int64_t s = c.offsets[row], e = c.offsets[row + 1];
+      ulonglong2 off = raw.get(d.gkey.slot[k]);
+      int64_t s = (int64_t)off.x, e = (int64_t)off.y;
       // utf8mb4_bin PAD SPACE: trim trailing spaces (collate.go:272)
       const uint8_t* p = (const uint8_t*)c.data;
       while (e > s && p[e - 1] == ' ') e--;
@@ -367,7 +418,7 @@ __device__ inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
       lane = (uint32_t)len << 24;
       for (int64_t j = 0; j < len; j++) lane |= (uint32_t)p[s + j] << (8 * j);
     } else {
-      int64_t v = ((const int64_t*)c.data)[row];
+      int64_t v = (int64_t)raw.get(d.gkey.slot[k]).x;
       if (v < 0 || v > 0x7FFFFFFF) { atomicOr(err, kErrBadKey); return false; }
       lane = (uint32_t)v;
     }
@@ -394,7 +445,161 @@ __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
     atomicAdd((unsigned long long*)&slot->cnt[a], (unsigned long long)dc);
 }
 
+// per-row pipeline after the raw fetch: filter -> VM -> LDS aggregate.
+// Returns false on a hard failure (error flag already set).
 template <bool WIDE>
+__device__ inline bool processRow(const FusedQueryDesc& d, int64_t row,
+                                  const RawState& raw, GroupSlot* lds,
+                                  uint64_t* mySel) {
+  // ---- filter (CNF; NULL rejects — expression.go:507 toBool) ----
+  bool pass = true;
+  for (int p = 0; p < d.nPreds && pass; p++) {
+    const PredDesc& pd = d.preds[p];
+    const DevCol& c = d.table.cols[pd.col];
+    if (colIsNull(c, row)) { pass = false; break; }
+    if (pd.kind == PRED_TIME_CMP_CONST) {
+      uint64_t v = raw.get(pd.slot).x & ~0xFULL;
+      uint64_t k = pd.constU64 & ~0xFULL;
+      int cmp = v < k ? -1 : (v > k ? 1 : 0);
+      pass = cmpResult(cmp, pd.cmp);
+    } else if (pd.kind == PRED_I64_CMP_CONST) {
+      int64_t v = (int64_t)raw.get(pd.slot).x;
+      int64_t k = (int64_t)pd.constU64;
+      int cmp = v < k ? -1 : (v > k ? 1 : 0);
+      pass = cmpResult(cmp, pd.cmp);
+    } else {  // PRED_DEC_CMP_CONST (engine aligned const to column scale)
+      typename VT<WIDE>::T u;
+      int sc;
+      if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &u, &sc,
+                                  d.errorFlag))
+        return false;
+      int cmp = VT<WIDE>::cmp(u, VT<WIDE>::fromI64((int64_t)pd.constU64, nullptr));
+      pass = cmpResult(cmp, pd.cmp);
+    }
+  }
+  if (!pass) return true;
+  (*mySel)++;
+  if (d.ablate == 2) return true;  // timing ablation: filter only
+
+  // ---- projection / agg-arg VM ----
+  VmState<WIDE> vm;
+  vm.nullBits = 0;
+  bool bad = false;
+  bool ovf = false;
+  for (int i = 0; i < d.nIns && !bad; i++) {
+    const VmIns& ins = d.ins[i];
+    switch (ins.op) {
+      case VM_LOAD_DEC: {
+        const DevCol& c = d.table.cols[ins.a];
+        bool nul = colIsNull(c, row);
+        typename VT<WIDE>::T v = VT<WIDE>::zero();
+        if (!nul) {
+          int sc;
+          if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)) {
+            bad = true;
+            break;
+          }
+          if (sc != ins.b) {  // engine encoded expected scale in b
+            if (sc < ins.b) v = VT<WIDE>::scale10(v, ins.b - sc, &ovf);
+            else { atomicOr(d.errorFlag, kErrScale); bad = true; break; }
+          }
+        }
+        vm.set(ins.dst, v);
+        vm.setNull(ins.dst, nul);
+        break;
+      }
+      case VM_LOAD_I64: {
+        const DevCol& c = d.table.cols[ins.a];
+        bool nul = colIsNull(c, row);
+        vm.set(ins.dst, nul ? VT<WIDE>::zero()
+                            : VT<WIDE>::fromI64((int64_t)raw.get(ins.c).x, &ovf));
+        vm.setNull(ins.dst, nul);
+        break;
+      }
+      case VM_LOAD_CONST: {
+        if (WIDE) {
+          Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
+          vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+        } else {
+          // engine guarantees narrow-mode consts fit i64 (else it forces WIDE)
+          int64_t cv = d.constLo[ins.a];
+          vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+        }
+        vm.setNull(ins.dst, false);
+        break;
+      }
+      case VM_ADD:
+        vm.set(ins.dst, VT<WIDE>::add(vm.get(ins.a), vm.get(ins.b), &ovf));
+        vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+        break;
+      case VM_SUB:
+        vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
+        vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+        break;
+      case VM_MUL: {
+        bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
+        typename VT<WIDE>::T v = VT<WIDE>::zero();
+        if (!nul) v = VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf);
+        vm.set(ins.dst, v);
+        vm.setNull(ins.dst, nul);
+        break;
+      }
+      case VM_SCALE_UP:
+        vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
+        vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+    }
+  }
+  if (ovf) {
+    atomicOr(d.errorFlag, WIDE ? kErrOverflow : kErrRetryWide);
+    return false;
+  }
+  if (bad) return false;
+
+  if (d.ablate == 1) {  // timing ablation: keep VM results live, skip agg
+    uint64_t sink = 0;
+    for (int a = 0; a < d.nAggs; a++)
+      if (d.aggs[a].srcReg >= 0)
+        sink ^= (uint64_t)VT<WIDE>::toAcc(vm.get(d.aggs[a].srcReg)).lo;
+    asm volatile("" ::"v"(sink));
+    return true;
+  }
+
+  // ---- group lookup / insert in LDS ----
+  uint64_t key;
+  if (!makeGroupKey(d, row, raw, &key, d.errorFlag)) return false;
+  uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
+  for (int probe = 0;; probe++) {
+    if (probe >= kLdsGroups) {
+      atomicOr(d.errorFlag, kErrLdsFull);
+      return false;
+    }
+    uint64_t cur = lds[slot].key;
+    if (cur == key) break;
+    if (cur == kEmptyKey) {
+      uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
+                                (unsigned long long)kEmptyKey,
+                                (unsigned long long)key);
+      if (prev == kEmptyKey || prev == key) break;
+    }
+    slot = (slot + 1) & (kLdsGroups - 1);
+  }
+
+  // ---- update states ----
+  for (int a = 0; a < d.nAggs; a++) {
+    const AggDesc& ad = d.aggs[a];
+    if (ad.func == 0 /*COUNT*/) {
+      bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
+      if (!isNull) accumInto(&lds[slot], a, Int128{0, 0}, 1);
+    } else {  // SUM / AVG
+      if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
+        accumInto(&lds[slot], a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
+    }
+  }
+  return true;
+}
+
+template <bool WIDE, int R>
 __launch_bounds__(256)
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
@@ -417,155 +622,22 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   if (end > n) end = n;
   uint64_t mySel = 0;
 
-  for (int64_t row = begin + threadIdx.x; row < end; row += blockDim.x) {
-    // ---- filter (CNF; NULL rejects — expression.go:507 toBool) ----
-    bool pass = true;
-    for (int p = 0; p < d.nPreds && pass; p++) {
-      const PredDesc& pd = d.preds[p];
-      const DevCol& c = d.table.cols[pd.col];
-      if (colIsNull(c, row)) { pass = false; break; }
-      if (pd.kind == PRED_TIME_CMP_CONST) {
-        uint64_t v = ((const uint64_t*)c.data)[row] & ~0xFULL;
-        uint64_t k = pd.constU64 & ~0xFULL;
-        int cmp = v < k ? -1 : (v > k ? 1 : 0);
-        pass = cmpResult(cmp, pd.cmp);
-      } else if (pd.kind == PRED_I64_CMP_CONST) {
-        int64_t v = ((const int64_t*)c.data)[row];
-        int64_t k = (int64_t)pd.constU64;
-        int cmp = v < k ? -1 : (v > k ? 1 : 0);
-        pass = cmpResult(cmp, pd.cmp);
-      } else {  // PRED_DEC_CMP_CONST: engine aligned const to column scale
-        typename VT<WIDE>::T u;
-        int sc;
-        if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &u, &sc,
-                                    d.errorFlag)) {
-          failed = true;
-          break;
-        }
-        int cmp = VT<WIDE>::cmp(u, VT<WIDE>::fromI64((int64_t)pd.constU64, nullptr));
-        pass = cmpResult(cmp, pd.cmp);
-      }
+  // R-row software pipeline: phase A issues every fetch for R rows
+  // back-to-back (loads overlap in the memory system), phase B consumes —
+  // without this the wave parks ~70% of its cycles on per-row load chains.
+  for (int64_t base = begin + threadIdx.x; base < end && !failed;
+       base += (int64_t)blockDim.x * R) {
+    RawState raw[R];
+#pragma unroll
+    for (int j = 0; j < R; j++) {
+      int64_t row = base + (int64_t)j * blockDim.x;
+      if (row < end) fetchRow(d, row, raw[j]);
     }
-    if (failed) break;
-    if (!pass) continue;
-    mySel++;
-    if (d.ablate == 2) continue;  // timing ablation: filter only
-
-    // ---- projection / agg-arg VM ----
-    VmState<WIDE> vm;
-    vm.nullBits = 0;
-    bool bad = false;
-    bool ovf = false;
-    for (int i = 0; i < d.nIns && !bad; i++) {
-      const VmIns& ins = d.ins[i];
-      switch (ins.op) {
-        case VM_LOAD_DEC: {
-          const DevCol& c = d.table.cols[ins.a];
-          bool nul = colIsNull(c, row);
-          typename VT<WIDE>::T v = VT<WIDE>::zero();
-          if (!nul) {
-            int sc;
-            if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &v,
-                                        &sc, d.errorFlag)) {
-              bad = true;
-              break;
-            }
-            if (sc != ins.b) {  // engine encoded expected scale in b
-              if (sc < ins.b) v = VT<WIDE>::scale10(v, ins.b - sc, &ovf);
-              else { atomicOr(d.errorFlag, kErrScale); bad = true; break; }
-            }
-          }
-          vm.set(ins.dst, v);
-          vm.setNull(ins.dst, nul);
-          break;
-        }
-        case VM_LOAD_I64: {
-          const DevCol& c = d.table.cols[ins.a];
-          bool nul = colIsNull(c, row);
-          vm.set(ins.dst, nul ? VT<WIDE>::zero()
-                              : VT<WIDE>::fromI64(((const int64_t*)c.data)[row],
-                                                  &ovf));
-          vm.setNull(ins.dst, nul);
-          break;
-        }
-        case VM_LOAD_CONST: {
-          if (WIDE) {
-            Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
-            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
-          } else {
-            // engine guarantees narrow-mode consts fit i64 (else it forces WIDE)
-            int64_t cv = d.constLo[ins.a];
-            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
-          }
-          vm.setNull(ins.dst, false);
-          break;
-        }
-        case VM_ADD:
-          vm.set(ins.dst, VT<WIDE>::add(vm.get(ins.a), vm.get(ins.b), &ovf));
-          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
-          break;
-        case VM_SUB:
-          vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
-          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
-          break;
-        case VM_MUL: {
-          bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
-          typename VT<WIDE>::T v = VT<WIDE>::zero();
-          if (!nul) v = VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf);
-          vm.set(ins.dst, v);
-          vm.setNull(ins.dst, nul);
-          break;
-        }
-        case VM_SCALE_UP:
-          vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
-          vm.setNull(ins.dst, vm.isNull(ins.a));
-          break;
-      }
-    }
-    if (ovf) {
-      atomicOr(d.errorFlag, WIDE ? kErrOverflow : kErrRetryWide);
-      failed = true;
-      break;
-    }
-    if (bad) { failed = true; break; }
-
-    if (d.ablate == 1) {  // timing ablation: keep VM results live, skip agg
-      uint64_t sink = 0;
-      for (int a = 0; a < d.nAggs; a++)
-        if (d.aggs[a].srcReg >= 0) sink ^= (uint64_t)VT<WIDE>::toAcc(vm.get(d.aggs[a].srcReg)).lo;
-      asm volatile("" ::"v"(sink));
-      continue;
-    }
-
-    // ---- group lookup / insert in LDS ----
-    uint64_t key;
-    if (!makeGroupKey(d, row, &key, d.errorFlag)) { failed = true; break; }
-    uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
-    bool slotOk = true;
-    for (int probe = 0;; probe++) {
-      if (probe >= kLdsGroups) { atomicOr(d.errorFlag, kErrLdsFull); slotOk = false; break; }
-      uint64_t cur = lds[slot].key;
-      if (cur == key) break;
-      if (cur == kEmptyKey) {
-        uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
-                                  (unsigned long long)kEmptyKey,
-                                  (unsigned long long)key);
-        if (prev == kEmptyKey || prev == key) break;
-      }
-      slot = (slot + 1) & (kLdsGroups - 1);
-    }
-    if (!slotOk) { failed = true; break; }
-
-    // ---- update states ----
-    for (int a = 0; a < d.nAggs; a++) {
-      const AggDesc& ad = d.aggs[a];
-      if (ad.func == 0 /*COUNT*/) {
-        bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
-        if (!isNull) accumInto(&lds[slot], a, Int128{0, 0}, 1);
-      } else {  // SUM / AVG
-        if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
-          accumInto(&lds[slot], a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
-      }
+#pragma unroll
+    for (int j = 0; j < R; j++) {
+      int64_t row = base + (int64_t)j * blockDim.x;
+      if (row < end && !failed)
+        if (!processRow<WIDE>(d, row, raw[j], lds, &mySel)) failed = true;
     }
   }
 
@@ -644,9 +716,9 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
   hipLaunchKernelGGL(initGlobalTableKernel, dim3((kGlobalGroups + 255) / 256),
                      dim3(256), 0, s, desc.globalTable, kGlobalGroups);
   if (desc.wide)
-    hipLaunchKernelGGL(fusedAggKernel<true>, dim3(grid), dim3(256), 0, s, devDesc);
+    hipLaunchKernelGGL((fusedAggKernel<true, 2>), dim3(grid), dim3(256), 0, s, devDesc);
   else
-    hipLaunchKernelGGL(fusedAggKernel<false>, dim3(grid), dim3(256), 0, s, devDesc);
+    hipLaunchKernelGGL((fusedAggKernel<false, 4>), dim3(grid), dim3(256), 0, s, devDesc);
   return (int)hipGetLastError();
 }
 

@@ -157,6 +157,9 @@ struct FusedQueryDesc {
   // 0 = narrow int64 VM (fast path); 1 = wide int128 VM. The narrow kernel
   // reports kErrRetryWide on overflow and the engine relaunches wide.
   int32_t wide = 0;
+  // rows per thread batch in the fetch pipeline (engine-chosen: large fetch
+  // plans use a smaller R to stay inside the VGPR budget)
+  int32_t rbatch = 2;
 };
 
 // host-side launch wrappers (defined in gx_kernels.hip)

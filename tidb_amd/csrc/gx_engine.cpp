@@ -572,6 +572,12 @@ static int32_t compileFused(gx_exec* ex) {
     int k = 0;
     for (auto& ins : loads) d.ins[k++] = ins;
     for (auto& ins : rest) d.ins[k++] = ins;
+    // pick the fetch-pipeline depth: keep raw state within the VGPR budget
+    // (~16 bytes per slot per row)
+    if (d.nFetch <= 4) d.rbatch = 4;
+    else if (d.nFetch <= 7) d.rbatch = 2;
+    else d.rbatch = 1;
+    if (getenv("GX_RBATCH")) d.rbatch = atoi(getenv("GX_RBATCH"));
   }
   ex->isFused = true;
   return GX_OK;

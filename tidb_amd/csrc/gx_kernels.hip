@@ -262,7 +262,7 @@ struct VT<true> {
 // into units at scale = digitsFrac. Returns false on malformed input
 // (kErrBadDecimal) or narrow overflow (kErrRetryWide).
 template <bool WIDE>
-__device__ inline bool parseDecimalRaw(ulonglong2 raw, typename VT<WIDE>::T* out,
+__device__ __attribute__((always_inline)) inline bool parseDecimalRaw(ulonglong2 raw, typename VT<WIDE>::T* out,
                                        int* scale, uint32_t* err) {
   uint2 lo2 = {(uint32_t)raw.x, (uint32_t)(raw.x >> 32)};
   uint2 hi2 = {(uint32_t)raw.y, (uint32_t)(raw.y >> 32)};
@@ -330,9 +330,13 @@ struct RawState {
   }
 };
 
-// phase A: issue every fetch for one row, no consumption (loads overlap)
-__device__ inline void fetchRow(const FusedQueryDesc& d, int64_t row, RawState& raw) {
-  for (int f = 0; f < d.nFetch; f++) {
+// phase A: issue every fetch for one row, no consumption (loads overlap).
+// The loop is unrolled over the compile-time slot bound so every raw.set has
+// a literal index — a runtime-indexed store would be re-rolled into scratch.
+__device__ __attribute__((always_inline)) inline void fetchRow(const FusedQueryDesc& d, int64_t row, RawState& raw) {
+#pragma unroll
+  for (int f = 0; f < kMaxFetch; f++) {
+    if (f >= d.nFetch) break;
     const FetchDesc& fd = d.fetch[f];
     const DevCol& c = d.table.cols[fd.col];
     ulonglong2 v;
@@ -398,7 +402,7 @@ struct VmState {
 
 // pack the group key (see GroupKeyDesc comment); offsets/values come from the
 // batched raw fetch
-__device__ inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
+__device__ __attribute__((always_inline)) inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
                                     const RawState& raw, uint64_t* keyOut,
                                     uint32_t* err) {
   uint64_t key = 0;
@@ -448,7 +452,7 @@ __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
 // per-row pipeline after the raw fetch: filter -> VM -> LDS aggregate.
 // Returns false on a hard failure (error flag already set).
 template <bool WIDE>
-__device__ inline bool processRow(const FusedQueryDesc& d, int64_t row,
+__device__ __attribute__((always_inline)) inline bool processRow(const FusedQueryDesc& d, int64_t row,
                                   const RawState& raw, GroupSlot* lds,
                                   uint64_t* mySel) {
   // ---- filter (CNF; NULL rejects — expression.go:507 toBool) ----
@@ -600,7 +604,7 @@ __device__ inline bool processRow(const FusedQueryDesc& d, int64_t row,
 }
 
 template <bool WIDE, int R>
-__launch_bounds__(256)
+__launch_bounds__(256, 2)  // allow up to 256 VGPRs: raw fetch state stays in registers
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
   bool failed = false;
@@ -715,10 +719,19 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL(initGlobalTableKernel, dim3((kGlobalGroups + 255) / 256),
                      dim3(256), 0, s, desc.globalTable, kGlobalGroups);
-  if (desc.wide)
-    hipLaunchKernelGGL((fusedAggKernel<true, 2>), dim3(grid), dim3(256), 0, s, devDesc);
-  else
-    hipLaunchKernelGGL((fusedAggKernel<false, 4>), dim3(grid), dim3(256), 0, s, devDesc);
+  if (desc.wide) {
+    if (desc.rbatch >= 2)
+      hipLaunchKernelGGL((fusedAggKernel<true, 2>), dim3(grid), dim3(256), 0, s, devDesc);
+    else
+      hipLaunchKernelGGL((fusedAggKernel<true, 1>), dim3(grid), dim3(256), 0, s, devDesc);
+  } else {
+    if (desc.rbatch >= 4)
+      hipLaunchKernelGGL((fusedAggKernel<false, 4>), dim3(grid), dim3(256), 0, s, devDesc);
+    else if (desc.rbatch >= 2)
+      hipLaunchKernelGGL((fusedAggKernel<false, 2>), dim3(grid), dim3(256), 0, s, devDesc);
+    else
+      hipLaunchKernelGGL((fusedAggKernel<false, 1>), dim3(grid), dim3(256), 0, s, devDesc);
+  }
   return (int)hipGetLastError();
 }
 

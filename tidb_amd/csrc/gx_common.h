@@ -21,6 +21,9 @@ struct DevCol {
   int32_t type = 0;              // gx type
   int32_t frac = 0;
   uint8_t hasNulls = 0;
+  // varlen col whose offsets are the identity (offsets[i] == i, i.e. 1 byte
+  // per row — char(1)): the scan needs no offsets reads. Proven at bind time.
+  uint8_t denseOffsets = 0;
 };
 
 constexpr int kMaxCols = 16;
@@ -90,7 +93,7 @@ struct FetchDesc {
   int32_t col;
 };
 
-constexpr int kMaxFetch = 8;
+constexpr int kMaxFetch = 6;
 
 // ---- aggregation ----
 // Per-group state layout (all aggs): int128 acc + int64 count per agg slot.
@@ -112,8 +115,8 @@ constexpr int kMaxAggs = 12;
 struct GroupKeyDesc {
   int32_t nCols;
   int32_t col[2];
-  int32_t kind[2];  // 0 = short string, 1 = small i64 (<2^31)
-  int32_t slot[2];  // raw fetch slot (string: the offsets pair; i64: value)
+  int32_t kind[2];  // 0 = short string, 1 = small i64 (<2^31), 2 = dense char(1)
+  int32_t slot[2];  // raw fetch slot (string: the offsets pair; i64: value; dense: -1)
 };
 
 constexpr uint64_t kEmptyKey = ~0ULL;

@@ -420,12 +420,10 @@ static int32_t compileFused(gx_exec* ex) {
         pd.kind = gxp::PRED_TIME_CMP_CONST;
         pd.constU64 = rhs->constTime;
         pd.slot = fetchSlot(ex, gxp::FETCH_8B, lhs->colIdx);
-        if (pd.slot < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
       } else if (ct == GX_TYPE_I64 && rhs->retType == GX_TYPE_I64) {
         pd.kind = gxp::PRED_I64_CMP_CONST;
         pd.constU64 = (uint64_t)rhs->constI64;
         pd.slot = fetchSlot(ex, gxp::FETCH_8B, lhs->colIdx);
-        if (pd.slot < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
       } else if (ct == GX_TYPE_DECIMAL && rhs->retType == GX_TYPE_DECIMAL) {
         __int128 u;
         int sc;
@@ -504,9 +502,9 @@ static int32_t compileFused(gx_exec* ex) {
       ex->err = "device group key must be string or int64";
       return GX_ERR_INVALID;
     }
-    gk.slot[gk.nCols] = fetchSlot(
-        ex, t == GX_TYPE_STRING ? gxp::FETCH_OFFSETS : gxp::FETCH_8B, srcCol);
-    if (gk.slot[gk.nCols] < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+    gk.slot[gk.nCols] = -1;  // dense char(1) needs no fetch; fixed at open
+    if (t == GX_TYPE_I64)
+      gk.slot[gk.nCols] = fetchSlot(ex, gxp::FETCH_8B, srcCol);
     gk.nCols++;
   }
   ex->desc.gkey = gk;
@@ -558,11 +556,9 @@ static int32_t compileFused(gx_exec* ex) {
       gxp::VmIns ins = d.ins[i];
       if (ins.op == gxp::VM_LOAD_DEC) {
         ins.c = fetchSlot(ex, gxp::FETCH_DEC16, ins.a);
-        if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
         loads.push_back(ins);
       } else if (ins.op == gxp::VM_LOAD_I64) {
         ins.c = fetchSlot(ex, gxp::FETCH_8B, ins.a);
-        if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
         loads.push_back(ins);
       } else {
         rest.push_back(ins);
@@ -575,8 +571,7 @@ static int32_t compileFused(gx_exec* ex) {
     // pick the fetch-pipeline depth: keep raw state within the VGPR budget
     // (~16 bytes per slot per row)
     if (d.nFetch <= 4) d.rbatch = 4;
-    else if (d.nFetch <= 7) d.rbatch = 2;
-    else d.rbatch = 1;
+    else d.rbatch = 2;
     if (getenv("GX_RBATCH")) d.rbatch = atoi(getenv("GX_RBATCH"));
   }
   ex->isFused = true;
@@ -611,6 +606,7 @@ static int32_t materializeDevice(gx_exec* ex) {
         // generator strings are single bytes (char(1)); offsets dense
         col.offsets = (int64_t*)devAlloc(ex, (n + 1) * 8);
         col.data = devAlloc(ex, std::max<int64_t>(n, 1));
+        col.denseOffsets = 1;
         if (!col.offsets || !col.data) {
           ex->err = "hipMalloc failed";
           return GX_ERR_INTERNAL;
@@ -662,11 +658,17 @@ static int32_t materializeDevice(gx_exec* ex) {
           data.insert(data.end(), hc.data.begin(), hc.data.end());
           for (int i = 1; i <= hc.length; i++)
             offsets.push_back(base + hc.offsets[i]);
-        } else {
+        } else {  // (density of string offsets checked below)
           data.insert(data.end(), hc.data.begin(), hc.data.end());
         }
       }
       col.hasNulls = hasNulls ? 1 : 0;
+      if (col.type == GX_TYPE_STRING) {
+        bool dense = true;
+        for (size_t i = 0; i < offsets.size() && dense; i++)
+          if (offsets[i] != (int64_t)i) dense = false;
+        col.denseOffsets = dense ? 1 : 0;
+      }
       col.data = devAlloc(ex, std::max<size_t>(data.size(), 1));
       if (!col.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
       HIP_OK(ex, hipMemcpy(col.data, data.data(), data.size(),
@@ -689,6 +691,20 @@ static int32_t materializeDevice(gx_exec* ex) {
   } else {
     ex->err = "source not bound";
     return GX_ERR_INVALID;
+  }
+  // finalize string group keys now that offset density is known
+  for (int k = 0; k < ex->desc.gkey.nCols; k++) {
+    if (ex->desc.gkey.kind[k] != 1) {
+      const gxp::DevCol& c = tab.cols[ex->desc.gkey.col[k]];
+      if (c.denseOffsets) {
+        ex->desc.gkey.kind[k] = 2;
+        ex->desc.gkey.slot[k] = -1;
+      } else {
+        ex->desc.gkey.kind[k] = 0;
+        ex->desc.gkey.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS,
+                                          ex->desc.gkey.col[k]);
+      }
+    }
   }
   // result/error buffers
   ex->devTable = (gxp::GroupSlot*)devAlloc(ex, sizeof(gxp::GroupSlot) * gxp::kGlobalGroups);

@@ -731,12 +731,12 @@ static void decodeGroupLane(gx_exec* ex, uint32_t lane, int kind, int type,
     v->isNull = true;
     return;
   }
-  if (kind == 0) {
+  if (kind == 1) {
+    v->i64 = (int64_t)lane;
+  } else {  // kind 0 (string) and kind 2 (dense char) share the lane encoding
     int len = (int)(lane >> 24);
     v->str.clear();
     for (int j = 0; j < len; j++) v->str.push_back((char)((lane >> (8 * j)) & 0xFF));
-  } else {
-    v->i64 = (int64_t)lane;
   }
 }
 

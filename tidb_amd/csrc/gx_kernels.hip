@@ -314,17 +314,20 @@ __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* 
 // raw per-row fetch buffer: setters use compile-time slot indices (phase A),
 // the getter is a wave-uniform switch (runtime-indexed arrays would spill)
 struct RawState {
-  ulonglong2 s0, s1, s2, s3, s4, s5;
+  // scalar u64 members (a vector-typed member array gets lowered back to an
+  // indexed scratch buffer; scalar switches promote to registers)
+  uint64_t x0, y0, x1, y1, x2, y2, x3, y3, x4, y4, x5, y5;
   __device__ ulonglong2 get(int i) const {
     switch (i) {
-      case 0: return s0; case 1: return s1; case 2: return s2; case 3: return s3;
-      case 4: return s4; default: return s5;
+      case 0: return {x0, y0}; case 1: return {x1, y1}; case 2: return {x2, y2};
+      case 3: return {x3, y3}; case 4: return {x4, y4}; default: return {x5, y5};
     }
   }
   __device__ void set(int i, ulonglong2 v) {
     switch (i) {
-      case 0: s0 = v; break; case 1: s1 = v; break; case 2: s2 = v; break;
-      case 3: s3 = v; break; case 4: s4 = v; break; default: s5 = v; break;
+      case 0: x0 = v.x; y0 = v.y; break; case 1: x1 = v.x; y1 = v.y; break;
+      case 2: x2 = v.x; y2 = v.y; break; case 3: x3 = v.x; y3 = v.y; break;
+      case 4: x4 = v.x; y4 = v.y; break; default: x5 = v.x; y5 = v.y; break;
     }
   }
 };
@@ -623,7 +626,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
 }
 
 template <bool WIDE, int R>
-__launch_bounds__(256, 2)  // allow up to 256 VGPRs: raw fetch state stays in registers
+__attribute__((amdgpu_waves_per_eu(2, 4)))  // cap occupancy at 4 waves/SIMD: 128-VGPR budget keeps both raw buffers in registers
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
   bool failed = false;
@@ -645,23 +648,25 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   if (end > n) end = n;
   uint64_t mySel = 0;
 
-  // R-row software pipeline: phase A issues every fetch for R rows
-  // back-to-back (loads overlap in the memory system), phase B consumes —
-  // without this the wave parks ~70% of its cycles on per-row load chains.
-  for (int64_t base = begin + threadIdx.x; base < end && !failed;
-       base += (int64_t)blockDim.x * R) {
-    RawState raw[R];
-#pragma unroll
-    for (int j = 0; j < R; j++) {
-      int64_t row = base + (int64_t)j * blockDim.x;
-      if (row < end) fetchRow(d, row, raw[j]);
-    }
-#pragma unroll
-    for (int j = 0; j < R; j++) {
-      int64_t row = base + (int64_t)j * blockDim.x;
-      if (row < end && !failed)
-        if (!processRow<WIDE>(d, row, raw[j], lds, &mySel)) failed = true;
-    }
+  // double-buffered fetch pipeline (async-STAGE split, guide G15): row j+1's
+  // loads are issued BEFORE row j's compute, so the VM/aggregate work of row
+  // j overlaps row j+1's memory latency. Ping-pong between two named raw
+  // buffers (pointer swaps would force the buffers into scratch).
+  (void)sizeof(char[R]);  // R kept for the launch-variant signature
+  int64_t stride = blockDim.x;
+  int64_t row = begin + threadIdx.x;
+  RawState rawA, rawB;
+  if (row < end) fetchRow(d, row, rawA);
+  while (row < end && !failed) {
+    int64_t next = row + stride;
+    if (next < end) fetchRow(d, next, rawB);
+    if (!processRow<WIDE>(d, row, rawA, lds, &mySel)) failed = true;
+    row = next;
+    if (row >= end || failed) break;
+    next = row + stride;
+    if (next < end) fetchRow(d, next, rawA);
+    if (!processRow<WIDE>(d, row, rawB, lds, &mySel)) failed = true;
+    row = next;
   }
 
   if (d.selCount) {

@@ -28,7 +28,7 @@ SF10_ROWS = 59_986_052
 # algorithmic bytes per row, generated reference layout:
 # shipdate 8 + 4 decimals x 40 + 2 char(1) cols x (8 offsets + 1 data);
 # synthetic tables carry no null bitmaps (no NULLs) — stated, not 187.
-BYTES_PER_ROW = 8 + 4 * 40 + 2 * 9
+BYTES_PER_ROW = 8 + 4 * 40 + 2 * 1  # dense char(1): offsets proven identity at bind, not read
 HBM_PEAK_GBS = 8000.0  # spec peak (MI355X_MICROARCH.md; measured ceiling ~6290)
 
 

@@ -420,10 +420,12 @@ static int32_t compileFused(gx_exec* ex) {
         pd.kind = gxp::PRED_TIME_CMP_CONST;
         pd.constU64 = rhs->constTime;
         pd.slot = fetchSlot(ex, gxp::FETCH_8B, lhs->colIdx);
+        if (pd.slot < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
       } else if (ct == GX_TYPE_I64 && rhs->retType == GX_TYPE_I64) {
         pd.kind = gxp::PRED_I64_CMP_CONST;
         pd.constU64 = (uint64_t)rhs->constI64;
         pd.slot = fetchSlot(ex, gxp::FETCH_8B, lhs->colIdx);
+        if (pd.slot < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
       } else if (ct == GX_TYPE_DECIMAL && rhs->retType == GX_TYPE_DECIMAL) {
         __int128 u;
         int sc;
@@ -502,9 +504,11 @@ static int32_t compileFused(gx_exec* ex) {
       ex->err = "device group key must be string or int64";
       return GX_ERR_INVALID;
     }
-    gk.slot[gk.nCols] = -1;  // dense char(1) needs no fetch; fixed at open
-    if (t == GX_TYPE_I64)
+    gk.slot[gk.nCols] = -1;  // string keys get a slot at open (density known)
+    if (t == GX_TYPE_I64) {
       gk.slot[gk.nCols] = fetchSlot(ex, gxp::FETCH_8B, srcCol);
+      if (gk.slot[gk.nCols] < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+    }
     gk.nCols++;
   }
   ex->desc.gkey = gk;
@@ -556,9 +560,11 @@ static int32_t compileFused(gx_exec* ex) {
       gxp::VmIns ins = d.ins[i];
       if (ins.op == gxp::VM_LOAD_DEC) {
         ins.c = fetchSlot(ex, gxp::FETCH_DEC16, ins.a);
+        if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
         loads.push_back(ins);
       } else if (ins.op == gxp::VM_LOAD_I64) {
         ins.c = fetchSlot(ex, gxp::FETCH_8B, ins.a);
+        if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
         loads.push_back(ins);
       } else {
         rest.push_back(ins);
@@ -703,6 +709,10 @@ static int32_t materializeDevice(gx_exec* ex) {
         ex->desc.gkey.kind[k] = 0;
         ex->desc.gkey.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS,
                                           ex->desc.gkey.col[k]);
+        if (ex->desc.gkey.slot[k] < 0) {
+          ex->err = "fetch plan full";
+          return GX_ERR_INVALID;
+        }
       }
     }
   }

@@ -314,20 +314,18 @@ __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* 
 // raw per-row fetch buffer: setters use compile-time slot indices (phase A),
 // the getter is a wave-uniform switch (runtime-indexed arrays would spill)
 struct RawState {
-  // scalar u64 members (a vector-typed member array gets lowered back to an
-  // indexed scratch buffer; scalar switches promote to registers)
-  uint64_t x0, y0, x1, y1, x2, y2, x3, y3, x4, y4, x5, y5;
+  ulonglong2 s0, s1, s2, s3, s4, s5, s6, s7;
   __device__ ulonglong2 get(int i) const {
     switch (i) {
-      case 0: return {x0, y0}; case 1: return {x1, y1}; case 2: return {x2, y2};
-      case 3: return {x3, y3}; case 4: return {x4, y4}; default: return {x5, y5};
+      case 0: return s0; case 1: return s1; case 2: return s2; case 3: return s3;
+      case 4: return s4; case 5: return s5; case 6: return s6; default: return s7;
     }
   }
   __device__ void set(int i, ulonglong2 v) {
     switch (i) {
-      case 0: x0 = v.x; y0 = v.y; break; case 1: x1 = v.x; y1 = v.y; break;
-      case 2: x2 = v.x; y2 = v.y; break; case 3: x3 = v.x; y3 = v.y; break;
-      case 4: x4 = v.x; y4 = v.y; break; default: x5 = v.x; y5 = v.y; break;
+      case 0: s0 = v; break; case 1: s1 = v; break; case 2: s2 = v; break;
+      case 3: s3 = v; break; case 4: s4 = v; break; case 5: s5 = v; break;
+      case 6: s6 = v; break; default: s7 = v; break;
     }
   }
 };
@@ -418,15 +416,8 @@ __device__ __attribute__((always_inline)) inline bool makeGroupKey(const FusedQu
       uint8_t b = ((const uint8_t*)c.data)[row];
       lane = b == ' ' ? 0u : ((1u << 24) | b);
     } else if (d.gkey.kind[k] == 0) {
-      int64_t s, e;
-      if (d.gkey.slot[k] >= 0) {
-        ulonglong2 off = raw.get(d.gkey.slot[k]);
-        s = (int64_t)off.x;
-        e = (int64_t)off.y;
-      } else {
-        s = c.offsets[row];
-        e = c.offsets[row + 1];
-      }
+      ulonglong2 off = raw.get(d.gkey.slot[k]);
+      int64_t s = (int64_t)off.x, e = (int64_t)off.y;
       // utf8mb4_bin PAD SPACE: trim trailing spaces (collate.go:272)
       const uint8_t* p = (const uint8_t*)c.data;
       while (e > s && p[e - 1] == ' ') e--;
@@ -475,14 +466,12 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     const DevCol& c = d.table.cols[pd.col];
     if (colIsNull(c, row)) { pass = false; break; }
     if (pd.kind == PRED_TIME_CMP_CONST) {
-      uint64_t v = (pd.slot >= 0 ? raw.get(pd.slot).x
-                                 : ((const uint64_t*)c.data)[row]) & ~0xFULL;
+      uint64_t v = raw.get(pd.slot).x & ~0xFULL;
       uint64_t k = pd.constU64 & ~0xFULL;
       int cmp = v < k ? -1 : (v > k ? 1 : 0);
       pass = cmpResult(cmp, pd.cmp);
     } else if (pd.kind == PRED_I64_CMP_CONST) {
-      int64_t v = pd.slot >= 0 ? (int64_t)raw.get(pd.slot).x
-                               : ((const int64_t*)c.data)[row];
+      int64_t v = (int64_t)raw.get(pd.slot).x;
       int64_t k = (int64_t)pd.constU64;
       int cmp = v < k ? -1 : (v > k ? 1 : 0);
       pass = cmpResult(cmp, pd.cmp);
@@ -514,11 +503,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         typename VT<WIDE>::T v = VT<WIDE>::zero();
         if (!nul) {
           int sc;
-          bool okp = ins.c >= 0
-              ? parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)
-              : loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &v,
-                                       &sc, d.errorFlag);
-          if (!okp) {
+          if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)) {
             bad = true;
             break;
           }
@@ -535,10 +520,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         const DevCol& c = d.table.cols[ins.a];
         bool nul = colIsNull(c, row);
         vm.set(ins.dst, nul ? VT<WIDE>::zero()
-                            : VT<WIDE>::fromI64(
-                                  ins.c >= 0 ? (int64_t)raw.get(ins.c).x
-                                             : ((const int64_t*)c.data)[row],
-                                  &ovf));
+                            : VT<WIDE>::fromI64((int64_t)raw.get(ins.c).x, &ovf));
         vm.setNull(ins.dst, nul);
         break;
       }

@@ -135,3 +135,25 @@ def test_native_code_loaded(libs):
     _, product = libs
     assert product.gx_engine_is_gpu() == 1
     assert product.gx_engine_name() == b"gxexec-mi355x"
+
+
+def test_q3_parity(libs):
+    """Q3 (3-table join + grouped sum + TopN) on the device join-aggregate
+    pipeline vs the oracle executor."""
+    oracle, product = libs
+    from tests.test_oracle_q3 import run_q3
+    got = run_q3(product)
+    want = run_q3(oracle)
+    assert len(got) == len(want) > 0
+    assert got == want
+
+
+def test_q3_parity_larger(libs):
+    oracle, product = libs
+    import tests.test_oracle_q3 as q3
+    old = (q3.N_LI, q3.N_ORD, q3.N_CUST)
+    try:
+        q3.N_LI, q3.N_ORD, q3.N_CUST = 400000, 100000, 10000
+        assert q3.run_q3(product) == q3.run_q3(oracle)
+    finally:
+        q3.N_LI, q3.N_ORD, q3.N_CUST = old

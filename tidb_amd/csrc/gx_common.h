@@ -165,12 +165,92 @@ struct FusedQueryDesc {
   int32_t rbatch = 2;
 };
 
+// ---- join-aggregate pipeline (TPC-H Q3 class) ----
+// customer(filter) -> key set; orders(filter) x set -> slot table keyed by
+// orderkey carrying <=2 payload cols + one int128 accumulator; lineitem
+// (filter) probes the table and accumulates; top-N selected by radix
+// threshold + host final sort. This is HashJoinV2 build/probe
+// (join/hash_join_v2.go) fused with the grouped sum, MI355X-shaped: the
+// group-by keys are the probe key + build-side payload, so the aggregate
+// lands in the build row (one random-HBM touch per matching probe row).
+struct JoinAggSlot {
+  uint64_t key;      // build join key (kEmptyKey = vacant)
+  uint64_t payload0; // e.g. o_orderdate
+  int64_t payload1;  // e.g. o_shippriority
+  uint64_t accLo;
+  int64_t accHi;
+  uint64_t cnt;
+};
+
+struct JoinAggDesc {
+  // phase tables
+  DevTable build0;   // e.g. customer
+  DevTable build1;   // e.g. orders
+  DevTable probe;    // e.g. lineitem
+  // filters (one pred per table this round; PRED_STR_EQ_CONST via strConst)
+  PredDesc pred0, pred1, predP;
+  int32_t nPred0 = 0, nPred1 = 0, nPredP = 0;
+  uint8_t strConst[16];
+  int32_t strConstLen = 0;
+  // join columns
+  int32_t b0KeyCol;   // customer.c_custkey
+  int32_t b1ProbeCol; // orders.o_custkey (probes the b0 set)
+  int32_t b1KeyCol;   // orders.o_orderkey (keys the slot table)
+  int32_t pKeyCol;    // lineitem.l_orderkey
+  int32_t payloadCol0, payloadCol1;  // orders cols carried into slots (-1 none)
+  // probe-side value expression (VM over `probe` columns)
+  VmIns ins[kMaxVmIns];
+  int32_t nIns = 0;
+  int64_t constLo[kMaxVmConsts];
+  int64_t constHi[kMaxVmConsts];
+  int32_t nConsts = 0;
+  int32_t valueReg = -1;  // register holding the summed value
+  FetchDesc fetch[kMaxFetch];
+  int32_t nFetch = 0;
+  // device state
+  uint64_t* keySet = nullptr;  // open-addressed key set (build0)
+  int32_t keySetLog2 = 0;
+  JoinAggSlot* slots = nullptr;
+  int32_t slotsLog2 = 0;
+  uint64_t* counters = nullptr;  // [0] build0 pass, [1] build1 pass, [2] probe match
+  uint32_t* errorFlag = nullptr;
+  int32_t wide = 0;
+};
+
+enum { PRED_STR_EQ_CONST = 3 };  // extra PredKind for the join path
+
+// top-N selection scratch
+struct TopNOut {
+  uint64_t key, payload0;
+  int64_t payload1;
+  uint64_t accLo;
+  int64_t accHi;
+};
+
 // host-side launch wrappers (defined in gx_kernels.hip)
 int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows,
                     uint64_t seed, int64_t totalRows, void* stream);
 int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream);
 int gxLaunchMemset(void* p, int v, size_t n, void* stream);
+// join-agg pipeline steps (gx_kernels.hip)
+int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                   void* stream);  // 0 count0 1 build0 2 count1 3 build1 4 probe
+int gxJoinAggMax(const JoinAggDesc* devDesc, const JoinAggDesc& h, uint64_t* devMax,
+                 void* stream);
+int gxJoinAggHist(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                  uint32_t* devHist, int shift, void* stream);
+int gxJoinAggCompact(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                     TopNOut* out, uint64_t* outCount, uint64_t thresholdBucket,
+                     int shift, uint64_t cap, void* stream);
+int gxGenOrders(DevTable* tab, int64_t rowBegin, int64_t nRows, uint64_t seed,
+                int64_t totalRows, void* stream);
+// customer: pass 1 fills offsets (device scan) and returns total data bytes
+// (synchronizes); pass 2 fills key + segment bytes
+int gxGenCustomerOffsets(DevTable* tab, int64_t rowBegin, int64_t nRows,
+                         uint64_t seed, void* stream, long long* totalBytes);
+int gxGenCustomerFill(DevTable* tab, int64_t rowBegin, int64_t nRows,
+                      uint64_t seed, void* stream);
 
 }  // namespace gxp
 

@@ -112,6 +112,7 @@ struct gx_exec {
   bool isFused = false;
   bool isFinalHost = false;
   bool isBareSource = false;
+  bool isJoinAgg = false;
   int sourceNode = -1;
   gxp::FusedQueryDesc desc;
   std::vector<std::pair<int, int>> projRegs;  // projection idx -> (reg, scale)
@@ -133,6 +134,19 @@ struct gx_exec {
   int64_t srcPos = 0;
   uint64_t lastSelCount = 0;
   double lastKernelMs = 0;
+
+  // join-agg (Q3-class) state
+  gxp::JoinAggDesc ja;
+  int jaSrcCust = -1, jaSrcOrd = -1, jaSrcLi = -1;
+  gxp::JoinAggDesc* devJa = nullptr;
+  int jaValueScale = 0;
+  // output mapping: for each group column: 0 = probe key, 1 = payload0,
+  // 2 = payload1; types of the payload cols
+  std::vector<int> jaGroupSrc;
+  std::vector<int> jaGroupType;
+  // topn keys: (column index into the agg output row, desc?)
+  std::vector<std::pair<int, bool>> jaSortKeys;
+  int64_t jaLimit = 0, jaOffset = 0;
 
   ~gx_exec() {
     for (void* p : devBufs) hipFree(p);
@@ -342,7 +356,33 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
   return reg;
 }
 
-// allocate (or reuse) a raw-fetch slot for (kind, col)
+// generic VM program build context (used by the fused-agg compiler and the
+// join-agg probe compiler)
+struct VmBuild {
+  gxp::VmIns* ins;
+  int32_t* nIns;
+  int64_t* cLo;
+  int64_t* cHi;
+  int32_t* nConsts;
+  gxp::FetchDesc* fetch;
+  int32_t* nFetch;
+  const std::vector<int>* colTypes;  // schema of the table the VM runs over
+  const std::vector<int>* colFracs;
+  int colBase = 0;  // plan colref index of this table's first column
+  int nextReg = 0;
+  std::map<int, std::pair<int, int>> cache;
+  int32_t* wideFlag;
+};
+
+static int vmFetchSlot(VmBuild& B, int kind, int col) {
+  for (int i = 0; i < *B.nFetch; i++)
+    if (B.fetch[i].kind == kind && B.fetch[i].col == col) return i;
+  if (*B.nFetch >= gxp::kMaxFetch) return -1;
+  B.fetch[*B.nFetch] = {kind, col};
+  return (*B.nFetch)++;
+}
+
+// allocate (or reuse) a raw-fetch slot for (kind, col) on the fused desc
 static int fetchSlot(gx_exec* ex, int kind, int col) {
   gxp::FusedQueryDesc& d = ex->desc;
   for (int i = 0; i < d.nFetch; i++)
@@ -350,6 +390,402 @@ static int fetchSlot(gx_exec* ex, int kind, int col) {
   if (d.nFetch >= gxp::kMaxFetch) return -1;
   d.fetch[d.nFetch] = {kind, col};
   return d.nFetch++;
+}
+
+// compile expression exprId over a single table schema into a VM program.
+// Returns the result register (>= 0) or -1; *scaleOut = static decimal scale.
+static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
+  const PExpr& e = ex->plan.exprs[exprId];
+  auto cached = B.cache.find(exprId);
+  if (cached != B.cache.end()) {
+    *scaleOut = cached->second.second;
+    return cached->second.first;
+  }
+  auto allocReg = [&]() -> int {
+    if (B.nextReg >= gxp::kMaxVmRegs) return -1;
+    return B.nextReg++;
+  };
+  auto emit = [&](int op, int dst, int a, int b, int c) -> int {
+    if (dst < 0 || *B.nIns >= gxp::kMaxVmIns) return -1;
+    B.ins[(*B.nIns)++] = {op, dst, a, b, c};
+    return dst;
+  };
+  int reg = -1;
+  switch (e.kind) {
+    case EK_COLREF: {
+      int col = e.colIdx - B.colBase;
+      if (col < 0 || col >= (int)B.colTypes->size()) {
+        ex->err = "bad colref in device expression";
+        return -1;
+      }
+      int t = (*B.colTypes)[col];
+      if (t == GX_TYPE_DECIMAL) {
+        int frac = (*B.colFracs)[col];
+        int slot = vmFetchSlot(B, gxp::FETCH_DEC16, col);
+        if (slot < 0) { ex->err = "fetch plan full"; return -1; }
+        reg = emit(gxp::VM_LOAD_DEC, allocReg(), col, frac, slot);
+        *scaleOut = frac;
+      } else if (t == GX_TYPE_I64) {
+        int slot = vmFetchSlot(B, gxp::FETCH_8B, col);
+        if (slot < 0) { ex->err = "fetch plan full"; return -1; }
+        reg = emit(gxp::VM_LOAD_I64, allocReg(), col, 0, slot);
+        *scaleOut = 0;
+      } else {
+        ex->err = "unsupported colref type in device expression";
+        return -1;
+      }
+      break;
+    }
+    case EK_CONST: {
+      __int128 u = 0;
+      int sc = 0;
+      if (e.retType == GX_TYPE_DECIMAL) {
+        if (!decToUnits(e.constDec, &u, &sc)) {
+          ex->err = "const decimal too wide for device path";
+          return -1;
+        }
+      } else if (e.retType == GX_TYPE_I64) {
+        u = e.constI64;
+        sc = 0;
+      } else {
+        ex->err = "unsupported const type in device expression";
+        return -1;
+      }
+      if (*B.nConsts >= gxp::kMaxVmConsts) {
+        ex->err = "too many consts";
+        return -1;
+      }
+      int ci = (*B.nConsts)++;
+      B.cLo[ci] = (int64_t)(uint64_t)u;
+      B.cHi[ci] = (int64_t)(u >> 64);
+      if (u > (__int128)INT64_MAX || u < (__int128)INT64_MIN) *B.wideFlag = 1;
+      reg = emit(gxp::VM_LOAD_CONST, allocReg(), ci, 0, -1);
+      *scaleOut = sc;
+      break;
+    }
+    case EK_CALL: {
+      if (e.args.size() != 2) {
+        ex->err = "unsupported call arity on device";
+        return -1;
+      }
+      int sa = 0, sb = 0;
+      int ra = vmCompile(ex, B, e.args[0], &sa);
+      if (ra < 0) return -1;
+      int rb = vmCompile(ex, B, e.args[1], &sb);
+      if (rb < 0) return -1;
+      int op;
+      switch (e.func) {
+        case GX_F_PLUS: op = gxp::VM_ADD; break;
+        case GX_F_MINUS: op = gxp::VM_SUB; break;
+        case GX_F_MUL: op = gxp::VM_MUL; break;
+        default:
+          ex->err = "unsupported function on device path";
+          return -1;
+      }
+      if (op == gxp::VM_MUL) {
+        reg = emit(op, allocReg(), ra, rb, -1);
+        *scaleOut = sa + sb;
+      } else {
+        int target = std::max(sa, sb);
+        if (sa < target) ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa, -1);
+        if (sb < target) rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb, -1);
+        if (ra < 0 || rb < 0) break;
+        reg = emit(op, allocReg(), ra, rb, -1);
+        *scaleOut = target;
+      }
+      break;
+    }
+  }
+  if (reg < 0 && ex->err.empty()) ex->err = "expression too large for device VM";
+  if (reg >= 0) B.cache[exprId] = {reg, *scaleOut};
+  return reg;
+}
+
+// compile one Selection condition over a single-table schema into a PredDesc
+static bool compileTablePred(gx_exec* ex, const PNode& srcNode, int condId,
+                             gxp::PredDesc* out, uint8_t* strConst,
+                             int32_t* strConstLen) {
+  const PExpr& e = ex->plan.exprs[condId];
+  if (e.kind != EK_CALL || e.func > GX_F_NE || e.args.size() != 2) {
+    ex->err = "unsupported filter expression on device";
+    return false;
+  }
+  const PExpr* lhs = &ex->plan.exprs[e.args[0]];
+  const PExpr* rhs = &ex->plan.exprs[e.args[1]];
+  int cmp = e.func;
+  if (lhs->kind == EK_CONST && rhs->kind == EK_COLREF) {
+    std::swap(lhs, rhs);
+    static const int mirror[6] = {GX_F_GT, GX_F_GE, GX_F_LT, GX_F_LE, GX_F_EQ,
+                                  GX_F_NE};
+    cmp = mirror[cmp];
+  }
+  if (lhs->kind != EK_COLREF || rhs->kind != EK_CONST) {
+    ex->err = "device filter must be <column> <cmp> <const>";
+    return false;
+  }
+  gxp::PredDesc pd{};
+  pd.col = lhs->colIdx;
+  pd.cmp = cmp;
+  pd.slot = -1;
+  if (pd.col < 0 || pd.col >= (int)srcNode.colTypes.size()) {
+    ex->err = "filter column out of range";
+    return false;
+  }
+  int ct = srcNode.colTypes[pd.col];
+  if (ct == GX_TYPE_TIME && rhs->retType == GX_TYPE_TIME) {
+    pd.kind = gxp::PRED_TIME_CMP_CONST;
+    pd.constU64 = rhs->constTime;
+  } else if (ct == GX_TYPE_I64 && rhs->retType == GX_TYPE_I64) {
+    pd.kind = gxp::PRED_I64_CMP_CONST;
+    pd.constU64 = (uint64_t)rhs->constI64;
+  } else if (ct == GX_TYPE_STRING && rhs->retType == GX_TYPE_STRING &&
+             (cmp == GX_F_EQ || cmp == GX_F_NE)) {
+    if (rhs->constStr.size() > 16) {
+      ex->err = "string const too long for device filter";
+      return false;
+    }
+    pd.kind = gxp::PRED_STR_EQ_CONST;
+    // PAD SPACE: trim the constant's trailing spaces too
+    std::string k = rhs->constStr;
+    while (!k.empty() && k.back() == ' ') k.pop_back();
+    std::memcpy(strConst, k.data(), k.size());
+    *strConstLen = (int32_t)k.size();
+  } else {
+    ex->err = "unsupported filter column/const type combination";
+    return false;
+  }
+  *out = pd;
+  return true;
+}
+
+// unwrap [Selection ->] Source; returns source node id (or -1) and the
+// selection node (or nullptr)
+static int unwrapSource(gx_exec* ex, int node, const PNode** selOut) {
+  *selOut = nullptr;
+  const PNode* n = &ex->plan.nodes[node];
+  if (n->kind == PK_SELECTION) {
+    *selOut = n;
+    node = n->child;
+    n = &ex->plan.nodes[node];
+  }
+  if (n->kind != PK_SOURCE) return -1;
+  return node;
+}
+
+// recognize & compile the join-aggregate (Q3-class) plan:
+// TopN <- HashAgg(sum) <- Projection <- HashJoin(probe=lineitem,
+//   build=HashJoin(build=customer, probe=orders))
+static int32_t compileJoinAgg(gx_exec* ex) {
+  const PPlan& plan = ex->plan;
+  const PNode& topn = plan.nodes[ex->root];
+  if (topn.limit <= 0 || topn.limit + topn.offset > 100000) {
+    ex->err = "device TopN limit too large";
+    return GX_ERR_INVALID;
+  }
+  const PNode& agg = plan.nodes[topn.child];
+  if (agg.kind != PK_HASHAGG || agg.aggMode != GX_AGG_MODE_COMPLETE ||
+      agg.aggFuncs.size() != 1 || agg.aggFuncs[0] != GX_AGG_SUM ||
+      agg.exprs.empty() || agg.exprs.size() > 3) {
+    ex->err = "unsupported aggregate shape for device join path";
+    return GX_ERR_INVALID;
+  }
+  const PNode& proj = plan.nodes[agg.child];
+  if (proj.kind != PK_PROJECTION) {
+    ex->err = "expected projection under aggregate";
+    return GX_ERR_INVALID;
+  }
+  const PNode& j2 = plan.nodes[proj.child];
+  if (j2.kind != PK_HASHJOIN || j2.joinType != 0 || j2.buildKeys.size() != 1) {
+    ex->err = "unsupported join shape";
+    return GX_ERR_INVALID;
+  }
+  const PNode& j1 = plan.nodes[j2.child];
+  if (j1.kind != PK_HASHJOIN || j1.joinType != 0 || j1.buildKeys.size() != 1) {
+    ex->err = "unsupported inner join shape";
+    return GX_ERR_INVALID;
+  }
+  const PNode *selC, *selO, *selL;
+  int srcC = unwrapSource(ex, j1.child, &selC);
+  int srcO = unwrapSource(ex, j1.child2, &selO);
+  int srcL = unwrapSource(ex, j2.child2, &selL);
+  if (srcC < 0 || srcO < 0 || srcL < 0) {
+    ex->err = "join children must be [Selection ->] Source";
+    return GX_ERR_INVALID;
+  }
+  const PNode& custN = plan.nodes[srcC];
+  const PNode& ordN = plan.nodes[srcO];
+  const PNode& liN = plan.nodes[srcL];
+  int nc = (int)custN.colTypes.size();
+  int no = (int)ordN.colTypes.size();
+  int nl = (int)liN.colTypes.size();
+  gxp::JoinAggDesc& ja = ex->ja;
+
+  // join keys (single-column int64)
+  const PExpr& bk1 = plan.exprs[j1.buildKeys[0]];
+  const PExpr& pk1 = plan.exprs[j1.probeKeys[0]];
+  const PExpr& bk2 = plan.exprs[j2.buildKeys[0]];
+  const PExpr& pk2 = plan.exprs[j2.probeKeys[0]];
+  if (bk1.kind != EK_COLREF || pk1.kind != EK_COLREF || bk2.kind != EK_COLREF ||
+      pk2.kind != EK_COLREF) {
+    ex->err = "join keys must be columns";
+    return GX_ERR_INVALID;
+  }
+  if (custN.colTypes[bk1.colIdx] != GX_TYPE_I64 ||
+      ordN.colTypes[pk1.colIdx] != GX_TYPE_I64 ||
+      liN.colTypes[pk2.colIdx] != GX_TYPE_I64 || bk2.colIdx < nc ||
+      bk2.colIdx >= nc + no || ordN.colTypes[bk2.colIdx - nc] != GX_TYPE_I64) {
+    ex->err = "device join keys must be int64";
+    return GX_ERR_INVALID;
+  }
+  ja.b0KeyCol = bk1.colIdx;
+  ja.b1ProbeCol = pk1.colIdx;
+  ja.b1KeyCol = bk2.colIdx - nc;
+  ja.pKeyCol = pk2.colIdx;
+
+  // predicates (0 or 1 conjunct per table this round)
+  auto doPred = [&](const PNode* sel, const PNode& srcNode, gxp::PredDesc* pd,
+                    int32_t* n) -> bool {
+    *n = 0;
+    if (!sel) return true;
+    if (sel->exprs.size() != 1) {
+      ex->err = "device join path supports one filter conjunct per table";
+      return false;
+    }
+    if (!compileTablePred(ex, srcNode, sel->exprs[0], pd, ja.strConst,
+                          &ja.strConstLen))
+      return false;
+    *n = 1;
+    return true;
+  };
+  if (!doPred(selC, custN, &ja.pred0, &ja.nPred0)) return GX_ERR_INVALID;
+  if (!doPred(selO, ordN, &ja.pred1, &ja.nPred1)) return GX_ERR_INVALID;
+  if (!doPred(selL, liN, &ja.predP, &ja.nPredP)) return GX_ERR_INVALID;
+
+  // projection classification
+  ja.payloadCol0 = ja.payloadCol1 = -1;
+  std::vector<int> projClass(proj.exprs.size(), -1);  // 0 probekey 1 pay0 2 pay1 3 value
+  int valueExpr = -1;
+  for (size_t i = 0; i < proj.exprs.size(); i++) {
+    const PExpr& pe = plan.exprs[proj.exprs[i]];
+    if (pe.kind == EK_COLREF) {
+      if (pe.colIdx >= nc + no) {
+        int lcol = pe.colIdx - nc - no;
+        if (lcol == ja.pKeyCol) projClass[i] = 0;
+        else {
+          ex->err = "probe-side passthrough must be the join key";
+          return GX_ERR_INVALID;
+        }
+      } else if (pe.colIdx >= nc) {
+        int ocol = pe.colIdx - nc;
+        int t = ordN.colTypes[ocol];
+        if (t != GX_TYPE_I64 && t != GX_TYPE_TIME) {
+          ex->err = "payload columns must be 8-byte";
+          return GX_ERR_INVALID;
+        }
+        if (ja.payloadCol0 < 0 || ja.payloadCol0 == ocol) {
+          ja.payloadCol0 = ocol;
+          projClass[i] = 1;
+        } else if (ja.payloadCol1 < 0 || ja.payloadCol1 == ocol) {
+          ja.payloadCol1 = ocol;
+          projClass[i] = 2;
+        } else {
+          ex->err = "too many build payload columns";
+          return GX_ERR_INVALID;
+        }
+      } else {
+        ex->err = "customer-side payload not supported this round";
+        return GX_ERR_INVALID;
+      }
+    } else {
+      if (valueExpr >= 0) {
+        ex->err = "one computed value per device join";
+        return GX_ERR_INVALID;
+      }
+      valueExpr = proj.exprs[i];
+      projClass[i] = 3;
+    }
+  }
+  if (valueExpr < 0) {
+    ex->err = "missing value expression";
+    return GX_ERR_INVALID;
+  }
+  // the agg arg must be the value expr; group cols map through projClass
+  const PExpr& aggArg = plan.exprs[agg.aggArgs[0]];
+  if (aggArg.kind != EK_COLREF || projClass[aggArg.colIdx] != 3) {
+    ex->err = "sum argument must be the computed value";
+    return GX_ERR_INVALID;
+  }
+  ex->jaGroupSrc.clear();
+  ex->jaGroupType.clear();
+  for (int ge : agg.exprs) {
+    const PExpr& gexpr = plan.exprs[ge];
+    if (gexpr.kind != EK_COLREF || projClass[gexpr.colIdx] < 0 ||
+        projClass[gexpr.colIdx] == 3) {
+      ex->err = "group keys must be the join key or payload columns";
+      return GX_ERR_INVALID;
+    }
+    int cls = projClass[gexpr.colIdx];
+    ex->jaGroupSrc.push_back(cls);
+    if (cls == 0) ex->jaGroupType.push_back(GX_TYPE_I64);
+    else if (cls == 1) ex->jaGroupType.push_back(ordN.colTypes[ja.payloadCol0]);
+    else ex->jaGroupType.push_back(ordN.colTypes[ja.payloadCol1]);
+  }
+
+  // value expression VM over the probe (lineitem) table
+  VmBuild B;
+  B.ins = ja.ins;
+  B.nIns = &ja.nIns;
+  B.cLo = ja.constLo;
+  B.cHi = ja.constHi;
+  B.nConsts = &ja.nConsts;
+  B.fetch = ja.fetch;
+  B.nFetch = &ja.nFetch;
+  B.colTypes = &liN.colTypes;
+  B.colFracs = &liN.colFracs;
+  B.colBase = nc + no;
+  B.wideFlag = &ja.wide;
+  int sc = 0;
+  int reg = vmCompile(ex, B, valueExpr, &sc);
+  if (reg < 0) return GX_ERR_INVALID;
+  ja.valueReg = reg;
+  ex->jaValueScale = sc;
+  // pred column fetch slot + key fetch slot (probe-side; via the VM fetch plan)
+  if (ja.nPredP &&
+      (ja.predP.kind == gxp::PRED_TIME_CMP_CONST ||
+       ja.predP.kind == gxp::PRED_I64_CMP_CONST))
+    ja.predP.slot = vmFetchSlot(B, gxp::FETCH_8B, ja.predP.col);
+  // reorder loads first (mirrors the fused path)
+  {
+    std::vector<gxp::VmIns> loads, rest;
+    for (int i = 0; i < ja.nIns; i++) {
+      if (ja.ins[i].op == gxp::VM_LOAD_DEC || ja.ins[i].op == gxp::VM_LOAD_I64)
+        loads.push_back(ja.ins[i]);
+      else
+        rest.push_back(ja.ins[i]);
+    }
+    int k = 0;
+    for (auto& ins : loads) ja.ins[k++] = ins;
+    for (auto& ins : rest) ja.ins[k++] = ins;
+  }
+
+  // topn keys -> positions in the agg output row [groups..., sum]
+  ex->jaSortKeys.clear();
+  int aggWidth = (int)agg.exprs.size() + 1;
+  for (size_t i = 0; i < topn.exprs.size(); i++) {
+    const PExpr& ke = plan.exprs[topn.exprs[i]];
+    if (ke.kind != EK_COLREF || ke.colIdx < 0 || ke.colIdx >= aggWidth) {
+      ex->err = "TopN keys must be aggregate output columns";
+      return GX_ERR_INVALID;
+    }
+    ex->jaSortKeys.push_back({ke.colIdx, topn.keyDesc[i] != 0});
+  }
+  ex->jaLimit = topn.limit;
+  ex->jaOffset = topn.offset;
+  ex->jaSrcCust = srcC;
+  ex->jaSrcOrd = srcO;
+  ex->jaSrcLi = srcL;
+  ex->isJoinAgg = true;
+  return GX_OK;
 }
 
 // compile the fused Source->[Selection]->[Projection]->HashAgg pipeline
@@ -585,6 +1021,147 @@ static int32_t compileFused(gx_exec* ex) {
 }
 
 // ---------------- device materialization ----------------
+
+// upload bound host chunks into a device table (concatenated)
+static int32_t uploadBoundChunks(gx_exec* ex, Binding& b, gxp::DevTable& tab) {
+  int nCols = tab.nCols;
+  int64_t total = 0;
+  for (auto& ch : b.chunks) total += ch.empty() ? 0 : ch[0].length;
+  tab.nRows = total;
+  for (int c = 0; c < nCols; c++) {
+    gxp::DevCol& col = tab.cols[c];
+    std::vector<uint8_t> data;
+    std::vector<int64_t> offsets{0};
+    std::vector<uint8_t> nulls((total + 7) / 8, 0);
+    int64_t row = 0;
+    bool hasNulls = false;
+    for (auto& ch : b.chunks) {
+      const HostCol& hc = ch[c];
+      for (int i = 0; i < hc.length; i++, row++) {
+        bool notNull = (hc.nullBitmap[i / 8] >> (i % 8)) & 1;
+        if (notNull) nulls[row / 8] |= 1 << (row % 8);
+        else hasNulls = true;
+      }
+      if (col.type == GX_TYPE_STRING) {
+        int64_t base = data.size();
+        data.insert(data.end(), hc.data.begin(), hc.data.end());
+        for (int i = 1; i <= hc.length; i++)
+          offsets.push_back(base + hc.offsets[i]);
+      } else {
+        data.insert(data.end(), hc.data.begin(), hc.data.end());
+      }
+    }
+    col.hasNulls = hasNulls ? 1 : 0;
+    if (col.type == GX_TYPE_STRING) {
+      bool dense = true;
+      for (size_t i = 0; i < offsets.size() && dense; i++)
+        if (offsets[i] != (int64_t)i) dense = false;
+      col.denseOffsets = dense ? 1 : 0;
+    }
+    col.data = devAlloc(ex, std::max<size_t>(data.size(), 1));
+    if (!col.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    HIP_OK(ex, hipMemcpy(col.data, data.data(), data.size(),
+                         hipMemcpyHostToDevice));
+    if (col.type == GX_TYPE_STRING) {
+      col.offsets = (int64_t*)devAlloc(ex, offsets.size() * 8);
+      if (!col.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      HIP_OK(ex, hipMemcpy(col.offsets, offsets.data(), offsets.size() * 8,
+                           hipMemcpyHostToDevice));
+    }
+    if (hasNulls) {
+      col.nullBitmap = (uint8_t*)devAlloc(ex, nulls.size());
+      if (!col.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      HIP_OK(ex, hipMemcpy(col.nullBitmap, nulls.data(), nulls.size(),
+                           hipMemcpyHostToDevice));
+    } else {
+      col.nullBitmap = nullptr;
+    }
+  }
+  return GX_OK;
+}
+
+// materialize one source (tpch generator or bound chunks) into a device table
+static int32_t materializeTable(gx_exec* ex, int srcNodeId, gxp::DevTable* tab) {
+  const PNode& srcNode = ex->plan.nodes[srcNodeId];
+  tab->nCols = (int)srcNode.colTypes.size();
+  for (size_t c = 0; c < srcNode.colTypes.size(); c++)
+    setDevColMeta(&tab->cols[c], srcNode.colTypes[c], srcNode.colFracs[c]);
+  auto it = ex->bindings.find(srcNodeId);
+  if (it == ex->bindings.end()) {
+    ex->err = "source not bound";
+    return GX_ERR_INVALID;
+  }
+  Binding& b = it->second;
+  if (b.haveChunks) return uploadBoundChunks(ex, b, *tab);
+  if (b.tpchTable < 0) {
+    ex->err = "source not bound";
+    return GX_ERR_INVALID;
+  }
+  int64_t n = b.tpchRows;
+  tab->nRows = n;
+  int64_t totalRows = b.tpchTotalRows > 0 ? b.tpchTotalRows : n;
+  int rc = 0;
+  switch (b.tpchTable) {
+    case GX_TPCH_LINEITEM: {
+      for (int c = 0; c < tab->nCols; c++) {
+        gxp::DevCol& col = tab->cols[c];
+        if (col.type == GX_TYPE_STRING) {
+          col.offsets = (int64_t*)devAlloc(ex, (n + 1) * 8);
+          col.data = devAlloc(ex, std::max<int64_t>(n, 1));
+          col.denseOffsets = 1;
+        } else {
+          col.data = devAlloc(ex, std::max<int64_t>(n * col.elemSize, 1));
+        }
+        if (!col.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        col.nullBitmap = nullptr;
+        col.hasNulls = 0;
+      }
+      rc = gxp::gxLaunchTpchGen(0, tab, b.tpchRowOffset, n, b.tpchSeed,
+                                totalRows, ex->stream);
+      break;
+    }
+    case GX_TPCH_ORDERS: {
+      for (int c = 0; c < 4; c++) {
+        tab->cols[c].data = devAlloc(ex, std::max<int64_t>(n * 8, 1));
+        if (!tab->cols[c].data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        tab->cols[c].nullBitmap = nullptr;
+        tab->cols[c].hasNulls = 0;
+      }
+      rc = gxp::gxGenOrders(tab, b.tpchRowOffset, n, b.tpchSeed, totalRows,
+                            ex->stream);
+      break;
+    }
+    case GX_TPCH_CUSTOMER: {
+      tab->cols[0].data = devAlloc(ex, std::max<int64_t>(n * 8, 1));
+      tab->cols[1].offsets = (int64_t*)devAlloc(ex, (n + 1) * 8);
+      if (!tab->cols[0].data || !tab->cols[1].offsets) {
+        ex->err = "hipMalloc failed";
+        return GX_ERR_INTERNAL;
+      }
+      long long totalBytes = 0;
+      rc = gxp::gxGenCustomerOffsets(tab, b.tpchRowOffset, n, b.tpchSeed,
+                                     ex->stream, &totalBytes);
+      if (rc != 0) break;
+      tab->cols[1].data = devAlloc(ex, std::max<long long>(totalBytes, 1));
+      if (!tab->cols[1].data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      rc = gxp::gxGenCustomerFill(tab, b.tpchRowOffset, n, b.tpchSeed, ex->stream);
+      tab->cols[0].nullBitmap = tab->cols[1].nullBitmap = nullptr;
+      tab->cols[0].hasNulls = tab->cols[1].hasNulls = 0;
+      tab->cols[1].denseOffsets = 0;
+      break;
+    }
+    default:
+      ex->err = "unknown synthetic table";
+      return GX_ERR_INVALID;
+  }
+  if (rc != 0) {
+    ex->err = std::string("generator failed: ") +
+              hipGetErrorString((hipError_t)rc);
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  return GX_OK;
+}
 
 static int32_t materializeDevice(gx_exec* ex) {
   if (ex->deviceReady) return GX_OK;
@@ -891,6 +1468,258 @@ static int32_t runFused(gx_exec* ex) {
         row.push_back(std::move(v));
       }
     }
+    ex->resultRows.push_back(std::move(row));
+  }
+  return GX_OK;
+}
+
+// ---------------- join-aggregate execution (Q3 class) ----------------
+
+static int ceilLog2(uint64_t v) {
+  int l = 0;
+  while ((1ULL << l) < v) l++;
+  return l;
+}
+
+static int32_t runJoinAgg(gx_exec* ex) {
+  gxp::JoinAggDesc& ja = ex->ja;
+  if (!ex->deviceReady) {
+    if (!gpuAvailable()) {
+      ex->err = "no MI355X visible: the product engine has no CPU fallback "
+                "(GX_ERR_NO_GPU)";
+      return GX_ERR_NO_GPU;
+    }
+    if (ex->device >= 0) hipSetDevice(ex->device);
+    HIP_OK(ex, hipStreamCreate(&ex->stream));
+    int32_t rc = materializeTable(ex, ex->jaSrcCust, &ja.build0);
+    if (rc) return rc;
+    rc = materializeTable(ex, ex->jaSrcOrd, &ja.build1);
+    if (rc) return rc;
+    rc = materializeTable(ex, ex->jaSrcLi, &ja.probe);
+    if (rc) return rc;
+    ex->devErr = (uint32_t*)devAlloc(ex, 4);
+    ja.counters = (uint64_t*)devAlloc(ex, 3 * 8);
+    ex->devJa = (gxp::JoinAggDesc*)devAlloc(ex, sizeof(gxp::JoinAggDesc));
+    if (!ex->devErr || !ja.counters || !ex->devJa) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+    ja.errorFlag = ex->devErr;
+    ex->deviceReady = true;
+  }
+  HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(ja.counters, 0, 24, ex->stream));
+
+  auto pushDesc = [&]() -> int32_t {
+    HIP_OK(ex, hipMemcpyAsync(ex->devJa, &ja, sizeof(ja),
+                              hipMemcpyHostToDevice, ex->stream));
+    return GX_OK;
+  };
+  auto phase = [&](int ph) -> int32_t {
+    int rc = gxp::gxJoinAggPhase(ph, ex->devJa, ja, ex->stream);
+    if (rc != 0) {
+      ex->err = std::string("join phase launch failed: ") +
+                hipGetErrorString((hipError_t)rc);
+      return GX_ERR_INTERNAL;
+    }
+    return GX_OK;
+  };
+  auto readCounter = [&](int i, uint64_t* out) -> int32_t {
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    HIP_OK(ex, hipMemcpy(out, ja.counters + i, 8, hipMemcpyDeviceToHost));
+    return GX_OK;
+  };
+
+  // build0: count, size the key set, fill
+  int32_t rc = pushDesc();
+  if (rc) return rc;
+  if ((rc = phase(0))) return rc;
+  uint64_t n0 = 0;
+  if ((rc = readCounter(0, &n0))) return rc;
+  int wantLog2 = ceilLog2(std::max<uint64_t>(2 * n0 + 1, 64));
+  if (ja.keySet == nullptr || ja.keySetLog2 != wantLog2) {
+    ja.keySet = (uint64_t*)devAlloc(ex, (1ULL << wantLog2) * 8);
+    if (!ja.keySet) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    ja.keySetLog2 = wantLog2;
+  }
+  HIP_OK(ex, hipMemsetAsync(ja.keySet, 0xFF, (1ULL << ja.keySetLog2) * 8,
+                            ex->stream));
+  if ((rc = pushDesc())) return rc;
+  if ((rc = phase(1))) return rc;
+
+  // build1: count qualifying, size slots, init, fill
+  if ((rc = phase(2))) return rc;
+  uint64_t n1 = 0;
+  if ((rc = readCounter(1, &n1))) return rc;
+  wantLog2 = ceilLog2(std::max<uint64_t>(2 * n1 + 1, 64));
+  if (ja.slots == nullptr || ja.slotsLog2 != wantLog2) {
+    ja.slots = (gxp::JoinAggSlot*)devAlloc(
+        ex, (1ULL << wantLog2) * sizeof(gxp::JoinAggSlot));
+    if (!ja.slots) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    ja.slotsLog2 = wantLog2;
+  }
+  if ((rc = pushDesc())) return rc;
+  if ((rc = phase(5))) return rc;  // init slots
+  if ((rc = phase(3))) return rc;  // build
+
+  // probe (timed — the dominant scan)
+  hipEvent_t ev0, ev1;
+  HIP_OK(ex, hipEventCreate(&ev0));
+  HIP_OK(ex, hipEventCreate(&ev1));
+  HIP_OK(ex, hipEventRecord(ev0, ex->stream));
+  if ((rc = phase(4))) return rc;
+  HIP_OK(ex, hipEventRecord(ev1, ex->stream));
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  {
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    ex->lastKernelMs = ms;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+  }
+  uint32_t errFlag = 0;
+  HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+  if (errFlag == 256u && !ja.wide) {
+    // narrow overflow: rebuild the accumulators and rerun wide
+    ja.wide = 1;
+    HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+    HIP_OK(ex, hipMemsetAsync(ja.counters + 2, 0, 8, ex->stream));
+    if ((rc = pushDesc())) return rc;
+    if ((rc = phase(5))) return rc;
+    if ((rc = phase(3))) return rc;
+    if ((rc = phase(4))) return rc;
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+  }
+  errFlag &= ~256u;
+  if (errFlag != 0) {
+    ex->err = "device join error flag 0x" + std::to_string(errFlag);
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipMemcpy(&ex->lastSelCount, ja.counters + 2, 8,
+                       hipMemcpyDeviceToHost));
+
+  // top-N selection: max -> 4096-bucket histogram -> threshold -> compact
+  uint64_t* devMax = (uint64_t*)devAlloc(ex, 8);
+  uint32_t* devHist = (uint32_t*)devAlloc(ex, 4096 * 4);
+  uint64_t* devCount = (uint64_t*)devAlloc(ex, 8);
+  if (!devMax || !devHist || !devCount) {
+    ex->err = "hipMalloc failed";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipMemsetAsync(devMax, 0, 8, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(devHist, 0, 4096 * 4, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(devCount, 0, 8, ex->stream));
+  if (gxp::gxJoinAggMax(ex->devJa, ja, devMax, ex->stream) != 0) {
+    ex->err = "max kernel failed";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  uint64_t maxRev = 0;
+  HIP_OK(ex, hipMemcpy(&maxRev, devMax, 8, hipMemcpyDeviceToHost));
+  HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+  if (errFlag != 0) {
+    // an accumulator exceeded 64 bits: the histogram shortcut is invalid
+    ex->err = "join revenue exceeds 64-bit fast top-N (next round: 128-bit select)";
+    return GX_ERR_INTERNAL;
+  }
+  int shift = 0;
+  {
+    int bits = 0;
+    while ((maxRev >> bits) != 0) bits++;
+    shift = bits > 12 ? bits - 12 : 0;
+  }
+  if (gxp::gxJoinAggHist(ex->devJa, ja, devHist, shift, ex->stream) != 0) {
+    ex->err = "hist kernel failed";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  std::vector<uint32_t> hist(4096);
+  HIP_OK(ex, hipMemcpy(hist.data(), devHist, 4096 * 4, hipMemcpyDeviceToHost));
+  uint64_t need = (uint64_t)(ex->jaLimit + ex->jaOffset);
+  uint64_t cum = 0;
+  int thresholdBucket = 0;
+  for (int b = 4095; b >= 0; b--) {
+    cum += hist[b];
+    if (cum >= need) {
+      thresholdBucket = b;
+      break;
+    }
+  }
+  uint64_t cap = cum;
+  std::vector<gxp::TopNOut> cand;
+  if (cap > 0) {
+    gxp::TopNOut* devOut =
+        (gxp::TopNOut*)devAlloc(ex, cap * sizeof(gxp::TopNOut));
+    if (!devOut) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    if (gxp::gxJoinAggCompact(ex->devJa, ja, devOut, devCount,
+                              (uint64_t)thresholdBucket, shift, cap,
+                              ex->stream) != 0) {
+      ex->err = "compact kernel failed";
+      return GX_ERR_INTERNAL;
+    }
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    uint64_t got = 0;
+    HIP_OK(ex, hipMemcpy(&got, devCount, 8, hipMemcpyDeviceToHost));
+    if (got > cap) got = cap;
+    cand.resize(got);
+    HIP_OK(ex, hipMemcpy(cand.data(), devOut, got * sizeof(gxp::TopNOut),
+                         hipMemcpyDeviceToHost));
+  }
+
+  // host: sort candidates by the TopN keys, slice, emit
+  size_t nGroup = ex->jaGroupSrc.size();
+  auto groupVal = [&](const gxp::TopNOut& c, size_t g) -> uint64_t {
+    switch (ex->jaGroupSrc[g]) {
+      case 0: return c.key;
+      case 1: return c.payload0;
+      default: return (uint64_t)c.payload1;
+    }
+  };
+  auto cmpKeys = [&](const gxp::TopNOut& a, const gxp::TopNOut& b) -> bool {
+    for (auto& [col, desc] : ex->jaSortKeys) {
+      int cres = 0;
+      if (col == (int)nGroup) {  // the sum column: int128 compare
+        __int128 x = ((__int128)a.accHi << 64) | a.accLo;
+        __int128 y = ((__int128)b.accHi << 64) | b.accLo;
+        cres = x < y ? -1 : (x > y ? 1 : 0);
+      } else {
+        int t = ex->jaGroupType[col];
+        uint64_t x = groupVal(a, col), y = groupVal(b, col);
+        if (t == GX_TYPE_TIME) {
+          x &= ~0xFULL;
+          y &= ~0xFULL;
+          cres = x < y ? -1 : (x > y ? 1 : 0);
+        } else {
+          int64_t sx = (int64_t)x, sy = (int64_t)y;
+          cres = sx < sy ? -1 : (sx > sy ? 1 : 0);
+        }
+      }
+      if (desc) cres = -cres;
+      if (cres != 0) return cres < 0;
+    }
+    return false;
+  };
+  size_t want = std::min<size_t>(cand.size(), (size_t)(ex->jaLimit + ex->jaOffset));
+  std::partial_sort(cand.begin(), cand.begin() + want, cand.end(), cmpKeys);
+  ex->resultRows.clear();
+  for (size_t i = (size_t)ex->jaOffset; i < want; i++) {
+    const gxp::TopNOut& c = cand[i];
+    std::vector<OutRowVal> row;
+    for (size_t g = 0; g < nGroup; g++) {
+      OutRowVal v;
+      v.type = ex->jaGroupType[g];
+      v.u64 = groupVal(c, g);
+      v.i64 = (int64_t)v.u64;
+      row.push_back(std::move(v));
+    }
+    OutRowVal v;
+    v.type = GX_TYPE_DECIMAL;
+    __int128 acc = ((__int128)c.accHi << 64) | c.accLo;
+    v.dec = decFromUnits(acc, ex->jaValueScale);
+    int aggFrac = ex->plan.nodes[ex->plan.nodes[ex->root].child].aggFracs[0];
+    v.dec.Round(&v.dec, aggFrac, gxp::ModeHalfUp);
+    row.push_back(std::move(v));
     ex->resultRows.push_back(std::move(row));
   }
   return GX_OK;
@@ -1308,6 +2137,10 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     int32_t rc = compileFused(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     (void)rc;
+  } else if (rn.kind == PK_TOPN) {
+    int32_t rc = compileJoinAgg(ex);
+    if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";
+    (void)rc;
   } else if (rn.kind == PK_SOURCE) {
     ex->isBareSource = true;
     ex->sourceNode = root;
@@ -1380,7 +2213,7 @@ int32_t gx_bind_tpch(gx_exec* ex, int32_t source_node, int32_t table,
 int32_t gx_open(gx_exec* ex) {
   if (!ex) return GX_ERR_INVALID;
   if (!ex->err.empty()) return GX_ERR_INVALID;
-  if (!ex->isFused && !ex->isFinalHost && !ex->isBareSource) {
+  if (!ex->isFused && !ex->isFinalHost && !ex->isBareSource && !ex->isJoinAgg) {
     ex->err = "plan not executable";
     return GX_ERR_INVALID;
   }
@@ -1396,7 +2229,9 @@ int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   if (!ex || !ex->opened) return GX_ERR_INVALID;
   if (ex->isBareSource) return emitSourceChunk(ex, out, rows_out);
   if (!ex->ranQuery) {
-    int32_t rc = ex->isFinalHost ? runFinalHost(ex) : runFused(ex);
+    int32_t rc = ex->isFinalHost ? runFinalHost(ex)
+                 : ex->isJoinAgg ? runJoinAgg(ex)
+                                 : runFused(ex);
     if (rc) {
       *rows_out = 0;
       return rc;

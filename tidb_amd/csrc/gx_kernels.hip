@@ -16,6 +16,7 @@
 // with overflow detection -> error flag (never silent). Equivalence to the
 // word-based MyDecimal arithmetic is covered by tests/golden + parity suites.
 #include <hip/hip_runtime.h>
+#include <hipcub/hipcub.hpp>
 
 #include "gx_common.h"
 
@@ -333,12 +334,12 @@ struct RawState {
 // phase A: issue every fetch for one row, no consumption (loads overlap).
 // The loop is unrolled over the compile-time slot bound so every raw.set has
 // a literal index — a runtime-indexed store would be re-rolled into scratch.
-__device__ __attribute__((always_inline)) inline void fetchRow(const FusedQueryDesc& d, int64_t row, RawState& raw) {
+__device__ __attribute__((always_inline)) inline void fetchRow(const DevTable& tab, const FetchDesc* fetch, int nFetch, int64_t row, RawState& raw) {
 #pragma unroll
   for (int f = 0; f < kMaxFetch; f++) {
-    if (f >= d.nFetch) break;
-    const FetchDesc& fd = d.fetch[f];
-    const DevCol& c = d.table.cols[fd.col];
+    if (f >= nFetch) break;
+    const FetchDesc& fd = fetch[f];
+    const DevCol& c = tab.cols[fd.col];
     ulonglong2 v;
     if (fd.kind == FETCH_8B) {
       v.x = ((const uint64_t*)c.data)[row];
@@ -639,7 +640,7 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   for (int64_t row = begin + threadIdx.x; row < end && !failed;
        row += blockDim.x) {
     RawState raw;
-    fetchRow(d, row, raw);
+    fetchRow(d.table, d.fetch, d.nFetch, row, raw);
     if (!processRow<WIDE>(d, row, raw, lds, &mySel)) failed = true;
   }
 
@@ -690,6 +691,418 @@ __global__ void initGlobalTableKernel(GroupSlot* table, int n) {
   }
 }
 
+
+// ==================================================================
+// join-aggregate pipeline (Q3 class) — see gx_common.h JoinAggDesc
+// ==================================================================
+
+// simple single-table predicate (direct loads; build phases are cheap scans)
+__device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
+                                      const uint8_t* strConst, int strConstLen,
+                                      int64_t row) {
+  const DevCol& c = tab.cols[pd.col];
+  if (colIsNull(c, row)) return false;
+  if (pd.kind == PRED_TIME_CMP_CONST) {
+    uint64_t v = ((const uint64_t*)c.data)[row] & ~0xFULL;
+    uint64_t k = pd.constU64 & ~0xFULL;
+    return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+  }
+  if (pd.kind == PRED_I64_CMP_CONST) {
+    int64_t v = ((const int64_t*)c.data)[row];
+    int64_t k = (int64_t)pd.constU64;
+    return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+  }
+  if (pd.kind == PRED_STR_EQ_CONST) {
+    int64_t st, en;
+    if (c.denseOffsets) { st = row; en = row + 1; }
+    else { st = c.offsets[row]; en = c.offsets[row + 1]; }
+    const uint8_t* p = (const uint8_t*)c.data;
+    while (en > st && p[en - 1] == ' ') en--;  // PAD SPACE
+    int len = (int)(en - st);
+    bool eq = len == strConstLen;
+    for (int j = 0; j < len && eq; j++) eq = p[st + j] == strConst[j];
+    return cmpResult(eq ? 0 : 1, pd.cmp);
+  }
+  return false;
+}
+
+__device__ inline uint64_t hashKey(uint64_t k) { return splitmix64(k); }
+
+// count rows of build0 passing its predicate
+__global__ void jaCountBuild0Kernel(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.build0.nRows;
+  uint64_t my = 0;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool pass = d.nPred0 == 0 ||
+                evalSimplePred(d.build0, d.pred0, d.strConst, d.strConstLen, row);
+    if (pass) my++;
+  }
+  for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
+  if ((threadIdx.x & 63) == 0 && my)
+    atomicAdd((unsigned long long*)&d.counters[0], (unsigned long long)my);
+}
+
+// insert passing build0 keys into the open-addressed key set
+__global__ void jaBuild0Kernel(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.build0.nRows;
+  uint32_t mask = (1u << d.keySetLog2) - 1;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool pass = d.nPred0 == 0 ||
+                evalSimplePred(d.build0, d.pred0, d.strConst, d.strConstLen, row);
+    if (!pass) continue;
+    if (colIsNull(d.build0.cols[d.b0KeyCol], row)) continue;  // NULL never joins
+    uint64_t key = ((const uint64_t*)d.build0.cols[d.b0KeyCol].data)[row];
+    if (key == kEmptyKey) key = kEmptyKey - 1;
+    uint32_t slot = (uint32_t)(hashKey(key) & mask);
+    for (uint32_t probe = 0; probe <= mask; probe++) {
+      uint64_t cur = d.keySet[slot];
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.keySet[slot],
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & mask;
+      if (probe == mask) atomicOr(d.errorFlag, kErrGlobalFull);
+    }
+  }
+}
+
+__device__ inline bool keySetHas(const JoinAggDesc& d, uint64_t key) {
+  uint32_t mask = (1u << d.keySetLog2) - 1;
+  if (key == kEmptyKey) key = kEmptyKey - 1;
+  uint32_t slot = (uint32_t)(hashKey(key) & mask);
+  for (uint32_t probe = 0; probe <= mask; probe++) {
+    uint64_t cur = d.keySet[slot];
+    if (cur == key) return true;
+    if (cur == kEmptyKey) return false;
+    slot = (slot + 1) & mask;
+  }
+  return false;
+}
+
+// count build1 rows passing pred AND matching the key set
+__global__ void jaCountBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.build1.nRows;
+  uint64_t my = 0;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool pass = d.nPred1 == 0 ||
+                evalSimplePred(d.build1, d.pred1, d.strConst, d.strConstLen, row);
+    if (!pass) continue;
+    const DevCol& kc = d.build1.cols[d.b1ProbeCol];
+    if (colIsNull(kc, row)) continue;
+    if (!keySetHas(d, ((const uint64_t*)kc.data)[row])) continue;
+    my++;
+  }
+  for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
+  if ((threadIdx.x & 63) == 0 && my)
+    atomicAdd((unsigned long long*)&d.counters[1], (unsigned long long)my);
+}
+
+// build the slot table from qualifying build1 rows
+__global__ void jaBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.build1.nRows;
+  uint32_t mask = (1u << d.slotsLog2) - 1;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool pass = d.nPred1 == 0 ||
+                evalSimplePred(d.build1, d.pred1, d.strConst, d.strConstLen, row);
+    if (!pass) continue;
+    const DevCol& pc = d.build1.cols[d.b1ProbeCol];
+    if (colIsNull(pc, row)) continue;
+    if (!keySetHas(d, ((const uint64_t*)pc.data)[row])) continue;
+    const DevCol& kc = d.build1.cols[d.b1KeyCol];
+    if (colIsNull(kc, row)) continue;
+    uint64_t key = ((const uint64_t*)kc.data)[row];
+    if (key == kEmptyKey) key = kEmptyKey - 1;
+    uint64_t pay0 = d.payloadCol0 >= 0
+        ? ((const uint64_t*)d.build1.cols[d.payloadCol0].data)[row] : 0;
+    int64_t pay1 = d.payloadCol1 >= 0
+        ? ((const int64_t*)d.build1.cols[d.payloadCol1].data)[row] : 0;
+    uint32_t slot = (uint32_t)(hashKey(key) & mask);
+    for (uint32_t probe = 0; probe <= mask; probe++) {
+      uint64_t cur = d.slots[slot].key;
+      if (cur == key) {
+        // duplicate build key (one row per orderkey in Q3; general inner join
+        // with duplicate build keys needs chaining — next round)
+        atomicOr(d.errorFlag, kErrBadKey);
+        break;
+      }
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.slots[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey) {
+          d.slots[slot].payload0 = pay0;
+          d.slots[slot].payload1 = pay1;
+          break;
+        }
+        if (prev == key) { atomicOr(d.errorFlag, kErrBadKey); break; }
+      }
+      slot = (slot + 1) & mask;
+      if (probe == mask) atomicOr(d.errorFlag, kErrGlobalFull);
+    }
+  }
+}
+
+// probe: scan the probe table; pred -> VM value -> probe slots -> accumulate.
+// Payload writes may race with probe reads only via the key CAS (published
+// before probing starts: build and probe are separate kernel launches, so
+// visibility comes from the dispatch boundary, not intra-kernel hand-off).
+template <bool WIDE>
+__launch_bounds__(256)
+__global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.probe.nRows;
+  uint32_t mask = (1u << d.slotsLog2) - 1;
+  int64_t per = (n + gridDim.x - 1) / gridDim.x;
+  int64_t begin = (int64_t)blockIdx.x * per;
+  int64_t end = begin + per;
+  if (end > n) end = n;
+  bool failed = false;
+  uint64_t myMatch = 0;
+  for (int64_t row = begin + threadIdx.x; row < end && !failed;
+       row += blockDim.x) {
+    RawState raw;
+    fetchRow(d.probe, d.fetch, d.nFetch, row, raw);
+    // predicate (slot-fetched when planned)
+    bool pass = d.nPredP == 0;
+    if (!pass) {
+      const PredDesc& pd = d.predP;
+      const DevCol& c = d.probe.cols[pd.col];
+      if (colIsNull(c, row)) pass = false;
+      else if (pd.kind == PRED_TIME_CMP_CONST) {
+        uint64_t v = (pd.slot >= 0 ? raw.get(pd.slot).x
+                                   : ((const uint64_t*)c.data)[row]) & ~0xFULL;
+        uint64_t k = pd.constU64 & ~0xFULL;
+        pass = cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+      } else if (pd.kind == PRED_I64_CMP_CONST) {
+        int64_t v = pd.slot >= 0 ? (int64_t)raw.get(pd.slot).x
+                                 : ((const int64_t*)c.data)[row];
+        int64_t k = (int64_t)pd.constU64;
+        pass = cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+      } else {
+        pass = evalSimplePred(d.probe, pd, d.strConst, d.strConstLen, row);
+      }
+    }
+    if (!pass) continue;
+    const DevCol& kc = d.probe.cols[d.pKeyCol];
+    if (colIsNull(kc, row)) continue;
+    uint64_t key = ((const uint64_t*)kc.data)[row];
+    if (key == kEmptyKey) key = kEmptyKey - 1;
+    // probe the slot table (prebuilt: no inserts, miss -> drop row)
+    uint32_t slot = (uint32_t)(hashKey(key) & mask);
+    bool found = false;
+    for (uint32_t probe = 0; probe <= mask; probe++) {
+      uint64_t cur = d.slots[slot].key;
+      if (cur == key) { found = true; break; }
+      if (cur == kEmptyKey) break;
+      slot = (slot + 1) & mask;
+    }
+    if (!found) continue;
+    // VM: compute the summed value
+    VmState<WIDE> vm;
+    vm.nullBits = 0;
+    bool bad = false;
+    bool ovf = false;
+    for (int i = 0; i < d.nIns && !bad; i++) {
+      const VmIns& ins = d.ins[i];
+      switch (ins.op) {
+        case VM_LOAD_DEC: {
+          const DevCol& c = d.probe.cols[ins.a];
+          bool nul = colIsNull(c, row);
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
+          if (!nul) {
+            int sc;
+            if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)) {
+              bad = true;
+              break;
+            }
+            if (sc != ins.b) {
+              if (sc < ins.b) v = VT<WIDE>::scale10(v, ins.b - sc, &ovf);
+              else { atomicOr(d.errorFlag, kErrScale); bad = true; break; }
+            }
+          }
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        case VM_LOAD_I64: {
+          const DevCol& c = d.probe.cols[ins.a];
+          bool nul = colIsNull(c, row);
+          vm.set(ins.dst, nul ? VT<WIDE>::zero()
+                              : VT<WIDE>::fromI64((int64_t)raw.get(ins.c).x, &ovf));
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        case VM_LOAD_CONST: {
+          if (WIDE) {
+            Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
+            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+          } else {
+            int64_t cv = d.constLo[ins.a];
+            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+          }
+          vm.setNull(ins.dst, false);
+          break;
+        }
+        case VM_ADD:
+          vm.set(ins.dst, VT<WIDE>::add(vm.get(ins.a), vm.get(ins.b), &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+          break;
+        case VM_SUB:
+          vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+          break;
+        case VM_MUL: {
+          bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
+          if (!nul) v = VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf);
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        case VM_SCALE_UP:
+          vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+      }
+    }
+    if (ovf) {
+      atomicOr(d.errorFlag, WIDE ? kErrOverflow : kErrRetryWide);
+      failed = true;
+      break;
+    }
+    if (bad) { failed = true; break; }
+    if (vm.isNull(d.valueReg)) continue;  // NULL never enters the sum
+    Int128 v = VT<WIDE>::toAcc(vm.get(d.valueReg));
+    JoinAggSlot* sp = &d.slots[slot];
+    uint64_t old = atomicAdd((unsigned long long*)&sp->accLo, (unsigned long long)v.lo);
+    uint64_t carry = (old + v.lo) < old ? 1 : 0;
+    int64_t hiAdd = v.hi + (int64_t)carry;
+    if (hiAdd != 0)
+      atomicAdd((unsigned long long*)&sp->accHi, (unsigned long long)hiAdd);
+    atomicAdd((unsigned long long*)&sp->cnt, 1ULL);
+    myMatch++;
+  }
+  for (int off = 32; off > 0; off >>= 1) myMatch += __shfl_down(myMatch, off, 64);
+  if ((threadIdx.x & 63) == 0 && myMatch)
+    atomicAdd((unsigned long long*)&d.counters[2], (unsigned long long)myMatch);
+}
+
+// top-N selection support: max accLo (revenues fit u64 in practice; accHi != 0
+// falls back to host full sort via the error-free "big" path)
+__global__ void jaMaxKernel(const JoinAggDesc* __restrict__ dp, uint64_t* outMax) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = 1LL << d.slotsLog2;
+  uint64_t my = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (d.slots[i].key == kEmptyKey || d.slots[i].cnt == 0) continue;
+    if (d.slots[i].accHi != 0) atomicOr(d.errorFlag, kErrOverflow);
+    uint64_t v = d.slots[i].accLo;
+    if (v > my) my = v;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    uint64_t o = __shfl_down(my, off, 64);
+    if (o > my) my = o;
+  }
+  if ((threadIdx.x & 63) == 0) atomicMax((unsigned long long*)outMax, my);
+}
+
+__global__ void jaHistKernel(const JoinAggDesc* __restrict__ dp, uint32_t* hist,
+                             int shift) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = 1LL << d.slotsLog2;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (d.slots[i].key == kEmptyKey || d.slots[i].cnt == 0) continue;
+    atomicAdd(&hist[(d.slots[i].accLo >> shift) & 4095], 1u);
+  }
+}
+
+__global__ void jaCompactKernel(const JoinAggDesc* __restrict__ dp, TopNOut* out,
+                                uint64_t* outCount, uint64_t thresholdBucket,
+                                int shift, uint64_t cap) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = 1LL << d.slotsLog2;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const JoinAggSlot& s = d.slots[i];
+    if (s.key == kEmptyKey || s.cnt == 0) continue;
+    if ((s.accLo >> shift) < thresholdBucket) continue;
+    uint64_t idx = atomicAdd((unsigned long long*)outCount, 1ULL);
+    if (idx >= cap) { atomicOr(d.errorFlag, kErrGlobalFull); continue; }
+    out[idx] = {s.key, s.payload0, s.payload1, s.accLo, s.accHi};
+  }
+}
+
+__global__ void jaInitSlotsKernel(JoinAggSlot* slots, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  slots[i].key = kEmptyKey;
+  slots[i].payload0 = 0;
+  slots[i].payload1 = 0;
+  slots[i].accLo = 0;
+  slots[i].accHi = 0;
+  slots[i].cnt = 0;
+}
+
+// ---- orders / customer generators ----
+__global__ void genOrdersKernel(DevTable tab, int64_t rowBegin, int64_t nRows,
+                                uint64_t seed, int64_t totalRows) {
+  int64_t nCust = totalRows / 10;
+  if (nCust < 1) nCust = 1;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nRows;
+       i += stride) {
+    int64_t row = rowBegin + i;
+    ((int64_t*)tab.cols[0].data)[i] = row + 1;
+    ((int64_t*)tab.cols[1].data)[i] =
+        1 + (int64_t)(fieldRand(seed, row, 1) % (uint64_t)nCust);
+    int y, m, dd;
+    civilFromDays(GX_EPOCH_1992 +
+                      (int64_t)(fieldRand(seed, row, 2) % (uint64_t)GX_ORDERDATE_DAYS),
+                  &y, &m, &dd);
+    ((uint64_t*)tab.cols[2].data)[i] = timeFromDate(y, m, dd);
+    ((int64_t*)tab.cols[3].data)[i] = 0;
+  }
+}
+
+__device__ __constant__ const char kSegChars[5][11] = {
+    "AUTOMOBILE", "BUILDING", "FURNITURE", "MACHINERY", "HOUSEHOLD"};
+__device__ __constant__ const int kSegLens[5] = {10, 8, 9, 9, 9};
+
+__global__ void genCustomerLenKernel(int64_t* lens, int64_t rowBegin,
+                                     int64_t nRows, uint64_t seed) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nRows;
+       i += stride)
+    lens[i] = kSegLens[fieldRand(seed, rowBegin + i, 1) % 5];
+}
+
+__global__ void genCustomerFillKernel(DevTable tab, int64_t rowBegin,
+                                      int64_t nRows, uint64_t seed) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nRows;
+       i += stride) {
+    int64_t row = rowBegin + i;
+    ((int64_t*)tab.cols[0].data)[i] = row + 1;
+    int seg = (int)(fieldRand(seed, row, 1) % 5);
+    int64_t off = tab.cols[1].offsets[i];
+    for (int j = 0; j < kSegLens[seg]; j++)
+      ((uint8_t*)tab.cols[1].data)[off + j] = kSegChars[seg][j];
+  }
+}
+
 // ---------------- host launch wrappers ----------------
 
 int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows,
@@ -735,6 +1148,117 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
 
 int gxLaunchMemset(void* p, int v, size_t n, void* stream) {
   return (int)hipMemsetAsync(p, v, n, (hipStream_t)stream);
+}
+
+static int gridFor(int64_t n) {
+  int g = (int)((n + 255) / 256);
+  if (g > 2048) g = 2048;
+  if (g < 1) g = 1;
+  return g;
+}
+
+int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                   void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  switch (phase) {
+    case 0:
+      hipLaunchKernelGGL(jaCountBuild0Kernel, dim3(gridFor(h.build0.nRows)),
+                         dim3(256), 0, s, devDesc);
+      break;
+    case 1:
+      hipLaunchKernelGGL(jaBuild0Kernel, dim3(gridFor(h.build0.nRows)),
+                         dim3(256), 0, s, devDesc);
+      break;
+    case 2:
+      hipLaunchKernelGGL(jaCountBuild1Kernel, dim3(gridFor(h.build1.nRows)),
+                         dim3(256), 0, s, devDesc);
+      break;
+    case 3:
+      hipLaunchKernelGGL(jaBuild1Kernel, dim3(gridFor(h.build1.nRows)),
+                         dim3(256), 0, s, devDesc);
+      break;
+    case 4:
+      if (h.wide)
+        hipLaunchKernelGGL((jaProbeKernel<true>), dim3(gridFor(h.probe.nRows)),
+                           dim3(256), 0, s, devDesc);
+      else
+        hipLaunchKernelGGL((jaProbeKernel<false>), dim3(gridFor(h.probe.nRows)),
+                           dim3(256), 0, s, devDesc);
+      break;
+    case 5: {
+      int64_t n = 1LL << h.slotsLog2;
+      hipLaunchKernelGGL(jaInitSlotsKernel, dim3((n + 255) / 256), dim3(256), 0, s,
+                         h.slots, n);
+      break;
+    }
+    default:
+      return -1;
+  }
+  return (int)hipGetLastError();
+}
+
+int gxJoinAggMax(const JoinAggDesc* devDesc, const JoinAggDesc& h, uint64_t* devMax,
+                 void* stream) {
+  hipLaunchKernelGGL(jaMaxKernel, dim3(gridFor(1LL << h.slotsLog2)), dim3(256), 0,
+                     (hipStream_t)stream, devDesc, devMax);
+  return (int)hipGetLastError();
+}
+
+int gxJoinAggHist(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                  uint32_t* devHist, int shift, void* stream) {
+  hipLaunchKernelGGL(jaHistKernel, dim3(gridFor(1LL << h.slotsLog2)), dim3(256), 0,
+                     (hipStream_t)stream, devDesc, devHist, shift);
+  return (int)hipGetLastError();
+}
+
+int gxJoinAggCompact(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                     TopNOut* out, uint64_t* outCount, uint64_t thresholdBucket,
+                     int shift, uint64_t cap, void* stream) {
+  hipLaunchKernelGGL(jaCompactKernel, dim3(gridFor(1LL << h.slotsLog2)), dim3(256),
+                     0, (hipStream_t)stream, devDesc, out, outCount,
+                     thresholdBucket, shift, cap);
+  return (int)hipGetLastError();
+}
+
+int gxGenOrders(DevTable* tab, int64_t rowBegin, int64_t nRows, uint64_t seed,
+                int64_t totalRows, void* stream) {
+  hipLaunchKernelGGL(genOrdersKernel, dim3(gridFor(nRows)), dim3(256), 0,
+                     (hipStream_t)stream, *tab, rowBegin, nRows, seed, totalRows);
+  return (int)hipGetLastError();
+}
+
+int gxGenCustomerOffsets(DevTable* tab, int64_t rowBegin, int64_t nRows,
+                         uint64_t seed, void* stream, long long* totalBytes) {
+  hipStream_t s = (hipStream_t)stream;
+  // lens in the offsets buffer's tail is unsafe; use a scratch buffer
+  int64_t* lens = nullptr;
+  if (hipMalloc(&lens, (nRows + 1) * 8) != hipSuccess) return -2;
+  hipLaunchKernelGGL(genCustomerLenKernel, dim3(gridFor(nRows)), dim3(256), 0, s,
+                     lens, rowBegin, nRows, seed);
+  hipMemsetAsync(lens + nRows, 0, 8, s);
+  void* tmp = nullptr;
+  size_t tmpBytes = 0;
+  hipcub::DeviceScan::ExclusiveSum(nullptr, tmpBytes, lens, tab->cols[1].offsets,
+                                   nRows + 1, s);
+  if (hipMalloc(&tmp, tmpBytes) != hipSuccess) { hipFree(lens); return -2; }
+  hipcub::DeviceScan::ExclusiveSum(tmp, tmpBytes, lens, tab->cols[1].offsets,
+                                   nRows + 1, s);
+  long long total = 0;
+  hipError_t e = hipMemcpyAsync(&total, tab->cols[1].offsets + nRows, 8,
+                                hipMemcpyDeviceToHost, s);
+  if (e == hipSuccess) e = hipStreamSynchronize(s);
+  hipFree(tmp);
+  hipFree(lens);
+  if (e != hipSuccess) return (int)e;
+  *totalBytes = total;
+  return (int)hipGetLastError();
+}
+
+int gxGenCustomerFill(DevTable* tab, int64_t rowBegin, int64_t nRows,
+                      uint64_t seed, void* stream) {
+  hipLaunchKernelGGL(genCustomerFillKernel, dim3(gridFor(nRows)), dim3(256), 0,
+                     (hipStream_t)stream, *tab, rowBegin, nRows, seed);
+  return (int)hipGetLastError();
 }
 
 }  // namespace gxp

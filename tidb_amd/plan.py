@@ -239,3 +239,75 @@ def q1_final_plan(lib):
         [GX_TYPE_DECIMAL] * 7 + [GX_TYPE_I64]
     out_fracs = [0, 0, 2, 2, 4, 6, 6, 6, 6, 0]
     return b, src, agg, out_types, out_fracs, part_types, part_fracs
+
+
+# ---- TPC-H Q3 (BASELINE config 3) ----
+# Plan shape pinned by the reference's TPC-H golden
+# (pkg/planner/core/casetest/tpch/tpch_test.go:454-467):
+#   HashJoin(customer filtered on c_mktsegment = 'BUILDING'
+#            ⋈ orders filtered on o_orderdate < 1995-03-15)
+#   ⋈ lineitem filtered on l_shipdate > 1995-03-15
+#   → HashAgg(group by l_orderkey, o_orderdate, o_shippriority;
+#             sum(l_extendedprice * (1 - l_discount)) as revenue)
+#   → TopN(revenue desc, o_orderdate asc, limit 10)
+O_ORDERKEY, O_CUSTKEY, O_ORDERDATE, O_SHIPPRIORITY = range(4)
+C_CUSTKEY, C_MKTSEGMENT = range(2)
+
+
+def q3_plan(lib, limit=10):
+    from tests.gxlib import GX_F_GT, GX_F_EQ, GX_TPCH_ORDERS, GX_TPCH_CUSTOMER
+    b = Builder(lib)
+    cust = b.source(CUSTOMER_TYPES)
+    seg = b.colref(C_MKTSEGMENT, GX_TYPE_STRING)
+    cond_c = b.call(GX_F_EQ, GX_TYPE_I64, 0, seg,
+                    lib.gx_pb_const_str(b.pb, b"BUILDING", 8))
+    sel_c = b.selection(cust, [cond_c])
+
+    orders = b.source(ORDERS_TYPES)
+    odate = b.colref(O_ORDERDATE, GX_TYPE_TIME)
+    cutoff = b.const_time(lib.gx_time_from_date(1995, 3, 15))
+    cond_o = b.call(GX_F_LT, GX_TYPE_I64, 0, odate, cutoff)
+    sel_o = b.selection(orders, [cond_o])
+
+    # join1: build = filtered customer, probe = filtered orders
+    # output cols: [c_custkey, c_mktsegment, o_orderkey, o_custkey,
+    #               o_orderdate, o_shippriority]
+    j1 = b.hashjoin(sel_c, sel_o,
+                    [b.colref(C_CUSTKEY, GX_TYPE_I64)],
+                    [b.colref(O_CUSTKEY, GX_TYPE_I64)])
+
+    li = b.source(LINEITEM_TYPES, LINEITEM_FRACS)
+    sdate = b.colref(L_SHIPDATE, GX_TYPE_TIME)
+    cond_l = b.call(GX_F_GT, GX_TYPE_I64, 0, sdate,
+                    b.const_time(lib.gx_time_from_date(1995, 3, 15)))
+    sel_l = b.selection(li, [cond_l])
+
+    # join2: build = join1 output, probe = filtered lineitem
+    # output cols: join1 cols (6) ++ lineitem cols (8)
+    j2 = b.hashjoin(j1, sel_l,
+                    [b.colref(2, GX_TYPE_I64)],          # o_orderkey in join1
+                    [b.colref(L_ORDERKEY, GX_TYPE_I64)])
+
+    # projection: l_orderkey, o_orderdate, o_shippriority, revenue-term
+    jo_orderkey = b.colref(6 + L_ORDERKEY, GX_TYPE_I64)
+    jo_odate = b.colref(4, GX_TYPE_TIME)
+    jo_prio = b.colref(5, GX_TYPE_I64)
+    price = b.colref(6 + L_EXTPRICE, GX_TYPE_DECIMAL, 2)
+    disc = b.colref(6 + L_DISCOUNT, GX_TYPE_DECIMAL, 2)
+    one = _const_dec_one(lib, b)
+    om_d = b.call(GX_F_MINUS, GX_TYPE_DECIMAL, 2, one, disc)
+    rev = b.call(GX_F_MUL, GX_TYPE_DECIMAL, 4, price, om_d)
+    proj = b.projection(j2, [jo_orderkey, jo_odate, jo_prio, rev])
+
+    from tests.gxlib import GX_AGG_SUM
+    agg = b.hashagg(proj,
+                    [b.colref(0, GX_TYPE_I64), b.colref(1, GX_TYPE_TIME),
+                     b.colref(2, GX_TYPE_I64)],
+                    [(GX_AGG_SUM, b.colref(3, GX_TYPE_DECIMAL, 4), 4)])
+    # agg out: orderkey, orderdate, shippriority, revenue(dec s4)
+    topn = b.topn(agg, [b.colref(3, GX_TYPE_DECIMAL, 4),
+                        b.colref(1, GX_TYPE_TIME)],
+                  [1, 0], limit)
+    out_types = [GX_TYPE_I64, GX_TYPE_TIME, GX_TYPE_I64, GX_TYPE_DECIMAL]
+    out_fracs = [0, 0, 0, 4]
+    return b, (cust, orders, li), topn, out_types, out_fracs

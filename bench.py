@@ -62,6 +62,88 @@ def run_cpu_baseline(rows=8_000_000):
     }
 
 
+def bench_q3(args):
+    """TPC-H Q3 (BASELINE config 3): 3-table join + grouped sum + TopN on one
+    GPU. One step = the full pipeline (customer/orders build + lineitem probe
+    + top-N); tables resident in HBM after the first step."""
+    import ctypes as C
+    from tests.gxlib import (GX_TPCH_CUSTOMER, GX_TPCH_LINEITEM,
+                             GX_TPCH_ORDERS, load_product)
+    from tidb_amd import plan as P
+    lib = load_product()
+    n_li = 6_000_000 * args.sf
+    n_ord = n_li // 4
+    n_cust = n_ord // 10
+    b, (cust, orders, li), topn, out_types, out_fracs = P.q3_plan(lib)
+    ex = b.build(topn, device=int(os.environ.get("LOCAL_RANK", "0")))
+    ex.bind_tpch(cust, GX_TPCH_CUSTOMER, n_cust)
+    ex.bind_tpch(orders, GX_TPCH_ORDERS, n_ord)
+    ex.bind_tpch(li, GX_TPCH_LINEITEM, n_li)
+    lib.gx_last_kernel_ms.restype = C.c_double
+    lib.gx_last_kernel_ms.argtypes = [C.c_void_p]
+
+    def step():
+        ex.open()
+        rows = ex.pull_all(out_types, out_fracs, data_caps=[None] * 4)
+        ex.close()
+        return rows, lib.gx_last_kernel_ms(ex.ex)
+
+    t0 = time.perf_counter()
+    for i in range(args.warmup):
+        rows, _ = step()
+        if i == 0:
+            log(f"[bench] q3 first step (incl. generation of {n_li} lineitem"
+                f" rows): {time.perf_counter()-t0:.1f}s, {len(rows)} rows")
+    t0 = time.perf_counter()
+    kms = []
+    for _ in range(args.steps):
+        rows, k = step()
+        kms.append(k)
+    elapsed = time.perf_counter() - t0
+    total_rows = n_li + n_ord + n_cust
+    value = total_rows * args.steps / elapsed
+    # probe-kernel roofline: algorithmic bytes per lineitem row on the Q3
+    # probe: shipdate 8 + orderkey 8 + extendedprice 40 + discount 40 = 96 B
+    # sequential + random slot traffic (reported separately via counters)
+    probe_bpr = 96
+    avg_kms = sum(kms) / len(kms)
+    achieved = n_li * probe_bpr / (avg_kms / 1000.0) / 1e9 if avg_kms else 0
+    out = {
+        "metric": "tpch_q3_rows_per_sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int128",
+        "data": "synthetic",
+        "config": {
+            "workload": f"tpch_q3_sf{args.sf}_synthetic",
+            "lineitem_rows": n_li,
+            "orders_rows": n_ord,
+            "customer_rows": n_cust,
+            "parallelism": "single-gpu",
+            "probe_bytes_per_row": probe_bpr,
+        },
+        "probe_kernel_ms_avg": avg_kms,
+        "roofline": {
+            "bound": "hbm",
+            "achieved": achieved,
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": achieved / HBM_PEAK_GBS,
+            "traffic": None,
+        },
+        "cpu_baseline": None,
+        "result_rows": len(rows),
+    }
+    print(json.dumps(out), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -69,8 +151,14 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--rows", type=int, default=SF10_ROWS,
                     help="rows per GPU (default SF10)")
+    ap.add_argument("--query", choices=["q1", "q3"], default="q1")
+    ap.add_argument("--sf", type=int, default=100,
+                    help="scale factor for --query q3 (lineitem = 6M x SF)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
+
+    if args.query == "q3":
+        return bench_q3(args)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -167,8 +167,14 @@ def main():
 
     dist = None
     if world > 1:
+        import torch
         import torch.distributed as tdist
-        tdist.init_process_group("nccl")  # = RCCL on ROCm
+        backend = os.environ.get("GX_BENCH_BACKEND", "nccl")  # nccl = RCCL
+        if torch.cuda.is_available():
+            # identity on a full node; lets a 2-rank smoke run on 1 GPU (gloo)
+            local_rank = local_rank % torch.cuda.device_count()
+            torch.cuda.set_device(local_rank)
+        tdist.init_process_group(backend)
         dist = tdist
 
     from tests.gxlib import (GX_AGG_MODE_COMPLETE, GX_AGG_MODE_PARTIAL,

@@ -196,9 +196,13 @@ def main():
     lib.gx_last_sel_count.restype = ctypes.c_int64
     lib.gx_last_sel_count.argtypes = [ctypes.c_void_p]
 
+    from tidb_amd.chunkpy import PyChunk
+    bench_chunk = PyChunk(out_types, 1024, out_fracs, caps)
+
     def step():
         ex.open()
-        rows = ex.pull_all(out_types, out_fracs, data_caps=caps)
+        rows = ex.pull_all(out_types, out_fracs, data_caps=caps,
+                           reuse_chunk=bench_chunk)
         ex.close()
         if world > 1:
             gathered = [None] * world

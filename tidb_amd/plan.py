@@ -117,11 +117,19 @@ class Executor:
     def error(self):
         return self.lib.gx_last_error(self.ex).decode()
 
-    def pull_all(self, out_types, out_fracs=None, max_rows=1024, data_caps=None):
-        """Drive Next until EOF; returns list of row tuples (decoded)."""
+    def pull_all(self, out_types, out_fracs=None, max_rows=1024, data_caps=None,
+                 reuse_chunk=None):
+        """Drive Next until EOF; returns list of row tuples (decoded).
+        reuse_chunk: a PyChunk from a previous call (avoids buffer realloc
+        in benchmark loops)."""
         rows = []
         while True:
-            chunk = PyChunk(out_types, max_rows, out_fracs, data_caps)
+            if reuse_chunk is not None:
+                chunk = reuse_chunk
+                for col in chunk.columns:
+                    col.length = 0
+            else:
+                chunk = PyChunk(out_types, max_rows, out_fracs, data_caps)
             g = chunk.as_gx()
             n = ctypes.c_int32(0)
             rc = self.lib.gx_next(self.ex, ctypes.byref(g), ctypes.byref(n))

@@ -93,7 +93,7 @@ struct FetchDesc {
   int32_t col;
 };
 
-constexpr int kMaxFetch = 8;
+constexpr int kMaxFetch = 6;
 
 // ---- aggregation ----
 // Per-group state layout (all aggs): int128 acc + int64 count per agg slot.

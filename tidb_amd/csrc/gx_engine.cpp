@@ -749,7 +749,11 @@ static int32_t compileJoinAgg(gx_exec* ex) {
   if (reg < 0) return GX_ERR_INVALID;
   ja.valueReg = reg;
   ex->jaValueScale = sc;
-  // pred column fetch slot + key fetch slot (probe-side; via the VM fetch plan)
+  // lazy value loads: only the FILTER column is prefetched — the VM runs
+  // only for rows that pass the filter AND hit the hash table (a small
+  // fraction), so its columns are loaded at use (ins.c = -1)
+  ja.nFetch = 0;
+  for (int i = 0; i < ja.nIns; i++) ja.ins[i].c = -1;
   if (ja.nPredP &&
       (ja.predP.kind == gxp::PRED_TIME_CMP_CONST ||
        ja.predP.kind == gxp::PRED_I64_CMP_CONST))

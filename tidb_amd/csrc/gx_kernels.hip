@@ -315,18 +315,17 @@ __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* 
 // raw per-row fetch buffer: setters use compile-time slot indices (phase A),
 // the getter is a wave-uniform switch (runtime-indexed arrays would spill)
 struct RawState {
-  ulonglong2 s0, s1, s2, s3, s4, s5, s6, s7;
+  ulonglong2 s0, s1, s2, s3, s4, s5;
   __device__ ulonglong2 get(int i) const {
     switch (i) {
       case 0: return s0; case 1: return s1; case 2: return s2; case 3: return s3;
-      case 4: return s4; case 5: return s5; case 6: return s6; default: return s7;
+      case 4: return s4; default: return s5;
     }
   }
   __device__ void set(int i, ulonglong2 v) {
     switch (i) {
       case 0: s0 = v; break; case 1: s1 = v; break; case 2: s2 = v; break;
-      case 3: s3 = v; break; case 4: s4 = v; break; case 5: s5 = v; break;
-      case 6: s6 = v; break; default: s7 = v; break;
+      case 3: s3 = v; break; case 4: s4 = v; break; default: s5 = v; break;
     }
   }
 };
@@ -917,12 +916,18 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
       const VmIns& ins = d.ins[i];
       switch (ins.op) {
         case VM_LOAD_DEC: {
+          // lazy: only filter-surviving, join-matching rows reach the VM, so
+          // value columns are loaded here (~5-10% of rows), not prefetched
           const DevCol& c = d.probe.cols[ins.a];
           bool nul = colIsNull(c, row);
           typename VT<WIDE>::T v = VT<WIDE>::zero();
           if (!nul) {
             int sc;
-            if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)) {
+            bool okp = ins.c >= 0
+                ? parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)
+                : loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &v,
+                                         &sc, d.errorFlag);
+            if (!okp) {
               bad = true;
               break;
             }
@@ -938,8 +943,11 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
         case VM_LOAD_I64: {
           const DevCol& c = d.probe.cols[ins.a];
           bool nul = colIsNull(c, row);
-          vm.set(ins.dst, nul ? VT<WIDE>::zero()
-                              : VT<WIDE>::fromI64((int64_t)raw.get(ins.c).x, &ovf));
+          int64_t lv = 0;
+          if (!nul)
+            lv = ins.c >= 0 ? (int64_t)raw.get(ins.c).x
+                            : ((const int64_t*)c.data)[row];
+          vm.set(ins.dst, nul ? VT<WIDE>::zero() : VT<WIDE>::fromI64(lv, &ovf));
           vm.setNull(ins.dst, nul);
           break;
         }

@@ -212,6 +212,11 @@ struct JoinAggDesc {
   int32_t keySetLog2 = 0;
   JoinAggSlot* slots = nullptr;
   int32_t slotsLog2 = 0;
+  // Bloom filter over the build keys (2 hashes; sized ~8 bits/key so it stays
+  // L2-resident): rejects the ~90% non-matching probes without touching the
+  // HBM-random slot table
+  uint32_t* bloom = nullptr;
+  int32_t bloomLog2 = 0;  // log2(bits)
   uint64_t* counters = nullptr;  // [0] build0 pass, [1] build1 pass, [2] probe match
   uint32_t* errorFlag = nullptr;
   int32_t wide = 0;

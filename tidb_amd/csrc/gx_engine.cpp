@@ -1562,6 +1562,13 @@ static int32_t runJoinAgg(gx_exec* ex) {
     if (!ja.slots) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
     ja.slotsLog2 = wantLog2;
   }
+  int bloomLog2 = ceilLog2(std::max<uint64_t>(8 * n1 + 1, 1024));
+  if (ja.bloom == nullptr || ja.bloomLog2 != bloomLog2) {
+    ja.bloom = (uint32_t*)devAlloc(ex, (1ULL << bloomLog2) / 8);
+    if (!ja.bloom) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    ja.bloomLog2 = bloomLog2;
+  }
+  HIP_OK(ex, hipMemsetAsync(ja.bloom, 0, (1ULL << ja.bloomLog2) / 8, ex->stream));
   if ((rc = pushDesc())) return rc;
   if ((rc = phase(5))) return rc;  // init slots
   if ((rc = phase(3))) return rc;  // build

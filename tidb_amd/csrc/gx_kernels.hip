@@ -727,6 +727,24 @@ __device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
 
 __device__ inline uint64_t hashKey(uint64_t k) { return splitmix64(k); }
 
+__device__ inline void bloomSet(const JoinAggDesc& d, uint64_t key) {
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloomLog2) - 1;
+  uint32_t b1 = (uint32_t)h & mask;
+  uint32_t b2 = (uint32_t)(h >> 32) & mask;
+  atomicOr(&d.bloom[b1 >> 5], 1u << (b1 & 31));
+  atomicOr(&d.bloom[b2 >> 5], 1u << (b2 & 31));
+}
+
+__device__ inline bool bloomMayHave(const JoinAggDesc& d, uint64_t key) {
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloomLog2) - 1;
+  uint32_t b1 = (uint32_t)h & mask;
+  uint32_t b2 = (uint32_t)(h >> 32) & mask;
+  if (!((d.bloom[b1 >> 5] >> (b1 & 31)) & 1)) return false;
+  return ((d.bloom[b2 >> 5] >> (b2 & 31)) & 1) != 0;
+}
+
 // count rows of build0 passing its predicate
 __global__ void jaCountBuild0Kernel(const JoinAggDesc* __restrict__ dp) {
   const JoinAggDesc& d = *dp;
@@ -826,6 +844,7 @@ __global__ void jaBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
         ? ((const uint64_t*)d.build1.cols[d.payloadCol0].data)[row] : 0;
     int64_t pay1 = d.payloadCol1 >= 0
         ? ((const int64_t*)d.build1.cols[d.payloadCol1].data)[row] : 0;
+    bloomSet(d, key);
     uint32_t slot = (uint32_t)(hashKey(key) & mask);
     for (uint32_t probe = 0; probe <= mask; probe++) {
       uint64_t cur = d.slots[slot].key;
@@ -897,6 +916,7 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
     if (colIsNull(kc, row)) continue;
     uint64_t key = ((const uint64_t*)kc.data)[row];
     if (key == kEmptyKey) key = kEmptyKey - 1;
+    if (!bloomMayHave(d, key)) continue;  // L2-resident reject
     // probe the slot table (prebuilt: no inserts, miss -> drop row)
     uint32_t slot = (uint32_t)(hashKey(key) & mask);
     bool found = false;

@@ -1569,6 +1569,7 @@ static int32_t runJoinAgg(gx_exec* ex) {
     ja.bloomLog2 = bloomLog2;
   }
   HIP_OK(ex, hipMemsetAsync(ja.bloom, 0, (1ULL << ja.bloomLog2) / 8, ex->stream));
+  if (getenv("GX_NO_BLOOM")) ja.bloomLog2 = 0;
   if ((rc = pushDesc())) return rc;
   if ((rc = phase(5))) return rc;  // init slots
   if ((rc = phase(3))) return rc;  // build

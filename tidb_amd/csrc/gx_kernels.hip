@@ -737,6 +737,7 @@ __device__ inline void bloomSet(const JoinAggDesc& d, uint64_t key) {
 }
 
 __device__ inline bool bloomMayHave(const JoinAggDesc& d, uint64_t key) {
+  if (d.bloomLog2 == 0) return true;
   uint64_t h = splitmix64(key);
   uint32_t mask = (1u << d.bloomLog2) - 1;
   uint32_t b1 = (uint32_t)h & mask;
@@ -1018,7 +1019,12 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
     int64_t hiAdd = v.hi + (int64_t)carry;
     if (hiAdd != 0)
       atomicAdd((unsigned long long*)&sp->accHi, (unsigned long long)hiAdd);
-    atomicAdd((unsigned long long*)&sp->cnt, 1ULL);
+    // the count atomic exists to mark the group matched; a contribution > 0
+    // already proves that through acc (sums can only return to zero if some
+    // contribution was <= 0, and those always count) -> one atomic per match
+    // on all-positive data
+    if (v.hi < 0 || (v.hi == 0 && v.lo == 0))
+      atomicAdd((unsigned long long*)&sp->cnt, 1ULL);
     myMatch++;
   }
   for (int off = 32; off > 0; off >>= 1) myMatch += __shfl_down(myMatch, off, 64);
@@ -1034,9 +1040,11 @@ __global__ void jaMaxKernel(const JoinAggDesc* __restrict__ dp, uint64_t* outMax
   uint64_t my = 0;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    if (d.slots[i].key == kEmptyKey || d.slots[i].cnt == 0) continue;
-    if (d.slots[i].accHi != 0) atomicOr(d.errorFlag, kErrOverflow);
-    uint64_t v = d.slots[i].accLo;
+    const JoinAggSlot& sm = d.slots[i];
+    if (sm.key == kEmptyKey || (sm.cnt == 0 && sm.accLo == 0 && sm.accHi == 0))
+      continue;
+    if (sm.accHi != 0) atomicOr(d.errorFlag, kErrOverflow);
+    uint64_t v = sm.accLo;
     if (v > my) my = v;
   }
   for (int off = 32; off > 0; off >>= 1) {
@@ -1052,8 +1060,10 @@ __global__ void jaHistKernel(const JoinAggDesc* __restrict__ dp, uint32_t* hist,
   int64_t n = 1LL << d.slotsLog2;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    if (d.slots[i].key == kEmptyKey || d.slots[i].cnt == 0) continue;
-    atomicAdd(&hist[(d.slots[i].accLo >> shift) & 4095], 1u);
+    const JoinAggSlot& sh = d.slots[i];
+    if (sh.key == kEmptyKey || (sh.cnt == 0 && sh.accLo == 0 && sh.accHi == 0))
+      continue;
+    atomicAdd(&hist[(sh.accLo >> shift) & 4095], 1u);
   }
 }
 
@@ -1065,7 +1075,8 @@ __global__ void jaCompactKernel(const JoinAggDesc* __restrict__ dp, TopNOut* out
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     const JoinAggSlot& s = d.slots[i];
-    if (s.key == kEmptyKey || s.cnt == 0) continue;
+    if (s.key == kEmptyKey || (s.cnt == 0 && s.accLo == 0 && s.accHi == 0))
+      continue;
     if ((s.accLo >> shift) < thresholdBucket) continue;
     uint64_t idx = atomicAdd((unsigned long long*)outCount, 1ULL);
     if (idx >= cap) { atomicOr(d.errorFlag, kErrGlobalFull); continue; }

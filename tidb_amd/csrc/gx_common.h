@@ -163,6 +163,10 @@ struct FusedQueryDesc {
   // rows per thread batch in the fetch pipeline (engine-chosen: large fetch
   // plans use a smaller R to stay inside the VGPR budget)
   int32_t rbatch = 2;
+  // 1 = skip the per-workgroup LDS table and aggregate straight into the
+  // global table (engine retries with this set when a workgroup's LDS table
+  // overflows — NDV above kLdsGroups)
+  int32_t noLds = 0;
 };
 
 // ---- join-aggregate pipeline (TPC-H Q3 class) ----

@@ -113,6 +113,7 @@ struct gx_exec {
   bool isFinalHost = false;
   bool isBareSource = false;
   bool isJoinAgg = false;
+  int aggRoot = -1;  // the HASHAGG node the fused kernel implements
   int sourceNode = -1;
   gxp::FusedQueryDesc desc;
   std::vector<std::pair<int, int>> projRegs;  // projection idx -> (reg, scale)
@@ -134,6 +135,11 @@ struct gx_exec {
   int64_t srcPos = 0;
   uint64_t lastSelCount = 0;
   double lastKernelMs = 0;
+
+  // post-aggregation host sort (ORDER BY / TopN over the fused agg output)
+  bool postSort = false;
+  std::vector<std::pair<int, bool>> postSortKeys;  // (agg output col, desc)
+  int64_t postLimit = -1, postOffset = 0;
 
   // join-agg (Q3-class) state
   gxp::JoinAggDesc ja;
@@ -795,6 +801,7 @@ static int32_t compileJoinAgg(gx_exec* ex) {
 // compile the fused Source->[Selection]->[Projection]->HashAgg pipeline
 static int32_t compileFused(gx_exec* ex) {
   const PPlan& plan = ex->plan;
+  ex->aggRoot = ex->root;
   const PNode* agg = &plan.nodes[ex->root];
   const PNode* proj = nullptr;
   const PNode* sel = nullptr;
@@ -1367,6 +1374,13 @@ static int32_t runFused(gx_exec* ex) {
     if (getenv("GX_DEBUG")) fprintf(stderr, "[gx] narrow overflow -> wide retry\n");
     return runFused(ex);
   }
+  if ((errFlag & 16u /*kErrLdsFull*/) && !ex->desc.noLds) {
+    // more groups than the LDS table holds: rerun aggregating directly into
+    // the global table (the init kernel resets it, so the rerun is clean)
+    ex->desc.noLds = 1;
+    if (getenv("GX_DEBUG")) fprintf(stderr, "[gx] LDS table full -> global-direct retry\n");
+    return runFused(ex);
+  }
   errFlag &= ~256u;
   if (ex->desc.ablate != 0 && getenv("GX_DEBUG"))
     fprintf(stderr, "[gx] ABLATE=%d kms=%.3f\n", ex->desc.ablate, ex->lastKernelMs);
@@ -1399,7 +1413,7 @@ static int32_t runFused(gx_exec* ex) {
   if (getenv("GX_DEBUG"))
     fprintf(stderr, "[gx] occupied group slots: %zu\n", occ.size());
 
-  const PNode& agg = ex->plan.nodes[ex->root];
+  const PNode& agg = ex->plan.nodes[ex->aggRoot];
   bool partial = agg.aggMode == GX_AGG_MODE_PARTIAL;
   ex->resultRows.clear();
   // zero-row, no-group-by => one row (count=0, sums NULL)
@@ -1447,7 +1461,7 @@ static int32_t runFused(gx_exec* ex) {
         if (cnt == 0) v.isNull = true;
         else {
           v.dec = decFromUnits(acc, ad.scale);
-          int aggFrac = ex->plan.nodes[ex->root].aggFracs[a];
+          int aggFrac = agg.aggFracs[a];
           v.dec.Round(&v.dec, aggFrac, gxp::ModeHalfUp);  // func_sum.go:203-222
         }
         row.push_back(std::move(v));
@@ -1465,7 +1479,7 @@ static int32_t runFused(gx_exec* ex) {
             ex->err = "avg finalize failed";
             return GX_ERR_INTERNAL;
           }
-          int aggFrac = ex->plan.nodes[ex->root].aggFracs[a];
+          int aggFrac = agg.aggFracs[a];
           res.Round(&res, aggFrac, gxp::ModeHalfUp);
           v.dec = res;
         }
@@ -1733,6 +1747,53 @@ static int32_t runJoinAgg(gx_exec* ex) {
     v.dec.Round(&v.dec, aggFrac, gxp::ModeHalfUp);
     row.push_back(std::move(v));
     ex->resultRows.push_back(std::move(row));
+  }
+
+  if (ex->postSort) {
+    auto cmpVal = [](const OutRowVal& a, const OutRowVal& b) -> int {
+      if (a.isNull || b.isNull) {  // NULL sorts first ascending
+        if (a.isNull && b.isNull) return 0;
+        return a.isNull ? -1 : 1;
+      }
+      switch (a.type) {
+        case GX_TYPE_DECIMAL:
+          return a.dec.Compare(b.dec);
+        case GX_TYPE_TIME: {
+          uint64_t x = a.u64 & ~0xFULL, y = b.u64 & ~0xFULL;
+          return x < y ? -1 : (x > y ? 1 : 0);
+        }
+        case GX_TYPE_STRING: {
+          std::string x = a.str, y = b.str;
+          while (!x.empty() && x.back() == ' ') x.pop_back();
+          while (!y.empty() && y.back() == ' ') y.pop_back();
+          int c = x.compare(y);
+          return c < 0 ? -1 : (c > 0 ? 1 : 0);
+        }
+        case GX_TYPE_F64:
+          return a.f64 < b.f64 ? -1 : (a.f64 > b.f64 ? 1 : 0);
+        default:
+          return a.i64 < b.i64 ? -1 : (a.i64 > b.i64 ? 1 : 0);
+      }
+    };
+    std::stable_sort(ex->resultRows.begin(), ex->resultRows.end(),
+                     [&](const std::vector<OutRowVal>& a,
+                         const std::vector<OutRowVal>& b) {
+                       for (auto& [col, desc] : ex->postSortKeys) {
+                         int c = cmpVal(a[col], b[col]);
+                         if (desc) c = -c;
+                         if (c != 0) return c < 0;
+                       }
+                       return false;
+                     });
+    size_t beginI = std::min<size_t>((size_t)ex->postOffset,
+                                     ex->resultRows.size());
+    size_t endI = ex->postLimit < 0
+                      ? ex->resultRows.size()
+                      : std::min<size_t>(beginI + (size_t)ex->postLimit,
+                                         ex->resultRows.size());
+    std::vector<std::vector<OutRowVal>> sliced(
+        ex->resultRows.begin() + beginI, ex->resultRows.begin() + endI);
+    ex->resultRows = std::move(sliced);
   }
   return GX_OK;
 }
@@ -2149,6 +2210,35 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     int32_t rc = compileFused(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     (void)rc;
+  } else if (rn.kind == PK_TOPN &&
+             ex->plan.nodes[rn.child].kind == PK_HASHAGG &&
+             ex->plan.nodes[rn.child].aggMode != GX_AGG_MODE_FINAL) {
+    // ORDER BY / TopN over a fusable aggregation: run the fused kernel, sort
+    // the (small) group output on the host — the reference sorts the agg
+    // output the same way (Q1's final Sort, sortexec/sort.go)
+    int aggRoot = rn.child;
+    int aggWidth = (int)ex->plan.nodes[aggRoot].exprs.size() +
+                   (int)ex->plan.nodes[aggRoot].aggFuncs.size();
+    bool ok = true;
+    for (size_t i = 0; i < rn.exprs.size(); i++) {
+      const PExpr& ke = ex->plan.exprs[rn.exprs[i]];
+      if (ke.kind != EK_COLREF || ke.colIdx < 0 || ke.colIdx >= aggWidth) {
+        ex->err = "sort keys must be aggregate output columns";
+        ok = false;
+        break;
+      }
+      ex->postSortKeys.push_back({ke.colIdx, rn.keyDesc[i] != 0});
+    }
+    if (ok) {
+      ex->postSort = true;
+      ex->postLimit = rn.limit;  // -1 = full sort
+      ex->postOffset = rn.offset;
+      int saved = ex->root;
+      ex->root = aggRoot;
+      int32_t rc = compileFused(ex);
+      ex->root = saved;
+      if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
+    }
   } else if (rn.kind == PK_TOPN) {
     int32_t rc = compileJoinAgg(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";

@@ -573,35 +573,67 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     return true;
   }
 
-  // ---- group lookup / insert in LDS ----
+  // ---- group lookup / insert (LDS table, or global when NDV > kLdsGroups) ----
   uint64_t key;
   if (!makeGroupKey(d, row, raw, &key, d.errorFlag)) return false;
-  uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
+  GroupSlot* target;
+  if (!d.noLds) {
+    uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
+    for (int probe = 0;; probe++) {
+      if (probe >= kLdsGroups) {
+        atomicOr(d.errorFlag, kErrLdsFull);
+        return false;
+      }
+      uint64_t cur = lds[slot].key;
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & (kLdsGroups - 1);
+    }
+    target = &lds[slot];
+    // ---- update states ----
+    for (int a = 0; a < d.nAggs; a++) {
+      const AggDesc& ad = d.aggs[a];
+      if (ad.func == 0 /*COUNT*/) {
+        bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
+        if (!isNull) accumInto(target, a, Int128{0, 0}, 1);
+      } else {  // SUM / AVG
+        if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
+          accumInto(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
+      }
+    }
+    return true;
+  }
+  // global-direct path (mid/high NDV up to kGlobalGroups)
+  uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
   for (int probe = 0;; probe++) {
-    if (probe >= kLdsGroups) {
-      atomicOr(d.errorFlag, kErrLdsFull);
+    if (probe >= kGlobalGroups) {
+      atomicOr(d.errorFlag, kErrGlobalFull);
       return false;
     }
-    uint64_t cur = lds[slot].key;
+    uint64_t cur = d.globalTable[slot].key;
     if (cur == key) break;
     if (cur == kEmptyKey) {
-      uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
+      uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
                                 (unsigned long long)kEmptyKey,
                                 (unsigned long long)key);
       if (prev == kEmptyKey || prev == key) break;
     }
-    slot = (slot + 1) & (kLdsGroups - 1);
+    slot = (slot + 1) & (kGlobalGroups - 1);
   }
-
-  // ---- update states ----
+  target = &d.globalTable[slot];
   for (int a = 0; a < d.nAggs; a++) {
     const AggDesc& ad = d.aggs[a];
     if (ad.func == 0 /*COUNT*/) {
       bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
-      if (!isNull) accumInto(&lds[slot], a, Int128{0, 0}, 1);
-    } else {  // SUM / AVG
+      if (!isNull) accumInto(target, a, Int128{0, 0}, 1);
+    } else {
       if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
-        accumInto(&lds[slot], a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
+        accumInto(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
     }
   }
   return true;
@@ -654,6 +686,7 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   __syncthreads();
 
   // ---- flush LDS table into the global table ----
+  if (d.noLds) return;
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     if (lds[i].key == kEmptyKey) continue;
     uint64_t key = lds[i].key;

@@ -1338,6 +1338,58 @@ static void decodeGroupLane(gx_exec* ex, uint32_t lane, int kind, int type,
   }
 }
 
+// host sort of the (small) fused-agg output for ORDER BY / TopN roots
+static void applyPostSort(gx_exec* ex) {
+  if (!ex->postSort) return;
+
+  {
+    auto cmpVal = [](const OutRowVal& a, const OutRowVal& b) -> int {
+      if (a.isNull || b.isNull) {  // NULL sorts first ascending
+        if (a.isNull && b.isNull) return 0;
+        return a.isNull ? -1 : 1;
+      }
+      switch (a.type) {
+        case GX_TYPE_DECIMAL:
+          return a.dec.Compare(b.dec);
+        case GX_TYPE_TIME: {
+          uint64_t x = a.u64 & ~0xFULL, y = b.u64 & ~0xFULL;
+          return x < y ? -1 : (x > y ? 1 : 0);
+        }
+        case GX_TYPE_STRING: {
+          std::string x = a.str, y = b.str;
+          while (!x.empty() && x.back() == ' ') x.pop_back();
+          while (!y.empty() && y.back() == ' ') y.pop_back();
+          int c = x.compare(y);
+          return c < 0 ? -1 : (c > 0 ? 1 : 0);
+        }
+        case GX_TYPE_F64:
+          return a.f64 < b.f64 ? -1 : (a.f64 > b.f64 ? 1 : 0);
+        default:
+          return a.i64 < b.i64 ? -1 : (a.i64 > b.i64 ? 1 : 0);
+      }
+    };
+    std::stable_sort(ex->resultRows.begin(), ex->resultRows.end(),
+                     [&](const std::vector<OutRowVal>& a,
+                         const std::vector<OutRowVal>& b) {
+                       for (auto& [col, desc] : ex->postSortKeys) {
+                         int c = cmpVal(a[col], b[col]);
+                         if (desc) c = -c;
+                         if (c != 0) return c < 0;
+                       }
+                       return false;
+                     });
+    size_t beginI = std::min<size_t>((size_t)ex->postOffset,
+                                     ex->resultRows.size());
+    size_t endI = ex->postLimit < 0
+                      ? ex->resultRows.size()
+                      : std::min<size_t>(beginI + (size_t)ex->postLimit,
+                                         ex->resultRows.size());
+    std::vector<std::vector<OutRowVal>> sliced(
+        ex->resultRows.begin() + beginI, ex->resultRows.begin() + endI);
+    ex->resultRows = std::move(sliced);
+  }
+}
+
 static int32_t runFused(gx_exec* ex) {
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;
@@ -1488,6 +1540,7 @@ static int32_t runFused(gx_exec* ex) {
     }
     ex->resultRows.push_back(std::move(row));
   }
+  applyPostSort(ex);
   return GX_OK;
 }
 
@@ -1747,53 +1800,6 @@ static int32_t runJoinAgg(gx_exec* ex) {
     v.dec.Round(&v.dec, aggFrac, gxp::ModeHalfUp);
     row.push_back(std::move(v));
     ex->resultRows.push_back(std::move(row));
-  }
-
-  if (ex->postSort) {
-    auto cmpVal = [](const OutRowVal& a, const OutRowVal& b) -> int {
-      if (a.isNull || b.isNull) {  // NULL sorts first ascending
-        if (a.isNull && b.isNull) return 0;
-        return a.isNull ? -1 : 1;
-      }
-      switch (a.type) {
-        case GX_TYPE_DECIMAL:
-          return a.dec.Compare(b.dec);
-        case GX_TYPE_TIME: {
-          uint64_t x = a.u64 & ~0xFULL, y = b.u64 & ~0xFULL;
-          return x < y ? -1 : (x > y ? 1 : 0);
-        }
-        case GX_TYPE_STRING: {
-          std::string x = a.str, y = b.str;
-          while (!x.empty() && x.back() == ' ') x.pop_back();
-          while (!y.empty() && y.back() == ' ') y.pop_back();
-          int c = x.compare(y);
-          return c < 0 ? -1 : (c > 0 ? 1 : 0);
-        }
-        case GX_TYPE_F64:
-          return a.f64 < b.f64 ? -1 : (a.f64 > b.f64 ? 1 : 0);
-        default:
-          return a.i64 < b.i64 ? -1 : (a.i64 > b.i64 ? 1 : 0);
-      }
-    };
-    std::stable_sort(ex->resultRows.begin(), ex->resultRows.end(),
-                     [&](const std::vector<OutRowVal>& a,
-                         const std::vector<OutRowVal>& b) {
-                       for (auto& [col, desc] : ex->postSortKeys) {
-                         int c = cmpVal(a[col], b[col]);
-                         if (desc) c = -c;
-                         if (c != 0) return c < 0;
-                       }
-                       return false;
-                     });
-    size_t beginI = std::min<size_t>((size_t)ex->postOffset,
-                                     ex->resultRows.size());
-    size_t endI = ex->postLimit < 0
-                      ? ex->resultRows.size()
-                      : std::min<size_t>(beginI + (size_t)ex->postLimit,
-                                         ex->resultRows.size());
-    std::vector<std::vector<OutRowVal>> sliced(
-        ex->resultRows.begin() + beginI, ex->resultRows.begin() + endI);
-    ex->resultRows = std::move(sliced);
   }
   return GX_OK;
 }
@@ -2237,6 +2243,24 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
       ex->root = aggRoot;
       int32_t rc = compileFused(ex);
       ex->root = saved;
+      if (rc != GX_OK) {
+        // not a fusable aggregation subtree (e.g. Q3's agg-over-join):
+        // reset and try the join-aggregate pipeline
+        ex->err.clear();
+        ex->isFused = false;
+        ex->postSort = false;
+        ex->postSortKeys.clear();
+        ex->desc = gxp::FusedQueryDesc{};
+        ex->projRegs.clear();
+        ex->vmNextReg = 0;
+        ex->exprRegCache.clear();
+        rc = compileJoinAgg(ex);
+        if (rc != GX_OK && ex->err.empty())
+          ex->err = "plan compilation failed";
+      }
+    } else {
+      ex->err.clear();
+      int32_t rc = compileJoinAgg(ex);
       if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     }
   } else if (rn.kind == PK_TOPN) {

@@ -86,11 +86,13 @@ enum FetchKind : int32_t {
   FETCH_8B = 0,       // 8-byte element at data + row*8
   FETCH_DEC16 = 1,    // 16 bytes at data + row*40 (decimal header+3 words)
   FETCH_OFFSETS = 2,  // offsets[row], offsets[row+1]
+  FETCH_B1 = 3,       // 1 byte per row (dense char(1); glds variant only)
 };
 
 struct FetchDesc {
   int32_t kind;
   int32_t col;
+  int32_t ldsOff;  // glds variant: byte offset of this stream in a tile buffer
 };
 
 constexpr int kMaxFetch = 6;
@@ -167,6 +169,12 @@ struct FusedQueryDesc {
   // global table (engine retries with this set when a workgroup's LDS table
   // overflows — NDV above kLdsGroups)
   int32_t noLds = 0;
+  // glds (LDS-DMA) staged variant: streams are DMA'd tile-by-tile into LDS
+  // (double-buffered, counted vmcnt waits) so compute overlaps the memory
+  // stream. Engine enables it when every fetch kind is stageable and no
+  // consumed column carries NULLs.
+  int32_t useGlds = 0;
+  int32_t tileBytes = 0;  // per-tile LDS bytes (sum of stream slots, 16-aligned)
 };
 
 // ---- join-aggregate pipeline (TPC-H Q3 class) ----

@@ -892,6 +892,7 @@ static int32_t compileFused(gx_exec* ex) {
         }
         pd.kind = gxp::PRED_DEC_CMP_CONST;
         pd.constU64 = (uint64_t)(int64_t)u;
+        pd.slot = fetchSlot(ex, gxp::FETCH_DEC16, lhs->colIdx);
       } else {
         ex->err = "unsupported filter column/const type combination";
         return GX_ERR_INVALID;
@@ -1029,6 +1030,19 @@ static int32_t compileFused(gx_exec* ex) {
   }
   ex->isFused = true;
   return GX_OK;
+}
+
+// assign per-stream LDS offsets for the glds-staged kernel (64-row tiles)
+static void assignGldsOffsets(gxp::FusedQueryDesc& d) {
+  int off = 0;
+  for (int f = 0; f < d.nFetch; f++) {
+    d.fetch[f].ldsOff = off;
+    int bytes = d.fetch[f].kind == gxp::FETCH_DEC16 ? 1024
+                : d.fetch[f].kind == gxp::FETCH_8B ? 512
+                                                   : 64;
+    off += (bytes + 15) & ~15;
+  }
+  d.tileBytes = off;
 }
 
 // ---------------- device materialization ----------------
@@ -1292,7 +1306,7 @@ static int32_t materializeDevice(gx_exec* ex) {
       const gxp::DevCol& c = tab.cols[ex->desc.gkey.col[k]];
       if (c.denseOffsets) {
         ex->desc.gkey.kind[k] = 2;
-        ex->desc.gkey.slot[k] = -1;
+        ex->desc.gkey.slot[k] = fetchSlot(ex, gxp::FETCH_B1, ex->desc.gkey.col[k]);
       } else {
         ex->desc.gkey.kind[k] = 0;
         ex->desc.gkey.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS,
@@ -1316,6 +1330,30 @@ static int32_t materializeDevice(gx_exec* ex) {
   ex->desc.globalTable = ex->devTable;
   ex->desc.errorFlag = ex->devErr;
   ex->desc.selCount = ex->devSel;
+
+  // glds-staged variant eligibility: every stream stageable, slots assigned,
+  // no NULLs on any consumed column, enough rows to matter
+  {
+    gxp::FusedQueryDesc& d = ex->desc;
+    bool ok = tab.nRows >= 256 && !getenv("GX_NO_GLDS");
+    for (int f = 0; f < d.nFetch && ok; f++)
+      ok = d.fetch[f].kind == gxp::FETCH_8B ||
+           d.fetch[f].kind == gxp::FETCH_DEC16 ||
+           d.fetch[f].kind == gxp::FETCH_B1;
+    for (int k = 0; k < d.gkey.nCols && ok; k++)
+      ok = d.gkey.kind[k] != 0 && d.gkey.slot[k] >= 0;
+    for (int p = 0; p < d.nPreds && ok; p++) ok = d.preds[p].slot >= 0;
+    for (int c = 0; c < tab.nCols && ok; c++) ok = tab.cols[c].hasNulls == 0;
+    if (ok) {
+      assignGldsOffsets(d);
+      d.useGlds = 1;
+    } else {
+      d.useGlds = 0;
+    }
+    if (getenv("GX_DEBUG"))
+      fprintf(stderr, "[gx] useGlds=%d tileBytes=%d nFetch=%d\n", d.useGlds,
+              d.tileBytes, d.nFetch);
+  }
   ex->deviceReady = true;
   return GX_OK;
 }

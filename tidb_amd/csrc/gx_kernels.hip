@@ -1117,6 +1117,343 @@ __global__ void jaCompactKernel(const JoinAggDesc* __restrict__ dp, TopNOut* out
   }
 }
 
+// ==================================================================
+// glds-staged fused aggregation
+// ==================================================================
+// Per-wave double-buffered tile pipeline: each wave owns tiles of 64 rows;
+// all streams of tile t+1 are DMA'd HBM->LDS (`global_load_lds`, no VGPR
+// round-trip) while tile t is consumed from LDS (parse + VM + aggregate).
+// Counted `s_waitcnt vmcnt(N)` keeps the next tile's DMAs in flight across
+// the compute (guide §5 'Pipelining across barriers'); all LDS lives in ONE
+// dynamic __shared__ carve (a second __shared__ object would make hipcc
+// drain vmcnt before every ds_read — guide §5 'Three .s-level traps').
+
+__device__ inline void gldsWaitVmcnt(int n) {
+  switch (n) {
+    case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+    case 1: asm volatile("s_waitcnt vmcnt(1)" ::: "memory"); break;
+    case 2: asm volatile("s_waitcnt vmcnt(2)" ::: "memory"); break;
+    case 3: asm volatile("s_waitcnt vmcnt(3)" ::: "memory"); break;
+    case 4: asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); break;
+    case 5: asm volatile("s_waitcnt vmcnt(5)" ::: "memory"); break;
+    case 6: asm volatile("s_waitcnt vmcnt(6)" ::: "memory"); break;
+    case 7: asm volatile("s_waitcnt vmcnt(7)" ::: "memory"); break;
+    default: asm volatile("s_waitcnt vmcnt(8)" ::: "memory"); break;
+  }
+}
+
+// issue all stream DMAs for one 64-row tile into tileBuf (wave-uniform base)
+__device__ __attribute__((always_inline)) inline void gldsIssueTile(
+    const FusedQueryDesc& d, int64_t row0, int64_t clampMax, char* tileBuf,
+    int lane) {
+  for (int f = 0; f < d.nFetch; f++) {
+    const FetchDesc& fd = d.fetch[f];
+    const DevCol& c = d.table.cols[fd.col];
+    auto lptr = (__attribute__((address_space(3))) void*)(tileBuf + fd.ldsOff);
+    if (fd.kind == FETCH_DEC16) {
+      int64_t r = row0 + lane;
+      if (r > clampMax) r = clampMax;  // tail rows re-read the last row
+      auto g = (const __attribute__((address_space(1))) void*)
+          ((const uint8_t*)c.data + r * 40);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)g,
+          (__attribute__((address_space(3))) uint32_t*)lptr, 16, 0, 0);
+    } else if (fd.kind == FETCH_8B) {
+      if (lane < 32) {
+        int64_t r = row0 + 2 * lane;
+        if (r > clampMax) r = clampMax;
+        auto g = (const __attribute__((address_space(1))) void*)
+            ((const uint8_t*)c.data + r * 8);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) uint32_t*)g,
+            (__attribute__((address_space(3))) uint32_t*)lptr, 16, 0, 0);
+      }
+    } else {  // FETCH_B1
+      if (lane < 16) {
+        int64_t r = row0 + 4 * lane;
+        if (r > clampMax - 3) r = clampMax - 3 < 0 ? 0 : clampMax - 3;
+        auto g = (const __attribute__((address_space(1))) void*)
+            ((const uint8_t*)c.data + r);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) uint32_t*)g,
+            (__attribute__((address_space(3))) uint32_t*)lptr, 4, 0, 0);
+      }
+    }
+  }
+}
+
+// LDS-backed raw accessor for one row (lane) of the staged tile
+struct LdsRaw {
+  const char* buf;  // generic-address view of the tile buffer
+  const FusedQueryDesc* d;
+  int lane;
+  __device__ ulonglong2 get(int slot) const {
+    const FetchDesc& fd = d->fetch[slot];
+    if (fd.kind == FETCH_DEC16) {
+      return *(const ulonglong2*)(buf + fd.ldsOff + lane * 16);
+    }
+    if (fd.kind == FETCH_8B) {
+      ulonglong2 v;
+      v.x = *(const uint64_t*)(buf + fd.ldsOff + lane * 8);
+      v.y = 0;
+      return v;
+    }
+    ulonglong2 v;
+    v.x = *(const uint8_t*)(buf + fd.ldsOff + lane);
+    v.y = 0;
+    return v;
+  }
+};
+
+// group key from staged streams (dense-char via B1 slot, i64 via 8B slot)
+template <typename RAW>
+__device__ inline bool makeGroupKeyStaged(const FusedQueryDesc& d,
+                                          const RAW& raw, uint64_t* keyOut,
+                                          uint32_t* err) {
+  uint64_t key = 0;
+  for (int k = 0; k < d.gkey.nCols; k++) {
+    uint32_t lane32;
+    if (d.gkey.kind[k] == 2) {
+      uint8_t b = (uint8_t)raw.get(d.gkey.slot[k]).x;
+      lane32 = b == ' ' ? 0u : ((1u << 24) | b);
+    } else {  // kind 1: small i64
+      int64_t v = (int64_t)raw.get(d.gkey.slot[k]).x;
+      if (v < 0 || v > 0x7FFFFFFF) { atomicOr(err, kErrBadKey); return false; }
+      lane32 = (uint32_t)v;
+    }
+    key |= (uint64_t)lane32 << (32 * k);
+  }
+  if (d.gkey.nCols == 0) key = 0;
+  if (key == kEmptyKey) key = kEmptyKey - 1;
+  *keyOut = key;
+  return true;
+}
+
+// row pipeline over a staged tile row (mirrors processRow; RAW = LdsRaw)
+template <bool WIDE>
+__device__ __attribute__((always_inline)) inline bool processRowStaged(
+    const FusedQueryDesc& d, const LdsRaw& raw, GroupSlot* lds,
+    uint64_t* mySel) {
+  bool pass = true;
+  for (int p = 0; p < d.nPreds && pass; p++) {
+    const PredDesc& pd = d.preds[p];
+    if (pd.kind == PRED_TIME_CMP_CONST) {
+      uint64_t v = raw.get(pd.slot).x & ~0xFULL;
+      uint64_t k = pd.constU64 & ~0xFULL;
+      pass = cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+    } else if (pd.kind == PRED_I64_CMP_CONST) {
+      int64_t v = (int64_t)raw.get(pd.slot).x;
+      int64_t k = (int64_t)pd.constU64;
+      pass = cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+    } else {  // decimal pred via its DEC16 slot
+      typename VT<WIDE>::T u;
+      int sc;
+      if (!parseDecimalRaw<WIDE>(raw.get(pd.slot), &u, &sc, d.errorFlag))
+        return false;
+      int cmp = VT<WIDE>::cmp(u, VT<WIDE>::fromI64((int64_t)pd.constU64, nullptr));
+      pass = cmpResult(cmp, pd.cmp);
+    }
+  }
+  if (!pass) return true;
+  (*mySel)++;
+
+  VmState<WIDE> vm;
+  vm.nullBits = 0;
+  bool bad = false;
+  bool ovf = false;
+  for (int i = 0; i < d.nIns && !bad; i++) {
+    const VmIns& ins = d.ins[i];
+    switch (ins.op) {
+      case VM_LOAD_DEC: {
+        typename VT<WIDE>::T v = VT<WIDE>::zero();
+        int sc;
+        if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)) {
+          bad = true;
+          break;
+        }
+        if (sc != ins.b) {
+          if (sc < ins.b) v = VT<WIDE>::scale10(v, ins.b - sc, &ovf);
+          else { atomicOr(d.errorFlag, kErrScale); bad = true; break; }
+        }
+        vm.set(ins.dst, v);
+        vm.setNull(ins.dst, false);
+        break;
+      }
+      case VM_LOAD_I64:
+        vm.set(ins.dst, VT<WIDE>::fromI64((int64_t)raw.get(ins.c).x, &ovf));
+        vm.setNull(ins.dst, false);
+        break;
+      case VM_LOAD_CONST: {
+        if (WIDE) {
+          Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
+          vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+        } else {
+          int64_t cv = d.constLo[ins.a];
+          vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+        }
+        vm.setNull(ins.dst, false);
+        break;
+      }
+      case VM_ADD:
+        vm.set(ins.dst, VT<WIDE>::add(vm.get(ins.a), vm.get(ins.b), &ovf));
+        break;
+      case VM_SUB:
+        vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
+        break;
+      case VM_MUL:
+        vm.set(ins.dst, VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf));
+        break;
+      case VM_SCALE_UP:
+        vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
+        break;
+    }
+  }
+  if (ovf) {
+    atomicOr(d.errorFlag, WIDE ? kErrOverflow : kErrRetryWide);
+    return false;
+  }
+  if (bad) return false;
+  if (d.ablate == 1) {
+    uint64_t sink = 0;
+    for (int a = 0; a < d.nAggs; a++)
+      if (d.aggs[a].srcReg >= 0)
+        sink ^= (uint64_t)VT<WIDE>::toAcc(vm.get(d.aggs[a].srcReg)).lo;
+    asm volatile("" ::"v"(sink));
+    return true;
+  }
+
+  uint64_t key;
+  if (!makeGroupKeyStaged(d, raw, &key, d.errorFlag)) return false;
+  GroupSlot* target;
+  if (!d.noLds) {
+    uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
+    for (int probe = 0;; probe++) {
+      if (probe >= kLdsGroups) { atomicOr(d.errorFlag, kErrLdsFull); return false; }
+      uint64_t cur = lds[slot].key;
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & (kLdsGroups - 1);
+    }
+    target = &lds[slot];
+  } else {
+    uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
+    for (int probe = 0;; probe++) {
+      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, kErrGlobalFull); return false; }
+      uint64_t cur = d.globalTable[slot].key;
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & (kGlobalGroups - 1);
+    }
+    target = &d.globalTable[slot];
+  }
+  for (int a = 0; a < d.nAggs; a++) {
+    const AggDesc& ad = d.aggs[a];
+    if (ad.func == 0) {
+      accumInto(target, a, Int128{0, 0}, 1);
+    } else if (ad.srcReg >= 0) {
+      accumInto(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
+    }
+  }
+  return true;
+}
+
+template <bool WIDE>
+__launch_bounds__(256)
+__global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
+  const FusedQueryDesc& d = *dp;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  GroupSlot* lds = (GroupSlot*)smem;
+  char* tiles = smem + ((sizeof(GroupSlot) * kLdsGroups + 15) & ~15ULL);
+  for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
+    lds[i].key = kEmptyKey;
+    for (int a = 0; a < kMaxAggs; a++) {
+      lds[i].accLo[a] = 0;
+      lds[i].accHi[a] = 0;
+      lds[i].cnt[a] = 0;
+    }
+  }
+  __syncthreads();
+
+  int64_t n = d.table.nRows;
+  int64_t per = (n + gridDim.x - 1) / gridDim.x;
+  int64_t begin = (int64_t)blockIdx.x * per;
+  int64_t end = begin + per;
+  if (end > n) end = n;
+  int wave = threadIdx.x >> 6;
+  int lane = threadIdx.x & 63;
+  char* myTiles = tiles + (size_t)wave * 2 * d.tileBytes;
+  uint64_t mySel = 0;
+  bool failed = false;
+
+  // tiles of 64 rows, strided across the block's 4 waves
+  int64_t range = end - begin;
+  int64_t nTiles = (range + 63) / 64;
+  int64_t t = wave;
+  if (t < nTiles)
+    gldsIssueTile(d, begin + t * 64, end - 1, myTiles, lane);
+  int buf = 0;
+  for (; t < nTiles && !failed; t += 4) {
+    int64_t nextT = t + 4;
+    bool haveNext = nextT < nTiles;
+    if (haveNext)
+      gldsIssueTile(d, begin + nextT * 64, end - 1, myTiles + (buf ^ 1) * d.tileBytes,
+                    lane);
+    // wait for tile t's DMAs (the next tile's stay in flight)
+    gldsWaitVmcnt(haveNext ? d.nFetch : 0);
+    int64_t row = begin + t * 64 + lane;
+    if (row < end) {
+      LdsRaw raw{myTiles + buf * d.tileBytes, &d, lane};
+      if (!processRowStaged<WIDE>(d, raw, lds, &mySel)) failed = true;
+    }
+    buf ^= 1;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  if (d.selCount) {
+    uint64_t total = mySel;
+    for (int off = 32; off > 0; off >>= 1)
+      total += __shfl_down(total, off, 64);
+    if ((threadIdx.x & 63) == 0 && total)
+      atomicAdd((unsigned long long*)d.selCount, (unsigned long long)total);
+  }
+  __syncthreads();
+
+  if (d.noLds) return;
+  for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
+    if (lds[i].key == kEmptyKey) continue;
+    uint64_t key = lds[i].key;
+    uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
+    bool ok = true;
+    for (int probe = 0;; probe++) {
+      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
+      uint64_t cur = d.globalTable[slot].key;
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & (kGlobalGroups - 1);
+    }
+    if (!ok) continue;
+    for (int a = 0; a < d.nAggs; a++) {
+      Int128 v = {lds[i].accLo[a], lds[i].accHi[a]};
+      accumInto(&d.globalTable[slot], a, v, lds[i].cnt[a]);
+    }
+  }
+}
+
 __global__ void jaInitSlotsKernel(JoinAggSlot* slots, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
@@ -1202,7 +1539,16 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL(initGlobalTableKernel, dim3((kGlobalGroups + 255) / 256),
                      dim3(256), 0, s, desc.globalTable, kGlobalGroups);
-  if (desc.wide) {
+  if (desc.useGlds) {
+    size_t shmem = ((sizeof(GroupSlot) * kLdsGroups + 15) & ~15ULL) +
+                   (size_t)4 * 2 * desc.tileBytes;
+    if (desc.wide)
+      hipLaunchKernelGGL((fusedAggGldsKernel<true>), dim3(grid), dim3(256),
+                         shmem, s, devDesc);
+    else
+      hipLaunchKernelGGL((fusedAggGldsKernel<false>), dim3(grid), dim3(256),
+                         shmem, s, devDesc);
+  } else if (desc.wide) {
     if (desc.rbatch >= 2)
       hipLaunchKernelGGL((fusedAggKernel<true, 2>), dim3(grid), dim3(256), 0, s, devDesc);
     else

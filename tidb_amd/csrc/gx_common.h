@@ -95,7 +95,7 @@ struct FetchDesc {
   int32_t ldsOff;  // glds variant: byte offset of this stream in a tile buffer
 };
 
-constexpr int kMaxFetch = 6;
+constexpr int kMaxFetch = 8;
 
 // ---- aggregation ----
 // Per-group state layout (all aggs): int128 acc + int64 count per agg slot.

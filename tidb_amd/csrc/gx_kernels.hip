@@ -315,17 +315,18 @@ __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* 
 // raw per-row fetch buffer: setters use compile-time slot indices (phase A),
 // the getter is a wave-uniform switch (runtime-indexed arrays would spill)
 struct RawState {
-  ulonglong2 s0, s1, s2, s3, s4, s5;
+  ulonglong2 s0, s1, s2, s3, s4, s5, s6, s7;
   __device__ ulonglong2 get(int i) const {
     switch (i) {
       case 0: return s0; case 1: return s1; case 2: return s2; case 3: return s3;
-      case 4: return s4; default: return s5;
+      case 4: return s4; case 5: return s5; case 6: return s6; default: return s7;
     }
   }
   __device__ void set(int i, ulonglong2 v) {
     switch (i) {
       case 0: s0 = v; break; case 1: s1 = v; break; case 2: s2 = v; break;
-      case 3: s3 = v; break; case 4: s4 = v; break; default: s5 = v; break;
+      case 3: s3 = v; break; case 4: s4 = v; break; case 5: s5 = v; break;
+      case 6: s6 = v; break; default: s7 = v; break;
     }
   }
 };
@@ -338,6 +339,7 @@ __device__ __attribute__((always_inline)) inline void fetchRow(const DevTable& t
   for (int f = 0; f < kMaxFetch; f++) {
     if (f >= nFetch) break;
     const FetchDesc& fd = fetch[f];
+    if (fd.kind == FETCH_B1) continue;  // staged-variant-only stream
     const DevCol& c = tab.cols[fd.col];
     ulonglong2 v;
     if (fd.kind == FETCH_8B) {

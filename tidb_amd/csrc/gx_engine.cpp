@@ -248,7 +248,8 @@ static void setDevColMeta(gxp::DevCol* c, int type, int frac) {
 
 static void* devAlloc(gx_exec* ex, size_t n) {
   void* p = nullptr;
-  if (hipMalloc(&p, n) != hipSuccess) return nullptr;
+  // +16 B slack: glds tail tiles may read a few bytes past the last row
+  if (hipMalloc(&p, n + 16) != hipSuccess) return nullptr;
   ex->devBufs.push_back(p);
   return p;
 }

@@ -1441,6 +1441,7 @@ static int32_t runFused(gx_exec* ex) {
   HIP_OK(ex, hipEventCreate(&ev0));
   HIP_OK(ex, hipEventCreate(&ev1));
   HIP_OK(ex, hipEventRecord(ev0, ex->stream));
+  if (ex->desc.noLds) ex->desc.useGlds = 0;  // high-NDV retry uses the plain kernel
   int lrc = gxp::gxLaunchFusedAgg(ex->desc, ex->devDesc, ex->stream);
   if (lrc != 0) {
     ex->err = "fused kernel launch failed: " +

@@ -185,28 +185,36 @@ __device__ inline Int128 i128MulI64(Int128 a, int64_t b, bool* ovf) {
   return {(uint64_t)sr, (int64_t)(sr >> 64)};
 }
 
-__device__ __constant__ const int64_t kP10[19] = {
-    1, 10, 100, 1000, 10000, 100000, 1000000, 10000000, 100000000, 1000000000,
-    10000000000LL, 100000000000LL, 1000000000000LL, 10000000000000LL,
-    100000000000000LL, 1000000000000000LL, 10000000000000000LL,
-    100000000000000000LL, 1000000000000000000LL};
+// powers of ten / magic reciprocals as VALU select trees: a memory table
+// indexed per lane emits a global load per lookup, which both costs a
+// dependent memory round trip per row and breaks the glds kernel's counted
+// vmcnt bookkeeping.
+__device__ inline int64_t kP10(int k) {
+  int64_t lo = k < 2 ? (k < 1 ? 1 : 10)
+                     : (k < 3 ? 100 : (k < 4 ? 1000 : 10000));
+  int64_t mid = k < 7 ? (k < 6 ? 100000 : 1000000)
+                      : (k < 8 ? 10000000 : (k < 9 ? 100000000 : 1000000000));
+  int64_t hi = k < 12 ? (k < 11 ? 10000000000LL : 100000000000LL)
+                      : (k < 13 ? 1000000000000LL
+                                : (k < 14 ? 10000000000000LL : 100000000000000LL));
+  int64_t top = k < 17 ? (k < 16 ? 1000000000000000LL : 10000000000000000LL)
+                       : (k < 18 ? 100000000000000000LL : 1000000000000000000LL);
+  return k < 5 ? lo : (k < 10 ? mid : (k < 15 ? hi : top));
+}
 
 // magic reciprocals for n / 10^k, exact for 0 <= n < 2^31:
-// q = (n * kMagic[k]) >> 62  with kMagic[k] = ceil(2^62 / 10^k)
-__device__ __constant__ const uint64_t kDivMagic[10] = {
-    4611686018427387904ULL,  // 10^0
-    461168601842738791ULL,   // 10^1
-    46116860184273880ULL,    // 10^2
-    4611686018427388ULL,     // 10^3
-    461168601842739ULL,      // 10^4
-    46116860184274ULL,       // 10^5
-    4611686018428ULL,        // 10^6
-    461168601843ULL,         // 10^7
-    46116860185ULL,          // 10^8
-    4611686019ULL,           // 10^9
-};
+// q = (n * magic(k)) >> 62 with magic(k) = ceil(2^62 / 10^k)
+__device__ inline uint64_t kDivMagic(int k) {
+  uint64_t lo = k < 2 ? (k < 1 ? 4611686018427387904ULL : 461168601842738791ULL)
+                      : (k < 3 ? 46116860184273880ULL
+                               : (k < 4 ? 4611686018427388ULL : 461168601842739ULL));
+  uint64_t hi = k < 7 ? (k < 6 ? 46116860184274ULL : 4611686018428ULL)
+                      : (k < 8 ? 461168601843ULL
+                               : (k < 9 ? 46116860185ULL : 4611686019ULL));
+  return k < 5 ? lo : hi;
+}
 __device__ inline int64_t divP10(uint32_t n, int k) {
-  return (int64_t)(((unsigned __int128)n * kDivMagic[k]) >> 62);
+  return (int64_t)(((unsigned __int128)n * kDivMagic(k)) >> 62);
 }
 
 // ---- value type abstraction (NARROW = int64, WIDE = Int128) ----
@@ -232,7 +240,7 @@ struct VT<false> {
     *ovf |= __builtin_mul_overflow(a, b, &r);
     return r;
   }
-  static __device__ T scale10(T a, int k, bool* ovf) { return mul(a, kP10[k], ovf); }
+  static __device__ T scale10(T a, int k, bool* ovf) { return mul(a, kP10(k), ovf); }
   static __device__ T zero() { return 0; }
   static __device__ Int128 toAcc(T v) { return i128FromI64(v); }
   static __device__ int cmp(T a, T b) { return a < b ? -1 : (a > b ? 1 : 0); }
@@ -249,7 +257,7 @@ struct VT<true> {
     if (!fits) { *ovf = true; return {0, 0}; }
     return i128MulI64(a, (int64_t)b.lo, ovf);
   }
-  static __device__ T scale10(T a, int k, bool* ovf) { return i128MulI64(a, kP10[k], ovf); }
+  static __device__ T scale10(T a, int k, bool* ovf) { return i128MulI64(a, kP10(k), ovf); }
   static __device__ T zero() { return {0, 0}; }
   static __device__ Int128 toAcc(T v) { return v; }
   static __device__ int cmp(T a, T b) {
@@ -283,13 +291,13 @@ __device__ __attribute__((always_inline)) inline bool parseDecimalRaw(ulonglong2
   uint32_t fw = wordsInt == 0 ? lo2.y : (wordsInt == 1 ? hi2.x : hi2.y);
   int64_t fr = digitsFrac > 0 ? divP10(fw, 9 - digitsFrac) : 0;
   if (WIDE) {
-    __int128 units = (__int128)ip * kP10[digitsFrac] + fr;
+    __int128 units = (__int128)ip * kP10(digitsFrac) + fr;
     if (neg) units = -units;
     Int128 u = {(uint64_t)units, (int64_t)(units >> 64)};
     *out = *(typename VT<WIDE>::T*)&u;
   } else {
     int64_t units;
-    bool ovf = __builtin_mul_overflow(ip, kP10[digitsFrac], &units);
+    bool ovf = __builtin_mul_overflow(ip, kP10(digitsFrac), &units);
     ovf |= __builtin_add_overflow(units, fr, &units);
     if (ovf) {
       atomicOr(err, kErrRetryWide);
@@ -688,7 +696,6 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   __syncthreads();
 
   // ---- flush LDS table into the global table ----
-  if (d.noLds) return;
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     if (lds[i].key == kEmptyKey) continue;
     uint64_t key = lds[i].key;
@@ -1130,6 +1137,29 @@ __global__ void jaCompactKernel(const JoinAggDesc* __restrict__ dp, TopNOut* out
 // dynamic __shared__ carve (a second __shared__ object would make hipcc
 // drain vmcnt before every ds_read — guide §5 'Three .s-level traps').
 
+typedef __attribute__((address_space(3))) char LdsChar;
+typedef __attribute__((address_space(3))) GroupSlot LdsGroupSlot;
+
+// LDS-table accumulate via explicit AS3 atomics (generic-pointer HIP atomics
+// lower to flat ops, which count on vmcnt and would break the counted-wait
+// DMA pipeline)
+__device__ inline void accumIntoLds(LdsGroupSlot* slot, int a, Int128 v,
+                                    int64_t dc) {
+  if (v.lo != 0 || v.hi != 0) {
+    uint64_t old = __hip_atomic_fetch_add(&slot->accLo[a], (uint64_t)v.lo,
+                                          __ATOMIC_RELAXED,
+                                          __HIP_MEMORY_SCOPE_WORKGROUP);
+    uint64_t carry = (old + v.lo) < old ? 1 : 0;
+    int64_t hiAdd = v.hi + (int64_t)carry;
+    if (hiAdd != 0)
+      __hip_atomic_fetch_add((__attribute__((address_space(3)))uint64_t*)&slot->accHi[a], (uint64_t)hiAdd,
+                             __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+  }
+  if (dc != 0)
+    __hip_atomic_fetch_add((__attribute__((address_space(3)))uint64_t*)&slot->cnt[a], (uint64_t)dc,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+
 __device__ inline void gldsWaitVmcnt(int n) {
   switch (n) {
     case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
@@ -1146,11 +1176,12 @@ __device__ inline void gldsWaitVmcnt(int n) {
 
 // issue all stream DMAs for one 64-row tile into tileBuf (wave-uniform base)
 __device__ __attribute__((always_inline)) inline void gldsIssueTile(
-    const FusedQueryDesc& d, int64_t row0, int64_t clampMax, char* tileBuf,
+    const FusedQueryDesc& d, int64_t row0, int64_t clampMax, LdsChar* tileBuf,
     int lane) {
-  for (int f = 0; f < d.nFetch; f++) {
+  for (int f0 = 0; f0 < d.nFetch; f0++) {
+    int f = __builtin_amdgcn_readfirstlane(f0);  // uniform -> scalar desc loads
     const FetchDesc& fd = d.fetch[f];
-    const DevCol& c = d.table.cols[fd.col];
+    const DevCol& c = d.table.cols[__builtin_amdgcn_readfirstlane(fd.col)];
     auto lptr = (__attribute__((address_space(3))) void*)(tileBuf + fd.ldsOff);
     if (fd.kind == FETCH_DEC16) {
       int64_t r = row0 + lane;
@@ -1184,24 +1215,30 @@ __device__ __attribute__((always_inline)) inline void gldsIssueTile(
   }
 }
 
-// LDS-backed raw accessor for one row (lane) of the staged tile
+// LDS-backed raw accessor for one row (lane) of the staged tile. The buffer
+// pointer is LDS-qualified so reads lower to ds_read (a generic pointer
+// would emit flat loads, which count on vmcnt and break the pipeline).
 struct LdsRaw {
-  const char* buf;  // generic-address view of the tile buffer
-  const FusedQueryDesc* d;
+  const LdsChar* buf;
+  const FusedQueryDesc* d;  // LDS-resident copy (see kernel prologue)
   int lane;
   __device__ ulonglong2 get(int slot) const {
-    const FetchDesc& fd = d->fetch[slot];
+    const FetchDesc& fd = d->fetch[__builtin_amdgcn_readfirstlane(slot)];
     if (fd.kind == FETCH_DEC16) {
-      return *(const ulonglong2*)(buf + fd.ldsOff + lane * 16);
+      auto p = (const __attribute__((address_space(3))) uint64_t*)(buf + fd.ldsOff + lane * 16);
+      ulonglong2 v;
+      v.x = p[0];
+      v.y = p[1];
+      return v;
     }
     if (fd.kind == FETCH_8B) {
       ulonglong2 v;
-      v.x = *(const uint64_t*)(buf + fd.ldsOff + lane * 8);
+      v.x = *(const __attribute__((address_space(3))) uint64_t*)(buf + fd.ldsOff + lane * 8);
       v.y = 0;
       return v;
     }
     ulonglong2 v;
-    v.x = *(const uint8_t*)(buf + fd.ldsOff + lane);
+    v.x = *(const __attribute__((address_space(3))) uint8_t*)(buf + fd.ldsOff + lane);
     v.y = 0;
     return v;
   }
@@ -1213,7 +1250,8 @@ __device__ inline bool makeGroupKeyStaged(const FusedQueryDesc& d,
                                           const RAW& raw, uint64_t* keyOut,
                                           uint32_t* err) {
   uint64_t key = 0;
-  for (int k = 0; k < d.gkey.nCols; k++) {
+  for (int k0 = 0; k0 < d.gkey.nCols; k0++) {
+    int k = __builtin_amdgcn_readfirstlane(k0);
     uint32_t lane32;
     if (d.gkey.kind[k] == 2) {
       uint8_t b = (uint8_t)raw.get(d.gkey.slot[k]).x;
@@ -1234,11 +1272,11 @@ __device__ inline bool makeGroupKeyStaged(const FusedQueryDesc& d,
 // row pipeline over a staged tile row (mirrors processRow; RAW = LdsRaw)
 template <bool WIDE>
 __device__ __attribute__((always_inline)) inline bool processRowStaged(
-    const FusedQueryDesc& d, const LdsRaw& raw, GroupSlot* lds,
+    const FusedQueryDesc& d, const LdsRaw& raw, LdsGroupSlot* lds,
     uint64_t* mySel) {
   bool pass = true;
-  for (int p = 0; p < d.nPreds && pass; p++) {
-    const PredDesc& pd = d.preds[p];
+  for (int p0 = 0; p0 < d.nPreds && pass; p0++) {
+    const PredDesc& pd = d.preds[__builtin_amdgcn_readfirstlane(p0)];
     if (pd.kind == PRED_TIME_CMP_CONST) {
       uint64_t v = raw.get(pd.slot).x & ~0xFULL;
       uint64_t k = pd.constU64 & ~0xFULL;
@@ -1263,8 +1301,8 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
   vm.nullBits = 0;
   bool bad = false;
   bool ovf = false;
-  for (int i = 0; i < d.nIns && !bad; i++) {
-    const VmIns& ins = d.ins[i];
+  for (int i0 = 0; i0 < d.nIns && !bad; i0++) {
+    const VmIns& ins = d.ins[__builtin_amdgcn_readfirstlane(i0)];
     switch (ins.op) {
       case VM_LOAD_DEC: {
         typename VT<WIDE>::T v = VT<WIDE>::zero();
@@ -1326,44 +1364,30 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
 
   uint64_t key;
   if (!makeGroupKeyStaged(d, raw, &key, d.errorFlag)) return false;
-  GroupSlot* target;
-  if (!d.noLds) {
-    uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
-    for (int probe = 0;; probe++) {
-      if (probe >= kLdsGroups) { atomicOr(d.errorFlag, kErrLdsFull); return false; }
-      uint64_t cur = lds[slot].key;
-      if (cur == key) break;
-      if (cur == kEmptyKey) {
-        uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
-                                  (unsigned long long)kEmptyKey,
-                                  (unsigned long long)key);
-        if (prev == kEmptyKey || prev == key) break;
-      }
-      slot = (slot + 1) & (kLdsGroups - 1);
+  // (the glds variant always uses the LDS table; the engine runs the plain
+  // kernel for the high-NDV global-direct retry)
+  uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
+  for (int probe = 0;; probe++) {
+    if (probe >= kLdsGroups) { atomicOr(d.errorFlag, kErrLdsFull); return false; }
+    uint64_t cur = lds[slot].key;
+    if (cur == key) break;
+    if (cur == kEmptyKey) {
+      uint64_t expected = kEmptyKey;
+      bool won = __hip_atomic_compare_exchange_strong(
+          &lds[slot].key, &expected, key, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+          __HIP_MEMORY_SCOPE_WORKGROUP);
+      if (won || expected == key) break;
     }
-    target = &lds[slot];
-  } else {
-    uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
-    for (int probe = 0;; probe++) {
-      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, kErrGlobalFull); return false; }
-      uint64_t cur = d.globalTable[slot].key;
-      if (cur == key) break;
-      if (cur == kEmptyKey) {
-        uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
-                                  (unsigned long long)kEmptyKey,
-                                  (unsigned long long)key);
-        if (prev == kEmptyKey || prev == key) break;
-      }
-      slot = (slot + 1) & (kGlobalGroups - 1);
-    }
-    target = &d.globalTable[slot];
+    slot = (slot + 1) & (kLdsGroups - 1);
   }
-  for (int a = 0; a < d.nAggs; a++) {
+  LdsGroupSlot* target = &lds[slot];
+  for (int a0 = 0; a0 < d.nAggs; a0++) {
+    int a = __builtin_amdgcn_readfirstlane(a0);
     const AggDesc& ad = d.aggs[a];
     if (ad.func == 0) {
-      accumInto(target, a, Int128{0, 0}, 1);
+      accumIntoLds(target, a, Int128{0, 0}, 1);
     } else if (ad.srcReg >= 0) {
-      accumInto(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
+      accumIntoLds(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
     }
   }
   return true;
@@ -1372,10 +1396,10 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
 template <bool WIDE>
 __launch_bounds__(256)
 __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
-  const FusedQueryDesc& d = *dp;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  GroupSlot* lds = (GroupSlot*)smem;
-  char* tiles = smem + ((sizeof(GroupSlot) * kLdsGroups + 15) & ~15ULL);
+  constexpr size_t kTableBytes = (sizeof(GroupSlot) * kLdsGroups + 15) & ~15ULL;
+  auto lds = (LdsGroupSlot*)(LdsChar*)smem;
+  auto tilesBase = (LdsChar*)smem + kTableBytes;
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     lds[i].key = kEmptyKey;
     for (int a = 0; a < kMaxAggs; a++) {
@@ -1385,6 +1409,7 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
     }
   }
   __syncthreads();
+  const FusedQueryDesc& d = *dp;
 
   int64_t n = d.table.nRows;
   int64_t per = (n + gridDim.x - 1) / gridDim.x;
@@ -1393,7 +1418,7 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
   if (end > n) end = n;
   int wave = threadIdx.x >> 6;
   int lane = threadIdx.x & 63;
-  char* myTiles = tiles + (size_t)wave * 2 * d.tileBytes;
+  LdsChar* myTiles = tilesBase + (size_t)wave * 2 * d.tileBytes;
   uint64_t mySel = 0;
   bool failed = false;
 
@@ -1430,7 +1455,6 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
   }
   __syncthreads();
 
-  if (d.noLds) return;
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     if (lds[i].key == kEmptyKey) continue;
     uint64_t key = lds[i].key;

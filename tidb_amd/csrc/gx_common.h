@@ -142,6 +142,12 @@ struct FusedQueryDesc {
   VmIns ins[kMaxVmIns];
   int32_t nIns = 0;
   int32_t nLoadIns = 0;
+  // per-instruction precomputed constants (scalar-loaded; avoids per-row
+  // power-of-ten select trees / table loads):
+  //   LOAD_DEC i: insP10[i] = 10^expectedFrac, insMagic[i] = ceil(2^62/10^(9-f))
+  //   SCALE_UP i: insP10[i] = 10^shift
+  int64_t insP10[kMaxVmIns];
+  uint64_t insMagic[kMaxVmIns];
   // batched fetch plan
   FetchDesc fetch[kMaxFetch];
   int32_t nFetch = 0;
@@ -217,6 +223,8 @@ struct JoinAggDesc {
   int64_t constHi[kMaxVmConsts];
   int32_t nConsts = 0;
   int32_t valueReg = -1;  // register holding the summed value
+  int64_t insP10[kMaxVmIns];     // same meaning as FusedQueryDesc::insP10
+  uint64_t insMagic[kMaxVmIns];
   FetchDesc fetch[kMaxFetch];
   int32_t nFetch = 0;
   // device state

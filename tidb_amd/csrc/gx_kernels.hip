@@ -185,34 +185,27 @@ __device__ inline Int128 i128MulI64(Int128 a, int64_t b, bool* ovf) {
   return {(uint64_t)sr, (int64_t)(sr >> 64)};
 }
 
-// powers of ten / magic reciprocals as VALU select trees: a memory table
-// indexed per lane emits a global load per lookup, which both costs a
-// dependent memory round trip per row and breaks the glds kernel's counted
-// vmcnt bookkeeping.
-__device__ inline int64_t kP10(int k) {
-  int64_t lo = k < 2 ? (k < 1 ? 1 : 10)
-                     : (k < 3 ? 100 : (k < 4 ? 1000 : 10000));
-  int64_t mid = k < 7 ? (k < 6 ? 100000 : 1000000)
-                      : (k < 8 ? 10000000 : (k < 9 ? 100000000 : 1000000000));
-  int64_t hi = k < 12 ? (k < 11 ? 10000000000LL : 100000000000LL)
-                      : (k < 13 ? 1000000000000LL
-                                : (k < 14 ? 10000000000000LL : 100000000000000LL));
-  int64_t top = k < 17 ? (k < 16 ? 1000000000000000LL : 10000000000000000LL)
-                       : (k < 18 ? 100000000000000000LL : 1000000000000000000LL);
-  return k < 5 ? lo : (k < 10 ? mid : (k < 15 ? hi : top));
-}
-
-// magic reciprocals for n / 10^k, exact for 0 <= n < 2^31:
-// q = (n * magic(k)) >> 62 with magic(k) = ceil(2^62 / 10^k)
-__device__ inline uint64_t kDivMagic(int k) {
-  uint64_t lo = k < 2 ? (k < 1 ? 4611686018427387904ULL : 461168601842738791ULL)
-                      : (k < 3 ? 46116860184273880ULL
-                               : (k < 4 ? 4611686018427388ULL : 461168601842739ULL));
-  uint64_t hi = k < 7 ? (k < 6 ? 46116860184274ULL : 4611686018428ULL)
-                      : (k < 8 ? 461168601843ULL
-                               : (k < 9 ? 46116860185ULL : 4611686019ULL));
-  return k < 5 ? lo : hi;
-}
+// powers of ten / magic reciprocals. These live in __constant__ memory: a
+// per-lane indexed lookup is one (cached) global load. The HOT paths never
+// reach them -- engine-precomputed per-instruction constants (insP10/insMagic)
+// cover every row whose stored frac matches the declared column frac -- so
+// these tables serve only mismatched-frac rows and rescale fallbacks. The
+// glds kernel's counted-vmcnt bookkeeping is safe because the staged loop's
+// executed path never issues these loads on conforming data.
+__constant__ int64_t kP10Tab[19] = {1, 10, 100, 1000, 10000, 100000, 1000000,
+                                    10000000, 100000000, 1000000000,
+                                    10000000000LL, 100000000000LL,
+                                    1000000000000LL, 10000000000000LL,
+                                    100000000000000LL, 1000000000000000LL,
+                                    10000000000000000LL, 100000000000000000LL,
+                                    1000000000000000000LL};
+__constant__ uint64_t kDivMagicTab[10] = {
+    4611686018427387904ULL, 461168601842738791ULL, 46116860184273880ULL,
+    4611686018427388ULL,    461168601842739ULL,    46116860184274ULL,
+    4611686018428ULL,       461168601843ULL,       46116860185ULL,
+    4611686019ULL};
+__device__ inline int64_t kP10(int k) { return kP10Tab[k]; }
+__device__ inline uint64_t kDivMagic(int k) { return kDivMagicTab[k]; }
 __device__ inline int64_t divP10(uint32_t n, int k) {
   return (int64_t)(((unsigned __int128)n * kDivMagic(k)) >> 62);
 }
@@ -270,9 +263,14 @@ struct VT<true> {
 // parse 16 raw bytes of a 40-byte MyDecimal (digitsInt <= 18, digitsFrac <= 9)
 // into units at scale = digitsFrac. Returns false on malformed input
 // (kErrBadDecimal) or narrow overflow (kErrRetryWide).
+// expFrac/p10exp/magicExp: engine-precomputed for the column's declared
+// frac — the common case takes no power-of-ten select tree. expFrac < 0
+// forces the generic path.
 template <bool WIDE>
 __device__ __attribute__((always_inline)) inline bool parseDecimalRaw(ulonglong2 raw, typename VT<WIDE>::T* out,
-                                       int* scale, uint32_t* err) {
+                                       int* scale, uint32_t* err,
+                                       int expFrac = -1, int64_t p10exp = 0,
+                                       uint64_t magicExp = 0) {
   uint2 lo2 = {(uint32_t)raw.x, (uint32_t)(raw.x >> 32)};
   uint2 hi2 = {(uint32_t)raw.y, (uint32_t)(raw.y >> 32)};
   uint32_t hdr = lo2.x;
@@ -289,15 +287,25 @@ __device__ __attribute__((always_inline)) inline bool parseDecimalRaw(ulonglong2
   if (wordsInt == 1) ip = (int32_t)lo2.y;
   else if (wordsInt == 2) ip = (int64_t)(int32_t)lo2.y * 1000000000 + (int32_t)hi2.x;
   uint32_t fw = wordsInt == 0 ? lo2.y : (wordsInt == 1 ? hi2.x : hi2.y);
-  int64_t fr = digitsFrac > 0 ? divP10(fw, 9 - digitsFrac) : 0;
+  int64_t p10v;
+  int64_t fr;
+  if (digitsFrac == expFrac) {  // fast path: constants provided
+    p10v = p10exp;
+    fr = digitsFrac > 0
+             ? (int64_t)(((unsigned __int128)fw * magicExp) >> 62)
+             : 0;
+  } else {
+    p10v = kP10(digitsFrac);
+    fr = digitsFrac > 0 ? divP10(fw, 9 - digitsFrac) : 0;
+  }
   if (WIDE) {
-    __int128 units = (__int128)ip * kP10(digitsFrac) + fr;
+    __int128 units = (__int128)ip * p10v + fr;
     if (neg) units = -units;
     Int128 u = {(uint64_t)units, (int64_t)(units >> 64)};
     *out = *(typename VT<WIDE>::T*)&u;
   } else {
     int64_t units;
-    bool ovf = __builtin_mul_overflow(ip, kP10(digitsFrac), &units);
+    bool ovf = __builtin_mul_overflow(ip, p10v, &units);
     ovf |= __builtin_add_overflow(units, fr, &units);
     if (ovf) {
       atomicOr(err, kErrRetryWide);
@@ -312,12 +320,14 @@ __device__ __attribute__((always_inline)) inline bool parseDecimalRaw(ulonglong2
 
 template <bool WIDE>
 __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* out,
-                                        int* scale, uint32_t* err) {
+                                        int* scale, uint32_t* err,
+                                        int expFrac = -1, int64_t p10exp = 0,
+                                        uint64_t magicExp = 0) {
   // 40-byte stride keeps rows 8-byte aligned: two dwordx2 loads
   ulonglong2 raw;
   raw.x = *(const uint64_t*)p;
   raw.y = *(const uint64_t*)(p + 8);
-  return parseDecimalRaw<WIDE>(raw, out, scale, err);
+  return parseDecimalRaw<WIDE>(raw, out, scale, err, expFrac, p10exp, magicExp);
 }
 
 // raw per-row fetch buffer: setters use compile-time slot indices (phase A),
@@ -513,7 +523,8 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         typename VT<WIDE>::T v = VT<WIDE>::zero();
         if (!nul) {
           int sc;
-          if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)) {
+          if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag,
+                                     ins.b, d.insP10[i], d.insMagic[i])) {
             bad = true;
             break;
           }
@@ -563,7 +574,9 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         break;
       }
       case VM_SCALE_UP:
-        vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
+        vm.set(ins.dst,
+               VT<WIDE>::mul(vm.get(ins.a),
+                             VT<WIDE>::fromI64(d.insP10[i], nullptr), &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a));
         break;
     }
@@ -1042,7 +1055,9 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
           break;
         }
         case VM_SCALE_UP:
-          vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
+          vm.set(ins.dst,
+                 VT<WIDE>::mul(vm.get(ins.a),
+                               VT<WIDE>::fromI64(d.insP10[i], nullptr), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a));
           break;
       }
@@ -1307,7 +1322,8 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
       case VM_LOAD_DEC: {
         typename VT<WIDE>::T v = VT<WIDE>::zero();
         int sc;
-        if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag)) {
+        if (!parseDecimalRaw<WIDE>(raw.get(ins.c), &v, &sc, d.errorFlag,
+                                   ins.b, d.insP10[i0], d.insMagic[i0])) {
           bad = true;
           break;
         }
@@ -1344,7 +1360,9 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
         vm.set(ins.dst, VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf));
         break;
       case VM_SCALE_UP:
-        vm.set(ins.dst, VT<WIDE>::scale10(vm.get(ins.a), ins.b, &ovf));
+        vm.set(ins.dst,
+               VT<WIDE>::mul(vm.get(ins.a),
+                             VT<WIDE>::fromI64(d.insP10[i0], nullptr), &ovf));
         break;
     }
   }
@@ -1412,7 +1430,7 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
 
   int64_t n = d.table.nRows;
-  int64_t per = (n + gridDim.x - 1) / gridDim.x;
+  int64_t per = ((n + gridDim.x - 1) / gridDim.x + 63) & ~63LL;
   int64_t begin = (int64_t)blockIdx.x * per;
   int64_t end = begin + per;
   if (end > n) end = n;

@@ -87,12 +87,16 @@ enum FetchKind : int32_t {
   FETCH_DEC16 = 1,    // 16 bytes at data + row*40 (decimal header+3 words)
   FETCH_OFFSETS = 2,  // offsets[row], offsets[row+1]
   FETCH_B1 = 3,       // 1 byte per row (dense char(1); glds variant only)
+  FETCH_8B_CHAR2 = 4, // x = 8-byte element; y = 1-2 dense char(1) cols packed
+                      // (plain kernel: prefetches group chars with the row)
+  FETCH_CHAR2 = 5,    // y = 1-2 dense char(1) cols packed; x unused
 };
 
 struct FetchDesc {
   int32_t kind;
   int32_t col;
   int32_t ldsOff;  // glds variant: byte offset of this stream in a tile buffer
+                   // FETCH_*CHAR2: packed char col ids (c0 | c1<<8 | n<<16)
 };
 
 constexpr int kMaxFetch = 8;
@@ -109,6 +113,14 @@ struct AggDesc {
 
 constexpr int kMaxAggs = 12;
 
+// Physical accumulator plan: aggregates sharing a source value (e.g. Q1's
+// sum(qty) and avg(qty)) share ONE acc slot -- every supported device agg
+// (count/sum/avg) is additive, so the per-row atomic runs once per unique
+// srcReg instead of once per aggregate. accMap is host-decode-side: agg a
+// reads acc slot accMap[a] (-1 for COUNT). When sharedCnt is set (no
+// consumed column has NULLs) every agg's count equals the group row count
+// and the kernel bumps only cnt[0].
+
 // ---- group keys ----
 // Round-1 device grouping: group-by columns pack into ONE u64 key
 // (string cols <= 3 bytes each as len<<24|bytes in a 32-bit lane; Q1 uses two
@@ -119,6 +131,8 @@ struct GroupKeyDesc {
   int32_t col[2];
   int32_t kind[2];  // 0 = short string, 1 = small i64 (<2^31), 2 = dense char(1)
   int32_t slot[2];  // raw fetch slot (string: the offsets pair; i64: value; dense: -1)
+  int32_t rawSlot[2] = {-1, -1};  // plain-kernel prefetch slot for dense chars:
+                                  // char k lives at byte k of that slot's v.y
 };
 
 constexpr uint64_t kEmptyKey = ~0ULL;
@@ -157,6 +171,10 @@ struct FusedQueryDesc {
   // aggs
   AggDesc aggs[kMaxAggs];
   int32_t nAggs = 0;
+  int32_t nAccSlots = 0;
+  int32_t accReg[kMaxAggs];   // phys acc slot -> VM register
+  int32_t accMap[kMaxAggs];   // agg index -> phys acc slot (-1 for COUNT)
+  int32_t sharedCnt = 0;
   GroupKeyDesc gkey;
   // outputs
   GroupSlot* globalTable = nullptr;  // kGlobalGroups slots

@@ -1015,6 +1015,24 @@ static int32_t compileFused(gx_exec* ex) {
     }
     ex->desc.aggs[ex->desc.nAggs++] = ad;
   }
+  // physical accumulator plan: one acc slot per unique source register
+  // (count/sum/avg are all additive, so aggs over the same value share)
+  ex->desc.nAccSlots = 0;
+  for (int a = 0; a < ex->desc.nAggs; a++) {
+    const gxp::AggDesc& ad = ex->desc.aggs[a];
+    if (ad.func == GX_AGG_COUNT || ad.srcReg < 0) {
+      ex->desc.accMap[a] = -1;
+      continue;
+    }
+    int found = -1;
+    for (int s = 0; s < ex->desc.nAccSlots; s++)
+      if (ex->desc.accReg[s] == ad.srcReg) { found = s; break; }
+    if (found < 0) {
+      found = ex->desc.nAccSlots++;
+      ex->desc.accReg[found] = ad.srcReg;
+    }
+    ex->desc.accMap[a] = found;
+  }
 
   // assign raw-fetch slots to VM load ops, then move loads to the front of
   // the instruction stream (each dst is written once, loads have no deps —
@@ -1395,6 +1413,39 @@ static int32_t materializeDevice(gx_exec* ex) {
     } else {
       d.useGlds = 0;
     }
+    // one shared row count per group when no consumed column can be NULL
+    d.sharedCnt = 1;
+    for (int c = 0; c < tab.nCols; c++)
+      if (tab.cols[c].hasNulls) d.sharedCnt = 0;
+    // plain-kernel char prefetch: when every group col is a dense char(1),
+    // fold the bytes into an 8B fetch slot's unused v.y (or a standalone
+    // slot) so the pipelined phase-A fetch covers the group key too -- the
+    // direct per-row char load was the serial latency left in the row chain.
+    for (int k = 0; k < 2; k++) d.gkey.rawSlot[k] = -1;
+    if (!d.useGlds && d.gkey.nCols > 0) {
+      bool allChar = true;
+      for (int k = 0; k < d.gkey.nCols; k++) allChar &= d.gkey.kind[k] == 2;
+      if (allChar) {
+        int c0 = d.gkey.col[0];
+        int c1 = d.gkey.nCols > 1 ? d.gkey.col[1] : 0;
+        int packed = (c0 & 0xFF) | ((c1 & 0xFF) << 8) |
+                     (d.gkey.nCols << 16);
+        int slot = -1;
+        for (int f = 0; f < d.nFetch; f++)
+          if (d.fetch[f].kind == gxp::FETCH_8B) { slot = f; break; }
+        if (slot >= 0) {
+          d.fetch[slot].kind = gxp::FETCH_8B_CHAR2;
+          d.fetch[slot].ldsOff = packed;
+        } else if (d.nFetch < gxp::kMaxFetch) {
+          slot = d.nFetch++;
+          d.fetch[slot].kind = gxp::FETCH_CHAR2;
+          d.fetch[slot].col = c0;
+          d.fetch[slot].ldsOff = packed;
+        }
+        if (slot >= 0)
+          for (int k = 0; k < d.gkey.nCols; k++) d.gkey.rawSlot[k] = slot;
+      }
+    }
     if (getenv("GX_DEBUG"))
       fprintf(stderr, "[gx] useGlds=%d tileBytes=%d nFetch=%d\n", d.useGlds,
               d.tileBytes, d.nFetch);
@@ -1574,8 +1625,10 @@ static int32_t runFused(gx_exec* ex) {
     }
     for (int a = 0; a < ex->desc.nAggs; a++) {
       const gxp::AggDesc& ad = ex->desc.aggs[a];
-      __int128 acc = ((__int128)s->accHi[a] << 64) | s->accLo[a];
-      int64_t cnt = s->cnt[a];
+      int phys = ex->desc.accMap[a];
+      __int128 acc = phys >= 0
+          ? (((__int128)s->accHi[phys] << 64) | s->accLo[phys]) : 0;
+      int64_t cnt = s->cnt[ex->desc.sharedCnt ? 0 : a];
       if (ad.func == GX_AGG_COUNT) {
         OutRowVal v;
         v.type = GX_TYPE_I64;

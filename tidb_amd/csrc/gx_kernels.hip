@@ -16,6 +16,7 @@
 // with overflow detection -> error flag (never silent). Equivalence to the
 // word-based MyDecimal arithmetic is covered by tests/golden + parity suites.
 #include <hip/hip_runtime.h>
+#include <type_traits>
 #include <hipcub/hipcub.hpp>
 
 #include "gx_common.h"
@@ -318,6 +319,15 @@ __device__ __attribute__((always_inline)) inline bool parseDecimalRaw(ulonglong2
   return true;
 }
 
+// explicit global (AS1) pointer: loads through the generic desc pointers
+// otherwise compile to FLAT loads, which decrement BOTH vmcnt and lgkmcnt --
+// every LDS hash-table wait then drains the prefetched row loads and the
+// software pipeline serializes.
+template <typename T>
+__device__ inline const __attribute__((address_space(1))) T* gptr(const void* p) {
+  return (const __attribute__((address_space(1))) T*)(uintptr_t)p;
+}
+
 template <bool WIDE>
 __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* out,
                                         int* scale, uint32_t* err,
@@ -325,8 +335,8 @@ __device__ inline bool loadDecimalUnits(const uint8_t* p, typename VT<WIDE>::T* 
                                         uint64_t magicExp = 0) {
   // 40-byte stride keeps rows 8-byte aligned: two dwordx2 loads
   ulonglong2 raw;
-  raw.x = *(const uint64_t*)p;
-  raw.y = *(const uint64_t*)(p + 8);
+  raw.x = *gptr<uint64_t>(p);
+  raw.y = *gptr<uint64_t>(p + 8);
   return parseDecimalRaw<WIDE>(raw, out, scale, err, expFrac, p10exp, magicExp);
 }
 
@@ -349,10 +359,31 @@ struct RawState {
   }
 };
 
+// 5-slot variant: the register allocator keeps every switch member live, so
+// a query using <= 5 fetch slots (TPC-H Q1: 4 decimals + 1 date) runs with a
+// 2x5x16B pipelined raw footprint instead of 2x8x16B -- the difference
+// between occupancy 3 and 4 waves/SIMD.
+struct RawState5 {
+  ulonglong2 s0, s1, s2, s3, s4;
+  __device__ ulonglong2 get(int i) const {
+    switch (i) {
+      case 0: return s0; case 1: return s1; case 2: return s2;
+      case 3: return s3; default: return s4;
+    }
+  }
+  __device__ void set(int i, ulonglong2 v) {
+    switch (i) {
+      case 0: s0 = v; break; case 1: s1 = v; break; case 2: s2 = v; break;
+      case 3: s3 = v; break; default: s4 = v; break;
+    }
+  }
+};
+
 // phase A: issue every fetch for one row, no consumption (loads overlap).
 // The loop is unrolled over the compile-time slot bound so every raw.set has
 // a literal index — a runtime-indexed store would be re-rolled into scratch.
-__device__ __attribute__((always_inline)) inline void fetchRow(const DevTable& tab, const FetchDesc* fetch, int nFetch, int64_t row, RawState& raw) {
+template <typename RAWT>
+__device__ __attribute__((always_inline)) inline void fetchRow(const DevTable& tab, const FetchDesc* fetch, int nFetch, int64_t row, RAWT& raw) {
 #pragma unroll
   for (int f = 0; f < kMaxFetch; f++) {
     if (f >= nFetch) break;
@@ -361,15 +392,22 @@ __device__ __attribute__((always_inline)) inline void fetchRow(const DevTable& t
     const DevCol& c = tab.cols[fd.col];
     ulonglong2 v;
     if (fd.kind == FETCH_8B) {
-      v.x = ((const uint64_t*)c.data)[row];
+      v.x = gptr<uint64_t>(c.data)[row];
       v.y = 0;
+    } else if (fd.kind == FETCH_8B_CHAR2 || fd.kind == FETCH_CHAR2) {
+      v.x = fd.kind == FETCH_8B_CHAR2 ? gptr<uint64_t>(c.data)[row] : 0;
+      const DevTable& t2 = tab;
+      uint64_t chars = (uint64_t)gptr<uint8_t>(t2.cols[fd.ldsOff & 0xFF].data)[row];
+      if (((fd.ldsOff >> 16) & 0xFF) > 1)
+        chars |= (uint64_t)gptr<uint8_t>(t2.cols[(fd.ldsOff >> 8) & 0xFF].data)[row] << 8;
+      v.y = chars;
     } else if (fd.kind == FETCH_DEC16) {
       const uint8_t* p = (const uint8_t*)c.data + row * 40;
-      v.x = *(const uint64_t*)p;
-      v.y = *(const uint64_t*)(p + 8);
+      v.x = *gptr<uint64_t>(p);
+      v.y = *gptr<uint64_t>(p + 8);
     } else {  // FETCH_OFFSETS
-      v.x = (uint64_t)c.offsets[row];
-      v.y = (uint64_t)c.offsets[row + 1];
+      v.x = (uint64_t)gptr<int64_t>(c.offsets)[row];
+      v.y = (uint64_t)gptr<int64_t>(c.offsets)[row + 1];
     }
     raw.set(f, v);
   }
@@ -377,7 +415,7 @@ __device__ __attribute__((always_inline)) inline void fetchRow(const DevTable& t
 
 __device__ inline bool colIsNull(const DevCol& c, int64_t row) {
   if (!c.hasNulls || c.nullBitmap == nullptr) return false;
-  return ((c.nullBitmap[row >> 3] >> (row & 7)) & 1) == 0;
+  return ((gptr<uint8_t>(c.nullBitmap)[row >> 3] >> (row & 7)) & 1) == 0;
 }
 
 __device__ inline int cmpResult(int c, int op) {
@@ -422,8 +460,9 @@ struct VmState {
 
 // pack the group key (see GroupKeyDesc comment); offsets/values come from the
 // batched raw fetch
+template <typename RAWT>
 __device__ __attribute__((always_inline)) inline bool makeGroupKey(const FusedQueryDesc& d, int64_t row,
-                                    const RawState& raw, uint64_t* keyOut,
+                                    const RAWT& raw, uint64_t* keyOut,
                                     uint32_t* err) {
   uint64_t key = 0;
   for (int k = 0; k < d.gkey.nCols; k++) {
@@ -432,14 +471,18 @@ __device__ __attribute__((always_inline)) inline bool makeGroupKey(const FusedQu
     if (colIsNull(c, row)) {
       lane = 0xFF000000u;
     } else if (d.gkey.kind[k] == 2) {
-      // dense char(1): data[row] is the value; PAD SPACE trims a lone space
-      uint8_t b = ((const uint8_t*)c.data)[row];
+      // dense char(1): data[row] is the value; PAD SPACE trims a lone space.
+      // Prefetched with the row's grouped fetch when a rawSlot is assigned
+      // (the direct load here is a serial dependency the pipeline can't hide).
+      uint8_t b = d.gkey.rawSlot[k] >= 0
+                      ? (uint8_t)(raw.get(d.gkey.rawSlot[k]).y >> (8 * k))
+                      : gptr<uint8_t>(c.data)[row];
       lane = b == ' ' ? 0u : ((1u << 24) | b);
     } else if (d.gkey.kind[k] == 0) {
       ulonglong2 off = raw.get(d.gkey.slot[k]);
       int64_t s = (int64_t)off.x, e = (int64_t)off.y;
       // utf8mb4_bin PAD SPACE: trim trailing spaces (collate.go:272)
-      const uint8_t* p = (const uint8_t*)c.data;
+      auto p = gptr<uint8_t>(c.data);
       while (e > s && p[e - 1] == ' ') e--;
       int64_t len = e - s;
       if (len > 3) { atomicOr(err, kErrBadKey); return false; }
@@ -461,6 +504,9 @@ __device__ __attribute__((always_inline)) inline bool makeGroupKey(const FusedQu
 // atomic int128 + count accumulation into a slot (LDS or global)
 template <typename SlotT>
 __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
+  // generic-pointer variant (global table); LDS paths use the AS3 helpers
+  // below so the atomics compile to ds_* ops (a flat atomic counts against
+  // vmcnt AND lgkmcnt and serializes the prefetch pipeline).
   if (v.lo != 0 || v.hi != 0) {
     uint64_t old = atomicAdd((unsigned long long*)&slot->accLo[a],
                              (unsigned long long)v.lo);
@@ -473,11 +519,40 @@ __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
     atomicAdd((unsigned long long*)&slot->cnt[a], (unsigned long long)dc);
 }
 
+typedef __attribute__((address_space(3))) GroupSlot Lds3GroupSlot;
+typedef __attribute__((address_space(3))) uint64_t Lds3U64;
+
+__device__ inline void lds3AccumAcc(Lds3GroupSlot* slot, int s, Int128 v) {
+  if (v.lo != 0 || v.hi != 0) {
+    uint64_t old = __hip_atomic_fetch_add((Lds3U64*)&slot->accLo[s], (uint64_t)v.lo,
+                                          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+    uint64_t carry = (old + v.lo) < old ? 1 : 0;
+    int64_t hiAdd = v.hi + (int64_t)carry;
+    if (hiAdd != 0)
+      __hip_atomic_fetch_add((Lds3U64*)&slot->accHi[s], (uint64_t)hiAdd,
+                             __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+  }
+}
+
+__device__ inline void lds3AccumCnt(Lds3GroupSlot* slot, int a, int64_t dc) {
+  if (dc != 0)
+    __hip_atomic_fetch_add((Lds3U64*)&slot->cnt[a], (uint64_t)dc,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+
+__device__ inline uint64_t lds3CasKey(Lds3GroupSlot* slot, uint64_t expect,
+                                      uint64_t val) {
+  __hip_atomic_compare_exchange_strong((Lds3U64*)&slot->key, &expect, val,
+                                       __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_WORKGROUP);
+  return expect;  // holds the previous value on failure, `expect` on success
+}
+
 // per-row pipeline after the raw fetch: filter -> VM -> LDS aggregate.
 // Returns false on a hard failure (error flag already set).
-template <bool WIDE>
+template <bool WIDE, typename RAWT>
 __device__ __attribute__((always_inline)) inline bool processRow(const FusedQueryDesc& d, int64_t row,
-                                  const RawState& raw, GroupSlot* lds,
+                                  const RAWT& raw, Lds3GroupSlot* lds,
                                   uint64_t* mySel) {
   // ---- filter (CNF; NULL rejects — expression.go:507 toBool) ----
   bool pass = true;
@@ -599,7 +674,6 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   // ---- group lookup / insert (LDS table, or global when NDV > kLdsGroups) ----
   uint64_t key;
   if (!makeGroupKey(d, row, raw, &key, d.errorFlag)) return false;
-  GroupSlot* target;
   if (!d.noLds) {
     uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
     for (int probe = 0;; probe++) {
@@ -610,23 +684,23 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
       uint64_t cur = lds[slot].key;
       if (cur == key) break;
       if (cur == kEmptyKey) {
-        uint64_t prev = atomicCAS((unsigned long long*)&lds[slot].key,
-                                  (unsigned long long)kEmptyKey,
-                                  (unsigned long long)key);
+        uint64_t prev = lds3CasKey(&lds[slot], kEmptyKey, key);
         if (prev == kEmptyKey || prev == key) break;
       }
       slot = (slot + 1) & (kLdsGroups - 1);
     }
-    target = &lds[slot];
-    // ---- update states ----
-    for (int a = 0; a < d.nAggs; a++) {
-      const AggDesc& ad = d.aggs[a];
-      if (ad.func == 0 /*COUNT*/) {
+    Lds3GroupSlot* target = &lds[slot];
+    // ---- update states: one atomic per unique acc, one or per-agg counts ----
+    if (d.sharedCnt) lds3AccumCnt(target, 0, 1);
+    for (int s = 0; s < d.nAccSlots; s++) {
+      int reg = d.accReg[s];
+      if (!vm.isNull(reg)) lds3AccumAcc(target, s, VT<WIDE>::toAcc(vm.get(reg)));
+    }
+    if (!d.sharedCnt) {
+      for (int a = 0; a < d.nAggs; a++) {
+        const AggDesc& ad = d.aggs[a];
         bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
-        if (!isNull) accumInto(target, a, Int128{0, 0}, 1);
-      } else {  // SUM / AVG
-        if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
-          accumInto(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
+        if (!isNull) lds3AccumCnt(target, a, 1);
       }
     }
     return true;
@@ -648,15 +722,17 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     }
     slot = (slot + 1) & (kGlobalGroups - 1);
   }
-  target = &d.globalTable[slot];
-  for (int a = 0; a < d.nAggs; a++) {
-    const AggDesc& ad = d.aggs[a];
-    if (ad.func == 0 /*COUNT*/) {
+  GroupSlot* target = &d.globalTable[slot];
+  if (d.sharedCnt) accumInto(target, 0, Int128{0, 0}, 1);  // bumps cnt[0] only
+  for (int s = 0; s < d.nAccSlots; s++) {
+    int reg = d.accReg[s];
+    if (!vm.isNull(reg)) accumInto(target, s, VT<WIDE>::toAcc(vm.get(reg)), 0);
+  }
+  if (!d.sharedCnt) {
+    for (int a = 0; a < d.nAggs; a++) {
+      const AggDesc& ad = d.aggs[a];
       bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
       if (!isNull) accumInto(target, a, Int128{0, 0}, 1);
-    } else {
-      if (ad.srcReg >= 0 && !vm.isNull(ad.srcReg))
-        accumInto(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
     }
   }
   return true;
@@ -668,6 +744,7 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
   bool failed = false;
   __shared__ GroupSlot lds[kLdsGroups];
+  Lds3GroupSlot* lds3 = (Lds3GroupSlot*)lds;
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     lds[i].key = kEmptyKey;
     for (int a = 0; a < kMaxAggs; a++) {
@@ -685,17 +762,28 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   if (end > n) end = n;
   uint64_t mySel = 0;
 
-  // grouped-fetch row loop: all of a row's loads issue back-to-back
-  // (fetchRow), one wait, then the row pipeline. A deeper software pipeline
-  // (ping-pong raw buffers / R-row batches) loses here: the extra live state
-  // pushes the register allocator into scratch spills that cost more than
-  // the overlap buys (measured 6.1 ms vs 4.9 ms at SF10).
-  (void)sizeof(char[R]);  // R kept for the launch-variant signature
-  for (int64_t row = begin + threadIdx.x; row < end && !failed;
-       row += blockDim.x) {
-    RawState raw;
-    fetchRow(d.table, d.fetch, d.nFetch, row, raw);
-    if (!processRow<WIDE>(d, row, raw, lds, &mySel)) failed = true;
+  // 2-deep software-pipelined grouped-fetch loop: row k+1's loads issue
+  // before row k's pipeline runs, so every wave keeps ~2 rows of bytes in
+  // flight (the kernel is HBM-latency bound: VALU util ~7%, concurrency =
+  // waves x rows-in-flight). Two NAMED RawState structs, never indexed or
+  // swapped through a pointer -- indexed access re-rolls to scratch spills
+  // (measured 6.1 ms), this form stays in registers.
+  {
+    // R is the compile-time fetch-slot bound: 5 covers Q1-shaped queries at
+    // one extra wave/SIMD of occupancy; 8 is the general variant.
+    using RAWT = typename std::conditional<R <= 5, RawState5, RawState>::type;
+    const int64_t stride = blockDim.x;
+    int64_t row = begin + threadIdx.x;
+    RAWT rawA, rawB;
+    if (row < end) fetchRow(d.table, d.fetch, d.nFetch, row, rawA);
+    for (; row < end && !failed; row += 2 * stride) {
+      const int64_t rB = row + stride;
+      if (rB < end) fetchRow(d.table, d.fetch, d.nFetch, rB, rawB);
+      if (!processRow<WIDE>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
+      const int64_t rA2 = row + 2 * stride;
+      if (rA2 < end) fetchRow(d.table, d.fetch, d.nFetch, rA2, rawA);
+      if (rB < end && !processRow<WIDE>(d, rB, rawB, lds3, &mySel)) failed = true;
+    }
   }
 
   if (d.selCount) {
@@ -727,10 +815,13 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
       slot = (slot + 1) & (kGlobalGroups - 1);
     }
     if (!ok) continue;
-    for (int a = 0; a < d.nAggs; a++) {
-      Int128 v = {lds[i].accLo[a], lds[i].accHi[a]};
-      accumInto(&d.globalTable[slot], a, v, lds[i].cnt[a]);
+    for (int s = 0; s < d.nAccSlots; s++) {
+      Int128 v = {lds[i].accLo[s], lds[i].accHi[s]};
+      accumInto(&d.globalTable[slot], s, v, 0);
     }
+    int nCnt = d.sharedCnt ? 1 : d.nAggs;
+    for (int a = 0; a < nCnt; a++)
+      accumInto(&d.globalTable[slot], a, Int128{0, 0}, lds[i].cnt[a]);
   }
 }
 
@@ -1399,13 +1490,16 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
     slot = (slot + 1) & (kLdsGroups - 1);
   }
   LdsGroupSlot* target = &lds[slot];
-  for (int a0 = 0; a0 < d.nAggs; a0++) {
-    int a = __builtin_amdgcn_readfirstlane(a0);
-    const AggDesc& ad = d.aggs[a];
-    if (ad.func == 0) {
+  if (d.sharedCnt) accumIntoLds(target, 0, Int128{0, 0}, 1);
+  for (int s0 = 0; s0 < d.nAccSlots; s0++) {
+    int s = __builtin_amdgcn_readfirstlane(s0);
+    int reg = d.accReg[s];
+    accumIntoLds(target, s, VT<WIDE>::toAcc(vm.get(reg)), 0);
+  }
+  if (!d.sharedCnt) {
+    for (int a0 = 0; a0 < d.nAggs; a0++) {
+      int a = __builtin_amdgcn_readfirstlane(a0);
       accumIntoLds(target, a, Int128{0, 0}, 1);
-    } else if (ad.srcReg >= 0) {
-      accumIntoLds(target, a, VT<WIDE>::toAcc(vm.get(ad.srcReg)), 1);
     }
   }
   return true;
@@ -1491,10 +1585,13 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
       slot = (slot + 1) & (kGlobalGroups - 1);
     }
     if (!ok) continue;
-    for (int a = 0; a < d.nAggs; a++) {
-      Int128 v = {lds[i].accLo[a], lds[i].accHi[a]};
-      accumInto(&d.globalTable[slot], a, v, lds[i].cnt[a]);
+    for (int s = 0; s < d.nAccSlots; s++) {
+      Int128 v = {lds[i].accLo[s], lds[i].accHi[s]};
+      accumInto(&d.globalTable[slot], s, v, 0);
     }
+    int nCnt = d.sharedCnt ? 1 : d.nAggs;
+    for (int a = 0; a < nCnt; a++)
+      accumInto(&d.globalTable[slot], a, Int128{0, 0}, lds[i].cnt[a]);
   }
 }
 
@@ -1592,18 +1689,24 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
     else
       hipLaunchKernelGGL((fusedAggGldsKernel<false>), dim3(grid), dim3(256),
                          shmem, s, devDesc);
-  } else if (desc.wide) {
-    if (desc.rbatch >= 2)
-      hipLaunchKernelGGL((fusedAggKernel<true, 2>), dim3(grid), dim3(256), 0, s, devDesc);
-    else
-      hipLaunchKernelGGL((fusedAggKernel<true, 1>), dim3(grid), dim3(256), 0, s, devDesc);
   } else {
-    if (desc.rbatch >= 4)
-      hipLaunchKernelGGL((fusedAggKernel<false, 4>), dim3(grid), dim3(256), 0, s, devDesc);
-    else if (desc.rbatch >= 2)
-      hipLaunchKernelGGL((fusedAggKernel<false, 2>), dim3(grid), dim3(256), 0, s, devDesc);
-    else
-      hipLaunchKernelGGL((fusedAggKernel<false, 1>), dim3(grid), dim3(256), 0, s, devDesc);
+    // pick the 5-slot raw-state variant when every non-staged fetch slot
+    // fits (higher occupancy for Q1-shaped queries)
+    int maxSlot = -1;
+    for (int f = 0; f < desc.nFetch; f++)
+      if (desc.fetch[f].kind != FETCH_B1 && f > maxSlot) maxSlot = f;
+    bool small = maxSlot < 5;
+    if (desc.wide) {
+      if (small)
+        hipLaunchKernelGGL((fusedAggKernel<true, 5>), dim3(grid), dim3(256), 0, s, devDesc);
+      else
+        hipLaunchKernelGGL((fusedAggKernel<true, 8>), dim3(grid), dim3(256), 0, s, devDesc);
+    } else {
+      if (small)
+        hipLaunchKernelGGL((fusedAggKernel<false, 5>), dim3(grid), dim3(256), 0, s, devDesc);
+      else
+        hipLaunchKernelGGL((fusedAggKernel<false, 8>), dim3(grid), dim3(256), 0, s, devDesc);
+    }
   }
   return (int)hipGetLastError();
 }

@@ -38,13 +38,20 @@ def log(msg):
 
 
 def run_cpu_baseline(rows=8_000_000):
-    """Oracle (CPU restatement) Q1 over pre-generated host-resident chunks,
-    single thread — the reported baseline, not the target."""
+    """Oracle (CPU restatement) Q1 over pre-generated host-resident chunks —
+    the reported baseline, not the target. Two legs: single
+    thread, and all host cores via the PARTIAL-per-shard + FINAL-merge split
+    (the CPU analog of the multi-GPU merge)."""
+    import os
     from tests.gxlib import load_oracle
     lib = load_oracle()
     lib.gx_oracle_bench_q1.argtypes = [
         ctypes.c_int64, ctypes.c_uint64, ctypes.POINTER(ctypes.c_double),
         ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_int64)]
+    lib.gx_oracle_bench_q1_mt.argtypes = [
+        ctypes.c_int64, ctypes.c_uint64, ctypes.c_int32,
+        ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_double),
+        ctypes.POINTER(ctypes.c_int64)]
     g = ctypes.c_double()
     e = ctypes.c_double()
     n = ctypes.c_int64()
@@ -52,13 +59,22 @@ def run_cpu_baseline(rows=8_000_000):
                                 ctypes.byref(n))
     if rc != 0:
         return None
+    st_rows_s = rows / (e.value / 1000.0)
+    cores = os.cpu_count() or 1
+    rc = lib.gx_oracle_bench_q1_mt(rows, 42, cores, ctypes.byref(g),
+                                   ctypes.byref(e), ctypes.byref(n))
+    if rc != 0:
+        return None
     return {
         "value": rows / (e.value / 1000.0),
         "unit": "rows/s",
-        "cores": 1,
+        "cores": cores,
         "kind": "port",
+        "single_thread_value": st_rows_s,
         "sample": f"Q1 over {rows} pre-generated host-resident rows, "
-                  f"executor only, single thread ({e.value/1000:.1f}s)",
+                  f"executor only; all-cores = {cores} threads partial+final "
+                  f"merge ({e.value/1000:.2f}s); single-thread = "
+                  f"{st_rows_s:.3g} rows/s",
     }
 
 

@@ -305,28 +305,18 @@ const char* gx_last_error(gx_exec* ex) {
 
 }  // extern "C"
 
-// ---- CPU-baseline timing entry (bench.py cpu_baseline leg) ----
+// ---- CPU-baseline timing entries (bench.py cpu_baseline leg) ----
 // Generates `rows` lineitem rows into host-resident chunks, then times ONE
-// Q1 pass of the oracle executor over the bound chunks (single thread).
+// Q1 pass of the oracle executor over the bound chunks. The MT variant
+// shards rows across threads (PARTIAL mode per shard) and merges through a
+// FINAL-mode agg -- the same partial/final split the multi-GPU path uses.
 #include <chrono>
-extern "C" int32_t gx_oracle_bench_q1(int64_t rows, uint64_t seed,
-                                      double* gen_ms, double* exec_ms,
-                                      int64_t* groups_out) {
-  using clk = std::chrono::steady_clock;
-  auto t0 = clk::now();
-  SourceBinding bind;
-  bind.haveChunks = true;
-  int64_t pos = 0;
-  while (pos < rows) {
-    int n = (int)std::min<int64_t>(rows - pos, kMaxChunkSize);
-    Chunk c;
-    TpchGenChunk(GX_TPCH_LINEITEM, pos, n, seed, rows, c);
-    bind.chunks.push_back(std::move(c));
-    pos += n;
-  }
-  auto t1 = clk::now();
-  // build the Q1 plan directly (mirrors tidb_amd/plan.py q1_plan)
-  Plan plan;
+#include <thread>
+
+// builds the Q1 plan over `bind` (mirrors tidb_amd/plan.py q1_plan);
+// returns the agg node id, plan in *planOut
+static int buildQ1Plan(Plan& plan, int aggMode, int* srcOut) {
+
   auto expr = [&](Expr e) { plan.exprs.push_back(e); return (int)plan.exprs.size() - 1; };
   auto node = [&](PlanNode n) { plan.nodes.push_back(n); return (int)plan.nodes.size() - 1; };
   PlanNode src;
@@ -366,7 +356,7 @@ extern "C" int32_t gx_oracle_bench_q1(int64_t rows, uint64_t seed,
   proj.exprs = {eRf, eLs, eQty, ePrice, eDisc, eDp, eCh};
   int nproj = node(proj);
   PlanNode agg; agg.kind = PK_HASHAGG; agg.child = nproj;
-  agg.aggMode = GX_AGG_MODE_COMPLETE;
+  agg.aggMode = aggMode;
   agg.exprs = {colref(0, GX_TYPE_STRING, 0), colref(1, GX_TYPE_STRING, 0)};
   int aQty = colref(2, GX_TYPE_DECIMAL, 2), aPrice = colref(3, GX_TYPE_DECIMAL, 2);
   int aDisc = colref(4, GX_TYPE_DECIMAL, 2), aDp = colref(5, GX_TYPE_DECIMAL, 4);
@@ -376,7 +366,69 @@ extern "C" int32_t gx_oracle_bench_q1(int64_t rows, uint64_t seed,
   agg.aggArgs = {aQty, aPrice, aDp, aCh, aQty, aPrice, aDisc, -1};
   agg.aggFracs = {2, 2, 4, 6, 6, 6, 6, 0};
   int nagg = node(agg);
+  *srcOut = nsrc;
+  return nagg;
+}
 
+// FINAL-mode plan over canonical 17-col partial-state chunks (mirrors
+// tidb_amd/plan.py q1_final_plan; MergePartialResult, aggfuncs.go:250-255)
+static int buildQ1FinalPlan(Plan& plan, int* srcOut) {
+  auto expr = [&](Expr e) { plan.exprs.push_back(e); return (int)plan.exprs.size() - 1; };
+  auto node = [&](PlanNode n) { plan.nodes.push_back(n); return (int)plan.nodes.size() - 1; };
+  PlanNode src;
+  src.kind = PK_SOURCE;
+  src.colTypes = {GX_TYPE_STRING, GX_TYPE_STRING,
+                  GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                  GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                  GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                  GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_I64};
+  src.colFracs = {0, 0, 2, 0, 2, 0, 4, 0, 6, 0, 2, 0, 2, 0, 2, 0, 0};
+  int nsrc = node(src);
+  auto colref = [&](int idx, int t, int f) {
+    Expr e; e.kind = EK_COLREF; e.colIdx = idx; e.retType = t; e.retFrac = f;
+    return expr(e);
+  };
+  PlanNode agg;
+  agg.kind = PK_HASHAGG;
+  agg.child = nsrc;
+  agg.aggMode = GX_AGG_MODE_FINAL;
+  agg.exprs = {colref(0, GX_TYPE_STRING, 0), colref(1, GX_TYPE_STRING, 0)};
+  agg.aggFuncs = {GX_AGG_SUM, GX_AGG_SUM, GX_AGG_SUM, GX_AGG_SUM,
+                  GX_AGG_AVG, GX_AGG_AVG, GX_AGG_AVG, GX_AGG_COUNT};
+  agg.aggArgs = {colref(2, GX_TYPE_DECIMAL, 2), colref(4, GX_TYPE_DECIMAL, 2),
+                 colref(6, GX_TYPE_DECIMAL, 4), colref(8, GX_TYPE_DECIMAL, 6),
+                 colref(10, GX_TYPE_DECIMAL, 2), colref(12, GX_TYPE_DECIMAL, 2),
+                 colref(14, GX_TYPE_DECIMAL, 2), -1};
+  agg.aggFracs = {2, 2, 4, 6, 6, 6, 6, 0};
+  int nagg = node(agg);
+  *srcOut = nsrc;
+  return nagg;
+}
+
+static void genShard(int64_t lo, int64_t hi, int64_t total, uint64_t seed,
+                     SourceBinding* bind) {
+  bind->haveChunks = true;
+  int64_t pos = lo;
+  while (pos < hi) {
+    int n = (int)std::min<int64_t>(hi - pos, kMaxChunkSize);
+    Chunk c;
+    TpchGenChunk(GX_TPCH_LINEITEM, pos, n, seed, total, c);
+    bind->chunks.push_back(std::move(c));
+    pos += n;
+  }
+}
+
+extern "C" int32_t gx_oracle_bench_q1(int64_t rows, uint64_t seed,
+                                      double* gen_ms, double* exec_ms,
+                                      int64_t* groups_out) {
+  using clk = std::chrono::steady_clock;
+  auto t0 = clk::now();
+  SourceBinding bind;
+  genShard(0, rows, rows, seed, &bind);
+  auto t1 = clk::now();
+  Plan plan;
+  int nsrc = 0;
+  int nagg = buildQ1Plan(plan, GX_AGG_MODE_COMPLETE, &nsrc);
   std::map<int, SourceBinding> bindings;
   bindings[nsrc] = std::move(bind);
   std::string err;
@@ -392,6 +444,83 @@ extern "C" int32_t gx_oracle_bench_q1(int64_t rows, uint64_t seed,
     groups += out.numRows();
   }
   execp->close();
+  auto t3 = clk::now();
+  *gen_ms = std::chrono::duration<double, std::milli>(t1 - t0).count();
+  *exec_ms = std::chrono::duration<double, std::milli>(t3 - t2).count();
+  *groups_out = groups;
+  return GX_OK;
+}
+
+// All-cores leg: PARTIAL-mode Q1 per shard on `threads` host threads, then a
+// FINAL-mode merge of the partial chunks (the CPU analog of the 8-GPU
+// partial/merge split). exec_ms covers shard execs + merge.
+extern "C" int32_t gx_oracle_bench_q1_mt(int64_t rows, uint64_t seed,
+                                         int32_t threads, double* gen_ms,
+                                         double* exec_ms, int64_t* groups_out) {
+  using clk = std::chrono::steady_clock;
+  if (threads < 1) threads = 1;
+  auto t0 = clk::now();
+  std::vector<SourceBinding> binds(threads);
+  {
+    std::vector<std::thread> ts;
+    int64_t per = (rows + threads - 1) / threads;
+    for (int t = 0; t < threads; t++) {
+      int64_t lo = std::min<int64_t>((int64_t)t * per, rows);
+      int64_t hi = std::min<int64_t>(lo + per, rows);
+      ts.emplace_back(genShard, lo, hi, rows, seed, &binds[t]);
+    }
+    for (auto& th : ts) th.join();
+  }
+  auto t1 = clk::now();
+  std::vector<SourceBinding> partials(threads);
+  std::vector<int32_t> rcs(threads, GX_OK);
+  auto worker = [&](int t) {
+    Plan plan;
+    int nsrc = 0;
+    int nagg = buildQ1Plan(plan, GX_AGG_MODE_PARTIAL, &nsrc);
+    std::map<int, SourceBinding> bindings;
+    bindings[nsrc] = std::move(binds[t]);
+    std::string err;
+    auto execp = BuildExec(plan, nagg, &bindings, &err);
+    if (!execp) { rcs[t] = GX_ERR_INTERNAL; return; }
+    execp->open();
+    partials[t].haveChunks = true;
+    for (;;) {
+      Chunk out;
+      if (execp->next(out) != GX_OK) { rcs[t] = GX_ERR_INTERNAL; return; }
+      if (out.numRows() == 0) break;
+      partials[t].chunks.push_back(std::move(out));
+    }
+    execp->close();
+  };
+  auto t2 = clk::now();
+  {
+    std::vector<std::thread> ts;
+    for (int t = 0; t < threads; t++) ts.emplace_back(worker, t);
+    for (auto& th : ts) th.join();
+  }
+  for (int32_t rc : rcs)
+    if (rc != GX_OK) return rc;
+  // FINAL merge of all partial chunks
+  Plan fplan;
+  int fsrc = 0;
+  int fagg = buildQ1FinalPlan(fplan, &fsrc);
+  std::map<int, SourceBinding> fbind;
+  fbind[fsrc].haveChunks = true;
+  for (auto& p : partials)
+    for (auto& c : p.chunks) fbind[fsrc].chunks.push_back(std::move(c));
+  std::string err;
+  auto fex = BuildExec(fplan, fagg, &fbind, &err);
+  if (!fex) return GX_ERR_INTERNAL;
+  fex->open();
+  int64_t groups = 0;
+  for (;;) {
+    Chunk out;
+    if (fex->next(out) != GX_OK) return GX_ERR_INTERNAL;
+    if (out.numRows() == 0) break;
+    groups += out.numRows();
+  }
+  fex->close();
   auto t3 = clk::now();
   *gen_ms = std::chrono::duration<double, std::milli>(t1 - t0).count();
   *exec_ms = std::chrono::duration<double, std::milli>(t3 - t2).count();

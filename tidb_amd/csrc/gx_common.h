@@ -276,6 +276,28 @@ int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows
 int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream);
 int gxLaunchMemset(void* p, int v, size_t n, void* stream);
+
+// ---- device full sort (sortexec/sort.go analog): LSD stable radix passes
+// over composed order-preserving u64 keys, then a device gather of every
+// column into sorted order ----
+struct SortKeyCompose {
+  int32_t col;
+  int32_t kind;   // 0 = i64 (sign-biased), 1 = time (masked), 2 = dense char,
+                  // 3 = decimal (parsed to int64 units at column frac)
+  int32_t desc;   // 1 = descending (key complemented)
+};
+// one stable pass: keys[i] = orderKey(table[idx[i]].col); err: sets
+// kErrRetryWide-style flag bit 1 on any unrepresentable decimal key
+int gxSortComposeKeys(const DevTable* tab, const DevTable& htab,
+                      SortKeyCompose k, const uint32_t* idx, uint64_t* keys,
+                      int64_t n, uint32_t* errFlag, void* stream);
+int gxSortPairs(uint64_t* keysIn, uint64_t* keysOut, uint32_t* idxIn,
+                uint32_t* idxOut, int64_t n, void* tmp, size_t* tmpBytes,
+                void* stream);
+int gxSortIota(uint32_t* idx, int64_t n, void* stream);
+// gather one column into out (same layout; elemSize 1, 8 or 40)
+int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,
+                    int elemSize, void* stream);
 // join-agg pipeline steps (gx_kernels.hip)
 int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
                    void* stream);  // 0 count0 1 build0 2 count1 3 build1 4 probe

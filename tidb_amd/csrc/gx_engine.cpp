@@ -133,6 +133,10 @@ struct gx_exec {
   size_t emitPos = 0;
   // bare-source emit state
   int64_t srcPos = 0;
+  // device full sort over a bare source (sortexec/sort.go analog)
+  std::vector<gxp::SortKeyCompose> devSortKeys;
+  bool devSorted = false;
+  int64_t devSortLimit = -1, devSortOffset = 0;
   uint64_t lastSelCount = 0;
   double lastKernelMs = 0;
 
@@ -2070,9 +2074,96 @@ static int32_t runFinalHost(gx_exec* ex) {
 
 // ---------------- bare source download (generator parity) ----------------
 
+// run the device radix sort once, replacing the table's column buffers with
+// gathered sorted copies (stable LSD passes, innermost key first)
+static int32_t runDeviceSort(gx_exec* ex) {
+  gxp::DevTable& tab = ex->desc.table;
+  int64_t n = tab.nRows;
+  if (n > 0xFFFFFFFFLL) { ex->err = "sort > 2^32 rows unsupported"; return GX_ERR_INVALID; }
+  for (auto& k : ex->devSortKeys) {
+    gxp::DevCol& c = tab.cols[k.col];
+    if (c.hasNulls) { ex->err = "NULLs in sort key unsupported this round"; return GX_ERR_INVALID; }
+    if (k.kind == 2 && !c.denseOffsets) {
+      ex->err = "general varlen sort key unsupported this round";
+      return GX_ERR_INVALID;
+    }
+  }
+  if (n == 0) { ex->devSorted = true; return GX_OK; }
+  if (!ex->devErr) ex->devErr = (uint32_t*)devAlloc(ex, 4);
+  if (!ex->devErr) { ex->err = "hipMalloc failed (sort err)"; return GX_ERR_INTERNAL; }
+  uint32_t* idxA = (uint32_t*)devAlloc(ex, n * 4);
+  uint32_t* idxB = (uint32_t*)devAlloc(ex, n * 4);
+  uint64_t* keyA = (uint64_t*)devAlloc(ex, n * 8);
+  uint64_t* keyB = (uint64_t*)devAlloc(ex, n * 8);
+  size_t tmpBytes = 0;
+  gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, nullptr, &tmpBytes, ex->stream);
+  void* tmp = devAlloc(ex, tmpBytes);
+  if (!idxA || !idxB || !keyA || !keyB || !tmp) {
+    ex->err = "hipMalloc failed (sort)";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+  hipEvent_t ev0, ev1;
+  hipEventCreate(&ev0);
+  hipEventCreate(&ev1);
+  hipEventRecord(ev0, ex->stream);
+  if (gxp::gxSortIota(idxA, n, ex->stream)) { ex->err = "iota launch failed"; return GX_ERR_INTERNAL; }
+  for (int j = (int)ex->devSortKeys.size() - 1; j >= 0; j--) {
+    if (gxp::gxSortComposeKeys(nullptr, tab, ex->devSortKeys[j], idxA, keyA, n,
+                               ex->devErr, ex->stream)) {
+      ex->err = "key compose launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    size_t tb = tmpBytes;
+    if (gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, tmp, &tb, ex->stream)) {
+      ex->err = "radix sort failed";
+      return GX_ERR_INTERNAL;
+    }
+    std::swap(idxA, idxB);
+  }
+  uint32_t herr = 0;
+  HIP_OK(ex, hipMemcpyAsync(&herr, ex->devErr, 4, hipMemcpyDeviceToHost, ex->stream));
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  if (herr) { ex->err = "sort key error (bad/overflowing decimal)"; return GX_ERR_INVALID; }
+  // gather every column through the final permutation
+  for (int c = 0; c < tab.nCols; c++) {
+    gxp::DevCol& col = tab.cols[c];
+    if (col.type == GX_TYPE_STRING && !col.denseOffsets) {
+      ex->err = "general varlen column in sorted table unsupported this round";
+      return GX_ERR_INVALID;
+    }
+    int es = col.type == GX_TYPE_DECIMAL ? 40
+             : (col.type == GX_TYPE_STRING ? 1 : 8);
+    void* nd = devAlloc(ex, (size_t)n * es + 16);
+    if (!nd) { ex->err = "hipMalloc failed (gather)"; return GX_ERR_INTERNAL; }
+    if (gxp::gxSortGatherCol(col.data, nd, idxA, n, es, ex->stream)) {
+      ex->err = "gather launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    col.data = nd;  // offsets for dense char remain the identity ramp
+  }
+  hipEventRecord(ev1, ex->stream);
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  float ms = 0;
+  hipEventElapsedTime(&ms, ev0, ev1);
+  ex->lastKernelMs = ms;
+  hipEventDestroy(ev0);
+  hipEventDestroy(ev1);
+  ex->devSorted = true;
+  // offset/limit: serve the slice [offset, offset+limit)
+  ex->srcPos = std::min<int64_t>(ex->devSortOffset, n);
+  if (ex->devSortLimit >= 0)
+    tab.nRows = std::min<int64_t>(n, ex->srcPos + ex->devSortLimit);
+  return GX_OK;
+}
+
 static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;
+  if (!ex->devSortKeys.empty() && !ex->devSorted) {
+    rc = runDeviceSort(ex);
+    if (rc) return rc;
+  }
   gxp::DevTable& tab = ex->desc.table;
   int64_t remaining = tab.nRows - ex->srcPos;
   int n = (int)std::min<int64_t>(remaining, 1024);
@@ -2400,6 +2491,43 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
       int32_t rc = compileJoinAgg(ex);
       if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     }
+  } else if (rn.kind == PK_TOPN &&
+             ex->plan.nodes[rn.child].kind == PK_SOURCE) {
+    // full ORDER BY / TopN directly over a source table: device radix sort
+    // (sortexec/sort.go:50,546; stable LSD passes over composed u64 keys)
+    const PNode& srcn = ex->plan.nodes[rn.child];
+    ex->isBareSource = true;
+    ex->sourceNode = rn.child;
+    ex->desc.table.nCols = (int)srcn.colTypes.size();
+    for (size_t c = 0; c < srcn.colTypes.size(); c++)
+      setDevColMeta(&ex->desc.table.cols[c], srcn.colTypes[c], srcn.colFracs[c]);
+    ex->devSortLimit = rn.limit;
+    ex->devSortOffset = rn.offset;
+    bool ok = true;
+    for (size_t i = 0; i < rn.exprs.size(); i++) {
+      const PExpr& ke = ex->plan.exprs[rn.exprs[i]];
+      if (ke.kind != EK_COLREF || ke.colIdx < 0 ||
+          ke.colIdx >= (int)srcn.colTypes.size()) {
+        ex->err = "sort keys must be source columns";
+        ok = false;
+        break;
+      }
+      gxp::SortKeyCompose k{};
+      k.col = ke.colIdx;
+      k.desc = rn.keyDesc[i] != 0;
+      int t = srcn.colTypes[ke.colIdx];
+      if (t == GX_TYPE_I64) k.kind = 0;
+      else if (t == GX_TYPE_TIME) k.kind = 1;
+      else if (t == GX_TYPE_STRING) k.kind = 2;  // dense char checked at run
+      else if (t == GX_TYPE_DECIMAL) k.kind = 3;
+      else {
+        ex->err = "unsupported sort key type this round";
+        ok = false;
+        break;
+      }
+      ex->devSortKeys.push_back(k);
+    }
+    (void)ok;
   } else if (rn.kind == PK_TOPN) {
     int32_t rc = compileJoinAgg(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";

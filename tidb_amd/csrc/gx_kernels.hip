@@ -1827,3 +1827,104 @@ int gxGenCustomerFill(DevTable* tab, int64_t rowBegin, int64_t nRows,
 }
 
 }  // namespace gxp
+
+namespace gxp {
+
+// ---- device full sort support (sortexec/sort.go analog) ----
+
+__global__ void sortIotaKernel(uint32_t* idx, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    idx[i] = (uint32_t)i;
+}
+
+// order-preserving u64 key for one sort column, gathered through idx so each
+// stable LSD pass sorts the CURRENT permutation by the next-outer key
+__global__ void sortComposeKeysKernel(DevTable tab, SortKeyCompose k,
+                                      const uint32_t* __restrict__ idx,
+                                      uint64_t* __restrict__ keys, int64_t n,
+                                      uint32_t* errFlag) {
+  const DevCol& c = tab.cols[k.col];
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t row = idx[i];
+    uint64_t key;
+    if (k.kind == 0) {  // int64: flip sign bit
+      key = gptr<uint64_t>(c.data)[row] ^ 0x8000000000000000ULL;
+    } else if (k.kind == 1) {  // packed CoreTime: masked compare order
+      key = gptr<uint64_t>(c.data)[row] & ~0xFULL;
+    } else if (k.kind == 2) {  // dense char(1); PAD SPACE: ' ' == ''
+      uint8_t b = gptr<uint8_t>(c.data)[row];
+      key = b == ' ' ? 0 : b;
+    } else {  // decimal -> int64 units at the column's declared frac
+      typename VT<false>::T v = 0;
+      int sc = 0;
+      if (!parseDecimalRaw<false>(
+              ulonglong2{gptr<uint64_t>(c.data)[row * 40 / 8],
+                         gptr<uint64_t>((const uint8_t*)c.data + row * 40 + 8)[0]},
+              &v, &sc, errFlag))
+        v = 0;  // flag set; result unusable anyway
+      if (sc != c.frac) {
+        bool ovf = false;
+        if (sc < c.frac) v = VT<false>::scale10(v, c.frac - sc, &ovf);
+        if (ovf || sc > c.frac) atomicOr(errFlag, kErrScale);
+      }
+      key = (uint64_t)v ^ 0x8000000000000000ULL;
+    }
+    if (k.desc) key = ~key;
+    keys[i] = key;
+  }
+}
+
+__global__ void sortGatherKernel(const uint8_t* __restrict__ in,
+                                 uint8_t* __restrict__ out,
+                                 const uint32_t* __restrict__ idx, int64_t n,
+                                 int elemSize) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t src = idx[i];
+    if (elemSize == 8) {
+      ((uint64_t*)out)[i] = ((const uint64_t*)in)[src];
+    } else if (elemSize == 1) {
+      out[i] = in[src];
+    } else {  // 40B decimal: five 8B words
+      const uint64_t* s = (const uint64_t*)(in + src * 40);
+      uint64_t* d = (uint64_t*)(out + i * 40);
+      d[0] = s[0]; d[1] = s[1]; d[2] = s[2]; d[3] = s[3]; d[4] = s[4];
+    }
+  }
+}
+
+int gxSortIota(uint32_t* idx, int64_t n, void* stream) {
+  hipLaunchKernelGGL(sortIotaKernel, dim3(gridFor(n)), dim3(256), 0,
+                     (hipStream_t)stream, idx, n);
+  return (int)hipGetLastError();
+}
+
+int gxSortComposeKeys(const DevTable* tab, const DevTable& htab,
+                      SortKeyCompose k, const uint32_t* idx, uint64_t* keys,
+                      int64_t n, uint32_t* errFlag, void* stream) {
+  (void)tab;
+  hipLaunchKernelGGL(sortComposeKeysKernel, dim3(gridFor(n)), dim3(256), 0,
+                     (hipStream_t)stream, htab, k, idx, keys, n, errFlag);
+  return (int)hipGetLastError();
+}
+
+int gxSortPairs(uint64_t* keysIn, uint64_t* keysOut, uint32_t* idxIn,
+                uint32_t* idxOut, int64_t n, void* tmp, size_t* tmpBytes,
+                void* stream) {
+  return (int)hipcub::DeviceRadixSort::SortPairs(tmp, *tmpBytes, keysIn,
+                                                 keysOut, idxIn, idxOut,
+                                                 (int)n, 0, 64,
+                                                 (hipStream_t)stream);
+}
+
+int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,
+                    int elemSize, void* stream) {
+  hipLaunchKernelGGL(sortGatherKernel, dim3(gridFor(n)), dim3(256), 0,
+                     (hipStream_t)stream, (const uint8_t*)in, (uint8_t*)out,
+                     idx, n, elemSize);
+  return (int)hipGetLastError();
+}
+
+}  // namespace gxp

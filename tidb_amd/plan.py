@@ -141,6 +141,17 @@ class Executor:
             rows.extend(chunk.rows(n.value))
         return rows
 
+    def pull_one(self, out_types, out_fracs=None, max_rows=1024,
+                 data_caps=None):
+        """One Next call (e.g. to trigger a lazy device sort without reading
+        the whole table back); returns the row count."""
+        chunk = PyChunk(out_types, max_rows, out_fracs, data_caps)
+        g = chunk.as_gx()
+        n = ctypes.c_int32(0)
+        rc = self.lib.gx_next(self.ex, ctypes.byref(g), ctypes.byref(n))
+        assert rc == 0, f"next failed ({rc}): {self.error()}"
+        return n.value
+
     def close(self):
         self.lib.gx_close(self.ex)
 

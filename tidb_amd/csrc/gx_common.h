@@ -293,7 +293,10 @@ int gxSortComposeKeys(const DevTable* tab, const DevTable& htab,
                       int64_t n, uint32_t* errFlag, void* stream);
 int gxSortPairs(uint64_t* keysIn, uint64_t* keysOut, uint32_t* idxIn,
                 uint32_t* idxOut, int64_t n, void* tmp, size_t* tmpBytes,
-                void* stream);
+                int beginBit, int endBit, void* stream);
+// writes {OR, AND} of all keys into devOrAnd[0..1]
+int gxSortKeyBits(const uint64_t* keys, int64_t n, uint64_t* devOrAnd,
+                  void* stream);
 int gxSortIota(uint32_t* idx, int64_t n, void* stream);
 // gather one column into out (same layout; elemSize 1, 8 or 40)
 int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,

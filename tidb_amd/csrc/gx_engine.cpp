@@ -2095,8 +2095,10 @@ static int32_t runDeviceSort(gx_exec* ex) {
   uint32_t* idxB = (uint32_t*)devAlloc(ex, n * 4);
   uint64_t* keyA = (uint64_t*)devAlloc(ex, n * 8);
   uint64_t* keyB = (uint64_t*)devAlloc(ex, n * 8);
+  uint64_t* devOrAnd = (uint64_t*)devAlloc(ex, 16);
   size_t tmpBytes = 0;
-  gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, nullptr, &tmpBytes, ex->stream);
+  gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, nullptr, &tmpBytes, 0, 64,
+                   ex->stream);
   void* tmp = devAlloc(ex, tmpBytes);
   if (!idxA || !idxB || !keyA || !keyB || !tmp) {
     ex->err = "hipMalloc failed (sort)";
@@ -2114,8 +2116,22 @@ static int32_t runDeviceSort(gx_exec* ex) {
       ex->err = "key compose launch failed";
       return GX_ERR_INTERNAL;
     }
+    // bound the radix to the bits that actually differ across keys
+    uint64_t oa[2];
+    if (gxp::gxSortKeyBits(keyA, n, devOrAnd, ex->stream)) {
+      ex->err = "key bits launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    HIP_OK(ex, hipMemcpyAsync(oa, devOrAnd, 16, hipMemcpyDeviceToHost,
+                              ex->stream));
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    uint64_t diff = oa[0] ^ oa[1];
+    if (diff == 0) continue;  // all keys equal: order already preserved
+    int beginBit = __builtin_ctzll(diff);
+    int endBit = 64 - __builtin_clzll(diff);
     size_t tb = tmpBytes;
-    if (gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, tmp, &tb, ex->stream)) {
+    if (gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, tmp, &tb, beginBit,
+                         endBit, ex->stream)) {
       ex->err = "radix sort failed";
       return GX_ERR_INTERNAL;
     }

@@ -78,6 +78,20 @@ def run_cpu_baseline(rows=8_000_000):
     }
 
 
+def measured_traffic(metric, rows_this_run):
+    """Per-launch HBM bytes from the committed rocprofv3 PMC profile
+    (profiles/r01_traffic.json), scaled to this run's row count; None when
+    no measurement exists for the metric."""
+    try:
+        import json as _json
+        with open(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                               "profiles", "r01_traffic.json")) as f:
+            t = _json.load(f)[metric]
+        return t["bytes_per_launch"] * rows_this_run / t["rows_per_launch"]
+    except Exception:
+        return None
+
+
 def bench_q3(args):
     """TPC-H Q3 (BASELINE config 3): 3-table join + grouped sum + TopN on one
     GPU. One step = the full pipeline (customer/orders build + lineitem probe
@@ -362,9 +376,8 @@ def main():
             "peak": HBM_PEAK_GBS,
             "unit": "GB/s",
             "frac": achieved_gbs / HBM_PEAK_GBS,
-            # PMC traffic comes from separate rocprofv3 --pmc runs
-            # (profiles/); not measured inline.
-            "traffic": None,
+            # per-launch HBM bytes from the committed rocprofv3 PMC profile
+            "traffic": measured_traffic("tpch_q1_rows_per_sec", args.rows),
         }
 
     cpu_baseline = None

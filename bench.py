@@ -61,20 +61,21 @@ def run_cpu_baseline(rows=8_000_000):
         return None
     st_rows_s = rows / (e.value / 1000.0)
     cores = os.cpu_count() or 1
-    rc = lib.gx_oracle_bench_q1_mt(rows, 42, cores, ctypes.byref(g),
+    mt_rows = rows * 8 if cores >= 64 else rows  # keep the sample >~1s
+    rc = lib.gx_oracle_bench_q1_mt(mt_rows, 42, cores, ctypes.byref(g),
                                    ctypes.byref(e), ctypes.byref(n))
     if rc != 0:
         return None
     return {
-        "value": rows / (e.value / 1000.0),
+        "value": mt_rows / (e.value / 1000.0),
         "unit": "rows/s",
         "cores": cores,
         "kind": "port",
         "single_thread_value": st_rows_s,
-        "sample": f"Q1 over {rows} pre-generated host-resident rows, "
+        "sample": f"Q1 over {mt_rows} pre-generated host-resident rows, "
                   f"executor only; all-cores = {cores} threads partial+final "
-                  f"merge ({e.value/1000:.2f}s); single-thread = "
-                  f"{st_rows_s:.3g} rows/s",
+                  f"merge ({e.value/1000:.2f}s); single-thread over {rows} "
+                  f"rows = {st_rows_s:.3g} rows/s",
     }
 
 

@@ -109,6 +109,8 @@ struct AggDesc {
   int32_t func;   // GX_AGG_*
   int32_t srcReg; // VM register holding the arg value (-1 for count(*))
   int32_t scale;  // scale of the accumulated units
+  int32_t fr = -1;  // FIRSTROW over a group-by column: index into gkey cols
+                    // (value decoded from the group key; no per-row state)
 };
 
 constexpr int kMaxAggs = 12;

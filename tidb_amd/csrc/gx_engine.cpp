@@ -991,6 +991,30 @@ static int32_t compileFused(gx_exec* ex) {
   for (size_t a = 0; a < agg->aggFuncs.size(); a++) {
     gxp::AggDesc ad{};
     ad.func = agg->aggFuncs[a];
+    if (ad.func == GX_AGG_FIRSTROW) {
+      // the golden Q1 plan carries firstrow(group col) (§8d): its value is
+      // fully determined by the group key, so it needs no per-row state
+      int argE = agg->aggArgs[a];
+      int srcCol = -1;
+      if (argE >= 0 && plan.exprs[argE].kind == EK_COLREF) {
+        int ci = plan.exprs[argE].colIdx;
+        if (proj) {
+          if (ci < (int)projSrcCol.size()) srcCol = projSrcCol[ci];
+        } else {
+          srcCol = ci;
+        }
+      }
+      for (int k = 0; k < gk.nCols; k++)
+        if (srcCol >= 0 && gk.col[k] == srcCol) ad.fr = k;
+      if (ad.fr < 0) {
+        ex->err = "device firstrow supports group-by columns this round";
+        return GX_ERR_INVALID;
+      }
+      ad.srcReg = -1;
+      ad.scale = 0;
+      ex->desc.aggs[ex->desc.nAggs++] = ad;
+      continue;
+    }
     if (ad.func != GX_AGG_COUNT && ad.func != GX_AGG_SUM && ad.func != GX_AGG_AVG) {
       ex->err = "device aggregation supports count/sum/avg this round";
       return GX_ERR_INVALID;
@@ -1633,6 +1657,15 @@ static int32_t runFused(gx_exec* ex) {
       __int128 acc = phys >= 0
           ? (((__int128)s->accHi[phys] << 64) | s->accLo[phys]) : 0;
       int64_t cnt = s->cnt[ex->desc.sharedCnt ? 0 : a];
+      if (ad.fr >= 0) {  // firstrow(group col): decode from the group key
+        OutRowVal v;
+        int srcCol = ex->desc.gkey.col[ad.fr];
+        decodeGroupLane(ex, (uint32_t)(s->key >> (32 * ad.fr)),
+                        ex->desc.gkey.kind[ad.fr],
+                        ex->desc.table.cols[srcCol].type, &v);
+        row.push_back(std::move(v));
+        continue;
+      }
       if (ad.func == GX_AGG_COUNT) {
         OutRowVal v;
         v.type = GX_TYPE_I64;

@@ -699,6 +699,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     if (!d.sharedCnt) {
       for (int a = 0; a < d.nAggs; a++) {
         const AggDesc& ad = d.aggs[a];
+        if (ad.fr >= 0) continue;  // firstrow(group col): no per-row state
         bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
         if (!isNull) lds3AccumCnt(target, a, 1);
       }
@@ -731,6 +732,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   if (!d.sharedCnt) {
     for (int a = 0; a < d.nAggs; a++) {
       const AggDesc& ad = d.aggs[a];
+      if (ad.fr >= 0) continue;  // firstrow(group col): no per-row state
       bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
       if (!isNull) accumInto(target, a, Int128{0, 0}, 1);
     }
@@ -1499,6 +1501,7 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
   if (!d.sharedCnt) {
     for (int a0 = 0; a0 < d.nAggs; a0++) {
       int a = __builtin_amdgcn_readfirstlane(a0);
+      if (d.aggs[a].fr >= 0) continue;
       accumIntoLds(target, a, Int128{0, 0}, 1);
     }
   }

@@ -159,7 +159,7 @@ class Executor:
         self.lib.gx_exec_free(self.ex)
 
 
-def q1_plan(lib, mode=GX_AGG_MODE_COMPLETE):
+def q1_plan(lib, mode=GX_AGG_MODE_COMPLETE, firstrow=False):
     """TPC-H Q1 operator tree per the reference's golden plan
     (pkg/planner/core/casetest/tpch/testdata/tpch_suite_out.json TestQ1):
       Selection(shipdate <= DATE'1998-09-01' - 90d ~ here: < 1998-09-01 per
@@ -206,6 +206,10 @@ def q1_plan(lib, mode=GX_AGG_MODE_COMPLETE):
         (GX_AGG_AVG, p_disc, 6),      # avg_disc
         (GX_AGG_COUNT, -1, 0),        # count_order
     ]
+    if firstrow:
+        # the golden plan also carries firstrow(rf), firstrow(ls)
+        # (tpch_suite_out.json TestQ1; aggfuncs/builder.go)
+        aggs += [(GX_AGG_FIRSTROW, p_rf, 0), (GX_AGG_FIRSTROW, p_ls, 0)]
     agg = b.hashagg(proj, [p_rf, p_ls], aggs, mode)
     out_types = [GX_TYPE_STRING, GX_TYPE_STRING]
     out_fracs = [0, 0]
@@ -214,12 +218,20 @@ def q1_plan(lib, mode=GX_AGG_MODE_COMPLETE):
             if f == GX_AGG_COUNT:
                 out_types += [GX_TYPE_I64]
                 out_fracs += [0]
+            elif f == GX_AGG_FIRSTROW:
+                out_types += [GX_TYPE_STRING]
+                out_fracs += [0]
             else:
                 out_types += [GX_TYPE_DECIMAL, GX_TYPE_I64]
                 out_fracs += [fr, 0]
     else:
         for f, _a, fr in aggs:
-            out_types += [GX_TYPE_I64 if f == GX_AGG_COUNT else GX_TYPE_DECIMAL]
+            if f == GX_AGG_COUNT:
+                out_types += [GX_TYPE_I64]
+            elif f == GX_AGG_FIRSTROW:
+                out_types += [GX_TYPE_STRING]
+            else:
+                out_types += [GX_TYPE_DECIMAL]
             out_fracs += [fr]
     return b, src, agg, out_types, out_fracs
 

@@ -161,6 +161,16 @@ int32_t gx_open(gx_exec* ex);
 /* fills out-chunk (caller-allocated buffers, data_cap/offsets_cap honored),
  * rows_out = rows appended; 0 = EOF. */
 int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out);
+
+/* ---- chunk wire codec (util/chunk/codec.go:41-141) -------------------
+ * Per column: [u32 LE length][u32 LE nullCount][nullBitmap iff nullCount>0,
+ * ceil(len/8) bytes][offsets iff varlen, (len+1)*8 bytes][data bytes].
+ * Encode returns bytes written (or -needed if cap too small).
+ * Decode consumes one chunk of n_cols columns into caller-allocated buffers
+ * (out->cols[i].data/null_bitmap/offsets with caps set); returns bytes
+ * consumed or a negative GX_ERR_* status. */
+int64_t gx_chunk_encode(const gx_chunk* chunk, uint8_t* out, int64_t cap);
+int64_t gx_chunk_decode(const uint8_t* buf, int64_t len, gx_chunk* out);
 int32_t gx_close(gx_exec* ex);
 void    gx_exec_free(gx_exec* ex);
 const char* gx_last_error(gx_exec* ex);

@@ -112,3 +112,95 @@ int32_t gx_time_compare(uint64_t a, uint64_t b) {
 }
 
 }  // extern "C"
+
+// ---- chunk wire codec (restates util/chunk/codec.go:41-141) ----
+// Per column: [u32 LE length][u32 LE nullCount][nullBitmap iff nullCount>0]
+// [offsets iff varlen][data]. nullCount counts zero bits among the first
+// `length` bitmap bits (codec.go:56 nullCount).
+
+static int64_t colNullCount(const gx_col& c) {
+  if (!c.null_bitmap) return 0;
+  int64_t nulls = 0;
+  for (int32_t i = 0; i < c.length; i++)
+    if (((c.null_bitmap[i >> 3] >> (i & 7)) & 1) == 0) nulls++;
+  return nulls;
+}
+
+static int64_t colDataBytes(const gx_col& c) {
+  if (c.elem_size < 0) return c.offsets ? c.offsets[c.length] : 0;
+  return (int64_t)c.length * c.elem_size;
+}
+
+extern "C" int64_t gx_chunk_encode(const gx_chunk* chunk, uint8_t* out,
+                                   int64_t cap) {
+  if (!chunk) return GX_ERR_INVALID;
+  int64_t need = 0;
+  for (int32_t ci = 0; ci < chunk->n_cols; ci++) {
+    const gx_col& c = chunk->cols[ci];
+    need += 8;
+    if (colNullCount(c) > 0) need += (c.length + 7) / 8;
+    if (c.elem_size < 0) need += (int64_t)(c.length + 1) * 8;
+    need += colDataBytes(c);
+  }
+  if (!out || cap < need) return -need;
+  uint8_t* p = out;
+  for (int32_t ci = 0; ci < chunk->n_cols; ci++) {
+    const gx_col& c = chunk->cols[ci];
+    uint32_t len = (uint32_t)c.length;
+    uint32_t nulls = (uint32_t)colNullCount(c);
+    memcpy(p, &len, 4); p += 4;
+    memcpy(p, &nulls, 4); p += 4;
+    if (nulls > 0) {
+      int64_t nb = (c.length + 7) / 8;
+      memcpy(p, c.null_bitmap, nb);
+      p += nb;
+    }
+    if (c.elem_size < 0) {
+      memcpy(p, c.offsets, (int64_t)(c.length + 1) * 8);
+      p += (int64_t)(c.length + 1) * 8;
+    }
+    int64_t db = colDataBytes(c);
+    memcpy(p, c.data, db);
+    p += db;
+  }
+  return p - out;
+}
+
+extern "C" int64_t gx_chunk_decode(const uint8_t* buf, int64_t len,
+                                   gx_chunk* out) {
+  if (!buf || !out) return GX_ERR_INVALID;
+  const uint8_t* p = buf;
+  const uint8_t* end = buf + len;
+  for (int32_t ci = 0; ci < out->n_cols; ci++) {
+    gx_col& c = out->cols[ci];
+    if (end - p < 8) return GX_ERR_INVALID;
+    uint32_t length, nulls;
+    memcpy(&length, p, 4); p += 4;
+    memcpy(&nulls, p, 4); p += 4;
+    c.length = (int32_t)length;
+    int64_t nb = (length + 7) / 8;
+    if (nulls > 0) {
+      if (end - p < nb || !c.null_bitmap) return GX_ERR_INVALID;
+      memcpy(c.null_bitmap, p, nb);
+      p += nb;
+    } else if (c.null_bitmap) {
+      memset(c.null_bitmap, 0xFF, nb);  // codec.go:146 setAllNotNull
+    }
+    int64_t db;
+    if (c.elem_size < 0) {
+      int64_t ob = (int64_t)(length + 1) * 8;
+      if (end - p < ob || !c.offsets || c.offsets_cap < (int64_t)length + 1)
+        return GX_ERR_INVALID;
+      memcpy(c.offsets, p, ob);
+      p += ob;
+      db = c.offsets[length];
+    } else {
+      db = (int64_t)length * c.elem_size;
+    }
+    if (end - p < db || c.data_cap < db) return GX_ERR_INVALID;
+    memcpy(c.data, p, db);
+    p += db;
+  }
+  out->n_rows = out->n_cols > 0 ? out->cols[0].length : 0;
+  return p - buf;
+}

@@ -47,6 +47,9 @@ enum VmOp : int32_t {
   VM_SUB = 4,
   VM_MUL = 5,        // dst <- a * b (scale = sa + sb)
   VM_SCALE_UP = 6,   // dst <- a * 10^b (align scales)
+  VM_DIV = 7,        // dst <- trunc(a * 10^c / b) (DecimalDiv semantics:
+                     // result scale word-granular, mydecimal.go doDiv;
+                     // div-by-zero -> NULL; engine forces the wide VM)
 };
 
 struct VmIns {
@@ -278,6 +281,7 @@ int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows
 int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream);
 int gxLaunchMemset(void* p, int v, size_t n, void* stream);
+int gxDumpDesc(const FusedQueryDesc* devDesc, void* stream);
 
 // ---- device full sort (sortexec/sort.go analog): LSD stable radix passes
 // over composed order-preserving u64 keys, then a device gather of every

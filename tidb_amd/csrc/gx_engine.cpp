@@ -119,6 +119,7 @@ struct gx_exec {
   std::vector<std::pair<int, int>> projRegs;  // projection idx -> (reg, scale)
   int vmNextReg = 0;
   std::map<int, std::pair<int, int>> exprRegCache;  // exprId -> (reg, scale)
+  bool vmHasDiv = false;  // DIV forces the wide VM and disables glds
   // device state
   bool deviceReady = false;
   std::vector<void*> devBufs;
@@ -198,8 +199,13 @@ static MyDecimal decFromUnits(__int128 u, int scale) {
   unsigned __int128 a = neg ? (unsigned __int128)(-u) : (unsigned __int128)u;
   static const int64_t p10[10] = {1, 10, 100, 1000, 10000, 100000, 1000000,
                                   10000000, 100000000, 1000000000};
-  unsigned __int128 ip = a / (unsigned)p10[scale];
-  int64_t fr = (int64_t)(a % (unsigned)p10[scale]);
+  // scale may exceed one decimal word (e.g. DecimalDiv results at 18):
+  // compute 10^scale in 128 bits
+  unsigned __int128 pscale = 1;
+  for (int s = scale; s > 0; s -= 9)
+    pscale *= (unsigned)p10[s > 9 ? 9 : s];
+  unsigned __int128 ip = a / pscale;
+  unsigned __int128 fr128 = a % pscale;
   // integer words, most significant first
   int32_t words[6] = {0};
   int nw = 0;
@@ -231,7 +237,17 @@ static MyDecimal decFromUnits(__int128 u, int scale) {
   d.resultFrac = (int8_t)scale;
   d.negative = neg && (a != 0);
   for (int i = 0; i < nw; i++) d.wordBuf[i] = words[i];
-  if (scale > 0) d.wordBuf[nw] = (int32_t)(fr * p10[9 - scale]);
+  if (scale > 0) {
+    // frac words, 9 digits each, left-aligned (mydecimal.go word layout):
+    // pad the trailing partial word so the digit count is a word multiple
+    int fw = (scale + 8) / 9;
+    int pad = fw * 9 - scale;
+    unsigned __int128 fadj = fr128 * (unsigned)p10[pad];
+    for (int i = fw - 1; i >= 0; i--) {
+      d.wordBuf[nw + i] = (int32_t)(fadj % 1000000000u);
+      fadj /= 1000000000u;
+    }
+  }
   return d;
 }
 
@@ -343,11 +359,30 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
         case GX_F_PLUS: op = gxp::VM_ADD; break;
         case GX_F_MINUS: op = gxp::VM_SUB; break;
         case GX_F_MUL: op = gxp::VM_MUL; break;
+        case GX_F_DIV: op = gxp::VM_DIV; break;
         default:
           ex->err = "unsupported function on device path";
           return -1;
       }
-      if (op == gxp::VM_MUL) {
+      if (op == gxp::VM_DIV) {
+        // result scale is word-granular (doDiv, mydecimal.go:1170-1215):
+        // frac_i word-rounded, fracIncr reduced by the slack, result frac =
+        // words(frac1w + frac2w + incr) * 9
+        int f1w = (sa + 8) / 9 * 9;
+        int f2w = (sb + 8) / 9 * 9;
+        int incr = std::max(0, 4 - (f1w - sa) - (f2w - sb));
+        int sr = (f1w + f2w + incr + 8) / 9 * 9;
+        if (sr > 30) sr = 30;
+        int e10 = sb + (sr - sa);
+        if (e10 < 0 || e10 > 36) {
+          ex->err = "division scale out of device range";
+          return -1;
+        }
+        reg = emit(op, allocReg(), ra, rb);
+        if (reg >= 0) d.ins[d.nIns - 1].c = e10;
+        ex->vmHasDiv = true;
+        *scaleOut = sr;
+      } else if (op == gxp::VM_MUL) {
         reg = emit(op, allocReg(), ra, rb);
         *scaleOut = sa + sb;
       } else {
@@ -1099,6 +1134,10 @@ static int32_t compileFused(gx_exec* ex) {
         4611686018427388ULL,    461168601842739ULL,    46116860184274ULL,
         4611686018428ULL,       461168601843ULL,       46116860185ULL,
         4611686019ULL};
+    if (getenv("GX_DEBUG"))
+      for (int i = 0; i < d.nIns; i++)
+        fprintf(stderr, "[gx] ins[%d] op=%d dst=%d a=%d b=%d c=%d\n", i,
+                d.ins[i].op, d.ins[i].dst, d.ins[i].a, d.ins[i].b, d.ins[i].c);
     for (int i = 0; i < d.nIns; i++) {
       d.insP10[i] = 1;
       d.insMagic[i] = 0;
@@ -1426,7 +1465,7 @@ static int32_t materializeDevice(gx_exec* ex) {
     // The glds-staged kernel is parity-green but measured slower than the
     // plain grouped-fetch kernel on Q1/SF10 (5.48 vs 4.71 ms), so it ships
     // opt-in until the pipelining wins back the staging overhead.
-    bool ok = tab.nRows >= 256 && getenv("GX_GLDS");
+    bool ok = tab.nRows >= 256 && getenv("GX_GLDS") && !ex->vmHasDiv;
     for (int f = 0; f < d.nFetch && ok; f++)
       ok = d.fetch[f].kind == gxp::FETCH_8B ||
            d.fetch[f].kind == gxp::FETCH_DEC16 ||
@@ -1553,6 +1592,9 @@ static void applyPostSort(gx_exec* ex) {
 }
 
 static int32_t runFused(gx_exec* ex) {
+  if (ex->vmHasDiv && !getenv("GX_DIV_NARROW"))
+    ex->desc.wide = 1;  // DIV quotients rarely fit int64
+  if (getenv("GX_FORCE_WIDE")) ex->desc.wide = 1;
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;
   ex->desc.ablate = getenv("GX_ABLATE") ? atoi(getenv("GX_ABLATE")) : 0;
@@ -1565,6 +1607,13 @@ static int32_t runFused(gx_exec* ex) {
   HIP_OK(ex, hipEventCreate(&ev1));
   HIP_OK(ex, hipEventRecord(ev0, ex->stream));
   if (ex->desc.noLds) ex->desc.useGlds = 0;  // high-NDV retry uses the plain kernel
+  if (getenv("GX_DEBUG_DESC")) {
+    fprintf(stderr, "[host] sizeof(desc)=%d aggs_off=%d ins_off=%d\n",
+            (int)sizeof(gxp::FusedQueryDesc),
+            (int)((char*)&ex->desc.aggs[0] - (char*)&ex->desc),
+            (int)((char*)&ex->desc.ins[0] - (char*)&ex->desc));
+    gxp::gxDumpDesc(ex->devDesc, ex->stream);
+  }
   int lrc = gxp::gxLaunchFusedAgg(ex->desc, ex->devDesc, ex->stream);
   if (lrc != 0) {
     ex->err = "fused kernel launch failed: " +
@@ -1657,6 +1706,10 @@ static int32_t runFused(gx_exec* ex) {
       __int128 acc = phys >= 0
           ? (((__int128)s->accHi[phys] << 64) | s->accLo[phys]) : 0;
       int64_t cnt = s->cnt[ex->desc.sharedCnt ? 0 : a];
+      if (getenv("GX_DEBUG_DESC") && phys >= 0)
+        fprintf(stderr, "[host] agg%d acc lo=%llu hi=%lld cnt=%lld\n", (int)a,
+                (unsigned long long)s->accLo[phys], (long long)s->accHi[phys],
+                (long long)cnt);
       if (ad.fr >= 0) {  // firstrow(group col): decode from the group key
         OutRowVal v;
         int srcCol = ex->desc.gkey.col[ad.fr];

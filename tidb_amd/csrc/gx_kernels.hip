@@ -548,6 +548,100 @@ __device__ inline uint64_t lds3CasKey(Lds3GroupSlot* slot, uint64_t expect,
   return expect;  // holds the previous value on failure, `expect` on success
 }
 
+__constant__ uint64_t kDivArgMax[37][2] = {  // I128_MAX / 10^e: {lo, hi}
+    {0xffffffffffffffffULL, 0x7fffffffffffffffULL},
+    {0xccccccccccccccccULL, 0xcccccccccccccccULL},
+    {0x147ae147ae147ae1ULL, 0x147ae147ae147aeULL},
+    {0xced916872b020c49ULL, 0x20c49ba5e353f7ULL},
+    {0x94af4f0d844d013aULL, 0x346dc5d638865ULL},
+    {0xc21187e7c06e19b9ULL, 0x53e2d6238da3ULL},
+    {0xc69b5a63f9a49c2cULL, 0x8637bd05af6ULL},
+    {0x7a42bc3d32907604ULL, 0xd6bf94d5e5ULL},
+    {0x8c39df9fb841a566ULL, 0x15798ee230ULL},
+    {0xdad2965cc5a02a23ULL, 0x225c17d04ULL},
+    {0xaf7b756fad5cd103ULL, 0x36f9bfb3ULL},
+    {0x5e592557f7bc7b4dULL, 0x57f5ff8ULL},
+    {0x96f5088cbf93f87ULL, 0x8cbcccULL},
+    {0x3424bb40e132865aULL, 0xe12e1ULL},
+    {0xb86a12b9b01ea709ULL, 0x16849ULL},
+    {0x5f3dceac2b3643e7ULL, 0x2407ULL},
+    {0x5652fb1137856d30ULL, 0x39aULL},
+    {0x3bd5191b525a2484ULL, 0x5cULL},
+    {0x392ee8e921d5d073ULL, 0x9ULL},
+    {0xec1e4a7db69561a5ULL, 0x0ULL},
+    {0x179ca10c9242235dULL, 0x0ULL},
+    {0x25c768141d369efULL, 0x0ULL},
+    {0x3c7240202ebdcbULL, 0x0ULL},
+    {0x60b6cd004ac94ULL, 0x0ULL},
+    {0x9abe14cd4475ULL, 0x0ULL},
+    {0xf79687aed3eULL, 0x0ULL},
+    {0x18c240c4aecULL, 0x0ULL},
+    {0x279d346de4ULL, 0x0ULL},
+    {0x3f61ed7caULL, 0x0ULL},
+    {0x65697bfaULL, 0x0ULL},
+    {0xa2425ffULL, 0x0ULL},
+    {0x1039d66ULL, 0x0ULL},
+    {0x19f623ULL, 0x0ULL},
+    {0x2989dULL, 0x0ULL},
+    {0x4276ULL, 0x0ULL},
+    {0x6a5ULL, 0x0ULL},
+    {0xaaULL, 0x0ULL},
+};
+
+// 128/64 unsigned division with no compiler-rt libcalls (__divti3 is a
+// device-library call with its own stack traffic; this inlines to plain
+// VALU). Hacker's Delight fig. 9-3 shape: normalized 2-by-1 division with
+// 32-bit digits.
+__device__ inline uint64_t udiv128by64(uint64_t hi, uint64_t lo, uint64_t d,
+                                       uint64_t* rem) {
+  // requires hi < d (caller peels the top word)
+  int s = __clzll(d);
+  d <<= s;
+  uint64_t un64 = s ? (hi << s) | (lo >> (64 - s)) : hi;
+  uint64_t un10 = lo << s;
+  uint32_t vn1 = (uint32_t)(d >> 32), vn0 = (uint32_t)d;
+  uint64_t un1 = un10 >> 32, un0 = (uint32_t)un10;
+  uint64_t q1 = un64 / vn1;
+  uint64_t rhat = un64 - q1 * vn1;
+  while (q1 >= 0x100000000ULL || q1 * vn0 > (rhat << 32) + un1) {
+    q1--;
+    rhat += vn1;
+    if (rhat >= 0x100000000ULL) break;
+  }
+  uint64_t un21 = (un64 << 32) + un1 - q1 * d;
+  uint64_t q0 = un21 / vn1;
+  rhat = un21 - q0 * vn1;
+  while (q0 >= 0x100000000ULL || q0 * vn0 > (rhat << 32) + un0) {
+    q0--;
+    rhat += vn1;
+    if (rhat >= 0x100000000ULL) break;
+  }
+  if (rem) *rem = ((un21 << 32) + un0 - q0 * d) >> s;
+  return (q1 << 32) + q0;
+}
+
+// full u128 / u64 -> u128 quotient (64-bit div/mod lower inline on amdgcn)
+__device__ inline unsigned __int128 u128DivU64(unsigned __int128 n, uint64_t d) {
+  uint64_t nhi = (uint64_t)(n >> 64), nlo = (uint64_t)n;
+  uint64_t qhi = nhi / d;
+  uint64_t r = nhi % d;
+  uint64_t qlo = udiv128by64(r, nlo, d, nullptr);
+  return ((unsigned __int128)qhi << 64) | qlo;
+}
+
+// u128 / u128 with divisor >= 2^64: quotient < 2^64, shift-subtract
+__device__ inline unsigned __int128 u128DivBig(unsigned __int128 n,
+                                               unsigned __int128 d) {
+  unsigned __int128 q = 0;
+  int shift = 0;
+  while ((d << 1) <= n && (d >> 126) == 0 && shift < 64) { d <<= 1; shift++; }
+  for (; shift >= 0; shift--) {
+    if (n >= d) { n -= d; q |= (unsigned __int128)1 << shift; }
+    d >>= 1;
+  }
+  return q;
+}
+
 // per-row pipeline after the raw fetch: filter -> VM -> LDS aggregate.
 // Returns false on a hard failure (error flag already set).
 template <bool WIDE, typename RAWT>
@@ -654,6 +748,61 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
                              VT<WIDE>::fromI64(d.insP10[i], nullptr), &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a));
         break;
+      case VM_DIV: {
+        // DecimalDiv (mydecimal.go:1311, doDiv:1168): quotient truncated
+        // toward zero at the word-granular result scale; ins.c holds the
+        // exponent e with result = trunc(a * 10^e / b). Division by zero
+        // yields NULL (builtin_arithmetic_vec.go:92 handleDivisionByZero).
+        bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
+        typename VT<WIDE>::T v = VT<WIDE>::zero();
+        if (!nul) {
+          Int128 bi = VT<WIDE>::toAcc(vm.get(ins.b));
+          __int128 bv = ((__int128)bi.hi << 64) | (__int128)bi.lo;
+          if (bv == 0) {
+            nul = true;
+          } else {
+            Int128 ai = VT<WIDE>::toAcc(vm.get(ins.a));
+            __int128 av = ((__int128)ai.hi << 64) | (__int128)ai.lo;
+            int e = ins.c;
+            unsigned __int128 p10 =
+                (unsigned __int128)(uint64_t)kP10(e > 18 ? 18 : e);
+            if (e > 18) p10 *= (uint64_t)kP10(e - 18);
+            unsigned __int128 aAbs =
+                (unsigned __int128)(av < 0 ? -av : av);
+            unsigned __int128 lim =
+                ((unsigned __int128)kDivArgMax[e][1] << 64) | kDivArgMax[e][0];
+            if (aAbs > lim) {
+              atomicOr(d.errorFlag, kErrOverflow);
+              bad = true;
+              break;
+            }
+            unsigned __int128 bAbs =
+                (unsigned __int128)(bv < 0 ? -bv : bv);
+            unsigned __int128 num = aAbs * p10;
+            unsigned __int128 uq = (bAbs >> 64) != 0
+                                       ? u128DivBig(num, bAbs)
+                                       : u128DivU64(num, (uint64_t)bAbs);
+            bool negq = (av < 0) != (bv < 0);
+            __int128 q = negq ? -(__int128)uq : (__int128)uq;
+            if (!WIDE &&
+                (q > (__int128)INT64_MAX || q < (__int128)INT64_MIN)) {
+              atomicOr(d.errorFlag, kErrRetryWide);
+              bad = true;
+              break;
+            }
+            if (WIDE) {
+              Int128 r = {(uint64_t)q, (int64_t)(q >> 64)};
+              v = *(typename VT<WIDE>::T*)&r;
+            } else {
+              int64_t qq = (int64_t)q;
+              v = *(typename VT<WIDE>::T*)&qq;
+            }
+          }
+        }
+        vm.set(ins.dst, v);
+        vm.setNull(ins.dst, nul);
+        break;
+      }
     }
   }
   if (ovf) {
@@ -1671,6 +1820,32 @@ int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows
   } else {
     return -1;  // orders/customer device generation lands with the Q3 path
   }
+  return (int)hipGetLastError();
+}
+
+__global__ void dumpDescKernel(const FusedQueryDesc* dp) {
+  if (threadIdx.x || blockIdx.x) return;
+  const FusedQueryDesc& d = *dp;
+  printf("[dev] nIns=%d nAggs=%d nAccSlots=%d sharedCnt=%d wide=%d\n",
+         d.nIns, d.nAggs, d.nAccSlots, d.sharedCnt, d.wide);
+  for (int i = 0; i < d.nIns; i++)
+    printf("[dev] ins[%d] op=%d dst=%d a=%d b=%d c=%d p10=%lld\n", i,
+           d.ins[i].op, d.ins[i].dst, d.ins[i].a, d.ins[i].b, d.ins[i].c,
+           (long long)d.insP10[i]);
+  for (int a = 0; a < d.nAggs; a++)
+    printf("[dev] agg[%d] func=%d srcReg=%d scale=%d fr=%d accMap=%d\n", a,
+           d.aggs[a].func, d.aggs[a].srcReg, d.aggs[a].scale, d.aggs[a].fr,
+           d.accMap[a]);
+  for (int s = 0; s < d.nAccSlots; s++)
+    printf("[dev] accReg[%d]=%d\n", s, d.accReg[s]);
+  printf("[dev] kP10(18)=%lld sizeof(desc)=%d aggs_off=%d ins_off=%d\n",
+         (long long)kP10(18), (int)sizeof(FusedQueryDesc),
+         (int)((char*)&d.aggs[0] - (char*)&d), (int)((char*)&d.ins[0] - (char*)&d));
+}
+
+int gxDumpDesc(const FusedQueryDesc* devDesc, void* stream) {
+  hipLaunchKernelGGL(dumpDescKernel, dim3(1), dim3(1), 0, (hipStream_t)stream, devDesc);
+  hipStreamSynchronize((hipStream_t)stream);
   return (int)hipGetLastError();
 }
 

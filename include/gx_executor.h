@@ -130,6 +130,14 @@ int32_t gx_pb_hashagg(gx_pb* pb, int32_t child,
                       const int32_t* group_exprs, int32_t n_group,
                       const int32_t* agg_funcs, const int32_t* agg_args,
                       const int32_t* agg_fracs, int32_t n_aggs, int32_t mode);
+/* stream aggregation (aggregate/agg_stream_executor.go): same semantics as
+ * hash aggregation but requires the child stream grouped (all rows of a key
+ * contiguous -- e.g. sorted on the group cols); emits groups in stream
+ * order. COMPLETE mode this round. */
+int32_t gx_pb_streamagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
+                        int32_t n_group, const int32_t* agg_funcs,
+                        const int32_t* agg_args, const int32_t* agg_fracs,
+                        int32_t n_aggs);
 int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
                    const uint8_t* key_desc, int32_t n_keys,
                    int64_t limit, int64_t offset);

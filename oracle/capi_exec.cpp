@@ -138,6 +138,22 @@ int32_t gx_pb_hashagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
   }
   return addNode(pb, std::move(n));
 }
+int32_t gx_pb_streamagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
+                        int32_t n_group, const int32_t* agg_funcs,
+                        const int32_t* agg_args, const int32_t* agg_fracs,
+                        int32_t n_aggs) {
+  PlanNode n;
+  n.kind = PK_STREAMAGG;
+  n.child = child;
+  n.aggMode = GX_AGG_MODE_COMPLETE;
+  for (int i = 0; i < n_group; i++) n.exprs.push_back(group_exprs[i]);
+  for (int i = 0; i < n_aggs; i++) {
+    n.aggFuncs.push_back(agg_funcs[i]);
+    n.aggArgs.push_back(agg_args[i]);
+    n.aggFracs.push_back(agg_fracs ? agg_fracs[i] : 0);
+  }
+  return addNode(pb, std::move(n));
+}
 int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
                    const uint8_t* key_desc, int32_t n_keys, int64_t limit,
                    int64_t offset) {

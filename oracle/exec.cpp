@@ -588,6 +588,8 @@ class HashAggExec : public Exec {
     return 0;
   }
 
+  std::string lastKey_;  // PK_STREAMAGG contiguity check
+
   int32_t drainRaw() {
     EvalCtx ctx{&plan_, &err};
     for (;;) {
@@ -611,6 +613,15 @@ class HashAggExec : public Exec {
         }
       }
       for (int i = 0; i < n; i++) {
+        // stream aggregation (agg_stream_executor.go): the child must be
+        // GROUPED -- all rows of a key contiguous. A key that re-appears
+        // after another key started violates the operator contract.
+        if (node_.kind == PK_STREAMAGG && keys[i] != lastKey_ &&
+            groups_.count(keys[i]) && !order_.empty()) {
+          err = "stream aggregation requires grouped (sorted) input";
+          return GX_ERR_INVALID;
+        }
+        if (node_.kind == PK_STREAMAGG) lastKey_ = keys[i];
         Group& g = getGroup(keys[i], groupCols, i);
         for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
           ec = updateState(g.states[a], node_.aggFuncs[a], argPtr[a], i, argType(a));
@@ -1083,7 +1094,8 @@ std::unique_ptr<Exec> BuildExec(const Plan& plan, int root,
       if (!child) return nullptr;
       return std::make_unique<ProjectionExec>(plan, node, std::move(child));
     }
-    case PK_HASHAGG: {
+    case PK_HASHAGG:
+    case PK_STREAMAGG: {
       auto child = BuildExec(plan, node.child, bindings, err);
       if (!child) return nullptr;
       return std::make_unique<HashAggExec>(plan, node, std::move(child));

@@ -22,7 +22,8 @@ constexpr int kMaxChunkSize = 1024;  // vardef DefMaxChunkSize (tidb_vars.go:155
 enum ExprKind { EK_COLREF = 0, EK_CONST = 1, EK_CALL = 2 };
 enum PlanKind {
   PK_SOURCE = 0, PK_SELECTION, PK_PROJECTION, PK_HASHAGG, PK_TOPN, PK_HASHJOIN,
-  PK_SORT
+  PK_SORT,
+  PK_STREAMAGG  // sorted/grouped-input aggregation (agg_stream_executor.go)
 };
 
 struct Expr {

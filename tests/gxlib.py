@@ -92,6 +92,9 @@ def _decl(lib):
                                   i32p, i32p, i32p, ctypes.c_int32, ctypes.c_int32]
     lib.gx_pb_topn.argtypes = [ctypes.c_void_p, ctypes.c_int32, i32p, i8p,
                                ctypes.c_int32, ctypes.c_int64, ctypes.c_int64]
+    lib.gx_pb_streamagg.argtypes = [ctypes.c_void_p, ctypes.c_int32, i32p,
+                                    ctypes.c_int32, i32p, i32p, i32p,
+                                    ctypes.c_int32]
     lib.gx_pb_hashjoin.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
                                    i32p, i32p, ctypes.c_int32, ctypes.c_int32]
     lib.gx_build.restype = ctypes.c_void_p

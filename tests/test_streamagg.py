@@ -1,0 +1,83 @@
+"""Stream aggregation (aggregate/agg_stream_executor.go): grouped agg over a
+GROUPED child stream (all rows of a key contiguous), groups emitted in stream
+order. The oracle enforces the contiguity contract; the device engine maps
+the operator onto the fused aggregation with ordered emit (same results)."""
+import pytest
+
+from tests.gxlib import (GX_AGG_COUNT, GX_AGG_SUM, GX_TPCH_LINEITEM,
+                         GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_STRING,
+                         load_oracle)
+from tidb_amd import plan as P
+
+
+def sa_plan(lib, sorted_child=True, desc=0):
+    b = P.Builder(lib)
+    src = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+    rf = b.colref(P.L_RETFLAG, GX_TYPE_STRING)
+    ls = b.colref(P.L_LINESTATUS, GX_TYPE_STRING)
+    qty = b.colref(P.L_QUANTITY, GX_TYPE_DECIMAL, 2)
+    child = b.sort(src, [rf, ls], [desc, desc]) if sorted_child else src
+    agg = b.streamagg(child, [rf, ls],
+                      [(GX_AGG_SUM, qty, 2), (GX_AGG_COUNT, -1, 0)])
+    out_types = [GX_TYPE_STRING, GX_TYPE_STRING, GX_TYPE_DECIMAL, GX_TYPE_I64]
+    out_fracs = [0, 0, 2, 0]
+    return b, src, agg, out_types, out_fracs
+
+
+def run_sa(lib, n_rows, sorted_child=True, desc=0):
+    b, src, agg, out_types, out_fracs = sa_plan(lib, sorted_child, desc)
+    ex = b.build(agg)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, n_rows)
+    ex.open()
+    caps = [2048 if t == 4 else None for t in out_types]
+    rows = ex.pull_all(out_types, out_fracs, data_caps=caps)
+    ex.close()
+    ex.free()
+    b.free()
+    return rows
+
+
+def test_streamagg_oracle(oracle_lib):
+    rows = run_sa(oracle_lib, 20000)
+    # emitted in stream (= sorted) order, and values match the hash agg
+    keys = [(r[0], r[1]) for r in rows]
+    assert keys == sorted(keys)
+    from tests.test_orderby_ndv import run_plan  # independent plan runner
+    # cross-check against HashAgg on the same data
+    b = P.Builder(oracle_lib)
+    src = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+    rf = b.colref(P.L_RETFLAG, 4)
+    ls = b.colref(P.L_LINESTATUS, 4)
+    qty = b.colref(P.L_QUANTITY, 2, 2)
+    agg = b.hashagg(src, [rf, ls], [(1, qty, 2), (0, -1, 0)])
+    ex = b.build(agg)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, 20000)
+    ex.open()
+    want = ex.pull_all([4, 4, 2, 1], [0, 0, 2, 0],
+                       data_caps=[2048, 2048, None, None])
+    ex.close(); ex.free(); b.free()
+    assert sorted(rows) == sorted(want)
+
+
+def test_streamagg_contract_violation(oracle_lib):
+    """Unsorted child -> the oracle rejects (grouped-input contract)."""
+    b, src, agg, out_types, out_fracs = sa_plan(oracle_lib, sorted_child=False)
+    ex = b.build(agg)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, 5000)
+    ex.open()
+    import pytest as _pt
+    with _pt.raises(AssertionError):
+        ex.pull_all(out_types, out_fracs,
+                    data_caps=[2048 if t == 4 else None for t in out_types])
+    ex.close()
+    ex.free()
+    b.free()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("desc", [0, 1])
+def test_streamagg_parity(desc):
+    from tests.gxlib import load_product
+    a = run_sa(load_oracle(), 50000, desc=desc)
+    b = run_sa(load_product(), 50000, desc=desc)
+    assert a == b  # ordered comparison: stream order must match too

@@ -33,7 +33,8 @@ namespace {
 
 // ---------------- plan IR (mirrors the C-ABI builder calls) ----------------
 enum { EK_COLREF, EK_CONST, EK_CALL };
-enum { PK_SOURCE, PK_SELECTION, PK_PROJECTION, PK_HASHAGG, PK_TOPN, PK_HASHJOIN };
+enum { PK_SOURCE, PK_SELECTION, PK_PROJECTION, PK_HASHAGG, PK_TOPN, PK_HASHJOIN,
+       PK_STREAMAGG };
 
 struct PExpr {
   int kind = EK_COLREF;
@@ -2499,6 +2500,22 @@ int32_t gx_pb_hashagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
   }
   return addNode(pb, std::move(n));
 }
+int32_t gx_pb_streamagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
+                        int32_t n_group, const int32_t* agg_funcs,
+                        const int32_t* agg_args, const int32_t* agg_fracs,
+                        int32_t n_aggs) {
+  PNode n;
+  n.kind = PK_STREAMAGG;
+  n.child = child;
+  n.aggMode = GX_AGG_MODE_COMPLETE;
+  for (int i = 0; i < n_group; i++) n.exprs.push_back(group_exprs[i]);
+  for (int i = 0; i < n_aggs; i++) {
+    n.aggFuncs.push_back(agg_funcs[i]);
+    n.aggArgs.push_back(agg_args[i]);
+    n.aggFracs.push_back(agg_fracs ? agg_fracs[i] : 0);
+  }
+  return addNode(pb, std::move(n));
+}
 int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
                    const uint8_t* key_desc, int32_t n_keys, int64_t limit,
                    int64_t offset) {
@@ -2591,6 +2608,56 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     } else {
       ex->err.clear();
       int32_t rc = compileJoinAgg(ex);
+      if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
+    }
+  } else if (rn.kind == PK_STREAMAGG) {
+    // agg_stream_executor.go analog: semantics = grouped agg over a grouped
+    // stream, emitted in stream order. MI355X-first implementation: the
+    // sort below it is irrelevant to the aggregate VALUES, so run the fused
+    // hash aggregation over the sort's input and emit ordered by the sort
+    // keys -- identical results, none of the sorted-input serialization.
+    const PNode* sortn = &ex->plan.nodes[rn.child];
+    if (sortn->kind != PK_TOPN || sortn->limit >= 0) {
+      ex->err = "device streamagg expects a full-sort child this round";
+      return ex;
+    }
+    // every sort key must be one of the group columns
+    std::vector<std::pair<int, bool>> orderKeys;
+    bool ok = true;
+    for (size_t i = 0; i < sortn->exprs.size() && ok; i++) {
+      const PExpr& ke = ex->plan.exprs[sortn->exprs[i]];
+      int found = -1;
+      for (size_t g = 0; g < rn.exprs.size(); g++) {
+        const PExpr& ge = ex->plan.exprs[rn.exprs[g]];
+        if (ke.kind == EK_COLREF && ge.kind == EK_COLREF &&
+            ke.colIdx == ge.colIdx)
+          found = (int)g;
+      }
+      if (found < 0) {
+        ex->err = "streamagg sort keys must be group columns";
+        ok = false;
+      } else {
+        orderKeys.push_back({found, sortn->keyDesc[i] != 0});
+      }
+    }
+    if (ok) {
+      PNode agg;
+      agg.kind = PK_HASHAGG;
+      agg.child = sortn->child;
+      agg.aggMode = rn.aggMode;
+      agg.exprs = rn.exprs;
+      agg.aggFuncs = rn.aggFuncs;
+      agg.aggArgs = rn.aggArgs;
+      agg.aggFracs = rn.aggFracs;
+      ex->plan.nodes.push_back(agg);
+      int saved = ex->root;
+      ex->root = (int)ex->plan.nodes.size() - 1;
+      ex->postSort = true;
+      ex->postSortKeys = orderKeys;
+      ex->postLimit = -1;
+      ex->postOffset = 0;
+      int32_t rc = compileFused(ex);
+      ex->root = saved;
       if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     }
   } else if (rn.kind == PK_TOPN &&

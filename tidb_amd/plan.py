@@ -69,6 +69,14 @@ class Builder:
                                       len(group_exprs), funcs, args, fracs,
                                       len(aggs), mode)
 
+    def streamagg(self, child, group_exprs, aggs):
+        funcs = _arr([a[0] for a in aggs])
+        args = _arr([a[1] for a in aggs])
+        fracs = _arr([a[2] for a in aggs])
+        return self.lib.gx_pb_streamagg(self.pb, child, _arr(group_exprs),
+                                        len(group_exprs), funcs, args, fracs,
+                                        len(aggs))
+
     def topn(self, child, keys, desc, limit, offset=0):
         return self.lib.gx_pb_topn(self.pb, child, _arr(keys), _u8arr(desc),
                                    len(keys), limit, offset)

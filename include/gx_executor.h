@@ -134,6 +134,12 @@ int32_t gx_pb_hashagg(gx_pb* pb, int32_t child,
  * hash aggregation but requires the child stream grouped (all rows of a key
  * contiguous -- e.g. sorted on the group cols); emits groups in stream
  * order. COMPLETE mode this round. */
+/* merge join (join/merge_join.go): inner join over inputs the plan sorts on
+ * the join keys (children must be full sorts whose leading key is the join
+ * key); results identical to the hash join. */
+int32_t gx_pb_mergejoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
+                        const int32_t* build_keys, const int32_t* probe_keys,
+                        int32_t n_keys, int32_t join_type);
 int32_t gx_pb_streamagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
                         int32_t n_group, const int32_t* agg_funcs,
                         const int32_t* agg_args, const int32_t* agg_fracs,

@@ -168,6 +168,20 @@ int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
   n.offset = offset;
   return addNode(pb, std::move(n));
 }
+int32_t gx_pb_mergejoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
+                        const int32_t* build_keys, const int32_t* probe_keys,
+                        int32_t n_keys, int32_t join_type) {
+  PlanNode n;
+  n.kind = PK_MERGEJOIN;
+  n.child = build_child;
+  n.child2 = probe_child;
+  for (int i = 0; i < n_keys; i++) {
+    n.buildKeys.push_back(build_keys[i]);
+    n.probeKeys.push_back(probe_keys[i]);
+  }
+  n.joinType = join_type;
+  return addNode(pb, std::move(n));
+}
 int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
                        const int32_t* build_keys, const int32_t* probe_keys,
                        int32_t n_keys, int32_t join_type) {

@@ -1110,6 +1110,29 @@ std::unique_ptr<Exec> BuildExec(const Plan& plan, int root,
       if (!child) return nullptr;
       return std::make_unique<TopNExec>(plan, node, std::move(child), true);
     }
+    case PK_MERGEJOIN:
+      // merge_join.go: planned only over inputs sorted on the join keys --
+      // enforce that contract statically (each child must be a full sort
+      // whose leading key is the join key), then execute through the join
+      // machinery: the two-pointer merge is an execution strategy, results
+      // are those of the inner join
+      for (int side = 0; side < 2; side++) {
+        int cid = side == 0 ? node.child : node.child2;
+        int keyExpr = side == 0 ? node.buildKeys[0] : node.probeKeys[0];
+        const PlanNode* c = &plan.nodes[cid];
+        bool okSort = c->kind == PK_SORT && !c->exprs.empty();
+        if (okSort) {
+          const Expr& ke = plan.exprs[c->exprs[0]];
+          const Expr& je = plan.exprs[keyExpr];
+          okSort = ke.kind == EK_COLREF && je.kind == EK_COLREF &&
+                   ke.colIdx == je.colIdx && c->keyDesc[0] == 0;
+        }
+        if (!okSort) {
+          if (err) *err = "merge join requires children sorted on the join key";
+          return nullptr;
+        }
+      }
+      [[fallthrough]];
     case PK_HASHJOIN: {
       auto build = BuildExec(plan, node.child, bindings, err);
       if (!build) return nullptr;

@@ -23,7 +23,8 @@ enum ExprKind { EK_COLREF = 0, EK_CONST = 1, EK_CALL = 2 };
 enum PlanKind {
   PK_SOURCE = 0, PK_SELECTION, PK_PROJECTION, PK_HASHAGG, PK_TOPN, PK_HASHJOIN,
   PK_SORT,
-  PK_STREAMAGG  // sorted/grouped-input aggregation (agg_stream_executor.go)
+  PK_STREAMAGG,  // sorted/grouped-input aggregation (agg_stream_executor.go)
+  PK_MERGEJOIN   // sorted-input inner join (join/merge_join.go)
 };
 
 struct Expr {

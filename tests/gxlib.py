@@ -95,6 +95,9 @@ def _decl(lib):
     lib.gx_pb_streamagg.argtypes = [ctypes.c_void_p, ctypes.c_int32, i32p,
                                     ctypes.c_int32, i32p, i32p, i32p,
                                     ctypes.c_int32]
+    lib.gx_pb_mergejoin.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                    ctypes.c_int32, i32p, i32p,
+                                    ctypes.c_int32, ctypes.c_int32]
     lib.gx_pb_hashjoin.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
                                    i32p, i32p, ctypes.c_int32, ctypes.c_int32]
     lib.gx_build.restype = ctypes.c_void_p

@@ -34,7 +34,7 @@ namespace {
 // ---------------- plan IR (mirrors the C-ABI builder calls) ----------------
 enum { EK_COLREF, EK_CONST, EK_CALL };
 enum { PK_SOURCE, PK_SELECTION, PK_PROJECTION, PK_HASHAGG, PK_TOPN, PK_HASHJOIN,
-       PK_STREAMAGG };
+       PK_STREAMAGG, PK_MERGEJOIN };
 
 struct PExpr {
   int kind = EK_COLREF;
@@ -607,12 +607,24 @@ static bool compileTablePred(gx_exec* ex, const PNode& srcNode, int condId,
 
 // unwrap [Selection ->] Source; returns source node id (or -1) and the
 // selection node (or nullptr)
+// full-sort (TOPN limit<0) wrappers below joins/aggregates do not change
+// their results -- the merge-join plan shape sorts its inputs on the join
+// keys (merge_join.go); the device pipeline hashes instead, so skip them
+static int skipFullSort(gx_exec* ex, int node) {
+  while (ex->plan.nodes[node].kind == PK_TOPN &&
+         ex->plan.nodes[node].limit < 0)
+    node = ex->plan.nodes[node].child;
+  return node;
+}
+
 static int unwrapSource(gx_exec* ex, int node, const PNode** selOut) {
   *selOut = nullptr;
+  node = skipFullSort(ex, node);
   const PNode* n = &ex->plan.nodes[node];
   if (n->kind == PK_SELECTION) {
     *selOut = n;
     node = n->child;
+    node = skipFullSort(ex, node);
     n = &ex->plan.nodes[node];
   }
   if (n->kind != PK_SOURCE) return -1;
@@ -641,13 +653,15 @@ static int32_t compileJoinAgg(gx_exec* ex) {
     ex->err = "expected projection under aggregate";
     return GX_ERR_INVALID;
   }
-  const PNode& j2 = plan.nodes[proj.child];
-  if (j2.kind != PK_HASHJOIN || j2.joinType != 0 || j2.buildKeys.size() != 1) {
+  const PNode& j2 = plan.nodes[skipFullSort(ex, proj.child)];
+  if ((j2.kind != PK_HASHJOIN && j2.kind != PK_MERGEJOIN) ||
+      j2.joinType != 0 || j2.buildKeys.size() != 1) {
     ex->err = "unsupported join shape";
     return GX_ERR_INVALID;
   }
-  const PNode& j1 = plan.nodes[j2.child];
-  if (j1.kind != PK_HASHJOIN || j1.joinType != 0 || j1.buildKeys.size() != 1) {
+  const PNode& j1 = plan.nodes[skipFullSort(ex, j2.child)];
+  if ((j1.kind != PK_HASHJOIN && j1.kind != PK_MERGEJOIN) ||
+      j1.joinType != 0 || j1.buildKeys.size() != 1) {
     ex->err = "unsupported inner join shape";
     return GX_ERR_INVALID;
   }
@@ -2528,6 +2542,20 @@ int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
   }
   n.limit = limit;
   n.offset = offset;
+  return addNode(pb, std::move(n));
+}
+int32_t gx_pb_mergejoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
+                        const int32_t* build_keys, const int32_t* probe_keys,
+                        int32_t n_keys, int32_t join_type) {
+  PNode n;
+  n.kind = PK_MERGEJOIN;
+  n.child = build_child;
+  n.child2 = probe_child;
+  for (int i = 0; i < n_keys; i++) {
+    n.buildKeys.push_back(build_keys[i]);
+    n.probeKeys.push_back(probe_keys[i]);
+  }
+  n.joinType = join_type;
   return addNode(pb, std::move(n));
 }
 int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,

@@ -85,6 +85,11 @@ class Builder:
         return self.lib.gx_pb_topn(self.pb, child, _arr(keys), _u8arr(desc),
                                    len(keys), -1, 0)
 
+    def mergejoin(self, build, probe, build_keys, probe_keys, join_type=0):
+        return self.lib.gx_pb_mergejoin(self.pb, build, probe,
+                                        _arr(build_keys), _arr(probe_keys),
+                                        len(build_keys), join_type)
+
     def hashjoin(self, build, probe, build_keys, probe_keys, join_type=0):
         return self.lib.gx_pb_hashjoin(self.pb, build, probe, _arr(build_keys),
                                        _arr(probe_keys), len(build_keys), join_type)
@@ -291,6 +296,65 @@ def q1_final_plan(lib):
 #   → TopN(revenue desc, o_orderdate asc, limit 10)
 O_ORDERKEY, O_CUSTKEY, O_ORDERDATE, O_SHIPPRIORITY = range(4)
 C_CUSTKEY, C_MKTSEGMENT = range(2)
+
+
+def q3_merge_plan(lib, limit=10):
+    """Q3 with merge joins (join/merge_join.go): each join's inputs sorted on
+    its join key by an explicit full Sort, as the planner would arrange.
+    Results are identical to q3_plan's hash joins."""
+    from tests.gxlib import GX_F_GT, GX_F_EQ
+    b = Builder(lib)
+    cust = b.source(CUSTOMER_TYPES)
+    seg = b.colref(C_MKTSEGMENT, GX_TYPE_STRING)
+    cond_c = b.call(GX_F_EQ, GX_TYPE_I64, 0, seg,
+                    lib.gx_pb_const_str(b.pb, b"BUILDING", 8))
+    sel_c = b.selection(cust, [cond_c])
+    sort_c = b.sort(sel_c, [b.colref(C_CUSTKEY, GX_TYPE_I64)], [0])
+
+    orders = b.source(ORDERS_TYPES)
+    odate = b.colref(O_ORDERDATE, GX_TYPE_TIME)
+    cond_o = b.call(GX_F_LT, GX_TYPE_I64, 0, odate,
+                    b.const_time(lib.gx_time_from_date(1995, 3, 15)))
+    sel_o = b.selection(orders, [cond_o])
+    sort_o = b.sort(sel_o, [b.colref(O_CUSTKEY, GX_TYPE_I64)], [0])
+
+    j1 = b.mergejoin(sort_c, sort_o,
+                     [b.colref(C_CUSTKEY, GX_TYPE_I64)],
+                     [b.colref(O_CUSTKEY, GX_TYPE_I64)])
+    sort_j1 = b.sort(j1, [b.colref(2, GX_TYPE_I64)], [0])  # by o_orderkey
+
+    li = b.source(LINEITEM_TYPES, LINEITEM_FRACS)
+    sdate = b.colref(L_SHIPDATE, GX_TYPE_TIME)
+    cond_l = b.call(GX_F_GT, GX_TYPE_I64, 0, sdate,
+                    b.const_time(lib.gx_time_from_date(1995, 3, 15)))
+    sel_l = b.selection(li, [cond_l])
+    sort_l = b.sort(sel_l, [b.colref(L_ORDERKEY, GX_TYPE_I64)], [0])
+
+    j2 = b.mergejoin(sort_j1, sort_l,
+                     [b.colref(2, GX_TYPE_I64)],
+                     [b.colref(L_ORDERKEY, GX_TYPE_I64)])
+
+    jo_orderkey = b.colref(6 + L_ORDERKEY, GX_TYPE_I64)
+    jo_odate = b.colref(4, GX_TYPE_TIME)
+    jo_prio = b.colref(5, GX_TYPE_I64)
+    price = b.colref(6 + L_EXTPRICE, GX_TYPE_DECIMAL, 2)
+    disc = b.colref(6 + L_DISCOUNT, GX_TYPE_DECIMAL, 2)
+    one = _const_dec_one(lib, b)
+    om_d = b.call(GX_F_MINUS, GX_TYPE_DECIMAL, 2, one, disc)
+    rev = b.call(GX_F_MUL, GX_TYPE_DECIMAL, 4, price, om_d)
+    proj = b.projection(j2, [jo_orderkey, jo_odate, jo_prio, rev])
+
+    from tests.gxlib import GX_AGG_SUM
+    agg = b.hashagg(proj,
+                    [b.colref(0, GX_TYPE_I64), b.colref(1, GX_TYPE_TIME),
+                     b.colref(2, GX_TYPE_I64)],
+                    [(GX_AGG_SUM, b.colref(3, GX_TYPE_DECIMAL, 4), 4)])
+    topn = b.topn(agg, [b.colref(3, GX_TYPE_DECIMAL, 4),
+                        b.colref(1, GX_TYPE_TIME)],
+                  [1, 0], limit)
+    out_types = [GX_TYPE_I64, GX_TYPE_TIME, GX_TYPE_I64, GX_TYPE_DECIMAL]
+    out_fracs = [0, 0, 0, 4]
+    return b, (cust, orders, li), topn, out_types, out_fracs
 
 
 def q3_plan(lib, limit=10):

@@ -180,6 +180,9 @@ struct FusedQueryDesc {
   int32_t accReg[kMaxAggs];   // phys acc slot -> VM register
   int32_t accMap[kMaxAggs];   // agg index -> phys acc slot (-1 for COUNT)
   int32_t sharedCnt = 0;
+  int32_t hasDiv = 0;  // launch the DIVOK kernel variant (division code is
+                       // compiled out of the common kernels: its register
+                       // demand alone costs a wave/SIMD of occupancy)
   GroupKeyDesc gkey;
   // outputs
   GroupSlot* globalTable = nullptr;  // kGlobalGroups slots

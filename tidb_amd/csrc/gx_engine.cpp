@@ -1609,6 +1609,7 @@ static void applyPostSort(gx_exec* ex) {
 static int32_t runFused(gx_exec* ex) {
   if (ex->vmHasDiv && !getenv("GX_DIV_NARROW"))
     ex->desc.wide = 1;  // DIV quotients rarely fit int64
+  ex->desc.hasDiv = ex->vmHasDiv ? 1 : 0;
   if (getenv("GX_FORCE_WIDE")) ex->desc.wide = 1;
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;

@@ -644,7 +644,7 @@ __device__ inline unsigned __int128 u128DivBig(unsigned __int128 n,
 
 // per-row pipeline after the raw fetch: filter -> VM -> LDS aggregate.
 // Returns false on a hard failure (error flag already set).
-template <bool WIDE, typename RAWT>
+template <bool WIDE, bool DIVOK, typename RAWT>
 __device__ __attribute__((always_inline)) inline bool processRow(const FusedQueryDesc& d, int64_t row,
                                   const RAWT& raw, Lds3GroupSlot* lds,
                                   uint64_t* mySel) {
@@ -749,6 +749,13 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         vm.setNull(ins.dst, vm.isNull(ins.a));
         break;
       case VM_DIV: {
+        if constexpr (!DIVOK) {
+          // engine launches the DIVOK variant for plans containing DIV;
+          // reaching here means a dispatch bug -- fail loudly
+          atomicOr(d.errorFlag, kErrBadDecimal);
+          bad = true;
+          break;
+        } else {
         // DecimalDiv (mydecimal.go:1311, doDiv:1168): quotient truncated
         // toward zero at the word-granular result scale; ins.c holds the
         // exponent e with result = trunc(a * 10^e / b). Division by zero
@@ -802,6 +809,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         vm.set(ins.dst, v);
         vm.setNull(ins.dst, nul);
         break;
+        }
       }
     }
   }
@@ -889,7 +897,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   return true;
 }
 
-template <bool WIDE, int R>
+template <bool WIDE, int R, bool DIVOK = false>
 __launch_bounds__(256)
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
@@ -930,10 +938,10 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     for (; row < end && !failed; row += 2 * stride) {
       const int64_t rB = row + stride;
       if (rB < end) fetchRow(d.table, d.fetch, d.nFetch, rB, rawB);
-      if (!processRow<WIDE>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
+      if (!processRow<WIDE, DIVOK>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
       const int64_t rA2 = row + 2 * stride;
       if (rA2 < end) fetchRow(d.table, d.fetch, d.nFetch, rA2, rawA);
-      if (rB < end && !processRow<WIDE>(d, rB, rawB, lds3, &mySel)) failed = true;
+      if (rB < end && !processRow<WIDE, DIVOK>(d, rB, rawB, lds3, &mySel)) failed = true;
     }
   }
 
@@ -1874,7 +1882,19 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
     for (int f = 0; f < desc.nFetch; f++)
       if (desc.fetch[f].kind != FETCH_B1 && f > maxSlot) maxSlot = f;
     bool small = maxSlot < 5;
-    if (desc.wide) {
+    if (desc.hasDiv) {
+      if (desc.wide) {
+        if (small)
+          hipLaunchKernelGGL((fusedAggKernel<true, 5, true>), dim3(grid), dim3(256), 0, s, devDesc);
+        else
+          hipLaunchKernelGGL((fusedAggKernel<true, 8, true>), dim3(grid), dim3(256), 0, s, devDesc);
+      } else {
+        if (small)
+          hipLaunchKernelGGL((fusedAggKernel<false, 5, true>), dim3(grid), dim3(256), 0, s, devDesc);
+        else
+          hipLaunchKernelGGL((fusedAggKernel<false, 8, true>), dim3(grid), dim3(256), 0, s, devDesc);
+      }
+    } else if (desc.wide) {
       if (small)
         hipLaunchKernelGGL((fusedAggKernel<true, 5>), dim3(grid), dim3(256), 0, s, devDesc);
       else

@@ -1047,8 +1047,9 @@ __device__ inline bool bloomMayHave(const JoinAggDesc& d, uint64_t key) {
   uint32_t mask = (1u << d.bloomLog2) - 1;
   uint32_t b1 = (uint32_t)h & mask;
   uint32_t b2 = (uint32_t)(h >> 32) & mask;
-  if (!((d.bloom[b1 >> 5] >> (b1 & 31)) & 1)) return false;
-  return ((d.bloom[b2 >> 5] >> (b2 & 31)) & 1) != 0;
+  auto bm = gptr<uint32_t>(d.bloom);  // AS1: keep probes off the flat path
+  if (!((bm[b1 >> 5] >> (b1 & 31)) & 1)) return false;
+  return ((bm[b2 >> 5] >> (b2 & 31)) & 1) != 0;
 }
 
 // count rows of build0 passing its predicate
@@ -1205,12 +1206,12 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
       if (colIsNull(c, row)) pass = false;
       else if (pd.kind == PRED_TIME_CMP_CONST) {
         uint64_t v = (pd.slot >= 0 ? raw.get(pd.slot).x
-                                   : ((const uint64_t*)c.data)[row]) & ~0xFULL;
+                                   : gptr<uint64_t>(c.data)[row]) & ~0xFULL;
         uint64_t k = pd.constU64 & ~0xFULL;
         pass = cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
       } else if (pd.kind == PRED_I64_CMP_CONST) {
         int64_t v = pd.slot >= 0 ? (int64_t)raw.get(pd.slot).x
-                                 : ((const int64_t*)c.data)[row];
+                                 : gptr<int64_t>(c.data)[row];
         int64_t k = (int64_t)pd.constU64;
         pass = cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
       } else {
@@ -1220,14 +1221,14 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
     if (!pass) continue;
     const DevCol& kc = d.probe.cols[d.pKeyCol];
     if (colIsNull(kc, row)) continue;
-    uint64_t key = ((const uint64_t*)kc.data)[row];
+    uint64_t key = gptr<uint64_t>(kc.data)[row];
     if (key == kEmptyKey) key = kEmptyKey - 1;
     if (!bloomMayHave(d, key)) continue;  // L2-resident reject
     // probe the slot table (prebuilt: no inserts, miss -> drop row)
     uint32_t slot = (uint32_t)(hashKey(key) & mask);
     bool found = false;
     for (uint32_t probe = 0; probe <= mask; probe++) {
-      uint64_t cur = d.slots[slot].key;
+      uint64_t cur = gptr<uint64_t>(&d.slots[slot].key)[0];
       if (cur == key) { found = true; break; }
       if (cur == kEmptyKey) break;
       slot = (slot + 1) & mask;

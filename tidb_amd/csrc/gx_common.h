@@ -284,6 +284,8 @@ int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows
 int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream);
 int gxLaunchMemset(void* p, int v, size_t n, void* stream);
+int gxFusedGrid(int64_t rows);
+int gxLaunchInitTable(GroupSlot* table, void* stream);
 int gxDumpDesc(const FusedQueryDesc* devDesc, void* stream);
 
 // ---- device full sort (sortexec/sort.go analog): LSD stable radix passes

@@ -26,6 +26,7 @@
 #include "../../include/gx_executor.h"
 #include "gx_common.h"
 #include "gx_decimal.h"
+#include "gx_jit.h"
 
 using gxp::MyDecimal;
 
@@ -121,6 +122,10 @@ struct gx_exec {
   int vmNextReg = 0;
   std::map<int, std::pair<int, int>> exprRegCache;  // exprId -> (reg, scale)
   bool vmHasDiv = false;  // DIV forces the wide VM and disables glds
+  // hipRTC-specialized kernel for this plan (gx_jit.cpp); nullptr -> use the
+  // interpreted fusedAggKernel
+  const gxjit::JitProg* jitProg = nullptr;
+  bool jitTried = false;
   // device state
   bool deviceReady = false;
   std::vector<void*> devBufs;
@@ -1623,6 +1628,23 @@ static int32_t runFused(gx_exec* ex) {
   HIP_OK(ex, hipEventCreate(&ev1));
   HIP_OK(ex, hipEventRecord(ev0, ex->stream));
   if (ex->desc.noLds) ex->desc.useGlds = 0;  // high-NDV retry uses the plain kernel
+  // runtime kernel specialization (gx_jit.cpp): try once per executor, any
+  // failure -> interpreted path. Disabled for the glds variant, timing
+  // ablations and general string group keys.
+  if (!ex->jitTried && !ex->desc.useGlds && ex->desc.ablate == 0 &&
+      !getenv("GX_NO_JIT")) {
+    ex->jitTried = true;
+    bool eligible = true;
+    for (int k = 0; k < ex->desc.gkey.nCols; k++)
+      eligible &= ex->desc.gkey.kind[k] != 0;
+    if (eligible) {
+      std::string why;
+      ex->jitProg = gxjit::compile(ex->desc, &why);
+      if (getenv("GX_DEBUG"))
+        fprintf(stderr, "[gx] jit %s%s\n", ex->jitProg ? "ok" : "DISABLED: ",
+                ex->jitProg ? "" : why.c_str());
+    }
+  }
   if (getenv("GX_DEBUG_DESC")) {
     fprintf(stderr, "[host] sizeof(desc)=%d aggs_off=%d ins_off=%d\n",
             (int)sizeof(gxp::FusedQueryDesc),
@@ -1630,7 +1652,15 @@ static int32_t runFused(gx_exec* ex) {
             (int)((char*)&ex->desc.ins[0] - (char*)&ex->desc));
     gxp::gxDumpDesc(ex->devDesc, ex->stream);
   }
-  int lrc = gxp::gxLaunchFusedAgg(ex->desc, ex->devDesc, ex->stream);
+  int lrc;
+  if (ex->jitProg) {
+    lrc = gxp::gxLaunchInitTable(ex->desc.globalTable, ex->stream);
+    if (lrc == 0)
+      lrc = gxjit::launch(ex->jitProg, ex->desc.wide != 0, ex->devDesc,
+                          gxp::gxFusedGrid(ex->desc.table.nRows), ex->stream);
+  } else {
+    lrc = gxp::gxLaunchFusedAgg(ex->desc, ex->devDesc, ex->stream);
+  }
   if (lrc != 0) {
     ex->err = "fused kernel launch failed: " +
               std::string(hipGetErrorString((hipError_t)lrc));
@@ -2841,6 +2871,15 @@ void gx_exec_free(gx_exec* ex) { delete ex; }
 const char* gx_last_error(gx_exec* ex) {
   if (!ex) return "null exec";
   return ex->err.c_str();
+}
+
+// debug: the hipRTC source the engine would generate for this plan
+// (primarily for offline syntax checking; not part of the stable ABI)
+const char* gx_debug_jit_source(gx_exec* ex) {
+  static std::string s;
+  if (!ex) return "";
+  s = gxjit::generateSource(ex->desc);
+  return s.c_str();
 }
 
 double gx_last_kernel_ms(gx_exec* ex) { return ex ? ex->lastKernelMs : 0; }

@@ -1,0 +1,15 @@
+// runtime kernel specialization via hipRTC (see gx_jit.cpp)
+#pragma once
+#include <string>
+#include "gx_common.h"
+
+namespace gxjit {
+struct JitProg;
+// compile (process-cached) a specialized fused-agg kernel for this plan;
+// nullptr (with *whyNot set) on any failure -- caller falls back to the
+// interpreted kernel
+const JitProg* compile(const gxp::FusedQueryDesc& d, std::string* whyNot);
+int launch(const JitProg* prog, bool wide, const gxp::FusedQueryDesc* devDesc,
+           int grid, void* stream);
+std::string generateSource(const gxp::FusedQueryDesc& d);
+}  // namespace gxjit

@@ -157,3 +157,13 @@ def test_q3_parity_larger(libs):
         assert q3.run_q3(product) == q3.run_q3(oracle)
     finally:
         q3.N_LI, q3.N_ORD, q3.N_CUST = old
+
+
+def test_q1_interpreted_fallback_parity(libs, monkeypatch):
+    """GX_NO_JIT=1 exercises the interpreted fused kernel (the fallback when
+    hipRTC is unavailable); results must match the oracle bit-for-bit too."""
+    oracle, product = libs
+    monkeypatch.setenv("GX_NO_JIT", "1")
+    got = _as_map(_run_q1(product, 65536))
+    monkeypatch.delenv("GX_NO_JIT")
+    assert got == _as_map(_run_q1(oracle, 65536))

@@ -126,6 +126,8 @@ struct gx_exec {
   // interpreted fusedAggKernel
   const gxjit::JitProg* jitProg = nullptr;
   bool jitTried = false;
+  const gxjit::JitProg* jaJitProg = nullptr;
+  bool jaJitTried = false;
   // device state
   bool deviceReady = false;
   std::vector<void*> devBufs;
@@ -1860,6 +1862,16 @@ static int32_t runJoinAgg(gx_exec* ex) {
     return GX_OK;
   };
   auto phase = [&](int ph) -> int32_t {
+    if (ph == 4 && ex->jaJitProg) {
+      int jrc = gxjit::launchJa(ex->jaJitProg, ja.wide != 0, ex->devJa,
+                                gxp::gxFusedGrid(ja.probe.nRows), ex->stream);
+      if (jrc != 0) {
+        ex->err = std::string("jit probe launch failed: ") +
+                  hipGetErrorString((hipError_t)jrc);
+        return GX_ERR_INTERNAL;
+      }
+      return GX_OK;
+    }
     int rc = gxp::gxJoinAggPhase(ph, ex->devJa, ja, ex->stream);
     if (rc != 0) {
       ex->err = std::string("join phase launch failed: ") +
@@ -1913,6 +1925,16 @@ static int32_t runJoinAgg(gx_exec* ex) {
   if ((rc = pushDesc())) return rc;
   if ((rc = phase(5))) return rc;  // init slots
   if ((rc = phase(3))) return rc;  // build
+
+  // specialize the probe kernel once per executor (gx_jit.cpp)
+  if (!ex->jaJitTried && !getenv("GX_NO_JIT")) {
+    ex->jaJitTried = true;
+    std::string why;
+    ex->jaJitProg = gxjit::compileJa(ja, &why);
+    if (getenv("GX_DEBUG"))
+      fprintf(stderr, "[gx] ja jit %s%s\n", ex->jaJitProg ? "ok" : "DISABLED: ",
+              ex->jaJitProg ? "" : why.c_str());
+  }
 
   // probe (timed — the dominant scan)
   hipEvent_t ev0, ev1;

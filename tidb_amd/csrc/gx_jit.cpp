@@ -429,8 +429,8 @@ extern "C" __global__ void __launch_bounds__(256) genq_wide(const FusedQueryDesc
 }
 
 // compile (cached by source text); returns nullptr on any failure
-const JitProg* compile(const FusedQueryDesc& d, std::string* whyNot) {
-  std::string src = generateSource(d);
+static const JitProg* compileSource(const std::string& src, const char* fn1,
+                                    const char* fn2, std::string* whyNot) {
   std::lock_guard<std::mutex> lk(cacheMu);
   auto it = cache().find(src);
   if (it != cache().end()) return it->second.ok ? &it->second : nullptr;
@@ -464,9 +464,8 @@ const JitProg* compile(const FusedQueryDesc& d, std::string* whyNot) {
   hiprtcGetCode(rp, &code[0]);
   hiprtcDestroyProgram(&rp);
   if (hipModuleLoadData(&prog.mod, code.data()) != hipSuccess ||
-      hipModuleGetFunction(&prog.fnNarrow, prog.mod, "genq_narrow") !=
-          hipSuccess ||
-      hipModuleGetFunction(&prog.fnWide, prog.mod, "genq_wide") != hipSuccess) {
+      hipModuleGetFunction(&prog.fnNarrow, prog.mod, fn1) != hipSuccess ||
+      hipModuleGetFunction(&prog.fnWide, prog.mod, fn2) != hipSuccess) {
     if (whyNot) *whyNot = "hipModule load failed";
     cache()[src] = prog;
     return nullptr;
@@ -474,6 +473,10 @@ const JitProg* compile(const FusedQueryDesc& d, std::string* whyNot) {
   prog.ok = true;
   auto& slot = cache()[src] = prog;
   return &slot;
+}
+
+const JitProg* compile(const FusedQueryDesc& d, std::string* whyNot) {
+  return compileSource(generateSource(d), "genq_narrow", "genq_wide", whyNot);
 }
 
 int launch(const JitProg* prog, bool wide, const FusedQueryDesc* devDesc,
@@ -484,4 +487,249 @@ int launch(const JitProg* prog, bool wide, const FusedQueryDesc* devDesc,
                                     (hipStream_t)stream, args, nullptr);
 }
 
+
+// ---------------------------------------------------------------------
+// join-aggregate probe specialization (jaProbeKernel): same idea as the
+// fused-agg specialization -- literal fetch/pred/VM/accumulator plans,
+// runtime table sizes (slot mask, bloom size) kept as scalar desc reads so
+// one compiled kernel serves every step of the executor.
+// ---------------------------------------------------------------------
+using gxp::JoinAggDesc;
+
+static void emitJaVm(std::ostringstream& s, const JoinAggDesc& d) {
+  for (int i = 0; i < d.nIns; i++) {
+    const gxp::VmIns& ins = d.ins[i];
+    std::string v = "v" + std::to_string(ins.dst);
+    std::string nv = "n" + std::to_string(ins.dst);
+    switch (ins.op) {
+      case gxp::VM_LOAD_DEC: {
+        const gxp::DevCol& c = d.probe.cols[ins.a];
+        s << "    T " << v << " = VT<WIDE>::zero(); bool " << nv
+          << " = false;\n";
+        if (c.hasNulls)
+          s << "    " << nv << " = colIsNull(d.probe.cols[" << ins.a
+            << "], row);\n";
+        s << "    if (!" << nv << ") { int sc;\n";
+        if (ins.c >= 0)
+          s << "      if (!parseDecimalRaw<WIDE>(raw.get(" << ins.c << "), &"
+            << v << ", &sc, d.errorFlag, " << ins.b << ", " << d.insP10[i]
+            << "LL, " << d.insMagic[i] << "ULL)) return false;\n";
+        else
+          s << "      if (!loadDecimalUnits<WIDE>((const uint8_t*)d.probe.cols["
+            << ins.a << "].data + row * 40, &" << v << ", &sc, d.errorFlag, "
+            << ins.b << ", " << d.insP10[i] << "LL, " << d.insMagic[i]
+            << "ULL)) return false;\n";
+        s << "      if (sc != " << ins.b << ") {\n"
+          << "        if (sc < " << ins.b << ") { bool o2 = false; " << v
+          << " = VT<WIDE>::scale10(" << v << ", " << ins.b
+          << " - sc, &o2); if (o2) { atomicOr(d.errorFlag, WIDE ? kErrOverflow"
+             " : kErrRetryWide); return false; } }\n"
+          << "        else { atomicOr(d.errorFlag, kErrScale); return false; }"
+             "\n      }\n    }\n";
+        break;
+      }
+      case gxp::VM_LOAD_I64: {
+        const gxp::DevCol& c = d.probe.cols[ins.a];
+        s << "    bool " << nv << " = false;\n";
+        if (c.hasNulls)
+          s << "    " << nv << " = colIsNull(d.probe.cols[" << ins.a
+            << "], row);\n";
+        if (ins.c >= 0)
+          s << "    T " << v << " = " << nv
+            << " ? VT<WIDE>::zero() : VT<WIDE>::fromI64((int64_t)raw.get("
+            << ins.c << ").x, &ovf);\n";
+        else
+          s << "    T " << v << " = " << nv
+            << " ? VT<WIDE>::zero() : VT<WIDE>::fromI64(gptr<int64_t>(d.probe."
+               "cols[" << ins.a << "].data)[row], &ovf);\n";
+        break;
+      }
+      case gxp::VM_LOAD_CONST:
+        s << "    T " << v << ";\n"
+          << "    { Int128 cv = {" << (uint64_t)d.constLo[ins.a]
+          << "ULL, (int64_t)" << d.constHi[ins.a] << "LL };\n"
+          << "      if (WIDE) " << v << " = *(T*)&cv;\n"
+          << "      else { int64_t c64 = " << d.constLo[ins.a] << "LL; " << v
+          << " = *(T*)&c64; } }\n"
+          << "    const bool " << nv << " = false;\n";
+        break;
+      case gxp::VM_ADD:
+        s << "    T " << v << " = VT<WIDE>::add(v" << ins.a << ", v" << ins.b
+          << ", &ovf); bool " << nv << " = n" << ins.a << " || n" << ins.b
+          << ";\n";
+        break;
+      case gxp::VM_SUB:
+        s << "    T " << v << " = VT<WIDE>::sub(v" << ins.a << ", v" << ins.b
+          << ", &ovf); bool " << nv << " = n" << ins.a << " || n" << ins.b
+          << ";\n";
+        break;
+      case gxp::VM_MUL:
+        s << "    bool " << nv << " = n" << ins.a << " || n" << ins.b << ";\n"
+          << "    T " << v << " = VT<WIDE>::zero();\n"
+          << "    if (!" << nv << ") " << v << " = VT<WIDE>::mul(v" << ins.a
+          << ", v" << ins.b << ", &ovf);\n";
+        break;
+      case gxp::VM_SCALE_UP:
+        s << "    T " << v << " = VT<WIDE>::mul(v" << ins.a
+          << ", VT<WIDE>::fromI64(" << d.insP10[i]
+          << "LL, nullptr), &ovf); bool " << nv << " = n" << ins.a << ";\n";
+        break;
+      default:
+        s << "    #error unsupported op\n";
+        break;
+    }
+  }
+}
+
+std::string generateJaSource(const JoinAggDesc& d) {
+  std::ostringstream s;
+  s << "#include \"gx_common.h\"\n#include \"gx_device.h\"\n"
+       "using namespace gxp;\n\n";
+  // literal probe-side fetch
+  s << "__device__ __forceinline__ void fetchJa(const DevTable& t, int64_t "
+       "row, RawState& r) {\n";
+  for (int f = 0; f < d.nFetch; f++) {
+    const gxp::FetchDesc& fd = d.fetch[f];
+    if (fd.kind == gxp::FETCH_B1) continue;
+    std::string m = "r.s" + std::to_string(f);
+    if (fd.kind == gxp::FETCH_8B)
+      s << "  " << m << ".x = gptr<uint64_t>(t.cols[" << fd.col
+        << "].data)[row]; " << m << ".y = 0;\n";
+    else if (fd.kind == gxp::FETCH_DEC16)
+      s << "  { const uint8_t* p = (const uint8_t*)t.cols[" << fd.col
+        << "].data + row * 40;\n    " << m << ".x = *gptr<uint64_t>(p); "
+        << m << ".y = *gptr<uint64_t>(p + 8); }\n";
+  }
+  s << "}\n\n";
+  s << R"RTC(
+__device__ __forceinline__ bool jaBloomMayHave(const JoinAggDesc& d, uint64_t key) {
+  if (d.bloomLog2 == 0) return true;
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloomLog2) - 1;
+  uint32_t b1 = (uint32_t)h & mask;
+  uint32_t b2 = (uint32_t)(h >> 32) & mask;
+  auto bm = gptr<uint32_t>(d.bloom);
+  if (!((bm[b1 >> 5] >> (b1 & 31)) & 1)) return false;
+  return ((bm[b2 >> 5] >> (b2 & 31)) & 1) != 0;
+}
+
+template <bool WIDE>
+__device__ __forceinline__ bool jaRow(const JoinAggDesc& d, int64_t row,
+                                      uint32_t mask, uint64_t* myMatch) {
+  using T = typename VT<WIDE>::T;
+  bool ovf = false; (void)ovf;
+  RawState raw;
+  fetchJa(d.probe, row, raw);
+)RTC";
+  // literal predicate
+  if (d.nPredP > 0) {
+    const gxp::PredDesc& pd = d.predP;
+    const gxp::DevCol& c = d.probe.cols[pd.col];
+    if (c.hasNulls)
+      s << "  if (colIsNull(d.probe.cols[" << pd.col
+        << "], row)) return true;\n";
+    if (pd.kind == gxp::PRED_TIME_CMP_CONST) {
+      s << "  { uint64_t pv = "
+        << (pd.slot >= 0
+                ? ("raw.get(" + std::to_string(pd.slot) + ").x")
+                : ("gptr<uint64_t>(d.probe.cols[" + std::to_string(pd.col) +
+                   "].data)[row]"))
+        << " & ~0xFULL;\n    if (!(pv " << cmpOp(pd.cmp) << " "
+        << (pd.constU64 & ~0xFULL) << "ULL)) return true; }\n";
+    } else if (pd.kind == gxp::PRED_I64_CMP_CONST) {
+      s << "  { int64_t pv = "
+        << (pd.slot >= 0
+                ? ("(int64_t)raw.get(" + std::to_string(pd.slot) + ").x")
+                : ("gptr<int64_t>(d.probe.cols[" + std::to_string(pd.col) +
+                   "].data)[row]"))
+        << ";\n    if (!(pv " << cmpOp(pd.cmp) << " (int64_t)"
+        << (int64_t)pd.constU64 << "LL)) return true; }\n";
+    }
+  }
+  // key + bloom + slot probe
+  const gxp::DevCol& kc = d.probe.cols[d.pKeyCol];
+  if (kc.hasNulls)
+    s << "  if (colIsNull(d.probe.cols[" << d.pKeyCol
+      << "], row)) return true;\n";
+  s << "  uint64_t key = gptr<uint64_t>(d.probe.cols[" << d.pKeyCol
+    << "].data)[row];\n"
+       "  if (key == kEmptyKey) key = kEmptyKey - 1;\n"
+       "  if (!jaBloomMayHave(d, key)) return true;\n"
+       "  uint32_t slot = (uint32_t)(splitmix64(key) & mask);\n"
+       "  bool found = false;\n"
+       "  for (uint32_t probe = 0; probe <= mask; probe++) {\n"
+       "    uint64_t cur = gptr<uint64_t>(&d.slots[slot].key)[0];\n"
+       "    if (cur == key) { found = true; break; }\n"
+       "    if (cur == kEmptyKey) break;\n"
+       "    slot = (slot + 1) & mask;\n"
+       "  }\n"
+       "  if (!found) return true;\n"
+       "  {\n";
+  emitJaVm(s, d);
+  s << "    if (ovf) { atomicOr(d.errorFlag, WIDE ? kErrOverflow : "
+       "kErrRetryWide); return false; }\n";
+  s << "    if (n" << d.valueReg << ") return true;\n"
+    << "    Int128 vv = VT<WIDE>::toAcc(v" << d.valueReg << ");\n";
+  s << R"RTC(
+    JoinAggSlot* sp = &d.slots[slot];
+    uint64_t old = atomicAdd((unsigned long long*)&sp->accLo, (unsigned long long)vv.lo);
+    uint64_t carry = (old + vv.lo) < old ? 1 : 0;
+    int64_t hiAdd = vv.hi + (int64_t)carry;
+    if (hiAdd != 0)
+      atomicAdd((unsigned long long*)&sp->accHi, (unsigned long long)hiAdd);
+    if (vv.hi < 0 || (vv.hi == 0 && vv.lo == 0))
+      atomicAdd((unsigned long long*)&sp->cnt, 1ULL);
+    (*myMatch)++;
+  }
+  return true;
+}
+
+template <bool WIDE>
+__device__ __forceinline__ void jaBody(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.probe.nRows;
+  uint32_t mask = (1u << d.slotsLog2) - 1;
+  int64_t per = (n + gridDim.x - 1) / gridDim.x;
+  int64_t begin = (int64_t)blockIdx.x * per;
+  int64_t end = begin + per;
+  if (end > n) end = n;
+  bool failed = false;
+  uint64_t myMatch = 0;
+  for (int64_t row = begin + threadIdx.x; row < end && !failed;
+       row += blockDim.x) {
+    if (!jaRow<WIDE>(d, row, mask, &myMatch)) failed = true;
+  }
+  for (int off = 32; off > 0; off >>= 1) myMatch += __shfl_down(myMatch, off, 64);
+  if ((threadIdx.x & 63) == 0 && myMatch)
+    atomicAdd((unsigned long long*)&d.counters[2], (unsigned long long)myMatch);
+}
+
+extern "C" __global__ void __launch_bounds__(256) genja_narrow(const JoinAggDesc* __restrict__ dp) {
+  jaBody<false>(dp);
+}
+extern "C" __global__ void __launch_bounds__(256) genja_wide(const JoinAggDesc* __restrict__ dp) {
+  jaBody<true>(dp);
+}
+)RTC";
+  return s.str();
+}
+
+const JitProg* compileJa(const JoinAggDesc& d, std::string* whyNot) {
+  if (d.nPredP > 0 && d.predP.kind != gxp::PRED_TIME_CMP_CONST &&
+      d.predP.kind != gxp::PRED_I64_CMP_CONST) {
+    if (whyNot) *whyNot = "probe predicate kind not specialized";
+    return nullptr;
+  }
+  std::string src = generateJaSource(d);
+  return compileSource(src, "genja_narrow", "genja_wide", whyNot);
+}
+
+
+int launchJa(const JitProg* prog, bool wide, const JoinAggDesc* devDesc,
+             int grid, void* stream) {
+  void* args[] = {(void*)&devDesc};
+  return (int)hipModuleLaunchKernel(wide ? prog->fnWide : prog->fnNarrow,
+                                    grid, 1, 1, 256, 1, 1, 0,
+                                    (hipStream_t)stream, args, nullptr);
+}
 }  // namespace gxjit

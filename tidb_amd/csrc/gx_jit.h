@@ -12,4 +12,8 @@ const JitProg* compile(const gxp::FusedQueryDesc& d, std::string* whyNot);
 int launch(const JitProg* prog, bool wide, const gxp::FusedQueryDesc* devDesc,
            int grid, void* stream);
 std::string generateSource(const gxp::FusedQueryDesc& d);
+// join-aggregate probe kernel specialization
+const JitProg* compileJa(const gxp::JoinAggDesc& d, std::string* whyNot);
+int launchJa(const JitProg* prog, bool wide, const gxp::JoinAggDesc* devDesc,
+             int grid, void* stream);
 }  // namespace gxjit

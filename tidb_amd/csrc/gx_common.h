@@ -258,6 +258,11 @@ struct JoinAggDesc {
   int32_t keySetLog2 = 0;
   JoinAggSlot* slots = nullptr;
   int32_t slotsLog2 = 0;
+  // Bloom filter over the build0 key set (customer keys), probed by every
+  // build1 row before the 64 MB key-set random access -- ~20% hit rate on
+  // Q3 makes this the build phases' main traffic cut
+  uint32_t* bloom0 = nullptr;
+  int32_t bloom0Log2 = 0;
   // Bloom filter over the build keys (2 hashes; sized ~8 bits/key so it stays
   // L2-resident): rejects the ~90% non-matching probes without touching the
   // HBM-random slot table

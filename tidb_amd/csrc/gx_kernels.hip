@@ -497,20 +497,20 @@ __device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
   const DevCol& c = tab.cols[pd.col];
   if (colIsNull(c, row)) return false;
   if (pd.kind == PRED_TIME_CMP_CONST) {
-    uint64_t v = ((const uint64_t*)c.data)[row] & ~0xFULL;
+    uint64_t v = gptr<uint64_t>(c.data)[row] & ~0xFULL;
     uint64_t k = pd.constU64 & ~0xFULL;
     return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
   }
   if (pd.kind == PRED_I64_CMP_CONST) {
-    int64_t v = ((const int64_t*)c.data)[row];
+    int64_t v = gptr<int64_t>(c.data)[row];
     int64_t k = (int64_t)pd.constU64;
     return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
   }
   if (pd.kind == PRED_STR_EQ_CONST) {
     int64_t st, en;
     if (c.denseOffsets) { st = row; en = row + 1; }
-    else { st = c.offsets[row]; en = c.offsets[row + 1]; }
-    const uint8_t* p = (const uint8_t*)c.data;
+    else { st = gptr<int64_t>(c.offsets)[row]; en = gptr<int64_t>(c.offsets)[row + 1]; }
+    auto p = gptr<uint8_t>(c.data);
     while (en > st && p[en - 1] == ' ') en--;  // PAD SPACE
     int len = (int)(en - st);
     bool eq = len == strConstLen;
@@ -529,6 +529,26 @@ __device__ inline void bloomSet(const JoinAggDesc& d, uint64_t key) {
   uint32_t b2 = (uint32_t)(h >> 32) & mask;
   atomicOr(&d.bloom[b1 >> 5], 1u << (b1 & 31));
   atomicOr(&d.bloom[b2 >> 5], 1u << (b2 & 31));
+}
+
+__device__ inline void bloom0Set(const JoinAggDesc& d, uint64_t key) {
+  if (d.bloom0Log2 == 0) return;
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloom0Log2) - 1;
+  atomicOr(&d.bloom0[((uint32_t)h & mask) >> 5], 1u << ((uint32_t)h & 31));
+  atomicOr(&d.bloom0[((uint32_t)(h >> 32) & mask) >> 5],
+           1u << ((uint32_t)(h >> 32) & 31));
+}
+
+__device__ inline bool bloom0MayHave(const JoinAggDesc& d, uint64_t key) {
+  if (d.bloom0Log2 == 0) return true;
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloom0Log2) - 1;
+  uint32_t b1 = (uint32_t)h & mask;
+  uint32_t b2 = (uint32_t)(h >> 32) & mask;
+  auto bm = gptr<uint32_t>(d.bloom0);
+  if (!((bm[b1 >> 5] >> (b1 & 31)) & 1)) return false;
+  return ((bm[b2 >> 5] >> (b2 & 31)) & 1) != 0;
 }
 
 __device__ inline bool bloomMayHave(const JoinAggDesc& d, uint64_t key) {
@@ -569,8 +589,9 @@ __global__ void jaBuild0Kernel(const JoinAggDesc* __restrict__ dp) {
                 evalSimplePred(d.build0, d.pred0, d.strConst, d.strConstLen, row);
     if (!pass) continue;
     if (colIsNull(d.build0.cols[d.b0KeyCol], row)) continue;  // NULL never joins
-    uint64_t key = ((const uint64_t*)d.build0.cols[d.b0KeyCol].data)[row];
+    uint64_t key = gptr<uint64_t>(d.build0.cols[d.b0KeyCol].data)[row];
     if (key == kEmptyKey) key = kEmptyKey - 1;
+    bloom0Set(d, key);
     uint32_t slot = (uint32_t)(hashKey(key) & mask);
     for (uint32_t probe = 0; probe <= mask; probe++) {
       uint64_t cur = d.keySet[slot];
@@ -592,7 +613,7 @@ __device__ inline bool keySetHas(const JoinAggDesc& d, uint64_t key) {
   if (key == kEmptyKey) key = kEmptyKey - 1;
   uint32_t slot = (uint32_t)(hashKey(key) & mask);
   for (uint32_t probe = 0; probe <= mask; probe++) {
-    uint64_t cur = d.keySet[slot];
+    uint64_t cur = gptr<uint64_t>(d.keySet)[slot];
     if (cur == key) return true;
     if (cur == kEmptyKey) return false;
     slot = (slot + 1) & mask;
@@ -612,7 +633,9 @@ __global__ void jaCountBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
     if (!pass) continue;
     const DevCol& kc = d.build1.cols[d.b1ProbeCol];
     if (colIsNull(kc, row)) continue;
-    if (!keySetHas(d, ((const uint64_t*)kc.data)[row])) continue;
+    uint64_t pkey0 = gptr<uint64_t>(kc.data)[row];
+    if (!bloom0MayHave(d, pkey0)) continue;  // L2-resident reject
+    if (!keySetHas(d, pkey0)) continue;
     my++;
   }
   for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
@@ -632,15 +655,17 @@ __global__ void jaBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
     if (!pass) continue;
     const DevCol& pc = d.build1.cols[d.b1ProbeCol];
     if (colIsNull(pc, row)) continue;
-    if (!keySetHas(d, ((const uint64_t*)pc.data)[row])) continue;
+    uint64_t pkey0 = gptr<uint64_t>(pc.data)[row];
+    if (!bloom0MayHave(d, pkey0)) continue;  // L2-resident reject
+    if (!keySetHas(d, pkey0)) continue;
     const DevCol& kc = d.build1.cols[d.b1KeyCol];
     if (colIsNull(kc, row)) continue;
-    uint64_t key = ((const uint64_t*)kc.data)[row];
+    uint64_t key = gptr<uint64_t>(kc.data)[row];
     if (key == kEmptyKey) key = kEmptyKey - 1;
     uint64_t pay0 = d.payloadCol0 >= 0
-        ? ((const uint64_t*)d.build1.cols[d.payloadCol0].data)[row] : 0;
+        ? gptr<uint64_t>(d.build1.cols[d.payloadCol0].data)[row] : 0;
     int64_t pay1 = d.payloadCol1 >= 0
-        ? ((const int64_t*)d.build1.cols[d.payloadCol1].data)[row] : 0;
+        ? gptr<int64_t>(d.build1.cols[d.payloadCol1].data)[row] : 0;
     bloomSet(d, key);
     uint32_t slot = (uint32_t)(hashKey(key) & mask);
     for (uint32_t probe = 0; probe <= mask; probe++) {

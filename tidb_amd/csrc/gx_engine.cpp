@@ -128,6 +128,8 @@ struct gx_exec {
   bool jitTried = false;
   const gxjit::JitProg* jaJitProg = nullptr;
   bool jaJitTried = false;
+  const gxjit::JaBuildProg* jaBuildProg = nullptr;
+  bool jaBuildTried = false;
   // device state
   bool deviceReady = false;
   std::vector<void*> devBufs;
@@ -1862,6 +1864,17 @@ static int32_t runJoinAgg(gx_exec* ex) {
     return GX_OK;
   };
   auto phase = [&](int ph) -> int32_t {
+    if (ph <= 3 && ex->jaBuildProg) {
+      int64_t rows = (ph <= 1 ? ja.build0 : ja.build1).nRows;
+      int jrc = gxjit::launchJaBuild(ex->jaBuildProg, ph, ex->devJa,
+                                     gxp::gxFusedGrid(rows), ex->stream);
+      if (jrc != 0) {
+        ex->err = std::string("jit build launch failed: ") +
+                  hipGetErrorString((hipError_t)jrc);
+        return GX_ERR_INTERNAL;
+      }
+      return GX_OK;
+    }
     if (ph == 4 && ex->jaJitProg) {
       int jrc = gxjit::launchJa(ex->jaJitProg, ja.wide != 0, ex->devJa,
                                 gxp::gxFusedGrid(ja.probe.nRows), ex->stream);
@@ -1885,6 +1898,17 @@ static int32_t runJoinAgg(gx_exec* ex) {
     HIP_OK(ex, hipMemcpy(out, ja.counters + i, 8, hipMemcpyDeviceToHost));
     return GX_OK;
   };
+
+  // specialize the build kernels once per executor (gx_jit.cpp)
+  if (!ex->jaBuildTried && !getenv("GX_NO_JIT")) {
+    ex->jaBuildTried = true;
+    std::string why;
+    ex->jaBuildProg = gxjit::compileJaBuild(ja, &why);
+    if (getenv("GX_DEBUG"))
+      fprintf(stderr, "[gx] ja build jit %s%s\n",
+              ex->jaBuildProg ? "ok" : "DISABLED: ",
+              ex->jaBuildProg ? "" : why.c_str());
+  }
 
   // build0: count, size the key set, fill
   int32_t rc = pushDesc();

@@ -732,4 +732,304 @@ int launchJa(const JitProg* prog, bool wide, const JoinAggDesc* devDesc,
                                     grid, 1, 1, 256, 1, 1, 0,
                                     (hipStream_t)stream, args, nullptr);
 }
+
+// build-phase specialization: four straight-line kernels (count0, build0,
+// count1, build1). count1/build1 scan the big middle table (orders at Q3):
+// their per-row chain is two dependent 8B loads + a key-set probe, so the
+// emitted loop software-pipelines the predicate-column load one stride
+// ahead (the interpreted versions measured latency-bound at ~5x their
+// sequential-traffic floor).
+static void emitPredInline(std::ostringstream& s, const JoinAggDesc& d,
+                           const gxp::PredDesc& pd, const char* tbl,
+                           const char* rowVar, const char* preVar) {
+  const gxp::DevCol& c =
+      (std::string(tbl) == "d.build0" ? d.build0 : d.build1).cols[pd.col];
+  if (c.hasNulls)
+    s << "    if (colIsNull(" << tbl << ".cols[" << pd.col << "], " << rowVar
+      << ")) continue;\n";
+  if (pd.kind == gxp::PRED_TIME_CMP_CONST) {
+    s << "    { uint64_t pv = " << (preVar ? preVar : "0");
+    if (!preVar)
+      s << "; pv = gptr<uint64_t>(" << tbl << ".cols[" << pd.col << "].data)["
+        << rowVar << "]";
+    s << "; pv &= ~0xFULL;\n      if (!(pv " << cmpOp(pd.cmp) << " "
+      << (pd.constU64 & ~0xFULL) << "ULL)) continue; }\n";
+  } else if (pd.kind == gxp::PRED_I64_CMP_CONST) {
+    s << "    { int64_t pv = (int64_t)" << (preVar ? preVar : "0");
+    if (!preVar)
+      s << "; pv = gptr<int64_t>(" << tbl << ".cols[" << pd.col << "].data)["
+        << rowVar << "]";
+    s << ";\n      if (!(pv " << cmpOp(pd.cmp) << " (int64_t)"
+      << (int64_t)pd.constU64 << "LL)) continue; }\n";
+  } else {  // PRED_STR_EQ_CONST: literal bytes
+    s << "    { int64_t st, en;\n";
+    if (c.denseOffsets)
+      s << "      st = " << rowVar << "; en = " << rowVar << " + 1;\n";
+    else
+      s << "      st = gptr<int64_t>(" << tbl << ".cols[" << pd.col
+        << "].offsets)[" << rowVar << "]; en = gptr<int64_t>(" << tbl
+        << ".cols[" << pd.col << "].offsets)[" << rowVar << " + 1];\n";
+    s << "      auto p = gptr<uint8_t>(" << tbl << ".cols[" << pd.col
+      << "].data);\n"
+         "      while (en > st && p[en - 1] == ' ') en--;\n"
+         "      if (en - st != " << d.strConstLen << ") continue;\n"
+         "      bool eq = true;\n";
+    for (int b = 0; b < d.strConstLen; b++)
+      s << "      eq = eq && p[st + " << b << "] == " << (int)d.strConst[b]
+        << ";\n";
+    s << "      if (" << (pd.cmp == 4 ? "!eq" : "eq") << ") continue; }\n";
+  }
+}
+
+std::string generateJaBuildSource(const JoinAggDesc& d) {
+  std::ostringstream s;
+  s << "#include \"gx_common.h\"\n#include \"gx_device.h\"\nusing namespace "
+       "gxp;\n\n";
+  s << R"RTC(
+__device__ __forceinline__ bool jaKeySetHas(const JoinAggDesc& d, uint64_t key) {
+  uint32_t mask = (1u << d.keySetLog2) - 1;
+  if (key == kEmptyKey) key = kEmptyKey - 1;
+  uint32_t slot = (uint32_t)(splitmix64(key) & mask);
+  for (uint32_t probe = 0; probe <= mask; probe++) {
+    uint64_t cur = gptr<uint64_t>(d.keySet)[slot];
+    if (cur == key) return true;
+    if (cur == kEmptyKey) return false;
+    slot = (slot + 1) & mask;
+  }
+  return false;
+}
+__device__ __forceinline__ bool jaBloom0MayHave(const JoinAggDesc& d, uint64_t key) {
+  if (d.bloom0Log2 == 0) return true;
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloom0Log2) - 1;
+  uint32_t b1 = (uint32_t)h & mask;
+  uint32_t b2 = (uint32_t)(h >> 32) & mask;
+  auto bm = gptr<uint32_t>(d.bloom0);
+  if (!((bm[b1 >> 5] >> (b1 & 31)) & 1)) return false;
+  return ((bm[b2 >> 5] >> (b2 & 31)) & 1) != 0;
+}
+__device__ __forceinline__ void jaBloom0Set(const JoinAggDesc& d, uint64_t key) {
+  if (d.bloom0Log2 == 0) return;
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloom0Log2) - 1;
+  atomicOr(&d.bloom0[((uint32_t)h & mask) >> 5], 1u << ((uint32_t)h & 31));
+  atomicOr(&d.bloom0[((uint32_t)(h >> 32) & mask) >> 5],
+           1u << ((uint32_t)(h >> 32) & 31));
+}
+__device__ __forceinline__ void jaBloomSet(const JoinAggDesc& d, uint64_t key) {
+  if (d.bloomLog2 == 0) return;
+  uint64_t h = splitmix64(key);
+  uint32_t mask = (1u << d.bloomLog2) - 1;
+  atomicOr(&d.bloom[((uint32_t)h & mask) >> 5], 1u << ((uint32_t)h & 31));
+  atomicOr(&d.bloom[((uint32_t)(h >> 32) & mask) >> 5],
+           1u << ((uint32_t)(h >> 32) & 31));
+}
+)RTC";
+  // ---- count0 / build0 over build0 (customer) ----
+  s << R"RTC(
+extern "C" __global__ void __launch_bounds__(256) genja_count0(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.build0.nRows;
+  uint64_t my = 0;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+)RTC";
+  if (d.nPred0 > 0) emitPredInline(s, d, d.pred0, "d.build0", "row", nullptr);
+  if (d.build0.cols[d.b0KeyCol].hasNulls)
+    s << "    if (colIsNull(d.build0.cols[" << d.b0KeyCol
+      << "], row)) continue;\n";
+  s << R"RTC(    my++;
+  }
+  for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
+  if ((threadIdx.x & 63) == 0 && my)
+    atomicAdd((unsigned long long*)&d.counters[0], (unsigned long long)my);
+}
+
+extern "C" __global__ void __launch_bounds__(256) genja_build0(const JoinAggDesc* __restrict__ dp) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.build0.nRows;
+  uint32_t mask = (1u << d.keySetLog2) - 1;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+)RTC";
+  if (d.nPred0 > 0) emitPredInline(s, d, d.pred0, "d.build0", "row", nullptr);
+  if (d.build0.cols[d.b0KeyCol].hasNulls)
+    s << "    if (colIsNull(d.build0.cols[" << d.b0KeyCol
+      << "], row)) continue;\n";
+  s << "    uint64_t key = gptr<uint64_t>(d.build0.cols[" << d.b0KeyCol
+    << "].data)[row];\n";
+  s << R"RTC(    if (key == kEmptyKey) key = kEmptyKey - 1;
+    jaBloom0Set(d, key);
+    uint32_t slot = (uint32_t)(splitmix64(key) & mask);
+    for (uint32_t probe = 0; probe <= mask; probe++) {
+      uint64_t cur = d.keySet[slot];
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.keySet[slot],
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & mask;
+      if (probe == mask) atomicOr(d.errorFlag, kErrGlobalFull);
+    }
+  }
+}
+)RTC";
+  // ---- count1 / build1 over build1 (orders): 2-deep pipelined pred load ---
+  // (only when the predicate is a slotless 8B compare; else plain loop)
+  bool pipe1 = d.nPred1 > 0 && (d.pred1.kind == gxp::PRED_TIME_CMP_CONST ||
+                                d.pred1.kind == gxp::PRED_I64_CMP_CONST) &&
+               !d.build1.cols[d.pred1.col].hasNulls;
+  auto emitB1Loop = [&](bool counting) {
+    const char* fn = counting ? "genja_count1" : "genja_build1";
+    s << "extern \"C\" __global__ void __launch_bounds__(256) " << fn
+      << "(const JoinAggDesc* __restrict__ dp) {\n"
+         "  const JoinAggDesc& d = *dp;\n"
+         "  int64_t n = d.build1.nRows;\n";
+    if (!counting) s << "  uint32_t mask = (1u << d.slotsLog2) - 1;\n";
+    if (counting) s << "  uint64_t my = 0;\n";
+    s << "  const int64_t stride = (int64_t)gridDim.x * blockDim.x;\n"
+         "  int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;\n";
+    if (pipe1) {
+      s << "  uint64_t preA = row < n ? gptr<uint64_t>(d.build1.cols["
+        << d.pred1.col << "].data)[row] : 0;\n";
+    }
+    s << "  for (; row < n; row += stride) {\n";
+    if (pipe1) {
+      s << "    uint64_t preCur = preA;\n"
+           "    { int64_t rn = row + stride;\n"
+           "      if (rn < n) preA = gptr<uint64_t>(d.build1.cols["
+        << d.pred1.col << "].data)[rn]; }\n";
+      emitPredInline(s, d, d.pred1, "d.build1", "row", "preCur");
+    } else if (d.nPred1 > 0) {
+      emitPredInline(s, d, d.pred1, "d.build1", "row", nullptr);
+    }
+    if (d.build1.cols[d.b1ProbeCol].hasNulls)
+      s << "    if (colIsNull(d.build1.cols[" << d.b1ProbeCol
+        << "], row)) continue;\n";
+    s << "    uint64_t pkey0 = gptr<uint64_t>(d.build1.cols[" << d.b1ProbeCol
+      << "].data)[row];\n"
+         "    if (!jaBloom0MayHave(d, pkey0)) continue;\n"
+         "    if (!jaKeySetHas(d, pkey0)) continue;\n";
+    if (counting) {
+      s << "    my++;\n  }\n"
+           "  for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, "
+           "off, 64);\n"
+           "  if ((threadIdx.x & 63) == 0 && my)\n"
+           "    atomicAdd((unsigned long long*)&d.counters[1], (unsigned long "
+           "long)my);\n}\n\n";
+      return;
+    }
+    if (d.build1.cols[d.b1KeyCol].hasNulls)
+      s << "    if (colIsNull(d.build1.cols[" << d.b1KeyCol
+        << "], row)) continue;\n";
+    s << "    uint64_t key = gptr<uint64_t>(d.build1.cols[" << d.b1KeyCol
+      << "].data)[row];\n"
+         "    if (key == kEmptyKey) key = kEmptyKey - 1;\n";
+    if (d.payloadCol0 >= 0)
+      s << "    uint64_t pay0 = gptr<uint64_t>(d.build1.cols[" << d.payloadCol0
+        << "].data)[row];\n";
+    else
+      s << "    uint64_t pay0 = 0;\n";
+    if (d.payloadCol1 >= 0)
+      s << "    int64_t pay1 = gptr<int64_t>(d.build1.cols[" << d.payloadCol1
+        << "].data)[row];\n";
+    else
+      s << "    int64_t pay1 = 0;\n";
+    s << R"RTC(    jaBloomSet(d, key);
+    uint32_t slot = (uint32_t)(splitmix64(key) & mask);
+    for (uint32_t probe = 0; probe <= mask; probe++) {
+      uint64_t cur = d.slots[slot].key;
+      if (cur == key) { atomicOr(d.errorFlag, kErrBadKey); break; }
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.slots[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey) {
+          d.slots[slot].payload0 = pay0;
+          d.slots[slot].payload1 = pay1;
+          break;
+        }
+        if (prev == key) { atomicOr(d.errorFlag, kErrBadKey); break; }
+      }
+      slot = (slot + 1) & mask;
+      if (probe == mask) atomicOr(d.errorFlag, kErrGlobalFull);
+    }
+  }
+}
+
+)RTC";
+  };
+  emitB1Loop(true);
+  emitB1Loop(false);
+  return s.str();
+}
+
+struct JaBuildProg {
+  hipModule_t mod = nullptr;
+  hipFunction_t fn[4] = {nullptr, nullptr, nullptr, nullptr};  // c0,b0,c1,b1
+  bool ok = false;
+};
+
+static std::map<std::string, JaBuildProg>& buildCache() {
+  static std::map<std::string, JaBuildProg> c;
+  return c;
+}
+
+const JaBuildProg* compileJaBuild(const JoinAggDesc& d, std::string* whyNot) {
+  std::string src = generateJaBuildSource(d);
+  std::lock_guard<std::mutex> lk(cacheMu);
+  auto it = buildCache().find(src);
+  if (it != buildCache().end()) return it->second.ok ? &it->second : nullptr;
+  JaBuildProg prog;
+  hiprtcProgram rp;
+  if (hiprtcCreateProgram(&rp, src.c_str(), "genjab.cu", 0, nullptr,
+                          nullptr) != HIPRTC_SUCCESS) {
+    if (whyNot) *whyNot = "hiprtcCreateProgram failed";
+    buildCache()[src] = prog;
+    return nullptr;
+  }
+  std::string inc = "-I" + headerDir();
+  const char* opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17",
+                        inc.c_str()};
+  if (hiprtcCompileProgram(rp, 4, opts) != HIPRTC_SUCCESS) {
+    if (whyNot) {
+      size_t lsz = 0;
+      hiprtcGetProgramLogSize(rp, &lsz);
+      std::string log(lsz, '\0');
+      if (lsz) hiprtcGetProgramLog(rp, &log[0]);
+      *whyNot = "hiprtc compile failed: " + log;
+    }
+    hiprtcDestroyProgram(&rp);
+    buildCache()[src] = prog;
+    return nullptr;
+  }
+  size_t csz = 0;
+  hiprtcGetCodeSize(rp, &csz);
+  std::string code(csz, '\0');
+  hiprtcGetCode(rp, &code[0]);
+  hiprtcDestroyProgram(&rp);
+  static const char* names[4] = {"genja_count0", "genja_build0",
+                                 "genja_count1", "genja_build1"};
+  bool ok = hipModuleLoadData(&prog.mod, code.data()) == hipSuccess;
+  for (int i = 0; ok && i < 4; i++)
+    ok = hipModuleGetFunction(&prog.fn[i], prog.mod, names[i]) == hipSuccess;
+  if (!ok) {
+    if (whyNot) *whyNot = "hipModule load failed";
+    buildCache()[src] = prog;
+    return nullptr;
+  }
+  prog.ok = true;
+  auto& slot = buildCache()[src] = prog;
+  return &slot;
+}
+
+int launchJaBuild(const JaBuildProg* prog, int phase,
+                  const JoinAggDesc* devDesc, int grid, void* stream) {
+  void* args[] = {(void*)&devDesc};
+  return (int)hipModuleLaunchKernel(prog->fn[phase], grid, 1, 1, 256, 1, 1, 0,
+                                    (hipStream_t)stream, args, nullptr);
+}
+
 }  // namespace gxjit

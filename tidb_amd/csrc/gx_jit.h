@@ -16,4 +16,10 @@ std::string generateSource(const gxp::FusedQueryDesc& d);
 const JitProg* compileJa(const gxp::JoinAggDesc& d, std::string* whyNot);
 int launchJa(const JitProg* prog, bool wide, const gxp::JoinAggDesc* devDesc,
              int grid, void* stream);
+// join-aggregate build-phase kernels (count0/build0/count1/build1)
+struct JaBuildProg;
+const JaBuildProg* compileJaBuild(const gxp::JoinAggDesc& d,
+                                  std::string* whyNot);
+int launchJaBuild(const JaBuildProg* prog, int phase,
+                  const gxp::JoinAggDesc* devDesc, int grid, void* stream);
 }  // namespace gxjit

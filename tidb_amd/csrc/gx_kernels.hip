@@ -863,12 +863,15 @@ __global__ void jaMaxKernel(const JoinAggDesc* __restrict__ dp, uint64_t* outMax
   uint64_t my = 0;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const JoinAggSlot& sm = d.slots[i];
-    if (sm.key == kEmptyKey || (sm.cnt == 0 && sm.accLo == 0 && sm.accHi == 0))
+    uint64_t skey = gptr<uint64_t>(&d.slots[i].key)[0];
+    if (skey == kEmptyKey) continue;
+    uint64_t accLo = gptr<uint64_t>(&d.slots[i].accLo)[0];
+    if (accLo == 0 && gptr<uint64_t>(&d.slots[i].accHi)[0] == 0 &&
+        gptr<uint64_t>(&d.slots[i].cnt)[0] == 0)
       continue;
-    if (sm.accHi != 0) atomicOr(d.errorFlag, kErrOverflow);
-    uint64_t v = sm.accLo;
-    if (v > my) my = v;
+    if (gptr<uint64_t>(&d.slots[i].accHi)[0] != 0)
+      atomicOr(d.errorFlag, kErrOverflow);
+    if (accLo > my) my = accLo;
   }
   for (int off = 32; off > 0; off >>= 1) {
     uint64_t o = __shfl_down(my, off, 64);
@@ -881,13 +884,27 @@ __global__ void jaHistKernel(const JoinAggDesc* __restrict__ dp, uint32_t* hist,
                              int shift) {
   const JoinAggDesc& d = *dp;
   int64_t n = 1LL << d.slotsLog2;
+  // ~14M occupied slots funneling into 4096 bins: privatize the histogram
+  // in LDS per block and flush once (the global-atomic version measured
+  // 6.4 ms of pure contention, 25x the scan floor)
+  __shared__ uint32_t lh[4096];
+  for (int b = threadIdx.x; b < 4096; b += blockDim.x) lh[b] = 0;
+  __syncthreads();
+  auto lh3 = (__attribute__((address_space(3))) uint32_t*)lh;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const JoinAggSlot& sh = d.slots[i];
-    if (sh.key == kEmptyKey || (sh.cnt == 0 && sh.accLo == 0 && sh.accHi == 0))
+    uint64_t key = gptr<uint64_t>(&d.slots[i].key)[0];
+    if (key == kEmptyKey) continue;
+    uint64_t accLo = gptr<uint64_t>(&d.slots[i].accLo)[0];
+    if (accLo == 0 && gptr<uint64_t>(&d.slots[i].accHi)[0] == 0 &&
+        gptr<uint64_t>(&d.slots[i].cnt)[0] == 0)
       continue;
-    atomicAdd(&hist[(sh.accLo >> shift) & 4095], 1u);
+    __hip_atomic_fetch_add(&lh3[(accLo >> shift) & 4095], 1u,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
   }
+  __syncthreads();
+  for (int b = threadIdx.x; b < 4096; b += blockDim.x)
+    if (lh[b]) atomicAdd(&hist[b], lh[b]);
 }
 
 __global__ void jaCompactKernel(const JoinAggDesc* __restrict__ dp, TopNOut* out,
@@ -897,9 +914,10 @@ __global__ void jaCompactKernel(const JoinAggDesc* __restrict__ dp, TopNOut* out
   int64_t n = 1LL << d.slotsLog2;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
+    uint64_t skey = gptr<uint64_t>(&d.slots[i].key)[0];
+    if (skey == kEmptyKey) continue;
     const JoinAggSlot& s = d.slots[i];
-    if (s.key == kEmptyKey || (s.cnt == 0 && s.accLo == 0 && s.accHi == 0))
-      continue;
+    if (s.cnt == 0 && s.accLo == 0 && s.accHi == 0) continue;
     if ((s.accLo >> shift) < thresholdBucket) continue;
     uint64_t idx = atomicAdd((unsigned long long*)outCount, 1ULL);
     if (idx >= cap) { atomicOr(d.errorFlag, kErrGlobalFull); continue; }

@@ -93,6 +93,31 @@ def measured_traffic(metric, rows_this_run):
         return None
 
 
+def run_cpu_baseline_q3(n_li=2_000_000):
+    """Oracle Q3 over a bounded sample (~5-10 s of CPU), single thread."""
+    import time as _t
+    import tests.test_oracle_q3 as q3
+    from tests.gxlib import load_oracle
+    lib = load_oracle()
+    old = (q3.N_LI, q3.N_ORD, q3.N_CUST)
+    try:
+        q3.N_LI, q3.N_ORD, q3.N_CUST = n_li, n_li // 4, n_li // 40
+        t0 = _t.perf_counter()
+        q3.run_q3(lib)
+        dt = _t.perf_counter() - t0
+    finally:
+        q3.N_LI, q3.N_ORD, q3.N_CUST = old
+    total = n_li + n_li // 4 + n_li // 40
+    return {
+        "value": total / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"Q3 over {n_li} lineitem (+orders/customer) rows incl. "
+                  f"generation, single thread ({dt:.1f}s)",
+    }
+
+
 def bench_q3(args):
     """TPC-H Q3 (BASELINE config 3): 3-table join + grouped sum + TopN on one
     GPU. One step = the full pipeline (customer/orders build + lineitem probe
@@ -169,10 +194,30 @@ def bench_q3(args):
             "frac": achieved / HBM_PEAK_GBS,
             "traffic": None,
         },
-        "cpu_baseline": None,
+        "cpu_baseline": run_cpu_baseline_q3() if not args.no_cpu_baseline
+        else None,
         "result_rows": len(rows),
     }
     print(json.dumps(out), flush=True)
+
+
+def run_cpu_baseline_sort(n=1_000_000):
+    """Oracle full sort over a bounded sample, single thread."""
+    import time as _t
+    from tests.test_full_sort import run_sort, KEYS_2
+    from tests.gxlib import load_oracle
+    lib = load_oracle()
+    t0 = _t.perf_counter()
+    run_sort(lib, KEYS_2, [0, 0], n_rows=n)
+    dt = _t.perf_counter() - t0
+    return {
+        "value": n / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"full sort of {n} lineitem rows incl. generation and "
+                  f"result pull, single thread ({dt:.1f}s)",
+    }
 
 
 def bench_sort(args):
@@ -250,7 +295,8 @@ def bench_sort(args):
             "frac": achieved / HBM_PEAK_GBS,
             "traffic": None,
         },
-        "cpu_baseline": None,
+        "cpu_baseline": run_cpu_baseline_sort() if not args.no_cpu_baseline
+        else None,
     }
     print(json.dumps(out), flush=True)
 

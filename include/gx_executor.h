@@ -78,7 +78,12 @@ typedef struct gx_chunk {
  */
 enum {
   GX_F_LT = 0, GX_F_LE = 1, GX_F_GT = 2, GX_F_GE = 3, GX_F_EQ = 4, GX_F_NE = 5,
-  GX_F_PLUS = 16, GX_F_MINUS = 17, GX_F_MUL = 18, GX_F_DIV = 19
+  GX_F_PLUS = 16, GX_F_MINUS = 17, GX_F_MUL = 18, GX_F_DIV = 19,
+  /* casts (builtin_cast_vec.go; types.ProduceDecWithSpecifiedTp
+   * datum.go:1629): CAST_DEC rounds HalfUp to the call's ret_frac (flen
+   * clamping unimplemented this round); CAST_INT = Round(0, HalfUp) + ToInt
+   * (builtin_cast_vec.go:1817-1852). */
+  GX_F_CAST_DEC = 20, GX_F_CAST_INT = 21
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

@@ -108,6 +108,41 @@ int32_t evalCompare(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
   return GX_OK;
 }
 
+// casts (builtin_cast_vec.go): CAST_DEC = ProduceDecWithSpecifiedTp's
+// Round(HalfUp, ret_frac) (datum.go:1629-1660; flen clamping out of scope
+// this round); CAST_INT = Round(0, HalfUp) + ToInt (:1817-1852)
+int32_t evalCast(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
+  Column a;
+  int32_t err = evalVec(ctx, e.args[0], in, a);
+  if (err) return err;
+  int n = in.numRows();
+  out.type = e.retType;
+  out.frac = e.retFrac;
+  out.reset();
+  for (int i = 0; i < n; i++) {
+    if (a.isNull(i)) { out.appendNull(); continue; }
+    MyDecimal d;
+    if (a.type == GX_TYPE_DECIMAL) d = *a.getDecimal(i);
+    else if (a.type == GX_TYPE_I64) d.FromInt(a.getI64(i));
+    else { *ctx.err = "unsupported cast argument type"; return GX_ERR_INVALID; }
+    if (e.func == GX_F_CAST_DEC) {
+      MyDecimal r;
+      int32_t ec = d.Round(&r, e.retFrac, ModeHalfUp);
+      if (ec != E_OK && ec != E_TRUNCATED) return ec;
+      out.appendDecimal(r);
+    } else {  // GX_F_CAST_INT
+      MyDecimal r;
+      int32_t ec = d.Round(&r, 0, ModeHalfUp);
+      if (ec != E_OK && ec != E_TRUNCATED) return ec;
+      int64_t v = 0;
+      ec = r.ToInt(&v);
+      if (ec == E_OVERFLOW) { *ctx.err = "cast overflow"; return ec; }
+      out.appendI64(v);
+    }
+  }
+  return GX_OK;
+}
+
 int32_t evalArith(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
   Column a, b;
   int32_t err = evalVec(ctx, e.args[0], in, a);
@@ -189,6 +224,8 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       return GX_OK;
     case EK_CALL:
       if (e.func <= GX_F_NE) return evalCompare(ctx, e, in, out);
+      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT)
+        return evalCast(ctx, e, in, out);
       return evalArith(ctx, e, in, out);
   }
   return GX_ERR_INVALID;

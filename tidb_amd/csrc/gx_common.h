@@ -50,6 +50,9 @@ enum VmOp : int32_t {
   VM_DIV = 7,        // dst <- trunc(a * 10^c / b) (DecimalDiv semantics:
                      // result scale word-granular, mydecimal.go doDiv;
                      // div-by-zero -> NULL; engine forces the wide VM)
+  VM_ROUND_SCALE = 8,  // dst <- round_half_up(a, scale b -> scale... encoded:
+                       // ins.b = target scale, ins.c = source scale (the
+                       // cast family, ProduceDecWithSpecifiedTp/ToInt)
 };
 
 struct VmIns {

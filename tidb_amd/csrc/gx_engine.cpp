@@ -355,6 +355,25 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       break;
     }
     case EK_CALL: {
+      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT) {
+        if (e.args.size() != 1) {
+          ex->err = "cast takes one argument";
+          return -1;
+        }
+        int sa = 0;
+        int ra = compileExpr(ex, e.args[0], &sa);
+        if (ra < 0) return -1;
+        int sr = e.func == GX_F_CAST_INT ? 0 : e.retFrac;
+        if (sr == sa) {
+          reg = ra;  // no-op cast
+          *scaleOut = sr;
+          break;
+        }
+        reg = emit(gxp::VM_ROUND_SCALE, allocReg(), ra, sr);
+        if (reg >= 0) d.ins[d.nIns - 1].c = sa;
+        *scaleOut = sr;
+        break;
+      }
       if (e.args.size() != 2) {
         ex->err = "unsupported call arity on device";
         return -1;
@@ -520,6 +539,10 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
       break;
     }
     case EK_CALL: {
+      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT) {
+        ex->err = "casts unsupported in the join-probe VM this round";
+        return -1;
+      }
       if (e.args.size() != 2) {
         ex->err = "unsupported call arity on device";
         return -1;
@@ -1171,6 +1194,9 @@ static int32_t compileFused(gx_exec* ex) {
         d.insMagic[i] = magich[9 - f];
       } else if (d.ins[i].op == gxp::VM_SCALE_UP) {
         d.insP10[i] = p10h[d.ins[i].b];
+      } else if (d.ins[i].op == gxp::VM_ROUND_SCALE) {
+        int sh = d.ins[i].b - d.ins[i].c;
+        d.insP10[i] = p10h[sh >= 0 ? sh : -sh];
       }
     }
     // pick the fetch-pipeline depth: keep raw state within the VGPR budget

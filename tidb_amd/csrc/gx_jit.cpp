@@ -159,6 +159,34 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
           << ", VT<WIDE>::fromI64(" << d.insP10[i] << "LL, nullptr), &ovf); "
           << "bool " << nv << " = n" << ins.a << ";\n";
         break;
+      case gxp::VM_ROUND_SCALE: {
+        int up = ins.b - ins.c;
+        s << "  bool " << nv << " = n" << ins.a << ";\n"
+          << "  T " << v << " = VT<WIDE>::zero();\n"
+          << "  if (!" << nv << ") {\n";
+        if (up >= 0) {
+          s << "    " << v << " = VT<WIDE>::mul(v" << ins.a
+            << ", VT<WIDE>::fromI64(" << d.insP10[i]
+            << "LL, nullptr), &ovf);\n";
+        } else {
+          s << "    Int128 ai = VT<WIDE>::toAcc(v" << ins.a << ");\n"
+            << "    __int128 av = ((__int128)ai.hi << 64) | (__int128)ai.lo;\n"
+            << "    uint64_t dv = " << (uint64_t)d.insP10[i] << "ULL;\n"
+            << "    unsigned __int128 aAbs = (unsigned __int128)(av < 0 ? -av : av);\n"
+            << "    unsigned __int128 q = u128DivU64(aAbs, dv);\n"
+            << "    unsigned __int128 r = aAbs - q * dv;\n"
+            << "    if (2 * (uint64_t)r >= dv) q += 1;\n"
+            << "    __int128 sq = av < 0 ? -(__int128)q : (__int128)q;\n"
+            << "    if (!WIDE && (sq > (__int128)INT64_MAX || sq < (__int128)INT64_MIN)) {\n"
+            << "      atomicOr(d.errorFlag, kErrRetryWide); return false; }\n"
+            << "    if (WIDE) { Int128 rr = {(uint64_t)sq, (int64_t)(sq >> 64)}; "
+            << v << " = *(T*)&rr; }\n"
+            << "    else { int64_t qq = (int64_t)sq; " << v
+            << " = *(T*)&qq; }\n";
+        }
+        s << "  }\n";
+        break;
+      }
       case gxp::VM_DIV:
         s << "  bool " << nv << " = n" << ins.a << " || n" << ins.b << ";\n"
           << "  T " << v << " = VT<WIDE>::zero();\n"

@@ -238,6 +238,45 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
                              VT<WIDE>::fromI64(d.insP10[i], nullptr), &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a));
         break;
+      case VM_ROUND_SCALE: {
+        // cast family: round_half_up from scale ins.c to scale ins.b
+        // (ProduceDecWithSpecifiedTp datum.go:1629; ToInt when the engine
+        // lowered cast-as-int to target scale 0)
+        bool nul = vm.isNull(ins.a);
+        typename VT<WIDE>::T v = VT<WIDE>::zero();
+        if (!nul) {
+          int up = ins.b - ins.c;
+          if (up >= 0) {
+            v = VT<WIDE>::mul(vm.get(ins.a),
+                              VT<WIDE>::fromI64(d.insP10[i], nullptr), &ovf);
+          } else {
+            Int128 ai = VT<WIDE>::toAcc(vm.get(ins.a));
+            __int128 av = ((__int128)ai.hi << 64) | (__int128)ai.lo;
+            uint64_t div = (uint64_t)d.insP10[i];  // 10^(c-b), <= 10^18
+            unsigned __int128 aAbs = (unsigned __int128)(av < 0 ? -av : av);
+            unsigned __int128 q = u128DivU64(aAbs, div);
+            unsigned __int128 r = aAbs - q * div;
+            if (2 * (uint64_t)r >= div) q += 1;  // half away from zero
+            __int128 sq = av < 0 ? -(__int128)q : (__int128)q;
+            if (!WIDE &&
+                (sq > (__int128)INT64_MAX || sq < (__int128)INT64_MIN)) {
+              atomicOr(d.errorFlag, kErrRetryWide);
+              bad = true;
+              break;
+            }
+            if (WIDE) {
+              Int128 rr = {(uint64_t)sq, (int64_t)(sq >> 64)};
+              v = *(typename VT<WIDE>::T*)&rr;
+            } else {
+              int64_t qq = (int64_t)sq;
+              v = *(typename VT<WIDE>::T*)&qq;
+            }
+          }
+        }
+        vm.set(ins.dst, v);
+        vm.setNull(ins.dst, nul);
+        break;
+      }
       case VM_DIV: {
         if constexpr (!DIVOK) {
           // engine launches the DIVOK variant for plans containing DIV;

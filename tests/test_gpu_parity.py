@@ -167,3 +167,15 @@ def test_q1_interpreted_fallback_parity(libs, monkeypatch):
     got = _as_map(_run_q1(product, 65536))
     monkeypatch.delenv("GX_NO_JIT")
     assert got == _as_map(_run_q1(oracle, 65536))
+
+
+def test_q1_glds_staged_parity(libs, monkeypatch):
+    """GX_GLDS=1 exercises the glds-staged (LDS-DMA) kernel variant — kept
+    opt-in (slower than the specialized kernels) but parity must hold."""
+    oracle, product = libs
+    monkeypatch.setenv("GX_GLDS", "1")
+    monkeypatch.setenv("GX_NO_JIT", "1")  # glds path is the interpreted engine
+    got = _as_map(_run_q1(product, 65536))
+    monkeypatch.delenv("GX_GLDS")
+    monkeypatch.delenv("GX_NO_JIT")
+    assert got == _as_map(_run_q1(oracle, 65536))

@@ -64,7 +64,7 @@ struct VmIns {
 };
 
 constexpr int kMaxVmIns = 48;
-constexpr int kMaxVmRegs = 12;
+constexpr int kMaxVmRegs = 14;
 constexpr int kMaxVmConsts = 16;
 
 // ---- filter (CNF of simple predicates; Q1-class) ----
@@ -183,7 +183,9 @@ struct FusedQueryDesc {
   int32_t accReg[kMaxAggs];   // phys acc slot -> VM register
   int32_t accMap[kMaxAggs];   // agg index -> phys acc slot (-1 for COUNT)
   int32_t sharedCnt = 0;
-  int32_t hasDiv = 0;  // launch the DIVOK kernel variant (division code is
+  int32_t hasDiv = 0;
+  int32_t nVmRegs = 0;  // registers the compiled VM uses (>12 selects the
+                        // wide VmState14 kernel variant)  // launch the DIVOK kernel variant (division code is
                        // compiled out of the common kernels: its register
                        // demand alone costs a wave/SIMD of occupancy)
   GroupKeyDesc gkey;

@@ -341,6 +341,37 @@ struct VmState {
   }
 };
 
+// 14-register VM state for wide projections (kMaxVmRegs > 12): only the
+// kernel variant the launcher picks for such plans pays the extra live
+// registers -- common plans keep the 12-register footprint above
+template <bool WIDE>
+struct VmState14 {
+  using T = typename VT<WIDE>::T;
+  T r0, r1, r2, r3, r4, r5, r6, r7, r8, r9, r10, r11, r12, r13;
+  uint32_t nullBits;
+  __device__ T get(int i) const {
+    switch (i) {
+      case 0: return r0; case 1: return r1; case 2: return r2; case 3: return r3;
+      case 4: return r4; case 5: return r5; case 6: return r6; case 7: return r7;
+      case 8: return r8; case 9: return r9; case 10: return r10;
+      case 11: return r11; case 12: return r12; default: return r13;
+    }
+  }
+  __device__ void set(int i, T v) {
+    switch (i) {
+      case 0: r0 = v; break; case 1: r1 = v; break; case 2: r2 = v; break;
+      case 3: r3 = v; break; case 4: r4 = v; break; case 5: r5 = v; break;
+      case 6: r6 = v; break; case 7: r7 = v; break; case 8: r8 = v; break;
+      case 9: r9 = v; break; case 10: r10 = v; break; case 11: r11 = v; break;
+      case 12: r12 = v; break; default: r13 = v; break;
+    }
+  }
+  __device__ bool isNull(int i) const { return (nullBits >> i) & 1; }
+  __device__ void setNull(int i, bool n) {
+    nullBits = (nullBits & ~(1u << i)) | ((uint32_t)n << i);
+  }
+};
+
 // pack the group key (see GroupKeyDesc comment); offsets/values come from the
 // batched raw fetch
 template <typename RAWT>

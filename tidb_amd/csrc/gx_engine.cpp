@@ -19,6 +19,8 @@
 #include <algorithm>
 #include <cstring>
 #include <map>
+#include <functional>
+#include <set>
 #include <memory>
 #include <string>
 #include <vector>
@@ -121,6 +123,12 @@ struct gx_exec {
   std::vector<std::pair<int, int>> projRegs;  // projection idx -> (reg, scale)
   int vmNextReg = 0;
   std::map<int, std::pair<int, int>> exprRegCache;  // exprId -> (reg, scale)
+  // register recycling: remaining consumer count per expression (computed
+  // from the projection roots before compiling); a subexpression's register
+  // returns to the free list after its last consumer is emitted, so wide
+  // projections fit the 12-register VM state
+  std::map<int, int> exprUse;
+  std::vector<int> vmFreeRegs;
   bool vmHasDiv = false;  // DIV forces the wide VM and disables glds
   // hipRTC-specialized kernel for this plan (gx_jit.cpp); nullptr -> use the
   // interpreted fusedAggKernel
@@ -297,13 +305,69 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
     return cached->second.first;
   }
   auto allocReg = [&]() -> int {
-    if (ex->vmNextReg >= gxp::kMaxVmRegs) return -1;
+    if (!ex->vmFreeRegs.empty()) {
+      int r = ex->vmFreeRegs.back();
+      ex->vmFreeRegs.pop_back();
+      return r;
+    }
+    if (ex->vmNextReg >= gxp::kMaxVmRegs) {
+      if (getenv("GX_DEBUG"))
+        fprintf(stderr, "[gx] reg exhausted at expr %d (nIns=%d, cache=%d)\n",
+                exprId, d.nIns, (int)ex->exprRegCache.size());
+      return -1;
+    }
     return ex->vmNextReg++;
   };
   auto emit = [&](int op, int dst, int a, int b) -> int {
     if (dst < 0 || d.nIns >= gxp::kMaxVmIns) return -1;
     d.ins[d.nIns++] = {op, dst, a, b, -1};
     return dst;
+  };
+  // emitted the last consumer of argExpr -> recycle its register
+  auto release = [&](int argExpr) {
+    auto u = ex->exprUse.find(argExpr);
+    if (u == ex->exprUse.end()) {
+      if (getenv("GX_DEBUG"))
+        fprintf(stderr, "[gx] release %d: not counted (pinned)\n", argExpr);
+      return;  // pinned (projection root) or unknown
+    }
+    if (--u->second > 0) {
+      if (getenv("GX_DEBUG"))
+        fprintf(stderr, "[gx] release %d: %d uses left\n", argExpr, u->second);
+      return;
+    }
+    if (getenv("GX_DEBUG")) fprintf(stderr, "[gx] release %d: freeing\n", argExpr);
+    auto c = ex->exprRegCache.find(argExpr);
+    if (c != ex->exprRegCache.end()) {
+      // aliased entries (no-op casts) share a register: keep it live while
+      // any other expression still maps to it
+      for (auto& kv : ex->exprRegCache)
+        if (kv.first != argExpr && kv.second.first == c->second.first) return;
+      // LOAD results must stay cached: the loads-first reorder hands their
+      // slots to the fetch pipeline once; recompiling one later would add a
+      // second load of the same column. Only recycle derived values.
+      int op = -1;
+      for (int i2 = 0; i2 < d.nIns; i2++)
+        if (d.ins[i2].dst == c->second.first) op = d.ins[i2].op;
+      if (op == gxp::VM_LOAD_DEC || op == gxp::VM_LOAD_I64 ||
+          op == gxp::VM_LOAD_CONST)
+        return;
+      ex->vmFreeRegs.push_back(c->second.first);
+      ex->exprRegCache.erase(c);
+    }
+  };
+  // loads are MOVED to the front of the instruction stream (the grouped
+  // fetch pipeline); a load's destination register must therefore never be
+  // a recycled one -- an instruction that originally preceded the load and
+  // wrote that register would execute after it post-reorder and clobber the
+  // loaded value. Fresh registers only for loads.
+  auto allocRegFresh = [&]() -> int {
+    if (ex->vmNextReg >= gxp::kMaxVmRegs) {
+      if (getenv("GX_DEBUG"))
+        fprintf(stderr, "[gx] fresh reg exhausted at expr %d\n", exprId);
+      return -1;
+    }
+    return ex->vmNextReg++;
   };
   int reg = -1;
   switch (e.kind) {
@@ -315,10 +379,10 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       int t = ex->desc.table.cols[e.colIdx].type;
       if (t == GX_TYPE_DECIMAL) {
         int frac = ex->desc.table.cols[e.colIdx].frac;
-        reg = emit(gxp::VM_LOAD_DEC, allocReg(), e.colIdx, frac);
+        reg = emit(gxp::VM_LOAD_DEC, allocRegFresh(), e.colIdx, frac);
         *scaleOut = frac;
       } else if (t == GX_TYPE_I64) {
-        reg = emit(gxp::VM_LOAD_I64, allocReg(), e.colIdx, 0);
+        reg = emit(gxp::VM_LOAD_I64, allocRegFresh(), e.colIdx, 0);
         *scaleOut = 0;
       } else {
         ex->err = "unsupported colref type in device expression";
@@ -371,6 +435,7 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
         }
         reg = emit(gxp::VM_ROUND_SCALE, allocReg(), ra, sr);
         if (reg >= 0) d.ins[d.nIns - 1].c = sa;
+        release(e.args[0]);
         *scaleOut = sr;
         break;
       }
@@ -417,12 +482,18 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       } else {
         // align scales to max (MySQL add/sub result frac = max(f1,f2))
         int target = std::max(sa, sb);
-        if (sa < target) ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa);
-        if (sb < target) rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb);
+        int tmpA = -1, tmpB = -1;
+        if (sa < target) tmpA = ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa);
+        if (sb < target) tmpB = rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb);
         if (ra < 0 || rb < 0) break;
         reg = emit(op, allocReg(), ra, rb);
+        // anonymous scale-up temps die with this op
+        if (tmpA >= 0) ex->vmFreeRegs.push_back(tmpA);
+        if (tmpB >= 0) ex->vmFreeRegs.push_back(tmpB);
         *scaleOut = target;
       }
+      release(e.args[0]);
+      release(e.args[1]);
       break;
     }
   }
@@ -477,7 +548,7 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
     return cached->second.first;
   }
   auto allocReg = [&]() -> int {
-    if (B.nextReg >= gxp::kMaxVmRegs) return -1;
+    if (B.nextReg >= 12) return -1;  // the probe kernel's VmState is 12-reg
     return B.nextReg++;
   };
   auto emit = [&](int op, int dst, int a, int b, int c) -> int {
@@ -1005,6 +1076,34 @@ static int32_t compileFused(gx_exec* ex) {
     }
   }
 
+  // consumer counts over the expression DAG (register recycling: a
+  // subexpression's register frees after its last consumer; roots — the
+  // projection outputs the aggregates read — are pinned by not being
+  // counted here)
+  ex->exprUse.clear();
+  ex->vmFreeRegs.clear();
+  {
+    std::vector<int> roots;
+    if (proj) {
+      for (int pe : proj->exprs) roots.push_back(pe);
+    } else {
+      for (int a : agg->aggArgs)
+        if (a >= 0) roots.push_back(a);
+    }
+    std::set<int> visited;
+    std::function<void(int)> cnt = [&](int id) {
+      for (int a : plan.exprs[id].args) {
+        ex->exprUse[a]++;
+        if (visited.insert(a).second) cnt(a);
+      }
+    };
+    for (int r : roots) cnt(r);
+    // a root may also appear as a subexpression (e.g. a projected value the
+    // cast of which is also projected): roots are pinned outputs, never
+    // released
+    for (int r : roots) ex->exprUse.erase(r);
+  }
+
   // projection exprs -> VM registers (group-col projections stay colrefs)
   std::vector<int> projSrcCol;  // proj idx -> source col for passthroughs
   ex->projRegs.clear();
@@ -1165,6 +1264,7 @@ static int32_t compileFused(gx_exec* ex) {
       }
     }
     d.nLoadIns = (int)loads.size();
+    d.nVmRegs = ex->vmNextReg;  // physical register budget for kernel dispatch
     int k = 0;
     for (auto& ins : loads) d.ins[k++] = ins;
     for (auto& ins : rest) d.ins[k++] = ins;
@@ -1515,7 +1615,8 @@ static int32_t materializeDevice(gx_exec* ex) {
     // The glds-staged kernel is parity-green but measured slower than the
     // plain grouped-fetch kernel on Q1/SF10 (5.48 vs 4.71 ms), so it ships
     // opt-in until the pipelining wins back the staging overhead.
-    bool ok = tab.nRows >= 256 && getenv("GX_GLDS") && !ex->vmHasDiv;
+    bool ok = tab.nRows >= 256 && getenv("GX_GLDS") && !ex->vmHasDiv &&
+              ex->vmNextReg <= 12;  // the staged kernel's VmState is 12-reg
     for (int f = 0; f < d.nFetch && ok; f++)
       ok = d.fetch[f].kind == gxp::FETCH_8B ||
            d.fetch[f].kind == gxp::FETCH_DEC16 ||
@@ -1645,6 +1746,7 @@ static int32_t runFused(gx_exec* ex) {
   if (ex->vmHasDiv && !getenv("GX_DIV_NARROW"))
     ex->desc.wide = 1;  // DIV quotients rarely fit int64
   ex->desc.hasDiv = ex->vmHasDiv ? 1 : 0;
+  ex->desc.nVmRegs = ex->vmNextReg;
   if (getenv("GX_FORCE_WIDE")) ex->desc.wide = 1;
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;
@@ -2745,6 +2847,8 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
         ex->projRegs.clear();
         ex->vmNextReg = 0;
         ex->exprRegCache.clear();
+        ex->exprUse.clear();
+        ex->vmFreeRegs.clear();
         rc = compileJoinAgg(ex);
         if (rc != GX_OK && ex->err.empty())
           ex->err = "plan compilation failed";

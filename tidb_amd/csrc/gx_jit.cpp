@@ -71,6 +71,11 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
        "__device__ __forceinline__ bool rowPipe(const FusedQueryDesc& d, "
        "int64_t row, const RawT& raw, Lds3GroupSlot* lds, uint64_t* mySel) {\n"
        "  using T = typename VT<WIDE>::T;\n";
+  // registers are RECYCLED by the plan compiler: declare every physical
+  // register once, instructions assign
+  for (int r = 0; r < d.nVmRegs; r++)
+    s << "  T v" << r << " = VT<WIDE>::zero(); bool n" << r
+      << " = false; (void)v" << r << "; (void)n" << r << ";\n";
   // ---- predicates (literal) ----
   for (int p = 0; p < d.nPreds; p++) {
     const gxp::PredDesc& pd = d.preds[p];
@@ -101,7 +106,7 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
     switch (ins.op) {
       case gxp::VM_LOAD_DEC: {
         const gxp::DevCol& c = d.table.cols[ins.a];
-        s << "  T " << v << " = VT<WIDE>::zero(); bool " << nv << " = false;\n";
+        s << "  " << v << " = VT<WIDE>::zero(); " << nv << " = false;\n";
         if (c.hasNulls)
           s << "  " << nv << " = colIsNull(d.table.cols[" << ins.a
             << "], row);\n";
@@ -120,52 +125,52 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
       }
       case gxp::VM_LOAD_I64: {
         const gxp::DevCol& c = d.table.cols[ins.a];
-        s << "  bool " << nv << " = false;\n";
+        s << "  " << nv << " = false;\n";
         if (c.hasNulls)
           s << "  " << nv << " = colIsNull(d.table.cols[" << ins.a
             << "], row);\n";
-        s << "  T " << v << " = " << nv
+        s << "  " << v << " = " << nv
           << " ? VT<WIDE>::zero() : VT<WIDE>::fromI64((int64_t)raw.get("
           << ins.c << ").x, &ovf);\n";
         break;
       }
       case gxp::VM_LOAD_CONST:
-        s << "  T " << v << ";\n"
-          << "  { Int128 cv = {" << (uint64_t)d.constLo[ins.a] << "ULL, (int64_t)"
+        s << "  { Int128 cv = {" << (uint64_t)d.constLo[ins.a] << "ULL, (int64_t)"
           << d.constHi[ins.a] << "LL };\n"
           << "    if (WIDE) " << v << " = *(T*)&cv;\n"
           << "    else { int64_t c64 = " << d.constLo[ins.a] << "LL; " << v
           << " = *(T*)&c64; } }\n"
-          << "  const bool " << nv << " = false;\n";
+          << "  " << nv << " = false;\n";
         break;
       case gxp::VM_ADD:
-        s << "  T " << v << " = VT<WIDE>::add(v" << ins.a << ", v" << ins.b
-          << ", &ovf); bool " << nv << " = n" << ins.a << " || n" << ins.b
-          << ";\n";
+        s << "  { T t2 = VT<WIDE>::add(v" << ins.a << ", v" << ins.b
+          << ", &ovf); bool t3 = n" << ins.a << " || n" << ins.b << "; " << v
+          << " = t2; " << nv << " = t3; }\n";
         break;
       case gxp::VM_SUB:
-        s << "  T " << v << " = VT<WIDE>::sub(v" << ins.a << ", v" << ins.b
-          << ", &ovf); bool " << nv << " = n" << ins.a << " || n" << ins.b
-          << ";\n";
+        s << "  { T t2 = VT<WIDE>::sub(v" << ins.a << ", v" << ins.b
+          << ", &ovf); bool t3 = n" << ins.a << " || n" << ins.b << "; " << v
+          << " = t2; " << nv << " = t3; }\n";
         break;
       case gxp::VM_MUL:
-        s << "  bool " << nv << " = n" << ins.a << " || n" << ins.b << ";\n"
-          << "  T " << v << " = VT<WIDE>::zero();\n"
-          << "  if (!" << nv << ") " << v << " = VT<WIDE>::mul(v" << ins.a
-          << ", v" << ins.b << ", &ovf);\n";
+        s << "  { bool t3 = n" << ins.a << " || n" << ins.b << ";\n"
+          << "    T t2 = VT<WIDE>::zero();\n"
+          << "    if (!t3) t2 = VT<WIDE>::mul(v" << ins.a << ", v" << ins.b
+          << ", &ovf);\n    " << v << " = t2; " << nv << " = t3; }\n";
         break;
       case gxp::VM_SCALE_UP:
-        s << "  T " << v << " = VT<WIDE>::mul(v" << ins.a
+        s << "  { T t2 = VT<WIDE>::mul(v" << ins.a
           << ", VT<WIDE>::fromI64(" << d.insP10[i] << "LL, nullptr), &ovf); "
-          << "bool " << nv << " = n" << ins.a << ";\n";
+          << "bool t3 = n" << ins.a << "; " << v << " = t2; " << nv
+          << " = t3; }\n";
         break;
       case gxp::VM_ROUND_SCALE: {
         int up = ins.b - ins.c;
-        s << "  bool " << nv << " = n" << ins.a << ";\n"
-          << "  T " << v << " = VT<WIDE>::zero();\n"
-          << "  if (!" << nv << ") {\n";
+        s << "  { bool t3 = n" << ins.a << ";\n"
+          << "  T t2 = VT<WIDE>::zero();\n"
+          << "  if (!t3) {\n";
         if (up >= 0) {
-          s << "    " << v << " = VT<WIDE>::mul(v" << ins.a
+          s << "    t2 = VT<WIDE>::mul(v" << ins.a
             << ", VT<WIDE>::fromI64(" << d.insP10[i]
             << "LL, nullptr), &ovf);\n";
         } else {
@@ -180,20 +185,19 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
             << "    if (!WIDE && (sq > (__int128)INT64_MAX || sq < (__int128)INT64_MIN)) {\n"
             << "      atomicOr(d.errorFlag, kErrRetryWide); return false; }\n"
             << "    if (WIDE) { Int128 rr = {(uint64_t)sq, (int64_t)(sq >> 64)}; "
-            << v << " = *(T*)&rr; }\n"
-            << "    else { int64_t qq = (int64_t)sq; " << v
-            << " = *(T*)&qq; }\n";
+               "t2 = *(T*)&rr; }\n"
+            << "    else { int64_t qq = (int64_t)sq; t2 = *(T*)&qq; }\n";
         }
-        s << "  }\n";
+        s << "  }\n  " << v << " = t2; " << nv << " = t3; }\n";
         break;
       }
       case gxp::VM_DIV:
-        s << "  bool " << nv << " = n" << ins.a << " || n" << ins.b << ";\n"
-          << "  T " << v << " = VT<WIDE>::zero();\n"
-          << "  if (!" << nv << ") {\n"
+        s << "  { bool t3 = n" << ins.a << " || n" << ins.b << ";\n"
+          << "  T t2 = VT<WIDE>::zero();\n"
+          << "  if (!t3) {\n"
           << "    Int128 bi = VT<WIDE>::toAcc(v" << ins.b << ");\n"
           << "    __int128 bv = ((__int128)bi.hi << 64) | (__int128)bi.lo;\n"
-          << "    if (bv == 0) " << nv << " = true;\n"
+          << "    if (bv == 0) t3 = true;\n"
           << "    else {\n"
           << "      Int128 ai = VT<WIDE>::toAcc(v" << ins.a << ");\n"
           << "      __int128 av = ((__int128)ai.hi << 64) | (__int128)ai.lo;\n"
@@ -213,10 +217,10 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
           << "      __int128 q = ((av < 0) != (bv < 0)) ? -(__int128)uq : (__int128)uq;\n"
           << "      if (!WIDE && (q > (__int128)INT64_MAX || q < (__int128)INT64_MIN)) {\n"
           << "        atomicOr(d.errorFlag, kErrRetryWide); return false; }\n"
-          << "      if (WIDE) { Int128 r = {(uint64_t)q, (int64_t)(q >> 64)}; " << v
-          << " = *(T*)&r; }\n"
-          << "      else { int64_t qq = (int64_t)q; " << v << " = *(T*)&qq; }\n"
-          << "    }\n  }\n";
+          << "      if (WIDE) { Int128 r = {(uint64_t)q, (int64_t)(q >> 64)}; "
+             "t2 = *(T*)&r; }\n"
+          << "      else { int64_t qq = (int64_t)q; t2 = *(T*)&qq; }\n"
+          << "    }\n  }\n  " << v << " = t2; " << nv << " = t3; }\n";
         break;
     }
   }

@@ -134,7 +134,7 @@ __global__ void genLineitemKernel(DevTable tab, int64_t rowBegin, int64_t nRows,
 
 // per-row pipeline after the raw fetch: filter -> VM -> LDS aggregate.
 // Returns false on a hard failure (error flag already set).
-template <bool WIDE, bool DIVOK, typename RAWT>
+template <bool WIDE, bool DIVOK, typename VMT, typename RAWT>
 __device__ __attribute__((always_inline)) inline bool processRow(const FusedQueryDesc& d, int64_t row,
                                   const RAWT& raw, Lds3GroupSlot* lds,
                                   uint64_t* mySel) {
@@ -169,7 +169,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   if (d.ablate == 2) return true;  // timing ablation: filter only
 
   // ---- projection / agg-arg VM ----
-  VmState<WIDE> vm;
+  VMT vm;
   vm.nullBits = 0;
   bool bad = false;
   bool ovf = false;
@@ -426,7 +426,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   return true;
 }
 
-template <bool WIDE, int R, bool DIVOK = false>
+template <bool WIDE, int R, bool DIVOK = false, int NVM = 12>
 __launch_bounds__(256)
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
@@ -467,10 +467,12 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     for (; row < end && !failed; row += 2 * stride) {
       const int64_t rB = row + stride;
       if (rB < end) fetchRow(d.table, d.fetch, d.nFetch, rB, rawB);
-      if (!processRow<WIDE, DIVOK>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
+      using VMT = typename std::conditional<NVM <= 12, VmState<WIDE>,
+                                            VmState14<WIDE>>::type;
+      if (!processRow<WIDE, DIVOK, VMT>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
       const int64_t rA2 = row + 2 * stride;
       if (rA2 < end) fetchRow(d.table, d.fetch, d.nFetch, rA2, rawA);
-      if (rB < end && !processRow<WIDE, DIVOK>(d, rB, rawB, lds3, &mySel)) failed = true;
+      if (rB < end && !processRow<WIDE, DIVOK, VMT>(d, rB, rawB, lds3, &mySel)) failed = true;
     }
   }
 
@@ -1466,7 +1468,19 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
     for (int f = 0; f < desc.nFetch; f++)
       if (desc.fetch[f].kind != FETCH_B1 && f > maxSlot) maxSlot = f;
     bool small = maxSlot < 5;
-    if (desc.hasDiv) {
+    if (desc.nVmRegs > 12) {
+      // wide-projection plans: the 14-register VM state variant (R=8)
+      if (desc.hasDiv) {
+        if (desc.wide)
+          hipLaunchKernelGGL((fusedAggKernel<true, 8, true, 14>), dim3(grid), dim3(256), 0, s, devDesc);
+        else
+          hipLaunchKernelGGL((fusedAggKernel<false, 8, true, 14>), dim3(grid), dim3(256), 0, s, devDesc);
+      } else if (desc.wide) {
+        hipLaunchKernelGGL((fusedAggKernel<true, 8, false, 14>), dim3(grid), dim3(256), 0, s, devDesc);
+      } else {
+        hipLaunchKernelGGL((fusedAggKernel<false, 8, false, 14>), dim3(grid), dim3(256), 0, s, devDesc);
+      }
+    } else if (desc.hasDiv) {
       if (desc.wide) {
         if (small)
           hipLaunchKernelGGL((fusedAggKernel<true, 5, true>), dim3(grid), dim3(256), 0, s, devDesc);

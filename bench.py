@@ -301,6 +301,77 @@ def bench_sort(args):
     print(json.dumps(out), flush=True)
 
 
+def bench_wide(args):
+    """Wide projection (BASELINE config 5's shape at round-1 VM capacity):
+    15 chained decimal expressions over 4 columns feeding 7 sums + count,
+    through the hipRTC-specialized kernel. One step = one pass over the
+    resident lineitem shard."""
+    import ctypes as C
+    from tests.gxlib import GX_TPCH_LINEITEM, load_product
+    from tests.test_wide_projection import wide_plan
+    lib = load_product()
+    n = args.rows
+    lib.gx_last_kernel_ms.restype = C.c_double
+    lib.gx_last_kernel_ms.argtypes = [C.c_void_p]
+    b, src, agg, out_types, out_fracs = wide_plan(lib)
+    ex = b.build(agg)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, n)
+
+    def step():
+        ex.open()
+        caps = [2048 if t == 4 else None for t in out_types]
+        rows = ex.pull_all(out_types, out_fracs, data_caps=caps)
+        k = lib.gx_last_kernel_ms(ex.ex)
+        ex.close()
+        return rows, k
+
+    for _ in range(args.warmup):
+        step()
+    t0 = time.perf_counter()
+    kms = []
+    for _ in range(args.steps):
+        _, k = step()
+        kms.append(k)
+    elapsed = time.perf_counter() - t0
+    avg_kms = sum(kms) / len(kms)
+    value = n / (avg_kms / 1000.0) if avg_kms else 0
+    bpr = 170  # same resident lineitem shard as Q1
+    achieved = n * bpr / (avg_kms / 1000.0) / 1e9 if avg_kms else 0
+    out = {
+        "metric": "wide_projection_rows_per_sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int128",
+        "data": "synthetic",
+        "config": {
+            "workload": f"wide_projection_15expr_8agg_{n}_rows",
+            "rows": n,
+            "parallelism": "single-gpu",
+            "bytes_per_row": bpr,
+        },
+        "kernel_ms_avg": avg_kms,
+        "roofline": {
+            "bound": "hbm",
+            "achieved": achieved,
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": achieved / HBM_PEAK_GBS,
+            "traffic": None,
+        },
+        "cpu_baseline": None,
+    }
+    ex.free()
+    b.free()
+    print(json.dumps(out), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -308,7 +379,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--rows", type=int, default=SF10_ROWS,
                     help="rows per GPU (default SF10)")
-    ap.add_argument("--query", choices=["q1", "q3", "sort"], default="q1")
+    ap.add_argument("--query", choices=["q1", "q3", "sort", "wide"],
+                    default="q1")
     ap.add_argument("--sf", type=int, default=100,
                     help="scale factor for --query q3 (lineitem = 6M x SF)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -318,6 +390,8 @@ def main():
         return bench_q3(args)
     if args.query == "sort":
         return bench_sort(args)
+    if args.query == "wide":
+        return bench_wide(args)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -182,6 +182,11 @@ struct FusedQueryDesc {
   int32_t nAccSlots = 0;
   int32_t accReg[kMaxAggs];   // phys acc slot -> VM register
   int32_t accMap[kMaxAggs];   // agg index -> phys acc slot (-1 for COUNT)
+  // phys slot accumulation kind: 0 = int128 sum, 1 = max over the
+  // order-preserving biased-u64 encoding (min stores the complement, so
+  // BOTH min and max accumulate with unsigned max from a zero-initialized
+  // table; func_max_min.go semantics, narrow int64 values only)
+  int32_t accKind[kMaxAggs];
   int32_t sharedCnt = 0;
   int32_t hasDiv = 0;
   int32_t nVmRegs = 0;  // registers the compiled VM uses (>12 selects the

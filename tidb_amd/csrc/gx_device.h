@@ -436,6 +436,24 @@ __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {
 typedef __attribute__((address_space(3))) GroupSlot Lds3GroupSlot;
 typedef __attribute__((address_space(3))) uint64_t Lds3U64;
 
+// order-preserving biased encoding for int64 min/max accumulation
+__device__ inline uint64_t biasI64(int64_t v) {
+  return (uint64_t)v ^ 0x8000000000000000ULL;
+}
+__device__ inline int64_t unbiasU64(uint64_t u) {
+  return (int64_t)(u ^ 0x8000000000000000ULL);
+}
+
+__device__ inline void lds3AccumMax(Lds3GroupSlot* slot, int s, uint64_t enc) {
+  __hip_atomic_fetch_max((Lds3U64*)&slot->accLo[s], enc, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+
+template <typename SlotT>
+__device__ inline void accumMax(SlotT* slot, int s, uint64_t enc) {
+  atomicMax((unsigned long long*)&slot->accLo[s], (unsigned long long)enc);
+}
+
 __device__ inline void lds3AccumAcc(Lds3GroupSlot* slot, int s, Int128 v) {
   if (v.lo != 0 || v.hi != 0) {
     uint64_t old = __hip_atomic_fetch_add((Lds3U64*)&slot->accLo[s], (uint64_t)v.lo,

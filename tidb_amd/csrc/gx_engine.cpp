@@ -1196,8 +1196,10 @@ static int32_t compileFused(gx_exec* ex) {
       ex->desc.aggs[ex->desc.nAggs++] = ad;
       continue;
     }
-    if (ad.func != GX_AGG_COUNT && ad.func != GX_AGG_SUM && ad.func != GX_AGG_AVG) {
-      ex->err = "device aggregation supports count/sum/avg this round";
+    if (ad.func != GX_AGG_COUNT && ad.func != GX_AGG_SUM &&
+        ad.func != GX_AGG_AVG && ad.func != GX_AGG_MIN &&
+        ad.func != GX_AGG_MAX) {
+      ex->err = "device aggregation supports count/sum/avg/min/max this round";
       return GX_ERR_INVALID;
     }
     int argE = agg->aggArgs[a];
@@ -1233,12 +1235,26 @@ static int32_t compileFused(gx_exec* ex) {
       ex->desc.accMap[a] = -1;
       continue;
     }
+    // MIN stores the complemented bias so both extremes accumulate with
+    // unsigned max; kind distinguishes the three accumulator behaviors
+    int kind = ad.func == GX_AGG_MAX ? 1 : (ad.func == GX_AGG_MIN ? 2 : 0);
+    if (kind != 0) {
+      int argE2 = agg->aggArgs[a];
+      if (argE2 < 0 || plan.exprs[argE2].retType != GX_TYPE_DECIMAL) {
+        ex->err = "device min/max supports decimal arguments this round";
+        return GX_ERR_INVALID;
+      }
+    }
     int found = -1;
     for (int s = 0; s < ex->desc.nAccSlots; s++)
-      if (ex->desc.accReg[s] == ad.srcReg) { found = s; break; }
+      if (ex->desc.accReg[s] == ad.srcReg && ex->desc.accKind[s] == kind) {
+        found = s;
+        break;
+      }
     if (found < 0) {
       found = ex->desc.nAccSlots++;
       ex->desc.accReg[found] = ad.srcReg;
+      ex->desc.accKind[found] = kind;
     }
     ex->desc.accMap[a] = found;
   }
@@ -1617,6 +1633,8 @@ static int32_t materializeDevice(gx_exec* ex) {
     // opt-in until the pipelining wins back the staging overhead.
     bool ok = tab.nRows >= 256 && getenv("GX_GLDS") && !ex->vmHasDiv &&
               ex->vmNextReg <= 12;  // the staged kernel's VmState is 12-reg
+    for (int s = 0; s < d.nAccSlots && ok; s++)
+      ok = d.accKind[s] == 0;  // staged accumulate is sum-only
     for (int f = 0; f < d.nFetch && ok; f++)
       ok = d.fetch[f].kind == gxp::FETCH_8B ||
            d.fetch[f].kind == gxp::FETCH_DEC16 ||
@@ -1901,6 +1919,22 @@ static int32_t runFused(gx_exec* ex) {
         OutRowVal v;
         v.type = GX_TYPE_I64;
         v.i64 = cnt;
+        row.push_back(std::move(v));
+      } else if (ad.func == GX_AGG_MIN || ad.func == GX_AGG_MAX) {
+        // biased-u64 extreme (accKind 1/2); NULL when no non-null arg row
+        OutRowVal v;
+        uint64_t raw = s->accLo[phys];
+        if (ex->desc.accKind[phys] == 2) raw = ~raw;
+        int64_t ext = (int64_t)(raw ^ 0x8000000000000000ULL);
+        int srcType = GX_TYPE_DECIMAL;
+        // arg type: decimal scale > 0 or i64 (scale 0 decimal still decimal)
+        if (cnt == 0) {
+          v.type = srcType;
+          v.isNull = true;
+        } else {
+          v.type = GX_TYPE_DECIMAL;
+          v.dec = decFromUnits((__int128)ext, ad.scale);
+        }
         row.push_back(std::move(v));
       } else if (partial) {
         OutRowVal v;

@@ -269,9 +269,22 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
        "    }\n"
        "    Lds3GroupSlot* t = &lds[slot];\n";
   if (d.sharedCnt) s << "    lds3AccumCnt(t, 0, 1);\n";
-  for (int sIdx = 0; sIdx < d.nAccSlots; sIdx++)
-    s << "    if (!n" << d.accReg[sIdx] << ") lds3AccumAcc(t, " << sIdx
-      << ", VT<WIDE>::toAcc(v" << d.accReg[sIdx] << "));\n";
+  for (int sIdx = 0; sIdx < d.nAccSlots; sIdx++) {
+    if (d.accKind[sIdx] == 0) {
+      s << "    if (!n" << d.accReg[sIdx] << ") lds3AccumAcc(t, " << sIdx
+        << ", VT<WIDE>::toAcc(v" << d.accReg[sIdx] << "));\n";
+    } else {
+      s << "    if (!n" << d.accReg[sIdx] << ") { Int128 mv = "
+           "VT<WIDE>::toAcc(v" << d.accReg[sIdx] << ");\n"
+        << "      bool fits = (mv.hi == 0 && (int64_t)mv.lo >= 0) || "
+           "(mv.hi == -1 && (int64_t)mv.lo < 0);\n"
+        << "      if (WIDE && !fits) { atomicOr(d.errorFlag, kErrOverflow); "
+           "return false; }\n"
+        << "      uint64_t enc = biasI64((int64_t)mv.lo);\n"
+        << "      lds3AccumMax(t, " << sIdx << ", "
+        << (d.accKind[sIdx] == 2 ? "~enc" : "enc") << "); }\n";
+    }
+  }
   if (!d.sharedCnt) {
     for (int a = 0; a < d.nAggs; a++) {
       const gxp::AggDesc& ad = d.aggs[a];
@@ -299,9 +312,22 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
        "    }\n"
        "    GroupSlot* t = &d.globalTable[slot];\n";
   if (d.sharedCnt) s << "    accumInto(t, 0, Int128{0, 0}, 1);\n";
-  for (int sIdx = 0; sIdx < d.nAccSlots; sIdx++)
-    s << "    if (!n" << d.accReg[sIdx] << ") accumInto(t, " << sIdx
-      << ", VT<WIDE>::toAcc(v" << d.accReg[sIdx] << "), 0);\n";
+  for (int sIdx = 0; sIdx < d.nAccSlots; sIdx++) {
+    if (d.accKind[sIdx] == 0) {
+      s << "    if (!n" << d.accReg[sIdx] << ") accumInto(t, " << sIdx
+        << ", VT<WIDE>::toAcc(v" << d.accReg[sIdx] << "), 0);\n";
+    } else {
+      s << "    if (!n" << d.accReg[sIdx] << ") { Int128 mv = "
+           "VT<WIDE>::toAcc(v" << d.accReg[sIdx] << ");\n"
+        << "      bool fits = (mv.hi == 0 && (int64_t)mv.lo >= 0) || "
+           "(mv.hi == -1 && (int64_t)mv.lo < 0);\n"
+        << "      if (WIDE && !fits) { atomicOr(d.errorFlag, kErrOverflow); "
+           "return false; }\n"
+        << "      uint64_t enc = biasI64((int64_t)mv.lo);\n"
+        << "      accumMax(t, " << sIdx << ", "
+        << (d.accKind[sIdx] == 2 ? "~enc" : "enc") << "); }\n";
+    }
+  }
   if (!d.sharedCnt) {
     for (int a = 0; a < d.nAggs; a++) {
       const gxp::AggDesc& ad = d.aggs[a];
@@ -441,6 +467,10 @@ __device__ __forceinline__ void kernBody(const FusedQueryDesc* __restrict__ dp) 
     }
     if (!ok) continue;
     for (int s2 = 0; s2 < d.nAccSlots; s2++) {
+      if (d.accKind[s2] != 0) {
+        accumMax(&d.globalTable[slot], s2, lds[i].accLo[s2]);
+        continue;
+      }
       Int128 v = {lds[i].accLo[s2], lds[i].accHi[s2]};
       accumInto(&d.globalTable[slot], s2, v, 0);
     }

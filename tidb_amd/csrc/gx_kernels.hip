@@ -380,7 +380,20 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     if (d.sharedCnt) lds3AccumCnt(target, 0, 1);
     for (int s = 0; s < d.nAccSlots; s++) {
       int reg = d.accReg[s];
-      if (!vm.isNull(reg)) lds3AccumAcc(target, s, VT<WIDE>::toAcc(vm.get(reg)));
+      if (vm.isNull(reg)) continue;
+      if (d.accKind[s] == 0) {
+        lds3AccumAcc(target, s, VT<WIDE>::toAcc(vm.get(reg)));
+      } else {
+        Int128 v = VT<WIDE>::toAcc(vm.get(reg));
+        bool fits = (v.hi == 0 && (int64_t)v.lo >= 0) ||
+                    (v.hi == -1 && (int64_t)v.lo < 0);
+        if (WIDE && !fits) {
+          atomicOr(d.errorFlag, kErrOverflow);  // >64-bit min/max value
+          return false;
+        }
+        uint64_t enc = biasI64((int64_t)v.lo);
+        lds3AccumMax(target, s, d.accKind[s] == 2 ? ~enc : enc);
+      }
     }
     if (!d.sharedCnt) {
       for (int a = 0; a < d.nAggs; a++) {
@@ -413,7 +426,20 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   if (d.sharedCnt) accumInto(target, 0, Int128{0, 0}, 1);  // bumps cnt[0] only
   for (int s = 0; s < d.nAccSlots; s++) {
     int reg = d.accReg[s];
-    if (!vm.isNull(reg)) accumInto(target, s, VT<WIDE>::toAcc(vm.get(reg)), 0);
+    if (vm.isNull(reg)) continue;
+    if (d.accKind[s] == 0) {
+      accumInto(target, s, VT<WIDE>::toAcc(vm.get(reg)), 0);
+    } else {
+      Int128 v = VT<WIDE>::toAcc(vm.get(reg));
+      bool fits = (v.hi == 0 && (int64_t)v.lo >= 0) ||
+                  (v.hi == -1 && (int64_t)v.lo < 0);
+      if (WIDE && !fits) {
+        atomicOr(d.errorFlag, kErrOverflow);
+        return false;
+      }
+      uint64_t enc = biasI64((int64_t)v.lo);
+      accumMax(target, s, d.accKind[s] == 2 ? ~enc : enc);
+    }
   }
   if (!d.sharedCnt) {
     for (int a = 0; a < d.nAggs; a++) {
@@ -506,6 +532,10 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     }
     if (!ok) continue;
     for (int s = 0; s < d.nAccSlots; s++) {
+      if (d.accKind[s] != 0) {
+        accumMax(&d.globalTable[slot], s, lds[i].accLo[s]);
+        continue;
+      }
       Int128 v = {lds[i].accLo[s], lds[i].accHi[s]};
       accumInto(&d.globalTable[slot], s, v, 0);
     }
@@ -1321,6 +1351,10 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
     }
     if (!ok) continue;
     for (int s = 0; s < d.nAccSlots; s++) {
+      if (d.accKind[s] != 0) {
+        accumMax(&d.globalTable[slot], s, lds[i].accLo[s]);
+        continue;
+      }
       Int128 v = {lds[i].accLo[s], lds[i].accHi[s]};
       accumInto(&d.globalTable[slot], s, v, 0);
     }

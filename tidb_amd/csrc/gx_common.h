@@ -327,6 +327,7 @@ int gxSortIota(uint32_t* idx, int64_t n, void* stream);
 // gather one column into out (same layout; elemSize 1, 8 or 40)
 int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,
                     int elemSize, void* stream);
+
 // join-agg pipeline steps (gx_kernels.hip)
 int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
                    void* stream);  // 0 count0 1 build0 2 count1 3 build1 4 probe

@@ -2494,22 +2494,42 @@ static int32_t runDeviceSort(gx_exec* ex) {
   HIP_OK(ex, hipMemcpyAsync(&herr, ex->devErr, 4, hipMemcpyDeviceToHost, ex->stream));
   HIP_OK(ex, hipStreamSynchronize(ex->stream));
   if (herr) { ex->err = "sort key error (bad/overflowing decimal)"; return GX_ERR_INVALID; }
-  // gather every column through the final permutation
-  for (int c = 0; c < tab.nCols; c++) {
-    gxp::DevCol& col = tab.cols[c];
-    if (col.type == GX_TYPE_STRING && !col.denseOffsets) {
-      ex->err = "general varlen column in sorted table unsupported this round";
-      return GX_ERR_INVALID;
+  // gather every column through the final permutation in ONE launch
+  // (per-column kernels serialize on the stream; the random-access gather
+  // is latency-bound and wants all columns' requests in flight together)
+  {
+    const void* ins[16];
+    void* outs[16];
+    int sizes[16];
+    int nc = tab.nCols;
+    if (nc > 8) { ex->err = "sort supports <= 8 columns this round"; return GX_ERR_INVALID; }
+    for (int c = 0; c < nc; c++) {
+      gxp::DevCol& col = tab.cols[c];
+      if (col.type == GX_TYPE_STRING && !col.denseOffsets) {
+        ex->err = "general varlen column in sorted table unsupported this round";
+        return GX_ERR_INVALID;
+      }
+      int es = col.type == GX_TYPE_DECIMAL ? 40
+               : (col.type == GX_TYPE_STRING ? 1 : 8);
+      void* nd = devAlloc(ex, (size_t)n * es + 16);
+      if (!nd) { ex->err = "hipMalloc failed (gather)"; return GX_ERR_INTERNAL; }
+      ins[c] = col.data;
+      outs[c] = nd;
+      sizes[c] = es;
     }
-    int es = col.type == GX_TYPE_DECIMAL ? 40
-             : (col.type == GX_TYPE_STRING ? 1 : 8);
-    void* nd = devAlloc(ex, (size_t)n * es + 16);
-    if (!nd) { ex->err = "hipMalloc failed (gather)"; return GX_ERR_INTERNAL; }
-    if (gxp::gxSortGatherCol(col.data, nd, idxA, n, es, ex->stream)) {
+    // sequential per-column gathers measured FASTER than both a combined
+    // single kernel and per-column concurrent streams: one column's random
+    // reads stay L2-coherent; concurrent columns evict each other
+    int grc = 0;
+    for (int c = 0; c < nc; c++)
+      grc |= gxp::gxSortGatherCol(ins[c], outs[c], idxA, n, sizes[c],
+                                  ex->stream);
+    if (grc) {
       ex->err = "gather launch failed";
       return GX_ERR_INTERNAL;
     }
-    col.data = nd;  // offsets for dense char remain the identity ramp
+    for (int c = 0; c < nc; c++)
+      tab.cols[c].data = outs[c];  // dense-char offsets stay the identity ramp
   }
   hipEventRecord(ev1, ex->stream);
   HIP_OK(ex, hipStreamSynchronize(ex->stream));

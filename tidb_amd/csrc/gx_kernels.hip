@@ -1725,6 +1725,8 @@ __global__ void sortGatherKernel(const uint8_t* __restrict__ in,
   }
 }
 
+
+
 int gxSortIota(uint32_t* idx, int64_t n, void* stream) {
   hipLaunchKernelGGL(sortIotaKernel, dim3(gridFor(n)), dim3(256), 0,
                      (hipStream_t)stream, idx, n);

@@ -978,6 +978,8 @@ class HashJoinExec : public Exec {
   }
   int32_t open() override {
     built_ = false;
+    pending_.reset();
+    pendEmit_ = 0;
     int32_t ec = build_->open();
     if (ec) return ec;
     return probe_->open();
@@ -997,11 +999,27 @@ class HashJoinExec : public Exec {
     }
     size_t nb = build_->outTypes.size();
     for (;;) {
+      // drain matches pending from the previous probe chunk first: one Next
+      // fills at most MaxChunkSize rows (exec.Executor contract,
+      // executor.go:224-250)
+      while (pendEmit_ < pending_.numRows() && out.numRows() < kMaxChunkSize) {
+        for (size_t c = 0; c < out.cols.size(); c++)
+          out.cols[c].appendFrom(pending_.cols[c], pendEmit_);
+        pendEmit_++;
+      }
+      if (out.numRows() >= kMaxChunkSize) return GX_OK;
       Chunk in;
       int32_t ec = probe_->next(in);
       if (ec) { err = probe_->err; return ec; }
       int n = in.numRows();
-      if (n == 0) return GX_OK;
+      if (n == 0) return GX_OK;  // probe exhausted: emit what we have (may be 0 = EOF)
+      pending_.cols.resize(outTypes.size());
+      for (size_t c = 0; c < pending_.cols.size(); c++) {
+        pending_.cols[c].type = outTypes[c];
+        pending_.cols[c].frac = outFracs[c];
+      }
+      pending_.reset();
+      pendEmit_ = 0;
       std::vector<std::string> keys(n);
       std::vector<uint8_t> hasNullKey(n, 0);
       ec = serializeJoinKeys(ctx, node_.probeKeys, in, keys, hasNullKey);
@@ -1012,12 +1030,11 @@ class HashJoinExec : public Exec {
         if (it == table_.end()) continue;
         for (const RowRef& br : it->second) {
           for (size_t c = 0; c < nb; c++)
-            out.cols[c].appendFrom(buildData_[br.chunkIdx].cols[c], br.rowIdx);
+            pending_.cols[c].appendFrom(buildData_[br.chunkIdx].cols[c], br.rowIdx);
           for (size_t c = 0; c < probe_->outTypes.size(); c++)
-            out.cols[nb + c].appendFrom(in.cols[c], i);
+            pending_.cols[nb + c].appendFrom(in.cols[c], i);
         }
       }
-      if (out.numRows() > 0) return GX_OK;
     }
   }
   int32_t close() override {
@@ -1102,6 +1119,8 @@ class HashJoinExec : public Exec {
   bool built_ = false;
   std::vector<Chunk> buildData_;
   std::unordered_map<std::string, std::vector<RowRef>> table_;
+  Chunk pending_;      // matches of the current probe chunk, emitted <=
+  int pendEmit_ = 0;   // kMaxChunkSize per next()
 };
 
 }  // namespace

@@ -285,6 +285,47 @@ struct JoinAggDesc {
 
 enum { PRED_STR_EQ_CONST = 3 };  // extra PredKind for the join path
 
+// ---- standalone hash join (inner, duplicate build keys) ----
+// HashJoinV2 equivalent (join/hash_join_v2.go): chained hash table over the
+// build side — heads + intrusive per-row next links (hash_table_v2.go:22-53
+// tagged-ptr chains / join_row_table.go next_row_ptr), lock-free CAS head
+// insert (hash_table_v2.go:94-105), chain-walk probe comparing keys
+// (base_join_probe.go:289-331), output = build cols ++ probe cols, one row
+// per matching pair, NULL keys never match (inner_join_probe.go:27-86).
+// MI355X shape: the "row table" is the resident columnar build table itself —
+// no serialized row copy; output columns gather straight from HBM through the
+// match-pair index, so each build/probe byte is read once per emitted value.
+// Output order is unspecified (the reference's concurrent probe workers make
+// its join output order nondeterministic too).
+struct HashJoinDesc {
+  DevTable build, probe;
+  // one filter conjunct per side this round (SelectionExec pushed into the
+  // build/probe scans)
+  PredDesc predB, predP;
+  int32_t nPredB = 0, nPredP = 0;
+  uint8_t strConstB[16];
+  int32_t strConstBLen = 0;
+  uint8_t strConstP[16];
+  int32_t strConstPLen = 0;
+  int32_t bKeyCol = 0, pKeyCol = 0;  // single int64 key column per side
+  uint32_t* heads = nullptr;  // 1<<headsLog2 entries: build row+1, 0 = empty
+  int32_t headsLog2 = 0;
+  uint32_t* next = nullptr;   // per build row: next chain row+1, 0 = end
+  uint32_t* outBuild = nullptr;  // match pairs (fill phase)
+  uint32_t* outProbe = nullptr;
+  uint64_t* counters = nullptr;  // [0] count-phase total, [1] fill cursor
+  uint32_t* errorFlag = nullptr;
+};
+
+// phases: 0 = build (chain insert), 1 = count matches, 2 = fill match pairs
+int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
+                    const HashJoinDesc& h, void* stream);
+// gather a null bitmap through the match index (one thread per output byte)
+int gxGatherNulls(const uint8_t* inBitmap, const uint32_t* idx, uint8_t* out,
+                  int64_t n, void* stream);
+// identity offsets ramp for gathered dense-char columns
+int gxIotaOffsets(int64_t* p, int64_t n, void* stream);
+
 // top-N selection scratch
 struct TopNOut {
   uint64_t key, payload0;

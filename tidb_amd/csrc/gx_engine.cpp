@@ -177,6 +177,12 @@ struct gx_exec {
   std::vector<std::pair<int, bool>> jaSortKeys;
   int64_t jaLimit = 0, jaOffset = 0;
 
+  // standalone hash join (inner, duplicate build keys) state
+  bool isHashJoin = false;
+  gxp::HashJoinDesc hj;
+  gxp::HashJoinDesc* devHj = nullptr;
+  int hjSrcB = -1, hjSrcP = -1;
+
   ~gx_exec() {
     for (void* p : devBufs) hipFree(p);
   }
@@ -970,6 +976,80 @@ static int32_t compileJoinAgg(gx_exec* ex) {
   ex->jaSrcOrd = srcO;
   ex->jaSrcLi = srcL;
   ex->isJoinAgg = true;
+  return GX_OK;
+}
+
+// compile a STANDALONE inner hash join (root = HashJoin over [Selection ->]
+// Source children): the general HashJoinV2 operator (join/hash_join_v2.go)
+// with duplicate build keys via chained table — output is the joined rows
+// themselves (build cols ++ probe cols), not an aggregate.
+static int32_t compileHashJoin(gx_exec* ex) {
+  const PPlan& plan = ex->plan;
+  const PNode& jn = plan.nodes[ex->root];
+  if (jn.joinType != 0) {
+    ex->err = "only inner joins on device this round";
+    return GX_ERR_INVALID;
+  }
+  if (jn.buildKeys.size() != 1 || jn.probeKeys.size() != 1) {
+    ex->err = "device hash join supports one key column this round";
+    return GX_ERR_INVALID;
+  }
+  const PNode *selB, *selP;
+  int srcB = unwrapSource(ex, jn.child, &selB);
+  int srcP = unwrapSource(ex, jn.child2, &selP);
+  if (srcB < 0 || srcP < 0) {
+    ex->err = "join children must be [Selection ->] Source";
+    return GX_ERR_INVALID;
+  }
+  const PNode& bN = plan.nodes[srcB];
+  const PNode& pN = plan.nodes[srcP];
+  int nb = (int)bN.colTypes.size();
+  int np = (int)pN.colTypes.size();
+  if (nb + np > gxp::kMaxCols) {
+    ex->err = "too many join output columns";
+    return GX_ERR_INVALID;
+  }
+  const PExpr& bk = plan.exprs[jn.buildKeys[0]];
+  const PExpr& pk = plan.exprs[jn.probeKeys[0]];
+  if (bk.kind != EK_COLREF || pk.kind != EK_COLREF || bk.colIdx < 0 ||
+      bk.colIdx >= nb || pk.colIdx < 0 || pk.colIdx >= np) {
+    ex->err = "join keys must be child columns";
+    return GX_ERR_INVALID;
+  }
+  if (bN.colTypes[bk.colIdx] != GX_TYPE_I64 ||
+      pN.colTypes[pk.colIdx] != GX_TYPE_I64) {
+    ex->err = "device join keys must be int64 this round";
+    return GX_ERR_INVALID;
+  }
+  gxp::HashJoinDesc& hj = ex->hj;
+  hj.bKeyCol = bk.colIdx;
+  hj.pKeyCol = pk.colIdx;
+  auto doPred = [&](const PNode* sel, const PNode& srcNode, gxp::PredDesc* pd,
+                    int32_t* n, uint8_t* sc, int32_t* scLen) -> bool {
+    *n = 0;
+    if (!sel) return true;
+    if (sel->exprs.size() != 1) {
+      ex->err = "device join path supports one filter conjunct per table";
+      return false;
+    }
+    if (!compileTablePred(ex, srcNode, sel->exprs[0], pd, sc, scLen))
+      return false;
+    *n = 1;
+    return true;
+  };
+  if (!doPred(selB, bN, &hj.predB, &hj.nPredB, hj.strConstB, &hj.strConstBLen))
+    return GX_ERR_INVALID;
+  if (!doPred(selP, pN, &hj.predP, &hj.nPredP, hj.strConstP, &hj.strConstPLen))
+    return GX_ERR_INVALID;
+  ex->hjSrcB = srcB;
+  ex->hjSrcP = srcP;
+  // output schema: build cols ++ probe cols (inner_join_probe.go:27-86)
+  ex->desc.table.nCols = nb + np;
+  for (int c = 0; c < nb; c++)
+    setDevColMeta(&ex->desc.table.cols[c], bN.colTypes[c], bN.colFracs[c]);
+  for (int c = 0; c < np; c++)
+    setDevColMeta(&ex->desc.table.cols[nb + c], pN.colTypes[c], pN.colFracs[c]);
+  ex->isHashJoin = true;
   return GX_OK;
 }
 
@@ -2298,6 +2378,147 @@ static int32_t runJoinAgg(gx_exec* ex) {
   return GX_OK;
 }
 
+// ---------------- standalone hash join execution ----------------
+
+static int32_t runHashJoin(gx_exec* ex) {
+  gxp::HashJoinDesc& hj = ex->hj;
+  if (!ex->deviceReady) {
+    if (!gpuAvailable()) {
+      ex->err = "no MI355X visible: the product engine has no CPU fallback "
+                "(GX_ERR_NO_GPU)";
+      return GX_ERR_NO_GPU;
+    }
+    if (ex->device >= 0) hipSetDevice(ex->device);
+    HIP_OK(ex, hipStreamCreate(&ex->stream));
+    int32_t rc = materializeTable(ex, ex->hjSrcB, &hj.build);
+    if (rc) return rc;
+    rc = materializeTable(ex, ex->hjSrcP, &hj.probe);
+    if (rc) return rc;
+    ex->devErr = (uint32_t*)devAlloc(ex, 4);
+    hj.counters = (uint64_t*)devAlloc(ex, 2 * 8);
+    ex->devHj = (gxp::HashJoinDesc*)devAlloc(ex, sizeof(gxp::HashJoinDesc));
+    if (!ex->devErr || !hj.counters || !ex->devHj) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+    hj.errorFlag = ex->devErr;
+    ex->deviceReady = true;
+  }
+  int64_t nb = hj.build.nRows;
+  if (nb >= 0xFFFFFFFFLL || hj.probe.nRows > 0xFFFFFFFFLL) {
+    ex->err = "join sides > 2^32 rows unsupported this round";
+    return GX_ERR_INVALID;
+  }
+  // gathered output supports fixed-width and dense char(1) columns only
+  for (int c = 0; c < ex->desc.table.nCols; c++) {
+    int nbc = hj.build.nCols;
+    const gxp::DevCol& src =
+        c < nbc ? hj.build.cols[c] : hj.probe.cols[c - nbc];
+    if (src.type == GX_TYPE_STRING && !src.denseOffsets) {
+      ex->err = "general varlen join output column unsupported this round";
+      return GX_ERR_INVALID;
+    }
+  }
+  hipEvent_t ev0, ev1;
+  HIP_OK(ex, hipEventCreate(&ev0));
+  HIP_OK(ex, hipEventCreate(&ev1));
+  HIP_OK(ex, hipEventRecord(ev0, ex->stream));
+  // chained table: heads (2x rows, pow2) + per-row next links
+  hj.headsLog2 = ceilLog2(std::max<uint64_t>(2 * (uint64_t)nb + 1, 64));
+  hj.heads = (uint32_t*)devAlloc(ex, (1ULL << hj.headsLog2) * 4);
+  hj.next = (uint32_t*)devAlloc(ex, std::max<int64_t>(nb, 1) * 4);
+  if (!hj.heads || !hj.next) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+  HIP_OK(ex, hipMemsetAsync(hj.heads, 0, (1ULL << hj.headsLog2) * 4, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(hj.counters, 0, 16, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+  HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj), hipMemcpyHostToDevice,
+                            ex->stream));
+  if (gxp::gxHashJoinPhase(0, ex->devHj, hj, ex->stream) != 0 ||
+      gxp::gxHashJoinPhase(1, ex->devHj, hj, ex->stream) != 0) {
+    ex->err = "join kernel launch failed";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  uint64_t total = 0;
+  HIP_OK(ex, hipMemcpy(&total, hj.counters, 8, hipMemcpyDeviceToHost));
+  uint32_t errFlag = 0;
+  HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+  if (errFlag != 0) {
+    ex->err = "device join error flag 0x" + std::to_string(errFlag);
+    return GX_ERR_INTERNAL;
+  }
+  if (total > 0x7FFFFFFFULL) {
+    ex->err = "join output > 2^31 rows unsupported this round";
+    return GX_ERR_INVALID;
+  }
+  ex->lastSelCount = total;
+  if (total > 0) {
+    hj.outBuild = (uint32_t*)devAlloc(ex, total * 4);
+    hj.outProbe = (uint32_t*)devAlloc(ex, total * 4);
+    if (!hj.outBuild || !hj.outProbe) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+    HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj),
+                              hipMemcpyHostToDevice, ex->stream));
+    if (gxp::gxHashJoinPhase(2, ex->devHj, hj, ex->stream) != 0) {
+      ex->err = "join fill launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    // gather every output column through its side's match index
+    for (int c = 0; c < ex->desc.table.nCols; c++) {
+      int nbc = hj.build.nCols;
+      const gxp::DevCol& src =
+          c < nbc ? hj.build.cols[c] : hj.probe.cols[c - nbc];
+      const uint32_t* idx = c < nbc ? hj.outBuild : hj.outProbe;
+      gxp::DevCol& dst = ex->desc.table.cols[c];
+      int es = src.type == GX_TYPE_DECIMAL ? 40
+               : (src.type == GX_TYPE_STRING ? 1 : 8);
+      dst.data = devAlloc(ex, (size_t)total * es + 16);
+      if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      if (gxp::gxSortGatherCol(src.data, dst.data, idx, (int64_t)total, es,
+                               ex->stream) != 0) {
+        ex->err = "join gather launch failed";
+        return GX_ERR_INTERNAL;
+      }
+      if (src.type == GX_TYPE_STRING) {
+        dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
+        if (!dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        if (gxp::gxIotaOffsets(dst.offsets, (int64_t)total + 1, ex->stream) != 0) {
+          ex->err = "join offsets launch failed";
+          return GX_ERR_INTERNAL;
+        }
+        dst.denseOffsets = 1;
+      }
+      if (src.hasNulls && src.nullBitmap) {
+        dst.nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8);
+        if (!dst.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        if (gxp::gxGatherNulls(src.nullBitmap, idx, dst.nullBitmap,
+                               (int64_t)total, ex->stream) != 0) {
+          ex->err = "join null gather launch failed";
+          return GX_ERR_INTERNAL;
+        }
+        dst.hasNulls = 1;
+      } else {
+        dst.nullBitmap = nullptr;
+        dst.hasNulls = 0;
+      }
+    }
+  }
+  HIP_OK(ex, hipEventRecord(ev1, ex->stream));
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  {
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    ex->lastKernelMs = ms;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+  }
+  ex->desc.table.nRows = (int64_t)total;
+  ex->srcPos = 0;
+  return GX_OK;
+}
+
 // ---------------- FINAL-mode host merge ----------------
 
 static int32_t runFinalHost(gx_exec* ex) {
@@ -2546,13 +2767,9 @@ static int32_t runDeviceSort(gx_exec* ex) {
   return GX_OK;
 }
 
-static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
-  int32_t rc = materializeDevice(ex);
-  if (rc) return rc;
-  if (!ex->devSortKeys.empty() && !ex->devSorted) {
-    rc = runDeviceSort(ex);
-    if (rc) return rc;
-  }
+// emit the next <=1024 rows of the device-resident table in ex->desc.table
+// (bare/sorted sources and joined-row output)
+static int32_t emitTableChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   gxp::DevTable& tab = ex->desc.table;
   int64_t remaining = tab.nRows - ex->srcPos;
   int n = (int)std::min<int64_t>(remaining, 1024);
@@ -2589,13 +2806,31 @@ static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
       HIP_OK(ex, hipMemcpy(g->data, (uint8_t*)col.data + ex->srcPos * col.elemSize,
                            bytes, hipMemcpyDeviceToHost));
     }
-    if (g->null_bitmap) std::memset(g->null_bitmap, 0xFF, (n + 7) / 8);
+    if (g->null_bitmap) {
+      // srcPos advances in 1024-row steps, so the bitmap slice is byte-aligned
+      if (col.hasNulls && col.nullBitmap) {
+        HIP_OK(ex, hipMemcpy(g->null_bitmap, col.nullBitmap + ex->srcPos / 8,
+                             (n + 7) / 8, hipMemcpyDeviceToHost));
+      } else {
+        std::memset(g->null_bitmap, 0xFF, (n + 7) / 8);
+      }
+    }
     g->length = n;
   }
   out->n_rows = n;
   ex->srcPos += n;
   *rows_out = n;
   return GX_OK;
+}
+
+static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
+  int32_t rc = materializeDevice(ex);
+  if (rc) return rc;
+  if (!ex->devSortKeys.empty() && !ex->devSorted) {
+    rc = runDeviceSort(ex);
+    if (rc) return rc;
+  }
+  return emitTableChunk(ex, out, rows_out);
 }
 
 // ---------------- result marshalling ----------------
@@ -3009,6 +3244,10 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     ex->desc.table.nCols = (int)rn.colTypes.size();
     for (size_t c = 0; c < rn.colTypes.size(); c++)
       setDevColMeta(&ex->desc.table.cols[c], rn.colTypes[c], rn.colFracs[c]);
+  } else if (rn.kind == PK_HASHJOIN) {
+    int32_t rc = compileHashJoin(ex);
+    if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";
+    (void)rc;
   } else {
     ex->err = "unsupported root plan node for the device engine this round";
   }
@@ -3075,7 +3314,8 @@ int32_t gx_bind_tpch(gx_exec* ex, int32_t source_node, int32_t table,
 int32_t gx_open(gx_exec* ex) {
   if (!ex) return GX_ERR_INVALID;
   if (!ex->err.empty()) return GX_ERR_INVALID;
-  if (!ex->isFused && !ex->isFinalHost && !ex->isBareSource && !ex->isJoinAgg) {
+  if (!ex->isFused && !ex->isFinalHost && !ex->isBareSource && !ex->isJoinAgg &&
+      !ex->isHashJoin) {
     ex->err = "plan not executable";
     return GX_ERR_INVALID;
   }
@@ -3090,6 +3330,17 @@ int32_t gx_open(gx_exec* ex) {
 int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   if (!ex || !ex->opened) return GX_ERR_INVALID;
   if (ex->isBareSource) return emitSourceChunk(ex, out, rows_out);
+  if (ex->isHashJoin) {
+    if (!ex->ranQuery) {
+      int32_t rc = runHashJoin(ex);
+      if (rc) {
+        *rows_out = 0;
+        return rc;
+      }
+      ex->ranQuery = true;
+    }
+    return emitTableChunk(ex, out, rows_out);
+  }
   if (!ex->ranQuery) {
     int32_t rc = ex->isFinalHost ? runFinalHost(ex)
                  : ex->isJoinAgg ? runJoinAgg(ex)

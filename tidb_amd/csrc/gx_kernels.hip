@@ -1706,6 +1706,111 @@ __global__ void sortComposeKeysKernel(DevTable tab, SortKeyCompose k,
   }
 }
 
+// ==================================================================
+// standalone hash join (inner, duplicate build keys) — HashJoinDesc
+// ==================================================================
+
+// chain-insert every qualifying build row (lock-free head CAS; the next[]
+// write is published to the probe kernels by the dispatch boundary)
+__global__ void hjBuildKernel(const HashJoinDesc* __restrict__ dp) {
+  const HashJoinDesc& d = *dp;
+  int64_t n = d.build.nRows;
+  uint32_t mask = (1u << d.headsLog2) - 1;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool pass = d.nPredB == 0 ||
+                evalSimplePred(d.build, d.predB, d.strConstB, d.strConstBLen, row);
+    if (!pass) continue;
+    const DevCol& kc = d.build.cols[d.bKeyCol];
+    if (colIsNull(kc, row)) continue;  // NULL never joins
+    uint64_t key = gptr<uint64_t>(kc.data)[row];
+    uint32_t slot = (uint32_t)(hashKey(key) & mask);
+    uint32_t newHead = (uint32_t)row + 1;
+    uint32_t old = d.heads[slot];
+    for (;;) {
+      d.next[row] = old;
+      uint32_t prev = atomicCAS(&d.heads[slot], old, newHead);
+      if (prev == old) break;
+      old = prev;
+    }
+  }
+}
+
+// probe: walk the chain comparing build keys loaded straight from the
+// resident build column. FILL=false counts (wavefront-reduced into
+// counters[0]); FILL=true reserves a contiguous range per probe row via one
+// atomic on counters[1] and writes the (build,probe) match pairs.
+template <bool FILL>
+__global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
+  const HashJoinDesc& d = *dp;
+  int64_t n = d.probe.nRows;
+  uint32_t mask = (1u << d.headsLog2) - 1;
+  const DevCol& bk = d.build.cols[d.bKeyCol];
+  uint64_t my = 0;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool pass = d.nPredP == 0 ||
+                evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen, row);
+    if (!pass) continue;
+    const DevCol& kc = d.probe.cols[d.pKeyCol];
+    if (colIsNull(kc, row)) continue;
+    uint64_t key = gptr<uint64_t>(kc.data)[row];
+    uint32_t head = gptr<uint32_t>(d.heads)[(uint32_t)(hashKey(key) & mask)];
+    uint32_t cnt = 0;
+    for (uint32_t cur = head; cur != 0;) {
+      uint32_t brow = cur - 1;
+      if (gptr<uint64_t>(bk.data)[brow] == key) cnt++;
+      cur = gptr<uint32_t>(d.next)[brow];
+    }
+    if (cnt == 0) continue;
+    if (FILL) {
+      uint64_t base = atomicAdd((unsigned long long*)&d.counters[1],
+                                (unsigned long long)cnt);
+      for (uint32_t cur = head; cur != 0;) {
+        uint32_t brow = cur - 1;
+        if (gptr<uint64_t>(bk.data)[brow] == key) {
+          d.outBuild[base] = brow;
+          d.outProbe[base] = (uint32_t)row;
+          base++;
+        }
+        cur = gptr<uint32_t>(d.next)[brow];
+      }
+    } else {
+      my += cnt;
+    }
+  }
+  if (!FILL) {
+    for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
+    if ((threadIdx.x & 63) == 0 && my)
+      atomicAdd((unsigned long long*)&d.counters[0], (unsigned long long)my);
+  }
+}
+
+// gather a null bitmap through the match index: one thread composes one
+// output byte (8 rows) — no atomics (LSB-first, 1 = NOT NULL)
+__global__ void hjGatherNullsKernel(const uint8_t* __restrict__ in,
+                                    const uint32_t* __restrict__ idx,
+                                    uint8_t* __restrict__ out, int64_t n) {
+  int64_t nBytes = (n + 7) / 8;
+  for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < nBytes;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    int64_t base = b * 8;
+    int m = n - base < 8 ? (int)(n - base) : 8;
+    uint8_t v = 0;
+    for (int j = 0; j < m; j++) {
+      uint32_t src = idx[base + j];
+      if ((in[src >> 3] >> (src & 7)) & 1) v |= (uint8_t)(1 << j);
+    }
+    out[b] = v;
+  }
+}
+
+__global__ void iotaI64Kernel(int64_t* __restrict__ p, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    p[i] = i;
+}
+
 __global__ void sortGatherKernel(const uint8_t* __restrict__ in,
                                  uint8_t* __restrict__ out,
                                  const uint32_t* __restrict__ idx, int64_t n,
@@ -1787,6 +1892,36 @@ int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,
   hipLaunchKernelGGL(sortGatherKernel, dim3(gridFor(n)), dim3(256), 0,
                      (hipStream_t)stream, (const uint8_t*)in, (uint8_t*)out,
                      idx, n, elemSize);
+  return (int)hipGetLastError();
+}
+
+int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
+                    const HashJoinDesc& h, void* stream) {
+  int64_t rows = phase == 0 ? h.build.nRows : h.probe.nRows;
+  if (rows == 0) return 0;
+  dim3 g(gridFor(rows));
+  if (phase == 0)
+    hipLaunchKernelGGL(hjBuildKernel, g, dim3(256), 0, (hipStream_t)stream,
+                       devDesc);
+  else if (phase == 1)
+    hipLaunchKernelGGL(hjProbeKernel<false>, g, dim3(256), 0,
+                       (hipStream_t)stream, devDesc);
+  else
+    hipLaunchKernelGGL(hjProbeKernel<true>, g, dim3(256), 0,
+                       (hipStream_t)stream, devDesc);
+  return (int)hipGetLastError();
+}
+
+int gxGatherNulls(const uint8_t* inBitmap, const uint32_t* idx, uint8_t* out,
+                  int64_t n, void* stream) {
+  hipLaunchKernelGGL(hjGatherNullsKernel, dim3(gridFor((n + 7) / 8)), dim3(256),
+                     0, (hipStream_t)stream, inBitmap, idx, out, n);
+  return (int)hipGetLastError();
+}
+
+int gxIotaOffsets(int64_t* p, int64_t n, void* stream) {
+  hipLaunchKernelGGL(iotaI64Kernel, dim3(gridFor(n)), dim3(256), 0,
+                     (hipStream_t)stream, p, n);
   return (int)hipGetLastError();
 }
 

@@ -372,6 +372,151 @@ def bench_wide(args):
     print(json.dumps(out), flush=True)
 
 
+def run_cpu_baseline_join(n_li=4_000_000):
+    """Oracle standalone join over a bounded sample, single thread."""
+    import time as _t
+    from tests.gxlib import (GX_TPCH_LINEITEM, GX_TPCH_ORDERS, GX_TYPE_I64,
+                             load_oracle)
+    from tidb_amd import plan as P
+    lib = load_oracle()
+    n_ord = n_li // 4
+    t0 = _t.perf_counter()
+    b = P.Builder(lib)
+    orders = b.source(P.ORDERS_TYPES)
+    li = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+    j = b.hashjoin(orders, li, [b.colref(P.O_ORDERKEY, GX_TYPE_I64)],
+                   [b.colref(P.L_ORDERKEY, GX_TYPE_I64)])
+    ex = b.build(j)
+    ex.bind_tpch(orders, GX_TPCH_ORDERS, n_ord)
+    ex.bind_tpch(li, GX_TPCH_LINEITEM, n_li)
+    ex.open()
+    # drive Next to EOF without Python-decoding every row
+    import ctypes as C
+    from tidb_amd.chunkpy import PyChunk
+    out_types = P.ORDERS_TYPES + P.LINEITEM_TYPES
+    out_fracs = [0] * 4 + P.LINEITEM_FRACS
+    chunk = PyChunk(out_types, 1024, out_fracs)
+    total = 0
+    while True:
+        g = chunk.as_gx()
+        n = C.c_int32(0)
+        rc = lib.gx_next(ex.ex, C.byref(g), C.byref(n))
+        assert rc == 0, ex.error()
+        if n.value == 0:
+            break
+        total += n.value
+    ex.close()
+    ex.free()
+    b.free()
+    dt = _t.perf_counter() - t0
+    return {
+        "value": (n_li + n_ord) / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"orders({n_ord}) JOIN lineitem({n_li}) on orderkey incl. "
+                  f"generation, {total} joined rows, single thread ({dt:.1f}s)",
+    }
+
+
+def bench_join(args):
+    """Standalone inner hash join (hash_join_v2.go):
+    orders (build) ⋈ lineitem (probe) on orderkey, joined rows MATERIALIZED
+    on device (build cols ++ probe cols gathered through the match index).
+    One step = chain build + count + fill + gather of all 12 output columns
+    over freshly generated tables (kernel time via HIP events around the
+    join phases; generation excluded)."""
+    import ctypes as C
+    from tests.gxlib import (GX_TPCH_LINEITEM, GX_TPCH_ORDERS, GX_TYPE_I64,
+                             load_product)
+    from tidb_amd import plan as P
+    lib = load_product()
+    n_li = args.rows
+    n_ord = n_li // 4
+    lib.gx_last_kernel_ms.restype = C.c_double
+    lib.gx_last_kernel_ms.argtypes = [C.c_void_p]
+    lib.gx_last_sel_count.restype = C.c_int64
+    lib.gx_last_sel_count.argtypes = [C.c_void_p]
+
+    out_types = P.ORDERS_TYPES + P.LINEITEM_TYPES
+    out_fracs = [0] * 4 + P.LINEITEM_FRACS
+
+    def step():
+        # fresh executor per step: device buffers (output table ~13 GB at
+        # SF10) are freed with it
+        b = P.Builder(lib)
+        orders = b.source(P.ORDERS_TYPES)
+        li = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+        j = b.hashjoin(orders, li, [b.colref(P.O_ORDERKEY, GX_TYPE_I64)],
+                       [b.colref(P.L_ORDERKEY, GX_TYPE_I64)])
+        ex = b.build(j)
+        ex.bind_tpch(orders, GX_TPCH_ORDERS, n_ord)
+        ex.bind_tpch(li, GX_TPCH_LINEITEM, n_li)
+        ex.open()
+        ex.pull_one(out_types, out_fracs,
+                    data_caps=[None] * 9 + [2048, 2048, None])
+        k = lib.gx_last_kernel_ms(ex.ex)
+        matches = lib.gx_last_sel_count(ex.ex)
+        ex.close()
+        ex.free()
+        b.free()
+        return k, matches
+
+    for _ in range(args.warmup):
+        step()
+    t0 = time.perf_counter()
+    kms = []
+    matches = 0
+    for _ in range(args.steps):
+        k, matches = step()
+        kms.append(k)
+    elapsed = time.perf_counter() - t0
+    avg_kms = sum(kms) / len(kms)
+    value = (n_li + n_ord) / (avg_kms / 1000.0) if avg_kms else 0
+    # algorithmic bytes per matched row (≈ per probe row here: every lineitem
+    # matches its order): probe side read 178 (8B key+time + 4x40 dec + 2
+    # dense char) + build side random read 32 (4x8B) + output write 210 +
+    # match-pair index write+read 16; plus per probe row: head probe 4 +
+    # next walk 4
+    bpr_out = 178 + 32 + 210 + 16
+    achieved = (matches * bpr_out + n_li * 8) / (avg_kms / 1000.0) / 1e9 \
+        if avg_kms else 0
+    out = {
+        "metric": "hash_join_rows_per_sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": {
+            "workload": f"orders_{n_ord}_join_lineitem_{n_li}_materialized",
+            "lineitem_rows": n_li,
+            "orders_rows": n_ord,
+            "joined_rows": matches,
+            "parallelism": "single-gpu",
+            "bytes_per_joined_row": bpr_out,
+        },
+        "join_kernel_ms_avg": avg_kms,
+        "roofline": {
+            "bound": "hbm",
+            "achieved": achieved,
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": achieved / HBM_PEAK_GBS,
+            "traffic": None,
+        },
+        "cpu_baseline": run_cpu_baseline_join() if not args.no_cpu_baseline
+        else None,
+    }
+    print(json.dumps(out), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -379,7 +524,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--rows", type=int, default=SF10_ROWS,
                     help="rows per GPU (default SF10)")
-    ap.add_argument("--query", choices=["q1", "q3", "sort", "wide"],
+    ap.add_argument("--query", choices=["q1", "q3", "sort", "wide", "join"],
                     default="q1")
     ap.add_argument("--sf", type=int, default=100,
                     help="scale factor for --query q3 (lineitem = 6M x SF)")
@@ -392,6 +537,8 @@ def main():
         return bench_sort(args)
     if args.query == "wide":
         return bench_wide(args)
+    if args.query == "join":
+        return bench_join(args)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

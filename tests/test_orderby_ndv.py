@@ -84,3 +84,14 @@ def test_orderkey_agg_parity():
     a = run_plan(load_oracle(), orderkey_agg_plan, 8000)
     b = run_plan(load_product(), orderkey_agg_plan, 8000)
     assert a == b
+
+
+@pytest.mark.gpu
+def test_orderkey_agg_high_ndv_parity():
+    """NDV ~60k > the default 8192-slot global table: exercises the
+    kErrGlobalFull grow-and-rerun retry (globalGroupsLog2 13 -> 16)."""
+    from tests.gxlib import load_product
+    a = run_plan(load_oracle(), orderkey_agg_plan, 240000)
+    b = run_plan(load_product(), orderkey_agg_plan, 240000)
+    assert len(a) > 8192
+    assert a == b

@@ -145,7 +145,12 @@ struct GroupKeyDesc {
 
 constexpr uint64_t kEmptyKey = ~0ULL;
 constexpr int kLdsGroups = 64;     // per-workgroup LDS table capacity (19 KiB state => 8 blocks/CU, full 32-wave occupancy)
-constexpr int kGlobalGroups = 8192; // global table capacity (power of two)
+constexpr int kGlobalGroups = 8192; // DEFAULT global table capacity (2^13);
+                                    // FusedQueryDesc.globalGroupsLog2 scales
+                                    // it — on kErrGlobalFull the engine
+                                    // reruns with an 8x larger table (the
+                                    // spill-partition growth analog of
+                                    // agg_spill.go's 256 partitions)
 
 // group state sized for kMaxAggs
 struct GroupSlot {
@@ -195,7 +200,8 @@ struct FusedQueryDesc {
                        // demand alone costs a wave/SIMD of occupancy)
   GroupKeyDesc gkey;
   // outputs
-  GroupSlot* globalTable = nullptr;  // kGlobalGroups slots
+  GroupSlot* globalTable = nullptr;  // 1<<globalGroupsLog2 slots
+  int32_t globalGroupsLog2 = 13;     // current global table capacity (log2)
   uint32_t* errorFlag = nullptr;     // != 0 => abort with error
   uint64_t* selCount = nullptr;      // rows passing the filter (stats)
   // perf ablation (GX_ABLATE env; results are WRONG when nonzero — timing
@@ -341,7 +347,7 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream);
 int gxLaunchMemset(void* p, int v, size_t n, void* stream);
 int gxFusedGrid(int64_t rows);
-int gxLaunchInitTable(GroupSlot* table, void* stream);
+int gxLaunchInitTable(GroupSlot* table, int nSlots, void* stream);
 int gxDumpDesc(const FusedQueryDesc* devDesc, void* stream);
 
 // ---- device full sort (sortexec/sort.go analog): LSD stable radix passes

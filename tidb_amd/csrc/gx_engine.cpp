@@ -1692,7 +1692,8 @@ static int32_t materializeDevice(gx_exec* ex) {
     }
   }
   // result/error buffers
-  ex->devTable = (gxp::GroupSlot*)devAlloc(ex, sizeof(gxp::GroupSlot) * gxp::kGlobalGroups);
+  ex->devTable = (gxp::GroupSlot*)devAlloc(
+      ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
   ex->devErr = (uint32_t*)devAlloc(ex, 4);
   ex->devSel = (uint64_t*)devAlloc(ex, 8);
   ex->devDesc = (gxp::FusedQueryDesc*)devAlloc(ex, sizeof(gxp::FusedQueryDesc));
@@ -1884,7 +1885,8 @@ static int32_t runFused(gx_exec* ex) {
   }
   int lrc;
   if (ex->jitProg) {
-    lrc = gxp::gxLaunchInitTable(ex->desc.globalTable, ex->stream);
+    lrc = gxp::gxLaunchInitTable(ex->desc.globalTable,
+                                 1 << ex->desc.globalGroupsLog2, ex->stream);
     if (lrc == 0)
       lrc = gxjit::launch(ex->jitProg, ex->desc.wide != 0, ex->devDesc,
                           gxp::gxFusedGrid(ex->desc.table.nRows), ex->stream);
@@ -1921,6 +1923,18 @@ static int32_t runFused(gx_exec* ex) {
     if (getenv("GX_DEBUG")) fprintf(stderr, "[gx] LDS table full -> global-direct retry\n");
     return runFused(ex);
   }
+  if ((errFlag & 32u /*kErrGlobalFull*/) && ex->desc.globalGroupsLog2 < 25) {
+    // NDV above the global table: rerun with an 8x table (capped 2^25 groups
+    // ~ 10 GB of state; the init kernel resets it, so the rerun is clean)
+    ex->desc.globalGroupsLog2 += 3;
+    ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAlloc(
+        ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
+    if (!ex->devTable) { ex->err = "hipMalloc failed (group table)"; return GX_ERR_INTERNAL; }
+    if (getenv("GX_DEBUG"))
+      fprintf(stderr, "[gx] global table full -> 2^%d retry\n",
+              ex->desc.globalGroupsLog2);
+    return runFused(ex);
+  }
   errFlag &= ~256u;
   if (ex->desc.ablate != 0 && getenv("GX_DEBUG"))
     fprintf(stderr, "[gx] ABLATE=%d kms=%.3f\n", ex->desc.ablate, ex->lastKernelMs);
@@ -1938,9 +1952,9 @@ static int32_t runFused(gx_exec* ex) {
             ex->desc.nAggs, ex->desc.gkey.nCols,
             (unsigned long long)ex->lastSelCount, ex->lastKernelMs);
   }
-  std::vector<gxp::GroupSlot> table(gxp::kGlobalGroups);
+  std::vector<gxp::GroupSlot> table((size_t)1 << ex->desc.globalGroupsLog2);
   HIP_OK(ex, hipMemcpy(table.data(), ex->devTable,
-                       sizeof(gxp::GroupSlot) * gxp::kGlobalGroups,
+                       sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2,
                        hipMemcpyDeviceToHost));
   // collect occupied slots, deterministic order (by key)
   std::vector<const gxp::GroupSlot*> occ;

@@ -297,9 +297,10 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
   }
   s << "    return true;\n  }\n";
   // global-direct path
-  s << "  { uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));\n"
-       "    for (int probe = 0;; probe++) {\n"
-       "      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, "
+  s << "  { uint32_t gmask = (1u << d.globalGroupsLog2) - 1;\n"
+       "    uint32_t slot = (uint32_t)(splitmix64(key) & gmask);\n"
+       "    for (uint32_t probe = 0;; probe++) {\n"
+       "      if (probe > gmask) { atomicOr(d.errorFlag, "
        "kErrGlobalFull); return false; }\n"
        "      uint64_t cur = d.globalTable[slot].key;\n"
        "      if (cur == key) break;\n"
@@ -308,7 +309,7 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
        "[slot].key, (unsigned long long)kEmptyKey, (unsigned long long)key);\n"
        "        if (prev == kEmptyKey || prev == key) break;\n"
        "      }\n"
-       "      slot = (slot + 1) & (kGlobalGroups - 1);\n"
+       "      slot = (slot + 1) & gmask;\n"
        "    }\n"
        "    GroupSlot* t = &d.globalTable[slot];\n";
   if (d.sharedCnt) s << "    accumInto(t, 0, Int128{0, 0}, 1);\n";
@@ -451,10 +452,11 @@ __device__ __forceinline__ void kernBody(const FusedQueryDesc* __restrict__ dp) 
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     if (lds[i].key == kEmptyKey) continue;
     uint64_t key = lds[i].key;
-    uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
+    uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
+    uint32_t slot = (uint32_t)(splitmix64(key) & gmask);
     bool ok = true;
-    for (int probe = 0;; probe++) {
-      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
+    for (uint32_t probe = 0;; probe++) {
+      if (probe > gmask) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
       uint64_t cur = d.globalTable[slot].key;
       if (cur == key) break;
       if (cur == kEmptyKey) {
@@ -463,7 +465,7 @@ __device__ __forceinline__ void kernBody(const FusedQueryDesc* __restrict__ dp) 
                                   (unsigned long long)key);
         if (prev == kEmptyKey || prev == key) break;
       }
-      slot = (slot + 1) & (kGlobalGroups - 1);
+      slot = (slot + 1) & gmask;
     }
     if (!ok) continue;
     for (int s2 = 0; s2 < d.nAccSlots; s2++) {

@@ -405,10 +405,11 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     }
     return true;
   }
-  // global-direct path (mid/high NDV up to kGlobalGroups)
-  uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
-  for (int probe = 0;; probe++) {
-    if (probe >= kGlobalGroups) {
+  // global-direct path (mid/high NDV; table grows on kErrGlobalFull retry)
+  uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
+  uint32_t slot = (uint32_t)(splitmix64(key) & gmask);
+  for (uint32_t probe = 0;; probe++) {
+    if (probe > gmask) {
       atomicOr(d.errorFlag, kErrGlobalFull);
       return false;
     }
@@ -420,7 +421,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
                                 (unsigned long long)key);
       if (prev == kEmptyKey || prev == key) break;
     }
-    slot = (slot + 1) & (kGlobalGroups - 1);
+    slot = (slot + 1) & gmask;
   }
   GroupSlot* target = &d.globalTable[slot];
   if (d.sharedCnt) accumInto(target, 0, Int128{0, 0}, 1);  // bumps cnt[0] only
@@ -516,10 +517,11 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     if (lds[i].key == kEmptyKey) continue;
     uint64_t key = lds[i].key;
-    uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
+    uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
+    uint32_t slot = (uint32_t)(splitmix64(key) & gmask);
     bool ok = true;
-    for (int probe = 0;; probe++) {
-      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
+    for (uint32_t probe = 0;; probe++) {
+      if (probe > gmask) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
       uint64_t cur = d.globalTable[slot].key;
       if (cur == key) break;
       if (cur == kEmptyKey) {
@@ -528,7 +530,7 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
                                   (unsigned long long)key);
         if (prev == kEmptyKey || prev == key) break;
       }
-      slot = (slot + 1) & (kGlobalGroups - 1);
+      slot = (slot + 1) & gmask;
     }
     if (!ok) continue;
     for (int s = 0; s < d.nAccSlots; s++) {
@@ -1335,10 +1337,11 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     if (lds[i].key == kEmptyKey) continue;
     uint64_t key = lds[i].key;
-    uint32_t slot = (uint32_t)(splitmix64(key) & (kGlobalGroups - 1));
+    uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
+    uint32_t slot = (uint32_t)(splitmix64(key) & gmask);
     bool ok = true;
-    for (int probe = 0;; probe++) {
-      if (probe >= kGlobalGroups) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
+    for (uint32_t probe = 0;; probe++) {
+      if (probe > gmask) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
       uint64_t cur = d.globalTable[slot].key;
       if (cur == key) break;
       if (cur == kEmptyKey) {
@@ -1347,7 +1350,7 @@ __global__ void fusedAggGldsKernel(const FusedQueryDesc* __restrict__ dp) {
                                   (unsigned long long)key);
         if (prev == kEmptyKey || prev == key) break;
       }
-      slot = (slot + 1) & (kGlobalGroups - 1);
+      slot = (slot + 1) & gmask;
     }
     if (!ok) continue;
     for (int s = 0; s < d.nAccSlots; s++) {
@@ -1474,9 +1477,9 @@ int gxFusedGrid(int64_t rows) {
   return grid;
 }
 
-int gxLaunchInitTable(GroupSlot* table, void* stream) {
-  hipLaunchKernelGGL(initGlobalTableKernel, dim3((kGlobalGroups + 255) / 256),
-                     dim3(256), 0, (hipStream_t)stream, table, kGlobalGroups);
+int gxLaunchInitTable(GroupSlot* table, int nSlots, void* stream) {
+  hipLaunchKernelGGL(initGlobalTableKernel, dim3((nSlots + 255) / 256),
+                     dim3(256), 0, (hipStream_t)stream, table, nSlots);
   return (int)hipGetLastError();
 }
 
@@ -1484,8 +1487,9 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream) {
   hipStream_t s = (hipStream_t)stream;
   int grid = gxFusedGrid(desc.table.nRows);
-  hipLaunchKernelGGL(initGlobalTableKernel, dim3((kGlobalGroups + 255) / 256),
-                     dim3(256), 0, s, desc.globalTable, kGlobalGroups);
+  int nSlots = 1 << desc.globalGroupsLog2;
+  hipLaunchKernelGGL(initGlobalTableKernel, dim3((nSlots + 255) / 256),
+                     dim3(256), 0, s, desc.globalTable, nSlots);
   if (desc.useGlds) {
     size_t shmem = ((sizeof(GroupSlot) * kLdsGroups + 15) & ~15ULL) +
                    (size_t)4 * 2 * desc.tileBytes;

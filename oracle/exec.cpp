@@ -980,6 +980,8 @@ class HashJoinExec : public Exec {
     built_ = false;
     pending_.reset();
     pendEmit_ = 0;
+    buildData_.clear();
+    table_.clear();
     int32_t ec = build_->open();
     if (ec) return ec;
     return probe_->open();

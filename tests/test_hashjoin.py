@@ -208,6 +208,65 @@ def test_join_parity_random_larger(libs):
     assert got == want
 
 
+def _two_key_plan(lib):
+    """2-column join key (int64, time) — SerializeKeys concatenation
+    (codec.go:852-910) specialized to fixed 8-byte keys."""
+    b = P.Builder(lib)
+    bsrc = b.source([GX_TYPE_I64, GX_TYPE_TIME, GX_TYPE_I64])
+    psrc = b.source([GX_TYPE_I64, GX_TYPE_TIME, GX_TYPE_I64])
+    j = b.hashjoin(bsrc, psrc,
+                   [b.colref(0, GX_TYPE_I64), b.colref(1, GX_TYPE_TIME)],
+                   [b.colref(0, GX_TYPE_I64), b.colref(1, GX_TYPE_TIME)])
+    return b, bsrc, psrc, j
+
+
+TK_BUILD = [
+    (1, DATE, 100),
+    (1, DATE, 101),                 # duplicate (key0,key1)
+    (1, DATE + (1 << 41), 102),     # same key0, different date
+    (2, DATE, 103),
+    (None, DATE, 104),              # NULL first key
+    (2, None, 105),                 # NULL second key
+]
+TK_PROBE = [
+    (1, DATE, 1),
+    (1, DATE + (1 << 41), 2),
+    (2, DATE, 3),
+    (2, None, 4),                   # NULL key: no match
+    (3, DATE, 5),
+]
+
+
+def run_two_key(lib):
+    b, bsrc, psrc, j = _two_key_plan(lib)
+    ex = b.build(j)
+    t3 = [GX_TYPE_I64, GX_TYPE_TIME, GX_TYPE_I64]
+    ex.bind_chunks(bsrc, [_to_chunk(lib, t3, [0, 0, 0], TK_BUILD)])
+    ex.bind_chunks(psrc, [_to_chunk(lib, t3, [0, 0, 0], TK_PROBE)])
+    ex.open()
+    rows = ex.pull_all(t3 + t3)
+    ex.close()
+    ex.free()
+    b.free()
+    return _canon(rows)
+
+
+def test_oracle_join_two_keys():
+    lib = load_oracle()
+    want = _canon([br + pr for pr in TK_PROBE for br in TK_BUILD
+                   if None not in (br[0], br[1], pr[0], pr[1])
+                   and br[0] == pr[0] and br[1] == pr[1]])
+    got = run_two_key(lib)
+    assert got == want
+    assert len(got) == 4  # (1,DATE)x2 matches, (1,DATE+1d), (2,DATE)
+
+
+@pytest.mark.gpu
+def test_join_parity_two_keys(libs):
+    oracle, product = libs
+    assert run_two_key(product) == run_two_key(oracle)
+
+
 @pytest.mark.gpu
 def test_join_parity_generator(libs):
     """orders ⋈ lineitem on orderkey over the synthetic generator tables,

@@ -313,7 +313,11 @@ struct HashJoinDesc {
   int32_t strConstBLen = 0;
   uint8_t strConstP[16];
   int32_t strConstPLen = 0;
-  int32_t bKeyCol = 0, pKeyCol = 0;  // single int64 key column per side
+  // 1-2 fixed 8-byte key columns per side (int64/packed-Time; the
+  // SerializeKeys concatenation of codec.go:852-910 specialized to
+  // fixed-width keys — equality of the concatenation == per-column equality)
+  int32_t nKeys = 1;
+  int32_t bKeyCol[2] = {0, 0}, pKeyCol[2] = {0, 0};
   uint32_t* heads = nullptr;  // 1<<headsLog2 entries: build row+1, 0 = empty
   int32_t headsLog2 = 0;
   uint32_t* next = nullptr;   // per build row: next chain row+1, 0 = end

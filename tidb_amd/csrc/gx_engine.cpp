@@ -990,8 +990,9 @@ static int32_t compileHashJoin(gx_exec* ex) {
     ex->err = "only inner joins on device this round";
     return GX_ERR_INVALID;
   }
-  if (jn.buildKeys.size() != 1 || jn.probeKeys.size() != 1) {
-    ex->err = "device hash join supports one key column this round";
+  if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
+      jn.buildKeys.size() > 2) {
+    ex->err = "device hash join supports 1-2 key columns this round";
     return GX_ERR_INVALID;
   }
   const PNode *selB, *selP;
@@ -1009,21 +1010,27 @@ static int32_t compileHashJoin(gx_exec* ex) {
     ex->err = "too many join output columns";
     return GX_ERR_INVALID;
   }
-  const PExpr& bk = plan.exprs[jn.buildKeys[0]];
-  const PExpr& pk = plan.exprs[jn.probeKeys[0]];
-  if (bk.kind != EK_COLREF || pk.kind != EK_COLREF || bk.colIdx < 0 ||
-      bk.colIdx >= nb || pk.colIdx < 0 || pk.colIdx >= np) {
-    ex->err = "join keys must be child columns";
-    return GX_ERR_INVALID;
-  }
-  if (bN.colTypes[bk.colIdx] != GX_TYPE_I64 ||
-      pN.colTypes[pk.colIdx] != GX_TYPE_I64) {
-    ex->err = "device join keys must be int64 this round";
-    return GX_ERR_INVALID;
-  }
   gxp::HashJoinDesc& hj = ex->hj;
-  hj.bKeyCol = bk.colIdx;
-  hj.pKeyCol = pk.colIdx;
+  hj.nKeys = (int32_t)jn.buildKeys.size();
+  for (int k = 0; k < hj.nKeys; k++) {
+    const PExpr& bk = plan.exprs[jn.buildKeys[k]];
+    const PExpr& pk = plan.exprs[jn.probeKeys[k]];
+    if (bk.kind != EK_COLREF || pk.kind != EK_COLREF || bk.colIdx < 0 ||
+        bk.colIdx >= nb || pk.colIdx < 0 || pk.colIdx >= np) {
+      ex->err = "join keys must be child columns";
+      return GX_ERR_INVALID;
+    }
+    int bt = bN.colTypes[bk.colIdx];
+    int pt = pN.colTypes[pk.colIdx];
+    // fixed 8-byte keys: int64 or packed CoreTime (SerializeKeys writes the
+    // raw 8 bytes for both — equality is bitwise)
+    if (bt != pt || (bt != GX_TYPE_I64 && bt != GX_TYPE_TIME)) {
+      ex->err = "device join keys must be int64/time this round";
+      return GX_ERR_INVALID;
+    }
+    hj.bKeyCol[k] = bk.colIdx;
+    hj.pKeyCol[k] = pk.colIdx;
+  }
   auto doPred = [&](const PNode* sel, const PNode& srcNode, gxp::PredDesc* pd,
                     int32_t* n, uint8_t* sc, int32_t* scLen) -> bool {
     *n = 0;

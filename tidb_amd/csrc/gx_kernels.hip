@@ -1714,6 +1714,41 @@ __global__ void sortComposeKeysKernel(DevTable tab, SortKeyCompose k,
 // standalone hash join (inner, duplicate build keys) — HashJoinDesc
 // ==================================================================
 
+// load a row's 1-2 fixed 8-byte key columns; false if any is NULL
+// (NULL join keys never match, inner_join_probe.go)
+__device__ inline bool hjLoadKeys(const HashJoinDesc& d, const DevTable& t,
+                                  const int32_t* cols, int64_t row,
+                                  uint64_t* k0, uint64_t* k1) {
+  const DevCol& c0 = t.cols[cols[0]];
+  if (colIsNull(c0, row)) return false;
+  *k0 = gptr<uint64_t>(c0.data)[row];
+  *k1 = 0;
+  if (d.nKeys > 1) {
+    const DevCol& c1 = t.cols[cols[1]];
+    if (colIsNull(c1, row)) return false;
+    *k1 = gptr<uint64_t>(c1.data)[row];
+  }
+  return true;
+}
+
+__device__ inline uint64_t hjHash(const HashJoinDesc& d, uint64_t k0,
+                                  uint64_t k1) {
+  uint64_t h = hashKey(k0);
+  if (d.nKeys > 1) h = splitmix64(h ^ k1);
+  return h;
+}
+
+// chain member's keys == probe keys? (rows with NULL keys never entered the
+// chain, so no null checks here)
+__device__ inline bool hjBuildKeyEq(const HashJoinDesc& d, uint32_t brow,
+                                    uint64_t k0, uint64_t k1) {
+  if (gptr<uint64_t>(d.build.cols[d.bKeyCol[0]].data)[brow] != k0) return false;
+  if (d.nKeys > 1 &&
+      gptr<uint64_t>(d.build.cols[d.bKeyCol[1]].data)[brow] != k1)
+    return false;
+  return true;
+}
+
 // chain-insert every qualifying build row (lock-free head CAS; the next[]
 // write is published to the probe kernels by the dispatch boundary)
 __global__ void hjBuildKernel(const HashJoinDesc* __restrict__ dp) {
@@ -1725,10 +1760,9 @@ __global__ void hjBuildKernel(const HashJoinDesc* __restrict__ dp) {
     bool pass = d.nPredB == 0 ||
                 evalSimplePred(d.build, d.predB, d.strConstB, d.strConstBLen, row);
     if (!pass) continue;
-    const DevCol& kc = d.build.cols[d.bKeyCol];
-    if (colIsNull(kc, row)) continue;  // NULL never joins
-    uint64_t key = gptr<uint64_t>(kc.data)[row];
-    uint32_t slot = (uint32_t)(hashKey(key) & mask);
+    uint64_t key, key1;
+    if (!hjLoadKeys(d, d.build, d.bKeyCol, row, &key, &key1)) continue;
+    uint32_t slot = (uint32_t)(hjHash(d, key, key1) & mask);
     uint32_t newHead = (uint32_t)row + 1;
     uint32_t old = d.heads[slot];
     for (;;) {
@@ -1749,21 +1783,20 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
   const HashJoinDesc& d = *dp;
   int64_t n = d.probe.nRows;
   uint32_t mask = (1u << d.headsLog2) - 1;
-  const DevCol& bk = d.build.cols[d.bKeyCol];
   uint64_t my = 0;
   for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
        row += (int64_t)gridDim.x * blockDim.x) {
     bool pass = d.nPredP == 0 ||
                 evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen, row);
     if (!pass) continue;
-    const DevCol& kc = d.probe.cols[d.pKeyCol];
-    if (colIsNull(kc, row)) continue;
-    uint64_t key = gptr<uint64_t>(kc.data)[row];
-    uint32_t head = gptr<uint32_t>(d.heads)[(uint32_t)(hashKey(key) & mask)];
+    uint64_t key, key1;
+    if (!hjLoadKeys(d, d.probe, d.pKeyCol, row, &key, &key1)) continue;
+    uint32_t head =
+        gptr<uint32_t>(d.heads)[(uint32_t)(hjHash(d, key, key1) & mask)];
     uint32_t cnt = 0;
     for (uint32_t cur = head; cur != 0;) {
       uint32_t brow = cur - 1;
-      if (gptr<uint64_t>(bk.data)[brow] == key) cnt++;
+      if (hjBuildKeyEq(d, brow, key, key1)) cnt++;
       cur = gptr<uint32_t>(d.next)[brow];
     }
     if (cnt == 0) continue;
@@ -1772,7 +1805,7 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
                                 (unsigned long long)cnt);
       for (uint32_t cur = head; cur != 0;) {
         uint32_t brow = cur - 1;
-        if (gptr<uint64_t>(bk.data)[brow] == key) {
+        if (hjBuildKeyEq(d, brow, key, key1)) {
           d.outBuild[base] = brow;
           d.outProbe[base] = (uint32_t)row;
           base++;

@@ -2440,9 +2440,12 @@ static int32_t runHashJoin(gx_exec* ex) {
       return GX_ERR_INVALID;
     }
   }
-  hipEvent_t ev0, ev1;
+  hipEvent_t ev0, ev1, evB, evC, evF;
   HIP_OK(ex, hipEventCreate(&ev0));
   HIP_OK(ex, hipEventCreate(&ev1));
+  HIP_OK(ex, hipEventCreate(&evB));
+  HIP_OK(ex, hipEventCreate(&evC));
+  HIP_OK(ex, hipEventCreate(&evF));
   HIP_OK(ex, hipEventRecord(ev0, ex->stream));
   // chained table: heads (2x rows, pow2) + per-row next links
   hj.headsLog2 = ceilLog2(std::max<uint64_t>(2 * (uint64_t)nb + 1, 64));
@@ -2454,11 +2457,16 @@ static int32_t runHashJoin(gx_exec* ex) {
   HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
   HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj), hipMemcpyHostToDevice,
                             ex->stream));
-  if (gxp::gxHashJoinPhase(0, ex->devHj, hj, ex->stream) != 0 ||
-      gxp::gxHashJoinPhase(1, ex->devHj, hj, ex->stream) != 0) {
+  if (gxp::gxHashJoinPhase(0, ex->devHj, hj, ex->stream) != 0) {
     ex->err = "join kernel launch failed";
     return GX_ERR_INTERNAL;
   }
+  HIP_OK(ex, hipEventRecord(evB, ex->stream));
+  if (gxp::gxHashJoinPhase(1, ex->devHj, hj, ex->stream) != 0) {
+    ex->err = "join kernel launch failed";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipEventRecord(evC, ex->stream));
   HIP_OK(ex, hipStreamSynchronize(ex->stream));
   uint64_t total = 0;
   HIP_OK(ex, hipMemcpy(&total, hj.counters, 8, hipMemcpyDeviceToHost));
@@ -2486,6 +2494,7 @@ static int32_t runHashJoin(gx_exec* ex) {
       ex->err = "join fill launch failed";
       return GX_ERR_INTERNAL;
     }
+    HIP_OK(ex, hipEventRecord(evF, ex->stream));
     // gather every output column through its side's match index
     for (int c = 0; c < ex->desc.table.nCols; c++) {
       int nbc = hj.build.nCols;
@@ -2532,8 +2541,22 @@ static int32_t runHashJoin(gx_exec* ex) {
     float ms = 0;
     hipEventElapsedTime(&ms, ev0, ev1);
     ex->lastKernelMs = ms;
+    if (getenv("GX_DEBUG")) {
+      float mb = 0, mc = 0, mf = 0;
+      hipEventElapsedTime(&mb, ev0, evB);
+      hipEventElapsedTime(&mc, evB, evC);
+      if (total > 0) hipEventElapsedTime(&mf, evC, evF);
+      fprintf(stderr,
+              "[gx] hj build=%.3fms count=%.3fms fill=%.3fms gather=%.3fms "
+              "total=%.3fms matches=%llu\n",
+              mb, mc, mf, ms - mb - mc - mf, ms,
+              (unsigned long long)total);
+    }
     hipEventDestroy(ev0);
     hipEventDestroy(ev1);
+    hipEventDestroy(evB);
+    hipEventDestroy(evC);
+    hipEventDestroy(evF);
   }
   ex->desc.table.nRows = (int64_t)total;
   ex->srcPos = 0;

@@ -1783,34 +1783,57 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
   const HashJoinDesc& d = *dp;
   int64_t n = d.probe.nRows;
   uint32_t mask = (1u << d.headsLog2) - 1;
+  int lane = threadIdx.x & 63;
   uint64_t my = 0;
-  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
-       row += (int64_t)gridDim.x * blockDim.x) {
-    bool pass = d.nPredP == 0 ||
-                evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen, row);
-    if (!pass) continue;
-    uint64_t key, key1;
-    if (!hjLoadKeys(d, d.probe, d.pKeyCol, row, &key, &key1)) continue;
-    uint32_t head =
-        gptr<uint32_t>(d.heads)[(uint32_t)(hjHash(d, key, key1) & mask)];
+  // wave-uniform loop: FILL reserves output ranges with ONE atomic per
+  // wavefront (a per-row atomic on the shared cursor serializes ~100M
+  // atomics/s — measured 369 ms at 30M rows before aggregation)
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;;
+       row += stride) {
+    bool active = row < n;
+    if (__ballot(active) == 0) break;
     uint32_t cnt = 0;
-    for (uint32_t cur = head; cur != 0;) {
-      uint32_t brow = cur - 1;
-      if (hjBuildKeyEq(d, brow, key, key1)) cnt++;
-      cur = gptr<uint32_t>(d.next)[brow];
-    }
-    if (cnt == 0) continue;
-    if (FILL) {
-      uint64_t base = atomicAdd((unsigned long long*)&d.counters[1],
-                                (unsigned long long)cnt);
-      for (uint32_t cur = head; cur != 0;) {
-        uint32_t brow = cur - 1;
-        if (hjBuildKeyEq(d, brow, key, key1)) {
-          d.outBuild[base] = brow;
-          d.outProbe[base] = (uint32_t)row;
-          base++;
+    uint32_t head = 0;
+    uint64_t key = 0, key1 = 0;
+    if (active) {
+      bool pass = d.nPredP == 0 ||
+                  evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen,
+                                 row);
+      if (pass && hjLoadKeys(d, d.probe, d.pKeyCol, row, &key, &key1)) {
+        head = gptr<uint32_t>(d.heads)[(uint32_t)(hjHash(d, key, key1) & mask)];
+        for (uint32_t cur = head; cur != 0;) {
+          uint32_t brow = cur - 1;
+          if (hjBuildKeyEq(d, brow, key, key1)) cnt++;
+          cur = gptr<uint32_t>(d.next)[brow];
         }
-        cur = gptr<uint32_t>(d.next)[brow];
+      }
+    }
+    if (FILL) {
+      // exclusive wave prefix sum of cnt -> per-lane slice of one wave-wide
+      // reservation on the shared cursor
+      uint64_t pre = cnt;
+      for (int off = 1; off < 64; off <<= 1) {
+        uint64_t t = __shfl_up(pre, off, 64);
+        if (lane >= off) pre += t;
+      }
+      uint64_t waveTotal = __shfl(pre, 63, 64);
+      if (waveTotal == 0) continue;
+      uint64_t base = 0;
+      if (lane == 63)
+        base = atomicAdd((unsigned long long*)&d.counters[1],
+                         (unsigned long long)waveTotal);
+      base = __shfl(base, 63, 64) + (pre - cnt);
+      if (cnt) {
+        for (uint32_t cur = head; cur != 0;) {
+          uint32_t brow = cur - 1;
+          if (hjBuildKeyEq(d, brow, key, key1)) {
+            d.outBuild[base] = brow;
+            d.outProbe[base] = (uint32_t)row;
+            base++;
+          }
+          cur = gptr<uint32_t>(d.next)[brow];
+        }
       }
     } else {
       my += cnt;
@@ -1818,7 +1841,7 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
   }
   if (!FILL) {
     for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
-    if ((threadIdx.x & 63) == 0 && my)
+    if (lane == 0 && my)
       atomicAdd((unsigned long long*)&d.counters[0], (unsigned long long)my);
   }
 }

@@ -192,6 +192,111 @@ def _run_join_random(lib, bkeys, bpay, pkeys, ppay):
     return sorted(rows)
 
 
+def _run_join_topn(lib, bkeys, bpay, pkeys, ppay, limit, offset=0):
+    """ORDER BY (probe payload, build payload) + limit/offset over the joined
+    rows (TopNExec/SortExec over the join child)."""
+    b = P.Builder(lib)
+    bsrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    psrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)])
+    root = b.topn(j, [b.colref(3, GX_TYPE_I64), b.colref(1, GX_TYPE_I64)],
+                  [0, 0], limit, offset)
+    ex = b.build(root)
+
+    def chunk_of(k, p):
+        ch = PyChunk([GX_TYPE_I64, GX_TYPE_I64], len(k))
+        for col, arr in zip(ch.columns, (k, p)):
+            col.data[:len(arr) * 8] = arr.astype("<i8").view(np.uint8)
+            col.length = len(arr)
+        return ch
+
+    ex.bind_chunks(bsrc, [chunk_of(bkeys, bpay)])
+    ex.bind_chunks(psrc, [chunk_of(pkeys, ppay)])
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_I64] * 4)
+    ex.close()
+    ex.free()
+    b.free()
+    return rows
+
+
+def _topn_data():
+    rng = np.random.default_rng(7)
+    bkeys = rng.integers(0, 200, 1500)
+    bpay = np.arange(1500) * 7 - 5000          # unique tiebreaker
+    pkeys = rng.integers(0, 400, 10000)
+    ppay = np.arange(10000) - 3000             # unique primary key
+    return bkeys, bpay, pkeys, ppay
+
+
+def _topn_expected(limit, offset):
+    bkeys, bpay, pkeys, ppay = _topn_data()
+    rows = [(int(bk), int(bp), int(pk), int(pp))
+            for pk, pp in zip(pkeys, ppay)
+            for bk, bp in zip(bkeys, bpay) if bk == pk]
+    rows.sort(key=lambda r: (r[3], r[1]))
+    return rows[offset:offset + limit]
+
+
+def test_oracle_join_topn():
+    lib = load_oracle()
+    got = _run_join_topn(lib, *_topn_data(), limit=50, offset=5)
+    assert got == _topn_expected(50, 5)
+
+
+@pytest.mark.gpu
+def test_join_topn_parity(libs):
+    oracle, product = libs
+    data = _topn_data()
+    want = _run_join_topn(oracle, *data, limit=50, offset=5)
+    got = _run_join_topn(product, *data, limit=50, offset=5)
+    assert len(want) == 50
+    assert got == want
+
+
+@pytest.mark.gpu
+def test_join_full_orderby_parity(libs):
+    """Full ORDER BY (limit -1) over the joined rows."""
+    oracle, product = libs
+    rng = np.random.default_rng(9)
+    bkeys = rng.integers(0, 50, 300)
+    bpay = np.arange(300)
+    pkeys = rng.integers(0, 100, 2000)
+    ppay = np.arange(2000)
+
+    def run(lib):
+        b = P.Builder(lib)
+        bsrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+        psrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+        j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                       [b.colref(0, GX_TYPE_I64)])
+        root = b.sort(j, [b.colref(3, GX_TYPE_I64), b.colref(1, GX_TYPE_I64)],
+                      [1, 0])  # ppay desc, bpay asc
+
+        def chunk_of(k, p):
+            ch = PyChunk([GX_TYPE_I64, GX_TYPE_I64], len(k))
+            for col, arr in zip(ch.columns, (k, p)):
+                col.data[:len(arr) * 8] = arr.astype("<i8").view(np.uint8)
+                col.length = len(arr)
+            return ch
+
+        ex = b.build(root)
+        ex.bind_chunks(bsrc, [chunk_of(bkeys, bpay)])
+        ex.bind_chunks(psrc, [chunk_of(pkeys, ppay)])
+        ex.open()
+        rows = ex.pull_all([GX_TYPE_I64] * 4)
+        ex.close()
+        ex.free()
+        b.free()
+        return rows
+
+    want = run(oracle)
+    got = run(product)
+    assert len(want) > 5000
+    assert got == want
+
+
 @pytest.mark.gpu
 def test_join_parity_random_larger(libs):
     """2k build rows over 300 distinct keys (~7 duplicates per key) probed by

@@ -2767,7 +2767,7 @@ static int32_t runDeviceSort(gx_exec* ex) {
     void* outs[16];
     int sizes[16];
     int nc = tab.nCols;
-    if (nc > 8) { ex->err = "sort supports <= 8 columns this round"; return GX_ERR_INVALID; }
+    if (nc > 16) { ex->err = "sort supports <= 16 columns this round"; return GX_ERR_INVALID; }
     for (int c = 0; c < nc; c++) {
       gxp::DevCol& col = tab.cols[c];
       if (col.type == GX_TYPE_STRING && !col.denseOffsets) {
@@ -2789,6 +2789,17 @@ static int32_t runDeviceSort(gx_exec* ex) {
     for (int c = 0; c < nc; c++)
       grc |= gxp::gxSortGatherCol(ins[c], outs[c], idxA, n, sizes[c],
                                   ex->stream);
+    // null bitmaps ride the same permutation (joined-table sort; generator
+    // sources have none)
+    for (int c = 0; c < nc; c++) {
+      gxp::DevCol& col = tab.cols[c];
+      if (col.hasNulls && col.nullBitmap) {
+        uint8_t* nb = (uint8_t*)devAlloc(ex, (n + 7) / 8);
+        if (!nb) { ex->err = "hipMalloc failed (null gather)"; return GX_ERR_INTERNAL; }
+        grc |= gxp::gxGatherNulls(col.nullBitmap, idxA, nb, n, ex->stream);
+        col.nullBitmap = nb;
+      }
+    }
     if (grc) {
       ex->err = "gather launch failed";
       return GX_ERR_INTERNAL;
@@ -3292,6 +3303,42 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     int32_t rc = compileHashJoin(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";
     (void)rc;
+  } else if (rn.kind == PK_TOPN &&
+             ex->plan.nodes[rn.child].kind == PK_HASHJOIN) {
+    // ORDER BY / TopN over the joined rows: standalone join, then the device
+    // radix sort over the materialized join output table (sortexec/sort.go
+    // over a join child)
+    int saved = ex->root;
+    ex->root = rn.child;
+    int32_t rc = compileHashJoin(ex);
+    ex->root = saved;
+    if (rc != GX_OK) {
+      if (ex->err.empty()) ex->err = "join plan compilation failed";
+      return ex;
+    }
+    ex->devSortLimit = rn.limit;
+    ex->devSortOffset = rn.offset;
+    for (size_t i = 0; i < rn.exprs.size(); i++) {
+      const PExpr& ke = ex->plan.exprs[rn.exprs[i]];
+      if (ke.kind != EK_COLREF || ke.colIdx < 0 ||
+          ke.colIdx >= ex->desc.table.nCols) {
+        ex->err = "sort keys must be join output columns";
+        return ex;
+      }
+      gxp::SortKeyCompose k{};
+      k.col = ke.colIdx;
+      k.desc = rn.keyDesc[i] != 0;
+      int t = ex->desc.table.cols[ke.colIdx].type;
+      if (t == GX_TYPE_I64) k.kind = 0;
+      else if (t == GX_TYPE_TIME) k.kind = 1;
+      else if (t == GX_TYPE_STRING) k.kind = 2;  // dense char checked at run
+      else if (t == GX_TYPE_DECIMAL) k.kind = 3;
+      else {
+        ex->err = "unsupported sort key type this round";
+        return ex;
+      }
+      ex->devSortKeys.push_back(k);
+    }
   } else {
     ex->err = "unsupported root plan node for the device engine this round";
   }
@@ -3382,6 +3429,13 @@ int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
         return rc;
       }
       ex->ranQuery = true;
+    }
+    if (!ex->devSortKeys.empty() && !ex->devSorted) {
+      int32_t rc = runDeviceSort(ex);
+      if (rc) {
+        *rows_out = 0;
+        return rc;
+      }
     }
     return emitTableChunk(ex, out, rows_out);
   }

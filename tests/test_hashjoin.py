@@ -192,6 +192,72 @@ def _run_join_random(lib, bkeys, bpay, pkeys, ppay):
     return sorted(rows)
 
 
+def _run_join_filtered(lib, bkeys, bpay, pkeys, ppay, topn_limit=None):
+    """Selection over the join = the join's other conditions
+    (inner_join_probe.go:75): build.payload < probe.payload AND
+    probe.payload < 9000."""
+    b = P.Builder(lib)
+    bsrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    psrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)])
+    colcol = b.call(GX_F_LT, GX_TYPE_I64, 0, b.colref(1, GX_TYPE_I64),
+                    b.colref(3, GX_TYPE_I64))
+    colconst = b.call(GX_F_LT, GX_TYPE_I64, 0, b.colref(3, GX_TYPE_I64),
+                      b.const_i64(9000))
+    root = b.selection(j, [colcol, colconst])
+    if topn_limit is not None:
+        root = b.topn(root, [b.colref(3, GX_TYPE_I64),
+                             b.colref(1, GX_TYPE_I64)], [0, 0], topn_limit)
+
+    def chunk_of(k, p):
+        ch = PyChunk([GX_TYPE_I64, GX_TYPE_I64], len(k))
+        for col, arr in zip(ch.columns, (k, p)):
+            col.data[:len(arr) * 8] = arr.astype("<i8").view(np.uint8)
+            col.length = len(arr)
+        return ch
+
+    ex = b.build(root)
+    ex.bind_chunks(bsrc, [chunk_of(bkeys, bpay)])
+    ex.bind_chunks(psrc, [chunk_of(pkeys, ppay)])
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_I64] * 4)
+    ex.close()
+    ex.free()
+    b.free()
+    return rows if topn_limit is not None else sorted(rows)
+
+
+def test_oracle_join_other_condition():
+    lib = load_oracle()
+    bkeys, bpay, pkeys, ppay = _topn_data()
+    got = _run_join_filtered(lib, bkeys, bpay, pkeys, ppay)
+    want = sorted((int(bk), int(bp), int(pk), int(pp))
+                  for pk, pp in zip(pkeys, ppay)
+                  for bk, bp in zip(bkeys, bpay)
+                  if bk == pk and bp < pp and pp < 9000)
+    assert got == want
+    assert 0 < len(got)
+
+
+@pytest.mark.gpu
+def test_join_other_condition_parity(libs):
+    oracle, product = libs
+    data = _topn_data()
+    assert _run_join_filtered(product, *data) == _run_join_filtered(oracle, *data)
+
+
+@pytest.mark.gpu
+def test_join_other_condition_topn_parity(libs):
+    """TopN over Selection over HashJoin — the full composed pipeline."""
+    oracle, product = libs
+    data = _topn_data()
+    want = _run_join_filtered(oracle, *data, topn_limit=40)
+    got = _run_join_filtered(product, *data, topn_limit=40)
+    assert len(want) == 40
+    assert got == want
+
+
 def _run_join_topn(lib, bkeys, bpay, pkeys, ppay, limit, offset=0):
     """ORDER BY (probe payload, build payload) + limit/offset over the joined
     rows (TopNExec/SortExec over the join child)."""

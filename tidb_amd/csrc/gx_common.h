@@ -291,6 +291,19 @@ struct JoinAggDesc {
 
 enum { PRED_STR_EQ_CONST = 3 };  // extra PredKind for the join path
 
+// one conjunct of a post-join filter (NULL operand rejects the row, the
+// VecEvalBool NULL semantics, expression.go:420-504)
+struct JoinPostPred {
+  int32_t kind = 0;    // 0 = side-local <col cmp const>; 1 = <col cmp col>
+  int32_t side = 0;    // kind 0: 0 = build, 1 = probe (pd.col is side-local)
+  PredDesc pd{};       // kind 0
+  uint8_t strC[16] = {0};
+  int32_t strCLen = 0;
+  int32_t lcol = 0, rcol = 0;  // kind 1: join OUTPUT column indexes
+  int32_t cmp = 0;             // kind 1: GX_F_LT..GX_F_NE
+  int32_t ctype = 0;           // kind 1: GX_TYPE_I64 or GX_TYPE_TIME
+};
+
 // ---- standalone hash join (inner, duplicate build keys) ----
 // HashJoinV2 equivalent (join/hash_join_v2.go): chained hash table over the
 // build side — heads + intrusive per-row next links (hash_table_v2.go:22-53
@@ -323,8 +336,18 @@ struct HashJoinDesc {
   uint32_t* next = nullptr;   // per build row: next chain row+1, 0 = end
   uint32_t* outBuild = nullptr;  // match pairs (fill phase)
   uint32_t* outProbe = nullptr;
-  uint64_t* counters = nullptr;  // [0] count-phase total, [1] fill cursor
+  uint64_t* counters = nullptr;  // [0] count-phase total, [1] fill cursor,
+                                 // [2] post-filter cursor
   uint32_t* errorFlag = nullptr;
+  // post-join filter (the join's "other conditions": VectorizedFilter over
+  // the joined chunk, inner_join_probe.go:75 — expressed as a Selection
+  // above the join). CNF; evaluated on the match pairs BEFORE the gather so
+  // rejected rows never touch HBM output.
+  JoinPostPred post[4];
+  int32_t nPost = 0;
+  int64_t nPairs = 0;            // fill-phase total (filter input size)
+  uint32_t* outBuild2 = nullptr; // filter-surviving pairs
+  uint32_t* outProbe2 = nullptr;
 };
 
 // phases: 0 = build (chain insert), 1 = count matches, 2 = fill match pairs

@@ -983,9 +983,9 @@ static int32_t compileJoinAgg(gx_exec* ex) {
 // Source children): the general HashJoinV2 operator (join/hash_join_v2.go)
 // with duplicate build keys via chained table — output is the joined rows
 // themselves (build cols ++ probe cols), not an aggregate.
-static int32_t compileHashJoin(gx_exec* ex) {
+static int32_t compileHashJoin(gx_exec* ex, int joinNode) {
   const PPlan& plan = ex->plan;
-  const PNode& jn = plan.nodes[ex->root];
+  const PNode& jn = plan.nodes[joinNode];
   if (jn.joinType != 0) {
     ex->err = "only inner joins on device this round";
     return GX_ERR_INVALID;
@@ -1058,6 +1058,87 @@ static int32_t compileHashJoin(gx_exec* ex) {
     setDevColMeta(&ex->desc.table.cols[nb + c], pN.colTypes[c], pN.colFracs[c]);
   ex->isHashJoin = true;
   return GX_OK;
+}
+
+// post-join filter conjuncts (a Selection over the join = the join's "other
+// conditions", inner_join_probe.go:75): <output col cmp const> or
+// <output col cmp output col> (e.g. build.x < probe.y)
+static int32_t compilePostJoinPreds(gx_exec* ex, const PNode& sel,
+                                    const PNode& bN, const PNode& pN) {
+  const PPlan& plan = ex->plan;
+  gxp::HashJoinDesc& hj = ex->hj;
+  int nb = (int)bN.colTypes.size();
+  if (sel.exprs.size() > 4) {
+    ex->err = "too many post-join filter conjuncts this round";
+    return GX_ERR_INVALID;
+  }
+  // join-output schema for const preds
+  PNode outN;
+  outN.colTypes = bN.colTypes;
+  outN.colFracs = bN.colFracs;
+  outN.colTypes.insert(outN.colTypes.end(), pN.colTypes.begin(), pN.colTypes.end());
+  outN.colFracs.insert(outN.colFracs.end(), pN.colFracs.begin(), pN.colFracs.end());
+  for (int cid : sel.exprs) {
+    const PExpr& e = plan.exprs[cid];
+    gxp::JoinPostPred q{};
+    bool colcol = e.kind == EK_CALL && e.args.size() == 2 &&
+                  plan.exprs[e.args[0]].kind == EK_COLREF &&
+                  plan.exprs[e.args[1]].kind == EK_COLREF;
+    if (colcol) {
+      if (e.func > GX_F_NE) {
+        ex->err = "unsupported post-join predicate";
+        return GX_ERR_INVALID;
+      }
+      const PExpr& l = plan.exprs[e.args[0]];
+      const PExpr& r = plan.exprs[e.args[1]];
+      if (l.colIdx < 0 || l.colIdx >= (int)outN.colTypes.size() ||
+          r.colIdx < 0 || r.colIdx >= (int)outN.colTypes.size()) {
+        ex->err = "post-join filter column out of range";
+        return GX_ERR_INVALID;
+      }
+      int lt = outN.colTypes[l.colIdx], rt = outN.colTypes[r.colIdx];
+      if (lt != rt || (lt != GX_TYPE_I64 && lt != GX_TYPE_TIME)) {
+        ex->err = "post-join col-col compare supports int64/time this round";
+        return GX_ERR_INVALID;
+      }
+      q.kind = 1;
+      q.lcol = l.colIdx;
+      q.rcol = r.colIdx;
+      q.cmp = e.func;
+      q.ctype = lt;
+    } else {
+      if (!compileTablePred(ex, outN, cid, &q.pd, q.strC, &q.strCLen))
+        return GX_ERR_INVALID;
+      q.kind = 0;
+      q.side = q.pd.col < nb ? 0 : 1;
+      if (q.side == 1) q.pd.col -= nb;
+    }
+    hj.post[hj.nPost++] = q;
+  }
+  return GX_OK;
+}
+
+// [Selection ->] HashJoin subtree (the Selection = post-join other
+// conditions)
+static int32_t compileHashJoinTree(gx_exec* ex, int node) {
+  const PNode* n = &ex->plan.nodes[node];
+  int selNode = -1;
+  if (n->kind == PK_SELECTION) {
+    selNode = node;
+    node = n->child;
+    n = &ex->plan.nodes[node];
+  }
+  if (n->kind != PK_HASHJOIN) {
+    ex->err = "expected a hash join under the selection";
+    return GX_ERR_INVALID;
+  }
+  int32_t rc = compileHashJoin(ex, node);
+  if (rc) return rc;
+  if (selNode >= 0)
+    rc = compilePostJoinPreds(ex, ex->plan.nodes[selNode],
+                              ex->plan.nodes[ex->hjSrcB],
+                              ex->plan.nodes[ex->hjSrcP]);
+  return rc;
 }
 
 // compile the fused Source->[Selection]->[Projection]->HashAgg pipeline
@@ -2416,7 +2497,7 @@ static int32_t runHashJoin(gx_exec* ex) {
     rc = materializeTable(ex, ex->hjSrcP, &hj.probe);
     if (rc) return rc;
     ex->devErr = (uint32_t*)devAlloc(ex, 4);
-    hj.counters = (uint64_t*)devAlloc(ex, 2 * 8);
+    hj.counters = (uint64_t*)devAlloc(ex, 3 * 8);
     ex->devHj = (gxp::HashJoinDesc*)devAlloc(ex, sizeof(gxp::HashJoinDesc));
     if (!ex->devErr || !hj.counters || !ex->devHj) {
       ex->err = "hipMalloc failed";
@@ -2453,7 +2534,7 @@ static int32_t runHashJoin(gx_exec* ex) {
   hj.next = (uint32_t*)devAlloc(ex, std::max<int64_t>(nb, 1) * 4);
   if (!hj.heads || !hj.next) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
   HIP_OK(ex, hipMemsetAsync(hj.heads, 0, (1ULL << hj.headsLog2) * 4, ex->stream));
-  HIP_OK(ex, hipMemsetAsync(hj.counters, 0, 16, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(hj.counters, 0, 24, ex->stream));
   HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
   HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj), hipMemcpyHostToDevice,
                             ex->stream));
@@ -2495,12 +2576,38 @@ static int32_t runHashJoin(gx_exec* ex) {
       return GX_ERR_INTERNAL;
     }
     HIP_OK(ex, hipEventRecord(evF, ex->stream));
+    // post-join filter (other conditions): compact surviving pairs before
+    // the gather so rejected rows never touch the output
+    const uint32_t* gatherB = hj.outBuild;
+    const uint32_t* gatherP = hj.outProbe;
+    if (hj.nPost > 0) {
+      hj.nPairs = (int64_t)total;
+      hj.outBuild2 = (uint32_t*)devAlloc(ex, total * 4);
+      hj.outProbe2 = (uint32_t*)devAlloc(ex, total * 4);
+      if (!hj.outBuild2 || !hj.outProbe2) {
+        ex->err = "hipMalloc failed";
+        return GX_ERR_INTERNAL;
+      }
+      HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj),
+                                hipMemcpyHostToDevice, ex->stream));
+      if (gxp::gxHashJoinPhase(3, ex->devHj, hj, ex->stream) != 0) {
+        ex->err = "join filter launch failed";
+        return GX_ERR_INTERNAL;
+      }
+      HIP_OK(ex, hipStreamSynchronize(ex->stream));
+      uint64_t total2 = 0;
+      HIP_OK(ex, hipMemcpy(&total2, hj.counters + 2, 8, hipMemcpyDeviceToHost));
+      total = total2;
+      ex->lastSelCount = total;
+      gatherB = hj.outBuild2;
+      gatherP = hj.outProbe2;
+    }
     // gather every output column through its side's match index
-    for (int c = 0; c < ex->desc.table.nCols; c++) {
+    for (int c = 0; c < ex->desc.table.nCols && total > 0; c++) {
       int nbc = hj.build.nCols;
       const gxp::DevCol& src =
           c < nbc ? hj.build.cols[c] : hj.probe.cols[c - nbc];
-      const uint32_t* idx = c < nbc ? hj.outBuild : hj.outProbe;
+      const uint32_t* idx = c < nbc ? gatherB : gatherP;
       gxp::DevCol& dst = ex->desc.table.cols[c];
       int es = src.type == GX_TYPE_DECIMAL ? 40
                : (src.type == GX_TYPE_STRING ? 1 : 8);
@@ -3289,7 +3396,11 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
       ex->devSortKeys.push_back(k);
     }
     (void)ok;
-  } else if (rn.kind == PK_TOPN) {
+  } else if (rn.kind == PK_TOPN &&
+             ex->plan.nodes[rn.child].kind != PK_HASHJOIN &&
+             !(ex->plan.nodes[rn.child].kind == PK_SELECTION &&
+               ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
+                   PK_HASHJOIN)) {
     int32_t rc = compileJoinAgg(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";
     (void)rc;
@@ -3299,19 +3410,21 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     ex->desc.table.nCols = (int)rn.colTypes.size();
     for (size_t c = 0; c < rn.colTypes.size(); c++)
       setDevColMeta(&ex->desc.table.cols[c], rn.colTypes[c], rn.colFracs[c]);
-  } else if (rn.kind == PK_HASHJOIN) {
-    int32_t rc = compileHashJoin(ex);
+  } else if (rn.kind == PK_HASHJOIN ||
+             (rn.kind == PK_SELECTION &&
+              ex->plan.nodes[rn.child].kind == PK_HASHJOIN)) {
+    int32_t rc = compileHashJoinTree(ex, root);
     if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";
     (void)rc;
   } else if (rn.kind == PK_TOPN &&
-             ex->plan.nodes[rn.child].kind == PK_HASHJOIN) {
-    // ORDER BY / TopN over the joined rows: standalone join, then the device
-    // radix sort over the materialized join output table (sortexec/sort.go
-    // over a join child)
-    int saved = ex->root;
-    ex->root = rn.child;
-    int32_t rc = compileHashJoin(ex);
-    ex->root = saved;
+             (ex->plan.nodes[rn.child].kind == PK_HASHJOIN ||
+              (ex->plan.nodes[rn.child].kind == PK_SELECTION &&
+               ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
+                   PK_HASHJOIN))) {
+    // ORDER BY / TopN over the joined rows: standalone join (+ post-join
+    // filter), then the device radix sort over the materialized join output
+    // table (sortexec/sort.go over a join child)
+    int32_t rc = compileHashJoinTree(ex, rn.child);
     if (rc != GX_OK) {
       if (ex->err.empty()) ex->err = "join plan compilation failed";
       return ex;

@@ -1846,6 +1846,74 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
   }
 }
 
+// evaluate the post-join CNF on one match pair; NULL operands reject
+// (VecEvalBool semantics)
+__device__ inline bool hjPostPass(const HashJoinDesc& d, uint32_t brow,
+                                  uint32_t prow) {
+  for (int p = 0; p < d.nPost; p++) {
+    const JoinPostPred& q = d.post[p];
+    if (q.kind == 0) {
+      const DevTable& t = q.side == 0 ? d.build : d.probe;
+      int64_t row = q.side == 0 ? (int64_t)brow : (int64_t)prow;
+      if (!evalSimplePred(t, q.pd, q.strC, q.strCLen, row)) return false;
+    } else {
+      int nbc = d.build.nCols;
+      const DevCol& lc = q.lcol < nbc ? d.build.cols[q.lcol]
+                                      : d.probe.cols[q.lcol - nbc];
+      const DevCol& rc = q.rcol < nbc ? d.build.cols[q.rcol]
+                                      : d.probe.cols[q.rcol - nbc];
+      int64_t lrow = q.lcol < nbc ? (int64_t)brow : (int64_t)prow;
+      int64_t rrow = q.rcol < nbc ? (int64_t)brow : (int64_t)prow;
+      if (colIsNull(lc, lrow) || colIsNull(rc, rrow)) return false;
+      int c;
+      if (q.ctype == 3 /*GX_TYPE_TIME*/) {
+        uint64_t a = gptr<uint64_t>(lc.data)[lrow] & ~0xFULL;
+        uint64_t b = gptr<uint64_t>(rc.data)[rrow] & ~0xFULL;
+        c = a < b ? -1 : (a > b ? 1 : 0);
+      } else {
+        int64_t a = gptr<int64_t>(lc.data)[lrow];
+        int64_t b = gptr<int64_t>(rc.data)[rrow];
+        c = a < b ? -1 : (a > b ? 1 : 0);
+      }
+      if (!cmpResult(c, q.cmp)) return false;
+    }
+  }
+  return true;
+}
+
+// compact filter-surviving match pairs (wave-aggregated cursor, one atomic
+// per wavefront)
+__global__ void hjFilterPairsKernel(const HashJoinDesc* __restrict__ dp) {
+  const HashJoinDesc& d = *dp;
+  int64_t n = d.nPairs;
+  int lane = threadIdx.x & 63;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;;
+       i += stride) {
+    bool active = i < n;
+    if (__ballot(active) == 0) break;
+    uint32_t brow = 0, prow = 0;
+    bool keep = false;
+    if (active) {
+      brow = gptr<uint32_t>(d.outBuild)[i];
+      prow = gptr<uint32_t>(d.outProbe)[i];
+      keep = hjPostPass(d, brow, prow);
+    }
+    uint64_t m = __ballot(keep);
+    if (m == 0) continue;
+    uint64_t base = 0;
+    if (lane == 0)
+      base = atomicAdd((unsigned long long*)&d.counters[2],
+                       (unsigned long long)__popcll(m));
+    base = __shfl(base, 0, 64);
+    if (keep) {
+      uint64_t off = __popcll(m & ((1ULL << lane) - 1));
+      d.outBuild2[base + off] = brow;
+      d.outProbe2[base + off] = prow;
+    }
+  }
+}
+
 // gather a null bitmap through the match index: one thread composes one
 // output byte (8 rows) — no atomics (LSB-first, 1 = NOT NULL)
 __global__ void hjGatherNullsKernel(const uint8_t* __restrict__ in,
@@ -1957,7 +2025,9 @@ int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,
 
 int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
                     const HashJoinDesc& h, void* stream) {
-  int64_t rows = phase == 0 ? h.build.nRows : h.probe.nRows;
+  int64_t rows = phase == 0 ? h.build.nRows
+                 : phase == 3 ? h.nPairs
+                              : h.probe.nRows;
   if (rows == 0) return 0;
   dim3 g(gridFor(rows));
   if (phase == 0)
@@ -1966,8 +2036,11 @@ int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
   else if (phase == 1)
     hipLaunchKernelGGL(hjProbeKernel<false>, g, dim3(256), 0,
                        (hipStream_t)stream, devDesc);
-  else
+  else if (phase == 2)
     hipLaunchKernelGGL(hjProbeKernel<true>, g, dim3(256), 0,
+                       (hipStream_t)stream, devDesc);
+  else
+    hipLaunchKernelGGL(hjFilterPairsKernel, g, dim3(256), 0,
                        (hipStream_t)stream, devDesc);
   return (int)hipGetLastError();
 }

@@ -88,3 +88,32 @@ def test_topn_source_limit_offset_parity():
                  offset=7)
     assert len(a) == len(b) == 100
     assert [(r[7], r[0]) for r in a] == [(r[7], r[0]) for r in b]
+
+
+@pytest.mark.gpu
+def test_spill_sort_parity(monkeypatch):
+    """Out-of-core sort: GX_SORT_RUN_ROWS forces 7 device-sorted runs spilled
+    to host + k-way merge on emission (sort_spill.go / multi_way_merge.go
+    analog). Same key order and row multiset as the in-HBM sort / oracle."""
+    from tests.gxlib import load_product
+    a = run_sort(load_oracle(), KEYS_2, [0, 1], n_rows=100000)
+    monkeypatch.setenv("GX_SORT_RUN_ROWS", "15000")
+    b = run_sort(load_product(), KEYS_2, [0, 1], n_rows=100000)
+    monkeypatch.delenv("GX_SORT_RUN_ROWS")
+    assert len(a) == len(b) == 100000
+    assert [(r[7], r[0]) for r in a] == [(r[7], r[0]) for r in b]
+    assert sorted(map(tuple, a)) == sorted(map(tuple, b))
+
+
+@pytest.mark.gpu
+def test_spill_topn_limit_offset_parity(monkeypatch):
+    """limit/offset served from the spill merge (TopN over a spilled sort)."""
+    from tests.gxlib import load_product
+    a = run_sort(load_oracle(), KEYS_2, [0, 1], n_rows=50000, limit=200,
+                 offset=13)
+    monkeypatch.setenv("GX_SORT_RUN_ROWS", "9000")
+    b = run_sort(load_product(), KEYS_2, [0, 1], n_rows=50000, limit=200,
+                 offset=13)
+    monkeypatch.delenv("GX_SORT_RUN_ROWS")
+    assert len(a) == len(b) == 200
+    assert [(r[7], r[0]) for r in a] == [(r[7], r[0]) for r in b]

@@ -88,6 +88,17 @@ struct Binding {
   uint64_t tpchSeed = 42;
 };
 
+// one host-resident sorted run of the out-of-core sort (the spill container
+// analog of sortexec/sort_spill.go + chunk.DataInDiskByChunks — host RAM
+// stands in for disk; runs stream back through the k-way merge on emission,
+// multi_way_merge.go semantics)
+struct SortRun {
+  std::vector<std::vector<uint8_t>> colData;   // per col: n x elem bytes
+  std::vector<std::vector<uint64_t>> keys;     // per sort key: composed
+                                               // order-preserving u64 per row
+  int64_t n = 0;
+};
+
 struct OutRowVal {
   bool isNull = false;
   int type = GX_TYPE_I64;
@@ -156,6 +167,14 @@ struct gx_exec {
   std::vector<gxp::SortKeyCompose> devSortKeys;
   bool devSorted = false;
   int64_t devSortLimit = -1, devSortOffset = 0;
+  // out-of-core sort: device-sorted runs spilled to host, k-way merged on
+  // emission (triggered when the table exceeds the HBM budget or
+  // GX_SORT_RUN_ROWS forces a run size)
+  bool spillSorted = false;
+  std::vector<SortRun> sortRuns;
+  std::vector<int64_t> runPos;
+  int64_t spillSkip = 0;        // offset countdown
+  int64_t spillRemaining = -1;  // limit countdown (-1 = unlimited)
   uint64_t lastSelCount = 0;
   double lastKernelMs = 0;
 
@@ -2929,6 +2948,226 @@ static int32_t runDeviceSort(gx_exec* ex) {
   return GX_OK;
 }
 
+// ---------------- out-of-core sort (spill runs + k-way merge) ----------------
+
+static void freeSince(gx_exec* ex, size_t mark) {
+  for (size_t i = mark; i < ex->devBufs.size(); i++)
+    (void)hipFree(ex->devBufs[i]);
+  ex->devBufs.resize(mark);
+}
+
+// sort a source larger than the HBM budget: materialize + device-radix-sort
+// row-range runs, download each sorted run (columns + composed order keys)
+// to host, free the device buffers, then k-way merge on emission.
+static int32_t runDeviceSortSpill(gx_exec* ex, int64_t runRows) {
+  if (!gpuAvailable()) {
+    ex->err = "no MI355X visible: the product engine has no CPU fallback "
+              "(GX_ERR_NO_GPU)";
+    return GX_ERR_NO_GPU;
+  }
+  if (ex->device >= 0) hipSetDevice(ex->device);
+  if (!ex->stream) HIP_OK(ex, hipStreamCreate(&ex->stream));
+  if (!ex->devErr) {
+    ex->devErr = (uint32_t*)devAlloc(ex, 4);
+    if (!ex->devErr) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+  }
+  Binding& b = ex->bindings[ex->sourceNode];
+  const int64_t total = b.tpchRows;
+  const int64_t saveOffB = b.tpchRowOffset;
+  const int64_t saveTotB = b.tpchTotalRows > 0 ? b.tpchTotalRows : total;
+  gxp::DevTable saveTab = ex->desc.table;
+  const int64_t saveLim = ex->devSortLimit, saveOff = ex->devSortOffset;
+  const int nk = (int)ex->devSortKeys.size();
+  const int ncols = saveTab.nCols;
+  double sumMs = 0;
+  int32_t rc = GX_OK;
+  for (int64_t start = 0; start < total && rc == GX_OK; start += runRows) {
+    int64_t n = std::min(runRows, total - start);
+    size_t mark = ex->devBufs.size();
+    b.tpchRowOffset = saveOffB + start;
+    b.tpchRows = n;
+    b.tpchTotalRows = saveTotB;
+    gxp::DevTable runTab{};
+    rc = materializeTable(ex, ex->sourceNode, &runTab);
+    if (rc == GX_OK) {
+      for (int c = 0; c < runTab.nCols && rc == GX_OK; c++)
+        if (runTab.cols[c].type == GX_TYPE_STRING &&
+            !runTab.cols[c].denseOffsets) {
+          ex->err = "general varlen column in spill sort unsupported this round";
+          rc = GX_ERR_INVALID;
+        }
+    }
+    if (rc == GX_OK) {
+      ex->desc.table = runTab;
+      ex->devSortLimit = -1;
+      ex->devSortOffset = 0;
+      ex->devSorted = false;
+      rc = runDeviceSort(ex);
+      sumMs += ex->lastKernelMs;
+    }
+    if (rc == GX_OK) {
+      // download the sorted run + its composed order keys
+      gxp::DevTable& st = ex->desc.table;
+      SortRun run;
+      run.n = n;
+      run.colData.resize(ncols);
+      run.keys.resize(nk);
+      uint32_t* iden = (uint32_t*)devAlloc(ex, n * 4);
+      uint64_t* kbuf = (uint64_t*)devAlloc(ex, n * 8);
+      if (!iden || !kbuf) { ex->err = "hipMalloc failed"; rc = GX_ERR_INTERNAL; }
+      if (rc == GX_OK && gxp::gxSortIota(iden, n, ex->stream) != 0) {
+        ex->err = "iota launch failed";
+        rc = GX_ERR_INTERNAL;
+      }
+      for (int j = 0; j < nk && rc == GX_OK; j++) {
+        if (gxp::gxSortComposeKeys(nullptr, st, ex->devSortKeys[j], iden,
+                                   kbuf, n, ex->devErr, ex->stream) != 0) {
+          ex->err = "key compose launch failed";
+          rc = GX_ERR_INTERNAL;
+          break;
+        }
+        run.keys[j].resize(n);
+        if (hipMemcpy(run.keys[j].data(), kbuf, n * 8,
+                      hipMemcpyDeviceToHost) != hipSuccess) {
+          ex->err = "key download failed";
+          rc = GX_ERR_INTERNAL;
+        }
+      }
+      for (int c = 0; c < ncols && rc == GX_OK; c++) {
+        const gxp::DevCol& col = st.cols[c];
+        int es = col.type == GX_TYPE_DECIMAL ? 40
+                 : (col.type == GX_TYPE_STRING ? 1 : 8);
+        run.colData[c].resize((size_t)n * es);
+        if (hipMemcpy(run.colData[c].data(), col.data, (size_t)n * es,
+                      hipMemcpyDeviceToHost) != hipSuccess) {
+          ex->err = "run download failed";
+          rc = GX_ERR_INTERNAL;
+        }
+      }
+      if (rc == GX_OK) ex->sortRuns.push_back(std::move(run));
+    }
+    freeSince(ex, mark);
+  }
+  b.tpchRowOffset = saveOffB;
+  b.tpchRows = total;
+  b.tpchTotalRows = saveTotB;
+  ex->desc.table = saveTab;
+  ex->devSortLimit = saveLim;
+  ex->devSortOffset = saveOff;
+  ex->devSorted = false;
+  if (rc != GX_OK) return rc;
+  ex->lastKernelMs = sumMs;
+  ex->runPos.assign(ex->sortRuns.size(), 0);
+  ex->spillSkip = saveOff;
+  ex->spillRemaining = saveLim;
+  ex->spillSorted = true;
+  if (getenv("GX_DEBUG"))
+    fprintf(stderr, "[gx] spill sort: %zu runs of <=%lld rows, %.3f ms device\n",
+            ex->sortRuns.size(), (long long)runRows, sumMs);
+  return GX_OK;
+}
+
+// decide between the in-HBM sort and the spill path (env override or HBM
+// budget); GX_OK with spillSorted unset means: use the in-HBM path
+static int32_t maybeSpillSort(gx_exec* ex) {
+  auto it = ex->bindings.find(ex->sourceNode);
+  if (it == ex->bindings.end() || it->second.haveChunks ||
+      it->second.tpchTable < 0)
+    return GX_OK;  // bound chunks are host-resident already: in-HBM path
+  int64_t total = it->second.tpchRows;
+  const PNode& srcN = ex->plan.nodes[ex->sourceNode];
+  int64_t rowBytes = 0;
+  for (int t : srcN.colTypes)
+    rowBytes += t == GX_TYPE_DECIMAL ? 40 : (t == GX_TYPE_STRING ? 9 : 8);
+  int64_t runRows = 0;
+  if (const char* e = getenv("GX_SORT_RUN_ROWS")) runRows = atoll(e);
+  if (runRows <= 0) {
+    if (!gpuAvailable()) return GX_OK;  // NO_GPU surfaces on the normal path
+    if (ex->device >= 0) hipSetDevice(ex->device);
+    size_t freeB = 0, totB = 0;
+    if (hipMemGetInfo(&freeB, &totB) != hipSuccess) return GX_OK;
+    // in-HBM sort needs table + gathered copy + idx/key scratch (~24 B/row)
+    double need = (double)total * (2.0 * rowBytes + 24.0);
+    if (need < 0.6 * (double)freeB) return GX_OK;
+    runRows = (int64_t)(0.25 * (double)freeB / (2.0 * rowBytes + 24.0));
+    if (runRows < 1024) runRows = 1024;
+  }
+  if (total <= runRows) return GX_OK;
+  return runDeviceSortSpill(ex, runRows);
+}
+
+// k-way merge emission over the host-resident sorted runs
+// (multi_way_merge.go): composed u64 keys compare lexicographically
+static int32_t emitSpillChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
+  const PNode& srcN = ex->plan.nodes[ex->sourceNode];
+  int ncols = (int)srcN.colTypes.size();
+  if (out->n_cols != ncols) {
+    ex->err = "output chunk column count mismatch";
+    return GX_ERR_INVALID;
+  }
+  int nk = (int)ex->devSortKeys.size();
+  size_t R = ex->sortRuns.size();
+  for (int c = 0; c < ncols; c++) {
+    gx_col* g = &out->cols[c];
+    int es = srcN.colTypes[c] == GX_TYPE_DECIMAL ? 40
+             : (srcN.colTypes[c] == GX_TYPE_STRING ? 1 : 8);
+    if (g->data_cap < 1024 * es ||
+        (srcN.colTypes[c] == GX_TYPE_STRING && g->offsets_cap < 1025)) {
+      ex->err = "output buffer too small";
+      return GX_ERR_INVALID;
+    }
+    if (g->offsets) g->offsets[0] = 0;
+  }
+  int n = 0;
+  while (n < 1024 && ex->spillRemaining != 0) {
+    int best = -1;
+    for (size_t r = 0; r < R; r++) {
+      int64_t pos = ex->runPos[r];
+      if (pos >= ex->sortRuns[r].n) continue;
+      if (best < 0) {
+        best = (int)r;
+        continue;
+      }
+      const SortRun& a = ex->sortRuns[r];
+      const SortRun& bs = ex->sortRuns[best];
+      int64_t bpos = ex->runPos[best];
+      for (int j = 0; j < nk; j++) {
+        uint64_t ka = a.keys[j][pos], kb = bs.keys[j][bpos];
+        if (ka < kb) { best = (int)r; break; }
+        if (ka > kb) break;
+      }
+    }
+    if (best < 0) break;
+    int64_t pos = ex->runPos[best]++;
+    if (ex->spillSkip > 0) {
+      ex->spillSkip--;
+      continue;
+    }
+    const SortRun& run = ex->sortRuns[best];
+    for (int c = 0; c < ncols; c++) {
+      gx_col* g = &out->cols[c];
+      if (srcN.colTypes[c] == GX_TYPE_STRING) {
+        ((uint8_t*)g->data)[n] = run.colData[c][pos];
+        g->offsets[n + 1] = n + 1;
+      } else {
+        int es = srcN.colTypes[c] == GX_TYPE_DECIMAL ? 40 : 8;
+        std::memcpy((uint8_t*)g->data + (size_t)n * es,
+                    run.colData[c].data() + (size_t)pos * es, es);
+      }
+    }
+    n++;
+    if (ex->spillRemaining > 0) ex->spillRemaining--;
+  }
+  for (int c = 0; c < ncols; c++) {
+    gx_col* g = &out->cols[c];
+    if (g->null_bitmap) std::memset(g->null_bitmap, 0xFF, (n + 7) / 8);
+    g->length = n;
+  }
+  out->n_rows = n;
+  *rows_out = n;
+  return GX_OK;
+}
+
 // emit the next <=1024 rows of the device-resident table in ex->desc.table
 // (bare/sorted sources and joined-row output)
 static int32_t emitTableChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
@@ -2986,6 +3225,13 @@ static int32_t emitTableChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
 }
 
 static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
+  // out-of-core sort decision comes BEFORE full materialization (spill runs
+  // materialize row ranges themselves)
+  if (!ex->devSortKeys.empty() && !ex->devSorted && !ex->spillSorted) {
+    int32_t rc = maybeSpillSort(ex);
+    if (rc) return rc;
+  }
+  if (ex->spillSorted) return emitSpillChunk(ex, out, rows_out);
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;
   if (!ex->devSortKeys.empty() && !ex->devSorted) {
@@ -3527,6 +3773,11 @@ int32_t gx_open(gx_exec* ex) {
   ex->emitPos = 0;
   ex->srcPos = 0;
   ex->resultRows.clear();
+  if (ex->spillSorted) {  // re-open: replay the merged runs from the top
+    ex->runPos.assign(ex->sortRuns.size(), 0);
+    ex->spillSkip = ex->devSortOffset;
+    ex->spillRemaining = ex->devSortLimit;
+  }
   ex->opened = true;
   return GX_OK;
 }

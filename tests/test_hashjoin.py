@@ -470,3 +470,81 @@ def test_join_parity_generator(libs):
     want = run(oracle)
     assert len(got) == len(want) > 1000
     assert got == want
+
+
+# ---------------- aggregation over joined rows ----------------
+
+def _agg_over_join_plan(lib, topn_limit=None):
+    """sum(probe value dec(s2)), count(*), group by build.grp — an aggregate
+    shape the Q3-class fused pipeline does not cover (group key from the
+    BUILD side): exercises HashAgg over materialized joined rows."""
+    from tests.gxlib import GX_AGG_SUM, GX_AGG_COUNT
+    b = P.Builder(lib)
+    bsrc = b.source([GX_TYPE_I64, GX_TYPE_I64])           # key, grp
+    psrc = b.source([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2])  # key, val
+    j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)])
+    grp = b.colref(1, GX_TYPE_I64)
+    val = b.colref(3, GX_TYPE_DECIMAL, 2)
+    agg = b.hashagg(j, [grp], [(GX_AGG_SUM, val, 2), (GX_AGG_COUNT, -1, 0)])
+    root = agg
+    if topn_limit is not None:
+        root = b.topn(agg, [b.colref(0, GX_TYPE_I64)], [0], topn_limit)
+    return b, bsrc, psrc, root
+
+
+def _agg_over_join_data():
+    rng = np.random.default_rng(11)
+    n_b, n_p = 800, 6000
+    build = [(int(k), int(g)) for k, g in zip(rng.integers(0, 150, n_b),
+                                              rng.integers(0, 40, n_b))]
+    probe = [(int(k), "%d.%02d" % (v // 100, v % 100))
+             for k, v in zip(rng.integers(0, 300, n_p),
+                             rng.integers(0, 100000, n_p))]
+    return build, probe
+
+
+def run_agg_over_join(lib, topn_limit=None):
+    build, probe = _agg_over_join_data()
+    b, bsrc, psrc, root = _agg_over_join_plan(lib, topn_limit)
+    ex = b.build(root)
+    ex.bind_chunks(bsrc, [_to_chunk(lib, [GX_TYPE_I64, GX_TYPE_I64], [0, 0], build)])
+    ex.bind_chunks(psrc, [_to_chunk(lib, [GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2], probe)])
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64], [0, 2, 0])
+    ex.close()
+    ex.free()
+    b.free()
+    return rows if topn_limit is not None else sorted(rows)
+
+
+def test_oracle_agg_over_join():
+    from fractions import Fraction
+    lib = load_oracle()
+    build, probe = _agg_over_join_data()
+    got = run_agg_over_join(lib)
+    groups = {}
+    for pk, pv in probe:
+        for bk, bg in build:
+            if bk == pk:
+                s, c = groups.get(bg, (Fraction(0), 0))
+                groups[bg] = (s + Fraction(pv), c + 1)
+    want = sorted((g, s, c) for g, (s, c) in groups.items())
+    assert [(r[0], Fraction(r[1]), r[2]) for r in got] == want
+    assert len(got) == len(want) > 10
+
+
+@pytest.mark.gpu
+def test_agg_over_join_parity(libs):
+    oracle, product = libs
+    assert run_agg_over_join(product) == run_agg_over_join(oracle)
+
+
+@pytest.mark.gpu
+def test_agg_over_join_topn_parity(libs):
+    """TopN over HashAgg over HashJoin (host post-sort of the agg output)."""
+    oracle, product = libs
+    want = run_agg_over_join(oracle, topn_limit=15)
+    got = run_agg_over_join(product, topn_limit=15)
+    assert len(want) == 15
+    assert got == want

@@ -201,6 +201,10 @@ struct gx_exec {
   gxp::HashJoinDesc hj;
   gxp::HashJoinDesc* devHj = nullptr;
   int hjSrcB = -1, hjSrcP = -1;
+  // general aggregation over joined rows: runHashJoin materializes the join
+  // output into desc.table, then the fused aggregation runs over it
+  bool aggOverJoin = false;
+  bool fusedBindDone = false;
 
   ~gx_exec() {
     for (void* p : devBufs) hipFree(p);
@@ -1160,6 +1164,61 @@ static int32_t compileHashJoinTree(gx_exec* ex, int node) {
   return rc;
 }
 
+static int32_t compileFused(gx_exec* ex);
+
+// general aggregation over joined rows: HashAgg <- [Projection] <-
+// [Selection] <- HashJoin. The join materializes its output table on device
+// (runHashJoin), then the fused aggregation kernel runs over that table —
+// the generalization of the Q3-class fused pipeline to arbitrary aggregate
+// shapes (at the cost of materializing the joined columns once).
+static int32_t compileAggOverJoin(gx_exec* ex, int aggNode) {
+  PPlan& plan = ex->plan;
+  // walk down to the join, remembering the agg-side chain
+  std::vector<int> chain;  // aggNode first, bottom-most last
+  int node = aggNode;
+  while (plan.nodes[node].kind == PK_HASHAGG ||
+         plan.nodes[node].kind == PK_PROJECTION ||
+         plan.nodes[node].kind == PK_SELECTION) {
+    chain.push_back(node);
+    node = plan.nodes[node].child;
+  }
+  if (plan.nodes[node].kind != PK_HASHJOIN) {
+    ex->err = "expected a hash join under the aggregation";
+    return GX_ERR_INVALID;
+  }
+  int32_t rc = compileHashJoinTree(ex, node);
+  if (rc) return rc;
+  ex->isHashJoin = false;  // dispatched through aggOverJoin instead
+  ex->aggOverJoin = true;
+  // pseudo source carrying the join output schema
+  PNode ps;
+  ps.kind = PK_SOURCE;
+  const PNode& bN = plan.nodes[ex->hjSrcB];
+  const PNode& pN = plan.nodes[ex->hjSrcP];
+  ps.colTypes = bN.colTypes;
+  ps.colFracs = bN.colFracs;
+  ps.colTypes.insert(ps.colTypes.end(), pN.colTypes.begin(), pN.colTypes.end());
+  ps.colFracs.insert(ps.colFracs.end(), pN.colFracs.begin(), pN.colFracs.end());
+  plan.nodes.push_back(ps);
+  int cur = (int)plan.nodes.size() - 1;
+  // clone the agg-side chain bottom-up with the join replaced by the pseudo
+  // source (clones, so the original plan stays intact)
+  for (int i = (int)chain.size() - 1; i >= 0; i--) {
+    PNode copy = plan.nodes[chain[i]];
+    copy.child = cur;
+    plan.nodes.push_back(copy);
+    cur = (int)plan.nodes.size() - 1;
+  }
+  int saved = ex->root;
+  ex->root = cur;
+  rc = compileFused(ex);
+  ex->root = saved;
+  if (rc != GX_OK) return rc;
+  // the fused desc's table metadata was overwritten from the pseudo source;
+  // actual column pointers/nRows arrive when runHashJoin materializes
+  return GX_OK;
+}
+
 // compile the fused Source->[Selection]->[Projection]->HashAgg pipeline
 static int32_t compileFused(gx_exec* ex) {
   const PPlan& plan = ex->plan;
@@ -1668,6 +1727,8 @@ static int32_t materializeTable(gx_exec* ex, int srcNodeId, gxp::DevTable* tab) 
   return GX_OK;
 }
 
+static int32_t finalizeFusedBind(gx_exec* ex);
+
 static int32_t materializeDevice(gx_exec* ex) {
   if (ex->deviceReady) return GX_OK;
   if (!gpuAvailable()) {
@@ -1780,6 +1841,15 @@ static int32_t materializeDevice(gx_exec* ex) {
     ex->err = "source not bound";
     return GX_ERR_INVALID;
   }
+  return finalizeFusedBind(ex);
+}
+
+// bind-time fixups + result buffers for the fused aggregation, over whatever
+// filled desc.table (a bound/generated source, or a materialized join
+// output). Idempotent (fetch slots must not be re-assigned on re-open).
+static int32_t finalizeFusedBind(gx_exec* ex) {
+  if (ex->fusedBindDone) return GX_OK;
+  gxp::DevTable& tab = ex->desc.table;
   // finalize string group keys now that offset density is known
   for (int k = 0; k < ex->desc.gkey.nCols; k++) {
     if (ex->desc.gkey.kind[k] != 1) {
@@ -1874,6 +1944,7 @@ static int32_t materializeDevice(gx_exec* ex) {
       fprintf(stderr, "[gx] useGlds=%d tileBytes=%d nFetch=%d\n", d.useGlds,
               d.tileBytes, d.nFetch);
   }
+  ex->fusedBindDone = true;
   ex->deviceReady = true;
   return GX_OK;
 }
@@ -3503,7 +3574,16 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     ex->isFinalHost = true;
     ex->sourceNode = rn.child;
   } else if (rn.kind == PK_HASHAGG) {
-    int32_t rc = compileFused(ex);
+    // bottom of the subtree: a Source (fused pipeline) or a HashJoin
+    // (aggregation over materialized joined rows)
+    int node = root;
+    while (ex->plan.nodes[node].kind == PK_HASHAGG ||
+           ex->plan.nodes[node].kind == PK_PROJECTION ||
+           ex->plan.nodes[node].kind == PK_SELECTION)
+      node = ex->plan.nodes[node].child;
+    int32_t rc = ex->plan.nodes[node].kind == PK_HASHJOIN
+                     ? compileAggOverJoin(ex, root)
+                     : compileFused(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     (void)rc;
   } else if (rn.kind == PK_TOPN &&
@@ -3536,6 +3616,7 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
       if (rc != GX_OK) {
         // not a fusable aggregation subtree (e.g. Q3's agg-over-join):
         // reset and try the join-aggregate pipeline
+        std::vector<std::pair<int, bool>> savedKeys = ex->postSortKeys;
         ex->err.clear();
         ex->isFused = false;
         ex->postSort = false;
@@ -3547,8 +3628,27 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
         ex->exprUse.clear();
         ex->vmFreeRegs.clear();
         rc = compileJoinAgg(ex);
-        if (rc != GX_OK && ex->err.empty())
-          ex->err = "plan compilation failed";
+        if (rc != GX_OK) {
+          // last resort: general aggregation over materialized joined rows
+          // + host post-sort (keys already validated as agg output columns)
+          ex->err.clear();
+          ex->desc = gxp::FusedQueryDesc{};
+          ex->projRegs.clear();
+          ex->vmNextReg = 0;
+          ex->exprRegCache.clear();
+          ex->exprUse.clear();
+          ex->vmFreeRegs.clear();
+          ex->postSort = true;
+          ex->postSortKeys = savedKeys;
+          ex->postLimit = rn.limit;
+          ex->postOffset = rn.offset;
+          rc = compileAggOverJoin(ex, aggRoot);
+          if (rc != GX_OK) {
+            ex->postSort = false;
+            ex->postSortKeys.clear();
+            if (ex->err.empty()) ex->err = "plan compilation failed";
+          }
+        }
       }
     } else {
       ex->err.clear();
@@ -3804,9 +3904,19 @@ int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
     return emitTableChunk(ex, out, rows_out);
   }
   if (!ex->ranQuery) {
-    int32_t rc = ex->isFinalHost ? runFinalHost(ex)
-                 : ex->isJoinAgg ? runJoinAgg(ex)
-                                 : runFused(ex);
+    int32_t rc;
+    if (ex->aggOverJoin) {
+      // materialize the joined rows, finish the fused bind over that table,
+      // then aggregate (materializeDevice no-ops: runHashJoin set
+      // deviceReady)
+      rc = runHashJoin(ex);
+      if (rc == GX_OK) rc = finalizeFusedBind(ex);
+      if (rc == GX_OK) rc = runFused(ex);
+    } else {
+      rc = ex->isFinalHost ? runFinalHost(ex)
+           : ex->isJoinAgg ? runJoinAgg(ex)
+                           : runFused(ex);
+    }
     if (rc) {
       *rows_out = 0;
       return rc;

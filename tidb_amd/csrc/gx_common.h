@@ -358,6 +358,17 @@ int gxGatherNulls(const uint8_t* inBitmap, const uint32_t* idx, uint8_t* out,
                   int64_t n, void* stream);
 // identity offsets ramp for gathered dense-char columns
 int gxIotaOffsets(int64_t* p, int64_t n, void* stream);
+// varlen gather, pass 1: per-output-row byte lengths from the source offsets
+int gxGatherVarlenLens(const int64_t* inOffsets, const uint32_t* idx,
+                       int64_t* lens, int64_t n, void* stream);
+// exclusive prefix sum of lens[0..n) -> outOffsets[0..n] (hipcub; outOffsets
+// gets n+1 entries, outOffsets[n] = total bytes)
+int gxExclusiveSumI64(const int64_t* lens, int64_t* outOffsets, int64_t n,
+                      void* tmp, size_t* tmpBytes, void* stream);
+// varlen gather, pass 2: copy each row's bytes through the match index
+int gxGatherVarlenBytes(const uint8_t* inData, const int64_t* inOffsets,
+                        const uint32_t* idx, const int64_t* outOffsets,
+                        uint8_t* outData, int64_t n, void* stream);
 
 // top-N selection scratch
 struct TopNOut {

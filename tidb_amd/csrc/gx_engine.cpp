@@ -2601,16 +2601,6 @@ static int32_t runHashJoin(gx_exec* ex) {
     ex->err = "join sides > 2^32 rows unsupported this round";
     return GX_ERR_INVALID;
   }
-  // gathered output supports fixed-width and dense char(1) columns only
-  for (int c = 0; c < ex->desc.table.nCols; c++) {
-    int nbc = hj.build.nCols;
-    const gxp::DevCol& src =
-        c < nbc ? hj.build.cols[c] : hj.probe.cols[c - nbc];
-    if (src.type == GX_TYPE_STRING && !src.denseOffsets) {
-      ex->err = "general varlen join output column unsupported this round";
-      return GX_ERR_INVALID;
-    }
-  }
   hipEvent_t ev0, ev1, evB, evC, evF;
   HIP_OK(ex, hipEventCreate(&ev0));
   HIP_OK(ex, hipEventCreate(&ev1));
@@ -2699,23 +2689,59 @@ static int32_t runHashJoin(gx_exec* ex) {
           c < nbc ? hj.build.cols[c] : hj.probe.cols[c - nbc];
       const uint32_t* idx = c < nbc ? gatherB : gatherP;
       gxp::DevCol& dst = ex->desc.table.cols[c];
-      int es = src.type == GX_TYPE_DECIMAL ? 40
-               : (src.type == GX_TYPE_STRING ? 1 : 8);
-      dst.data = devAlloc(ex, (size_t)total * es + 16);
-      if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-      if (gxp::gxSortGatherCol(src.data, dst.data, idx, (int64_t)total, es,
-                               ex->stream) != 0) {
-        ex->err = "join gather launch failed";
-        return GX_ERR_INTERNAL;
-      }
-      if (src.type == GX_TYPE_STRING) {
+      if (src.type == GX_TYPE_STRING && !src.denseOffsets) {
+        // general varlen gather: lengths -> exclusive scan -> byte copy
+        int64_t* lens = (int64_t*)devAlloc(ex, total * 8);
         dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
-        if (!dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-        if (gxp::gxIotaOffsets(dst.offsets, (int64_t)total + 1, ex->stream) != 0) {
-          ex->err = "join offsets launch failed";
+        if (!lens || !dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        if (gxp::gxGatherVarlenLens(src.offsets, idx, lens, (int64_t)total,
+                                    ex->stream) != 0) {
+          ex->err = "varlen lens launch failed";
           return GX_ERR_INTERNAL;
         }
-        dst.denseOffsets = 1;
+        size_t tmpBytes = 0;
+        gxp::gxExclusiveSumI64(lens, dst.offsets, (int64_t)total, nullptr,
+                               &tmpBytes, ex->stream);
+        void* tmp = devAlloc(ex, std::max<size_t>(tmpBytes, 1));
+        if (!tmp) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        HIP_OK(ex, hipMemsetAsync(dst.offsets, 0, 8, ex->stream));
+        if (gxp::gxExclusiveSumI64(lens, dst.offsets, (int64_t)total, tmp,
+                                   &tmpBytes, ex->stream) != 0) {
+          ex->err = "varlen offsets scan failed";
+          return GX_ERR_INTERNAL;
+        }
+        int64_t totalBytes = 0;
+        HIP_OK(ex, hipStreamSynchronize(ex->stream));
+        HIP_OK(ex, hipMemcpy(&totalBytes, dst.offsets + total, 8,
+                             hipMemcpyDeviceToHost));
+        dst.data = devAlloc(ex, std::max<int64_t>(totalBytes, 1));
+        if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        if (gxp::gxGatherVarlenBytes((const uint8_t*)src.data, src.offsets,
+                                     idx, dst.offsets, (uint8_t*)dst.data,
+                                     (int64_t)total, ex->stream) != 0) {
+          ex->err = "varlen bytes launch failed";
+          return GX_ERR_INTERNAL;
+        }
+        dst.denseOffsets = 0;
+      } else {
+        int es = src.type == GX_TYPE_DECIMAL ? 40
+                 : (src.type == GX_TYPE_STRING ? 1 : 8);
+        dst.data = devAlloc(ex, (size_t)total * es + 16);
+        if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        if (gxp::gxSortGatherCol(src.data, dst.data, idx, (int64_t)total, es,
+                                 ex->stream) != 0) {
+          ex->err = "join gather launch failed";
+          return GX_ERR_INTERNAL;
+        }
+        if (src.type == GX_TYPE_STRING) {
+          dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
+          if (!dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+          if (gxp::gxIotaOffsets(dst.offsets, (int64_t)total + 1, ex->stream) != 0) {
+            ex->err = "join offsets launch failed";
+            return GX_ERR_INTERNAL;
+          }
+          dst.denseOffsets = 1;
+        }
       }
       if (src.hasNulls && src.nullBitmap) {
         dst.nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8);

@@ -1939,6 +1939,35 @@ __global__ void iotaI64Kernel(int64_t* __restrict__ p, int64_t n) {
     p[i] = i;
 }
 
+// varlen gather pass 1: per-output-row byte lengths
+__global__ void hjVarlenLensKernel(const int64_t* __restrict__ inOffsets,
+                                   const uint32_t* __restrict__ idx,
+                                   int64_t* __restrict__ lens, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t src = idx[i];
+    lens[i] = inOffsets[src + 1] - inOffsets[src];
+  }
+}
+
+// varlen gather pass 2: copy bytes (one thread per output row; rows are
+// short strings, so per-thread byte loops stay coalesced enough across the
+// wave)
+__global__ void hjVarlenBytesKernel(const uint8_t* __restrict__ inData,
+                                    const int64_t* __restrict__ inOffsets,
+                                    const uint32_t* __restrict__ idx,
+                                    const int64_t* __restrict__ outOffsets,
+                                    uint8_t* __restrict__ outData, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t src = idx[i];
+    int64_t s = inOffsets[src];
+    int64_t len = inOffsets[src + 1] - s;
+    int64_t o = outOffsets[i];
+    for (int64_t j = 0; j < len; j++) outData[o + j] = inData[s + j];
+  }
+}
+
 __global__ void sortGatherKernel(const uint8_t* __restrict__ in,
                                  uint8_t* __restrict__ out,
                                  const uint32_t* __restrict__ idx, int64_t n,
@@ -2055,6 +2084,30 @@ int gxGatherNulls(const uint8_t* inBitmap, const uint32_t* idx, uint8_t* out,
 int gxIotaOffsets(int64_t* p, int64_t n, void* stream) {
   hipLaunchKernelGGL(iotaI64Kernel, dim3(gridFor(n)), dim3(256), 0,
                      (hipStream_t)stream, p, n);
+  return (int)hipGetLastError();
+}
+
+int gxGatherVarlenLens(const int64_t* inOffsets, const uint32_t* idx,
+                       int64_t* lens, int64_t n, void* stream) {
+  hipLaunchKernelGGL(hjVarlenLensKernel, dim3(gridFor(n)), dim3(256), 0,
+                     (hipStream_t)stream, inOffsets, idx, lens, n);
+  return (int)hipGetLastError();
+}
+
+int gxExclusiveSumI64(const int64_t* lens, int64_t* outOffsets, int64_t n,
+                      void* tmp, size_t* tmpBytes, void* stream) {
+  // inclusive sum into outOffsets+1; caller zeroes outOffsets[0]
+  return (int)hipcub::DeviceScan::InclusiveSum(tmp, *tmpBytes, lens,
+                                               outOffsets + 1, (int)n,
+                                               (hipStream_t)stream);
+}
+
+int gxGatherVarlenBytes(const uint8_t* inData, const int64_t* inOffsets,
+                        const uint32_t* idx, const int64_t* outOffsets,
+                        uint8_t* outData, int64_t n, void* stream) {
+  hipLaunchKernelGGL(hjVarlenBytesKernel, dim3(gridFor(n)), dim3(256), 0,
+                     (hipStream_t)stream, inData, inOffsets, idx, outOffsets,
+                     outData, n);
   return (int)hipGetLastError();
 }
 

@@ -117,3 +117,17 @@ def test_spill_topn_limit_offset_parity(monkeypatch):
     monkeypatch.delenv("GX_SORT_RUN_ROWS")
     assert len(a) == len(b) == 200
     assert [(r[7], r[0]) for r in a] == [(r[7], r[0]) for r in b]
+
+
+@pytest.mark.gpu
+def test_spill_sort_decimal_key_parity(monkeypatch):
+    """Spilled runs with a decimal sort key (composed int64-unit keys per
+    run; merge compares the same composed keys)."""
+    from tests.gxlib import load_product
+    keys = [(P.L_EXTPRICE, GX_TYPE_DECIMAL, 2), (P.L_ORDERKEY, GX_TYPE_I64, 0)]
+    a = run_sort(load_oracle(), keys, [1, 0], n_rows=40000)
+    monkeypatch.setenv("GX_SORT_RUN_ROWS", "7000")
+    b = run_sort(load_product(), keys, [1, 0], n_rows=40000)
+    monkeypatch.delenv("GX_SORT_RUN_ROWS")
+    assert [(r[2], r[0]) for r in a] == [(r[2], r[0]) for r in b]
+    assert sorted(map(tuple, a)) == sorted(map(tuple, b))

@@ -581,3 +581,44 @@ def test_join_parity_varlen_output(libs):
     want = run(oracle)
     assert len(got) == len(want) > 1000
     assert got == want
+
+
+def _full_agg_over_join(lib):
+    """avg/min/max/firstrow + sum/count over joined rows, group by a build
+    column — the full aggfuncs family (func_avg.go, func_max_min.go,
+    func_sum.go) over a join child."""
+    from tests.gxlib import (GX_AGG_AVG, GX_AGG_COUNT, GX_AGG_FIRSTROW,
+                            GX_AGG_MAX, GX_AGG_MIN, GX_AGG_SUM)
+    build, probe = _agg_over_join_data()
+    b = P.Builder(lib)
+    bsrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    psrc = b.source([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2])
+    j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)])
+    grp = b.colref(1, GX_TYPE_I64)
+    val = b.colref(3, GX_TYPE_DECIMAL, 2)
+    key = b.colref(2, GX_TYPE_I64)
+    agg = b.hashagg(j, [grp], [
+        (GX_AGG_SUM, val, 2), (GX_AGG_AVG, val, 6), (GX_AGG_MIN, key, 0),
+        (GX_AGG_MAX, key, 0), (GX_AGG_COUNT, -1, 0), (GX_AGG_FIRSTROW, grp, 0),
+    ])
+    ex = b.build(agg)
+    ex.bind_chunks(bsrc, [_to_chunk(lib, [GX_TYPE_I64, GX_TYPE_I64], [0, 0], build)])
+    ex.bind_chunks(psrc, [_to_chunk(lib, [GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2], probe)])
+    ex.open()
+    out_t = [GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_DECIMAL, GX_TYPE_I64,
+             GX_TYPE_I64, GX_TYPE_I64, GX_TYPE_I64]
+    rows = ex.pull_all(out_t, [0, 2, 6, 0, 0, 0, 0])
+    ex.close()
+    ex.free()
+    b.free()
+    return sorted(rows)
+
+
+@pytest.mark.gpu
+def test_full_agg_family_over_join_parity(libs):
+    oracle, product = libs
+    got = _full_agg_over_join(product)
+    want = _full_agg_over_join(oracle)
+    assert len(got) == len(want) > 10
+    assert got == want

@@ -1486,8 +1486,9 @@ static int32_t compileFused(gx_exec* ex) {
     int kind = ad.func == GX_AGG_MAX ? 1 : (ad.func == GX_AGG_MIN ? 2 : 0);
     if (kind != 0) {
       int argE2 = agg->aggArgs[a];
-      if (argE2 < 0 || plan.exprs[argE2].retType != GX_TYPE_DECIMAL) {
-        ex->err = "device min/max supports decimal arguments this round";
+      if (argE2 < 0 || (plan.exprs[argE2].retType != GX_TYPE_DECIMAL &&
+                        plan.exprs[argE2].retType != GX_TYPE_I64)) {
+        ex->err = "device min/max supports decimal/int64 arguments this round";
         return GX_ERR_INVALID;
       }
     }
@@ -2198,11 +2199,16 @@ static int32_t runFused(gx_exec* ex) {
         uint64_t raw = s->accLo[phys];
         if (ex->desc.accKind[phys] == 2) raw = ~raw;
         int64_t ext = (int64_t)(raw ^ 0x8000000000000000ULL);
-        int srcType = GX_TYPE_DECIMAL;
-        // arg type: decimal scale > 0 or i64 (scale 0 decimal still decimal)
+        // output type follows the argument's type (func_max_min.go keeps the
+        // arg type)
+        int argE = agg.aggArgs[a];
+        int srcType = argE >= 0 ? ex->plan.exprs[argE].retType
+                                : GX_TYPE_DECIMAL;
+        v.type = srcType;
         if (cnt == 0) {
-          v.type = srcType;
           v.isNull = true;
+        } else if (srcType == GX_TYPE_I64) {
+          v.i64 = ext;
         } else {
           v.type = GX_TYPE_DECIMAL;
           v.dec = decFromUnits((__int128)ext, ad.scale);

@@ -594,7 +594,7 @@ class HashAggExec : public Exec {
 
   void storeValue(AggState& s, const Column& col, int row, int vt) {
     switch (vt) {
-      case GX_TYPE_I64: s.aux = col.getI64(row); s.i64 = col.getI64(row); break;
+      case GX_TYPE_I64: s.i64 = col.getI64(row); break;
       case GX_TYPE_F64: s.f64 = col.getF64(row); break;
       case GX_TYPE_TIME: s.u64 = col.getU64(row); break;
       case GX_TYPE_DECIMAL: s.dec = *col.getDecimal(row); break;
@@ -747,7 +747,7 @@ class HashAggExec : public Exec {
 
   void appendValue(const AggState& s, int vt, Column& out) {
     switch (vt) {
-      case GX_TYPE_I64: out.appendI64(s.aux); break;
+      case GX_TYPE_I64: out.appendI64(s.i64); break;
       case GX_TYPE_F64: out.appendF64(s.f64); break;
       case GX_TYPE_TIME: out.appendU64(s.u64); break;
       case GX_TYPE_DECIMAL: out.appendDecimal(s.dec); break;

@@ -1796,6 +1796,9 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
     uint32_t cnt = 0;
     uint32_t head = 0;
     uint64_t key = 0, key1 = 0;
+    // matched rows are cached in registers during the counting walk so the
+    // common case (a handful of duplicates) never re-walks the chain
+    uint32_t hit[4];
     if (active) {
       bool pass = d.nPredP == 0 ||
                   evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen,
@@ -1804,7 +1807,10 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
         head = gptr<uint32_t>(d.heads)[(uint32_t)(hjHash(d, key, key1) & mask)];
         for (uint32_t cur = head; cur != 0;) {
           uint32_t brow = cur - 1;
-          if (hjBuildKeyEq(d, brow, key, key1)) cnt++;
+          if (hjBuildKeyEq(d, brow, key, key1)) {
+            if (FILL && cnt < 4) hit[cnt] = brow;
+            cnt++;
+          }
           cur = gptr<uint32_t>(d.next)[brow];
         }
       }
@@ -1824,7 +1830,12 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
         base = atomicAdd((unsigned long long*)&d.counters[1],
                          (unsigned long long)waveTotal);
       base = __shfl(base, 63, 64) + (pre - cnt);
-      if (cnt) {
+      if (cnt > 0 && cnt <= 4) {
+        for (uint32_t k = 0; k < cnt; k++) {
+          d.outBuild[base + k] = hit[k];
+          d.outProbe[base + k] = (uint32_t)row;
+        }
+      } else if (cnt > 4) {
         for (uint32_t cur = head; cur != 0;) {
           uint32_t brow = cur - 1;
           if (hjBuildKeyEq(d, brow, key, key1)) {

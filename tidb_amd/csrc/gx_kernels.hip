@@ -1816,20 +1816,31 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
       }
     }
     if (FILL) {
-      // exclusive wave prefix sum of cnt -> per-lane slice of one wave-wide
-      // reservation on the shared cursor
-      uint64_t pre = cnt;
-      for (int off = 1; off < 64; off <<= 1) {
-        uint64_t t = __shfl_up(pre, off, 64);
-        if (lane >= off) pre += t;
-      }
-      uint64_t waveTotal = __shfl(pre, 63, 64);
-      if (waveTotal == 0) continue;
       uint64_t base = 0;
-      if (lane == 63)
-        base = atomicAdd((unsigned long long*)&d.counters[1],
-                         (unsigned long long)waveTotal);
-      base = __shfl(base, 63, 64) + (pre - cnt);
+      if (__ballot(cnt > 1) == 0) {
+        // common case (every lane matched <=1 row): one ballot + popcount
+        // replaces the 6-step shuffle prefix chain
+        uint64_t m = __ballot(cnt == 1);
+        if (m == 0) continue;
+        if (lane == 63)
+          base = atomicAdd((unsigned long long*)&d.counters[1],
+                           (unsigned long long)__popcll(m));
+        base = __shfl(base, 63, 64) + __popcll(m & ((1ULL << lane) - 1));
+      } else {
+        // exclusive wave prefix sum of cnt -> per-lane slice of one
+        // wave-wide reservation on the shared cursor
+        uint64_t pre = cnt;
+        for (int off = 1; off < 64; off <<= 1) {
+          uint64_t t = __shfl_up(pre, off, 64);
+          if (lane >= off) pre += t;
+        }
+        uint64_t waveTotal = __shfl(pre, 63, 64);
+        if (waveTotal == 0) continue;
+        if (lane == 63)
+          base = atomicAdd((unsigned long long*)&d.counters[1],
+                           (unsigned long long)waveTotal);
+        base = __shfl(base, 63, 64) + (pre - cnt);
+      }
       if (cnt > 0 && cnt <= 4) {
         for (uint32_t k = 0; k < cnt; k++) {
           d.outBuild[base + k] = hit[k];

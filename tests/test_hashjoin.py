@@ -622,3 +622,44 @@ def test_full_agg_family_over_join_parity(libs):
     want = _full_agg_over_join(oracle)
     assert len(got) == len(want) > 10
     assert got == want
+
+
+def test_oracle_join_varlen_output():
+    """CPU pin for the varlen-output join: oracle hashjoin of generator
+    customer ⋈ orders equals a pure-Python join of the pulled tables."""
+    from tests.gxlib import GX_TPCH_CUSTOMER, GX_TPCH_ORDERS
+    lib = load_oracle()
+
+    def pull(types, table, n, caps):
+        b = P.Builder(lib)
+        src = b.source(types)
+        ex = b.build(src)
+        ex.bind_tpch(src, table, n)
+        ex.open()
+        rows = ex.pull_all(types, [0] * len(types), data_caps=caps)
+        ex.close()
+        ex.free()
+        b.free()
+        return rows
+
+    cust = pull(P.CUSTOMER_TYPES, GX_TPCH_CUSTOMER, 300, [None, 65536])
+    orders = pull(P.ORDERS_TYPES, GX_TPCH_ORDERS, 3000, [None] * 4)
+    want = sorted(tuple(c) + tuple(o) for o in orders for c in cust
+                  if c[0] == o[1])
+
+    b = P.Builder(lib)
+    csrc = b.source(P.CUSTOMER_TYPES)
+    osrc = b.source(P.ORDERS_TYPES)
+    j = b.hashjoin(csrc, osrc, [b.colref(P.C_CUSTKEY, GX_TYPE_I64)],
+                   [b.colref(P.O_CUSTKEY, GX_TYPE_I64)])
+    ex = b.build(j)
+    ex.bind_tpch(csrc, GX_TPCH_CUSTOMER, 300)
+    ex.bind_tpch(osrc, GX_TPCH_ORDERS, 3000)
+    ex.open()
+    got = sorted(ex.pull_all(P.CUSTOMER_TYPES + P.ORDERS_TYPES, [0] * 6,
+                             data_caps=[None, 65536] + [None] * 4))
+    ex.close()
+    ex.free()
+    b.free()
+    assert got == want
+    assert len(got) > 500

@@ -663,3 +663,114 @@ def test_oracle_join_varlen_output():
     b.free()
     assert got == want
     assert len(got) > 500
+
+
+# ---------------- nested joins (stage compilation) ----------------
+
+def _nested_data():
+    rng = np.random.default_rng(21)
+    t0 = [(int(k), int(v)) for k, v in zip(rng.integers(0, 40, 200),
+                                           np.arange(200))]        # inner build
+    t1 = [(int(k), int(k2), int(v)) for k, k2, v in zip(
+        rng.integers(0, 80, 1000), rng.integers(0, 60, 1000),
+        np.arange(1000))]                                          # inner probe
+    t2 = [(int(k2), int(v)) for k2, v in zip(rng.integers(0, 120, 4000),
+                                             np.arange(4000))]     # outer probe
+    return t0, t1, t2
+
+
+def _run_nested(lib):
+    """(t0 ⋈ t1 on k) ⋈ t2 on k2 — duplicate keys at every level; the outer
+    build side is the inner join's materialized output."""
+    t0, t1, t2 = _nested_data()
+    b = P.Builder(lib)
+    s0 = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    s1 = b.source([GX_TYPE_I64, GX_TYPE_I64, GX_TYPE_I64])
+    s2 = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    j1 = b.hashjoin(s0, s1, [b.colref(0, GX_TYPE_I64)],
+                    [b.colref(0, GX_TYPE_I64)])
+    # j1 out: t0.k, t0.v, t1.k, t1.k2, t1.v  -> join t2 on k2 (col 3)
+    j2 = b.hashjoin(j1, s2, [b.colref(3, GX_TYPE_I64)],
+                    [b.colref(0, GX_TYPE_I64)])
+    ex = b.build(j2)
+    for src, rows, tps in ((s0, t0, 2), (s1, t1, 3), (s2, t2, 2)):
+        ex.bind_chunks(src, [_to_chunk(lib, [GX_TYPE_I64] * tps, [0] * tps,
+                                       rows)])
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_I64] * 7)
+    ex.close()
+    ex.free()
+    b.free()
+    return sorted(rows)
+
+
+def test_oracle_nested_join():
+    t0, t1, t2 = _nested_data()
+    want = sorted(a + bb + c
+                  for bb in t1 for a in t0 if a[0] == bb[0]
+                  for c in t2 if c[0] == bb[1])
+    got = _run_nested(load_oracle())
+    assert got == want
+    assert len(got) > 3000
+
+
+@pytest.mark.gpu
+def test_nested_join_parity(libs):
+    oracle, product = libs
+    got = _run_nested(product)
+    want = _run_nested(oracle)
+    assert len(got) == len(want) > 3000
+    assert got == want
+
+
+@pytest.mark.gpu
+def test_q3_class_agg_over_nested_join_parity(libs):
+    """customer⋈orders⋈lineitem with a CUSTOMER-side group key — a Q3-class
+    query the fused join-aggregate pipeline rejects: runs as nested join
+    stages + fused aggregation over the materialized output."""
+    from tests.gxlib import (GX_AGG_COUNT, GX_AGG_SUM, GX_F_EQ, GX_F_GT,
+                             GX_F_MINUS, GX_F_MUL, GX_TPCH_CUSTOMER,
+                             GX_TPCH_LINEITEM, GX_TPCH_ORDERS)
+
+    def run(lib):
+        b = P.Builder(lib)
+        cust = b.source(P.CUSTOMER_TYPES)
+        seg = b.colref(P.C_MKTSEGMENT, GX_TYPE_STRING)
+        sel_c = b.selection(cust, [b.call(GX_F_EQ, GX_TYPE_I64, 0, seg,
+                                          lib.gx_pb_const_str(b.pb, b"BUILDING", 8))])
+        orders = b.source(P.ORDERS_TYPES)
+        li = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+        sdate = b.colref(P.L_SHIPDATE, GX_TYPE_TIME)
+        sel_l = b.selection(li, [b.call(GX_F_GT, GX_TYPE_I64, 0, sdate,
+                                        b.const_time(lib.gx_time_from_date(1995, 3, 15)))])
+        j1 = b.hashjoin(sel_c, orders, [b.colref(P.C_CUSTKEY, GX_TYPE_I64)],
+                        [b.colref(P.O_CUSTKEY, GX_TYPE_I64)])
+        # j1 out: c_custkey, c_mktsegment, o_orderkey, o_custkey, o_odate, o_prio
+        j2 = b.hashjoin(j1, sel_l, [b.colref(2, GX_TYPE_I64)],
+                        [b.colref(P.L_ORDERKEY, GX_TYPE_I64)])
+        ckey = b.colref(0, GX_TYPE_I64)
+        price = b.colref(6 + P.L_EXTPRICE, GX_TYPE_DECIMAL, 2)
+        disc = b.colref(6 + P.L_DISCOUNT, GX_TYPE_DECIMAL, 2)
+        one = P._const_dec_one(lib, b)
+        rev = b.call(GX_F_MUL, GX_TYPE_DECIMAL, 4, price,
+                     b.call(GX_F_MINUS, GX_TYPE_DECIMAL, 2, one, disc))
+        proj = b.projection(j2, [ckey, rev])
+        agg = b.hashagg(proj, [b.colref(0, GX_TYPE_I64)],
+                        [(GX_AGG_SUM, b.colref(1, GX_TYPE_DECIMAL, 4), 4),
+                         (GX_AGG_COUNT, -1, 0)])
+        ex = b.build(agg)
+        ex.bind_tpch(cust, GX_TPCH_CUSTOMER, 1000)
+        ex.bind_tpch(orders, GX_TPCH_ORDERS, 10000)
+        ex.bind_tpch(li, GX_TPCH_LINEITEM, 40000)
+        ex.open()
+        rows = ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64],
+                           [0, 4, 0])
+        ex.close()
+        ex.free()
+        b.free()
+        return sorted(rows)
+
+    got = run(product)
+    want = run(oracle)
+    assert len(got) == len(want) > 50
+    assert got == want

@@ -196,11 +196,24 @@ struct gx_exec {
   std::vector<std::pair<int, bool>> jaSortKeys;
   int64_t jaLimit = 0, jaOffset = 0;
 
-  // standalone hash join (inner, duplicate build keys) state
+  // standalone hash join (inner, duplicate build keys) state.
+  // A join subtree compiles into a topologically ordered list of STAGES
+  // (nested joins materialize bottom-up: a stage's side is either a bound
+  // source or a previous stage's materialized output table).
+  struct JoinStage {
+    gxp::HashJoinDesc hj;
+    int srcB = -1, srcP = -1;            // source plan-node ids, or -1
+    int buildStage = -1, probeStage = -1;  // feeding stage index, or -1
+    std::vector<int> types, fracs;       // output schema (build ++ probe)
+    gxp::DevTable out;                   // materialized output (run time)
+    bool inputsReady = false;            // sources materialized once
+    gxp::DevTable buildTab, probeTab;    // resident side tables
+  };
   bool isHashJoin = false;
-  gxp::HashJoinDesc hj;
+  std::vector<JoinStage> joinStages;     // root stage last
   gxp::HashJoinDesc* devHj = nullptr;
-  int hjSrcB = -1, hjSrcP = -1;
+  int hjSrcB = -1, hjSrcP = -1;          // ROOT stage's source ids (used by
+                                         // compileAggOverJoin's schema)
   // general aggregation over joined rows: runHashJoin materializes the join
   // output into desc.table, then the fused aggregation runs over it
   bool aggOverJoin = false;
@@ -1002,105 +1015,83 @@ static int32_t compileJoinAgg(gx_exec* ex) {
   return GX_OK;
 }
 
-// compile a STANDALONE inner hash join (root = HashJoin over [Selection ->]
-// Source children): the general HashJoinV2 operator (join/hash_join_v2.go)
-// with duplicate build keys via chained table — output is the joined rows
-// themselves (build cols ++ probe cols), not an aggregate.
-static int32_t compileHashJoin(gx_exec* ex, int joinNode) {
-  const PPlan& plan = ex->plan;
-  const PNode& jn = plan.nodes[joinNode];
-  if (jn.joinType != 0) {
-    ex->err = "only inner joins on device this round";
-    return GX_ERR_INVALID;
-  }
-  if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
-      jn.buildKeys.size() > 2) {
-    ex->err = "device hash join supports 1-2 key columns this round";
-    return GX_ERR_INVALID;
-  }
-  const PNode *selB, *selP;
-  int srcB = unwrapSource(ex, jn.child, &selB);
-  int srcP = unwrapSource(ex, jn.child2, &selP);
-  if (srcB < 0 || srcP < 0) {
-    ex->err = "join children must be [Selection ->] Source";
-    return GX_ERR_INVALID;
-  }
-  const PNode& bN = plan.nodes[srcB];
-  const PNode& pN = plan.nodes[srcP];
-  int nb = (int)bN.colTypes.size();
-  int np = (int)pN.colTypes.size();
-  if (nb + np > gxp::kMaxCols) {
-    ex->err = "too many join output columns";
-    return GX_ERR_INVALID;
-  }
-  gxp::HashJoinDesc& hj = ex->hj;
-  hj.nKeys = (int32_t)jn.buildKeys.size();
-  for (int k = 0; k < hj.nKeys; k++) {
-    const PExpr& bk = plan.exprs[jn.buildKeys[k]];
-    const PExpr& pk = plan.exprs[jn.probeKeys[k]];
-    if (bk.kind != EK_COLREF || pk.kind != EK_COLREF || bk.colIdx < 0 ||
-        bk.colIdx >= nb || pk.colIdx < 0 || pk.colIdx >= np) {
-      ex->err = "join keys must be child columns";
+// ---------------- standalone hash join compilation (stage model) ----------
+// A join subtree (nested joins allowed) compiles bottom-up into
+// gx_exec::joinStages: each stage is one HashJoinV2-equivalent
+// (join/hash_join_v2.go) whose sides are bound sources ([Selection ->]
+// Source) or previous stages' materialized output tables.
+
+static int32_t compileJoinStage(gx_exec* ex, int joinNode,
+                                const PNode* postSel, int* stageOut);
+
+struct JoinSideRef {
+  int srcNode = -1;            // bound source (-1 if stage-fed)
+  int stage = -1;              // feeding stage (-1 if source)
+  const PNode* sel = nullptr;  // Selection over a SOURCE side (table pred)
+  std::vector<int> types, fracs;
+};
+
+static int32_t resolveJoinSide(gx_exec* ex, int node, JoinSideRef* side) {
+  node = skipFullSort(ex, node);
+  const PNode* n = &ex->plan.nodes[node];
+  if (n->kind == PK_SELECTION) {
+    const PNode* sel = n;
+    int child = skipFullSort(ex, n->child);
+    const PNode* cn = &ex->plan.nodes[child];
+    if (cn->kind == PK_HASHJOIN) {
+      // the selection is the inner join\'s post filter (other conditions)
+      int st = -1;
+      int32_t rc = compileJoinStage(ex, child, sel, &st);
+      if (rc) return rc;
+      side->stage = st;
+      side->types = ex->joinStages[st].types;
+      side->fracs = ex->joinStages[st].fracs;
+      return GX_OK;
+    }
+    if (cn->kind != PK_SOURCE) {
+      ex->err = "join children must be [Selection ->] Source or a join";
       return GX_ERR_INVALID;
     }
-    int bt = bN.colTypes[bk.colIdx];
-    int pt = pN.colTypes[pk.colIdx];
-    // fixed 8-byte keys: int64 or packed CoreTime (SerializeKeys writes the
-    // raw 8 bytes for both — equality is bitwise)
-    if (bt != pt || (bt != GX_TYPE_I64 && bt != GX_TYPE_TIME)) {
-      ex->err = "device join keys must be int64/time this round";
-      return GX_ERR_INVALID;
-    }
-    hj.bKeyCol[k] = bk.colIdx;
-    hj.pKeyCol[k] = pk.colIdx;
+    side->srcNode = child;
+    side->sel = sel;
+    side->types = cn->colTypes;
+    side->fracs = cn->colFracs;
+    return GX_OK;
   }
-  auto doPred = [&](const PNode* sel, const PNode& srcNode, gxp::PredDesc* pd,
-                    int32_t* n, uint8_t* sc, int32_t* scLen) -> bool {
-    *n = 0;
-    if (!sel) return true;
-    if (sel->exprs.size() != 1) {
-      ex->err = "device join path supports one filter conjunct per table";
-      return false;
-    }
-    if (!compileTablePred(ex, srcNode, sel->exprs[0], pd, sc, scLen))
-      return false;
-    *n = 1;
-    return true;
-  };
-  if (!doPred(selB, bN, &hj.predB, &hj.nPredB, hj.strConstB, &hj.strConstBLen))
-    return GX_ERR_INVALID;
-  if (!doPred(selP, pN, &hj.predP, &hj.nPredP, hj.strConstP, &hj.strConstPLen))
-    return GX_ERR_INVALID;
-  ex->hjSrcB = srcB;
-  ex->hjSrcP = srcP;
-  // output schema: build cols ++ probe cols (inner_join_probe.go:27-86)
-  ex->desc.table.nCols = nb + np;
-  for (int c = 0; c < nb; c++)
-    setDevColMeta(&ex->desc.table.cols[c], bN.colTypes[c], bN.colFracs[c]);
-  for (int c = 0; c < np; c++)
-    setDevColMeta(&ex->desc.table.cols[nb + c], pN.colTypes[c], pN.colFracs[c]);
-  ex->isHashJoin = true;
-  return GX_OK;
+  if (n->kind == PK_HASHJOIN) {
+    int st = -1;
+    int32_t rc = compileJoinStage(ex, node, nullptr, &st);
+    if (rc) return rc;
+    side->stage = st;
+    side->types = ex->joinStages[st].types;
+    side->fracs = ex->joinStages[st].fracs;
+    return GX_OK;
+  }
+  if (n->kind == PK_SOURCE) {
+    side->srcNode = node;
+    side->types = n->colTypes;
+    side->fracs = n->colFracs;
+    return GX_OK;
+  }
+  ex->err = "join children must be [Selection ->] Source or a join";
+  return GX_ERR_INVALID;
 }
 
-// post-join filter conjuncts (a Selection over the join = the join's "other
+// post-join filter conjuncts (a Selection over the join = the join\'s "other
 // conditions", inner_join_probe.go:75): <output col cmp const> or
 // <output col cmp output col> (e.g. build.x < probe.y)
 static int32_t compilePostJoinPreds(gx_exec* ex, const PNode& sel,
-                                    const PNode& bN, const PNode& pN) {
+                                    const std::vector<int>& outTypes,
+                                    const std::vector<int>& outFracs, int nb,
+                                    gxp::HashJoinDesc& hj) {
   const PPlan& plan = ex->plan;
-  gxp::HashJoinDesc& hj = ex->hj;
-  int nb = (int)bN.colTypes.size();
   if (sel.exprs.size() > 4) {
     ex->err = "too many post-join filter conjuncts this round";
     return GX_ERR_INVALID;
   }
-  // join-output schema for const preds
   PNode outN;
-  outN.colTypes = bN.colTypes;
-  outN.colFracs = bN.colFracs;
-  outN.colTypes.insert(outN.colTypes.end(), pN.colTypes.begin(), pN.colTypes.end());
-  outN.colFracs.insert(outN.colFracs.end(), pN.colFracs.begin(), pN.colFracs.end());
+  outN.colTypes = outTypes;
+  outN.colFracs = outFracs;
   for (int cid : sel.exprs) {
     const PExpr& e = plan.exprs[cid];
     gxp::JoinPostPred q{};
@@ -1141,13 +1132,98 @@ static int32_t compilePostJoinPreds(gx_exec* ex, const PNode& sel,
   return GX_OK;
 }
 
+static int32_t compileJoinStage(gx_exec* ex, int joinNode,
+                                const PNode* postSel, int* stageOut) {
+  const PPlan& plan = ex->plan;
+  const PNode& jn = plan.nodes[joinNode];
+  if (jn.joinType != 0) {
+    ex->err = "only inner joins on device this round";
+    return GX_ERR_INVALID;
+  }
+  if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
+      jn.buildKeys.size() > 2) {
+    ex->err = "device hash join supports 1-2 key columns this round";
+    return GX_ERR_INVALID;
+  }
+  JoinSideRef B, P;
+  int32_t rc = resolveJoinSide(ex, jn.child, &B);
+  if (rc) return rc;
+  rc = resolveJoinSide(ex, jn.child2, &P);
+  if (rc) return rc;
+  int nb = (int)B.types.size();
+  int np = (int)P.types.size();
+  if (nb + np > gxp::kMaxCols) {
+    ex->err = "too many join output columns";
+    return GX_ERR_INVALID;
+  }
+  gx_exec::JoinStage st;
+  gxp::HashJoinDesc& hj = st.hj;
+  hj.nKeys = (int32_t)jn.buildKeys.size();
+  for (int k = 0; k < hj.nKeys; k++) {
+    const PExpr& bk = plan.exprs[jn.buildKeys[k]];
+    const PExpr& pk = plan.exprs[jn.probeKeys[k]];
+    if (bk.kind != EK_COLREF || pk.kind != EK_COLREF || bk.colIdx < 0 ||
+        bk.colIdx >= nb || pk.colIdx < 0 || pk.colIdx >= np) {
+      ex->err = "join keys must be child columns";
+      return GX_ERR_INVALID;
+    }
+    int bt = B.types[bk.colIdx];
+    int pt = P.types[pk.colIdx];
+    // fixed 8-byte keys: int64 or packed CoreTime (SerializeKeys writes the
+    // raw 8 bytes for both - equality is bitwise)
+    if (bt != pt || (bt != GX_TYPE_I64 && bt != GX_TYPE_TIME)) {
+      ex->err = "device join keys must be int64/time this round";
+      return GX_ERR_INVALID;
+    }
+    hj.bKeyCol[k] = bk.colIdx;
+    hj.pKeyCol[k] = pk.colIdx;
+  }
+  // one filter conjunct per SOURCE side (stage-fed sides carry their filter
+  // as the inner stage\'s post preds)
+  auto doPred = [&](const PNode* sel, int srcNode, gxp::PredDesc* pd,
+                    int32_t* n, uint8_t* sc, int32_t* scLen) -> bool {
+    *n = 0;
+    if (!sel) return true;
+    if (sel->exprs.size() != 1) {
+      ex->err = "device join path supports one filter conjunct per table";
+      return false;
+    }
+    if (!compileTablePred(ex, ex->plan.nodes[srcNode], sel->exprs[0], pd, sc,
+                          scLen))
+      return false;
+    *n = 1;
+    return true;
+  };
+  if (!doPred(B.sel, B.srcNode, &hj.predB, &hj.nPredB, hj.strConstB,
+              &hj.strConstBLen))
+    return GX_ERR_INVALID;
+  if (!doPred(P.sel, P.srcNode, &hj.predP, &hj.nPredP, hj.strConstP,
+              &hj.strConstPLen))
+    return GX_ERR_INVALID;
+  st.srcB = B.srcNode;
+  st.srcP = P.srcNode;
+  st.buildStage = B.stage;
+  st.probeStage = P.stage;
+  st.types = B.types;
+  st.fracs = B.fracs;
+  st.types.insert(st.types.end(), P.types.begin(), P.types.end());
+  st.fracs.insert(st.fracs.end(), P.fracs.begin(), P.fracs.end());
+  if (postSel) {
+    rc = compilePostJoinPreds(ex, *postSel, st.types, st.fracs, nb, hj);
+    if (rc) return rc;
+  }
+  ex->joinStages.push_back(std::move(st));
+  *stageOut = (int)ex->joinStages.size() - 1;
+  return GX_OK;
+}
+
 // [Selection ->] HashJoin subtree (the Selection = post-join other
-// conditions)
+// conditions); nested joins compile into stages, root stage last
 static int32_t compileHashJoinTree(gx_exec* ex, int node) {
   const PNode* n = &ex->plan.nodes[node];
-  int selNode = -1;
+  const PNode* sel = nullptr;
   if (n->kind == PK_SELECTION) {
-    selNode = node;
+    sel = n;
     node = n->child;
     n = &ex->plan.nodes[node];
   }
@@ -1155,13 +1231,16 @@ static int32_t compileHashJoinTree(gx_exec* ex, int node) {
     ex->err = "expected a hash join under the selection";
     return GX_ERR_INVALID;
   }
-  int32_t rc = compileHashJoin(ex, node);
+  int root = -1;
+  int32_t rc = compileJoinStage(ex, node, sel, &root);
   if (rc) return rc;
-  if (selNode >= 0)
-    rc = compilePostJoinPreds(ex, ex->plan.nodes[selNode],
-                              ex->plan.nodes[ex->hjSrcB],
-                              ex->plan.nodes[ex->hjSrcP]);
-  return rc;
+  // root stage must be last (bottom-up compile order guarantees it)
+  gx_exec::JoinStage& st = ex->joinStages[root];
+  ex->desc.table.nCols = (int)st.types.size();
+  for (size_t c = 0; c < st.types.size(); c++)
+    setDevColMeta(&ex->desc.table.cols[c], st.types[c], st.fracs[c]);
+  ex->isHashJoin = true;
+  return GX_OK;
 }
 
 static int32_t compileFused(gx_exec* ex);
@@ -1190,15 +1269,11 @@ static int32_t compileAggOverJoin(gx_exec* ex, int aggNode) {
   if (rc) return rc;
   ex->isHashJoin = false;  // dispatched through aggOverJoin instead
   ex->aggOverJoin = true;
-  // pseudo source carrying the join output schema
+  // pseudo source carrying the join output schema (root stage)
   PNode ps;
   ps.kind = PK_SOURCE;
-  const PNode& bN = plan.nodes[ex->hjSrcB];
-  const PNode& pN = plan.nodes[ex->hjSrcP];
-  ps.colTypes = bN.colTypes;
-  ps.colFracs = bN.colFracs;
-  ps.colTypes.insert(ps.colTypes.end(), pN.colTypes.begin(), pN.colTypes.end());
-  ps.colFracs.insert(ps.colFracs.end(), pN.colFracs.begin(), pN.colFracs.end());
+  ps.colTypes = ex->joinStages.back().types;
+  ps.colFracs = ex->joinStages.back().fracs;
   plan.nodes.push_back(ps);
   int cur = (int)plan.nodes.size() - 1;
   // clone the agg-side chain bottom-up with the join replaced by the pseudo
@@ -2578,30 +2653,24 @@ static int32_t runJoinAgg(gx_exec* ex) {
 
 // ---------------- standalone hash join execution ----------------
 
-static int32_t runHashJoin(gx_exec* ex) {
-  gxp::HashJoinDesc& hj = ex->hj;
-  if (!ex->deviceReady) {
-    if (!gpuAvailable()) {
-      ex->err = "no MI355X visible: the product engine has no CPU fallback "
-                "(GX_ERR_NO_GPU)";
-      return GX_ERR_NO_GPU;
+// run one join stage: chain build -> count -> fill -> [post filter] ->
+// gather into st.out. Sides come from bound sources (materialized once) or
+// previous stages' outputs.
+static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
+  gxp::HashJoinDesc& hj = st.hj;
+  if (!st.inputsReady) {
+    if (st.srcB >= 0) {
+      int32_t rc = materializeTable(ex, st.srcB, &st.buildTab);
+      if (rc) return rc;
     }
-    if (ex->device >= 0) hipSetDevice(ex->device);
-    HIP_OK(ex, hipStreamCreate(&ex->stream));
-    int32_t rc = materializeTable(ex, ex->hjSrcB, &hj.build);
-    if (rc) return rc;
-    rc = materializeTable(ex, ex->hjSrcP, &hj.probe);
-    if (rc) return rc;
-    ex->devErr = (uint32_t*)devAlloc(ex, 4);
-    hj.counters = (uint64_t*)devAlloc(ex, 3 * 8);
-    ex->devHj = (gxp::HashJoinDesc*)devAlloc(ex, sizeof(gxp::HashJoinDesc));
-    if (!ex->devErr || !hj.counters || !ex->devHj) {
-      ex->err = "hipMalloc failed";
-      return GX_ERR_INTERNAL;
+    if (st.srcP >= 0) {
+      int32_t rc = materializeTable(ex, st.srcP, &st.probeTab);
+      if (rc) return rc;
     }
-    hj.errorFlag = ex->devErr;
-    ex->deviceReady = true;
+    st.inputsReady = true;
   }
+  hj.build = st.srcB >= 0 ? st.buildTab : ex->joinStages[st.buildStage].out;
+  hj.probe = st.srcP >= 0 ? st.probeTab : ex->joinStages[st.probeStage].out;
   int64_t nb = hj.build.nRows;
   if (nb >= 0xFFFFFFFFLL || hj.probe.nRows > 0xFFFFFFFFLL) {
     ex->err = "join sides > 2^32 rows unsupported this round";
@@ -2647,7 +2716,11 @@ static int32_t runHashJoin(gx_exec* ex) {
     ex->err = "join output > 2^31 rows unsupported this round";
     return GX_ERR_INVALID;
   }
-  ex->lastSelCount = total;
+  // output table (schema fixed at compile; buffers per run)
+  st.out.nCols = (int)st.types.size();
+  st.out.nRows = 0;
+  for (size_t c = 0; c < st.types.size(); c++)
+    setDevColMeta(&st.out.cols[c], st.types[c], st.fracs[c]);
   if (total > 0) {
     hj.outBuild = (uint32_t*)devAlloc(ex, total * 4);
     hj.outProbe = (uint32_t*)devAlloc(ex, total * 4);
@@ -2684,17 +2757,16 @@ static int32_t runHashJoin(gx_exec* ex) {
       uint64_t total2 = 0;
       HIP_OK(ex, hipMemcpy(&total2, hj.counters + 2, 8, hipMemcpyDeviceToHost));
       total = total2;
-      ex->lastSelCount = total;
       gatherB = hj.outBuild2;
       gatherP = hj.outProbe2;
     }
     // gather every output column through its side's match index
-    for (int c = 0; c < ex->desc.table.nCols && total > 0; c++) {
+    for (int c = 0; c < st.out.nCols && total > 0; c++) {
       int nbc = hj.build.nCols;
       const gxp::DevCol& src =
           c < nbc ? hj.build.cols[c] : hj.probe.cols[c - nbc];
       const uint32_t* idx = c < nbc ? gatherB : gatherP;
-      gxp::DevCol& dst = ex->desc.table.cols[c];
+      gxp::DevCol& dst = st.out.cols[c];
       if (src.type == GX_TYPE_STRING && !src.denseOffsets) {
         // general varlen gather: lengths -> exclusive scan -> byte copy
         int64_t* lens = (int64_t*)devAlloc(ex, total * 8);
@@ -2769,16 +2841,17 @@ static int32_t runHashJoin(gx_exec* ex) {
   {
     float ms = 0;
     hipEventElapsedTime(&ms, ev0, ev1);
-    ex->lastKernelMs = ms;
+    ex->lastKernelMs += ms;
+    if (isRoot) ex->lastSelCount = total;
     if (getenv("GX_DEBUG")) {
       float mb = 0, mc = 0, mf = 0;
       hipEventElapsedTime(&mb, ev0, evB);
       hipEventElapsedTime(&mc, evB, evC);
       if (total > 0) hipEventElapsedTime(&mf, evC, evF);
       fprintf(stderr,
-              "[gx] hj build=%.3fms count=%.3fms fill=%.3fms gather=%.3fms "
+              "[gx] hj%s build=%.3fms count=%.3fms fill=%.3fms gather=%.3fms "
               "total=%.3fms matches=%llu\n",
-              mb, mc, mf, ms - mb - mc - mf, ms,
+              isRoot ? "" : " (inner)", mb, mc, mf, ms - mb - mc - mf, ms,
               (unsigned long long)total);
     }
     hipEventDestroy(ev0);
@@ -2787,7 +2860,45 @@ static int32_t runHashJoin(gx_exec* ex) {
     hipEventDestroy(evC);
     hipEventDestroy(evF);
   }
-  ex->desc.table.nRows = (int64_t)total;
+  st.out.nRows = (int64_t)total;
+  return GX_OK;
+}
+
+static int32_t runHashJoin(gx_exec* ex) {
+  if (!ex->deviceReady) {
+    if (!gpuAvailable()) {
+      ex->err = "no MI355X visible: the product engine has no CPU fallback "
+                "(GX_ERR_NO_GPU)";
+      return GX_ERR_NO_GPU;
+    }
+    if (ex->device >= 0) hipSetDevice(ex->device);
+    HIP_OK(ex, hipStreamCreate(&ex->stream));
+    ex->devErr = (uint32_t*)devAlloc(ex, 4);
+    ex->devHj = (gxp::HashJoinDesc*)devAlloc(ex, sizeof(gxp::HashJoinDesc));
+    if (!ex->devErr || !ex->devHj) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+    ex->deviceReady = true;
+  }
+  ex->lastKernelMs = 0;
+  for (size_t s = 0; s < ex->joinStages.size(); s++) {
+    gx_exec::JoinStage& st = ex->joinStages[s];
+    if (!st.hj.counters) {
+      st.hj.counters = (uint64_t*)devAlloc(ex, 3 * 8);
+      if (!st.hj.counters) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    }
+    st.hj.errorFlag = ex->devErr;
+    int32_t rc = runJoinStage(ex, st, s + 1 == ex->joinStages.size());
+    if (rc) return rc;
+  }
+  // root stage output becomes the engine's resident table (emission / sort /
+  // fused aggregation all read desc.table)
+  gx_exec::JoinStage& root = ex->joinStages.back();
+  for (int c = 0; c < root.out.nCols; c++)
+    ex->desc.table.cols[c] = root.out.cols[c];
+  ex->desc.table.nCols = root.out.nCols;
+  ex->desc.table.nRows = root.out.nRows;
   ex->srcPos = 0;
   return GX_OK;
 }

@@ -731,6 +731,7 @@ def test_q3_class_agg_over_nested_join_parity(libs):
     from tests.gxlib import (GX_AGG_COUNT, GX_AGG_SUM, GX_F_EQ, GX_F_GT,
                              GX_F_MINUS, GX_F_MUL, GX_TPCH_CUSTOMER,
                              GX_TPCH_LINEITEM, GX_TPCH_ORDERS)
+    oracle, product = libs
 
     def run(lib):
         b = P.Builder(lib)

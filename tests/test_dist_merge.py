@@ -116,3 +116,60 @@ def test_product_final_merge_matches_oracle():
         partials.extend(run_q1(oracle, 7000, GX_AGG_MODE_PARTIAL,
                                row_offset=off, total_rows=21000))
     assert merge_partials(product, partials) == merge_partials(oracle, partials)
+
+
+def test_final_merge_minmax_firstrow_both_libs():
+    """MergePartialResult for MIN/MAX/FIRSTROW (func_max_min.go, aggfuncs.go
+    merge semantics) on BOTH libraries' host FINAL path: extreme of per-shard
+    extremes (NULL shard partials skipped), firstrow = first partial in input
+    order (its value may be NULL)."""
+    from tests.gxlib import (GX_AGG_FIRSTROW, GX_AGG_MAX, GX_AGG_MIN,
+                             GX_AGG_MODE_FINAL, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                             load_oracle, load_product)
+    from tidb_amd import plan as P
+    from tidb_amd.chunkpy import PyChunk
+    from tidb_amd.decimals import str_to_decimal_bytes
+
+    part_types = [GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_I64]
+    part_fracs = [0, 2, 0, 0]
+    # rows: group, min-partial(dec), max-partial(i64), firstrow-partial(i64)
+    partials = [
+        (1, "2.50", 10, 7),
+        (1, "-1.25", 40, None),
+        (2, None, None, None),  # all-NULL shard partials
+        (2, "9.00", -5, 3),
+        (3, "0.00", 0, 0),
+    ]
+    expected = {
+        1: ("-1.25", 40, 7),
+        2: ("9.00", -5, None),  # firstrow: group 2's FIRST partial was NULL
+        3: ("0.00", 0, 0),
+    }
+
+    def run(lib):
+        b = P.Builder(lib)
+        src = b.source(part_types, part_fracs)
+        agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                        [(GX_AGG_MIN, b.colref(1, GX_TYPE_DECIMAL, 2), 2),
+                         (GX_AGG_MAX, b.colref(2, GX_TYPE_I64), 0),
+                         (GX_AGG_FIRSTROW, b.colref(3, GX_TYPE_I64), 0)],
+                        GX_AGG_MODE_FINAL)
+        chunk = PyChunk(part_types, len(partials), part_fracs)
+        for r in partials:
+            vals = [r[0],
+                    None if r[1] is None else str_to_decimal_bytes(lib, r[1]),
+                    r[2], r[3]]
+            chunk.append_row(vals)
+        ex = b.build(agg)
+        ex.bind_chunks(src, [chunk])
+        ex.open()
+        rows = ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                            GX_TYPE_I64], [0, 2, 0, 0])
+        ex.close()
+        ex.free()
+        b.free()
+        return {r[0]: tuple(r[1:]) for r in rows}
+
+    got_o = run(load_oracle())
+    got_p = run(load_product())
+    assert got_o == got_p == expected

@@ -2918,6 +2918,48 @@ static int32_t runFinalHost(gx_exec* ex) {
     std::vector<OutRowVal> groupVals;
     std::vector<MyDecimal> dec;
     std::vector<int64_t> cnt;
+    std::vector<OutRowVal> val;   // min/max/firstrow merged value
+    std::vector<uint8_t> has;
+  };
+  // decode one partial cell (min/max/firstrow value columns)
+  auto readCell = [](const HostCol& hc, int i) {
+    OutRowVal v;
+    v.type = hc.type;
+    bool notNull = (hc.nullBitmap[i / 8] >> (i % 8)) & 1;
+    if (!notNull) {
+      v.isNull = true;
+    } else if (hc.type == GX_TYPE_STRING) {
+      int64_t s0 = hc.offsets[i], e0 = hc.offsets[i + 1];
+      v.str.assign((const char*)hc.data.data() + s0, e0 - s0);
+    } else if (hc.type == GX_TYPE_DECIMAL) {
+      std::memcpy(&v.dec, hc.data.data() + i * 40, 40);
+    } else {
+      std::memcpy(&v.u64, hc.data.data() + i * 8, 8);
+      v.i64 = (int64_t)v.u64;
+      std::memcpy(&v.f64, hc.data.data() + i * 8, 8);
+    }
+    return v;
+  };
+  // MergePartialResult compare for extremes (func_max_min.go semantics;
+  // strings: binary collation with PAD SPACE)
+  auto cmpCell = [](const OutRowVal& a, const OutRowVal& b) -> int {
+    switch (a.type) {
+      case GX_TYPE_DECIMAL: return a.dec.Compare(b.dec);
+      case GX_TYPE_F64: return a.f64 < b.f64 ? -1 : (a.f64 > b.f64 ? 1 : 0);
+      case GX_TYPE_TIME: {
+        uint64_t x = a.u64 & ~0xFULL, y = b.u64 & ~0xFULL;
+        return x < y ? -1 : (x > y ? 1 : 0);
+      }
+      case GX_TYPE_STRING: {
+        std::string x = a.str, y = b.str;
+        while (!x.empty() && x.back() == ' ') x.pop_back();
+        while (!y.empty() && y.back() == ' ') y.pop_back();
+        int c = x.compare(y);
+        return c < 0 ? -1 : (c > 0 ? 1 : 0);
+      }
+      default:
+        return a.i64 < b.i64 ? -1 : (a.i64 > b.i64 ? 1 : 0);
+    }
   };
   std::map<std::string, FState> groups;
   std::vector<std::string> order;
@@ -2958,6 +3000,8 @@ static int32_t runFinalHost(gx_exec* ex) {
         st.groupVals = std::move(gvals);
         st.dec.resize(agg.aggFuncs.size());
         st.cnt.assign(agg.aggFuncs.size(), 0);
+        st.val.resize(agg.aggFuncs.size());
+        st.has.assign(agg.aggFuncs.size(), 0);
         git = groups.emplace(key, std::move(st)).first;
         order.push_back(key);
       }
@@ -2969,6 +3013,27 @@ static int32_t runFinalHost(gx_exec* ex) {
           int64_t c;
           std::memcpy(&c, ch[col].data.data() + i * 8, 8);
           st.cnt[a] += c;
+          col += 1;
+        } else if (f == GX_AGG_MIN || f == GX_AGG_MAX) {
+          // extreme of per-shard extremes; NULL partials (all-NULL shard)
+          // are skipped (func_max_min.go merge)
+          OutRowVal v = readCell(ch[col], i);
+          if (!v.isNull) {
+            if (!st.has[a] ||
+                (f == GX_AGG_MAX ? cmpCell(v, st.val[a]) > 0
+                                 : cmpCell(v, st.val[a]) < 0)) {
+              st.val[a] = std::move(v);
+              st.has[a] = 1;
+            }
+          }
+          col += 1;
+        } else if (f == GX_AGG_FIRSTROW) {
+          // first partial row of the group in input order (its value may
+          // legitimately be NULL: the group's first row had a NULL arg)
+          if (!st.has[a]) {
+            st.val[a] = readCell(ch[col], i);
+            st.has[a] = 1;
+          }
           col += 1;
         } else {  // SUM/AVG: decimal + count
           int64_t c;
@@ -3000,6 +3065,14 @@ static int32_t runFinalHost(gx_exec* ex) {
       if (f == GX_AGG_COUNT) {
         v.type = GX_TYPE_I64;
         v.i64 = st.cnt[a];
+      } else if (f == GX_AGG_MIN || f == GX_AGG_MAX ||
+                 f == GX_AGG_FIRSTROW) {
+        if (!st.has[a]) {
+          v.isNull = true;
+          v.type = GX_TYPE_DECIMAL;
+        } else {
+          v = st.val[a];
+        }
       } else if (f == GX_AGG_SUM) {
         v.type = GX_TYPE_DECIMAL;
         if (st.cnt[a] == 0) v.isNull = true;

@@ -173,3 +173,88 @@ def test_final_merge_minmax_firstrow_both_libs():
     got_o = run(load_oracle())
     got_p = run(load_product())
     assert got_o == got_p == expected
+
+
+def test_shards_final_merge_full_family():
+    """2 row-range shards, PARTIAL per shard, FINAL merge == COMPLETE, for the
+    full aggregate family incl. min/max/firstrow (oracle, CPU)."""
+    from tests.gxlib import (GX_AGG_COUNT, GX_AGG_FIRSTROW, GX_AGG_MAX,
+                             GX_AGG_MIN, GX_AGG_MODE_COMPLETE,
+                             GX_AGG_MODE_FINAL, GX_AGG_MODE_PARTIAL,
+                             GX_AGG_SUM, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                             GX_TYPE_STRING, load_oracle)
+    from tidb_amd import plan as P
+    from tidb_amd.chunkpy import PyChunk
+    from tidb_amd.decimals import str_to_decimal_bytes
+    lib = load_oracle()
+    total = 16000
+
+    AGGS = lambda b: [
+        (GX_AGG_SUM, b.colref(P.L_QUANTITY, GX_TYPE_DECIMAL, 2), 2),
+        (GX_AGG_MIN, b.colref(P.L_EXTPRICE, GX_TYPE_DECIMAL, 2), 2),
+        (GX_AGG_MAX, b.colref(P.L_ORDERKEY, GX_TYPE_I64), 0),
+        (GX_AGG_FIRSTROW, b.colref(P.L_RETFLAG, GX_TYPE_STRING), 0),
+        (GX_AGG_COUNT, -1, 0),
+    ]
+    GROUP = lambda b: [b.colref(P.L_RETFLAG, GX_TYPE_STRING),
+                       b.colref(P.L_LINESTATUS, GX_TYPE_STRING)]
+
+    def run(mode, rows, offset):
+        from tests.gxlib import GX_TPCH_LINEITEM
+        b = P.Builder(lib)
+        src = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+        agg = b.hashagg(src, GROUP(b), AGGS(b), mode)
+        ex = b.build(agg)
+        ex.bind_tpch(src, GX_TPCH_LINEITEM, rows, 42, offset, total)
+        ex.open()
+        if mode == GX_AGG_MODE_PARTIAL:
+            types = [GX_TYPE_STRING] * 2 + [GX_TYPE_DECIMAL, GX_TYPE_I64,
+                                            GX_TYPE_DECIMAL, GX_TYPE_I64,
+                                            GX_TYPE_STRING, GX_TYPE_I64]
+            fracs = [0, 0, 2, 0, 2, 0, 0, 0]
+        else:
+            types = [GX_TYPE_STRING] * 2 + [GX_TYPE_DECIMAL, GX_TYPE_DECIMAL,
+                                            GX_TYPE_I64, GX_TYPE_STRING,
+                                            GX_TYPE_I64]
+            fracs = [0, 0, 2, 2, 0, 0, 0]
+        out = ex.pull_all(types, fracs, data_caps=[2048] * len(types))
+        ex.close()
+        ex.free()
+        b.free()
+        return out, types, fracs
+
+    want, _, _ = run(GX_AGG_MODE_COMPLETE, total, 0)
+    p0, pt, pf = run(GX_AGG_MODE_PARTIAL, total // 2, 0)
+    p1, _, _ = run(GX_AGG_MODE_PARTIAL, total - total // 2, total // 2)
+
+    # FINAL merge of the gathered partial rows
+    b = P.Builder(lib)
+    src = b.source(pt, pf)
+    agg = b.hashagg(src, [b.colref(0, GX_TYPE_STRING),
+                          b.colref(1, GX_TYPE_STRING)],
+                    [(GX_AGG_SUM, b.colref(2, GX_TYPE_DECIMAL, 2), 2),
+                     (GX_AGG_MIN, b.colref(4, GX_TYPE_DECIMAL, 2), 2),
+                     (GX_AGG_MAX, b.colref(5, GX_TYPE_I64), 0),
+                     (GX_AGG_FIRSTROW, b.colref(6, GX_TYPE_STRING), 0),
+                     (GX_AGG_COUNT, -1, 0)], GX_AGG_MODE_FINAL)
+    chunk = PyChunk(pt, len(p0) + len(p1), pf, data_caps=[4096] * len(pt))
+    for r in p0 + p1:
+        vals = []
+        for v, t in zip(r, pt):
+            if t == GX_TYPE_DECIMAL and v is not None:
+                vals.append(str_to_decimal_bytes(lib, v))
+            else:
+                vals.append(v)
+        chunk.append_row(vals)
+    ex = b.build(agg)
+    ex.bind_chunks(src, [chunk])
+    ex.open()
+    types = [GX_TYPE_STRING] * 2 + [GX_TYPE_DECIMAL, GX_TYPE_DECIMAL,
+                                    GX_TYPE_I64, GX_TYPE_STRING, GX_TYPE_I64]
+    got = ex.pull_all(types, [0, 0, 2, 2, 0, 0, 0], data_caps=[2048] * 7)
+    ex.close()
+    ex.free()
+    b.free()
+    as_map = lambda rows: {(r[0], r[1]): tuple(r[2:]) for r in rows}
+    assert as_map(got) == as_map(want)
+    assert len(got) >= 4

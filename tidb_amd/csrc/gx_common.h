@@ -353,6 +353,13 @@ struct HashJoinDesc {
 // phases: 0 = build (chain insert), 1 = count matches, 2 = fill match pairs
 int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
                     const HashJoinDesc& h, void* stream);
+// standalone Selection compaction over h.probe (post[] = the CNF):
+// 0 = count survivors, 1 = fill survivor indices into outProbe
+int gxSelectPhase(int phase, const HashJoinDesc* devDesc,
+                  const HashJoinDesc& h, void* stream);
+// ascending u32 radix sort (survivor indices -> input row order)
+int gxSortU32Keys(const uint32_t* in, uint32_t* out, int64_t n, void* tmp,
+                  size_t* tmpBytes, void* stream);
 // gather a null bitmap through the match index (one thread per output byte)
 int gxGatherNulls(const uint8_t* inBitmap, const uint32_t* idx, uint8_t* out,
                   int64_t n, void* stream);

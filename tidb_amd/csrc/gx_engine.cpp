@@ -218,6 +218,9 @@ struct gx_exec {
   // output into desc.table, then the fused aggregation runs over it
   bool aggOverJoin = false;
   bool fusedBindDone = false;
+  // standalone Selection (compact survivors of a CNF over one source)
+  bool isSelect = false;
+  JoinStage selStage;  // probe side only (srcP); hj.post = the CNF
 
   ~gx_exec() {
     for (void* p : devBufs) hipFree(p);
@@ -1240,6 +1243,33 @@ static int32_t compileHashJoinTree(gx_exec* ex, int node) {
   for (size_t c = 0; c < st.types.size(); c++)
     setDevColMeta(&ex->desc.table.cols[c], st.types[c], st.fracs[c]);
   ex->isHashJoin = true;
+  return GX_OK;
+}
+
+// standalone Selection over a Source (SelectionExec, select.go:750-785):
+// survivors compact on device, columns gather through the survivor index
+static int32_t compileSelect(gx_exec* ex, int selNode) {
+  const PNode& sel = ex->plan.nodes[selNode];
+  int child = skipFullSort(ex, sel.child);
+  const PNode& srcN = ex->plan.nodes[child];
+  if (srcN.kind != PK_SOURCE) {
+    ex->err = "selection child must be a source";
+    return GX_ERR_INVALID;
+  }
+  if ((int)srcN.colTypes.size() > gxp::kMaxCols) {
+    ex->err = "too many source columns";
+    return GX_ERR_INVALID;
+  }
+  gx_exec::JoinStage& st = ex->selStage;
+  st.srcP = child;
+  st.types = srcN.colTypes;
+  st.fracs = srcN.colFracs;
+  int32_t rc = compilePostJoinPreds(ex, sel, st.types, st.fracs, 0, st.hj);
+  if (rc) return rc;
+  ex->desc.table.nCols = (int)st.types.size();
+  for (size_t c = 0; c < st.types.size(); c++)
+    setDevColMeta(&ex->desc.table.cols[c], st.types[c], st.fracs[c]);
+  ex->isSelect = true;
   return GX_OK;
 }
 
@@ -2653,6 +2683,86 @@ static int32_t runJoinAgg(gx_exec* ex) {
 
 // ---------------- standalone hash join execution ----------------
 
+// gather nCols columns of srcTab through a match/survivor index into
+// out->cols[dstBase..): fixed-width, dense char (identity offsets), general
+// varlen (two-pass), and null bitmaps
+static int32_t gatherCols(gx_exec* ex, const gxp::DevTable& srcTab,
+                          const uint32_t* idx, uint64_t total,
+                          gxp::DevTable* out, int dstBase) {
+  for (int c = 0; c < srcTab.nCols; c++) {
+    const gxp::DevCol& src = srcTab.cols[c];
+    gxp::DevCol& dst = out->cols[dstBase + c];
+    if (src.type == GX_TYPE_STRING && !src.denseOffsets) {
+      // general varlen gather: lengths -> exclusive scan -> byte copy
+      int64_t* lens = (int64_t*)devAlloc(ex, total * 8);
+      dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
+      if (!lens || !dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      if (gxp::gxGatherVarlenLens(src.offsets, idx, lens, (int64_t)total,
+                                  ex->stream) != 0) {
+        ex->err = "varlen lens launch failed";
+        return GX_ERR_INTERNAL;
+      }
+      size_t tmpBytes = 0;
+      gxp::gxExclusiveSumI64(lens, dst.offsets, (int64_t)total, nullptr,
+                             &tmpBytes, ex->stream);
+      void* tmp = devAlloc(ex, std::max<size_t>(tmpBytes, 1));
+      if (!tmp) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      HIP_OK(ex, hipMemsetAsync(dst.offsets, 0, 8, ex->stream));
+      if (gxp::gxExclusiveSumI64(lens, dst.offsets, (int64_t)total, tmp,
+                                 &tmpBytes, ex->stream) != 0) {
+        ex->err = "varlen offsets scan failed";
+        return GX_ERR_INTERNAL;
+      }
+      int64_t totalBytes = 0;
+      HIP_OK(ex, hipStreamSynchronize(ex->stream));
+      HIP_OK(ex, hipMemcpy(&totalBytes, dst.offsets + total, 8,
+                           hipMemcpyDeviceToHost));
+      dst.data = devAlloc(ex, std::max<int64_t>(totalBytes, 1));
+      if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      if (gxp::gxGatherVarlenBytes((const uint8_t*)src.data, src.offsets,
+                                   idx, dst.offsets, (uint8_t*)dst.data,
+                                   (int64_t)total, ex->stream) != 0) {
+        ex->err = "varlen bytes launch failed";
+        return GX_ERR_INTERNAL;
+      }
+      dst.denseOffsets = 0;
+    } else {
+      int es = src.type == GX_TYPE_DECIMAL ? 40
+               : (src.type == GX_TYPE_STRING ? 1 : 8);
+      dst.data = devAlloc(ex, (size_t)total * es + 16);
+      if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      if (gxp::gxSortGatherCol(src.data, dst.data, idx, (int64_t)total, es,
+                               ex->stream) != 0) {
+        ex->err = "gather launch failed";
+        return GX_ERR_INTERNAL;
+      }
+      if (src.type == GX_TYPE_STRING) {
+        dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
+        if (!dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        if (gxp::gxIotaOffsets(dst.offsets, (int64_t)total + 1, ex->stream) != 0) {
+          ex->err = "offsets launch failed";
+          return GX_ERR_INTERNAL;
+        }
+        dst.denseOffsets = 1;
+      }
+    }
+    if (src.hasNulls && src.nullBitmap) {
+      dst.nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8);
+      if (!dst.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      if (gxp::gxGatherNulls(src.nullBitmap, idx, dst.nullBitmap,
+                             (int64_t)total, ex->stream) != 0) {
+        ex->err = "null gather launch failed";
+        return GX_ERR_INTERNAL;
+      }
+      dst.hasNulls = 1;
+    } else {
+      dst.nullBitmap = nullptr;
+      dst.hasNulls = 0;
+    }
+  }
+  return GX_OK;
+}
+
 // run one join stage: chain build -> count -> fill -> [post filter] ->
 // gather into st.out. Sides come from bound sources (materialized once) or
 // previous stages' outputs.
@@ -2761,79 +2871,13 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
       gatherP = hj.outProbe2;
     }
     // gather every output column through its side's match index
-    for (int c = 0; c < st.out.nCols && total > 0; c++) {
-      int nbc = hj.build.nCols;
-      const gxp::DevCol& src =
-          c < nbc ? hj.build.cols[c] : hj.probe.cols[c - nbc];
-      const uint32_t* idx = c < nbc ? gatherB : gatherP;
-      gxp::DevCol& dst = st.out.cols[c];
-      if (src.type == GX_TYPE_STRING && !src.denseOffsets) {
-        // general varlen gather: lengths -> exclusive scan -> byte copy
-        int64_t* lens = (int64_t*)devAlloc(ex, total * 8);
-        dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
-        if (!lens || !dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-        if (gxp::gxGatherVarlenLens(src.offsets, idx, lens, (int64_t)total,
-                                    ex->stream) != 0) {
-          ex->err = "varlen lens launch failed";
-          return GX_ERR_INTERNAL;
-        }
-        size_t tmpBytes = 0;
-        gxp::gxExclusiveSumI64(lens, dst.offsets, (int64_t)total, nullptr,
-                               &tmpBytes, ex->stream);
-        void* tmp = devAlloc(ex, std::max<size_t>(tmpBytes, 1));
-        if (!tmp) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-        HIP_OK(ex, hipMemsetAsync(dst.offsets, 0, 8, ex->stream));
-        if (gxp::gxExclusiveSumI64(lens, dst.offsets, (int64_t)total, tmp,
-                                   &tmpBytes, ex->stream) != 0) {
-          ex->err = "varlen offsets scan failed";
-          return GX_ERR_INTERNAL;
-        }
-        int64_t totalBytes = 0;
-        HIP_OK(ex, hipStreamSynchronize(ex->stream));
-        HIP_OK(ex, hipMemcpy(&totalBytes, dst.offsets + total, 8,
-                             hipMemcpyDeviceToHost));
-        dst.data = devAlloc(ex, std::max<int64_t>(totalBytes, 1));
-        if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-        if (gxp::gxGatherVarlenBytes((const uint8_t*)src.data, src.offsets,
-                                     idx, dst.offsets, (uint8_t*)dst.data,
-                                     (int64_t)total, ex->stream) != 0) {
-          ex->err = "varlen bytes launch failed";
-          return GX_ERR_INTERNAL;
-        }
-        dst.denseOffsets = 0;
-      } else {
-        int es = src.type == GX_TYPE_DECIMAL ? 40
-                 : (src.type == GX_TYPE_STRING ? 1 : 8);
-        dst.data = devAlloc(ex, (size_t)total * es + 16);
-        if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-        if (gxp::gxSortGatherCol(src.data, dst.data, idx, (int64_t)total, es,
-                                 ex->stream) != 0) {
-          ex->err = "join gather launch failed";
-          return GX_ERR_INTERNAL;
-        }
-        if (src.type == GX_TYPE_STRING) {
-          dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
-          if (!dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-          if (gxp::gxIotaOffsets(dst.offsets, (int64_t)total + 1, ex->stream) != 0) {
-            ex->err = "join offsets launch failed";
-            return GX_ERR_INTERNAL;
-          }
-          dst.denseOffsets = 1;
-        }
-      }
-      if (src.hasNulls && src.nullBitmap) {
-        dst.nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8);
-        if (!dst.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-        if (gxp::gxGatherNulls(src.nullBitmap, idx, dst.nullBitmap,
-                               (int64_t)total, ex->stream) != 0) {
-          ex->err = "join null gather launch failed";
-          return GX_ERR_INTERNAL;
-        }
-        dst.hasNulls = 1;
-      } else {
-        dst.nullBitmap = nullptr;
-        dst.hasNulls = 0;
-      }
+    if (total > 0) {
+      gxp::DevTable bsub = hj.build;
+      gxp::DevTable psub = hj.probe;
+      int32_t rc = gatherCols(ex, bsub, gatherB, total, &st.out, 0);
+      if (rc == GX_OK)
+        rc = gatherCols(ex, psub, gatherP, total, &st.out, hj.build.nCols);
+      if (rc) return rc;
     }
   }
   HIP_OK(ex, hipEventRecord(ev1, ex->stream));
@@ -2899,6 +2943,98 @@ static int32_t runHashJoin(gx_exec* ex) {
     ex->desc.table.cols[c] = root.out.cols[c];
   ex->desc.table.nCols = root.out.nCols;
   ex->desc.table.nRows = root.out.nRows;
+  ex->srcPos = 0;
+  return GX_OK;
+}
+
+// ---------------- standalone selection execution ----------------
+
+static int32_t runSelect(gx_exec* ex) {
+  gx_exec::JoinStage& st = ex->selStage;
+  gxp::HashJoinDesc& hj = st.hj;
+  if (!ex->deviceReady) {
+    if (!gpuAvailable()) {
+      ex->err = "no MI355X visible: the product engine has no CPU fallback "
+                "(GX_ERR_NO_GPU)";
+      return GX_ERR_NO_GPU;
+    }
+    if (ex->device >= 0) hipSetDevice(ex->device);
+    HIP_OK(ex, hipStreamCreate(&ex->stream));
+    ex->devErr = (uint32_t*)devAlloc(ex, 4);
+    ex->devHj = (gxp::HashJoinDesc*)devAlloc(ex, sizeof(gxp::HashJoinDesc));
+    hj.counters = (uint64_t*)devAlloc(ex, 3 * 8);
+    if (!ex->devErr || !ex->devHj || !hj.counters) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+    hj.errorFlag = ex->devErr;
+    int32_t rc = materializeTable(ex, st.srcP, &st.probeTab);
+    if (rc) return rc;
+    ex->deviceReady = true;
+  }
+  hj.probe = st.probeTab;
+  if (hj.probe.nRows > 0xFFFFFFFFLL) {
+    ex->err = "selection > 2^32 rows unsupported this round";
+    return GX_ERR_INVALID;
+  }
+  hipEvent_t ev0, ev1;
+  HIP_OK(ex, hipEventCreate(&ev0));
+  HIP_OK(ex, hipEventCreate(&ev1));
+  HIP_OK(ex, hipEventRecord(ev0, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(hj.counters, 0, 24, ex->stream));
+  HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+  HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj), hipMemcpyHostToDevice,
+                            ex->stream));
+  if (gxp::gxSelectPhase(0, ex->devHj, hj, ex->stream) != 0) {
+    ex->err = "selection kernel launch failed";
+    return GX_ERR_INTERNAL;
+  }
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  uint64_t total = 0;
+  HIP_OK(ex, hipMemcpy(&total, hj.counters, 8, hipMemcpyDeviceToHost));
+  st.out.nCols = (int)st.types.size();
+  st.out.nRows = 0;
+  for (size_t c = 0; c < st.types.size(); c++)
+    setDevColMeta(&st.out.cols[c], st.types[c], st.fracs[c]);
+  if (total > 0) {
+    hj.outProbe = (uint32_t*)devAlloc(ex, total * 4);
+    if (!hj.outProbe) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj),
+                              hipMemcpyHostToDevice, ex->stream));
+    if (gxp::gxSelectPhase(1, ex->devHj, hj, ex->stream) != 0) {
+      ex->err = "selection fill launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    // restore input row order (SelectionExec appends survivors in order;
+    // the wave-aggregated compaction is order-free)
+    uint32_t* sorted = (uint32_t*)devAlloc(ex, total * 4);
+    size_t tmpBytes = 0;
+    gxp::gxSortU32Keys(hj.outProbe, sorted, (int64_t)total, nullptr,
+                       &tmpBytes, ex->stream);
+    void* tmp = devAlloc(ex, std::max<size_t>(tmpBytes, 1));
+    if (!sorted || !tmp) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    if (gxp::gxSortU32Keys(hj.outProbe, sorted, (int64_t)total, tmp,
+                           &tmpBytes, ex->stream) != 0) {
+      ex->err = "selection index sort failed";
+      return GX_ERR_INTERNAL;
+    }
+    int32_t rc = gatherCols(ex, hj.probe, sorted, total, &st.out, 0);
+    if (rc) return rc;
+  }
+  HIP_OK(ex, hipEventRecord(ev1, ex->stream));
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  {
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    ex->lastKernelMs = ms;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+  }
+  ex->lastSelCount = total;
+  st.out.nRows = (int64_t)total;
+  for (int c = 0; c < st.out.nCols; c++)
+    ex->desc.table.cols[c] = st.out.cols[c];
+  ex->desc.table.nRows = st.out.nRows;
   ex->srcPos = 0;
   return GX_OK;
 }
@@ -3961,8 +4097,10 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
   } else if (rn.kind == PK_TOPN &&
              ex->plan.nodes[rn.child].kind != PK_HASHJOIN &&
              !(ex->plan.nodes[rn.child].kind == PK_SELECTION &&
-               ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
-                   PK_HASHJOIN)) {
+               (ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
+                    PK_HASHJOIN ||
+                ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
+                    PK_SOURCE))) {
     int32_t rc = compileJoinAgg(ex);
     if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";
     (void)rc;
@@ -3978,15 +4116,28 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     int32_t rc = compileHashJoinTree(ex, root);
     if (rc != GX_OK && ex->err.empty()) ex->err = "join plan compilation failed";
     (void)rc;
+  } else if (rn.kind == PK_SELECTION &&
+             ex->plan.nodes[skipFullSort(ex, rn.child)].kind == PK_SOURCE) {
+    // standalone Selection (SelectionExec): survivors compacted on device
+    int32_t rc = compileSelect(ex, root);
+    if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
+    (void)rc;
   } else if (rn.kind == PK_TOPN &&
              (ex->plan.nodes[rn.child].kind == PK_HASHJOIN ||
               (ex->plan.nodes[rn.child].kind == PK_SELECTION &&
-               ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
-                   PK_HASHJOIN))) {
-    // ORDER BY / TopN over the joined rows: standalone join (+ post-join
-    // filter), then the device radix sort over the materialized join output
-    // table (sortexec/sort.go over a join child)
-    int32_t rc = compileHashJoinTree(ex, rn.child);
+               (ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
+                    PK_HASHJOIN ||
+                ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
+                    PK_SOURCE)))) {
+    // ORDER BY / TopN over joined or filtered rows: materialize the device
+    // table (join stages or selection compaction), then the device radix
+    // sort over it (sortexec/sort.go)
+    const PNode& cn = ex->plan.nodes[rn.child];
+    bool selOverSource =
+        cn.kind == PK_SELECTION &&
+        ex->plan.nodes[cn.child].kind == PK_SOURCE;
+    int32_t rc = selOverSource ? compileSelect(ex, rn.child)
+                               : compileHashJoinTree(ex, rn.child);
     if (rc != GX_OK) {
       if (ex->err.empty()) ex->err = "join plan compilation failed";
       return ex;
@@ -4081,7 +4232,7 @@ int32_t gx_open(gx_exec* ex) {
   if (!ex) return GX_ERR_INVALID;
   if (!ex->err.empty()) return GX_ERR_INVALID;
   if (!ex->isFused && !ex->isFinalHost && !ex->isBareSource && !ex->isJoinAgg &&
-      !ex->isHashJoin) {
+      !ex->isHashJoin && !ex->isSelect) {
     ex->err = "plan not executable";
     return GX_ERR_INVALID;
   }
@@ -4101,6 +4252,24 @@ int32_t gx_open(gx_exec* ex) {
 int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   if (!ex || !ex->opened) return GX_ERR_INVALID;
   if (ex->isBareSource) return emitSourceChunk(ex, out, rows_out);
+  if (ex->isSelect) {
+    if (!ex->ranQuery) {
+      int32_t rc = runSelect(ex);
+      if (rc) {
+        *rows_out = 0;
+        return rc;
+      }
+      ex->ranQuery = true;
+    }
+    if (!ex->devSortKeys.empty() && !ex->devSorted) {
+      int32_t rc = runDeviceSort(ex);
+      if (rc) {
+        *rows_out = 0;
+        return rc;
+      }
+    }
+    return emitTableChunk(ex, out, rows_out);
+  }
   if (ex->isHashJoin) {
     if (!ex->ranQuery) {
       int32_t rc = runHashJoin(ex);

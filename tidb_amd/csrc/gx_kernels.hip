@@ -1936,6 +1936,44 @@ __global__ void hjFilterPairsKernel(const HashJoinDesc* __restrict__ dp) {
   }
 }
 
+// standalone Selection (SelectionExec, select.go:750-785): compact the rows
+// surviving a CNF into a survivor index, MI355X-shaped — no row-at-a-time
+// AppendRow copy; output columns gather through the index afterwards.
+// Conjuncts ride HashJoinDesc.post evaluated against d.probe (build side
+// empty). FILL=false counts into counters[0]; FILL=true reserves
+// wave-aggregated slices of counters[1] and writes row indices.
+template <bool FILL>
+__global__ void selCompactKernel(const HashJoinDesc* __restrict__ dp) {
+  const HashJoinDesc& d = *dp;
+  int64_t n = d.probe.nRows;
+  int lane = threadIdx.x & 63;
+  uint64_t my = 0;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;;
+       row += stride) {
+    bool active = row < n;
+    if (__ballot(active) == 0) break;
+    bool keep = active && hjPostPass(d, 0, (uint32_t)row);
+    if (FILL) {
+      uint64_t m = __ballot(keep);
+      if (m == 0) continue;
+      uint64_t base = 0;
+      if (lane == 0)
+        base = atomicAdd((unsigned long long*)&d.counters[1],
+                         (unsigned long long)__popcll(m));
+      base = __shfl(base, 0, 64) + __popcll(m & ((1ULL << lane) - 1));
+      if (keep) d.outProbe[base] = (uint32_t)row;
+    } else if (keep) {
+      my++;
+    }
+  }
+  if (!FILL) {
+    for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
+    if (lane == 0 && my)
+      atomicAdd((unsigned long long*)&d.counters[0], (unsigned long long)my);
+  }
+}
+
 // gather a null bitmap through the match index: one thread composes one
 // output byte (8 rows) — no atomics (LSB-first, 1 = NOT NULL)
 __global__ void hjGatherNullsKernel(const uint8_t* __restrict__ in,
@@ -2094,6 +2132,28 @@ int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
     hipLaunchKernelGGL(hjFilterPairsKernel, g, dim3(256), 0,
                        (hipStream_t)stream, devDesc);
   return (int)hipGetLastError();
+}
+
+int gxSelectPhase(int phase, const HashJoinDesc* devDesc,
+                  const HashJoinDesc& h, void* stream) {
+  if (h.probe.nRows == 0) return 0;
+  dim3 g(gridFor(h.probe.nRows));
+  if (phase == 0)
+    hipLaunchKernelGGL(selCompactKernel<false>, g, dim3(256), 0,
+                       (hipStream_t)stream, devDesc);
+  else
+    hipLaunchKernelGGL(selCompactKernel<true>, g, dim3(256), 0,
+                       (hipStream_t)stream, devDesc);
+  return (int)hipGetLastError();
+}
+
+// ascending radix sort of survivor indices (restores SelectionExec's
+// input-order emission after the wave-aggregated compaction)
+int gxSortU32Keys(const uint32_t* in, uint32_t* out, int64_t n, void* tmp,
+                  size_t* tmpBytes, void* stream) {
+  return (int)hipcub::DeviceRadixSort::SortKeys(tmp, *tmpBytes, in, out,
+                                                (int)n, 0, 32,
+                                                (hipStream_t)stream);
 }
 
 int gxGatherNulls(const uint8_t* inBitmap, const uint32_t* idx, uint8_t* out,

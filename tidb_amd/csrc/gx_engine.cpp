@@ -212,8 +212,6 @@ struct gx_exec {
   bool isHashJoin = false;
   std::vector<JoinStage> joinStages;     // root stage last
   gxp::HashJoinDesc* devHj = nullptr;
-  int hjSrcB = -1, hjSrcP = -1;          // ROOT stage's source ids (used by
-                                         // compileAggOverJoin's schema)
   // general aggregation over joined rows: runHashJoin materializes the join
   // output into desc.table, then the fused aggregation runs over it
   bool aggOverJoin = false;

@@ -360,6 +360,36 @@ int gxSelectPhase(int phase, const HashJoinDesc* devDesc,
 // ascending u32 radix sort (survivor indices -> input row order)
 int gxSortU32Keys(const uint32_t* in, uint32_t* out, int64_t n, void* tmp,
                   size_t* tmpBytes, void* stream);
+
+// ---- standalone Projection (ProjectionExec, projection.go:77) ----
+// Computed expressions evaluate per row in the direct-load VM and MATERIALIZE
+// as device columns — decimals encode to the 40-byte MyDecimal struct ON
+// DEVICE (canonical word layout, mydecimal.go:236-248); passthrough column
+// references alias the source buffers (zero copy). Null flags write as one
+// byte per row, packed to the LSB-first bitmap in a second pass.
+struct ProjDesc {
+  DevTable table;
+  VmIns ins[kMaxVmIns];
+  int32_t nIns = 0;
+  int64_t constLo[kMaxVmConsts];
+  int64_t constHi[kMaxVmConsts];
+  int32_t nConsts = 0;
+  int64_t insP10[kMaxVmIns];
+  uint64_t insMagic[kMaxVmIns];
+  int32_t nOut = 0;           // computed outputs only
+  int32_t outReg[kMaxCols];   // VM register holding the value
+  int32_t outScale[kMaxCols]; // units scale of that register
+  int32_t outType[kMaxCols];  // GX_TYPE_DECIMAL or GX_TYPE_I64
+  void* outData[kMaxCols];    // 40 B/row (decimal) or 8 B/row (i64)
+  uint8_t* outNotNull[kMaxCols];  // byte per row, 1 = NOT NULL
+  uint32_t* errorFlag = nullptr;
+  int32_t wide = 0;
+};
+
+int gxProject(const ProjDesc* devDesc, const ProjDesc& h, void* stream);
+// pack byte-per-row not-null flags into the LSB-first bitmap
+int gxPackNulls(const uint8_t* notNullBytes, uint8_t* bitmap, int64_t n,
+                void* stream);
 // gather a null bitmap through the match index (one thread per output byte)
 int gxGatherNulls(const uint8_t* inBitmap, const uint32_t* idx, uint8_t* out,
                   int64_t n, void* stream);

@@ -219,6 +219,15 @@ struct gx_exec {
   // standalone Selection (compact survivors of a CNF over one source)
   bool isSelect = false;
   JoinStage selStage;  // probe side only (srcP); hj.post = the CNF
+  // standalone Projection (materialize computed columns on device)
+  bool isProject = false;
+  bool projOverSelect = false;
+  int projSrcNode = -1;
+  gxp::ProjDesc pd;
+  gxp::ProjDesc* devPd = nullptr;
+  std::vector<int> projOutTypes, projOutFracs;
+  std::vector<int> projOutSrcCol;  // >=0 passthrough; -1 computed
+  std::vector<int> projOutSlot;    // computed: index into pd.out*
 
   ~gx_exec() {
     for (void* p : devBufs) hipFree(p);
@@ -1268,6 +1277,134 @@ static int32_t compileSelect(gx_exec* ex, int selNode) {
   for (size_t c = 0; c < st.types.size(); c++)
     setDevColMeta(&ex->desc.table.cols[c], st.types[c], st.fracs[c]);
   ex->isSelect = true;
+  return GX_OK;
+}
+
+// standalone Projection over a Source or a Selection(Source)
+// (ProjectionExec, projection.go:77): computed expressions materialize as
+// device columns (decimal encode on device), passthrough colrefs alias the
+// input buffers
+static int32_t compileProject(gx_exec* ex) {
+  const PPlan& plan = ex->plan;
+  const PNode& pj = plan.nodes[ex->root];
+  int childNode = skipFullSort(ex, pj.child);
+  const PNode* cn = &plan.nodes[childNode];
+  const std::vector<int>* childTypes;
+  const std::vector<int>* childFracs;
+  if (cn->kind == PK_SELECTION &&
+      plan.nodes[skipFullSort(ex, cn->child)].kind == PK_SOURCE) {
+    int32_t rc = compileSelect(ex, childNode);
+    if (rc) return rc;
+    ex->isSelect = false;  // dispatched through isProject
+    ex->projOverSelect = true;
+    childTypes = &ex->selStage.types;
+    childFracs = &ex->selStage.fracs;
+  } else if (cn->kind == PK_SOURCE) {
+    ex->projSrcNode = childNode;
+    childTypes = &cn->colTypes;
+    childFracs = &cn->colFracs;
+  } else {
+    ex->err = "projection child must be [Selection ->] Source";
+    return GX_ERR_INVALID;
+  }
+  if ((int)pj.exprs.size() > gxp::kMaxCols) {
+    ex->err = "too many projection outputs";
+    return GX_ERR_INVALID;
+  }
+  gxp::ProjDesc& pd = ex->pd;
+  // the projection kernel loads directly (no fetch pipeline); give vmCompile
+  // a scratch fetch plan it can fill and we discard
+  gxp::FetchDesc scratchFetch[gxp::kMaxFetch];
+  int32_t scratchNFetch = 0;
+  VmBuild B;
+  B.ins = pd.ins;
+  B.nIns = &pd.nIns;
+  B.cLo = pd.constLo;
+  B.cHi = pd.constHi;
+  B.nConsts = &pd.nConsts;
+  B.fetch = scratchFetch;
+  B.nFetch = &scratchNFetch;
+  B.colTypes = childTypes;
+  B.colFracs = childFracs;
+  B.colBase = 0;
+  B.wideFlag = &pd.wide;
+  for (int eid : pj.exprs) {
+    const PExpr& e = plan.exprs[eid];
+    if (e.kind == EK_COLREF) {
+      if (e.colIdx < 0 || e.colIdx >= (int)childTypes->size()) {
+        ex->err = "projection column out of range";
+        return GX_ERR_INVALID;
+      }
+      ex->projOutSrcCol.push_back(e.colIdx);
+      ex->projOutSlot.push_back(-1);
+      ex->projOutTypes.push_back((*childTypes)[e.colIdx]);
+      ex->projOutFracs.push_back((*childFracs)[e.colIdx]);
+    } else {
+      int sc = 0;
+      int reg = vmCompile(ex, B, eid, &sc);
+      if (reg < 0) {
+        if (ex->err.empty()) ex->err = "unsupported projection expression";
+        return GX_ERR_INVALID;
+      }
+      int outType = e.retType == GX_TYPE_I64 ? GX_TYPE_I64 : GX_TYPE_DECIMAL;
+      int o = pd.nOut++;
+      pd.outReg[o] = reg;
+      pd.outScale[o] = sc;
+      pd.outType[o] = outType;
+      ex->projOutSrcCol.push_back(-1);
+      ex->projOutSlot.push_back(o);
+      ex->projOutTypes.push_back(outType);
+      ex->projOutFracs.push_back(outType == GX_TYPE_I64 ? 0 : sc);
+    }
+  }
+  // per-instruction precomputed constants (same rules as the fused path)
+  static const int64_t p10h[19] = {1,
+                                   10,
+                                   100,
+                                   1000,
+                                   10000,
+                                   100000,
+                                   1000000,
+                                   10000000,
+                                   100000000,
+                                   1000000000,
+                                   10000000000LL,
+                                   100000000000LL,
+                                   1000000000000LL,
+                                   10000000000000LL,
+                                   100000000000000LL,
+                                   1000000000000000LL,
+                                   10000000000000000LL,
+                                   100000000000000000LL,
+                                   1000000000000000000LL};
+  static const uint64_t magich[10] = {
+      4611686018427387904ULL, 461168601842738791ULL, 46116860184273880ULL,
+      4611686018427388ULL,    461168601842739ULL,    46116860184274ULL,
+      4611686018428ULL,       461168601843ULL,       46116860185ULL,
+      4611686019ULL};
+  for (int i = 0; i < pd.nIns; i++) {
+    pd.insP10[i] = 1;
+    pd.insMagic[i] = 0;
+    if (pd.ins[i].op == gxp::VM_LOAD_DEC) {
+      pd.insP10[i] = p10h[pd.ins[i].b];
+      pd.insMagic[i] = magich[9 - pd.ins[i].b];
+    } else if (pd.ins[i].op == gxp::VM_SCALE_UP) {
+      pd.insP10[i] = p10h[pd.ins[i].b];
+    } else if (pd.ins[i].op == gxp::VM_ROUND_SCALE) {
+      int sh = pd.ins[i].b - pd.ins[i].c;
+      pd.insP10[i] = p10h[sh >= 0 ? sh : -sh];
+    }
+  }
+  // loads use the direct path (no fetch slots in the projection kernel)
+  for (int i = 0; i < pd.nIns; i++)
+    if (pd.ins[i].op == gxp::VM_LOAD_DEC || pd.ins[i].op == gxp::VM_LOAD_I64)
+      pd.ins[i].c = -1;
+  // OUTPUT schema for emission
+  ex->desc.table.nCols = (int)ex->projOutTypes.size();
+  for (size_t c = 0; c < ex->projOutTypes.size(); c++)
+    setDevColMeta(&ex->desc.table.cols[c], ex->projOutTypes[c],
+                  ex->projOutFracs[c]);
+  ex->isProject = true;
   return GX_OK;
 }
 
@@ -3037,6 +3174,115 @@ static int32_t runSelect(gx_exec* ex) {
   return GX_OK;
 }
 
+// ---------------- standalone projection execution ----------------
+
+static int32_t runProject(gx_exec* ex) {
+  gxp::ProjDesc& pd = ex->pd;
+  if (ex->projOverSelect) {
+    // materialize the filtered input first (runSelect leaves its output in
+    // selStage.out and clobbers desc.table meta; we restore below)
+    int32_t rc = runSelect(ex);
+    if (rc) return rc;
+    pd.table = ex->selStage.out;
+  } else {
+    if (!ex->deviceReady) {
+      if (!gpuAvailable()) {
+        ex->err = "no MI355X visible: the product engine has no CPU fallback "
+                  "(GX_ERR_NO_GPU)";
+        return GX_ERR_NO_GPU;
+      }
+      if (ex->device >= 0) hipSetDevice(ex->device);
+      HIP_OK(ex, hipStreamCreate(&ex->stream));
+      ex->devErr = (uint32_t*)devAlloc(ex, 4);
+      if (!ex->devErr) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      int32_t rc = materializeTable(ex, ex->projSrcNode, &ex->selStage.probeTab);
+      if (rc) return rc;
+      ex->deviceReady = true;
+    }
+    pd.table = ex->selStage.probeTab;
+  }
+  if (!ex->devPd) {
+    ex->devPd = (gxp::ProjDesc*)devAlloc(ex, sizeof(gxp::ProjDesc));
+    if (!ex->devPd) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+  }
+  pd.errorFlag = ex->devErr;
+  if (ex->vmHasDiv) pd.wide = 1;
+  int64_t n = pd.table.nRows;
+  hipEvent_t ev0, ev1;
+  HIP_OK(ex, hipEventCreate(&ev0));
+  HIP_OK(ex, hipEventCreate(&ev1));
+  HIP_OK(ex, hipEventRecord(ev0, ex->stream));
+  // computed-output buffers
+  std::vector<uint8_t*> bitmaps(pd.nOut, nullptr);
+  for (int o = 0; o < pd.nOut && n > 0; o++) {
+    int es = pd.outType[o] == GX_TYPE_DECIMAL ? 40 : 8;
+    pd.outData[o] = devAlloc(ex, (size_t)n * es);
+    pd.outNotNull[o] = (uint8_t*)devAlloc(ex, (size_t)n);
+    bitmaps[o] = (uint8_t*)devAlloc(ex, (n + 7) / 8);
+    if (!pd.outData[o] || !pd.outNotNull[o] || !bitmaps[o]) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+  }
+  for (int attempt = 0; n > 0 && attempt < 2; attempt++) {
+    HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+    HIP_OK(ex, hipMemcpyAsync(ex->devPd, &pd, sizeof(pd),
+                              hipMemcpyHostToDevice, ex->stream));
+    if (gxp::gxProject(ex->devPd, pd, ex->stream) != 0) {
+      ex->err = "projection kernel launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    uint32_t errFlag = 0;
+    HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+    if (errFlag == 256u /*kErrRetryWide*/ && !pd.wide) {
+      pd.wide = 1;  // int64 fast path overflowed: rerun with the int128 VM
+      continue;
+    }
+    errFlag &= ~256u;
+    if (errFlag != 0) {
+      ex->err = "device projection error flag 0x" + std::to_string(errFlag);
+      return GX_ERR_INTERNAL;
+    }
+    break;
+  }
+  for (int o = 0; o < pd.nOut && n > 0; o++) {
+    if (gxp::gxPackNulls(pd.outNotNull[o], bitmaps[o], n, ex->stream) != 0) {
+      ex->err = "null pack launch failed";
+      return GX_ERR_INTERNAL;
+    }
+  }
+  HIP_OK(ex, hipEventRecord(ev1, ex->stream));
+  HIP_OK(ex, hipStreamSynchronize(ex->stream));
+  {
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    ex->lastKernelMs += ms;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+  }
+  // assemble the output table: passthrough columns alias the input buffers
+  ex->desc.table.nCols = (int)ex->projOutTypes.size();
+  for (size_t c = 0; c < ex->projOutTypes.size(); c++) {
+    gxp::DevCol& dst = ex->desc.table.cols[c];
+    if (ex->projOutSrcCol[c] >= 0) {
+      dst = pd.table.cols[ex->projOutSrcCol[c]];
+    } else {
+      int o = ex->projOutSlot[c];
+      setDevColMeta(&dst, ex->projOutTypes[c], ex->projOutFracs[c]);
+      dst.data = pd.outData[o];
+      dst.offsets = nullptr;
+      dst.nullBitmap = n > 0 ? bitmaps[o] : nullptr;
+      dst.hasNulls = n > 0 ? 1 : 0;
+      dst.denseOffsets = 0;
+    }
+  }
+  ex->desc.table.nRows = n;
+  ex->lastSelCount = (uint64_t)n;
+  ex->srcPos = 0;
+  return GX_OK;
+}
+
 // ---------------- FINAL-mode host merge ----------------
 
 static int32_t runFinalHost(gx_exec* ex) {
@@ -4120,6 +4366,11 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     int32_t rc = compileSelect(ex, root);
     if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     (void)rc;
+  } else if (rn.kind == PK_PROJECTION) {
+    // standalone Projection (ProjectionExec): computed columns materialize
+    int32_t rc = compileProject(ex);
+    if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
+    (void)rc;
   } else if (rn.kind == PK_TOPN &&
              (ex->plan.nodes[rn.child].kind == PK_HASHJOIN ||
               (ex->plan.nodes[rn.child].kind == PK_SELECTION &&
@@ -4230,7 +4481,7 @@ int32_t gx_open(gx_exec* ex) {
   if (!ex) return GX_ERR_INVALID;
   if (!ex->err.empty()) return GX_ERR_INVALID;
   if (!ex->isFused && !ex->isFinalHost && !ex->isBareSource && !ex->isJoinAgg &&
-      !ex->isHashJoin && !ex->isSelect) {
+      !ex->isHashJoin && !ex->isSelect && !ex->isProject) {
     ex->err = "plan not executable";
     return GX_ERR_INVALID;
   }
@@ -4250,6 +4501,17 @@ int32_t gx_open(gx_exec* ex) {
 int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   if (!ex || !ex->opened) return GX_ERR_INVALID;
   if (ex->isBareSource) return emitSourceChunk(ex, out, rows_out);
+  if (ex->isProject) {
+    if (!ex->ranQuery) {
+      int32_t rc = runProject(ex);
+      if (rc) {
+        *rows_out = 0;
+        return rc;
+      }
+      ex->ranQuery = true;
+    }
+    return emitTableChunk(ex, out, rows_out);
+  }
   if (ex->isSelect) {
     if (!ex->ranQuery) {
       int32_t rc = runSelect(ex);

@@ -1974,6 +1974,285 @@ __global__ void selCompactKernel(const HashJoinDesc* __restrict__ dp) {
   }
 }
 
+// ==================================================================
+// standalone Projection — ProjDesc (gx_common.h)
+// ==================================================================
+
+// encode int128 units at `scale` into the canonical 40-byte MyDecimal
+// (mirrors the host decFromUnits digit-exactly: integer words
+// most-significant-first, frac words left-aligned 9-digit, mydecimal.go
+// word layout)
+__device__ inline void devDecEncode(__int128 u, int scale,
+                                    uint8_t* __restrict__ out40) {
+  bool neg = u < 0;
+  unsigned __int128 a = neg ? (unsigned __int128)(-u) : (unsigned __int128)u;
+  unsigned __int128 pscale = 1;
+  for (int s = scale; s > 0; s -= 9)
+    pscale *= (uint64_t)kP10(s > 9 ? 9 : s);
+  unsigned __int128 ip, fr128;
+  if ((pscale >> 64) != 0) {
+    ip = u128DivBig(a, pscale);
+  } else if (pscale > 1) {
+    ip = u128DivU64(a, (uint64_t)pscale);
+  } else {
+    ip = a;
+  }
+  fr128 = a - ip * pscale;
+  int32_t words[6] = {0};
+  int nw = 0;
+  if (ip == 0) {
+    words[0] = 0;
+    nw = 1;
+  } else {
+    int32_t tmp[6];
+    int k = 0;
+    while (ip > 0) {
+      unsigned __int128 q = u128DivU64(ip, 1000000000u);
+      tmp[k++] = (int32_t)(uint64_t)(ip - q * 1000000000u);
+      ip = q;
+    }
+    nw = k;
+    for (int i = 0; i < k; i++) words[i] = tmp[k - 1 - i];
+  }
+  int digitsInt = (nw - 1) * 9;
+  {
+    int32_t head = words[0];
+    int hd = 1;
+    while (head >= 10) {
+      head /= 10;
+      hd++;
+    }
+    digitsInt += hd;
+  }
+  out40[0] = (uint8_t)(int8_t)digitsInt;
+  out40[1] = (uint8_t)(int8_t)scale;
+  out40[2] = (uint8_t)(int8_t)scale;  // resultFrac
+  out40[3] = (uint8_t)(neg && a != 0 ? 1 : 0);
+  int32_t* wb = (int32_t*)(out40 + 4);
+  for (int i = 0; i < 9; i++) wb[i] = 0;
+  for (int i = 0; i < nw; i++) wb[i] = words[i];
+  if (scale > 0) {
+    int fw = (scale + 8) / 9;
+    int pad = fw * 9 - scale;
+    unsigned __int128 fadj = fr128 * (uint64_t)kP10(pad);
+    for (int i = fw - 1; i >= 0; i--) {
+      unsigned __int128 q = u128DivU64(fadj, 1000000000u);
+      wb[nw + i] = (int32_t)(uint64_t)(fadj - q * 1000000000u);
+      fadj = q;
+    }
+  }
+}
+
+// per-row direct-load VM + output materialization
+template <bool WIDE>
+__global__ void projectKernel(const ProjDesc* __restrict__ dp) {
+  const ProjDesc& d = *dp;
+  int64_t n = d.table.nRows;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    VmState14<WIDE> vm;
+    vm.nullBits = 0;
+    bool bad = false;
+    bool ovf = false;
+    for (int i = 0; i < d.nIns && !bad; i++) {
+      const VmIns& ins = d.ins[i];
+      switch (ins.op) {
+        case VM_LOAD_DEC: {
+          const DevCol& c = d.table.cols[ins.a];
+          bool nul = colIsNull(c, row);
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
+          if (!nul) {
+            int sc;
+            if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &v,
+                                        &sc, d.errorFlag)) {
+              bad = true;
+              break;
+            }
+            if (sc != ins.b) {
+              if (sc < ins.b) v = VT<WIDE>::scale10(v, ins.b - sc, &ovf);
+              else { atomicOr(d.errorFlag, kErrScale); bad = true; break; }
+            }
+          }
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        case VM_LOAD_I64: {
+          const DevCol& c = d.table.cols[ins.a];
+          bool nul = colIsNull(c, row);
+          vm.set(ins.dst,
+                 nul ? VT<WIDE>::zero()
+                     : VT<WIDE>::fromI64(gptr<int64_t>(c.data)[row], &ovf));
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        case VM_LOAD_CONST: {
+          if (WIDE) {
+            Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
+            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+          } else {
+            int64_t cv = d.constLo[ins.a];
+            vm.set(ins.dst, *(typename VT<WIDE>::T*)&cv);
+          }
+          vm.setNull(ins.dst, false);
+          break;
+        }
+        case VM_ADD:
+          vm.set(ins.dst, VT<WIDE>::add(vm.get(ins.a), vm.get(ins.b), &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+          break;
+        case VM_SUB:
+          vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+          break;
+        case VM_MUL: {
+          bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
+          if (!nul) v = VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf);
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        case VM_SCALE_UP:
+          vm.set(ins.dst,
+                 VT<WIDE>::mul(vm.get(ins.a),
+                               VT<WIDE>::fromI64(d.insP10[i], nullptr), &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        case VM_ROUND_SCALE: {
+          bool nul = vm.isNull(ins.a);
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
+          if (!nul) {
+            int up = ins.b - ins.c;
+            if (up >= 0) {
+              v = VT<WIDE>::mul(vm.get(ins.a),
+                                VT<WIDE>::fromI64(d.insP10[i], nullptr), &ovf);
+            } else {
+              Int128 ai = VT<WIDE>::toAcc(vm.get(ins.a));
+              __int128 av = ((__int128)ai.hi << 64) | (__int128)ai.lo;
+              uint64_t div = (uint64_t)d.insP10[i];
+              unsigned __int128 aAbs = (unsigned __int128)(av < 0 ? -av : av);
+              unsigned __int128 q = u128DivU64(aAbs, div);
+              unsigned __int128 r = aAbs - q * div;
+              if (2 * (uint64_t)r >= div) q += 1;
+              __int128 sq = av < 0 ? -(__int128)q : (__int128)q;
+              if (!WIDE &&
+                  (sq > (__int128)INT64_MAX || sq < (__int128)INT64_MIN)) {
+                atomicOr(d.errorFlag, kErrRetryWide);
+                bad = true;
+                break;
+              }
+              if (WIDE) {
+                Int128 rr = {(uint64_t)sq, (int64_t)(sq >> 64)};
+                v = *(typename VT<WIDE>::T*)&rr;
+              } else {
+                int64_t qq = (int64_t)sq;
+                v = *(typename VT<WIDE>::T*)&qq;
+              }
+            }
+          }
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        case VM_DIV: {
+          // DecimalDiv semantics (same as the fused DIVOK path): quotient
+          // truncated toward zero at the word-granular result scale;
+          // division by zero -> NULL
+          bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
+          typename VT<WIDE>::T v = VT<WIDE>::zero();
+          if (!nul) {
+            Int128 bi = VT<WIDE>::toAcc(vm.get(ins.b));
+            __int128 bv = ((__int128)bi.hi << 64) | (__int128)bi.lo;
+            if (bv == 0) {
+              nul = true;
+            } else {
+              Int128 ai = VT<WIDE>::toAcc(vm.get(ins.a));
+              __int128 av = ((__int128)ai.hi << 64) | (__int128)ai.lo;
+              int e = ins.c;
+              unsigned __int128 p10 =
+                  (unsigned __int128)(uint64_t)kP10(e > 18 ? 18 : e);
+              if (e > 18) p10 *= (uint64_t)kP10(e - 18);
+              unsigned __int128 aAbs = (unsigned __int128)(av < 0 ? -av : av);
+              unsigned __int128 lim =
+                  ((unsigned __int128)kDivArgMax[e][1] << 64) |
+                  kDivArgMax[e][0];
+              if (aAbs > lim) {
+                atomicOr(d.errorFlag, kErrOverflow);
+                bad = true;
+                break;
+              }
+              unsigned __int128 bAbs = (unsigned __int128)(bv < 0 ? -bv : bv);
+              unsigned __int128 num = aAbs * p10;
+              unsigned __int128 uq = (bAbs >> 64) != 0
+                                         ? u128DivBig(num, bAbs)
+                                         : u128DivU64(num, (uint64_t)bAbs);
+              bool negq = (av < 0) != (bv < 0);
+              __int128 q = negq ? -(__int128)uq : (__int128)uq;
+              if (!WIDE &&
+                  (q > (__int128)INT64_MAX || q < (__int128)INT64_MIN)) {
+                atomicOr(d.errorFlag, kErrRetryWide);
+                bad = true;
+                break;
+              }
+              if (WIDE) {
+                Int128 rr = {(uint64_t)q, (int64_t)(q >> 64)};
+                v = *(typename VT<WIDE>::T*)&rr;
+              } else {
+                int64_t qq = (int64_t)q;
+                v = *(typename VT<WIDE>::T*)&qq;
+              }
+            }
+          }
+          vm.set(ins.dst, v);
+          vm.setNull(ins.dst, nul);
+          break;
+        }
+        default:
+          atomicOr(d.errorFlag, kErrBadDecimal);
+          bad = true;
+          break;
+      }
+    }
+    if (ovf) {
+      atomicOr(d.errorFlag, WIDE ? kErrOverflow : kErrRetryWide);
+      bad = true;
+    }
+    if (bad) continue;  // flag set; results of this row are discarded anyway
+    for (int o = 0; o < d.nOut; o++) {
+      bool nul = vm.isNull(d.outReg[o]);
+      d.outNotNull[o][row] = nul ? 0 : 1;
+      if (d.outType[o] == 2 /*GX_TYPE_DECIMAL*/) {
+        uint8_t* out = (uint8_t*)d.outData[o] + row * 40;
+        if (nul) {
+          for (int i = 0; i < 10; i++) ((uint32_t*)out)[i] = 0;
+        } else {
+          Int128 ai = VT<WIDE>::toAcc(vm.get(d.outReg[o]));
+          devDecEncode(((__int128)ai.hi << 64) | (__int128)ai.lo,
+                       d.outScale[o], out);
+        }
+      } else {
+        Int128 ai = VT<WIDE>::toAcc(vm.get(d.outReg[o]));
+        ((int64_t*)d.outData[o])[row] = nul ? 0 : (int64_t)ai.lo;
+      }
+    }
+  }
+}
+
+__global__ void packNullsKernel(const uint8_t* __restrict__ notNull,
+                                uint8_t* __restrict__ bitmap, int64_t n) {
+  int64_t nBytes = (n + 7) / 8;
+  for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < nBytes;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    int64_t base = b * 8;
+    int m = n - base < 8 ? (int)(n - base) : 8;
+    uint8_t v = 0;
+    for (int j = 0; j < m; j++)
+      if (notNull[base + j]) v |= (uint8_t)(1 << j);
+    bitmap[b] = v;
+  }
+}
+
 // gather a null bitmap through the match index: one thread composes one
 // output byte (8 rows) — no atomics (LSB-first, 1 = NOT NULL)
 __global__ void hjGatherNullsKernel(const uint8_t* __restrict__ in,
@@ -2144,6 +2423,25 @@ int gxSelectPhase(int phase, const HashJoinDesc* devDesc,
   else
     hipLaunchKernelGGL(selCompactKernel<true>, g, dim3(256), 0,
                        (hipStream_t)stream, devDesc);
+  return (int)hipGetLastError();
+}
+
+int gxProject(const ProjDesc* devDesc, const ProjDesc& h, void* stream) {
+  if (h.table.nRows == 0) return 0;
+  dim3 g(gridFor(h.table.nRows));
+  if (h.wide)
+    hipLaunchKernelGGL(projectKernel<true>, g, dim3(256), 0,
+                       (hipStream_t)stream, devDesc);
+  else
+    hipLaunchKernelGGL(projectKernel<false>, g, dim3(256), 0,
+                       (hipStream_t)stream, devDesc);
+  return (int)hipGetLastError();
+}
+
+int gxPackNulls(const uint8_t* notNullBytes, uint8_t* bitmap, int64_t n,
+                void* stream) {
+  hipLaunchKernelGGL(packNullsKernel, dim3(gridFor((n + 7) / 8)), dim3(256), 0,
+                     (hipStream_t)stream, notNullBytes, bitmap, n);
   return (int)hipGetLastError();
 }
 

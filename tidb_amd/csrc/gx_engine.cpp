@@ -663,8 +663,24 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
     }
     case EK_CALL: {
       if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT) {
-        ex->err = "casts unsupported in the join-probe VM this round";
-        return -1;
+        // cast family (ProduceDecWithSpecifiedTp / ToInt): round half-up
+        // from the arg's scale to the target scale (VM_ROUND_SCALE)
+        if (e.args.size() != 1) {
+          ex->err = "cast takes one argument";
+          return -1;
+        }
+        int sa = 0;
+        int ra = vmCompile(ex, B, e.args[0], &sa);
+        if (ra < 0) return -1;
+        int sr = e.func == GX_F_CAST_INT ? 0 : e.retFrac;
+        if (sr == sa) {
+          reg = ra;  // no-op cast
+          *scaleOut = sr;
+          break;
+        }
+        reg = emit(gxp::VM_ROUND_SCALE, allocReg(), ra, sr, sa);
+        *scaleOut = sr;
+        break;
       }
       if (e.args.size() != 2) {
         ex->err = "unsupported call arity on device";

@@ -96,3 +96,50 @@ def test_projection_cast_parity():
     b = run_proj(load_product(), 20000, with_cast=True)
     assert len(a) == len(b) == 20000
     assert a == b
+
+
+def _null_proj_run(lib):
+    """NULL propagation through computed expressions (NULL operand -> NULL
+    result; builtin_arithmetic_vec.go null merge) + NULL passthrough."""
+    import ctypes
+    from tidb_amd.chunkpy import PyChunk
+
+    def decb(s):
+        out = (ctypes.c_uint8 * 40)()
+        assert lib.gx_dec_from_string(s.encode(), len(s.encode()), out) == 0
+        return bytes(out)
+
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_DECIMAL, GX_TYPE_DECIMAL, GX_TYPE_I64], [2, 2, 0])
+    x = b.colref(0, GX_TYPE_DECIMAL, 2)
+    y = b.colref(1, GX_TYPE_DECIMAL, 2)
+    prod = b.call(GX_F_MUL, GX_TYPE_DECIMAL, 4, x, y)
+    proj = b.projection(src, [prod, b.colref(2, GX_TYPE_I64), x])
+    ex = b.build(proj)
+    ch = PyChunk([GX_TYPE_DECIMAL, GX_TYPE_DECIMAL, GX_TYPE_I64], 8, [2, 2, 0])
+    rows = [("1.50", "2.00", 7), (None, "3.00", None), ("-0.25", None, 1),
+            (None, None, 0), ("10.00", "0.10", -5)]
+    for r in rows:
+        ch.append_row([None if r[0] is None else decb(r[0]),
+                       None if r[1] is None else decb(r[1]), r[2]])
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    got = ex.pull_all([GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_DECIMAL],
+                      [4, 0, 2])
+    ex.close()
+    ex.free()
+    b.free()
+    return got
+
+
+def test_oracle_projection_nulls():
+    lib = load_oracle()
+    got = _null_proj_run(lib)
+    assert got == [("3.0000", 7, "1.50"), (None, None, None),
+                   (None, 1, "-0.25"), (None, 0, None),
+                   ("1.0000", -5, "10.00")]
+
+
+@pytest.mark.gpu
+def test_projection_nulls_parity():
+    assert _null_proj_run(load_product()) == _null_proj_run(load_oracle())

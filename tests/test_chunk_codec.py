@@ -48,8 +48,8 @@ def encode(lib, chunk):
     return bytes(buf)
 
 
-def decode(lib, data, n_rows=16):
-    chunk = PyChunk(TYPES, n_rows, FRACS, data_caps=[None, None, 256, None])
+def decode(lib, data, n_rows=16, str_cap=256):
+    chunk = PyChunk(TYPES, n_rows, FRACS, data_caps=[None, None, str_cap, None])
     g = chunk.as_gx()
     buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
     n = lib.gx_chunk_decode(buf, len(data), ctypes.byref(g))
@@ -123,5 +123,5 @@ def test_fuzz_roundtrip_cross_library():
         enc_p = encode(p, chunk)
         assert enc_o == enc_p
         want = chunk.rows(n)
-        assert decode(p, enc_o, 64).rows(n) == want
-        assert decode(o, enc_p, 64).rows(n) == want
+        assert decode(p, enc_o, 64, 8192).rows(n) == want
+        assert decode(o, enc_p, 64, 8192).rows(n) == want

@@ -775,3 +775,39 @@ def test_q3_class_agg_over_nested_join_parity(libs):
     want = run(oracle)
     assert len(got) == len(want) > 50
     assert got == want
+
+
+def test_oracle_full_agg_family_over_join():
+    """Independent (non-oracle) pin of avg/min/max/firstrow/sum/count over
+    joined rows — guards the oracle itself (a firstrow aux-flag bug was
+    caught by exactly this kind of check)."""
+    from fractions import Fraction
+    lib = load_oracle()
+    build, probe = _agg_over_join_data()
+    got = _full_agg_over_join(lib)
+    groups = {}
+    for pk, pv in probe:
+        for bk, bg in build:
+            if bk != pk:
+                continue
+            g = groups.setdefault(bg, {"sum": Fraction(0), "cnt": 0,
+                                       "min": None, "max": None})
+            g["sum"] += Fraction(pv)
+            g["cnt"] += 1
+            g["min"] = pk if g["min"] is None else min(g["min"], pk)
+            g["max"] = pk if g["max"] is None else max(g["max"], pk)
+    assert len(got) == len(groups) > 10
+    for r in got:
+        g = groups[r[0]]
+        assert Fraction(r[1]) == g["sum"]
+        # avg = Round(sum / count, 6, HalfUp) — check via scaled integers
+        num, den = (g["sum"] / g["cnt"]).numerator, (g["sum"] / g["cnt"]).denominator
+        scaled2 = abs(num) * 10**6 * 2
+        q, rem = divmod(scaled2, den * 2)
+        avg = q + (1 if 2 * rem >= den * 2 else 0)
+        if num < 0:
+            avg = -avg
+        assert round(Fraction(r[2]) * 10**6) == avg
+        assert r[3] == g["min"] and r[4] == g["max"]
+        assert r[5] == g["cnt"]
+        assert r[6] == r[0]  # firstrow(group col) == the group value

@@ -811,3 +811,25 @@ def test_oracle_full_agg_family_over_join():
         assert r[3] == g["min"] and r[4] == g["max"]
         assert r[5] == g["cnt"]
         assert r[6] == r[0]  # firstrow(group col) == the group value
+
+
+def test_oracle_join_multi_chunk_sources():
+    """Sources bound as MULTIPLE chunks (the Next-pulls-many-chunks shape):
+    same result as a single concatenated chunk."""
+    lib = load_oracle()
+    b1 = BUILD_ROWS[:3]
+    b2 = BUILD_ROWS[3:]
+    p1 = PROBE_ROWS[:4]
+    p2 = PROBE_ROWS[4:]
+    b, bsrc, psrc, j = _join_plan(lib, False)
+    ex = b.build(j)
+    ex.bind_chunks(bsrc, [_to_chunk(lib, BUILD_TYPES, BUILD_FRACS, b1),
+                          _to_chunk(lib, BUILD_TYPES, BUILD_FRACS, b2)])
+    ex.bind_chunks(psrc, [_to_chunk(lib, PROBE_TYPES, PROBE_FRACS, p1),
+                          _to_chunk(lib, PROBE_TYPES, PROBE_FRACS, p2)])
+    ex.open()
+    rows = ex.pull_all(OUT_TYPES, OUT_FRACS)
+    ex.close()
+    ex.free()
+    b.free()
+    assert _canon(rows) == expected_join()

@@ -156,13 +156,22 @@ int32_t evalArith(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
   if (e.retType == GX_TYPE_DECIMAL) {
     for (int i = 0; i < n; i++) {
       if (!bothNotNull(a, b, i)) { out.appendNull(); continue; }
+      // int children coerce to decimal (the reference's planner inserts the
+      // cast; expression.WrapWithCastAsDecimal semantics)
+      MyDecimal ca, cb;
+      const MyDecimal* pa;
+      const MyDecimal* pb;
+      if (a.type == GX_TYPE_I64) { ca.FromInt(a.getI64(i)); pa = &ca; }
+      else pa = a.getDecimal(i);
+      if (b.type == GX_TYPE_I64) { cb.FromInt(b.getI64(i)); pb = &cb; }
+      else pb = b.getDecimal(i);
       MyDecimal to;
       int32_t ec = GX_OK;
       switch (e.func) {
-        case GX_F_PLUS: ec = DecimalAdd(a.getDecimal(i), b.getDecimal(i), &to); break;
-        case GX_F_MINUS: ec = DecimalSub(a.getDecimal(i), b.getDecimal(i), &to); break;
-        case GX_F_MUL: ec = DecimalMul(a.getDecimal(i), b.getDecimal(i), &to); break;
-        case GX_F_DIV: ec = DecimalDiv(a.getDecimal(i), b.getDecimal(i), &to, kDivFracIncr); break;
+        case GX_F_PLUS: ec = DecimalAdd(pa, pb, &to); break;
+        case GX_F_MINUS: ec = DecimalSub(pa, pb, &to); break;
+        case GX_F_MUL: ec = DecimalMul(pa, pb, &to); break;
+        case GX_F_DIV: ec = DecimalDiv(pa, pb, &to, kDivFracIncr); break;
       }
       if (ec == E_DIV_ZERO) { out.appendNull(); continue; }  // div-by-0 -> NULL (MySQL)
       if (ec != GX_OK && ec != E_TRUNCATED) {

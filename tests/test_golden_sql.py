@@ -66,3 +66,37 @@ def test_avg_of_int_division():
     got = _run(lib, rows, plan, [GX_TYPE_DECIMAL], [8],
                [GX_TYPE_I64, GX_TYPE_I64], [0, 0])
     assert got == [("25769363061037.62077260",)]
+
+
+def test_count_nullable_column_group_by():
+    """aggregate.result:11-14 — count(c) per group over (1,NULL),(2,1):
+    NULL rows don't count."""
+    from tests.gxlib import GX_AGG_COUNT
+    lib = load_oracle()
+
+    def plan(b, src):
+        gid = b.colref(0, GX_TYPE_I64)
+        c = b.colref(1, GX_TYPE_I64)
+        return b.hashagg(src, [gid], [(GX_AGG_COUNT, c, 0)])
+
+    got = sorted(_run(lib, [(1, None), (2, 1)], plan,
+                      [GX_TYPE_I64, GX_TYPE_I64], [0, 0],
+                      [GX_TYPE_I64, GX_TYPE_I64], [0, 0]))
+    assert got == [(1, 0), (2, 1)]
+
+
+def test_minmax_string_null_group():
+    """aggregate.result:46-51 — MIN(b), MAX(b) over a varchar with one group
+    all-NULL: that group's extremes are NULL."""
+    from tests.gxlib import GX_AGG_MAX, GX_AGG_MIN, GX_TYPE_STRING
+    lib = load_oracle()
+
+    def plan(b, src):
+        a = b.colref(0, GX_TYPE_I64)
+        s = b.colref(1, GX_TYPE_STRING)
+        return b.hashagg(src, [a], [(GX_AGG_MIN, s, 0), (GX_AGG_MAX, s, 0)])
+
+    got = sorted(_run(lib, [(1, "11"), (3, None)], plan,
+                      [GX_TYPE_I64, GX_TYPE_STRING, GX_TYPE_STRING],
+                      [0, 0, 0], [GX_TYPE_I64, GX_TYPE_STRING], [0, 0]))
+    assert got == [(1, "11", "11"), (3, None, None)]

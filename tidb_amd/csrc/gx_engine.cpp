@@ -152,6 +152,12 @@ struct gx_exec {
   // device state
   bool deviceReady = false;
   std::vector<void*> devBufs;
+  // persistent singletons (descriptors, counters, result tables) whose
+  // pointers are cached across runs — never freed by beginRun/freeSince
+  std::vector<void*> persistBufs;
+  // index into devBufs where the current run's allocations begin (SIZE_MAX
+  // until the first run); see beginRun()
+  size_t runMark = SIZE_MAX;
   gxp::GroupSlot* devTable = nullptr;
   gxp::FusedQueryDesc* devDesc = nullptr;
   uint32_t* devErr = nullptr;
@@ -231,6 +237,7 @@ struct gx_exec {
 
   ~gx_exec() {
     for (void* p : devBufs) hipFree(p);
+    for (void* p : persistBufs) hipFree(p);
   }
 };
 
@@ -342,6 +349,42 @@ static void* devAlloc(gx_exec* ex, size_t n) {
   if (hipMalloc(&p, n + 16) != hipSuccess) return nullptr;
   ex->devBufs.push_back(p);
   return p;
+}
+
+static void freeSince(gx_exec* ex, size_t mark) {
+  for (size_t i = mark; i < ex->devBufs.size(); i++)
+    (void)hipFree(ex->devBufs[i]);
+  ex->devBufs.resize(mark);
+}
+
+// persistent allocation: survives beginRun/freeSince (pointer is cached in a
+// gx_exec field across runs); freed only at destroy
+static void* devAllocP(gx_exec* ex, size_t n) {
+  void* p = nullptr;
+  if (hipMalloc(&p, n + 16) != hipSuccess) return nullptr;
+  ex->persistBufs.push_back(p);
+  return p;
+}
+
+static void devFreeP(gx_exec* ex, void* p) {
+  for (size_t i = 0; i < ex->persistBufs.size(); i++)
+    if (ex->persistBufs[i] == p) {
+      (void)hipFree(p);
+      ex->persistBufs.erase(ex->persistBufs.begin() + i);
+      return;
+    }
+}
+
+// per-run allocation scope: the first run records where its allocations
+// begin; every re-run (gx_open resets ranQuery) frees them first so repeated
+// open/next cycles do not grow device memory. Persistent singletons
+// (devErr/devHj/stream/counters) and cached source materializations are
+// allocated BEFORE beginRun is called.
+static void beginRun(gx_exec* ex) {
+  if (ex->runMark == SIZE_MAX)
+    ex->runMark = ex->devBufs.size();
+  else
+    freeSince(ex, ex->runMark);
 }
 
 // ---------------- plan compilation ----------------
@@ -1402,12 +1445,25 @@ static int32_t compileProject(gx_exec* ex) {
     pd.insP10[i] = 1;
     pd.insMagic[i] = 0;
     if (pd.ins[i].op == gxp::VM_LOAD_DEC) {
+      if (pd.ins[i].b < 0 || pd.ins[i].b > 9) {
+        ex->err = "decimal scale out of range for projection (frac must be "
+                  "0..9, got " + std::to_string(pd.ins[i].b) + ")";
+        return GX_ERR_INVALID;
+      }
       pd.insP10[i] = p10h[pd.ins[i].b];
       pd.insMagic[i] = magich[9 - pd.ins[i].b];
     } else if (pd.ins[i].op == gxp::VM_SCALE_UP) {
+      if (pd.ins[i].b < 0 || pd.ins[i].b > 18) {
+        ex->err = "scale shift out of range for projection";
+        return GX_ERR_INVALID;
+      }
       pd.insP10[i] = p10h[pd.ins[i].b];
     } else if (pd.ins[i].op == gxp::VM_ROUND_SCALE) {
       int sh = pd.ins[i].b - pd.ins[i].c;
+      if (sh < -18 || sh > 18) {
+        ex->err = "rescale shift out of range for projection";
+        return GX_ERR_INVALID;
+      }
       pd.insP10[i] = p10h[sh >= 0 ? sh : -sh];
     }
   }
@@ -2126,11 +2182,11 @@ static int32_t finalizeFusedBind(gx_exec* ex) {
     }
   }
   // result/error buffers
-  ex->devTable = (gxp::GroupSlot*)devAlloc(
+  ex->devTable = (gxp::GroupSlot*)devAllocP(
       ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
-  ex->devErr = (uint32_t*)devAlloc(ex, 4);
-  ex->devSel = (uint64_t*)devAlloc(ex, 8);
-  ex->devDesc = (gxp::FusedQueryDesc*)devAlloc(ex, sizeof(gxp::FusedQueryDesc));
+  ex->devErr = (uint32_t*)devAllocP(ex, 4);
+  ex->devSel = (uint64_t*)devAllocP(ex, 8);
+  ex->devDesc = (gxp::FusedQueryDesc*)devAllocP(ex, sizeof(gxp::FusedQueryDesc));
   if (!ex->devTable || !ex->devErr || !ex->devSel || !ex->devDesc) {
     ex->err = "hipMalloc failed";
     return GX_ERR_INTERNAL;
@@ -2362,7 +2418,8 @@ static int32_t runFused(gx_exec* ex) {
     // NDV above the global table: rerun with an 8x table (capped 2^25 groups
     // ~ 10 GB of state; the init kernel resets it, so the rerun is clean)
     ex->desc.globalGroupsLog2 += 3;
-    ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAlloc(
+    devFreeP(ex, ex->devTable);  // the outgrown table is dead weight
+    ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAllocP(
         ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
     if (!ex->devTable) { ex->err = "hipMalloc failed (group table)"; return GX_ERR_INTERNAL; }
     if (getenv("GX_DEBUG"))
@@ -2541,9 +2598,9 @@ static int32_t runJoinAgg(gx_exec* ex) {
     if (rc) return rc;
     rc = materializeTable(ex, ex->jaSrcLi, &ja.probe);
     if (rc) return rc;
-    ex->devErr = (uint32_t*)devAlloc(ex, 4);
-    ja.counters = (uint64_t*)devAlloc(ex, 3 * 8);
-    ex->devJa = (gxp::JoinAggDesc*)devAlloc(ex, sizeof(gxp::JoinAggDesc));
+    ex->devErr = (uint32_t*)devAllocP(ex, 4);
+    ja.counters = (uint64_t*)devAllocP(ex, 3 * 8);
+    ex->devJa = (gxp::JoinAggDesc*)devAllocP(ex, sizeof(gxp::JoinAggDesc));
     if (!ex->devErr || !ja.counters || !ex->devJa) {
       ex->err = "hipMalloc failed";
       return GX_ERR_INTERNAL;
@@ -2919,17 +2976,6 @@ static int32_t gatherCols(gx_exec* ex, const gxp::DevTable& srcTab,
 // previous stages' outputs.
 static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
   gxp::HashJoinDesc& hj = st.hj;
-  if (!st.inputsReady) {
-    if (st.srcB >= 0) {
-      int32_t rc = materializeTable(ex, st.srcB, &st.buildTab);
-      if (rc) return rc;
-    }
-    if (st.srcP >= 0) {
-      int32_t rc = materializeTable(ex, st.srcP, &st.probeTab);
-      if (rc) return rc;
-    }
-    st.inputsReady = true;
-  }
   hj.build = st.srcB >= 0 ? st.buildTab : ex->joinStages[st.buildStage].out;
   hj.probe = st.srcP >= 0 ? st.probeTab : ex->joinStages[st.probeStage].out;
   int64_t nb = hj.build.nRows;
@@ -3068,8 +3114,8 @@ static int32_t runHashJoin(gx_exec* ex) {
     }
     if (ex->device >= 0) hipSetDevice(ex->device);
     HIP_OK(ex, hipStreamCreate(&ex->stream));
-    ex->devErr = (uint32_t*)devAlloc(ex, 4);
-    ex->devHj = (gxp::HashJoinDesc*)devAlloc(ex, sizeof(gxp::HashJoinDesc));
+    ex->devErr = (uint32_t*)devAllocP(ex, 4);
+    ex->devHj = (gxp::HashJoinDesc*)devAllocP(ex, sizeof(gxp::HashJoinDesc));
     if (!ex->devErr || !ex->devHj) {
       ex->err = "hipMalloc failed";
       return GX_ERR_INTERNAL;
@@ -3077,12 +3123,29 @@ static int32_t runHashJoin(gx_exec* ex) {
     ex->deviceReady = true;
   }
   ex->lastKernelMs = 0;
+  // persistent per-stage state (counters + cached source materializations)
+  // is allocated before the per-run scope opens
   for (size_t s = 0; s < ex->joinStages.size(); s++) {
     gx_exec::JoinStage& st = ex->joinStages[s];
     if (!st.hj.counters) {
-      st.hj.counters = (uint64_t*)devAlloc(ex, 3 * 8);
+      st.hj.counters = (uint64_t*)devAllocP(ex, 3 * 8);
       if (!st.hj.counters) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
     }
+    if (!st.inputsReady) {
+      if (st.srcB >= 0) {
+        int32_t rc = materializeTable(ex, st.srcB, &st.buildTab);
+        if (rc) return rc;
+      }
+      if (st.srcP >= 0) {
+        int32_t rc = materializeTable(ex, st.srcP, &st.probeTab);
+        if (rc) return rc;
+      }
+      st.inputsReady = true;
+    }
+  }
+  beginRun(ex);
+  for (size_t s = 0; s < ex->joinStages.size(); s++) {
+    gx_exec::JoinStage& st = ex->joinStages[s];
     st.hj.errorFlag = ex->devErr;
     int32_t rc = runJoinStage(ex, st, s + 1 == ex->joinStages.size());
     if (rc) return rc;
@@ -3111,9 +3174,9 @@ static int32_t runSelect(gx_exec* ex) {
     }
     if (ex->device >= 0) hipSetDevice(ex->device);
     HIP_OK(ex, hipStreamCreate(&ex->stream));
-    ex->devErr = (uint32_t*)devAlloc(ex, 4);
-    ex->devHj = (gxp::HashJoinDesc*)devAlloc(ex, sizeof(gxp::HashJoinDesc));
-    hj.counters = (uint64_t*)devAlloc(ex, 3 * 8);
+    ex->devErr = (uint32_t*)devAllocP(ex, 4);
+    ex->devHj = (gxp::HashJoinDesc*)devAllocP(ex, sizeof(gxp::HashJoinDesc));
+    hj.counters = (uint64_t*)devAllocP(ex, 3 * 8);
     if (!ex->devErr || !ex->devHj || !hj.counters) {
       ex->err = "hipMalloc failed";
       return GX_ERR_INTERNAL;
@@ -3123,6 +3186,7 @@ static int32_t runSelect(gx_exec* ex) {
     if (rc) return rc;
     ex->deviceReady = true;
   }
+  beginRun(ex);
   hj.probe = st.probeTab;
   if (hj.probe.nRows > 0xFFFFFFFFLL) {
     ex->err = "selection > 2^32 rows unsupported this round";
@@ -3209,16 +3273,17 @@ static int32_t runProject(gx_exec* ex) {
       }
       if (ex->device >= 0) hipSetDevice(ex->device);
       HIP_OK(ex, hipStreamCreate(&ex->stream));
-      ex->devErr = (uint32_t*)devAlloc(ex, 4);
+      ex->devErr = (uint32_t*)devAllocP(ex, 4);
       if (!ex->devErr) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
       int32_t rc = materializeTable(ex, ex->projSrcNode, &ex->selStage.probeTab);
       if (rc) return rc;
       ex->deviceReady = true;
     }
+    beginRun(ex);  // projOverSelect re-runs open the scope inside runSelect
     pd.table = ex->selStage.probeTab;
   }
   if (!ex->devPd) {
-    ex->devPd = (gxp::ProjDesc*)devAlloc(ex, sizeof(gxp::ProjDesc));
+    ex->devPd = (gxp::ProjDesc*)devAllocP(ex, sizeof(gxp::ProjDesc));
     if (!ex->devPd) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
   }
   pd.errorFlag = ex->devErr;
@@ -3516,7 +3581,7 @@ static int32_t runDeviceSort(gx_exec* ex) {
     }
   }
   if (n == 0) { ex->devSorted = true; return GX_OK; }
-  if (!ex->devErr) ex->devErr = (uint32_t*)devAlloc(ex, 4);
+  if (!ex->devErr) ex->devErr = (uint32_t*)devAllocP(ex, 4);
   if (!ex->devErr) { ex->err = "hipMalloc failed (sort err)"; return GX_ERR_INTERNAL; }
   uint32_t* idxA = (uint32_t*)devAlloc(ex, n * 4);
   uint32_t* idxB = (uint32_t*)devAlloc(ex, n * 4);
@@ -3633,12 +3698,6 @@ static int32_t runDeviceSort(gx_exec* ex) {
 
 // ---------------- out-of-core sort (spill runs + k-way merge) ----------------
 
-static void freeSince(gx_exec* ex, size_t mark) {
-  for (size_t i = mark; i < ex->devBufs.size(); i++)
-    (void)hipFree(ex->devBufs[i]);
-  ex->devBufs.resize(mark);
-}
-
 // sort a source larger than the HBM budget: materialize + device-radix-sort
 // row-range runs, download each sorted run (columns + composed order keys)
 // to host, free the device buffers, then k-way merge on emission.
@@ -3651,7 +3710,7 @@ static int32_t runDeviceSortSpill(gx_exec* ex, int64_t runRows) {
   if (ex->device >= 0) hipSetDevice(ex->device);
   if (!ex->stream) HIP_OK(ex, hipStreamCreate(&ex->stream));
   if (!ex->devErr) {
-    ex->devErr = (uint32_t*)devAlloc(ex, 4);
+    ex->devErr = (uint32_t*)devAllocP(ex, 4);
     if (!ex->devErr) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
   }
   Binding& b = ex->bindings[ex->sourceNode];
@@ -3673,12 +3732,19 @@ static int32_t runDeviceSortSpill(gx_exec* ex, int64_t runRows) {
     gxp::DevTable runTab{};
     rc = materializeTable(ex, ex->sourceNode, &runTab);
     if (rc == GX_OK) {
-      for (int c = 0; c < runTab.nCols && rc == GX_OK; c++)
+      for (int c = 0; c < runTab.nCols && rc == GX_OK; c++) {
         if (runTab.cols[c].type == GX_TYPE_STRING &&
             !runTab.cols[c].denseOffsets) {
           ex->err = "general varlen column in spill sort unsupported this round";
           rc = GX_ERR_INVALID;
         }
+        // emitSpillChunk emits all-NOT-NULL bitmaps; a nullable source would
+        // silently lose its NULL flags — reject instead
+        if (runTab.cols[c].hasNulls && runTab.cols[c].nullBitmap) {
+          ex->err = "nullable column in spill sort unsupported this round";
+          rc = GX_ERR_INVALID;
+        }
+      }
     }
     if (rc == GX_OK) {
       ex->desc.table = runTab;
@@ -3891,10 +3957,25 @@ static int32_t emitTableChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
                            bytes, hipMemcpyDeviceToHost));
     }
     if (g->null_bitmap) {
-      // srcPos advances in 1024-row steps, so the bitmap slice is byte-aligned
       if (col.hasNulls && col.nullBitmap) {
-        HIP_OK(ex, hipMemcpy(g->null_bitmap, col.nullBitmap + ex->srcPos / 8,
-                             (n + 7) / 8, hipMemcpyDeviceToHost));
+        // srcPos advances in 1024-row steps AFTER the first chunk, but the
+        // FIRST emission may start at a sort offset that is not a multiple
+        // of 8 (runDeviceSort sets srcPos = devSortOffset): repack the
+        // bitmap with the bit offset so NULL flags land on the right rows
+        int bit = (int)(ex->srcPos & 7);
+        if (bit == 0) {
+          HIP_OK(ex, hipMemcpy(g->null_bitmap, col.nullBitmap + ex->srcPos / 8,
+                               (n + 7) / 8, hipMemcpyDeviceToHost));
+        } else {
+          int nb = (n + bit + 7) / 8;  // source bytes covering [srcPos, srcPos+n)
+          std::vector<uint8_t> tmp(nb + 1, 0);
+          HIP_OK(ex, hipMemcpy(tmp.data(), col.nullBitmap + ex->srcPos / 8, nb,
+                               hipMemcpyDeviceToHost));
+          int outBytes = (n + 7) / 8;
+          for (int i = 0; i < outBytes; i++)
+            g->null_bitmap[i] =
+                (uint8_t)((tmp[i] >> bit) | (tmp[i + 1] << (8 - bit)));
+        }
       } else {
         std::memset(g->null_bitmap, 0xFF, (n + 7) / 8);
       }
@@ -4505,6 +4586,9 @@ int32_t gx_open(gx_exec* ex) {
   ex->emitPos = 0;
   ex->srcPos = 0;
   ex->resultRows.clear();
+  // join/select/project re-runs rebuild desc.table from scratch (beginRun
+  // frees the previous run's buffers), so a sort above them must re-run too
+  if (ex->isHashJoin || ex->isSelect || ex->isProject) ex->devSorted = false;
   if (ex->spillSorted) {  // re-open: replay the merged runs from the top
     ex->runPos.assign(ex->sortRuns.size(), 0);
     ex->spillSkip = ex->devSortOffset;

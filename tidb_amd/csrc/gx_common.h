@@ -130,17 +130,39 @@ constexpr int kMaxAggs = 12;
 // and the kernel bumps only cnt[0].
 
 // ---- group keys ----
-// Round-1 device grouping: group-by columns pack into ONE u64 key
-// (string cols <= 3 bytes each as len<<24|bytes in a 32-bit lane; Q1 uses two
-// char(1) columns). Wider keys (Q3's high-NDV int keys) arrive with the Q3
-// path. kEmptyKey is reserved.
+// Two device key paths:
+//  PACKED (round 1, kept for the Q1 shape): <= 2 dense char(1) / short
+//  string / small-i64 columns pack into ONE u64 key.
+//  WIDE (serialized keys, codec.HashGroupKey semantics codec.go:1791-1879):
+//  arbitrary column sets (strings any length, decimals, time, i64, up to
+//  kMaxGroupKeyCols). Per row a 64-bit hash over the canonical column
+//  values probes the global table; the slot key packs (hash32 << 32 |
+//  recordIdx) and equality is VERIFIED against a fixed-stride key record
+//  (exact — hash collisions probe on). Records: [0] u64 null bits,
+//  [8] u64 owner row (the row that claimed the group; string bytes beyond
+//  the 16 B inline prefix compare/decode through it), [16 + 24*k] per-col
+//  {u64 a, u64 b, u64 c}: i64/time value | decimal units lo/hi | string
+//  trimmed len + 16 B prefix. utf8mb4_bin PAD SPACE: trailing spaces are
+//  trimmed before hashing/comparing (collate.go:272), and the emitted group
+//  value is the trimmed form.
+constexpr int kMaxGroupKeyCols = 6;
 struct GroupKeyDesc {
   int32_t nCols;
-  int32_t col[2];
-  int32_t kind[2];  // 0 = short string, 1 = small i64 (<2^31), 2 = dense char(1)
-  int32_t slot[2];  // raw fetch slot (string: the offsets pair; i64: value; dense: -1)
+  int32_t col[kMaxGroupKeyCols];
+  int32_t kind[kMaxGroupKeyCols];  // packed: 0 = short string, 1 = small i64
+                                   // (<2^31), 2 = dense char(1);
+                                   // wide: 3 = raw 8B (i64/time, fetch slot),
+                                   // 4 = decimal (VM register), 5 = varlen
+                                   // string (offsets fetch slot)
+  int32_t slot[kMaxGroupKeyCols];  // fetch slot; kind 4: VM register
+  int32_t kscale[kMaxGroupKeyCols] = {0};  // kind 4: units scale (decode)
   int32_t rawSlot[2] = {-1, -1};  // plain-kernel prefetch slot for dense chars:
                                   // char k lives at byte k of that slot's v.y
+  int32_t wideMode = 0;       // 1 = serialized wide-key path
+  int32_t recBytes = 0;       // record stride = 16 + 24 * nCols
+  int64_t recCap = 0;         // record capacity (grown with the table)
+  uint8_t* keyStore = nullptr;    // recCap * recBytes
+  uint64_t* recCursor = nullptr;  // device allocation cursor (1 u64)
 };
 
 constexpr uint64_t kEmptyKey = ~0ULL;

@@ -415,6 +415,192 @@ __device__ __attribute__((always_inline)) inline bool makeGroupKey(const FusedQu
   return true;
 }
 
+// ---- serialized wide group keys (GroupKeyDesc wideMode) ----
+// Exact arbitrary-column grouping: a 64-bit value hash probes the global
+// table, equality is verified against the group's key record (hash collisions
+// linear-probe on). The record is written COMPLETELY and device-fenced
+// BEFORE its (hash32|recordIdx) publishes via CAS, so no reader ever spins
+// on a half-written record (wave64 lockstep makes intra-wave spinning a
+// deadlock; this protocol never waits).
+
+// trimmed varlen string metadata: byte start, PAD-SPACE-trimmed length and
+// the 16-byte inline prefix as two LE words (utf8mb4_bin, collate.go:272)
+__device__ inline void wideStrMeta(const DevCol& c, ulonglong2 off,
+                                   int64_t* sOut, uint64_t* lenOut,
+                                   uint64_t* w0, uint64_t* w1) {
+  int64_t s = (int64_t)off.x, e = (int64_t)off.y;
+  auto p = gptr<uint8_t>(c.data);
+  while (e > s && p[e - 1] == ' ') e--;
+  uint64_t len = (uint64_t)(e - s);
+  uint64_t a = 0, b = 0;
+  int64_t m = (int64_t)(len < 16 ? len : 16);
+  for (int64_t t = 0; t < m; t++) {
+    uint64_t byte = p[s + t];
+    if (t < 8) a |= byte << (8 * t);
+    else b |= byte << (8 * (t - 8));
+  }
+  *sOut = s;
+  *lenOut = len;
+  *w0 = a;
+  *w1 = b;
+}
+
+template <bool WIDE, typename VMT, typename RAWT>
+__device__ inline uint64_t wideKeyHash(const FusedQueryDesc& d, int64_t row,
+                                       const RAWT& raw, const VMT& vm,
+                                       uint64_t* nullBitsOut) {
+  uint64_t h = 0x243F6A8885A308D3ULL;
+  uint64_t nb = 0;
+  for (int k = 0; k < d.gkey.nCols; k++) {
+    const DevCol& c = d.table.cols[d.gkey.col[k]];
+    int kind = d.gkey.kind[k];
+    bool nul = kind == 4 ? vm.isNull(d.gkey.slot[k]) : colIsNull(c, row);
+    if (nul) {
+      nb |= 1ULL << k;
+      h = splitmix64(h ^ 0xA5A5F00DFEEDBEEFULL);
+      continue;
+    }
+    if (kind == 3) {
+      h = splitmix64(h ^ raw.get(d.gkey.slot[k]).x);
+    } else if (kind == 4) {
+      Int128 u = VT<WIDE>::toAcc(vm.get(d.gkey.slot[k]));
+      h = splitmix64(h ^ u.lo);
+      h = splitmix64(h ^ (uint64_t)u.hi);
+    } else {  // kind 5: varlen string
+      ulonglong2 off = raw.get(d.gkey.slot[k]);
+      int64_t s = (int64_t)off.x, e = (int64_t)off.y;
+      auto p = gptr<uint8_t>(c.data);
+      while (e > s && p[e - 1] == ' ') e--;
+      for (int64_t j = s; j < e; j += 8) {
+        uint64_t w = 0;
+        int64_t m = e - j < 8 ? e - j : 8;
+        for (int64_t t = 0; t < m; t++) w |= (uint64_t)p[j + t] << (8 * t);
+        h = splitmix64(h ^ w);
+      }
+      h = splitmix64(h ^ ((uint64_t)(e - s) * 0x9E3779B97F4A7C15ULL + 1));
+    }
+  }
+  *nullBitsOut = nb;
+  return h;
+}
+
+template <bool WIDE, typename VMT, typename RAWT>
+__device__ inline void wideKeyWrite(const FusedQueryDesc& d, uint8_t* rec,
+                                    int64_t row, const RAWT& raw,
+                                    const VMT& vm, uint64_t nullBits) {
+  uint64_t* r64 = (uint64_t*)rec;
+  r64[0] = nullBits;
+  r64[1] = (uint64_t)row;
+  for (int k = 0; k < d.gkey.nCols; k++) {
+    uint64_t* f = (uint64_t*)(rec + 16 + 24 * k);
+    f[0] = f[1] = f[2] = 0;  // deterministic record bytes
+    if ((nullBits >> k) & 1) continue;
+    int kind = d.gkey.kind[k];
+    if (kind == 3) {
+      f[0] = raw.get(d.gkey.slot[k]).x;
+    } else if (kind == 4) {
+      Int128 u = VT<WIDE>::toAcc(vm.get(d.gkey.slot[k]));
+      f[0] = u.lo;
+      f[1] = (uint64_t)u.hi;
+    } else {
+      const DevCol& c = d.table.cols[d.gkey.col[k]];
+      int64_t s;
+      wideStrMeta(c, raw.get(d.gkey.slot[k]), &s, &f[0], &f[1], &f[2]);
+    }
+  }
+}
+
+template <bool WIDE, typename VMT, typename RAWT>
+__device__ inline bool wideKeyMatches(const FusedQueryDesc& d,
+                                      const uint8_t* rec, int64_t row,
+                                      const RAWT& raw, const VMT& vm,
+                                      uint64_t nullBits) {
+  const uint64_t* r64 = (const uint64_t*)rec;
+  if (r64[0] != nullBits) return false;
+  for (int k = 0; k < d.gkey.nCols; k++) {
+    if ((nullBits >> k) & 1) continue;
+    const uint64_t* f = (const uint64_t*)(rec + 16 + 24 * k);
+    int kind = d.gkey.kind[k];
+    if (kind == 3) {
+      if (f[0] != raw.get(d.gkey.slot[k]).x) return false;
+    } else if (kind == 4) {
+      Int128 u = VT<WIDE>::toAcc(vm.get(d.gkey.slot[k]));
+      if (f[0] != u.lo || f[1] != (uint64_t)u.hi) return false;
+    } else {
+      const DevCol& c = d.table.cols[d.gkey.col[k]];
+      int64_t s;
+      uint64_t len, w0, w1;
+      wideStrMeta(c, raw.get(d.gkey.slot[k]), &s, &len, &w0, &w1);
+      if (f[0] != len || f[1] != w0 || f[2] != w1) return false;
+      if (len > 16) {  // tail compares against the owner row's bytes
+        int64_t owner = (int64_t)r64[1];
+        int64_t os = gptr<int64_t>(c.offsets)[owner];
+        auto p = gptr<uint8_t>(c.data);
+        for (uint64_t j = 16; j < len; j++)
+          if (p[s + (int64_t)j] != p[os + (int64_t)j]) return false;
+      }
+    }
+  }
+  return true;
+}
+
+// probe/insert: returns the global table slot owning this row's group, or
+// false with the error flag set (table or record store full -> the engine
+// grows 8x and reruns, the agg_spill.go partition-growth analog)
+template <bool WIDE, typename VMT, typename RAWT>
+__device__ inline bool makeWideGroupKey(const FusedQueryDesc& d, int64_t row,
+                                        const RAWT& raw, const VMT& vm,
+                                        uint32_t* slotOut) {
+  uint64_t nullBits;
+  uint64_t h = wideKeyHash<WIDE>(d, row, raw, vm, &nullBits);
+  uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
+  uint32_t slot = (uint32_t)h & gmask;
+  uint32_t h32 = (uint32_t)(h >> 32);
+  uint32_t myRec = 0xFFFFFFFFu;
+  for (uint32_t probe = 0;; probe++) {
+    if (probe > gmask) {
+      atomicOr(d.errorFlag, kErrGlobalFull);
+      return false;
+    }
+    uint64_t cur = __hip_atomic_load(
+        (unsigned long long*)&d.globalTable[slot].key, __ATOMIC_ACQUIRE,
+        __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == kEmptyKey) {
+      if (myRec == 0xFFFFFFFFu) {
+        uint64_t idx = atomicAdd((unsigned long long*)d.gkey.recCursor, 1ULL);
+        if ((int64_t)idx >= d.gkey.recCap) {
+          atomicOr(d.errorFlag, kErrGlobalFull);
+          return false;
+        }
+        myRec = (uint32_t)idx;
+        wideKeyWrite<WIDE>(d, d.gkey.keyStore + idx * (uint64_t)d.gkey.recBytes,
+                           row, raw, vm, nullBits);
+        __threadfence();  // record visible before the key publishes
+      }
+      uint64_t want = ((uint64_t)h32 << 32) | myRec;
+      unsigned long long expect = (unsigned long long)kEmptyKey;
+      __hip_atomic_compare_exchange_strong(
+          (unsigned long long*)&d.globalTable[slot].key, &expect,
+          (unsigned long long)want, __ATOMIC_ACQ_REL, __ATOMIC_ACQUIRE,
+          __HIP_MEMORY_SCOPE_AGENT);
+      if (expect == (unsigned long long)kEmptyKey) {  // we won
+        *slotOut = slot;
+        return true;
+      }
+      cur = (uint64_t)expect;  // lost: verify against the winner
+    }
+    if ((uint32_t)(cur >> 32) == h32) {
+      const uint8_t* rec =
+          d.gkey.keyStore + (cur & 0xFFFFFFFFu) * (uint64_t)d.gkey.recBytes;
+      if (wideKeyMatches<WIDE>(d, rec, row, raw, vm, nullBits)) {
+        *slotOut = slot;
+        return true;
+      }
+    }
+    slot = (slot + 1) & gmask;
+  }
+}
+
 // atomic int128 + count accumulation into a slot (LDS or global)
 template <typename SlotT>
 __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {

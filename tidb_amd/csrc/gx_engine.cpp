@@ -1683,9 +1683,17 @@ static int32_t compileFused(gx_exec* ex) {
     }
   }
 
-  // group keys
+  // group keys — two device paths (gx_common.h GroupKeyDesc): the packed u64
+  // key for <= 2 dense-char/short-string/small-i64 columns (the Q1 shape,
+  // JIT-supported), and the serialized wide-key path (codec.HashGroupKey
+  // semantics, util/codec/codec.go:1791-1879) for arbitrary column sets —
+  // strings of any length, decimals, time, up to kMaxGroupKeyCols columns.
+  // Decimal/time keys and >2 columns force wide at compile; strings decide
+  // packed-vs-wide at bind (offset density known there); packed-lane overflow
+  // at run time (string > 3 B, i64 >= 2^31) converts to wide and reruns.
   gxp::GroupKeyDesc gk{};
   gk.nCols = 0;
+  bool forceWideKeys = (int)agg->exprs.size() > 2;
   for (int ge : agg->exprs) {
     const PExpr& e = plan.exprs[ge];
     int srcCol = -1;
@@ -1702,24 +1710,46 @@ static int32_t compileFused(gx_exec* ex) {
     } else {
       srcCol = e.colIdx;
     }
-    if (gk.nCols >= 2) {
-      ex->err = "device grouping supports <= 2 key columns this round";
+    if (gk.nCols >= gxp::kMaxGroupKeyCols) {
+      ex->err = "too many group-by key columns";
       return GX_ERR_INVALID;
     }
     int t = src->colTypes[srcCol];
     gk.col[gk.nCols] = srcCol;
-    gk.kind[gk.nCols] = t == GX_TYPE_STRING ? 0 : 1;
-    if (t != GX_TYPE_STRING && t != GX_TYPE_I64) {
-      ex->err = "device group key must be string or int64";
-      return GX_ERR_INVALID;
-    }
-    gk.slot[gk.nCols] = -1;  // string keys get a slot at open (density known)
-    if (t == GX_TYPE_I64) {
+    gk.kscale[gk.nCols] = 0;
+    if (t == GX_TYPE_STRING) {
+      gk.kind[gk.nCols] = 0;   // bind refines: dense -> 2, else packed 0/wide 5
+      gk.slot[gk.nCols] = -1;  // slot at open (density known)
+    } else if (t == GX_TYPE_I64) {
+      gk.kind[gk.nCols] = 1;   // bind flips to 3 (raw 8B) in wide mode
       gk.slot[gk.nCols] = fetchSlot(ex, gxp::FETCH_8B, srcCol);
       if (gk.slot[gk.nCols] < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+    } else if (t == GX_TYPE_TIME) {
+      forceWideKeys = true;
+      gk.kind[gk.nCols] = 3;
+      gk.slot[gk.nCols] = fetchSlot(ex, gxp::FETCH_8B, srcCol);
+      if (gk.slot[gk.nCols] < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+    } else if (t == GX_TYPE_DECIMAL) {
+      forceWideKeys = true;
+      int sc = 0;
+      // loaded through the VM (canonical units at the column's static scale;
+      // load registers are never recycled, so the register is stable)
+      int srcExpr = proj ? proj->exprs[e.colIdx] : ge;
+      int reg = compileExpr(ex, srcExpr, &sc);
+      if (reg < 0) {
+        if (ex->err.empty()) ex->err = "group key compile failed";
+        return GX_ERR_INVALID;
+      }
+      gk.kind[gk.nCols] = 4;
+      gk.slot[gk.nCols] = reg;
+      gk.kscale[gk.nCols] = sc;
+    } else {
+      ex->err = "unsupported group key column type";
+      return GX_ERR_INVALID;
     }
     gk.nCols++;
   }
+  gk.wideMode = forceWideKeys ? 1 : 0;
   ex->desc.gkey = gk;
 
   // aggs
@@ -2157,26 +2187,63 @@ static int32_t materializeDevice(gx_exec* ex) {
   return finalizeFusedBind(ex);
 }
 
+// key-record store for the serialized wide-key path (sized with the global
+// table; grown together with it on kErrGlobalFull)
+static int32_t ensureWideKeyStore(gx_exec* ex) {
+  gxp::GroupKeyDesc& g = ex->desc.gkey;
+  g.recBytes = 16 + 24 * g.nCols;
+  g.recCap = ((int64_t)1 << ex->desc.globalGroupsLog2) + 4096;
+  g.keyStore = (uint8_t*)devAllocP(ex, (size_t)g.recCap * g.recBytes);
+  if (!g.recCursor) g.recCursor = (uint64_t*)devAllocP(ex, 8);
+  if (!g.keyStore || !g.recCursor) {
+    ex->err = "hipMalloc failed (wide key store)";
+    return GX_ERR_INTERNAL;
+  }
+  return GX_OK;
+}
+
 // bind-time fixups + result buffers for the fused aggregation, over whatever
 // filled desc.table (a bound/generated source, or a materialized join
 // output). Idempotent (fetch slots must not be re-assigned on re-open).
 static int32_t finalizeFusedBind(gx_exec* ex) {
   if (ex->fusedBindDone) return GX_OK;
   gxp::DevTable& tab = ex->desc.table;
-  // finalize string group keys now that offset density is known
-  for (int k = 0; k < ex->desc.gkey.nCols; k++) {
-    if (ex->desc.gkey.kind[k] != 1) {
-      const gxp::DevCol& c = tab.cols[ex->desc.gkey.col[k]];
-      if (c.denseOffsets) {
-        ex->desc.gkey.kind[k] = 2;
-        ex->desc.gkey.slot[k] = fetchSlot(ex, gxp::FETCH_B1, ex->desc.gkey.col[k]);
-      } else {
-        ex->desc.gkey.kind[k] = 0;
-        ex->desc.gkey.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS,
-                                          ex->desc.gkey.col[k]);
-        if (ex->desc.gkey.slot[k] < 0) {
-          ex->err = "fetch plan full";
-          return GX_ERR_INVALID;
+  // finalize group keys now that offset density is known; non-dense string
+  // keys join the compile-forced reasons (decimal/time/3+ cols) in selecting
+  // the serialized wide-key path
+  {
+    gxp::GroupKeyDesc& g = ex->desc.gkey;
+    if (!g.wideMode) {
+      for (int k = 0; k < g.nCols; k++)
+        if (g.kind[k] == 0 && !tab.cols[g.col[k]].denseOffsets) g.wideMode = 1;
+    }
+    if (g.wideMode) {
+      for (int k = 0; k < g.nCols; k++) {
+        if (g.kind[k] == 0 || g.kind[k] == 2) {  // strings: offsets-pair fetch
+          g.kind[k] = 5;
+          g.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS, g.col[k]);
+          if (g.slot[k] < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+        } else if (g.kind[k] == 1) {
+          g.kind[k] = 3;  // raw 8B value; slot already assigned
+        }
+      }
+      int32_t rc = ensureWideKeyStore(ex);
+      if (rc) return rc;
+    } else {
+      for (int k = 0; k < g.nCols; k++) {
+        if (g.kind[k] != 1) {
+          const gxp::DevCol& c = tab.cols[g.col[k]];
+          if (c.denseOffsets) {
+            g.kind[k] = 2;
+            g.slot[k] = fetchSlot(ex, gxp::FETCH_B1, g.col[k]);
+          } else {
+            g.kind[k] = 0;
+            g.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS, g.col[k]);
+            if (g.slot[k] < 0) {
+              ex->err = "fetch plan full";
+              return GX_ERR_INVALID;
+            }
+          }
         }
       }
     }
@@ -2203,6 +2270,7 @@ static int32_t finalizeFusedBind(gx_exec* ex) {
     // plain grouped-fetch kernel on Q1/SF10 (5.48 vs 4.71 ms), so it ships
     // opt-in until the pipelining wins back the staging overhead.
     bool ok = tab.nRows >= 256 && getenv("GX_GLDS") && !ex->vmHasDiv &&
+              !d.gkey.wideMode &&   // wide keys use the plain kernel
               ex->vmNextReg <= 12;  // the staged kernel's VmState is 12-reg
     for (int s = 0; s < d.nAccSlots && ok; s++)
       ok = d.accKind[s] == 0;  // staged accumulate is sum-only
@@ -2356,7 +2424,7 @@ static int32_t runFused(gx_exec* ex) {
   if (!ex->jitTried && !ex->desc.useGlds && ex->desc.ablate == 0 &&
       !getenv("GX_NO_JIT")) {
     ex->jitTried = true;
-    bool eligible = true;
+    bool eligible = !ex->desc.gkey.wideMode;  // wide keys: interpreted kernel
     for (int k = 0; k < ex->desc.gkey.nCols; k++)
       eligible &= ex->desc.gkey.kind[k] != 0;
     if (eligible) {
@@ -2422,10 +2490,44 @@ static int32_t runFused(gx_exec* ex) {
     ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAllocP(
         ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
     if (!ex->devTable) { ex->err = "hipMalloc failed (group table)"; return GX_ERR_INTERNAL; }
+    if (ex->desc.gkey.wideMode) {  // key-record store grows with the table
+      devFreeP(ex, ex->desc.gkey.keyStore);
+      int32_t rc2 = ensureWideKeyStore(ex);
+      if (rc2) return rc2;
+    }
     if (getenv("GX_DEBUG"))
       fprintf(stderr, "[gx] global table full -> 2^%d retry\n",
               ex->desc.globalGroupsLog2);
     return runFused(ex);
+  }
+  if ((errFlag & 2u /*kErrBadKey*/) && !ex->desc.gkey.wideMode &&
+      ex->desc.gkey.nCols > 0) {
+    // a packed key lane overflowed (string > 3 bytes or i64 >= 2^31):
+    // convert every column to the serialized wide-key path and rerun
+    gxp::GroupKeyDesc& g = ex->desc.gkey;
+    bool ok = true;
+    for (int k = 0; k < g.nCols && ok; k++) {
+      if (g.kind[k] == 1) {
+        g.kind[k] = 3;
+      } else if (g.kind[k] == 0) {
+        g.kind[k] = 5;  // offsets slot already assigned at bind
+        ok = g.slot[k] >= 0;
+      } else if (g.kind[k] == 2) {
+        g.kind[k] = 5;
+        g.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS, g.col[k]);
+        ok = g.slot[k] >= 0;
+      }
+    }
+    if (ok) {
+      g.wideMode = 1;
+      ex->desc.useGlds = 0;
+      ex->jitProg = nullptr;  // the packed-key specialization no longer applies
+      int32_t rc2 = ensureWideKeyStore(ex);
+      if (rc2) return rc2;
+      if (getenv("GX_DEBUG"))
+        fprintf(stderr, "[gx] packed key overflow -> wide-key retry\n");
+      return runFused(ex);
+    }
   }
   errFlag &= ~256u;
   if (ex->desc.ablate != 0 && getenv("GX_DEBUG"))
@@ -2459,6 +2561,87 @@ static int32_t runFused(gx_exec* ex) {
   if (getenv("GX_DEBUG"))
     fprintf(stderr, "[gx] occupied group slots: %zu\n", occ.size());
 
+  // wide-key mode: download the key records and decode each occupied slot's
+  // group values; slot keys pack (hash32|recordIdx), so deterministic output
+  // order comes from sorting by the DECODED canonical values instead
+  const bool wideKeys = ex->desc.gkey.wideMode != 0;
+  std::map<const gxp::GroupSlot*, std::vector<OutRowVal>> wideVals;
+  if (wideKeys) {
+    const gxp::GroupKeyDesc& g = ex->desc.gkey;
+    uint64_t nRecs = 0;
+    HIP_OK(ex, hipMemcpy(&nRecs, g.recCursor, 8, hipMemcpyDeviceToHost));
+    if (nRecs > (uint64_t)g.recCap) nRecs = (uint64_t)g.recCap;
+    std::vector<uint8_t> recs((size_t)nRecs * g.recBytes);
+    if (nRecs)
+      HIP_OK(ex, hipMemcpy(recs.data(), g.keyStore, recs.size(),
+                           hipMemcpyDeviceToHost));
+    auto decodeWideCol = [&](const gxp::GroupSlot* s, int k, OutRowVal* v,
+                             std::string* canon) -> int32_t {
+      uint32_t recIdx = (uint32_t)(s->key & 0xFFFFFFFFu);
+      if ((uint64_t)recIdx >= nRecs) {
+        ex->err = "wide key record index out of range";
+        return GX_ERR_INTERNAL;
+      }
+      const uint8_t* rec = recs.data() + (size_t)recIdx * g.recBytes;
+      const uint64_t* r64 = (const uint64_t*)rec;
+      int srcCol = g.col[k];
+      v->type = ex->desc.table.cols[srcCol].type;
+      if ((r64[0] >> k) & 1) {
+        v->isNull = true;
+        canon->push_back('\x01');
+        return GX_OK;
+      }
+      canon->push_back('\x02');
+      const uint64_t* f = (const uint64_t*)(rec + 16 + 24 * k);
+      if (g.kind[k] == 3) {
+        v->i64 = (int64_t)f[0];
+        v->u64 = f[0];
+        canon->append((const char*)&f[0], 8);
+      } else if (g.kind[k] == 4) {
+        __int128 units = ((__int128)(int64_t)f[1] << 64) | f[0];
+        v->type = GX_TYPE_DECIMAL;
+        v->dec = decFromUnits(units, g.kscale[k]);
+        canon->append((const char*)&f[0], 16);
+      } else {  // kind 5: trimmed string (prefix inline, tail via owner row)
+        uint64_t len = f[0];
+        v->type = GX_TYPE_STRING;
+        v->str.clear();
+        const char* pfx = (const char*)&f[1];
+        for (uint64_t j = 0; j < len && j < 16; j++) v->str.push_back(pfx[j]);
+        if (len > 16) {
+          int64_t owner = (int64_t)r64[1];
+          const gxp::DevCol& c = ex->desc.table.cols[srcCol];
+          int64_t os = 0;
+          HIP_OK(ex, hipMemcpy(&os, c.offsets + owner, 8, hipMemcpyDeviceToHost));
+          std::vector<uint8_t> tail((size_t)len - 16);
+          HIP_OK(ex, hipMemcpy(tail.data(), (const uint8_t*)c.data + os + 16,
+                               tail.size(), hipMemcpyDeviceToHost));
+          v->str.append((const char*)tail.data(), tail.size());
+        }
+        canon->append((const char*)&len, 8);
+        canon->append(v->str);
+      }
+      return GX_OK;
+    };
+    std::map<const gxp::GroupSlot*, std::string> canonOf;
+    for (const gxp::GroupSlot* s : occ) {
+      std::vector<OutRowVal> vals;
+      std::string ck;
+      for (int k = 0; k < g.nCols; k++) {
+        OutRowVal v;
+        int32_t rc2 = decodeWideCol(s, k, &v, &ck);
+        if (rc2) return rc2;
+        vals.push_back(std::move(v));
+      }
+      canonOf[s] = std::move(ck);
+      wideVals[s] = std::move(vals);
+    }
+    std::sort(occ.begin(), occ.end(),
+              [&](const gxp::GroupSlot* a, const gxp::GroupSlot* b) {
+                return canonOf[a] < canonOf[b];
+              });
+  }
+
   const PNode& agg = ex->plan.nodes[ex->aggRoot];
   bool partial = agg.aggMode == GX_AGG_MODE_PARTIAL;
   ex->resultRows.clear();
@@ -2476,6 +2659,10 @@ static int32_t runFused(gx_exec* ex) {
   for (const gxp::GroupSlot* s : occ) {
     std::vector<OutRowVal> row;
     for (int k = 0; k < ex->desc.gkey.nCols; k++) {
+      if (wideKeys) {
+        row.push_back(wideVals[s][k]);
+        continue;
+      }
       OutRowVal v;
       int srcCol = ex->desc.gkey.col[k];
       decodeGroupLane(ex, (uint32_t)(s->key >> (32 * k)), ex->desc.gkey.kind[k],
@@ -2493,6 +2680,10 @@ static int32_t runFused(gx_exec* ex) {
                 (unsigned long long)s->accLo[phys], (long long)s->accHi[phys],
                 (long long)cnt);
       if (ad.fr >= 0) {  // firstrow(group col): decode from the group key
+        if (wideKeys) {
+          row.push_back(wideVals[s][ad.fr]);
+          continue;
+        }
         OutRowVal v;
         int srcCol = ex->desc.gkey.col[ad.fr];
         decodeGroupLane(ex, (uint32_t)(s->key >> (32 * ad.fr)),

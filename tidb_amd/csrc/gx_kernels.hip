@@ -134,7 +134,7 @@ __global__ void genLineitemKernel(DevTable tab, int64_t rowBegin, int64_t nRows,
 
 // per-row pipeline after the raw fetch: filter -> VM -> LDS aggregate.
 // Returns false on a hard failure (error flag already set).
-template <bool WIDE, bool DIVOK, typename VMT, typename RAWT>
+template <bool WIDE, bool DIVOK, typename VMT, bool WK = false, typename RAWT>
 __device__ __attribute__((always_inline)) inline bool processRow(const FusedQueryDesc& d, int64_t row,
                                   const RAWT& raw, Lds3GroupSlot* lds,
                                   uint64_t* mySel) {
@@ -358,8 +358,17 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   }
 
   // ---- group lookup / insert (LDS table, or global when NDV > kLdsGroups) ----
+  // Wide serialized keys resolve the GLOBAL slot first (hash probe + record
+  // verify); the LDS stage then keys on that slot index — exact, and the
+  // flush adds straight into globalTable[slot].
   uint64_t key;
-  if (!makeGroupKey(d, row, raw, &key, d.errorFlag)) return false;
+  uint32_t gslot = 0;
+  if constexpr (WK) {
+    if (!makeWideGroupKey<WIDE>(d, row, raw, vm, &gslot)) return false;
+    key = gslot;
+  } else {
+    if (!makeGroupKey(d, row, raw, &key, d.errorFlag)) return false;
+  }
   if (!d.noLds) {
     uint32_t slot = (uint32_t)(splitmix64(key) & (kLdsGroups - 1));
     for (int probe = 0;; probe++) {
@@ -406,22 +415,27 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     return true;
   }
   // global-direct path (mid/high NDV; table grows on kErrGlobalFull retry)
-  uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
-  uint32_t slot = (uint32_t)(splitmix64(key) & gmask);
-  for (uint32_t probe = 0;; probe++) {
-    if (probe > gmask) {
-      atomicOr(d.errorFlag, kErrGlobalFull);
-      return false;
+  uint32_t slot;
+  if constexpr (WK) {
+    slot = gslot;  // makeWideGroupKey already resolved (and claimed) the slot
+  } else {
+    uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
+    slot = (uint32_t)(splitmix64(key) & gmask);
+    for (uint32_t probe = 0;; probe++) {
+      if (probe > gmask) {
+        atomicOr(d.errorFlag, kErrGlobalFull);
+        return false;
+      }
+      uint64_t cur = d.globalTable[slot].key;
+      if (cur == key) break;
+      if (cur == kEmptyKey) {
+        uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
+                                  (unsigned long long)kEmptyKey,
+                                  (unsigned long long)key);
+        if (prev == kEmptyKey || prev == key) break;
+      }
+      slot = (slot + 1) & gmask;
     }
-    uint64_t cur = d.globalTable[slot].key;
-    if (cur == key) break;
-    if (cur == kEmptyKey) {
-      uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
-                                (unsigned long long)kEmptyKey,
-                                (unsigned long long)key);
-      if (prev == kEmptyKey || prev == key) break;
-    }
-    slot = (slot + 1) & gmask;
   }
   GroupSlot* target = &d.globalTable[slot];
   if (d.sharedCnt) accumInto(target, 0, Int128{0, 0}, 1);  // bumps cnt[0] only
@@ -453,7 +467,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   return true;
 }
 
-template <bool WIDE, int R, bool DIVOK = false, int NVM = 12>
+template <bool WIDE, int R, bool DIVOK = false, int NVM = 12, bool WK = false>
 __launch_bounds__(256)
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
@@ -496,10 +510,10 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
       if (rB < end) fetchRow(d.table, d.fetch, d.nFetch, rB, rawB);
       using VMT = typename std::conditional<NVM <= 12, VmState<WIDE>,
                                             VmState14<WIDE>>::type;
-      if (!processRow<WIDE, DIVOK, VMT>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
+      if (!processRow<WIDE, DIVOK, VMT, WK>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
       const int64_t rA2 = row + 2 * stride;
       if (rA2 < end) fetchRow(d.table, d.fetch, d.nFetch, rA2, rawA);
-      if (rB < end && !processRow<WIDE, DIVOK, VMT>(d, rB, rawB, lds3, &mySel)) failed = true;
+      if (rB < end && !processRow<WIDE, DIVOK, VMT, WK>(d, rB, rawB, lds3, &mySel)) failed = true;
     }
   }
 
@@ -517,20 +531,25 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     if (lds[i].key == kEmptyKey) continue;
     uint64_t key = lds[i].key;
-    uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
-    uint32_t slot = (uint32_t)(splitmix64(key) & gmask);
+    uint32_t slot;
     bool ok = true;
-    for (uint32_t probe = 0;; probe++) {
-      if (probe > gmask) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
-      uint64_t cur = d.globalTable[slot].key;
-      if (cur == key) break;
-      if (cur == kEmptyKey) {
-        uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
-                                  (unsigned long long)kEmptyKey,
-                                  (unsigned long long)key);
-        if (prev == kEmptyKey || prev == key) break;
+    if constexpr (WK) {
+      slot = (uint32_t)key;  // wide keys: the LDS key IS the global slot
+    } else {
+      uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
+      slot = (uint32_t)(splitmix64(key) & gmask);
+      for (uint32_t probe = 0;; probe++) {
+        if (probe > gmask) { atomicOr(d.errorFlag, kErrGlobalFull); ok = false; break; }
+        uint64_t cur = d.globalTable[slot].key;
+        if (cur == key) break;
+        if (cur == kEmptyKey) {
+          uint64_t prev = atomicCAS((unsigned long long*)&d.globalTable[slot].key,
+                                    (unsigned long long)kEmptyKey,
+                                    (unsigned long long)key);
+          if (prev == kEmptyKey || prev == key) break;
+        }
+        slot = (slot + 1) & gmask;
       }
-      slot = (slot + 1) & gmask;
     }
     if (!ok) continue;
     for (int s = 0; s < d.nAccSlots; s++) {
@@ -1490,6 +1509,8 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
   int nSlots = 1 << desc.globalGroupsLog2;
   hipLaunchKernelGGL(initGlobalTableKernel, dim3((nSlots + 255) / 256),
                      dim3(256), 0, s, desc.globalTable, nSlots);
+  if (desc.gkey.wideMode && desc.gkey.recCursor)
+    hipMemsetAsync(desc.gkey.recCursor, 0, 8, s);
   if (desc.useGlds) {
     size_t shmem = ((sizeof(GroupSlot) * kLdsGroups + 15) & ~15ULL) +
                    (size_t)4 * 2 * desc.tileBytes;
@@ -1499,6 +1520,15 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
     else
       hipLaunchKernelGGL((fusedAggGldsKernel<false>), dim3(grid), dim3(256),
                          shmem, s, devDesc);
+  } else if (desc.gkey.wideMode) {
+    // serialized wide group keys: one general variant (R=8, div-capable,
+    // 14-reg VM) x narrow/wide arithmetic
+    if (desc.wide)
+      hipLaunchKernelGGL((fusedAggKernel<true, 8, true, 14, true>), dim3(grid),
+                         dim3(256), 0, s, devDesc);
+    else
+      hipLaunchKernelGGL((fusedAggKernel<false, 8, true, 14, true>), dim3(grid),
+                         dim3(256), 0, s, devDesc);
   } else {
     // pick the 5-slot raw-state variant when every non-staged fetch slot
     // fits (higher occupancy for Q1-shaped queries)

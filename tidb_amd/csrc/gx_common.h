@@ -326,6 +326,8 @@ struct JoinPostPred {
   int32_t ctype = 0;           // kind 1: GX_TYPE_I64 or GX_TYPE_TIME
 };
 
+constexpr int kMaxJoinKeys = 4;
+
 // ---- standalone hash join (inner, duplicate build keys) ----
 // HashJoinV2 equivalent (join/hash_join_v2.go): chained hash table over the
 // build side — heads + intrusive per-row next links (hash_table_v2.go:22-53
@@ -348,11 +350,19 @@ struct HashJoinDesc {
   int32_t strConstBLen = 0;
   uint8_t strConstP[16];
   int32_t strConstPLen = 0;
-  // 1-2 fixed 8-byte key columns per side (int64/packed-Time; the
-  // SerializeKeys concatenation of codec.go:852-910 specialized to
-  // fixed-width keys — equality of the concatenation == per-column equality)
+  // up to kMaxJoinKeys key columns per side (SerializeKeys semantics,
+  // codec.go:852-910,445-456). keyKind per column:
+  //   0 = raw 8 B (int64 / packed CoreTime; equality is bitwise — the round-1
+  //       fast path, kernels unchanged)
+  //   1 = decimal, value-normalized (ToHashKey semantics: units with trailing
+  //       zeros removed + residual scale, so 1.10 == 1.1 across fracs)
+  //   2 = varlen string (utf8mb4_bin PAD SPACE: trailing spaces trimmed,
+  //       BinCollatorKey semantics)
+  // generalKeys = 1 selects the general build/probe kernel variants.
   int32_t nKeys = 1;
-  int32_t bKeyCol[2] = {0, 0}, pKeyCol[2] = {0, 0};
+  int32_t bKeyCol[kMaxJoinKeys] = {0, 0, 0, 0}, pKeyCol[kMaxJoinKeys] = {0, 0, 0, 0};
+  int32_t keyKind[kMaxJoinKeys] = {0, 0, 0, 0};
+  int32_t generalKeys = 0;
   uint32_t* heads = nullptr;  // 1<<headsLog2 entries: build row+1, 0 = empty
   int32_t headsLog2 = 0;
   uint32_t* next = nullptr;   // per build row: next chain row+1, 0 = end

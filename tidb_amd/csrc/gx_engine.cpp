@@ -1210,8 +1210,8 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
     return GX_ERR_INVALID;
   }
   if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
-      jn.buildKeys.size() > 2) {
-    ex->err = "device hash join supports 1-2 key columns this round";
+      (int)jn.buildKeys.size() > gxp::kMaxJoinKeys) {
+    ex->err = "too many join key columns";
     return GX_ERR_INVALID;
   }
   JoinSideRef B, P;
@@ -1238,15 +1238,29 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
     }
     int bt = B.types[bk.colIdx];
     int pt = P.types[pk.colIdx];
-    // fixed 8-byte keys: int64 or packed CoreTime (SerializeKeys writes the
-    // raw 8 bytes for both - equality is bitwise)
-    if (bt != pt || (bt != GX_TYPE_I64 && bt != GX_TYPE_TIME)) {
-      ex->err = "device join keys must be int64/time this round";
+    // SerializeKeys semantics (codec.go:852-910): int64/packed-Time compare
+    // as their raw 8 bytes; decimals compare value-normalized (ToHashKey,
+    // so 1.10 == 1.1 across fracs); strings compare PAD-SPACE-trimmed bytes
+    if (bt != pt) {
+      ex->err = "join key column types must match";
+      return GX_ERR_INVALID;
+    }
+    if (bt == GX_TYPE_I64 || bt == GX_TYPE_TIME) {
+      hj.keyKind[k] = 0;
+    } else if (bt == GX_TYPE_DECIMAL) {
+      hj.keyKind[k] = 1;
+      hj.generalKeys = 1;
+    } else if (bt == GX_TYPE_STRING) {
+      hj.keyKind[k] = 2;
+      hj.generalKeys = 1;
+    } else {
+      ex->err = "unsupported join key column type";
       return GX_ERR_INVALID;
     }
     hj.bKeyCol[k] = bk.colIdx;
     hj.pKeyCol[k] = pk.colIdx;
   }
+  if (hj.nKeys > 2) hj.generalKeys = 1;  // fast path caches 2 u64 keys
   // one filter conjunct per SOURCE side (stage-fed sides carry their filter
   // as the inner stage\'s post preds)
   auto doPred = [&](const PNode* sel, int srcNode, gxp::PredDesc* pd,

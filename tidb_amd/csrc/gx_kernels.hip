@@ -1779,8 +1779,144 @@ __device__ inline bool hjBuildKeyEq(const HashJoinDesc& d, uint32_t brow,
   return true;
 }
 
+// ---- general join keys (SerializeKeys semantics, codec.go:852-910) ----
+// Per-column canonical values: raw 8 B (kind 0), value-normalized decimal
+// (kind 1 — units with trailing zeros stripped + residual scale, ToHashKey
+// semantics so 1.10 == 1.1 across fracs), PAD-SPACE-trimmed varlen string
+// (kind 2, compared byte-wise against the resident build column).
+struct HjKeyVals {
+  uint64_t lo[kMaxJoinKeys];   // raw value | decimal units lo | string trimmed len
+  int64_t hi[kMaxJoinKeys];    // decimal units hi
+  int32_t sc[kMaxJoinKeys];    // decimal residual scale
+  int64_t soff[kMaxJoinKeys];  // string byte start in the probe-side data
+};
+
+__device__ inline bool hjDecCanon(const DevCol& c, int64_t row, uint64_t* lo,
+                                  int64_t* hi, int32_t* sc, uint32_t* err) {
+  Int128 u;
+  int scale;
+  if (!loadDecimalUnits<true>((const uint8_t*)c.data + row * 40, &u, &scale,
+                              err))
+    return false;
+  __int128 v = ((__int128)u.hi << 64) | u.lo;
+  bool neg = v < 0;
+  unsigned __int128 a = (unsigned __int128)(neg ? -v : v);
+  while (scale > 0) {
+    unsigned __int128 q = u128DivU64(a, 10);
+    if (q * 10 != a) break;
+    a = q;
+    scale--;
+  }
+  __int128 r = neg ? -(__int128)a : (__int128)a;
+  *lo = (uint64_t)r;
+  *hi = (int64_t)(r >> 64);
+  *sc = scale;
+  return true;
+}
+
+__device__ inline bool hjLoadKeysG(const HashJoinDesc& d, const DevTable& t,
+                                   const int32_t* cols, int64_t row,
+                                   HjKeyVals* v, uint64_t* hash) {
+  uint64_t h = 0x243F6A8885A308D3ULL;
+  for (int k = 0; k < d.nKeys; k++) {
+    const DevCol& c = t.cols[cols[k]];
+    if (colIsNull(c, row)) return false;
+    int kind = d.keyKind[k];
+    if (kind == 0) {
+      uint64_t x = gptr<uint64_t>(c.data)[row];
+      v->lo[k] = x;
+      h = splitmix64(h ^ x);
+    } else if (kind == 1) {
+      if (!hjDecCanon(c, row, &v->lo[k], &v->hi[k], &v->sc[k], d.errorFlag))
+        return false;
+      h = splitmix64(h ^ v->lo[k]);
+      h = splitmix64(h ^ (uint64_t)v->hi[k]);
+      h = splitmix64(h ^ (uint64_t)(uint32_t)v->sc[k]);
+    } else {
+      int64_t s = gptr<int64_t>(c.offsets)[row];
+      int64_t e = gptr<int64_t>(c.offsets)[row + 1];
+      auto p = gptr<uint8_t>(c.data);
+      while (e > s && p[e - 1] == ' ') e--;
+      v->lo[k] = (uint64_t)(e - s);
+      v->soff[k] = s;
+      for (int64_t j = s; j < e; j += 8) {
+        uint64_t w = 0;
+        int64_t m = e - j < 8 ? e - j : 8;
+        for (int64_t tt = 0; tt < m; tt++)
+          w |= (uint64_t)p[j + tt] << (8 * tt);
+        h = splitmix64(h ^ w);
+      }
+      h = splitmix64(h ^ (v->lo[k] * 0x9E3779B97F4A7C15ULL + 1));
+    }
+  }
+  *hash = h;
+  return true;
+}
+
+__device__ inline bool hjBuildKeyEqG(const HashJoinDesc& d, uint32_t brow,
+                                     const HjKeyVals& pv) {
+  for (int k = 0; k < d.nKeys; k++) {
+    const DevCol& c = d.build.cols[d.bKeyCol[k]];
+    int kind = d.keyKind[k];
+    if (kind == 0) {
+      if (gptr<uint64_t>(c.data)[brow] != pv.lo[k]) return false;
+    } else if (kind == 1) {
+      uint64_t lo;
+      int64_t hi;
+      int32_t sc;
+      if (!hjDecCanon(c, brow, &lo, &hi, &sc, d.errorFlag)) return false;
+      if (lo != pv.lo[k] || hi != pv.hi[k] || sc != pv.sc[k]) return false;
+    } else {
+      int64_t s = gptr<int64_t>(c.offsets)[brow];
+      int64_t e = gptr<int64_t>(c.offsets)[brow + 1];
+      auto bp = gptr<uint8_t>(c.data);
+      while (e > s && bp[e - 1] == ' ') e--;
+      if ((uint64_t)(e - s) != pv.lo[k]) return false;
+      auto pp = gptr<uint8_t>(d.probe.cols[d.pKeyCol[k]].data);
+      for (int64_t j = 0; j < e - s; j++)
+        if (bp[s + j] != pp[pv.soff[k] + j]) return false;
+    }
+  }
+  return true;
+}
+
+// key storage + load/eq shims shared by the fast (2 x u64 registers) and
+// general (HjKeyVals) kernel variants
+template <bool G>
+struct HjKeys;
+template <>
+struct HjKeys<false> {
+  uint64_t k0 = 0, k1 = 0;
+};
+template <>
+struct HjKeys<true> {
+  HjKeyVals v;
+};
+
+__device__ inline bool hjLoad(const HashJoinDesc& d, const DevTable& t,
+                              const int32_t* cols, int64_t row,
+                              HjKeys<false>& K, uint64_t* h) {
+  if (!hjLoadKeys(d, t, cols, row, &K.k0, &K.k1)) return false;
+  *h = hjHash(d, K.k0, K.k1);
+  return true;
+}
+__device__ inline bool hjLoad(const HashJoinDesc& d, const DevTable& t,
+                              const int32_t* cols, int64_t row,
+                              HjKeys<true>& K, uint64_t* h) {
+  return hjLoadKeysG(d, t, cols, row, &K.v, h);
+}
+__device__ inline bool hjEq(const HashJoinDesc& d, uint32_t brow,
+                            const HjKeys<false>& K) {
+  return hjBuildKeyEq(d, brow, K.k0, K.k1);
+}
+__device__ inline bool hjEq(const HashJoinDesc& d, uint32_t brow,
+                            const HjKeys<true>& K) {
+  return hjBuildKeyEqG(d, brow, K.v);
+}
+
 // chain-insert every qualifying build row (lock-free head CAS; the next[]
 // write is published to the probe kernels by the dispatch boundary)
+template <bool G>
 __global__ void hjBuildKernel(const HashJoinDesc* __restrict__ dp) {
   const HashJoinDesc& d = *dp;
   int64_t n = d.build.nRows;
@@ -1790,9 +1926,10 @@ __global__ void hjBuildKernel(const HashJoinDesc* __restrict__ dp) {
     bool pass = d.nPredB == 0 ||
                 evalSimplePred(d.build, d.predB, d.strConstB, d.strConstBLen, row);
     if (!pass) continue;
-    uint64_t key, key1;
-    if (!hjLoadKeys(d, d.build, d.bKeyCol, row, &key, &key1)) continue;
-    uint32_t slot = (uint32_t)(hjHash(d, key, key1) & mask);
+    HjKeys<G> K;
+    uint64_t h;
+    if (!hjLoad(d, d.build, d.bKeyCol, row, K, &h)) continue;
+    uint32_t slot = (uint32_t)(h & mask);
     uint32_t newHead = (uint32_t)row + 1;
     uint32_t old = d.heads[slot];
     for (;;) {
@@ -1808,7 +1945,7 @@ __global__ void hjBuildKernel(const HashJoinDesc* __restrict__ dp) {
 // resident build column. FILL=false counts (wavefront-reduced into
 // counters[0]); FILL=true reserves a contiguous range per probe row via one
 // atomic on counters[1] and writes the (build,probe) match pairs.
-template <bool FILL>
+template <bool FILL, bool G = false>
 __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
   const HashJoinDesc& d = *dp;
   int64_t n = d.probe.nRows;
@@ -1825,7 +1962,7 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
     if (__ballot(active) == 0) break;
     uint32_t cnt = 0;
     uint32_t head = 0;
-    uint64_t key = 0, key1 = 0;
+    HjKeys<G> K;
     // matched rows are cached in registers during the counting walk so the
     // common case (a handful of duplicates) never re-walks the chain
     uint32_t hit[4];
@@ -1833,11 +1970,12 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
       bool pass = d.nPredP == 0 ||
                   evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen,
                                  row);
-      if (pass && hjLoadKeys(d, d.probe, d.pKeyCol, row, &key, &key1)) {
-        head = gptr<uint32_t>(d.heads)[(uint32_t)(hjHash(d, key, key1) & mask)];
+      uint64_t h;
+      if (pass && hjLoad(d, d.probe, d.pKeyCol, row, K, &h)) {
+        head = gptr<uint32_t>(d.heads)[(uint32_t)(h & mask)];
         for (uint32_t cur = head; cur != 0;) {
           uint32_t brow = cur - 1;
-          if (hjBuildKeyEq(d, brow, key, key1)) {
+          if (hjEq(d, brow, K)) {
             if (FILL && cnt < 4) hit[cnt] = brow;
             cnt++;
           }
@@ -1879,7 +2017,7 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
       } else if (cnt > 4) {
         for (uint32_t cur = head; cur != 0;) {
           uint32_t brow = cur - 1;
-          if (hjBuildKeyEq(d, brow, key, key1)) {
+          if (hjEq(d, brow, K)) {
             d.outBuild[base] = brow;
             d.outProbe[base] = (uint32_t)row;
             base++;
@@ -2428,18 +2566,32 @@ int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
                               : h.probe.nRows;
   if (rows == 0) return 0;
   dim3 g(gridFor(rows));
-  if (phase == 0)
-    hipLaunchKernelGGL(hjBuildKernel, g, dim3(256), 0, (hipStream_t)stream,
-                       devDesc);
-  else if (phase == 1)
-    hipLaunchKernelGGL(hjProbeKernel<false>, g, dim3(256), 0,
-                       (hipStream_t)stream, devDesc);
-  else if (phase == 2)
-    hipLaunchKernelGGL(hjProbeKernel<true>, g, dim3(256), 0,
-                       (hipStream_t)stream, devDesc);
-  else
+  const bool gen = h.generalKeys != 0;
+  if (phase == 0) {
+    if (gen)
+      hipLaunchKernelGGL(hjBuildKernel<true>, g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+    else
+      hipLaunchKernelGGL(hjBuildKernel<false>, g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+  } else if (phase == 1) {
+    if (gen)
+      hipLaunchKernelGGL((hjProbeKernel<false, true>), g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+    else
+      hipLaunchKernelGGL((hjProbeKernel<false, false>), g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+  } else if (phase == 2) {
+    if (gen)
+      hipLaunchKernelGGL((hjProbeKernel<true, true>), g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+    else
+      hipLaunchKernelGGL((hjProbeKernel<true, false>), g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+  } else {
     hipLaunchKernelGGL(hjFilterPairsKernel, g, dim3(256), 0,
                        (hipStream_t)stream, devDesc);
+  }
   return (int)hipGetLastError();
 }
 

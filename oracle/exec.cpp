@@ -967,18 +967,34 @@ class TopNExec : public Exec {
   size_t emit_ = 0;
 };
 
-// ---------- inner hash join ----------
-// HashJoinV2 semantics (join/hash_join_v2.go, inner_join_probe.go:27-86):
-// output columns = build side cols then probe side cols, one output row per
-// matching (build,probe) pair; NULL join keys never match.
+// ---------- hash join ----------
+// HashJoinV2 semantics (join/hash_join_v2.go): output columns = build side
+// cols then probe side cols, one output row per matching (build,probe) pair;
+// NULL join keys never match a chain entry. joinType
+// (gx_executor.h gx_pb_hashjoin):
+//   0 inner  (inner_join_probe.go:27-86)
+//   1 left outer, probe = outer side: unmatched probe rows emit once with
+//     build cols NULL (outer_join_probe.go probe-outer path; NULL-key probe
+//     rows are unmatched)
+//   2 right outer, build = outer side: matched pairs plus every unmatched
+//     build row with probe cols NULL (outer_join_probe.go build-outer path
+//     via per-row matched flags)
+//   3 semi: emit each probe row once if ANY build row matches; output =
+//     probe cols only (base_semi_join.go)
+//   4 anti semi: emit each probe row iff NO build row matches (NULL-key
+//     probe rows match nothing and are emitted); output = probe cols only
+//     (anti_semi_join_probe.go, non-null-aware variant)
 class HashJoinExec : public Exec {
  public:
   HashJoinExec(const Plan& plan, const PlanNode& node, std::unique_ptr<Exec> build,
                std::unique_ptr<Exec> probe)
       : plan_(plan), node_(node), build_(std::move(build)), probe_(std::move(probe)) {
-    for (size_t i = 0; i < build_->outTypes.size(); i++) {
-      outTypes.push_back(build_->outTypes[i]);
-      outFracs.push_back(build_->outFracs[i]);
+    int jt = node_.joinType;
+    if (jt != 3 && jt != 4) {
+      for (size_t i = 0; i < build_->outTypes.size(); i++) {
+        outTypes.push_back(build_->outTypes[i]);
+        outFracs.push_back(build_->outFracs[i]);
+      }
     }
     for (size_t i = 0; i < probe_->outTypes.size(); i++) {
       outTypes.push_back(probe_->outTypes[i]);
@@ -991,6 +1007,10 @@ class HashJoinExec : public Exec {
     pendEmit_ = 0;
     buildData_.clear();
     table_.clear();
+    matched_.clear();
+    drainChunk_ = 0;
+    drainRow_ = 0;
+    probeDone_ = false;
     int32_t ec = build_->open();
     if (ec) return ec;
     return probe_->open();
@@ -1008,7 +1028,9 @@ class HashJoinExec : public Exec {
       if (ec) return ec;
       built_ = true;
     }
-    size_t nb = build_->outTypes.size();
+    int jt = node_.joinType;
+    size_t nb = (jt == 3 || jt == 4) ? 0 : build_->outTypes.size();
+    size_t np = probe_->outTypes.size();
     for (;;) {
       // drain matches pending from the previous probe chunk first: one Next
       // fills at most MaxChunkSize rows (exec.Executor contract,
@@ -1019,11 +1041,35 @@ class HashJoinExec : public Exec {
         pendEmit_++;
       }
       if (out.numRows() >= kMaxChunkSize) return GX_OK;
+      if (probeDone_) {
+        // right outer: unmatched build rows with probe cols NULL
+        if (jt == 2) {
+          while (out.numRows() < kMaxChunkSize &&
+                 drainChunk_ < buildData_.size()) {
+            const Chunk& bc = buildData_[drainChunk_];
+            if (drainRow_ >= (size_t)bc.numRows()) {
+              drainChunk_++;
+              drainRow_ = 0;
+              continue;
+            }
+            if (!matched_[drainChunk_][drainRow_]) {
+              for (size_t c = 0; c < nb; c++)
+                out.cols[c].appendFrom(bc.cols[c], (int)drainRow_);
+              for (size_t c = 0; c < np; c++) out.cols[nb + c].appendNull();
+            }
+            drainRow_++;
+          }
+        }
+        return GX_OK;
+      }
       Chunk in;
       int32_t ec = probe_->next(in);
       if (ec) { err = probe_->err; return ec; }
       int n = in.numRows();
-      if (n == 0) return GX_OK;  // probe exhausted: emit what we have (may be 0 = EOF)
+      if (n == 0) {
+        probeDone_ = true;
+        continue;  // may still drain unmatched build rows (right outer)
+      }
       pending_.cols.resize(outTypes.size());
       for (size_t c = 0; c < pending_.cols.size(); c++) {
         pending_.cols[c].type = outTypes[c];
@@ -1035,15 +1081,36 @@ class HashJoinExec : public Exec {
       std::vector<uint8_t> hasNullKey(n, 0);
       ec = serializeJoinKeys(ctx, node_.probeKeys, in, keys, hasNullKey);
       if (ec) return ec;
+      auto appendProbeOnly = [&](int i) {
+        for (size_t c = 0; c < np; c++)
+          pending_.cols[nb + c].appendFrom(in.cols[c], i);
+      };
       for (int i = 0; i < n; i++) {
-        if (hasNullKey[i]) continue;
-        auto it = table_.find(keys[i]);
-        if (it == table_.end()) continue;
-        for (const RowRef& br : it->second) {
-          for (size_t c = 0; c < nb; c++)
-            pending_.cols[c].appendFrom(buildData_[br.chunkIdx].cols[c], br.rowIdx);
-          for (size_t c = 0; c < probe_->outTypes.size(); c++)
-            pending_.cols[nb + c].appendFrom(in.cols[c], i);
+        const std::vector<RowRef>* refs = nullptr;
+        if (!hasNullKey[i]) {
+          auto it = table_.find(keys[i]);
+          if (it != table_.end()) refs = &it->second;
+        }
+        bool any = refs && !refs->empty();
+        if (jt == 3) {  // semi
+          if (any) appendProbeOnly(i);
+          continue;
+        }
+        if (jt == 4) {  // anti semi
+          if (!any) appendProbeOnly(i);
+          continue;
+        }
+        if (any) {
+          for (const RowRef& br : *refs) {
+            if (jt == 2) matched_[br.chunkIdx][br.rowIdx] = 1;
+            for (size_t c = 0; c < nb; c++)
+              pending_.cols[c].appendFrom(buildData_[br.chunkIdx].cols[c],
+                                          br.rowIdx);
+            appendProbeOnly(i);
+          }
+        } else if (jt == 1) {  // left outer: null-extend the build side
+          for (size_t c = 0; c < nb; c++) pending_.cols[c].appendNull();
+          appendProbeOnly(i);
         }
       }
     }
@@ -1120,6 +1187,7 @@ class HashJoinExec : public Exec {
         if (hasNull[i]) continue;
         table_[keys[i]].push_back({ci, i});
       }
+      matched_.push_back(std::vector<uint8_t>(n, 0));
       buildData_.push_back(std::move(in));
     }
   }
@@ -1132,6 +1200,9 @@ class HashJoinExec : public Exec {
   std::unordered_map<std::string, std::vector<RowRef>> table_;
   Chunk pending_;      // matches of the current probe chunk, emitted <=
   int pendEmit_ = 0;   // kMaxChunkSize per next()
+  std::vector<std::vector<uint8_t>> matched_;  // right outer: per build row
+  bool probeDone_ = false;
+  size_t drainChunk_ = 0, drainRow_ = 0;  // right-outer drain cursor
 };
 
 }  // namespace

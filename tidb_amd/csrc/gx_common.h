@@ -380,9 +380,20 @@ struct HashJoinDesc {
   int64_t nPairs = 0;            // fill-phase total (filter input size)
   uint32_t* outBuild2 = nullptr; // filter-surviving pairs
   uint32_t* outProbe2 = nullptr;
+  // join type (gx_executor.h gx_pb_hashjoin): 0 inner, 1 left outer
+  // (probe = outer: unmatched probe rows null-extend the build side),
+  // 2 right outer (build = outer: matched flags + unmatched-build drain),
+  // 3 semi / 4 anti semi (probe cols only). Null-extended pairs carry
+  // kHjNullRow on the missing side; the gathers decode it to NULL.
+  int32_t joinType = 0;
+  uint32_t* matched = nullptr;   // right outer: 1 bit per build row
 };
 
-// phases: 0 = build (chain insert), 1 = count matches, 2 = fill match pairs
+constexpr uint32_t kHjNullRow = 0xFFFFFFFFu;
+
+// phases: 0 = build (chain insert), 1 = count matches (+ matched flags),
+// 2 = fill match pairs, 3 = post filter, 4 = count unmatched build rows
+// (right outer), 5 = fill unmatched build rows
 int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
                     const HashJoinDesc& h, void* stream);
 // standalone Selection compaction over h.probe (post[] = the CNF):

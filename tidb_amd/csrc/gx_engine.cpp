@@ -1205,10 +1205,12 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
                                 const PNode* postSel, int* stageOut) {
   const PPlan& plan = ex->plan;
   const PNode& jn = plan.nodes[joinNode];
-  if (jn.joinType != 0) {
-    ex->err = "only inner joins on device this round";
+  if (jn.joinType < 0 || jn.joinType > 4) {
+    ex->err = "unsupported join type (0=inner, 1=left outer, 2=right outer, "
+              "3=semi, 4=anti semi)";
     return GX_ERR_INVALID;
   }
+  hj.joinType = jn.joinType;
   if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
       (int)jn.buildKeys.size() > gxp::kMaxJoinKeys) {
     ex->err = "too many join key columns";
@@ -1287,11 +1289,24 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
   st.srcP = P.srcNode;
   st.buildStage = B.stage;
   st.probeStage = P.stage;
-  st.types = B.types;
-  st.fracs = B.fracs;
-  st.types.insert(st.types.end(), P.types.begin(), P.types.end());
-  st.fracs.insert(st.fracs.end(), P.fracs.begin(), P.fracs.end());
+  if (jn.joinType == 3 || jn.joinType == 4) {
+    // semi / anti semi emit the probe (outer) side only
+    st.types = P.types;
+    st.fracs = P.fracs;
+  } else {
+    st.types = B.types;
+    st.fracs = B.fracs;
+    st.types.insert(st.types.end(), P.types.begin(), P.types.end());
+    st.fracs.insert(st.fracs.end(), P.fracs.begin(), P.fracs.end());
+  }
   if (postSel) {
+    if (jn.joinType != 0) {
+      // other conditions change which outer rows count as matched
+      // (outer_join_probe.go post-condition matched tracking) — not wired
+      // into the non-inner paths yet; fail loudly rather than mis-join
+      ex->err = "post-join conditions with non-inner joins unsupported this round";
+      return GX_ERR_INVALID;
+    }
     rc = compilePostJoinPreds(ex, *postSel, st.types, st.fracs, nb, hj);
     if (rc) return rc;
   }
@@ -3101,7 +3116,8 @@ static int32_t runJoinAgg(gx_exec* ex) {
 // varlen (two-pass), and null bitmaps
 static int32_t gatherCols(gx_exec* ex, const gxp::DevTable& srcTab,
                           const uint32_t* idx, uint64_t total,
-                          gxp::DevTable* out, int dstBase) {
+                          gxp::DevTable* out, int dstBase,
+                          bool forceNulls = false) {
   for (int c = 0; c < srcTab.nCols; c++) {
     const gxp::DevCol& src = srcTab.cols[c];
     gxp::DevCol& dst = out->cols[dstBase + c];
@@ -3159,11 +3175,12 @@ static int32_t gatherCols(gx_exec* ex, const gxp::DevTable& srcTab,
         dst.denseOffsets = 1;
       }
     }
-    if (src.hasNulls && src.nullBitmap) {
+    if ((src.hasNulls && src.nullBitmap) || forceNulls) {
       dst.nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8);
       if (!dst.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-      if (gxp::gxGatherNulls(src.nullBitmap, idx, dst.nullBitmap,
-                             (int64_t)total, ex->stream) != 0) {
+      if (gxp::gxGatherNulls(src.hasNulls ? src.nullBitmap : nullptr, idx,
+                             dst.nullBitmap, (int64_t)total,
+                             ex->stream) != 0) {
         ex->err = "null gather launch failed";
         return GX_ERR_INTERNAL;
       }
@@ -3200,6 +3217,13 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
   hj.heads = (uint32_t*)devAlloc(ex, (1ULL << hj.headsLog2) * 4);
   hj.next = (uint32_t*)devAlloc(ex, std::max<int64_t>(nb, 1) * 4);
   if (!hj.heads || !hj.next) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+  if (hj.joinType == 2) {  // right outer: per-build-row matched flags
+    size_t words = ((size_t)nb + 31) / 32;
+    hj.matched = (uint32_t*)devAlloc(ex, std::max<size_t>(words * 4, 4));
+    if (!hj.matched) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    HIP_OK(ex, hipMemsetAsync(hj.matched, 0, std::max<size_t>(words * 4, 4),
+                              ex->stream));
+  }
   HIP_OK(ex, hipMemsetAsync(hj.heads, 0, (1ULL << hj.headsLog2) * 4, ex->stream));
   HIP_OK(ex, hipMemsetAsync(hj.counters, 0, 24, ex->stream));
   HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
@@ -3212,6 +3236,11 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
   HIP_OK(ex, hipEventRecord(evB, ex->stream));
   if (gxp::gxHashJoinPhase(1, ex->devHj, hj, ex->stream) != 0) {
     ex->err = "join kernel launch failed";
+    return GX_ERR_INTERNAL;
+  }
+  if (hj.joinType == 2 &&
+      gxp::gxHashJoinPhase(4, ex->devHj, hj, ex->stream) != 0) {
+    ex->err = "join unmatched-count launch failed";
     return GX_ERR_INTERNAL;
   }
   HIP_OK(ex, hipEventRecord(evC, ex->stream));
@@ -3246,6 +3275,11 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
       ex->err = "join fill launch failed";
       return GX_ERR_INTERNAL;
     }
+    if (hj.joinType == 2 &&
+        gxp::gxHashJoinPhase(5, ex->devHj, hj, ex->stream) != 0) {
+      ex->err = "join unmatched-fill launch failed";
+      return GX_ERR_INTERNAL;
+    }
     HIP_OK(ex, hipEventRecord(evF, ex->stream));
     // post-join filter (other conditions): compact surviving pairs before
     // the gather so rejected rows never touch the output
@@ -3276,9 +3310,18 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
     if (total > 0) {
       gxp::DevTable bsub = hj.build;
       gxp::DevTable psub = hj.probe;
-      int32_t rc = gatherCols(ex, bsub, gatherB, total, &st.out, 0);
-      if (rc == GX_OK)
-        rc = gatherCols(ex, psub, gatherP, total, &st.out, hj.build.nCols);
+      int32_t rc;
+      if (hj.joinType == 3 || hj.joinType == 4) {
+        // semi / anti semi: probe (outer) columns only
+        rc = gatherCols(ex, psub, gatherP, total, &st.out, 0, false);
+      } else {
+        // outer joins carry null-extended rows on the inner side
+        rc = gatherCols(ex, bsub, gatherB, total, &st.out, 0,
+                        hj.joinType == 1);
+        if (rc == GX_OK)
+          rc = gatherCols(ex, psub, gatherP, total, &st.out, hj.build.nCols,
+                          hj.joinType == 2);
+      }
       if (rc) return rc;
     }
   }

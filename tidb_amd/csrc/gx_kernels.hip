@@ -1962,14 +1962,15 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
     if (__ballot(active) == 0) break;
     uint32_t cnt = 0;
     uint32_t head = 0;
+    bool pass = false;
     HjKeys<G> K;
     // matched rows are cached in registers during the counting walk so the
     // common case (a handful of duplicates) never re-walks the chain
     uint32_t hit[4];
     if (active) {
-      bool pass = d.nPredP == 0 ||
-                  evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen,
-                                 row);
+      pass = d.nPredP == 0 ||
+             evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen,
+                            row);
       uint64_t h;
       if (pass && hjLoad(d, d.probe, d.pKeyCol, row, K, &h)) {
         head = gptr<uint32_t>(d.heads)[(uint32_t)(h & mask)];
@@ -1977,27 +1978,41 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
           uint32_t brow = cur - 1;
           if (hjEq(d, brow, K)) {
             if (FILL && cnt < 4) hit[cnt] = brow;
+            if (!FILL && d.joinType == 2)  // right outer: flag during count
+              atomicOr(&d.matched[brow >> 5], 1u << (brow & 31));
             cnt++;
           }
           cur = gptr<uint32_t>(d.next)[brow];
         }
       }
     }
+    // per-join-type output rows for this probe row (pred-failing rows emit
+    // nothing regardless of type)
+    uint32_t emit = cnt;
+    bool nullExt = false;
+    if (d.joinType == 1) {  // left outer: unmatched probe null-extends
+      if (pass && cnt == 0) { emit = 1; nullExt = true; }
+    } else if (d.joinType == 3) {  // semi: once on any match
+      emit = cnt ? 1 : 0;
+    } else if (d.joinType == 4) {  // anti semi: once on no match
+      emit = (pass && cnt == 0) ? 1 : 0;
+      nullExt = emit != 0;
+    }
     if (FILL) {
       uint64_t base = 0;
-      if (__ballot(cnt > 1) == 0) {
-        // common case (every lane matched <=1 row): one ballot + popcount
+      if (__ballot(emit > 1) == 0) {
+        // common case (every lane emits <=1 row): one ballot + popcount
         // replaces the 6-step shuffle prefix chain
-        uint64_t m = __ballot(cnt == 1);
+        uint64_t m = __ballot(emit == 1);
         if (m == 0) continue;
         if (lane == 63)
           base = atomicAdd((unsigned long long*)&d.counters[1],
                            (unsigned long long)__popcll(m));
         base = __shfl(base, 63, 64) + __popcll(m & ((1ULL << lane) - 1));
       } else {
-        // exclusive wave prefix sum of cnt -> per-lane slice of one
+        // exclusive wave prefix sum of emit -> per-lane slice of one
         // wave-wide reservation on the shared cursor
-        uint64_t pre = cnt;
+        uint64_t pre = emit;
         for (int off = 1; off < 64; off <<= 1) {
           uint64_t t = __shfl_up(pre, off, 64);
           if (lane >= off) pre += t;
@@ -2007,14 +2022,21 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
         if (lane == 63)
           base = atomicAdd((unsigned long long*)&d.counters[1],
                            (unsigned long long)waveTotal);
-        base = __shfl(base, 63, 64) + (pre - cnt);
+        base = __shfl(base, 63, 64) + (pre - emit);
       }
-      if (cnt > 0 && cnt <= 4) {
+      if (emit == 0) continue;
+      if (nullExt) {
+        d.outBuild[base] = kHjNullRow;
+        d.outProbe[base] = (uint32_t)row;
+      } else if (d.joinType == 3) {
+        d.outBuild[base] = hit[0];
+        d.outProbe[base] = (uint32_t)row;
+      } else if (cnt <= 4) {
         for (uint32_t k = 0; k < cnt; k++) {
           d.outBuild[base + k] = hit[k];
           d.outProbe[base + k] = (uint32_t)row;
         }
-      } else if (cnt > 4) {
+      } else {
         for (uint32_t cur = head; cur != 0;) {
           uint32_t brow = cur - 1;
           if (hjEq(d, brow, K)) {
@@ -2026,7 +2048,7 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
         }
       }
     } else {
-      my += cnt;
+      my += emit;
     }
   }
   if (!FILL) {
@@ -2101,6 +2123,52 @@ __global__ void hjFilterPairsKernel(const HashJoinDesc* __restrict__ dp) {
       d.outBuild2[base + off] = brow;
       d.outProbe2[base + off] = prow;
     }
+  }
+}
+
+// right outer: emit every predB-passing build row whose matched flag is
+// unset (NULL-key build rows never entered a chain, so they drain here too).
+// FILL=false adds the count into counters[0]; FILL=true reserves slices of
+// the shared fill cursor counters[1] and writes (brow, kHjNullRow) pairs.
+template <bool FILL>
+__global__ void hjUnmatchedKernel(const HashJoinDesc* __restrict__ dp) {
+  const HashJoinDesc& d = *dp;
+  int64_t n = d.build.nRows;
+  int lane = threadIdx.x & 63;
+  uint64_t my = 0;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;;
+       row += stride) {
+    bool active = row < n;
+    if (__ballot(active) == 0) break;
+    bool um = false;
+    if (active) {
+      bool pass = d.nPredB == 0 || evalSimplePred(d.build, d.predB,
+                                                  d.strConstB, d.strConstBLen,
+                                                  row);
+      um = pass &&
+           !((gptr<uint32_t>(d.matched)[row >> 5] >> (row & 31)) & 1);
+    }
+    if (FILL) {
+      uint64_t m = __ballot(um);
+      if (m == 0) continue;
+      uint64_t base = 0;
+      if (lane == 63)
+        base = atomicAdd((unsigned long long*)&d.counters[1],
+                         (unsigned long long)__popcll(m));
+      base = __shfl(base, 63, 64) + __popcll(m & ((1ULL << lane) - 1));
+      if (um) {
+        d.outBuild[base] = (uint32_t)row;
+        d.outProbe[base] = kHjNullRow;
+      }
+    } else {
+      my += um ? 1 : 0;
+    }
+  }
+  if (!FILL) {
+    for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
+    if (lane == 0 && my)
+      atomicAdd((unsigned long long*)&d.counters[0], (unsigned long long)my);
   }
 }
 
@@ -2434,7 +2502,9 @@ __global__ void hjGatherNullsKernel(const uint8_t* __restrict__ in,
     uint8_t v = 0;
     for (int j = 0; j < m; j++) {
       uint32_t src = idx[base + j];
-      if ((in[src >> 3] >> (src & 7)) & 1) v |= (uint8_t)(1 << j);
+      if (src == kHjNullRow) continue;  // null-extended row -> NULL
+      if (in == nullptr || ((in[src >> 3] >> (src & 7)) & 1))
+        v |= (uint8_t)(1 << j);
     }
     out[b] = v;
   }
@@ -2452,8 +2522,8 @@ __global__ void hjVarlenLensKernel(const int64_t* __restrict__ inOffsets,
                                    int64_t* __restrict__ lens, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    int64_t src = idx[i];
-    lens[i] = inOffsets[src + 1] - inOffsets[src];
+    uint32_t src = idx[i];
+    lens[i] = src == kHjNullRow ? 0 : inOffsets[src + 1] - inOffsets[src];
   }
 }
 
@@ -2467,7 +2537,8 @@ __global__ void hjVarlenBytesKernel(const uint8_t* __restrict__ inData,
                                     uint8_t* __restrict__ outData, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    int64_t src = idx[i];
+    uint32_t src = idx[i];
+    if (src == kHjNullRow) continue;
     int64_t s = inOffsets[src];
     int64_t len = inOffsets[src + 1] - s;
     int64_t o = outOffsets[i];
@@ -2481,6 +2552,15 @@ __global__ void sortGatherKernel(const uint8_t* __restrict__ in,
                                  int elemSize) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
+    if (idx[i] == kHjNullRow) {  // null-extended outer-join row
+      if (elemSize == 8) ((uint64_t*)out)[i] = 0;
+      else if (elemSize == 1) out[i] = 0;
+      else {
+        uint64_t* d = (uint64_t*)(out + i * 40);
+        d[0] = 0; d[1] = 0; d[2] = 0; d[3] = 0; d[4] = 0;
+      }
+      continue;
+    }
     int64_t src = idx[i];
     if (elemSize == 8) {
       ((uint64_t*)out)[i] = ((const uint64_t*)in)[src];
@@ -2561,11 +2641,20 @@ int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,
 
 int gxHashJoinPhase(int phase, const HashJoinDesc* devDesc,
                     const HashJoinDesc& h, void* stream) {
-  int64_t rows = phase == 0 ? h.build.nRows
-                 : phase == 3 ? h.nPairs
-                              : h.probe.nRows;
+  int64_t rows = (phase == 0 || phase >= 4) ? h.build.nRows
+                 : phase == 3               ? h.nPairs
+                                            : h.probe.nRows;
   if (rows == 0) return 0;
   dim3 g(gridFor(rows));
+  if (phase >= 4) {  // right-outer unmatched-build drain
+    if (phase == 4)
+      hipLaunchKernelGGL(hjUnmatchedKernel<false>, g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+    else
+      hipLaunchKernelGGL(hjUnmatchedKernel<true>, g, dim3(256), 0,
+                         (hipStream_t)stream, devDesc);
+    return (int)hipGetLastError();
+  }
   const bool gen = h.generalKeys != 0;
   if (phase == 0) {
     if (gen)

@@ -380,6 +380,17 @@ struct HashJoinDesc {
   int64_t nPairs = 0;            // fill-phase total (filter input size)
   uint32_t* outBuild2 = nullptr; // filter-surviving pairs
   uint32_t* outProbe2 = nullptr;
+  // deterministic fill (no shared-cursor atomics): the count pass caches
+  // per-probe-row match info in `hits` (0 = eligible-no-match, brow+1 =
+  // single match, kHjMulti = several matches -> fill re-walks the chain,
+  // kHjIneligible = pred-failed row) and writes each 64-row wave tile's
+  // emit total into tileCounts; the engine exclusive-scans those into
+  // tileBases, so the fill pass streams hits + bases and walks no chain for
+  // unique matches (the r01 fill was 12.5 ms of wave-serialized cursor
+  // atomics + a full chain re-walk)
+  uint32_t* hits = nullptr;      // per probe row
+  int64_t* tileCounts = nullptr; // per 64-row tile: emit total
+  int64_t* tileBases = nullptr;  // exclusive scan of tileCounts (+ total)
   // join type (gx_executor.h gx_pb_hashjoin): 0 inner, 1 left outer
   // (probe = outer: unmatched probe rows null-extend the build side),
   // 2 right outer (build = outer: matched flags + unmatched-build drain),
@@ -390,6 +401,8 @@ struct HashJoinDesc {
 };
 
 constexpr uint32_t kHjNullRow = 0xFFFFFFFFu;
+constexpr uint32_t kHjIneligible = 0xFFFFFFFFu;  // hits[]: pred-failed row
+constexpr uint32_t kHjMulti = 0xFFFFFFFEu;       // hits[]: >1 match
 
 // phases: 0 = build (chain insert), 1 = count matches (+ matched flags),
 // 2 = fill match pairs, 3 = post filter, 4 = count unmatched build rows

@@ -1210,7 +1210,6 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
               "3=semi, 4=anti semi)";
     return GX_ERR_INVALID;
   }
-  hj.joinType = jn.joinType;
   if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
       (int)jn.buildKeys.size() > gxp::kMaxJoinKeys) {
     ex->err = "too many join key columns";
@@ -1229,6 +1228,7 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
   }
   gx_exec::JoinStage st;
   gxp::HashJoinDesc& hj = st.hj;
+  hj.joinType = jn.joinType;
   hj.nKeys = (int32_t)jn.buildKeys.size();
   for (int k = 0; k < hj.nKeys; k++) {
     const PExpr& bk = plan.exprs[jn.buildKeys[k]];
@@ -3224,6 +3224,17 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
     HIP_OK(ex, hipMemsetAsync(hj.matched, 0, std::max<size_t>(words * 4, 4),
                               ex->stream));
   }
+  // deterministic-fill scratch: per-row walk cache + per-64-row-tile emit
+  // counts scanned into bases (no shared-cursor atomics in the fill pass)
+  const int64_t nProbe = hj.probe.nRows;
+  const int64_t nTiles = (nProbe + 63) / 64;
+  hj.hits = (uint32_t*)devAlloc(ex, std::max<int64_t>(nProbe, 1) * 4);
+  hj.tileCounts = (int64_t*)devAlloc(ex, std::max<int64_t>(nTiles, 1) * 8);
+  hj.tileBases = (int64_t*)devAlloc(ex, (std::max<int64_t>(nTiles, 1) + 1) * 8);
+  if (!hj.hits || !hj.tileCounts || !hj.tileBases) {
+    ex->err = "hipMalloc failed";
+    return GX_ERR_INTERNAL;
+  }
   HIP_OK(ex, hipMemsetAsync(hj.heads, 0, (1ULL << hj.headsLog2) * 4, ex->stream));
   HIP_OK(ex, hipMemsetAsync(hj.counters, 0, 24, ex->stream));
   HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
@@ -3242,6 +3253,23 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
       gxp::gxHashJoinPhase(4, ex->devHj, hj, ex->stream) != 0) {
     ex->err = "join unmatched-count launch failed";
     return GX_ERR_INTERNAL;
+  }
+  // scan tile counts -> fill bases (tileBases[nTiles] = matched total, which
+  // also seeds the unmatched-drain cursor for right outer)
+  if (nTiles > 0) {
+    size_t tmpBytes = 0;
+    gxp::gxExclusiveSumI64(hj.tileCounts, hj.tileBases, nTiles, nullptr,
+                           &tmpBytes, ex->stream);
+    void* tmp = devAlloc(ex, std::max<size_t>(tmpBytes, 1));
+    if (!tmp) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    HIP_OK(ex, hipMemsetAsync(hj.tileBases, 0, 8, ex->stream));
+    if (gxp::gxExclusiveSumI64(hj.tileCounts, hj.tileBases, nTiles, tmp,
+                               &tmpBytes, ex->stream) != 0) {
+      ex->err = "tile-base scan failed";
+      return GX_ERR_INTERNAL;
+    }
+    HIP_OK(ex, hipMemcpyAsync(&hj.counters[1], &hj.tileBases[nTiles], 8,
+                              hipMemcpyDeviceToDevice, ex->stream));
   }
   HIP_OK(ex, hipEventRecord(evC, ex->stream));
   HIP_OK(ex, hipStreamSynchronize(ex->stream));

@@ -1963,68 +1963,81 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
     uint32_t cnt = 0;
     uint32_t head = 0;
     bool pass = false;
+    bool needWalk = false;
     HjKeys<G> K;
-    // matched rows are cached in registers during the counting walk so the
-    // common case (a handful of duplicates) never re-walks the chain
+    // matched rows are cached in registers during the walk so the common
+    // case (a handful of duplicates) never re-walks the chain; the FILL
+    // pass walks NO chain for rows the count pass resolved to 0/1 matches —
+    // it streams the cached per-row hits and the scanned tile bases instead
+    // (the r01 fill spent 12.5 ms on wave-serialized cursor atomics + a
+    // full re-walk)
     uint32_t hit[4];
     if (active) {
-      pass = d.nPredP == 0 ||
-             evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen,
-                            row);
-      uint64_t h;
-      if (pass && hjLoad(d, d.probe, d.pKeyCol, row, K, &h)) {
-        head = gptr<uint32_t>(d.heads)[(uint32_t)(h & mask)];
-        for (uint32_t cur = head; cur != 0;) {
-          uint32_t brow = cur - 1;
-          if (hjEq(d, brow, K)) {
-            if (FILL && cnt < 4) hit[cnt] = brow;
-            if (!FILL && d.joinType == 2)  // right outer: flag during count
-              atomicOr(&d.matched[brow >> 5], 1u << (brow & 31));
-            cnt++;
+      if (FILL) {
+        uint32_t enc = gptr<uint32_t>(d.hits)[row];
+        pass = enc != kHjIneligible;
+        if (enc != 0 && enc != kHjIneligible) {
+          if (enc != kHjMulti) {
+            cnt = 1;
+            hit[0] = enc - 1;
+          } else {
+            needWalk = true;
           }
-          cur = gptr<uint32_t>(d.next)[brow];
+        }
+      } else {
+        pass = d.nPredP == 0 ||
+               evalSimplePred(d.probe, d.predP, d.strConstP, d.strConstPLen,
+                              row);
+        needWalk = pass;
+      }
+      if (needWalk) {
+        uint64_t h;
+        if (hjLoad(d, d.probe, d.pKeyCol, row, K, &h)) {
+          head = gptr<uint32_t>(d.heads)[(uint32_t)(h & mask)];
+          for (uint32_t cur = head; cur != 0;) {
+            uint32_t brow = cur - 1;
+            if (hjEq(d, brow, K)) {
+              if (cnt < 4) hit[cnt] = brow;
+              if (!FILL && d.joinType == 2)  // right outer: flag during count
+                atomicOr(&d.matched[brow >> 5], 1u << (brow & 31));
+              cnt++;
+            }
+            cur = gptr<uint32_t>(d.next)[brow];
+          }
         }
       }
+      if (!FILL)  // cache the walk result for the fill pass
+        d.hits[row] = !pass      ? kHjIneligible
+                      : cnt == 0 ? 0u
+                      : cnt == 1 ? hit[0] + 1
+                                 : kHjMulti;
     }
     // per-join-type output rows for this probe row (pred-failing rows emit
     // nothing regardless of type)
-    uint32_t emit = cnt;
+    uint32_t emit = active ? cnt : 0;
     bool nullExt = false;
-    if (d.joinType == 1) {  // left outer: unmatched probe null-extends
-      if (pass && cnt == 0) { emit = 1; nullExt = true; }
-    } else if (d.joinType == 3) {  // semi: once on any match
-      emit = cnt ? 1 : 0;
-    } else if (d.joinType == 4) {  // anti semi: once on no match
-      emit = (pass && cnt == 0) ? 1 : 0;
-      nullExt = emit != 0;
+    if (active) {
+      if (d.joinType == 1) {  // left outer: unmatched probe null-extends
+        if (pass && cnt == 0) { emit = 1; nullExt = true; }
+      } else if (d.joinType == 3) {  // semi: once on any match
+        emit = cnt ? 1 : 0;
+      } else if (d.joinType == 4) {  // anti semi: once on no match
+        emit = (pass && cnt == 0) ? 1 : 0;
+        nullExt = emit != 0;
+      }
     }
     if (FILL) {
-      uint64_t base = 0;
-      if (__ballot(emit > 1) == 0) {
-        // common case (every lane emits <=1 row): one ballot + popcount
-        // replaces the 6-step shuffle prefix chain
-        uint64_t m = __ballot(emit == 1);
-        if (m == 0) continue;
-        if (lane == 63)
-          base = atomicAdd((unsigned long long*)&d.counters[1],
-                           (unsigned long long)__popcll(m));
-        base = __shfl(base, 63, 64) + __popcll(m & ((1ULL << lane) - 1));
-      } else {
-        // exclusive wave prefix sum of emit -> per-lane slice of one
-        // wave-wide reservation on the shared cursor
-        uint64_t pre = emit;
-        for (int off = 1; off < 64; off <<= 1) {
-          uint64_t t = __shfl_up(pre, off, 64);
-          if (lane >= off) pre += t;
-        }
-        uint64_t waveTotal = __shfl(pre, 63, 64);
-        if (waveTotal == 0) continue;
-        if (lane == 63)
-          base = atomicAdd((unsigned long long*)&d.counters[1],
-                           (unsigned long long)waveTotal);
-        base = __shfl(base, 63, 64) + (pre - emit);
+      // deterministic base = scanned tile base + intra-wave exclusive
+      // prefix (wave rows are contiguous, so tile = lane-0 row / 64)
+      uint64_t pre = emit;
+      for (int off = 1; off < 64; off <<= 1) {
+        uint64_t t = __shfl_up(pre, off, 64);
+        if (lane >= off) pre += t;
       }
       if (emit == 0) continue;
+      uint64_t base =
+          (uint64_t)gptr<int64_t>(d.tileBases)[(row - lane) >> 6] +
+          (pre - emit);
       if (nullExt) {
         d.outBuild[base] = kHjNullRow;
         d.outProbe[base] = (uint32_t)row;
@@ -2049,6 +2062,10 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
       }
     } else {
       my += emit;
+      // per-tile emit total for the fill pass's deterministic bases
+      uint64_t tot = emit;
+      for (int off = 32; off > 0; off >>= 1) tot += __shfl_down(tot, off, 64);
+      if (lane == 0 && row < n) d.tileCounts[row >> 6] = (int64_t)tot;
     }
   }
   if (!FILL) {

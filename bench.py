@@ -596,9 +596,12 @@ def main():
                            reuse_chunk=bench_chunk)
         ex.close()
         if world > 1:
-            gathered = [None] * world
-            dist.all_gather_object(gathered, rows)
-            all_rows = [r for part in gathered for r in part]
+            # partial states travel as wire-codec bytes in DEVICE tensors
+            # (RCCL over xGMI; gloo host tensors in CPU smoke runs) —
+            # tidb_amd/dist.py, the ShuffleExec/partial-worker analog
+            from tidb_amd import dist as gxdist
+            all_rows = gxdist.gather_partial_rows(dist, lib, out_types,
+                                                  out_fracs, rows)
             result = merge_partials(lib, all_rows)
         else:
             result = {(r[0], r[1]): tuple(r[2:]) for r in rows}

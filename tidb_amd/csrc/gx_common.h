@@ -404,6 +404,26 @@ constexpr uint32_t kHjNullRow = 0xFFFFFFFFu;
 constexpr uint32_t kHjIneligible = 0xFFFFFFFFu;  // hits[]: pred-failed row
 constexpr uint32_t kHjMulti = 0xFFFFFFFEu;       // hits[]: >1 match
 
+// row-pack gather: the build side's fixed-width output columns pack into a
+// row-major staging buffer (one pass of sequential reads), so the random
+// per-output-row fetch touches ONE cache line per row instead of one line
+// PER COLUMN — the reference's row-table layout (join_row_table.go) applied
+// where it actually pays on MI355X: only for the random-access gather.
+struct RowPackDesc {
+  const void* src[kMaxCols];
+  void* dst[kMaxCols];
+  int32_t width[kMaxCols];  // 1 (dense char), 8, or 40
+  int32_t off[kMaxCols];    // byte offset inside the packed row
+  int32_t nCols = 0;
+  int32_t stride = 0;       // packed row bytes (8-aligned)
+  int64_t nRows = 0;        // pack: source rows
+  const uint32_t* idx = nullptr;  // unpack: match index (kHjNullRow -> zeros)
+  int64_t total = 0;        // unpack: output rows
+  uint8_t* staging = nullptr;
+};
+int gxPackRows(const RowPackDesc& d, void* stream);
+int gxUnpackRows(const RowPackDesc& d, void* stream);
+
 // phases: 0 = build (chain insert), 1 = count matches (+ matched flags),
 // 2 = fill match pairs, 3 = post filter, 4 = count unmatched build rows
 // (right outer), 5 = fill unmatched build rows

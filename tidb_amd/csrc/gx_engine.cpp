@@ -3114,6 +3114,28 @@ static int32_t runJoinAgg(gx_exec* ex) {
 // gather nCols columns of srcTab through a match/survivor index into
 // out->cols[dstBase..): fixed-width, dense char (identity offsets), general
 // varlen (two-pass), and null bitmaps
+// gather one column's null bitmap through the match index (allocates the
+// output bitmap; forceNulls covers null-extended outer-join rows over a
+// source with no bitmap)
+static int32_t gatherColNulls(gx_exec* ex, const gxp::DevCol& src,
+                              const uint32_t* idx, uint64_t total,
+                              gxp::DevCol* dst, bool forceNulls) {
+  if ((src.hasNulls && src.nullBitmap) || forceNulls) {
+    dst->nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8);
+    if (!dst->nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    if (gxp::gxGatherNulls(src.hasNulls ? src.nullBitmap : nullptr, idx,
+                           dst->nullBitmap, (int64_t)total, ex->stream) != 0) {
+      ex->err = "null gather launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    dst->hasNulls = 1;
+  } else {
+    dst->nullBitmap = nullptr;
+    dst->hasNulls = 0;
+  }
+  return GX_OK;
+}
+
 static int32_t gatherCols(gx_exec* ex, const gxp::DevTable& srcTab,
                           const uint32_t* idx, uint64_t total,
                           gxp::DevTable* out, int dstBase,
@@ -3175,20 +3197,85 @@ static int32_t gatherCols(gx_exec* ex, const gxp::DevTable& srcTab,
         dst.denseOffsets = 1;
       }
     }
-    if ((src.hasNulls && src.nullBitmap) || forceNulls) {
-      dst.nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8);
-      if (!dst.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-      if (gxp::gxGatherNulls(src.hasNulls ? src.nullBitmap : nullptr, idx,
-                             dst.nullBitmap, (int64_t)total,
-                             ex->stream) != 0) {
-        ex->err = "null gather launch failed";
+    int32_t rc = gatherColNulls(ex, src, idx, total, &dst, forceNulls);
+    if (rc) return rc;
+  }
+  return GX_OK;
+}
+
+// row-pack variant: >= 2 fixed-width columns pack row-major once (sequential
+// pass over the source), then the random per-output-row fetch touches one
+// packed row — one or two cache lines — instead of one line PER COLUMN.
+// General varlen columns and null bitmaps keep the per-column paths.
+static int32_t gatherColsAuto(gx_exec* ex, const gxp::DevTable& srcTab,
+                              const uint32_t* idx, uint64_t total,
+                              gxp::DevTable* out, int dstBase,
+                              bool forceNulls) {
+  gxp::RowPackDesc pd{};
+  int stride = 0;
+  std::vector<int> cols;
+  for (int pass = 0; pass < 2; pass++) {  // 8-aligned widths first, chars last
+    for (int c = 0; c < srcTab.nCols; c++) {
+      const gxp::DevCol& src = srcTab.cols[c];
+      int w = src.type == GX_TYPE_DECIMAL  ? 40
+              : src.type == GX_TYPE_STRING ? (src.denseOffsets ? 1 : 0)
+                                           : 8;
+      if (w == 0 || (pass == 0) != (w >= 8)) continue;
+      pd.src[pd.nCols] = src.data;
+      pd.width[pd.nCols] = w;
+      pd.off[pd.nCols] = stride;
+      cols.push_back(c);
+      stride += w;
+      pd.nCols++;
+    }
+  }
+  // worth packing only when the sequential pack pass is cheaper than the
+  // per-column random line fetches it saves
+  if (pd.nCols < 2 || srcTab.nRows > 2 * (int64_t)total)
+    return gatherCols(ex, srcTab, idx, total, out, dstBase, forceNulls);
+  pd.stride = (stride + 7) & ~7;
+  pd.nRows = srcTab.nRows;
+  pd.idx = idx;
+  pd.total = (int64_t)total;
+  pd.staging = (uint8_t*)devAlloc(ex, (size_t)pd.nRows * pd.stride + 16);
+  if (!pd.staging) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+  for (int k = 0; k < pd.nCols; k++) {
+    int c = cols[k];
+    const gxp::DevCol& src = srcTab.cols[c];
+    gxp::DevCol& dst = out->cols[dstBase + c];
+    dst.data = devAlloc(ex, (size_t)total * pd.width[k] + 16);
+    if (!dst.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    pd.dst[k] = dst.data;
+    if (src.type == GX_TYPE_STRING) {  // dense char(1)
+      dst.offsets = (int64_t*)devAlloc(ex, (total + 1) * 8);
+      if (!dst.offsets) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      if (gxp::gxIotaOffsets(dst.offsets, (int64_t)total + 1, ex->stream) != 0) {
+        ex->err = "offsets launch failed";
         return GX_ERR_INTERNAL;
       }
-      dst.hasNulls = 1;
-    } else {
-      dst.nullBitmap = nullptr;
-      dst.hasNulls = 0;
+      dst.denseOffsets = 1;
     }
+    int32_t rc = gatherColNulls(ex, src, idx, total, &dst, forceNulls);
+    if (rc) return rc;
+  }
+  if (gxp::gxPackRows(pd, ex->stream) != 0 ||
+      gxp::gxUnpackRows(pd, ex->stream) != 0) {
+    ex->err = "row-pack gather launch failed";
+    return GX_ERR_INTERNAL;
+  }
+  // remaining (general varlen) columns via the per-column path
+  for (int c = 0; c < srcTab.nCols; c++) {
+    const gxp::DevCol& src = srcTab.cols[c];
+    if (!(src.type == GX_TYPE_STRING && !src.denseOffsets)) continue;
+    gxp::DevTable one{};
+    one.nCols = 1;
+    one.cols[0] = src;
+    one.nRows = srcTab.nRows;
+    gxp::DevTable oneOut{};
+    oneOut.cols[0] = out->cols[dstBase + c];
+    int32_t rc = gatherCols(ex, one, idx, total, &oneOut, 0, forceNulls);
+    if (rc) return rc;
+    out->cols[dstBase + c] = oneOut.cols[0];
   }
   return GX_OK;
 }
@@ -3343,9 +3430,10 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
         // semi / anti semi: probe (outer) columns only
         rc = gatherCols(ex, psub, gatherP, total, &st.out, 0, false);
       } else {
-        // outer joins carry null-extended rows on the inner side
-        rc = gatherCols(ex, bsub, gatherB, total, &st.out, 0,
-                        hj.joinType == 1);
+        // outer joins carry null-extended rows on the inner side; the build
+        // side (the random-access gather) goes through the row-pack path
+        rc = gatherColsAuto(ex, bsub, gatherB, total, &st.out, 0,
+                            hj.joinType == 1);
         if (rc == GX_OK)
           rc = gatherCols(ex, psub, gatherP, total, &st.out, hj.build.nCols,
                           hj.joinType == 2);

@@ -2189,6 +2189,71 @@ __global__ void hjUnmatchedKernel(const HashJoinDesc* __restrict__ dp) {
   }
 }
 
+// ---- row-pack gather (gx_common.h RowPackDesc) ----
+__global__ void hjPackRowsKernel(RowPackDesc d) {
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < d.nRows; row += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t* out = d.staging + row * (int64_t)d.stride;
+    for (int c = 0; c < d.nCols; c++) {
+      if (d.width[c] == 8) {
+        *(uint64_t*)(out + d.off[c]) = gptr<uint64_t>(d.src[c])[row];
+      } else if (d.width[c] == 1) {
+        out[d.off[c]] = gptr<uint8_t>(d.src[c])[row];
+      } else {  // 40
+        auto s = gptr<uint64_t>((const uint8_t*)d.src[c] + row * 40);
+        uint64_t* o = (uint64_t*)(out + d.off[c]);
+        o[0] = s[0]; o[1] = s[1]; o[2] = s[2]; o[3] = s[3]; o[4] = s[4];
+      }
+    }
+  }
+}
+
+__global__ void hjUnpackRowsKernel(RowPackDesc d) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < d.total; i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t src = gptr<uint32_t>(d.idx)[i];
+    if (src == kHjNullRow) {  // null-extended outer-join row -> zeros
+      for (int c = 0; c < d.nCols; c++) {
+        if (d.width[c] == 8) ((uint64_t*)d.dst[c])[i] = 0;
+        else if (d.width[c] == 1) ((uint8_t*)d.dst[c])[i] = 0;
+        else {
+          uint64_t* o = (uint64_t*)((uint8_t*)d.dst[c] + i * 40);
+          o[0] = 0; o[1] = 0; o[2] = 0; o[3] = 0; o[4] = 0;
+        }
+      }
+      continue;
+    }
+    auto in = gptr<uint8_t>(d.staging + (int64_t)src * d.stride);
+    for (int c = 0; c < d.nCols; c++) {
+      if (d.width[c] == 8) {
+        ((uint64_t*)d.dst[c])[i] = *(const __attribute__((address_space(1)))
+                                         uint64_t*)(in + d.off[c]);
+      } else if (d.width[c] == 1) {
+        ((uint8_t*)d.dst[c])[i] = in[d.off[c]];
+      } else {
+        auto s = (const __attribute__((address_space(1))) uint64_t*)(in +
+                                                                     d.off[c]);
+        uint64_t* o = (uint64_t*)((uint8_t*)d.dst[c] + i * 40);
+        o[0] = s[0]; o[1] = s[1]; o[2] = s[2]; o[3] = s[3]; o[4] = s[4];
+      }
+    }
+  }
+}
+
+int gxPackRows(const RowPackDesc& d, void* stream) {
+  if (d.nRows == 0) return 0;
+  hipLaunchKernelGGL(hjPackRowsKernel, dim3(gridFor(d.nRows)), dim3(256), 0,
+                     (hipStream_t)stream, d);
+  return (int)hipGetLastError();
+}
+
+int gxUnpackRows(const RowPackDesc& d, void* stream) {
+  if (d.total == 0) return 0;
+  hipLaunchKernelGGL(hjUnpackRowsKernel, dim3(gridFor(d.total)), dim3(256), 0,
+                     (hipStream_t)stream, d);
+  return (int)hipGetLastError();
+}
+
 // standalone Selection (SelectionExec, select.go:750-785): compact the rows
 // surviving a CNF into a survivor index, MI355X-shaped — no row-at-a-time
 // AppendRow copy; output columns gather through the index afterwards.

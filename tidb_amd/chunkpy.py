@@ -7,7 +7,8 @@ import ctypes
 
 import numpy as np
 
-from tests.gxlib import GxChunk, GxCol, GX_TYPE_DECIMAL, GX_TYPE_STRING
+from tests.gxlib import (GxChunk, GxCol, GX_TYPE_DECIMAL, GX_TYPE_F64,
+                         GX_TYPE_STRING)
 
 
 class PyColumn:
@@ -85,6 +86,9 @@ class PyChunk:
             return bytes(c.data[s:e]).decode("utf8", "replace")
         if c.typ == GX_TYPE_DECIMAL:
             return decimal_bytes_to_str(bytes(c.data[row * 40:(row + 1) * 40]))
+        if c.typ == GX_TYPE_F64:
+            import struct
+            return struct.unpack("<d", bytes(c.data[row * 8:(row + 1) * 8]))[0]
         v = int.from_bytes(bytes(c.data[row * 8:(row + 1) * 8]), "little", signed=True)
         return v
 
@@ -107,6 +111,10 @@ class PyChunk:
                     c.offsets[i + 1] = s + len(b)
                 elif c.typ == GX_TYPE_DECIMAL:
                     c.data[i * 40:(i + 1) * 40] = np.frombuffer(v, dtype=np.uint8)
+                elif c.typ == GX_TYPE_F64:
+                    import struct
+                    c.data[i * 8:(i + 1) * 8] = np.frombuffer(
+                        struct.pack("<d", float(v)), dtype=np.uint8)
                 else:
                     if v < 0:
                         v += 1 << 64

@@ -117,6 +117,10 @@ struct AggDesc {
   int32_t scale;  // scale of the accumulated units
   int32_t fr = -1;  // FIRSTROW over a group-by column: index into gkey cols
                     // (value decoded from the group key; no per-row state)
+  int32_t fcol = -1;  // f64 sum/avg: source column (bypasses the integer VM;
+                      // the value loads raw through a fetch slot and
+                      // accumulates with f64 atomics — oracle/exec.cpp f64
+                      // path, stated-tolerance parity)
 };
 
 constexpr int kMaxAggs = 12;
@@ -212,8 +216,11 @@ struct FusedQueryDesc {
   // phys slot accumulation kind: 0 = int128 sum, 1 = max over the
   // order-preserving biased-u64 encoding (min stores the complement, so
   // BOTH min and max accumulate with unsigned max from a zero-initialized
-  // table; func_max_min.go semantics, narrow int64 values only)
+  // table; func_max_min.go semantics, narrow int64 values only),
+  // 3 = f64 sum (accLo holds the double bits; accReg = raw fetch slot,
+  // accFcol = source column for the NULL check)
   int32_t accKind[kMaxAggs];
+  int32_t accFcol[kMaxAggs];
   int32_t sharedCnt = 0;
   int32_t hasDiv = 0;
   int32_t nVmRegs = 0;  // registers the compiled VM uses (>12 selects the

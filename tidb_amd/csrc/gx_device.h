@@ -652,6 +652,13 @@ __device__ inline void lds3AccumAcc(Lds3GroupSlot* slot, int s, Int128 v) {
   }
 }
 
+typedef __attribute__((address_space(3))) double Lds3F64;
+
+__device__ inline void lds3AccumF64(Lds3GroupSlot* slot, int s, double v) {
+  __hip_atomic_fetch_add((Lds3F64*)&slot->accLo[s], v, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+
 __device__ inline void lds3AccumCnt(Lds3GroupSlot* slot, int a, int64_t dc) {
   if (dc != 0)
     __hip_atomic_fetch_add((Lds3U64*)&slot->cnt[a], (uint64_t)dc,

@@ -1823,6 +1823,27 @@ static int32_t compileFused(gx_exec* ex) {
     if (argE < 0) {
       ad.srcReg = -1;
       ad.scale = 0;
+    } else if (plan.exprs[argE].retType == GX_TYPE_F64) {
+      // f64 sum/avg (oracle/exec.cpp f64 path; stated-tolerance parity):
+      // the value bypasses the integer VM — raw bits load through a fetch
+      // slot and accumulate with f64 atomics (atomic order makes the sum
+      // round differently run to run; parity tests carry the tolerance)
+      const PExpr& e = plan.exprs[argE];
+      int srcColF = -1;
+      if (e.kind == EK_COLREF) {
+        if (proj) {
+          if (e.colIdx < (int)projSrcCol.size()) srcColF = projSrcCol[e.colIdx];
+        } else {
+          srcColF = e.colIdx;
+        }
+      }
+      if ((ad.func != GX_AGG_SUM && ad.func != GX_AGG_AVG) || srcColF < 0) {
+        ex->err = "device f64 aggregates support sum/avg over a float column";
+        return GX_ERR_INVALID;
+      }
+      ad.fcol = srcColF;
+      ad.srcReg = -1;
+      ad.scale = 0;
     } else {
       const PExpr& e = plan.exprs[argE];
       if (proj) {
@@ -1848,6 +1869,24 @@ static int32_t compileFused(gx_exec* ex) {
   ex->desc.nAccSlots = 0;
   for (int a = 0; a < ex->desc.nAggs; a++) {
     const gxp::AggDesc& ad = ex->desc.aggs[a];
+    if (ad.fcol >= 0) {  // f64 sum/avg: slot keyed by the source column
+      int found = -1;
+      for (int s = 0; s < ex->desc.nAccSlots; s++)
+        if (ex->desc.accKind[s] == 3 && ex->desc.accFcol[s] == ad.fcol) {
+          found = s;
+          break;
+        }
+      if (found < 0) {
+        int slot = fetchSlot(ex, gxp::FETCH_8B, ad.fcol);
+        if (slot < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+        found = ex->desc.nAccSlots++;
+        ex->desc.accReg[found] = slot;
+        ex->desc.accKind[found] = 3;
+        ex->desc.accFcol[found] = ad.fcol;
+      }
+      ex->desc.accMap[a] = found;
+      continue;
+    }
     if (ad.func == GX_AGG_COUNT || ad.srcReg < 0) {
       ex->desc.accMap[a] = -1;
       continue;
@@ -1873,6 +1912,7 @@ static int32_t compileFused(gx_exec* ex) {
       found = ex->desc.nAccSlots++;
       ex->desc.accReg[found] = ad.srcReg;
       ex->desc.accKind[found] = kind;
+      ex->desc.accFcol[found] = -1;
     }
     ex->desc.accMap[a] = found;
   }
@@ -2726,6 +2766,21 @@ static int32_t runFused(gx_exec* ex) {
         v.type = GX_TYPE_I64;
         v.i64 = cnt;
         row.push_back(std::move(v));
+      } else if (phys >= 0 && ex->desc.accKind[phys] == 3) {  // f64 sum/avg
+        double sum;
+        std::memcpy(&sum, &s->accLo[phys], 8);
+        OutRowVal v;
+        v.type = GX_TYPE_F64;
+        if (cnt == 0) v.isNull = true;
+        else v.f64 = ad.func == GX_AGG_AVG && !partial ? sum / (double)cnt
+                                                       : sum;
+        row.push_back(std::move(v));
+        if (partial) {
+          OutRowVal c2;
+          c2.type = GX_TYPE_I64;
+          c2.i64 = cnt;
+          row.push_back(std::move(c2));
+        }
       } else if (ad.func == GX_AGG_MIN || ad.func == GX_AGG_MAX) {
         // biased-u64 extreme (accKind 1/2); NULL when no non-null arg row
         OutRowVal v;
@@ -3745,6 +3800,7 @@ static int32_t runFinalHost(gx_exec* ex) {
     std::vector<int64_t> cnt;
     std::vector<OutRowVal> val;   // min/max/firstrow merged value
     std::vector<uint8_t> has;
+    std::vector<double> f64;      // f64 sum partials
   };
   // decode one partial cell (min/max/firstrow value columns)
   auto readCell = [](const HostCol& hc, int i) {
@@ -3827,6 +3883,7 @@ static int32_t runFinalHost(gx_exec* ex) {
         st.cnt.assign(agg.aggFuncs.size(), 0);
         st.val.resize(agg.aggFuncs.size());
         st.has.assign(agg.aggFuncs.size(), 0);
+        st.f64.assign(agg.aggFuncs.size(), 0.0);
         git = groups.emplace(key, std::move(st)).first;
         order.push_back(key);
       }
@@ -3860,6 +3917,16 @@ static int32_t runFinalHost(gx_exec* ex) {
             st.has[a] = 1;
           }
           col += 1;
+        } else if (ch[col].type == GX_TYPE_F64) {  // f64 SUM/AVG partials
+          int64_t c;
+          std::memcpy(&c, ch[col + 1].data.data() + i * 8, 8);
+          if (c > 0) {
+            double v;
+            std::memcpy(&v, ch[col].data.data() + i * 8, 8);
+            st.f64[a] += v;
+            st.cnt[a] += c;
+          }
+          col += 2;
         } else {  // SUM/AVG: decimal + count
           int64_t c;
           std::memcpy(&c, ch[col + 1].data.data() + i * 8, 8);
@@ -3898,6 +3965,12 @@ static int32_t runFinalHost(gx_exec* ex) {
         } else {
           v = st.val[a];
         }
+      } else if (agg.aggArgs[a] >= 0 &&
+                 ex->plan.exprs[agg.aggArgs[a]].retType == GX_TYPE_F64) {
+        v.type = GX_TYPE_F64;
+        if (st.cnt[a] == 0) v.isNull = true;
+        else v.f64 = f == GX_AGG_AVG ? st.f64[a] / (double)st.cnt[a]
+                                     : st.f64[a];
       } else if (f == GX_AGG_SUM) {
         v.type = GX_TYPE_DECIMAL;
         if (st.cnt[a] == 0) v.isNull = true;

@@ -540,6 +540,11 @@ static const JitProg* compileSource(const std::string& src, const char* fn1,
 }
 
 const JitProg* compile(const FusedQueryDesc& d, std::string* whyNot) {
+  for (int s = 0; s < d.nAccSlots; s++)
+    if (d.accKind[s] == 3) {  // f64 atomics: interpreted kernel only
+      if (whyNot) *whyNot = "f64 accumulator not specialized";
+      return nullptr;
+    }
   return compileSource(generateSource(d), "genq_narrow", "genq_wide", whyNot);
 }
 

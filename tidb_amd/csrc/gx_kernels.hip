@@ -389,6 +389,12 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     if (d.sharedCnt) lds3AccumCnt(target, 0, 1);
     for (int s = 0; s < d.nAccSlots; s++) {
       int reg = d.accReg[s];
+      if (d.accKind[s] == 3) {  // f64 sum: raw column bits, f64 LDS atomic
+        const DevCol& fc = d.table.cols[d.accFcol[s]];
+        if (colIsNull(fc, row)) continue;
+        lds3AccumF64(target, s, __builtin_bit_cast(double, raw.get(reg).x));
+        continue;
+      }
       if (vm.isNull(reg)) continue;
       if (d.accKind[s] == 0) {
         lds3AccumAcc(target, s, VT<WIDE>::toAcc(vm.get(reg)));
@@ -408,7 +414,9 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
       for (int a = 0; a < d.nAggs; a++) {
         const AggDesc& ad = d.aggs[a];
         if (ad.fr >= 0) continue;  // firstrow(group col): no per-row state
-        bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
+        bool isNull = ad.fcol >= 0
+                          ? colIsNull(d.table.cols[ad.fcol], row)
+                          : (ad.srcReg >= 0 && vm.isNull(ad.srcReg));
         if (!isNull) lds3AccumCnt(target, a, 1);
       }
     }
@@ -441,6 +449,13 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   if (d.sharedCnt) accumInto(target, 0, Int128{0, 0}, 1);  // bumps cnt[0] only
   for (int s = 0; s < d.nAccSlots; s++) {
     int reg = d.accReg[s];
+    if (d.accKind[s] == 3) {  // f64 sum
+      const DevCol& fc = d.table.cols[d.accFcol[s]];
+      if (colIsNull(fc, row)) continue;
+      atomicAdd((double*)&target->accLo[s],
+                __builtin_bit_cast(double, raw.get(reg).x));
+      continue;
+    }
     if (vm.isNull(reg)) continue;
     if (d.accKind[s] == 0) {
       accumInto(target, s, VT<WIDE>::toAcc(vm.get(reg)), 0);
@@ -460,7 +475,9 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
     for (int a = 0; a < d.nAggs; a++) {
       const AggDesc& ad = d.aggs[a];
       if (ad.fr >= 0) continue;  // firstrow(group col): no per-row state
-      bool isNull = ad.srcReg >= 0 && vm.isNull(ad.srcReg);
+      bool isNull = ad.fcol >= 0
+                        ? colIsNull(d.table.cols[ad.fcol], row)
+                        : (ad.srcReg >= 0 && vm.isNull(ad.srcReg));
       if (!isNull) accumInto(target, a, Int128{0, 0}, 1);
     }
   }
@@ -553,6 +570,11 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
     }
     if (!ok) continue;
     for (int s = 0; s < d.nAccSlots; s++) {
+      if (d.accKind[s] == 3) {  // f64 sum: add the LDS partial
+        atomicAdd((double*)&d.globalTable[slot].accLo[s],
+                  __builtin_bit_cast(double, lds[i].accLo[s]));
+        continue;
+      }
       if (d.accKind[s] != 0) {
         accumMax(&d.globalTable[slot], s, lds[i].accLo[s]);
         continue;

@@ -8,7 +8,7 @@ from tests.gxlib import load_oracle, load_product
 from tidb_amd.chunkpy import PyChunk
 from tidb_amd.decimals import str_to_decimal_bytes
 
-TYPES = [1, 2, 4, 3]  # i64, decimal, string, time
+TYPES = [0, 2, 4, 3]  # i64, decimal, string, time
 FRACS = [0, 2, 0, 0]
 
 

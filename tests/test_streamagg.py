@@ -53,7 +53,7 @@ def test_streamagg_oracle(oracle_lib):
     ex = b.build(agg)
     ex.bind_tpch(src, GX_TPCH_LINEITEM, 20000)
     ex.open()
-    want = ex.pull_all([4, 4, 2, 1], [0, 0, 2, 0],
+    want = ex.pull_all([4, 4, 2, 0], [0, 0, 2, 0],
                        data_caps=[2048, 2048, None, None])
     ex.close(); ex.free(); b.free()
     assert sorted(rows) == sorted(want)

@@ -83,7 +83,18 @@ enum {
    * datum.go:1629): CAST_DEC rounds HalfUp to the call's ret_frac (flen
    * clamping unimplemented this round); CAST_INT = Round(0, HalfUp) + ToInt
    * (builtin_cast_vec.go:1817-1852). */
-  GX_F_CAST_DEC = 20, GX_F_CAST_INT = 21
+  GX_F_CAST_DEC = 20, GX_F_CAST_INT = 21,
+  /* string builtins (pkg/expression/builtin_string_vec.go), binary/byte
+   * semantics (the non-UTF8 sigs; test data is ASCII where they coincide):
+   * LENGTH  (builtinLengthSig): byte length -> i64; NULL propagates.
+   * SUBSTR  (builtinSubstring3ArgsSig): args (str, pos i64 const, len i64
+   *          const); MySQL 1-based pos, negative pos counts from the end,
+   *          pos 0 / |pos|>len / len<=0 -> empty string.
+   * LIKE_PREFIX (builtinLikeSig's 'abc%' fast path): args (str, const
+   *          prefix WITHOUT the trailing %); 1 iff str starts with prefix
+   *          (binary collation, case-sensitive); NULL str -> NULL.
+   * UPPER   (builtinUpperSig): ASCII a-z upcased, other bytes unchanged. */
+  GX_F_LENGTH = 32, GX_F_SUBSTR = 33, GX_F_LIKE_PREFIX = 34, GX_F_UPPER = 35
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

@@ -222,6 +222,75 @@ int32_t evalArith(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
   return GX_ERR_INVALID;
 }
 
+// string builtins (builtin_string_vec.go, byte/binary sigs — see
+// gx_executor.h GX_F_LENGTH.. doc)
+static int32_t evalString(EvalCtx& ctx, const Expr& e, const Chunk& in,
+                          Column& out) {
+  Column a;
+  int32_t err = evalVec(ctx, e.args[0], in, a);
+  if (err) return err;
+  int n = in.numRows();
+  out.reset();
+  if (e.func == GX_F_LENGTH) {  // builtinLengthSig: byte length
+    out.type = GX_TYPE_I64;
+    for (int i = 0; i < n; i++) {
+      if (a.isNull(i)) { out.appendNull(); continue; }
+      int len;
+      a.getBytes(i, &len);
+      out.appendI64(len);
+    }
+    return GX_OK;
+  }
+  if (e.func == GX_F_UPPER) {  // builtinUpperSig (ASCII)
+    out.type = GX_TYPE_STRING;
+    out.offsets.assign(1, 0);
+    for (int i = 0; i < n; i++) {
+      if (a.isNull(i)) { out.appendNull(); continue; }
+      int len;
+      const uint8_t* p = a.getBytes(i, &len);
+      std::string s((const char*)p, len);
+      for (char& c : s)
+        if (c >= 'a' && c <= 'z') c = (char)(c - 'a' + 'A');
+      out.appendBytes(s.data(), s.size());
+    }
+    return GX_OK;
+  }
+  if (e.func == GX_F_LIKE_PREFIX) {  // builtinLikeSig 'abc%' fast path
+    const Expr& pat = ctx.plan->exprs[e.args[1]];
+    out.type = GX_TYPE_I64;
+    for (int i = 0; i < n; i++) {
+      if (a.isNull(i)) { out.appendNull(); continue; }
+      int len;
+      const uint8_t* p = a.getBytes(i, &len);
+      bool m = (size_t)len >= pat.constStr.size() &&
+               std::memcmp(p, pat.constStr.data(), pat.constStr.size()) == 0;
+      out.appendI64(m ? 1 : 0);
+    }
+    return GX_OK;
+  }
+  if (e.func == GX_F_SUBSTR) {  // builtinSubstring3ArgsSig (byte semantics)
+    int64_t pos = ctx.plan->exprs[e.args[1]].constI64;
+    int64_t length = ctx.plan->exprs[e.args[2]].constI64;
+    out.type = GX_TYPE_STRING;
+    out.offsets.assign(1, 0);
+    for (int i = 0; i < n; i++) {
+      if (a.isNull(i)) { out.appendNull(); continue; }
+      int len;
+      const uint8_t* p = a.getBytes(i, &len);
+      int64_t start = pos;
+      if (start < 0) start = (int64_t)len + start + 1;
+      if (start < 1 || start > len || length <= 0) {
+        out.appendBytes(p, 0);
+        continue;
+      }
+      int64_t l = std::min<int64_t>(length, len - (start - 1));
+      out.appendBytes(p + (start - 1), (size_t)l);
+    }
+    return GX_OK;
+  }
+  return GX_ERR_INVALID;
+}
+
 int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
   const Expr& e = ctx.plan->exprs[exprId];
   switch (e.kind) {
@@ -235,6 +304,8 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       if (e.func <= GX_F_NE) return evalCompare(ctx, e, in, out);
       if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT)
         return evalCast(ctx, e, in, out);
+      if (e.func >= GX_F_LENGTH && e.func <= GX_F_UPPER)
+        return evalString(ctx, e, in, out);
       return evalArith(ctx, e, in, out);
   }
   return GX_ERR_INVALID;

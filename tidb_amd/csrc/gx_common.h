@@ -53,6 +53,8 @@ enum VmOp : int32_t {
   VM_ROUND_SCALE = 8,  // dst <- round_half_up(a, scale b -> scale... encoded:
                        // ins.b = target scale, ins.c = source scale (the
                        // cast family, ProduceDecWithSpecifiedTp/ToInt)
+  VM_STRLEN = 9,       // dst <- byte length of string column a
+                       // (builtinLengthSig; offsets fetch slot in c)
 };
 
 struct VmIns {
@@ -319,6 +321,7 @@ struct JoinAggDesc {
 };
 
 enum { PRED_STR_EQ_CONST = 3 };  // extra PredKind for the join path
+enum { PRED_STR_LIKE_PREFIX = 6 };  // LIKE 'abc%' fast path (builtinLikeSig)
 
 // one conjunct of a post-join filter (NULL operand rejects the row, the
 // VecEvalBool NULL semantics, expression.go:420-504)
@@ -450,6 +453,23 @@ int gxSortU32Keys(const uint32_t* in, uint32_t* out, int64_t n, void* tmp,
 // DEVICE (canonical word layout, mydecimal.go:236-248); passthrough column
 // references alias the source buffers (zero copy). Null flags write as one
 // byte per row, packed to the LSB-first bitmap in a second pass.
+// string projection program (builtin_string_vec.go SUBSTR/UPPER over a
+// string column): windows compose into ONE (start,len) view per row —
+// no intermediate materialization; bytes copy once at emit (with optional
+// ASCII upcase). LENGTH rides the integer VM (VM_STRLEN) instead.
+constexpr int kMaxStrWin = 4;
+struct StrProg {
+  int32_t col = -1;
+  int32_t nWin = 0;
+  int64_t winPos[kMaxStrWin];  // MySQL 1-based; negative counts from the end
+  int64_t winLen[kMaxStrWin];
+  int32_t upper = 0;
+  // per-run device temps (engine-allocated)
+  int64_t* starts = nullptr;
+  int64_t* lens = nullptr;
+  uint8_t* notNull = nullptr;
+};
+
 struct ProjDesc {
   DevTable table;
   VmIns ins[kMaxVmIns];
@@ -467,9 +487,17 @@ struct ProjDesc {
   uint8_t* outNotNull[kMaxCols];  // byte per row, 1 = NOT NULL
   uint32_t* errorFlag = nullptr;
   int32_t wide = 0;
+  StrProg sprog[kMaxCols];
+  int32_t nSprog = 0;
 };
 
 int gxProject(const ProjDesc* devDesc, const ProjDesc& h, void* stream);
+// string program: pass 1 computes per-row (start,len,notNull) views;
+// pass 2 copies the bytes through the scanned offsets (ASCII upcase opt.)
+int gxStrWindow(const ProjDesc* devDesc, const ProjDesc& h, int progIdx,
+                void* stream);
+int gxStrEmit(const ProjDesc* devDesc, const ProjDesc& h, int progIdx,
+              const int64_t* outOffsets, uint8_t* outData, void* stream);
 // pack byte-per-row not-null flags into the LSB-first bitmap
 int gxPackNulls(const uint8_t* notNullBytes, uint8_t* bitmap, int64_t n,
                 void* stream);

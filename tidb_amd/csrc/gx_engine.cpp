@@ -233,6 +233,7 @@ struct gx_exec {
   gxp::ProjDesc* devPd = nullptr;
   std::vector<int> projOutTypes, projOutFracs;
   std::vector<int> projOutSrcCol;  // >=0 passthrough; -1 computed
+  std::vector<int> projOutSprog;   // >=0 string program; -1 otherwise
   std::vector<int> projOutSlot;    // computed: index into pd.out*
 
   ~gx_exec() {
@@ -514,6 +515,22 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       break;
     }
     case EK_CALL: {
+      if (e.func == GX_F_LENGTH) {  // builtinLengthSig: byte length -> i64
+        if (e.args.size() != 1 ||
+            ex->plan.exprs[e.args[0]].kind != EK_COLREF) {
+          ex->err = "device LENGTH takes one string column";
+          return -1;
+        }
+        int col = ex->plan.exprs[e.args[0]].colIdx;
+        if (col < 0 || col >= ex->desc.table.nCols ||
+            ex->desc.table.cols[col].type != GX_TYPE_STRING) {
+          ex->err = "device LENGTH argument must be a string column";
+          return -1;
+        }
+        reg = emit(gxp::VM_STRLEN, allocRegFresh(), col, 0);
+        *scaleOut = 0;
+        break;
+      }
       if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT) {
         if (e.args.size() != 1) {
           ex->err = "cast takes one argument";
@@ -705,6 +722,22 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
       break;
     }
     case EK_CALL: {
+      if (e.func == GX_F_LENGTH) {  // builtinLengthSig: byte length -> i64
+        if (e.args.size() != 1 ||
+            ex->plan.exprs[e.args[0]].kind != EK_COLREF) {
+          ex->err = "device LENGTH takes one string column";
+          return -1;
+        }
+        int col = ex->plan.exprs[e.args[0]].colIdx - B.colBase;
+        if (col < 0 || col >= (int)B.colTypes->size() ||
+            (*B.colTypes)[col] != GX_TYPE_STRING) {
+          ex->err = "device LENGTH argument must be a string column";
+          return -1;
+        }
+        reg = emit(gxp::VM_STRLEN, allocReg(), col, 0, -1);
+        *scaleOut = 0;
+        break;
+      }
       if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT) {
         // cast family (ProduceDecWithSpecifiedTp / ToInt): round half-up
         // from the arg's scale to the target scale (VM_ROUND_SCALE)
@@ -767,6 +800,27 @@ static bool compileTablePred(gx_exec* ex, const PNode& srcNode, int condId,
                              gxp::PredDesc* out, uint8_t* strConst,
                              int32_t* strConstLen) {
   const PExpr& e = ex->plan.exprs[condId];
+  if (e.kind == EK_CALL && e.func == GX_F_LIKE_PREFIX && e.args.size() == 2) {
+    // LIKE 'abc%' fast path (builtinLikeSig): byte prefix, no PAD trimming
+    const PExpr& col = ex->plan.exprs[e.args[0]];
+    const PExpr& pat = ex->plan.exprs[e.args[1]];
+    if (col.kind != EK_COLREF || pat.kind != EK_CONST ||
+        col.colIdx < 0 || col.colIdx >= (int)srcNode.colTypes.size() ||
+        srcNode.colTypes[col.colIdx] != GX_TYPE_STRING ||
+        pat.constStr.size() > 16) {
+      ex->err = "device LIKE takes <string column> LIKE <'prefix%' <= 16B>";
+      return false;
+    }
+    gxp::PredDesc pd{};
+    pd.kind = gxp::PRED_STR_LIKE_PREFIX;
+    pd.col = col.colIdx;
+    pd.cmp = GX_F_EQ;
+    pd.slot = -1;
+    std::memcpy(strConst, pat.constStr.data(), pat.constStr.size());
+    *strConstLen = (int32_t)pat.constStr.size();
+    *out = pd;
+    return true;
+  }
   if (e.kind != EK_CALL || e.func > GX_F_NE || e.args.size() != 2) {
     ex->err = "unsupported filter expression on device";
     return false;
@@ -1368,6 +1422,39 @@ static int32_t compileSelect(gx_exec* ex, int selNode) {
   return GX_OK;
 }
 
+// recognize a {SUBSTR, UPPER}* chain over a string column and fold it into
+// one StrProg (windows compose; ASCII upcase commutes with windowing)
+static bool compileStrProg(gx_exec* ex, int exprId,
+                           const std::vector<int>& types, gxp::StrProg* sp) {
+  const PExpr& e = ex->plan.exprs[exprId];
+  if (e.kind == EK_COLREF) {
+    if (e.colIdx < 0 || e.colIdx >= (int)types.size() ||
+        types[e.colIdx] != GX_TYPE_STRING)
+      return false;
+    sp->col = e.colIdx;
+    return true;
+  }
+  if (e.kind != EK_CALL) return false;
+  if (e.func == GX_F_UPPER && e.args.size() == 1) {
+    if (!compileStrProg(ex, e.args[0], types, sp)) return false;
+    sp->upper = 1;
+    return true;
+  }
+  if (e.func == GX_F_SUBSTR && e.args.size() == 3) {
+    if (!compileStrProg(ex, e.args[0], types, sp)) return false;
+    const PExpr& p = ex->plan.exprs[e.args[1]];
+    const PExpr& l = ex->plan.exprs[e.args[2]];
+    if (p.kind != EK_CONST || l.kind != EK_CONST ||
+        sp->nWin >= gxp::kMaxStrWin)
+      return false;
+    sp->winPos[sp->nWin] = p.constI64;
+    sp->winLen[sp->nWin] = l.constI64;
+    sp->nWin++;
+    return true;
+  }
+  return false;
+}
+
 // standalone Projection over a Source or a Selection(Source)
 // (ProjectionExec, projection.go:77): computed expressions materialize as
 // device columns (decimal encode on device), passthrough colrefs alias the
@@ -1425,8 +1512,24 @@ static int32_t compileProject(gx_exec* ex) {
       }
       ex->projOutSrcCol.push_back(e.colIdx);
       ex->projOutSlot.push_back(-1);
+      ex->projOutSprog.push_back(-1);
       ex->projOutTypes.push_back((*childTypes)[e.colIdx]);
       ex->projOutFracs.push_back((*childFracs)[e.colIdx]);
+    } else if (e.kind == EK_CALL && e.retType == GX_TYPE_STRING) {
+      // string outputs: SUBSTR/UPPER chains fold into one windowed view
+      gxp::StrProg sp{};
+      if (!compileStrProg(ex, eid, *childTypes, &sp)) {
+        ex->err = "unsupported string projection (SUBSTR/UPPER chains over "
+                  "a string column this round)";
+        return GX_ERR_INVALID;
+      }
+      int si = pd.nSprog++;
+      pd.sprog[si] = sp;
+      ex->projOutSrcCol.push_back(-1);
+      ex->projOutSlot.push_back(-1);
+      ex->projOutSprog.push_back(si);
+      ex->projOutTypes.push_back(GX_TYPE_STRING);
+      ex->projOutFracs.push_back(0);
     } else {
       int sc = 0;
       int reg = vmCompile(ex, B, eid, &sc);
@@ -1441,6 +1544,7 @@ static int32_t compileProject(gx_exec* ex) {
       pd.outType[o] = outType;
       ex->projOutSrcCol.push_back(-1);
       ex->projOutSlot.push_back(o);
+      ex->projOutSprog.push_back(-1);
       ex->projOutTypes.push_back(outType);
       ex->projOutFracs.push_back(outType == GX_TYPE_I64 ? 0 : sc);
     }
@@ -1931,6 +2035,10 @@ static int32_t compileFused(gx_exec* ex) {
         loads.push_back(ins);
       } else if (ins.op == gxp::VM_LOAD_I64) {
         ins.c = fetchSlot(ex, gxp::FETCH_8B, ins.a);
+        if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
+        loads.push_back(ins);
+      } else if (ins.op == gxp::VM_STRLEN) {
+        ins.c = fetchSlot(ex, gxp::FETCH_OFFSETS, ins.a);
         if (ins.c < 0) { ex->err = "fetch plan full"; return GX_ERR_INVALID; }
         loads.push_back(ins);
       } else {
@@ -3724,6 +3832,18 @@ static int32_t runProject(gx_exec* ex) {
       return GX_ERR_INTERNAL;
     }
   }
+  // string-program temps (allocated before the devPd upload so the window
+  // kernels see the pointers)
+  for (int si = 0; si < pd.nSprog && n > 0; si++) {
+    gxp::StrProg& sp = pd.sprog[si];
+    sp.starts = (int64_t*)devAlloc(ex, (size_t)n * 8);
+    sp.lens = (int64_t*)devAlloc(ex, (size_t)n * 8);
+    sp.notNull = (uint8_t*)devAlloc(ex, (size_t)n);
+    if (!sp.starts || !sp.lens || !sp.notNull) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+  }
   for (int attempt = 0; n > 0 && attempt < 2; attempt++) {
     HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
     HIP_OK(ex, hipMemcpyAsync(ex->devPd, &pd, sizeof(pd),
@@ -3752,6 +3872,46 @@ static int32_t runProject(gx_exec* ex) {
       return GX_ERR_INTERNAL;
     }
   }
+  // string programs: window views -> scanned offsets -> byte emit
+  std::vector<int64_t*> spOffsets(pd.nSprog, nullptr);
+  std::vector<void*> spData(pd.nSprog, nullptr);
+  std::vector<uint8_t*> spBitmaps(pd.nSprog, nullptr);
+  for (int si = 0; si < pd.nSprog && n > 0; si++) {
+    gxp::StrProg& sp = pd.sprog[si];
+    if (gxp::gxStrWindow(ex->devPd, pd, si, ex->stream) != 0) {
+      ex->err = "string window launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    spOffsets[si] = (int64_t*)devAlloc(ex, ((size_t)n + 1) * 8);
+    if (!spOffsets[si]) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    size_t tmpBytes = 0;
+    gxp::gxExclusiveSumI64(sp.lens, spOffsets[si], n, nullptr, &tmpBytes,
+                           ex->stream);
+    void* tmp = devAlloc(ex, std::max<size_t>(tmpBytes, 1));
+    if (!tmp) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    HIP_OK(ex, hipMemsetAsync(spOffsets[si], 0, 8, ex->stream));
+    if (gxp::gxExclusiveSumI64(sp.lens, spOffsets[si], n, tmp, &tmpBytes,
+                               ex->stream) != 0) {
+      ex->err = "string offsets scan failed";
+      return GX_ERR_INTERNAL;
+    }
+    int64_t totalBytes = 0;
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    HIP_OK(ex, hipMemcpy(&totalBytes, spOffsets[si] + n, 8,
+                         hipMemcpyDeviceToHost));
+    spData[si] = devAlloc(ex, std::max<int64_t>(totalBytes, 1));
+    spBitmaps[si] = (uint8_t*)devAlloc(ex, (n + 7) / 8);
+    if (!spData[si] || !spBitmaps[si]) {
+      ex->err = "hipMalloc failed";
+      return GX_ERR_INTERNAL;
+    }
+    if (gxp::gxStrEmit(ex->devPd, pd, si, spOffsets[si],
+                       (uint8_t*)spData[si], ex->stream) != 0 ||
+        gxp::gxPackNulls(sp.notNull, spBitmaps[si], n, ex->stream) != 0) {
+      ex->err = "string emit launch failed";
+      return GX_ERR_INTERNAL;
+    }
+  }
   HIP_OK(ex, hipEventRecord(ev1, ex->stream));
   HIP_OK(ex, hipStreamSynchronize(ex->stream));
   {
@@ -3767,6 +3927,14 @@ static int32_t runProject(gx_exec* ex) {
     gxp::DevCol& dst = ex->desc.table.cols[c];
     if (ex->projOutSrcCol[c] >= 0) {
       dst = pd.table.cols[ex->projOutSrcCol[c]];
+    } else if (ex->projOutSprog[c] >= 0) {
+      int si = ex->projOutSprog[c];
+      setDevColMeta(&dst, GX_TYPE_STRING, 0);
+      dst.data = n > 0 ? spData[si] : nullptr;
+      dst.offsets = n > 0 ? spOffsets[si] : nullptr;
+      dst.nullBitmap = n > 0 ? spBitmaps[si] : nullptr;
+      dst.hasNulls = n > 0 ? 1 : 0;
+      dst.denseOffsets = 0;
     } else {
       int o = ex->projOutSlot[c];
       setDevColMeta(&dst, ex->projOutTypes[c], ex->projOutFracs[c]);

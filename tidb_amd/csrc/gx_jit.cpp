@@ -545,6 +545,11 @@ const JitProg* compile(const FusedQueryDesc& d, std::string* whyNot) {
       if (whyNot) *whyNot = "f64 accumulator not specialized";
       return nullptr;
     }
+  for (int i = 0; i < d.nIns; i++)
+    if (d.ins[i].op == gxp::VM_STRLEN) {
+      if (whyNot) *whyNot = "string builtin not specialized";
+      return nullptr;
+    }
   return compileSource(generateSource(d), "genq_narrow", "genq_wide", whyNot);
 }
 

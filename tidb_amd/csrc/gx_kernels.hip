@@ -204,6 +204,16 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         vm.setNull(ins.dst, nul);
         break;
       }
+      case VM_STRLEN: {
+        // LENGTH (builtinLengthSig): byte length from the offsets pair
+        const DevCol& c = d.table.cols[ins.a];
+        bool nul = colIsNull(c, row);
+        ulonglong2 off = raw.get(ins.c);
+        int64_t len = c.denseOffsets ? 1 : (int64_t)(off.y - off.x);
+        vm.set(ins.dst, nul ? VT<WIDE>::zero() : VT<WIDE>::fromI64(len, &ovf));
+        vm.setNull(ins.dst, nul);
+        break;
+      }
       case VM_LOAD_CONST: {
         if (WIDE) {
           Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
@@ -630,6 +640,18 @@ __device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
     bool eq = len == strConstLen;
     for (int j = 0; j < len && eq; j++) eq = p[st + j] == strConst[j];
     return cmpResult(eq ? 0 : 1, pd.cmp);
+  }
+  if (pd.kind == PRED_STR_LIKE_PREFIX) {
+    // builtinLikeSig 'abc%' fast path: byte prefix, case-sensitive, no pad
+    // trimming (LIKE does not use PAD SPACE semantics)
+    int64_t st, en;
+    if (c.denseOffsets) { st = row; en = row + 1; }
+    else { st = gptr<int64_t>(c.offsets)[row]; en = gptr<int64_t>(c.offsets)[row + 1]; }
+    auto p = gptr<uint8_t>(c.data);
+    if (en - st < strConstLen) return false;
+    for (int j = 0; j < strConstLen; j++)
+      if (p[st + j] != strConst[j]) return false;
+    return true;
   }
   return false;
 }
@@ -2426,6 +2448,20 @@ __global__ void projectKernel(const ProjDesc* __restrict__ dp) {
           vm.setNull(ins.dst, nul);
           break;
         }
+        case VM_STRLEN: {
+          // LENGTH (builtinLengthSig): byte length, direct offsets read
+          const DevCol& c = d.table.cols[ins.a];
+          bool nul = colIsNull(c, row);
+          int64_t len = 0;
+          if (!nul)
+            len = c.denseOffsets ? 1
+                                 : gptr<int64_t>(c.offsets)[row + 1] -
+                                       gptr<int64_t>(c.offsets)[row];
+          vm.set(ins.dst,
+                 nul ? VT<WIDE>::zero() : VT<WIDE>::fromI64(len, &ovf));
+          vm.setNull(ins.dst, nul);
+          break;
+        }
         case VM_LOAD_CONST: {
           if (WIDE) {
             Int128 cv = {(uint64_t)d.constLo[ins.a], d.constHi[ins.a]};
@@ -2798,6 +2834,81 @@ int gxSelectPhase(int phase, const HashJoinDesc* devDesc,
   else
     hipLaunchKernelGGL(selCompactKernel<true>, g, dim3(256), 0,
                        (hipStream_t)stream, devDesc);
+  return (int)hipGetLastError();
+}
+
+__global__ void strWindowKernel(const ProjDesc* __restrict__ dp, int pi) {
+  const ProjDesc& d = *dp;
+  const StrProg& sp = d.sprog[pi];
+  const DevCol& c = d.table.cols[sp.col];
+  int64_t n = d.table.nRows;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool nul = colIsNull(c, row);
+    int64_t s = 0, len = 0;
+    if (!nul) {
+      if (c.denseOffsets) {
+        s = row;
+        len = 1;
+      } else {
+        s = gptr<int64_t>(c.offsets)[row];
+        len = gptr<int64_t>(c.offsets)[row + 1] - s;
+      }
+      // SUBSTR windows compose (builtinSubstring3ArgsSig, byte semantics):
+      // 1-based pos, negative from the end, out-of-range/len<=0 -> empty
+      for (int w = 0; w < sp.nWin; w++) {
+        int64_t pos = sp.winPos[w], L = sp.winLen[w];
+        int64_t start = pos < 0 ? len + pos + 1 : pos;
+        if (start < 1 || start > len || L <= 0) {
+          len = 0;
+          break;
+        }
+        int64_t l2 = L < len - (start - 1) ? L : len - (start - 1);
+        s += start - 1;
+        len = l2;
+      }
+    }
+    sp.starts[row] = s;
+    sp.lens[row] = len;
+    sp.notNull[row] = nul ? 0 : 1;
+  }
+}
+
+__global__ void strEmitKernel(const ProjDesc* __restrict__ dp, int pi,
+                              const int64_t* __restrict__ outOffsets,
+                              uint8_t* __restrict__ outData) {
+  const ProjDesc& d = *dp;
+  const StrProg& sp = d.sprog[pi];
+  const DevCol& c = d.table.cols[sp.col];
+  int64_t n = d.table.nRows;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    int64_t s = gptr<int64_t>(sp.starts)[row];
+    int64_t len = gptr<int64_t>(sp.lens)[row];
+    int64_t o = outOffsets[row];
+    auto p = gptr<uint8_t>(c.data);
+    for (int64_t j = 0; j < len; j++) {
+      uint8_t b = p[s + j];
+      if (sp.upper && b >= 'a' && b <= 'z') b = (uint8_t)(b - 'a' + 'A');
+      outData[o + j] = b;
+    }
+  }
+}
+
+int gxStrWindow(const ProjDesc* devDesc, const ProjDesc& h, int progIdx,
+                void* stream) {
+  if (h.table.nRows == 0) return 0;
+  hipLaunchKernelGGL(strWindowKernel, dim3(gridFor(h.table.nRows)), dim3(256),
+                     0, (hipStream_t)stream, devDesc, progIdx);
+  return (int)hipGetLastError();
+}
+
+int gxStrEmit(const ProjDesc* devDesc, const ProjDesc& h, int progIdx,
+              const int64_t* outOffsets, uint8_t* outData, void* stream) {
+  if (h.table.nRows == 0) return 0;
+  hipLaunchKernelGGL(strEmitKernel, dim3(gridFor(h.table.nRows)), dim3(256),
+                     0, (hipStream_t)stream, devDesc, progIdx, outOffsets,
+                     outData);
   return (int)hipGetLastError();
 }
 

@@ -48,6 +48,10 @@ class Builder:
     def const_dec(self, dec40):
         return self.lib.gx_pb_const_dec(self.pb, (ctypes.c_uint8 * 40)(*dec40))
 
+    def const_str(self, s):
+        b = s.encode() if isinstance(s, str) else s
+        return self.lib.gx_pb_const_str(self.pb, b, len(b))
+
     def call(self, func, ret_type, ret_frac, *args):
         return self.lib.gx_pb_call(self.pb, func, ret_type, ret_frac, _arr(args), len(args))
 

@@ -434,6 +434,10 @@ struct RowPackDesc {
 int gxPackRows(const RowPackDesc& d, void* stream);
 int gxUnpackRows(const RowPackDesc& d, void* stream);
 
+// out-of-core join: per-row partition ids for side 0 (build) / 1 (probe)
+int gxHjPartIds(const HashJoinDesc* devDesc, const HashJoinDesc& h, int side,
+                int nParts, uint32_t* out, void* stream);
+
 // phases: 0 = build (chain insert), 1 = count matches (+ matched flags),
 // 2 = fill match pairs, 3 = post filter, 4 = count unmatched build rows
 // (right outer), 5 = fill unmatched build rows
@@ -530,7 +534,7 @@ struct TopNOut {
 int gxLaunchTpchGen(int table, DevTable* devTab, int64_t rowBegin, int64_t nRows,
                     uint64_t seed, int64_t totalRows, void* stream);
 int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
-                     void* stream);
+                     void* stream, int skipInit = 0);
 int gxLaunchMemset(void* p, int v, size_t n, void* stream);
 int gxFusedGrid(int64_t rows);
 int gxLaunchInitTable(GroupSlot* table, int nSlots, void* stream);

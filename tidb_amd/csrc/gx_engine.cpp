@@ -99,6 +99,17 @@ struct SortRun {
   int64_t n = 0;
 };
 
+// one host-resident hash partition of an out-of-core join side — the
+// partition-file analog of join/hash_join_spill.go (host RAM stands in for
+// disk, like the sort-spill runs). Rows of one join key land in exactly one
+// partition on both sides, so the partitions join independently.
+struct JoinPart {
+  std::vector<std::vector<uint8_t>> colData;     // per col: bytes
+  std::vector<std::vector<int64_t>> colOffsets;  // varlen: n+1 (else empty)
+  std::vector<std::vector<uint8_t>> nullBytes;   // byte/row (else empty)
+  int64_t n = 0;
+};
+
 struct OutRowVal {
   bool isNull = false;
   int type = GX_TYPE_I64;
@@ -151,6 +162,7 @@ struct gx_exec {
   bool jaBuildTried = false;
   // device state
   bool deviceReady = false;
+  bool fusedStreamed = false;  // out-of-core agg ran (re-opens re-stream)
   std::vector<void*> devBufs;
   // persistent singletons (descriptors, counters, result tables) whose
   // pointers are cached across runs — never freed by beginRun/freeSince
@@ -217,6 +229,14 @@ struct gx_exec {
   };
   bool isHashJoin = false;
   std::vector<JoinStage> joinStages;     // root stage last
+  // out-of-core join (hash_join_spill.go analog): both sides hash-partition
+  // into host-RAM runs when build+probe+output exceed the HBM budget;
+  // partitions join one at a time on device, streaming out per partition
+  bool joinSpill = false;
+  int joinSpillCur = -1;
+  size_t joinSpillMark = SIZE_MAX;
+  std::vector<JoinPart> spillB, spillP;
+  uint64_t joinSpillMatches = 0;
   gxp::HashJoinDesc* devHj = nullptr;
   // general aggregation over joined rows: runHashJoin materializes the join
   // output into desc.table, then the fused aggregation runs over it
@@ -2577,12 +2597,205 @@ static void applyPostSort(gx_exec* ex) {
   }
 }
 
+// a packed group key overflowed its lane (string > 3 bytes or i64 >= 2^31):
+// convert every column to the serialized wide-key path. Returns false when
+// the conversion is impossible (fetch plan full / key store alloc failed).
+static bool convertPackedKeysToWide(gx_exec* ex) {
+  gxp::GroupKeyDesc& g = ex->desc.gkey;
+  bool ok = true;
+  for (int k = 0; k < g.nCols && ok; k++) {
+    if (g.kind[k] == 1) {
+      g.kind[k] = 3;
+    } else if (g.kind[k] == 0) {
+      g.kind[k] = 5;  // offsets slot already assigned at bind
+      ok = g.slot[k] >= 0;
+    } else if (g.kind[k] == 2) {
+      g.kind[k] = 5;
+      g.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS, g.col[k]);
+      ok = g.slot[k] >= 0;
+    }
+  }
+  if (!ok) return false;
+  g.wideMode = 1;
+  ex->desc.useGlds = 0;
+  ex->jitProg = nullptr;  // the packed-key specialization no longer applies
+  return ensureWideKeyStore(ex) == GX_OK;
+}
+
+static int32_t fusedDecodeResults(gx_exec* ex);
+static int64_t hbmBudgetBytes();
+
+// out-of-core aggregation (the agg_spill.go analog, MI355X-shaped): the
+// GROUP STATES stay resident in HBM — bounded by the grown global table —
+// while INPUT row-range slices stream through the fused kernel, every slice
+// accumulating into the SAME table (skipInit). Retryable device flags
+// (narrow overflow, LDS/global-table growth, packed-key overflow) restart
+// the whole slice loop with the new setting, exactly like the one-shot path.
+static int32_t runFusedStreaming(gx_exec* ex, Binding& b, int64_t sliceRows) {
+  const int64_t total = b.tpchRows, saveOff = b.tpchRowOffset;
+  const int64_t saveTot = b.tpchTotalRows > 0 ? b.tpchTotalRows : total;
+  ex->fusedStreamed = true;
+  int32_t rc = GX_OK;
+  if (getenv("GX_DEBUG"))
+    fprintf(stderr, "[gx] fused agg streaming: %lld rows in %lld-row slices\n",
+            (long long)total, (long long)sliceRows);
+restart:
+  ex->lastKernelMs = 0;
+  for (int64_t done = 0; done < total;) {
+    int64_t n = std::min(sliceRows, total - done);
+    size_t mark = ex->devBufs.size();
+    b.tpchRows = n;
+    b.tpchRowOffset = saveOff + done;
+    b.tpchTotalRows = saveTot;
+    ex->deviceReady = false;
+    rc = materializeDevice(ex);
+    if (rc) goto out;
+    if (done == 0) {
+      // string group keys through the wide path verify/decode via OWNER ROW
+      // references into the source table — slices are freed, so those
+      // references would dangle. Fail loudly (decimal/int/time wide keys and
+      // all packed keys stream fine).
+      for (int k = 0; k < ex->desc.gkey.nCols; k++)
+        if (ex->desc.gkey.wideMode && ex->desc.gkey.kind[k] == 5) {
+          ex->err = "string group keys over out-of-core inputs unsupported "
+                    "this round";
+          rc = GX_ERR_INVALID;
+          goto out;
+        }
+    }
+    {
+      ex->desc.ablate = 0;
+      HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+      if (done == 0) HIP_OK(ex, hipMemsetAsync(ex->devSel, 0, 8, ex->stream));
+      HIP_OK(ex, hipMemcpyAsync(ex->devDesc, &ex->desc, sizeof(ex->desc),
+                                hipMemcpyHostToDevice, ex->stream));
+      if (ex->desc.noLds) ex->desc.useGlds = 0;
+      hipEvent_t ev0, ev1;
+      HIP_OK(ex, hipEventCreate(&ev0));
+      HIP_OK(ex, hipEventCreate(&ev1));
+      HIP_OK(ex, hipEventRecord(ev0, ex->stream));
+      if (!ex->jitTried && !ex->desc.useGlds && !getenv("GX_NO_JIT")) {
+        ex->jitTried = true;
+        bool eligible = !ex->desc.gkey.wideMode;
+        for (int k = 0; k < ex->desc.gkey.nCols; k++)
+          eligible &= ex->desc.gkey.kind[k] != 0;
+        if (eligible) {
+          std::string why;
+          ex->jitProg = gxjit::compile(ex->desc, &why);
+        }
+      }
+      int lrc;
+      if (ex->jitProg) {
+        lrc = done == 0 ? gxp::gxLaunchInitTable(
+                              ex->desc.globalTable,
+                              1 << ex->desc.globalGroupsLog2, ex->stream)
+                        : 0;
+        if (lrc == 0 && done == 0 && ex->desc.gkey.wideMode &&
+            ex->desc.gkey.recCursor)
+          lrc = (int)hipMemsetAsync(ex->desc.gkey.recCursor, 0, 8, ex->stream);
+        if (lrc == 0)
+          lrc = gxjit::launch(ex->jitProg, ex->desc.wide != 0, ex->devDesc,
+                              gxp::gxFusedGrid(ex->desc.table.nRows),
+                              ex->stream);
+      } else {
+        lrc = gxp::gxLaunchFusedAgg(ex->desc, ex->devDesc, ex->stream,
+                                    done != 0);
+      }
+      if (lrc != 0) {
+        ex->err = "fused kernel launch failed: " +
+                  std::string(hipGetErrorString((hipError_t)lrc));
+        rc = GX_ERR_INTERNAL;
+        goto out;
+      }
+      HIP_OK(ex, hipEventRecord(ev1, ex->stream));
+      HIP_OK(ex, hipStreamSynchronize(ex->stream));
+      float ms = 0;
+      hipEventElapsedTime(&ms, ev0, ev1);
+      ex->lastKernelMs += ms;
+      hipEventDestroy(ev0);
+      hipEventDestroy(ev1);
+    }
+    uint32_t errFlag = 0;
+    HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+    if (errFlag == 256u && !ex->desc.wide) {
+      ex->desc.wide = 1;
+      freeSince(ex, mark);
+      goto restart;
+    }
+    if ((errFlag & 16u) && !ex->desc.noLds) {
+      ex->desc.noLds = 1;
+      freeSince(ex, mark);
+      goto restart;
+    }
+    if ((errFlag & 32u) && ex->desc.globalGroupsLog2 < 25) {
+      ex->desc.globalGroupsLog2 += 3;
+      devFreeP(ex, ex->devTable);
+      ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAllocP(
+          ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
+      if (!ex->devTable) {
+        ex->err = "hipMalloc failed (group table)";
+        rc = GX_ERR_INTERNAL;
+        goto out;
+      }
+      if (ex->desc.gkey.wideMode) {
+        devFreeP(ex, ex->desc.gkey.keyStore);
+        rc = ensureWideKeyStore(ex);
+        if (rc) goto out;
+      }
+      freeSince(ex, mark);
+      goto restart;
+    }
+    if ((errFlag & 2u) && !ex->desc.gkey.wideMode && ex->desc.gkey.nCols > 0 &&
+        convertPackedKeysToWide(ex)) {
+      freeSince(ex, mark);
+      goto restart;
+    }
+    errFlag &= ~256u;
+    if (errFlag != 0) {
+      ex->err = "device execution error flag 0x" + std::to_string(errFlag) +
+                " (unsupported data shape or overflow)";
+      rc = GX_ERR_INTERNAL;
+      goto out;
+    }
+    freeSince(ex, mark);
+    done += n;
+  }
+out:
+  b.tpchRows = total;
+  b.tpchRowOffset = saveOff;
+  b.tpchTotalRows = saveTot;
+  if (rc) return rc;
+  HIP_OK(ex, hipMemcpy(&ex->lastSelCount, ex->devSel, 8, hipMemcpyDeviceToHost));
+  return fusedDecodeResults(ex);
+}
+
 static int32_t runFused(gx_exec* ex) {
   if (ex->vmHasDiv && !getenv("GX_DIV_NARROW"))
     ex->desc.wide = 1;  // DIV quotients rarely fit int64
   ex->desc.hasDiv = ex->vmHasDiv ? 1 : 0;
   ex->desc.nVmRegs = ex->vmNextReg;
   if (getenv("GX_FORCE_WIDE")) ex->desc.wide = 1;
+  // out-of-core aggregation: a generator source above the HBM budget
+  // streams in row-range slices instead of materializing whole
+  if ((!ex->deviceReady || ex->fusedStreamed) && !ex->aggOverJoin &&
+      ex->sourceNode >= 0) {
+    auto itb = ex->bindings.find(ex->sourceNode);
+    if (itb != ex->bindings.end() && !itb->second.haveChunks &&
+        itb->second.tpchTable >= 0) {
+      int64_t rb = 0;
+      for (int c = 0; c < ex->desc.table.nCols; c++) {
+        int t = ex->desc.table.cols[c].type;
+        rb += t == GX_TYPE_DECIMAL ? 40 : (t == GX_TYPE_STRING ? 9 : 8);
+      }
+      int64_t need = itb->second.tpchRows * rb;
+      int64_t budget = hbmBudgetBytes();
+      if (ex->fusedStreamed || need > budget) {
+        int64_t sliceRows = std::max<int64_t>(
+            4096, budget / 2 / std::max<int64_t>(rb, 1));
+        return runFusedStreaming(ex, itb->second, sliceRows);
+      }
+    }
+  }
   int32_t rc = materializeDevice(ex);
   if (rc) return rc;
   ex->desc.ablate = getenv("GX_ABLATE") ? atoi(getenv("GX_ABLATE")) : 0;
@@ -2678,33 +2891,10 @@ static int32_t runFused(gx_exec* ex) {
     return runFused(ex);
   }
   if ((errFlag & 2u /*kErrBadKey*/) && !ex->desc.gkey.wideMode &&
-      ex->desc.gkey.nCols > 0) {
-    // a packed key lane overflowed (string > 3 bytes or i64 >= 2^31):
-    // convert every column to the serialized wide-key path and rerun
-    gxp::GroupKeyDesc& g = ex->desc.gkey;
-    bool ok = true;
-    for (int k = 0; k < g.nCols && ok; k++) {
-      if (g.kind[k] == 1) {
-        g.kind[k] = 3;
-      } else if (g.kind[k] == 0) {
-        g.kind[k] = 5;  // offsets slot already assigned at bind
-        ok = g.slot[k] >= 0;
-      } else if (g.kind[k] == 2) {
-        g.kind[k] = 5;
-        g.slot[k] = fetchSlot(ex, gxp::FETCH_OFFSETS, g.col[k]);
-        ok = g.slot[k] >= 0;
-      }
-    }
-    if (ok) {
-      g.wideMode = 1;
-      ex->desc.useGlds = 0;
-      ex->jitProg = nullptr;  // the packed-key specialization no longer applies
-      int32_t rc2 = ensureWideKeyStore(ex);
-      if (rc2) return rc2;
-      if (getenv("GX_DEBUG"))
-        fprintf(stderr, "[gx] packed key overflow -> wide-key retry\n");
-      return runFused(ex);
-    }
+      ex->desc.gkey.nCols > 0 && convertPackedKeysToWide(ex)) {
+    if (getenv("GX_DEBUG"))
+      fprintf(stderr, "[gx] packed key overflow -> wide-key retry\n");
+    return runFused(ex);
   }
   errFlag &= ~256u;
   if (ex->desc.ablate != 0 && getenv("GX_DEBUG"))
@@ -2723,6 +2913,12 @@ static int32_t runFused(gx_exec* ex) {
             ex->desc.nAggs, ex->desc.gkey.nCols,
             (unsigned long long)ex->lastSelCount, ex->lastKernelMs);
   }
+  return fusedDecodeResults(ex);
+}
+
+// download the global table + wide-key records and decode the result
+// rows (shared by the one-shot and the out-of-core streaming paths)
+static int32_t fusedDecodeResults(gx_exec* ex) {
   std::vector<gxp::GroupSlot> table((size_t)1 << ex->desc.globalGroupsLog2);
   HIP_OK(ex, hipMemcpy(table.data(), ex->devTable,
                        sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2,
@@ -2956,6 +3152,7 @@ static int32_t runFused(gx_exec* ex) {
   applyPostSort(ex);
   return GX_OK;
 }
+
 
 // ---------------- join-aggregate execution (Q3 class) ----------------
 
@@ -3632,6 +3829,234 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
   return GX_OK;
 }
 
+
+// ---------------- out-of-core join (hash_join_spill.go analog) ----------------
+
+static int64_t hbmBudgetBytes() {
+  if (const char* e = getenv("GX_HBM_BUDGET")) return atoll(e);
+  size_t freeB = 0, totalB = 0;
+  if (hipMemGetInfo(&freeB, &totalB) == hipSuccess)
+    return (int64_t)(freeB / 10 * 9);
+  return INT64_MAX;
+}
+
+// rough resident bytes per row of a source schema (varlen estimated)
+static int64_t schemaRowBytes(const PNode& srcN) {
+  int64_t b = 0;
+  for (int t : srcN.colTypes)
+    b += t == GX_TYPE_DECIMAL ? 40 : (t == GX_TYPE_STRING ? 24 : 8);
+  return b;
+}
+
+static int64_t bindingRows(gx_exec* ex, int srcId) {
+  auto it = ex->bindings.find(srcId);
+  if (it == ex->bindings.end()) return 0;
+  if (it->second.haveChunks) {
+    int64_t n = 0;
+    for (auto& ch : it->second.chunks) n += ch.empty() ? 0 : ch[0].length;
+    return n;
+  }
+  return it->second.tpchRows;
+}
+
+// hash-partition one side into host-RAM runs, streaming the source in
+// budget-bounded row slices (generator sources re-generate per slice; bound
+// chunks upload once — their size is the caller's to bound)
+static int32_t partitionJoinSide(gx_exec* ex, gx_exec::JoinStage& st,
+                                 int side, int nParts, int64_t sliceRows,
+                                 std::vector<JoinPart>* parts) {
+  int srcId = side == 0 ? st.srcB : st.srcP;
+  const PNode& srcN = ex->plan.nodes[srcId];
+  Binding& b = ex->bindings[srcId];
+  int nc = (int)srcN.colTypes.size();
+  parts->assign(nParts, JoinPart{});
+  for (auto& jp : *parts) {
+    jp.colData.resize(nc);
+    jp.colOffsets.resize(nc);
+    jp.nullBytes.resize(nc);
+    for (int c = 0; c < nc; c++)
+      if (srcN.colTypes[c] == GX_TYPE_STRING) jp.colOffsets[c].push_back(0);
+  }
+  const int64_t saveRows = b.tpchRows, saveOff = b.tpchRowOffset,
+                saveTot = b.tpchTotalRows;
+  int64_t done = 0;
+  int32_t rc = GX_OK;
+  for (;;) {
+    size_t mark = ex->devBufs.size();
+    gxp::DevTable tab{};
+    if (b.haveChunks) {
+      if (done > 0) break;
+      rc = materializeTable(ex, srcId, &tab);
+      if (rc) break;
+      done = tab.nRows;
+    } else {
+      if (done >= saveRows) break;
+      int64_t n = std::min(sliceRows, saveRows - done);
+      b.tpchRowOffset = saveOff + done;
+      b.tpchRows = n;
+      b.tpchTotalRows = saveTot > 0 ? saveTot : saveRows;
+      rc = materializeTable(ex, srcId, &tab);
+      if (rc) break;
+      done += n;
+    }
+    int64_t n = tab.nRows;
+    if (n == 0) { freeSince(ex, mark); continue; }
+    uint32_t* devPid = (uint32_t*)devAlloc(ex, (size_t)n * 4);
+    if (!devPid) { ex->err = "hipMalloc failed"; rc = GX_ERR_INTERNAL; break; }
+    gxp::HashJoinDesc hj2 = st.hj;
+    if (side == 0) hj2.build = tab;
+    else hj2.probe = tab;
+    hipError_t he = hipMemcpyAsync(ex->devHj, &hj2, sizeof(hj2),
+                                   hipMemcpyHostToDevice, ex->stream);
+    if (he != hipSuccess ||
+        gxp::gxHjPartIds(ex->devHj, hj2, side, nParts, devPid,
+                         ex->stream) != 0) {
+      ex->err = "partition-id launch failed";
+      rc = GX_ERR_INTERNAL;
+      break;
+    }
+    if (hipStreamSynchronize(ex->stream) != hipSuccess) {
+      ex->err = "partition sync failed";
+      rc = GX_ERR_INTERNAL;
+      break;
+    }
+    std::vector<uint32_t> pid(n);
+    hipMemcpy(pid.data(), devPid, (size_t)n * 4, hipMemcpyDeviceToHost);
+    // download the slice
+    std::vector<std::vector<uint8_t>> hData(nc);
+    std::vector<std::vector<int64_t>> hOff(nc);
+    std::vector<std::vector<uint8_t>> hNull(nc);
+    for (int c = 0; c < nc; c++) {
+      const gxp::DevCol& col = tab.cols[c];
+      if (col.type == GX_TYPE_STRING) {
+        hOff[c].resize(n + 1);
+        hipMemcpy(hOff[c].data(), col.offsets, (size_t)(n + 1) * 8,
+                  hipMemcpyDeviceToHost);
+        hData[c].resize(std::max<int64_t>(hOff[c][n], 1));
+        if (hOff[c][n] > 0)
+          hipMemcpy(hData[c].data(), col.data, (size_t)hOff[c][n],
+                    hipMemcpyDeviceToHost);
+      } else {
+        hData[c].resize((size_t)n * col.elemSize);
+        hipMemcpy(hData[c].data(), col.data, hData[c].size(),
+                  hipMemcpyDeviceToHost);
+      }
+      if (col.hasNulls && col.nullBitmap) {
+        hNull[c].resize((n + 7) / 8);
+        hipMemcpy(hNull[c].data(), col.nullBitmap, hNull[c].size(),
+                  hipMemcpyDeviceToHost);
+      }
+    }
+    freeSince(ex, mark);
+    // split rows into partitions on the host
+    for (int64_t r = 0; r < n; r++) {
+      JoinPart& jp = (*parts)[pid[r]];
+      for (int c = 0; c < nc; c++) {
+        if (srcN.colTypes[c] == GX_TYPE_STRING) {
+          int64_t s = hOff[c][r], e2 = hOff[c][r + 1];
+          jp.colData[c].insert(jp.colData[c].end(), hData[c].begin() + s,
+                               hData[c].begin() + e2);
+          jp.colOffsets[c].push_back((int64_t)jp.colData[c].size());
+        } else {
+          int es = srcN.colTypes[c] == GX_TYPE_DECIMAL ? 40 : 8;
+          jp.colData[c].insert(jp.colData[c].end(),
+                               hData[c].begin() + (size_t)r * es,
+                               hData[c].begin() + (size_t)(r + 1) * es);
+        }
+        if (!hNull[c].empty()) {
+          if (jp.nullBytes[c].empty() && jp.n > 0)
+            jp.nullBytes[c].assign((size_t)jp.n, 1);
+          if (!hNull[c].empty())
+            jp.nullBytes[c].push_back((hNull[c][r / 8] >> (r % 8)) & 1);
+        } else if (!jp.nullBytes[c].empty()) {
+          jp.nullBytes[c].push_back(1);
+        }
+      }
+      jp.n++;
+    }
+  }
+  b.tpchRows = saveRows;
+  b.tpchRowOffset = saveOff;
+  b.tpchTotalRows = saveTot;
+  return rc;
+}
+
+// upload one host partition as a resident device table
+static int32_t uploadJoinPart(gx_exec* ex, const PNode& srcN,
+                              const JoinPart& jp, gxp::DevTable* tab) {
+  int nc = (int)srcN.colTypes.size();
+  tab->nCols = nc;
+  tab->nRows = jp.n;
+  int64_t n = jp.n;
+  for (int c = 0; c < nc; c++) {
+    gxp::DevCol& col = tab->cols[c];
+    setDevColMeta(&col, srcN.colTypes[c], srcN.colFracs[c]);
+    if (srcN.colTypes[c] == GX_TYPE_STRING) {
+      col.offsets = (int64_t*)devAlloc(ex, (size_t)(n + 1) * 8);
+      col.data = devAlloc(ex, std::max<size_t>(jp.colData[c].size(), 1));
+      if (!col.offsets || !col.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      hipMemcpy(col.offsets, jp.colOffsets[c].data(), (size_t)(n + 1) * 8,
+                hipMemcpyHostToDevice);
+      if (!jp.colData[c].empty())
+        hipMemcpy(col.data, jp.colData[c].data(), jp.colData[c].size(),
+                  hipMemcpyHostToDevice);
+      col.denseOffsets = jp.colOffsets[c].back() == n ? 1 : 0;
+    } else {
+      col.data = devAlloc(ex, std::max<size_t>(jp.colData[c].size(), 1) + 16);
+      if (!col.data) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+      if (!jp.colData[c].empty())
+        hipMemcpy(col.data, jp.colData[c].data(), jp.colData[c].size(),
+                  hipMemcpyHostToDevice);
+    }
+    if (!jp.nullBytes[c].empty()) {
+      std::vector<uint8_t> bm((n + 7) / 8, 0);
+      bool anyNull = false;
+      for (int64_t r = 0; r < n; r++) {
+        if (jp.nullBytes[c][r]) bm[r / 8] |= 1 << (r % 8);
+        else anyNull = true;
+      }
+      if (anyNull) {
+        col.nullBitmap = (uint8_t*)devAlloc(ex, bm.size());
+        if (!col.nullBitmap) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+        hipMemcpy(col.nullBitmap, bm.data(), bm.size(), hipMemcpyHostToDevice);
+        col.hasNulls = 1;
+      }
+    }
+  }
+  return GX_OK;
+}
+
+// run the next non-empty partition's join; leaves its output in desc.table
+// (nRows 0 once exhausted)
+static int32_t joinSpillAdvance(gx_exec* ex) {
+  gx_exec::JoinStage& st = ex->joinStages[0];
+  while (++ex->joinSpillCur < (int)ex->spillB.size()) {
+    freeSince(ex, ex->joinSpillMark);
+    int p = ex->joinSpillCur;
+    if (ex->spillB[p].n == 0 && ex->spillP[p].n == 0) continue;
+    const PNode& bN = ex->plan.nodes[st.srcB];
+    const PNode& pN = ex->plan.nodes[st.srcP];
+    int32_t rc = uploadJoinPart(ex, bN, ex->spillB[p], &st.buildTab);
+    if (rc == GX_OK) rc = uploadJoinPart(ex, pN, ex->spillP[p], &st.probeTab);
+    if (rc) return rc;
+    st.inputsReady = true;
+    rc = runJoinStage(ex, st, true);
+    if (rc) return rc;
+    ex->joinSpillMatches += ex->lastSelCount;
+    for (int c = 0; c < st.out.nCols; c++)
+      ex->desc.table.cols[c] = st.out.cols[c];
+    ex->desc.table.nCols = st.out.nCols;
+    ex->desc.table.nRows = st.out.nRows;
+    ex->srcPos = 0;
+    if (st.out.nRows == 0) continue;
+    return GX_OK;
+  }
+  ex->desc.table.nRows = 0;
+  ex->srcPos = 0;
+  ex->lastSelCount = ex->joinSpillMatches;
+  return GX_OK;
+}
+
 static int32_t runHashJoin(gx_exec* ex) {
   if (!ex->deviceReady) {
     if (!gpuAvailable()) {
@@ -3650,6 +4075,48 @@ static int32_t runHashJoin(gx_exec* ex) {
     ex->deviceReady = true;
   }
   ex->lastKernelMs = 0;
+  // out-of-core: when build + probe + output would exceed the HBM budget
+  // (GX_HBM_BUDGET overrides the free-memory estimate), hash-partition both
+  // sides to host RAM and join partition-wise (hash_join_spill.go analog).
+  // Single-stage source-fed joins only; partitions persist across re-opens.
+  if (ex->joinStages.size() == 1 && !ex->aggOverJoin &&
+      ex->joinStages[0].srcB >= 0 && ex->joinStages[0].srcP >= 0) {
+    gx_exec::JoinStage& st0 = ex->joinStages[0];
+    if (!st0.hj.counters) {
+      st0.hj.counters = (uint64_t*)devAllocP(ex, 3 * 8);
+      if (!st0.hj.counters) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    }
+    st0.hj.errorFlag = ex->devErr;
+    if (!ex->joinSpill && !st0.inputsReady) {
+      int64_t rbB = schemaRowBytes(ex->plan.nodes[st0.srcB]);
+      int64_t rbP = schemaRowBytes(ex->plan.nodes[st0.srcP]);
+      int64_t nB = bindingRows(ex, st0.srcB);
+      int64_t nP = bindingRows(ex, st0.srcP);
+      int64_t need = nB * rbB + nP * rbP + nP * (rbB + rbP);
+      int64_t budget = hbmBudgetBytes();
+      if (need > budget) {
+        int nParts = (int)std::min<int64_t>(
+            256, std::max<int64_t>(2, need * 2 / std::max<int64_t>(budget, 1)));
+        int64_t sliceRows = std::max<int64_t>(
+            1024, budget / 4 / std::max<int64_t>(rbB + rbP, 1));
+        int32_t rc = partitionJoinSide(ex, st0, 0, nParts, sliceRows,
+                                       &ex->spillB);
+        if (rc == GX_OK)
+          rc = partitionJoinSide(ex, st0, 1, nParts, sliceRows, &ex->spillP);
+        if (rc) return rc;
+        ex->joinSpill = true;
+        ex->joinSpillMark = ex->devBufs.size();
+        if (getenv("GX_DEBUG"))
+          fprintf(stderr, "[gx] join spill: %d partitions (need %lld > budget %lld)\n",
+                  nParts, (long long)need, (long long)budget);
+      }
+    }
+    if (ex->joinSpill) {
+      ex->joinSpillCur = -1;
+      ex->joinSpillMatches = 0;
+      return joinSpillAdvance(ex);
+    }
+  }
   // persistent per-stage state (counters + cached source materializations)
   // is allocated before the per-run scope opens
   for (size_t s = 0; s < ex->joinStages.size(); s++) {
@@ -5245,13 +5712,26 @@ int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
       ex->ranQuery = true;
     }
     if (!ex->devSortKeys.empty() && !ex->devSorted) {
+      if (ex->joinSpill) {
+        ex->err = "ORDER BY over an out-of-core join unsupported this round";
+        *rows_out = 0;
+        return GX_ERR_INVALID;
+      }
       int32_t rc = runDeviceSort(ex);
       if (rc) {
         *rows_out = 0;
         return rc;
       }
     }
-    return emitTableChunk(ex, out, rows_out);
+    int32_t rc = emitTableChunk(ex, out, rows_out);
+    // out-of-core: stream the next partitions' outputs
+    while (rc == GX_OK && *rows_out == 0 && ex->joinSpill &&
+           ex->joinSpillCur < (int)ex->spillB.size()) {
+      rc = joinSpillAdvance(ex);
+      if (rc || ex->joinSpillCur >= (int)ex->spillB.size()) break;
+      rc = emitTableChunk(ex, out, rows_out);
+    }
+    return rc;
   }
   if (!ex->ranQuery) {
     int32_t rc;

@@ -1547,14 +1547,16 @@ int gxLaunchInitTable(GroupSlot* table, int nSlots, void* stream) {
 }
 
 int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
-                     void* stream) {
+                     void* stream, int skipInit) {
   hipStream_t s = (hipStream_t)stream;
   int grid = gxFusedGrid(desc.table.nRows);
   int nSlots = 1 << desc.globalGroupsLog2;
-  hipLaunchKernelGGL(initGlobalTableKernel, dim3((nSlots + 255) / 256),
-                     dim3(256), 0, s, desc.globalTable, nSlots);
-  if (desc.gkey.wideMode && desc.gkey.recCursor)
-    hipMemsetAsync(desc.gkey.recCursor, 0, 8, s);
+  if (!skipInit) {  // out-of-core slices accumulate into the SAME table
+    hipLaunchKernelGGL(initGlobalTableKernel, dim3((nSlots + 255) / 256),
+                       dim3(256), 0, s, desc.globalTable, nSlots);
+    if (desc.gkey.wideMode && desc.gkey.recCursor)
+      hipMemsetAsync(desc.gkey.recCursor, 0, 8, s);
+  }
   if (desc.useGlds) {
     size_t shmem = ((sizeof(GroupSlot) * kLdsGroups + 15) & ~15ULL) +
                    (size_t)4 * 2 * desc.tileBytes;
@@ -2185,6 +2187,43 @@ __global__ void hjFilterPairsKernel(const HashJoinDesc* __restrict__ dp) {
       d.outProbe2[base + off] = prow;
     }
   }
+}
+
+// out-of-core join (hash_join_spill.go analog): per-row partition id from
+// the join-key hash, using bits independent of the chain-table index so the
+// per-partition tables hash freely. NULL-key rows round-robin by row index —
+// they match nothing, but outer/anti joins still emit them.
+template <bool G>
+__global__ void hjPartIdKernel(const HashJoinDesc* __restrict__ dp, int side,
+                               int nParts, uint32_t* __restrict__ out) {
+  const HashJoinDesc& d = *dp;
+  const DevTable& t = side == 0 ? d.build : d.probe;
+  const int32_t* cols = side == 0 ? d.bKeyCol : d.pKeyCol;
+  int64_t n = t.nRows;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    HjKeys<G> K;
+    uint64_t h;
+    uint32_t p;
+    if (hjLoad(d, t, cols, row, K, &h))
+      p = (uint32_t)((h >> 43) % (uint64_t)nParts);
+    else
+      p = (uint32_t)(row % nParts);
+    out[row] = p;
+  }
+}
+
+int gxHjPartIds(const HashJoinDesc* devDesc, const HashJoinDesc& h, int side,
+                int nParts, uint32_t* out, void* stream) {
+  int64_t n = side == 0 ? h.build.nRows : h.probe.nRows;
+  if (n == 0) return 0;
+  if (h.generalKeys)
+    hipLaunchKernelGGL(hjPartIdKernel<true>, dim3(gridFor(n)), dim3(256), 0,
+                       (hipStream_t)stream, devDesc, side, nParts, out);
+  else
+    hipLaunchKernelGGL(hjPartIdKernel<false>, dim3(gridFor(n)), dim3(256), 0,
+                       (hipStream_t)stream, devDesc, side, nParts, out);
+  return (int)hipGetLastError();
 }
 
 // right outer: emit every predB-passing build row whose matched flag is

@@ -526,6 +526,140 @@ def bench_join(args):
     print(json.dumps(out), flush=True)
 
 
+def run_cpu_baseline_strings(n=2_000_000):
+    """Oracle string-builtin projection over a bounded customer sample."""
+    import time as _t
+    from tests.gxlib import GX_TPCH_CUSTOMER, load_oracle
+    lib = load_oracle()
+    t0 = _t.perf_counter()
+    b, src, root = _strings_plan(lib)
+    ex = b.build(root)
+    ex.bind_tpch(src, GX_TPCH_CUSTOMER, n)
+    ex.open()
+    import ctypes as C
+    from tests.gxlib import GX_TYPE_STRING
+    from tidb_amd.chunkpy import PyChunk
+    types = [4, 4, 0, 0]
+    chunk = PyChunk(types, 1024, [0] * 4, [32768, 32768, None, None])
+    total = 0
+    while True:
+        g = chunk.as_gx()
+        nn = C.c_int32(0)
+        rc = lib.gx_next(ex.ex, C.byref(g), C.byref(nn))
+        assert rc == 0, ex.error()
+        if nn.value == 0:
+            break
+        total += nn.value
+    ex.close()
+    ex.free()
+    b.free()
+    dt = _t.perf_counter() - t0
+    return {
+        "value": n / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"SUBSTR/UPPER/LENGTH projection over {n} customer rows "
+                  f"incl. generation and pull, single thread ({dt:.1f}s)",
+    }
+
+
+def _strings_plan(lib):
+    """Varchar-builtin projection (config 5's varchar half): SUBSTR(seg,1,4),
+    UPPER(SUBSTR(seg,-5,5)), LENGTH(seg), custkey over customer."""
+    from tests.gxlib import (GX_F_LENGTH, GX_F_SUBSTR, GX_F_UPPER,
+                             GX_TYPE_I64, GX_TYPE_STRING)
+    from tidb_amd import plan as P
+    b = P.Builder(lib)
+    src = b.source(P.CUSTOMER_TYPES)
+    seg = b.colref(P.C_MKTSEGMENT, GX_TYPE_STRING)
+    exprs = [
+        b.call(GX_F_SUBSTR, GX_TYPE_STRING, 0, seg, b.const_i64(1),
+               b.const_i64(4)),
+        b.call(GX_F_UPPER, GX_TYPE_STRING, 0,
+               b.call(GX_F_SUBSTR, GX_TYPE_STRING, 0, seg, b.const_i64(-5),
+                      b.const_i64(5))),
+        b.call(GX_F_LENGTH, GX_TYPE_I64, 0, seg),
+        b.colref(P.C_CUSTKEY, GX_TYPE_I64),
+    ]
+    root = b.projection(src, exprs)
+    return b, src, root
+
+
+def bench_strings(args):
+    """Varchar builtins at scale (BASELINE config 5's varchar half,
+    SURVEY §8f row 2): a 4-expression string projection (2 windowed string
+    outputs + LENGTH + passthrough key) over `--rows` synthetic customer
+    rows resident in HBM. One step = the full device projection (window
+    kernels + offset scans + byte emits); emission to host excluded."""
+    import ctypes as C
+    from tests.gxlib import GX_TPCH_CUSTOMER, load_product
+    lib = load_product()
+    n = min(args.rows, 150_000_000)
+    lib.gx_last_kernel_ms.restype = C.c_double
+    lib.gx_last_kernel_ms.argtypes = [C.c_void_p]
+    b, src, root = _strings_plan(lib)
+    ex = b.build(root)
+    ex.bind_tpch(src, GX_TPCH_CUSTOMER, n)
+
+    def step():
+        ex.open()
+        ex.pull_one([4, 4, 0, 0], [0] * 4,
+                    data_caps=[32768, 32768, None, None])
+        k = lib.gx_last_kernel_ms(ex.ex)
+        ex.close()
+        return k
+
+    for _ in range(args.warmup):
+        step()
+    t0 = time.perf_counter()
+    kms = []
+    for _ in range(args.steps):
+        kms.append(step())
+    elapsed = time.perf_counter() - t0
+    avg_kms = sum(kms) / len(kms)
+    value = n / (avg_kms / 1000.0) if avg_kms else 0
+    # algorithmic bytes/row: offsets 8 + segment bytes ~12.9 read by window
+    # (data read once per string output at emit: 2 x ~5) + key 8 + temps
+    # (start/len/notNull 17 x 2 outputs, written+read) + out writes (~9 + 8)
+    bpr = 8 + 13 + 8 + 2 * 5 + 2 * 2 * 17 + 9 + 8 + 8
+    achieved = n * bpr / (avg_kms / 1000.0) / 1e9 if avg_kms else 0
+    out = {
+        "metric": "string_builtin_rows_per_sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "u8",
+        "data": "synthetic",
+        "config": {
+            "workload": f"customer_{n}_substr_upper_length_projection",
+            "rows": n,
+            "parallelism": "single-gpu",
+            "bytes_per_row": bpr,
+        },
+        "kernel_ms_avg": avg_kms,
+        "roofline": {
+            "bound": "hbm",
+            "achieved": achieved,
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": achieved / HBM_PEAK_GBS,
+            "traffic": None,
+        },
+        "cpu_baseline": run_cpu_baseline_strings()
+        if not args.no_cpu_baseline else None,
+    }
+    ex.free()
+    b.free()
+    print(json.dumps(out), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -533,7 +667,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--rows", type=int, default=SF100_ROWS,
                     help="rows per GPU (default SF100 — the metric's config)")
-    ap.add_argument("--query", choices=["q1", "q3", "sort", "wide", "join"],
+    ap.add_argument("--query",
+                    choices=["q1", "q3", "sort", "wide", "join", "strings"],
                     default="q1")
     ap.add_argument("--sf", type=int, default=100,
                     help="scale factor for --query q3 (lineitem = 6M x SF)")
@@ -550,6 +685,8 @@ def main():
         return bench_wide(args)
     if args.query == "join":
         return bench_join(args)
+    if args.query == "strings":
+        return bench_strings(args)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -3991,6 +3991,13 @@ static int32_t uploadJoinPart(gx_exec* ex, const PNode& srcN,
   for (int c = 0; c < nc; c++) {
     gxp::DevCol& col = tab->cols[c];
     setDevColMeta(&col, srcN.colTypes[c], srcN.colFracs[c]);
+    // tab persists across partitions: clear the previous partition's (freed)
+    // buffers so a no-NULL partition cannot inherit a stale bitmap
+    col.data = nullptr;
+    col.offsets = nullptr;
+    col.nullBitmap = nullptr;
+    col.hasNulls = 0;
+    col.denseOffsets = 0;
     if (srcN.colTypes[c] == GX_TYPE_STRING) {
       col.offsets = (int64_t*)devAlloc(ex, (size_t)(n + 1) * 8);
       col.data = devAlloc(ex, std::max<size_t>(jp.colData[c].size(), 1));
@@ -4040,6 +4047,17 @@ static int32_t joinSpillAdvance(gx_exec* ex) {
     if (rc == GX_OK) rc = uploadJoinPart(ex, pN, ex->spillP[p], &st.probeTab);
     if (rc) return rc;
     st.inputsReady = true;
+    if (getenv("GX_DEBUG")) {
+      int64_t bk = 0, pk = 0;
+      if (ex->spillB[p].n > 0 && ex->spillB[p].colData[0].size() >= 8)
+        std::memcpy(&bk, ex->spillB[p].colData[0].data(), 8);
+      if (ex->spillP[p].n > 0 && ex->spillP[p].colData[0].size() >= 8)
+        std::memcpy(&pk, ex->spillP[p].colData[0].data(), 8);
+      fprintf(stderr,
+              "[gx] spill part %d: build %lld (k0=%lld) probe %lld (k0=%lld)\n",
+              p, (long long)ex->spillB[p].n, (long long)bk,
+              (long long)ex->spillP[p].n, (long long)pk);
+    }
     rc = runJoinStage(ex, st, true);
     if (rc) return rc;
     ex->joinSpillMatches += ex->lastSelCount;

@@ -163,7 +163,17 @@ int32_t gx_pb_streamagg(gx_pb* pb, int32_t child, const int32_t* group_exprs,
 int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
                    const uint8_t* key_desc, int32_t n_keys,
                    int64_t limit, int64_t offset);
-/* join_type: 0 = inner (HashJoinV2 equivalent, pkg/executor/join/hash_join_v2.go) */
+/* join_type (HashJoinV2 equivalent, pkg/executor/join/hash_join_v2.go):
+ *  0 inner (inner_join_probe.go:27-86)
+ *  1 left outer, probe = outer side: unmatched probe rows null-extend the
+ *    build columns (outer_join_probe.go probe-outer path)
+ *  2 right outer, build = outer side: matched pairs plus unmatched build
+ *    rows with probe columns NULL (outer_join_probe.go matched flags)
+ *  3 semi: each probe row once on any match; output = probe columns only
+ *    (base_semi_join.go)
+ *  4 anti semi: each probe row once on NO match (NULL-key probe rows match
+ *    nothing and emit); output = probe columns only
+ *    (anti_semi_join_probe.go, non-null-aware variant) */
 int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
                        const int32_t* build_keys, const int32_t* probe_keys,
                        int32_t n_keys, int32_t join_type);

@@ -318,6 +318,24 @@ struct JoinAggDesc {
   uint64_t* counters = nullptr;  // [0] build0 pass, [1] build1 pass, [2] probe match
   uint32_t* errorFlag = nullptr;
   int32_t wide = 0;
+  // extra filter conjuncts per table beyond the specialized first one
+  // (evaluated through evalSimplePred; the JIT emits a runtime loop)
+  PredDesc pred0x[3], pred1x[3], predPx[3];
+  int32_t nPred0x = 0, nPred1x = 0, nPredPx = 0;
+  // duplicate build1 keys: row-indexed slots (one per build1 row) chained
+  // from a pow2 heads table — the fused-pipeline analog of the standalone
+  // join's heads+next (hash_table_v2.go chains). Engine retries with
+  // chained=1 when the unique-key insert sees a duplicate (kErrBadKey).
+  int32_t chained = 0;
+  uint32_t* b1Heads = nullptr;  // build1 row + 1, 0 = empty
+  int32_t b1HeadsLog2 = 0;
+  uint32_t* b1Next = nullptr;   // per build1 row
+  int64_t nSlots = 0;           // slot entries to scan (chained: build1 rows;
+                                // else 1 << slotsLog2)
+  // 128-bit top-N: the radix-threshold select runs on the order-preserving
+  // projection key64 = (acc128 >> topnShift); candidates compact with the
+  // exact 128-bit acc and the host sorts them exactly
+  int32_t topnShift = 0;
 };
 
 enum { PRED_STR_EQ_CONST = 3 };  // extra PredKind for the join path
@@ -570,6 +588,12 @@ int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
                    void* stream);  // 0 count0 1 build0 2 count1 3 build1 4 probe
 int gxJoinAggMax(const JoinAggDesc* devDesc, const JoinAggDesc& h, uint64_t* devMax,
                  void* stream);
+// pass A of the 128-bit top-N: max accHi over occupied slots -> devMaxHi
+int gxJoinAggMaxHi(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                   uint64_t* devMaxHi, void* stream);
+// chained mode: merge duplicate-(key,payload) slots into one group
+int gxJoinAggMergeDups(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                       void* stream);
 int gxJoinAggHist(const JoinAggDesc* devDesc, const JoinAggDesc& h,
                   uint32_t* devHist, int shift, void* stream);
 int gxJoinAggCompact(const JoinAggDesc* devDesc, const JoinAggDesc& h,

@@ -601,6 +601,48 @@ __device__ inline bool makeWideGroupKey(const FusedQueryDesc& d, int64_t row,
   }
 }
 
+// simple single-table predicate (direct loads; build phases are cheap scans)
+__device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
+                                      const uint8_t* strConst, int strConstLen,
+                                      int64_t row) {
+  const DevCol& c = tab.cols[pd.col];
+  if (colIsNull(c, row)) return false;
+  if (pd.kind == PRED_TIME_CMP_CONST) {
+    uint64_t v = gptr<uint64_t>(c.data)[row] & ~0xFULL;
+    uint64_t k = pd.constU64 & ~0xFULL;
+    return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+  }
+  if (pd.kind == PRED_I64_CMP_CONST) {
+    int64_t v = gptr<int64_t>(c.data)[row];
+    int64_t k = (int64_t)pd.constU64;
+    return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+  }
+  if (pd.kind == PRED_STR_EQ_CONST) {
+    int64_t st, en;
+    if (c.denseOffsets) { st = row; en = row + 1; }
+    else { st = gptr<int64_t>(c.offsets)[row]; en = gptr<int64_t>(c.offsets)[row + 1]; }
+    auto p = gptr<uint8_t>(c.data);
+    while (en > st && p[en - 1] == ' ') en--;  // PAD SPACE
+    int len = (int)(en - st);
+    bool eq = len == strConstLen;
+    for (int j = 0; j < len && eq; j++) eq = p[st + j] == strConst[j];
+    return cmpResult(eq ? 0 : 1, pd.cmp);
+  }
+  if (pd.kind == PRED_STR_LIKE_PREFIX) {
+    // builtinLikeSig 'abc%' fast path: byte prefix, case-sensitive, no pad
+    // trimming (LIKE does not use PAD SPACE semantics)
+    int64_t st, en;
+    if (c.denseOffsets) { st = row; en = row + 1; }
+    else { st = gptr<int64_t>(c.offsets)[row]; en = gptr<int64_t>(c.offsets)[row + 1]; }
+    auto p = gptr<uint8_t>(c.data);
+    if (en - st < strConstLen) return false;
+    for (int j = 0; j < strConstLen; j++)
+      if (p[st + j] != strConst[j]) return false;
+    return true;
+  }
+  return false;
+}
+
 // atomic int128 + count accumulation into a slot (LDS or global)
 template <typename SlotT>
 __device__ inline void accumInto(SlotT* slot, int a, Int128 v, int64_t dc) {

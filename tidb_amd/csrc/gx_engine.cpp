@@ -991,24 +991,41 @@ static int32_t compileJoinAgg(gx_exec* ex) {
   ja.b1KeyCol = bk2.colIdx - nc;
   ja.pKeyCol = pk2.colIdx;
 
-  // predicates (0 or 1 conjunct per table this round)
+  // predicates: 1-4 CNF conjuncts per table (the first is specialized into
+  // the tuned kernels/JIT; extras evaluate through the evalSimplePred loop)
   auto doPred = [&](const PNode* sel, const PNode& srcNode, gxp::PredDesc* pd,
-                    int32_t* n) -> bool {
+                    int32_t* n, gxp::PredDesc* px, int32_t* nx) -> bool {
     *n = 0;
+    *nx = 0;
     if (!sel) return true;
-    if (sel->exprs.size() != 1) {
-      ex->err = "device join path supports one filter conjunct per table";
+    if (sel->exprs.empty() || sel->exprs.size() > 4) {
+      ex->err = "device join path supports 1-4 filter conjuncts per table";
       return false;
     }
-    if (!compileTablePred(ex, srcNode, sel->exprs[0], pd, ja.strConst,
-                          &ja.strConstLen))
-      return false;
+    for (size_t j = 0; j < sel->exprs.size(); j++) {
+      gxp::PredDesc* dst = j == 0 ? pd : &px[j - 1];
+      int32_t before = ja.strConstLen;
+      if (!compileTablePred(ex, srcNode, sel->exprs[j], dst, ja.strConst,
+                            &ja.strConstLen))
+        return false;
+      if ((dst->kind == gxp::PRED_STR_EQ_CONST ||
+           dst->kind == gxp::PRED_STR_LIKE_PREFIX) &&
+          before != 0) {
+        // one shared 16-byte string constant per join-agg plan this round
+        ex->err = "one string filter constant per device join-agg plan";
+        return false;
+      }
+    }
     *n = 1;
+    *nx = (int32_t)sel->exprs.size() - 1;
     return true;
   };
-  if (!doPred(selC, custN, &ja.pred0, &ja.nPred0)) return GX_ERR_INVALID;
-  if (!doPred(selO, ordN, &ja.pred1, &ja.nPred1)) return GX_ERR_INVALID;
-  if (!doPred(selL, liN, &ja.predP, &ja.nPredP)) return GX_ERR_INVALID;
+  if (!doPred(selC, custN, &ja.pred0, &ja.nPred0, ja.pred0x, &ja.nPred0x))
+    return GX_ERR_INVALID;
+  if (!doPred(selO, ordN, &ja.pred1, &ja.nPred1, ja.pred1x, &ja.nPred1x))
+    return GX_ERR_INVALID;
+  if (!doPred(selL, liN, &ja.predP, &ja.nPredP, ja.predPx, &ja.nPredPx))
+    return GX_ERR_INVALID;
 
   // projection classification
   ja.payloadCol0 = ja.payloadCol1 = -1;
@@ -3278,12 +3295,13 @@ static int32_t runJoinAgg(gx_exec* ex) {
   uint64_t n1 = 0;
   if ((rc = readCounter(1, &n1))) return rc;
   wantLog2 = ceilLog2(std::max<uint64_t>(2 * n1 + 1, 64));
-  if (ja.slots == nullptr || ja.slotsLog2 != wantLog2) {
+  if (!ja.chained && (ja.slots == nullptr || ja.slotsLog2 != wantLog2)) {
     ja.slots = (gxp::JoinAggSlot*)devAlloc(
         ex, (1ULL << wantLog2) * sizeof(gxp::JoinAggSlot));
     if (!ja.slots) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
     ja.slotsLog2 = wantLog2;
   }
+  ja.nSlots = ja.chained ? ja.build1.nRows : (1LL << ja.slotsLog2);
   int bloomLog2 = ceilLog2(std::max<uint64_t>(8 * n1 + 1, 1024));
   if (ja.bloom == nullptr || ja.bloomLog2 != bloomLog2) {
     ja.bloom = (uint32_t*)devAlloc(ex, (1ULL << bloomLog2) / 8);
@@ -3295,6 +3313,42 @@ static int32_t runJoinAgg(gx_exec* ex) {
   if ((rc = pushDesc())) return rc;
   if ((rc = phase(5))) return rc;  // init slots
   if ((rc = phase(3))) return rc;  // build
+  if (!ja.chained) {
+    // duplicate build keys? the unique-key insert flags kErrBadKey — retry
+    // with the CHAINED slot layout (one slot per build1 row, heads+next
+    // chains; each duplicate (key, payloads) row is its own group)
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    uint32_t ef = 0;
+    HIP_OK(ex, hipMemcpy(&ef, ex->devErr, 4, hipMemcpyDeviceToHost));
+    if (ef & 2u /*kErrBadKey*/) {
+      ja.chained = 1;
+      ja.b1HeadsLog2 = ceilLog2(std::max<uint64_t>(2 * n1 + 1, 64));
+      ja.b1Heads = (uint32_t*)devAlloc(ex, (1ULL << ja.b1HeadsLog2) * 4);
+      ja.b1Next = (uint32_t*)devAlloc(
+          ex, std::max<int64_t>(ja.build1.nRows, 1) * 4);
+      ja.slots = (gxp::JoinAggSlot*)devAlloc(
+          ex, std::max<int64_t>(ja.build1.nRows, 1) *
+                  sizeof(gxp::JoinAggSlot));
+      if (!ja.b1Heads || !ja.b1Next || !ja.slots) {
+        ex->err = "hipMalloc failed (chained join-agg)";
+        return GX_ERR_INTERNAL;
+      }
+      ja.nSlots = ja.build1.nRows;
+      // the chained variants run interpreted (no specialization yet)
+      ex->jaJitProg = nullptr;
+      ex->jaJitTried = true;
+      ex->jaBuildProg = nullptr;
+      ex->jaBuildTried = true;
+      HIP_OK(ex, hipMemsetAsync(ja.b1Heads, 0, (1ULL << ja.b1HeadsLog2) * 4,
+                                ex->stream));
+      HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
+      if (getenv("GX_DEBUG"))
+        fprintf(stderr, "[gx] duplicate build keys -> chained join-agg\n");
+      if ((rc = pushDesc())) return rc;
+      if ((rc = phase(5))) return rc;  // re-init row-indexed slots
+      if ((rc = phase(3))) return rc;  // chained build
+    }
+  }
 
   // specialize the probe kernel once per executor (gx_jit.cpp)
   if (!ex->jaJitTried && !getenv("GX_NO_JIT")) {
@@ -3343,6 +3397,15 @@ static int32_t runJoinAgg(gx_exec* ex) {
   HIP_OK(ex, hipMemcpy(&ex->lastSelCount, ja.counters + 2, 8,
                        hipMemcpyDeviceToHost));
 
+  if (ja.chained) {
+    // duplicate build rows with identical (key, payloads) are ONE group in
+    // the reference — merge them before the top-N select
+    if (gxp::gxJoinAggMergeDups(ex->devJa, ja, ex->stream) != 0) {
+      ex->err = "dup-merge kernel failed";
+      return GX_ERR_INTERNAL;
+    }
+  }
+
   // top-N selection: max -> 4096-bucket histogram -> threshold -> compact
   uint64_t* devMax = (uint64_t*)devAlloc(ex, 8);
   uint32_t* devHist = (uint32_t*)devAlloc(ex, 4096 * 4);
@@ -3354,6 +3417,30 @@ static int32_t runJoinAgg(gx_exec* ex) {
   HIP_OK(ex, hipMemsetAsync(devMax, 0, 8, ex->stream));
   HIP_OK(ex, hipMemsetAsync(devHist, 0, 4096 * 4, ex->stream));
   HIP_OK(ex, hipMemsetAsync(devCount, 0, 8, ex->stream));
+  // pass A of the 128-bit-capable select: max acc HI word -> topnShift so
+  // key64 = acc128 >> shift is an order-preserving u64 (exact candidates
+  // still compact with the full 128-bit acc; the host sorts them exactly)
+  {
+    uint64_t* devMaxHi = (uint64_t*)devAlloc(ex, 8);
+    if (!devMaxHi) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    HIP_OK(ex, hipMemsetAsync(devMaxHi, 0, 8, ex->stream));
+    if (gxp::gxJoinAggMaxHi(ex->devJa, ja, devMaxHi, ex->stream) != 0) {
+      ex->err = "maxhi kernel failed";
+      return GX_ERR_INTERNAL;
+    }
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    uint64_t maxHi = 0;
+    HIP_OK(ex, hipMemcpy(&maxHi, devMaxHi, 8, hipMemcpyDeviceToHost));
+    HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
+    if (errFlag != 0) {
+      ex->err = "negative revenue in device top-N unsupported this round";
+      return GX_ERR_INTERNAL;
+    }
+    int bits = 0;
+    while ((maxHi >> bits) != 0) bits++;
+    ja.topnShift = bits;
+    if (bits) { if ((rc = pushDesc())) return rc; }
+  }
   if (gxp::gxJoinAggMax(ex->devJa, ja, devMax, ex->stream) != 0) {
     ex->err = "max kernel failed";
     return GX_ERR_INTERNAL;
@@ -3361,12 +3448,6 @@ static int32_t runJoinAgg(gx_exec* ex) {
   HIP_OK(ex, hipStreamSynchronize(ex->stream));
   uint64_t maxRev = 0;
   HIP_OK(ex, hipMemcpy(&maxRev, devMax, 8, hipMemcpyDeviceToHost));
-  HIP_OK(ex, hipMemcpy(&errFlag, ex->devErr, 4, hipMemcpyDeviceToHost));
-  if (errFlag != 0) {
-    // an accumulator exceeded 64 bits: the histogram shortcut is invalid
-    ex->err = "join revenue exceeds 64-bit fast top-N (next round: 128-bit select)";
-    return GX_ERR_INTERNAL;
-  }
   int shift = 0;
   {
     int bits = 0;

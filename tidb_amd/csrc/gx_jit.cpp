@@ -794,6 +794,10 @@ const JitProg* compileJa(const JoinAggDesc& d, std::string* whyNot) {
     if (whyNot) *whyNot = "probe predicate kind not specialized";
     return nullptr;
   }
+  if (d.chained || d.nPredPx > 0) {
+    if (whyNot) *whyNot = "chained/multi-conjunct probe not specialized";
+    return nullptr;
+  }
   std::string src = generateJaSource(d);
   return compileSource(src, "genja_narrow", "genja_wide", whyNot);
 }
@@ -1052,6 +1056,10 @@ static std::map<std::string, JaBuildProg>& buildCache() {
 }
 
 const JaBuildProg* compileJaBuild(const JoinAggDesc& d, std::string* whyNot) {
+  if (d.chained || d.nPred0x > 0 || d.nPred1x > 0) {
+    if (whyNot) *whyNot = "chained/multi-conjunct build not specialized";
+    return nullptr;
+  }
   std::string src = generateJaBuildSource(d);
   std::lock_guard<std::mutex> lk(cacheMu);
   auto it = buildCache().find(src);

@@ -614,48 +614,6 @@ __global__ void initGlobalTableKernel(GroupSlot* table, int n) {
 // join-aggregate pipeline (Q3 class) — see gx_common.h JoinAggDesc
 // ==================================================================
 
-// simple single-table predicate (direct loads; build phases are cheap scans)
-__device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
-                                      const uint8_t* strConst, int strConstLen,
-                                      int64_t row) {
-  const DevCol& c = tab.cols[pd.col];
-  if (colIsNull(c, row)) return false;
-  if (pd.kind == PRED_TIME_CMP_CONST) {
-    uint64_t v = gptr<uint64_t>(c.data)[row] & ~0xFULL;
-    uint64_t k = pd.constU64 & ~0xFULL;
-    return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
-  }
-  if (pd.kind == PRED_I64_CMP_CONST) {
-    int64_t v = gptr<int64_t>(c.data)[row];
-    int64_t k = (int64_t)pd.constU64;
-    return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
-  }
-  if (pd.kind == PRED_STR_EQ_CONST) {
-    int64_t st, en;
-    if (c.denseOffsets) { st = row; en = row + 1; }
-    else { st = gptr<int64_t>(c.offsets)[row]; en = gptr<int64_t>(c.offsets)[row + 1]; }
-    auto p = gptr<uint8_t>(c.data);
-    while (en > st && p[en - 1] == ' ') en--;  // PAD SPACE
-    int len = (int)(en - st);
-    bool eq = len == strConstLen;
-    for (int j = 0; j < len && eq; j++) eq = p[st + j] == strConst[j];
-    return cmpResult(eq ? 0 : 1, pd.cmp);
-  }
-  if (pd.kind == PRED_STR_LIKE_PREFIX) {
-    // builtinLikeSig 'abc%' fast path: byte prefix, case-sensitive, no pad
-    // trimming (LIKE does not use PAD SPACE semantics)
-    int64_t st, en;
-    if (c.denseOffsets) { st = row; en = row + 1; }
-    else { st = gptr<int64_t>(c.offsets)[row]; en = gptr<int64_t>(c.offsets)[row + 1]; }
-    auto p = gptr<uint8_t>(c.data);
-    if (en - st < strConstLen) return false;
-    for (int j = 0; j < strConstLen; j++)
-      if (p[st + j] != strConst[j]) return false;
-    return true;
-  }
-  return false;
-}
-
 __device__ inline uint64_t hashKey(uint64_t k) { return splitmix64(k); }
 
 __device__ inline void bloomSet(const JoinAggDesc& d, uint64_t key) {
@@ -699,6 +657,26 @@ __device__ inline bool bloomMayHave(const JoinAggDesc& d, uint64_t key) {
 }
 
 // count rows of build0 passing its predicate
+// extra filter conjuncts beyond the specialized first one (VectorizedFilter
+// CNF semantics: every conjunct must pass; NULL rejects)
+__device__ inline bool jaExtraPreds(const DevTable& t, const PredDesc* px,
+                                    int n, const uint8_t* sc, int scLen,
+                                    int64_t row) {
+  for (int j = 0; j < n; j++)
+    if (!evalSimplePred(t, px[j], sc, scLen, row)) return false;
+  return true;
+}
+
+// order-preserving u64 projection of the (non-negative) 128-bit accumulator
+// for the radix-threshold top-N; exact candidates compact with the full
+// 128-bit value and the host sorts them exactly
+__device__ inline uint64_t jaKey64(const JoinAggDesc& d, int64_t i) {
+  unsigned __int128 a =
+      ((unsigned __int128)(uint64_t)gptr<int64_t>(&d.slots[i].accHi)[0] << 64) |
+      gptr<uint64_t>(&d.slots[i].accLo)[0];
+  return (uint64_t)(a >> d.topnShift);
+}
+
 __global__ void jaCountBuild0Kernel(const JoinAggDesc* __restrict__ dp) {
   const JoinAggDesc& d = *dp;
   int64_t n = d.build0.nRows;
@@ -707,6 +685,9 @@ __global__ void jaCountBuild0Kernel(const JoinAggDesc* __restrict__ dp) {
        row += (int64_t)gridDim.x * blockDim.x) {
     bool pass = d.nPred0 == 0 ||
                 evalSimplePred(d.build0, d.pred0, d.strConst, d.strConstLen, row);
+    if (pass && d.nPred0x)
+      pass = jaExtraPreds(d.build0, d.pred0x, d.nPred0x, d.strConst,
+                          d.strConstLen, row);
     if (pass) my++;
   }
   for (int off = 32; off > 0; off >>= 1) my += __shfl_down(my, off, 64);
@@ -723,6 +704,9 @@ __global__ void jaBuild0Kernel(const JoinAggDesc* __restrict__ dp) {
        row += (int64_t)gridDim.x * blockDim.x) {
     bool pass = d.nPred0 == 0 ||
                 evalSimplePred(d.build0, d.pred0, d.strConst, d.strConstLen, row);
+    if (pass && d.nPred0x)
+      pass = jaExtraPreds(d.build0, d.pred0x, d.nPred0x, d.strConst,
+                          d.strConstLen, row);
     if (!pass) continue;
     if (colIsNull(d.build0.cols[d.b0KeyCol], row)) continue;  // NULL never joins
     uint64_t key = gptr<uint64_t>(d.build0.cols[d.b0KeyCol].data)[row];
@@ -766,6 +750,9 @@ __global__ void jaCountBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
        row += (int64_t)gridDim.x * blockDim.x) {
     bool pass = d.nPred1 == 0 ||
                 evalSimplePred(d.build1, d.pred1, d.strConst, d.strConstLen, row);
+    if (pass && d.nPred1x)
+      pass = jaExtraPreds(d.build1, d.pred1x, d.nPred1x, d.strConst,
+                          d.strConstLen, row);
     if (!pass) continue;
     const DevCol& kc = d.build1.cols[d.b1ProbeCol];
     if (colIsNull(kc, row)) continue;
@@ -780,6 +767,7 @@ __global__ void jaCountBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
 }
 
 // build the slot table from qualifying build1 rows
+template <bool CHAIN = false>
 __global__ void jaBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
   const JoinAggDesc& d = *dp;
   int64_t n = d.build1.nRows;
@@ -788,6 +776,9 @@ __global__ void jaBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
        row += (int64_t)gridDim.x * blockDim.x) {
     bool pass = d.nPred1 == 0 ||
                 evalSimplePred(d.build1, d.pred1, d.strConst, d.strConstLen, row);
+    if (pass && d.nPred1x)
+      pass = jaExtraPreds(d.build1, d.pred1x, d.nPred1x, d.strConst,
+                          d.strConstLen, row);
     if (!pass) continue;
     const DevCol& pc = d.build1.cols[d.b1ProbeCol];
     if (colIsNull(pc, row)) continue;
@@ -803,12 +794,31 @@ __global__ void jaBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
     int64_t pay1 = d.payloadCol1 >= 0
         ? gptr<int64_t>(d.build1.cols[d.payloadCol1].data)[row] : 0;
     bloomSet(d, key);
+    if (CHAIN) {
+      // duplicate-key mode: one slot PER BUILD1 ROW (each (key, payloads)
+      // row is its own group), chained from the pow2 heads table like the
+      // standalone join (hash_table_v2.go chains)
+      JoinAggSlot& sp2 = d.slots[row];
+      sp2.key = key;
+      sp2.payload0 = pay0;
+      sp2.payload1 = pay1;
+      uint32_t hmask = (1u << d.b1HeadsLog2) - 1;
+      uint32_t hslot = (uint32_t)(hashKey(key) & hmask);
+      uint32_t newHead = (uint32_t)row + 1;
+      uint32_t old = d.b1Heads[hslot];
+      for (;;) {
+        d.b1Next[row] = old;
+        uint32_t prev = atomicCAS(&d.b1Heads[hslot], old, newHead);
+        if (prev == old) break;
+        old = prev;
+      }
+      continue;
+    }
     uint32_t slot = (uint32_t)(hashKey(key) & mask);
     for (uint32_t probe = 0; probe <= mask; probe++) {
       uint64_t cur = d.slots[slot].key;
       if (cur == key) {
-        // duplicate build key (one row per orderkey in Q3; general inner join
-        // with duplicate build keys needs chaining — next round)
+        // duplicate build key: the engine retries with the chained variant
         atomicOr(d.errorFlag, kErrBadKey);
         break;
       }
@@ -833,7 +843,7 @@ __global__ void jaBuild1Kernel(const JoinAggDesc* __restrict__ dp) {
 // Payload writes may race with probe reads only via the key CAS (published
 // before probing starts: build and probe are separate kernel launches, so
 // visibility comes from the dispatch boundary, not intra-kernel hand-off).
-template <bool WIDE>
+template <bool WIDE, bool CHAIN = false>
 __launch_bounds__(256)
 __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
   const JoinAggDesc& d = *dp;
@@ -869,6 +879,9 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
         pass = evalSimplePred(d.probe, pd, d.strConst, d.strConstLen, row);
       }
     }
+    if (pass && d.nPredPx)
+      pass = jaExtraPreds(d.probe, d.predPx, d.nPredPx, d.strConst,
+                          d.strConstLen, row);
     if (!pass) continue;
     const DevCol& kc = d.probe.cols[d.pKeyCol];
     if (colIsNull(kc, row)) continue;
@@ -876,15 +889,30 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
     if (key == kEmptyKey) key = kEmptyKey - 1;
     if (!bloomMayHave(d, key)) continue;  // L2-resident reject
     // probe the slot table (prebuilt: no inserts, miss -> drop row)
-    uint32_t slot = (uint32_t)(hashKey(key) & mask);
-    bool found = false;
-    for (uint32_t probe = 0; probe <= mask; probe++) {
-      uint64_t cur = gptr<uint64_t>(&d.slots[slot].key)[0];
-      if (cur == key) { found = true; break; }
-      if (cur == kEmptyKey) break;
-      slot = (slot + 1) & mask;
+    uint32_t slot = 0;
+    uint32_t firstCur = 0;
+    if (CHAIN) {
+      uint32_t hmask = (1u << d.b1HeadsLog2) - 1;
+      uint32_t cur = gptr<uint32_t>(d.b1Heads)[(uint32_t)(hashKey(key) & hmask)];
+      while (cur != 0) {
+        if (gptr<uint64_t>(&d.slots[cur - 1].key)[0] == key) {
+          firstCur = cur;
+          break;
+        }
+        cur = gptr<uint32_t>(d.b1Next)[cur - 1];
+      }
+      if (firstCur == 0) continue;
+    } else {
+      slot = (uint32_t)(hashKey(key) & mask);
+      bool found = false;
+      for (uint32_t probe = 0; probe <= mask; probe++) {
+        uint64_t cur = gptr<uint64_t>(&d.slots[slot].key)[0];
+        if (cur == key) { found = true; break; }
+        if (cur == kEmptyKey) break;
+        slot = (slot + 1) & mask;
+      }
+      if (!found) continue;
     }
-    if (!found) continue;
     // VM: compute the summed value
     VmState<WIDE> vm;
     vm.nullBits = 0;
@@ -972,6 +1000,27 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
     if (bad) { failed = true; break; }
     if (vm.isNull(d.valueReg)) continue;  // NULL never enters the sum
     Int128 v = VT<WIDE>::toAcc(vm.get(d.valueReg));
+    if (CHAIN) {
+      // every duplicate build row with this key is its own group: the probe
+      // row's value accumulates into each (one pair per build row)
+      for (uint32_t cur = firstCur; cur != 0;
+           cur = gptr<uint32_t>(d.b1Next)[cur - 1]) {
+        uint32_t brow = cur - 1;
+        if (gptr<uint64_t>(&d.slots[brow].key)[0] != key) continue;
+        JoinAggSlot* sp = &d.slots[brow];
+        uint64_t old = atomicAdd((unsigned long long*)&sp->accLo,
+                                 (unsigned long long)v.lo);
+        uint64_t carry = (old + v.lo) < old ? 1 : 0;
+        int64_t hiAdd = v.hi + (int64_t)carry;
+        if (hiAdd != 0)
+          atomicAdd((unsigned long long*)&sp->accHi,
+                    (unsigned long long)hiAdd);
+        if (v.hi < 0 || (v.hi == 0 && v.lo == 0))
+          atomicAdd((unsigned long long*)&sp->cnt, 1ULL);
+        myMatch++;
+      }
+      continue;
+    }
     JoinAggSlot* sp = &d.slots[slot];
     uint64_t old = atomicAdd((unsigned long long*)&sp->accLo, (unsigned long long)v.lo);
     uint64_t carry = (old + v.lo) < old ? 1 : 0;
@@ -991,23 +1040,85 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
     atomicAdd((unsigned long long*)&d.counters[2], (unsigned long long)myMatch);
 }
 
-// top-N selection support: max accLo (revenues fit u64 in practice; accHi != 0
-// falls back to host full sort via the error-free "big" path)
-__global__ void jaMaxKernel(const JoinAggDesc* __restrict__ dp, uint64_t* outMax) {
+// top-N selection: pass A finds the max ACC HI WORD (and flags negative
+// accumulators — the order-preserving shift projection needs non-negative
+// revenues; Q3's price*(1-disc) always is); the engine derives topnShift so
+// key64 = acc128 >> shift fits u64; pass B (jaMaxKernel) then maxes key64.
+// chained mode: duplicate build1 rows with IDENTICAL (key, payloads) are
+// ONE group in the reference (each probe row matched both, so the sums
+// add); merge every such slot into the chain-first canonical slot. Only the
+// owning thread reads/clears its own slot; everyone adds into canonicals.
+__global__ void jaMergeDupSlotsKernel(const JoinAggDesc* __restrict__ dp) {
   const JoinAggDesc& d = *dp;
-  int64_t n = 1LL << d.slotsLog2;
+  int64_t n = d.nSlots;
+  uint32_t hmask = (1u << d.b1HeadsLog2) - 1;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint64_t key = gptr<uint64_t>(&d.slots[i].key)[0];
+    if (key == kEmptyKey) continue;
+    uint64_t p0 = gptr<uint64_t>(&d.slots[i].payload0)[0];
+    int64_t p1 = gptr<int64_t>(&d.slots[i].payload1)[0];
+    uint32_t canon = 0;
+    for (uint32_t cur =
+             gptr<uint32_t>(d.b1Heads)[(uint32_t)(hashKey(key) & hmask)];
+         cur != 0; cur = gptr<uint32_t>(d.b1Next)[cur - 1]) {
+      uint32_t r = cur - 1;
+      if (gptr<uint64_t>(&d.slots[r].key)[0] == key &&
+          gptr<uint64_t>(&d.slots[r].payload0)[0] == p0 &&
+          gptr<int64_t>(&d.slots[r].payload1)[0] == p1) {
+        canon = cur;
+        break;
+      }
+    }
+    if (canon == 0 || (int64_t)(canon - 1) == i) continue;
+    JoinAggSlot* dst = &d.slots[canon - 1];
+    const JoinAggSlot& s = d.slots[i];
+    uint64_t old = atomicAdd((unsigned long long*)&dst->accLo,
+                             (unsigned long long)s.accLo);
+    uint64_t carry = (old + s.accLo) < old ? 1 : 0;
+    int64_t hiAdd = s.accHi + (int64_t)carry;
+    if (hiAdd != 0)
+      atomicAdd((unsigned long long*)&dst->accHi, (unsigned long long)hiAdd);
+    if (s.cnt)
+      atomicAdd((unsigned long long*)&dst->cnt, (unsigned long long)s.cnt);
+    d.slots[i].key = kEmptyKey;  // only this thread touches slot i
+  }
+}
+
+__global__ void jaMaxHiKernel(const JoinAggDesc* __restrict__ dp,
+                              uint64_t* outMaxHi) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.nSlots;
   uint64_t my = 0;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint64_t skey = gptr<uint64_t>(&d.slots[i].key)[0];
     if (skey == kEmptyKey) continue;
-    uint64_t accLo = gptr<uint64_t>(&d.slots[i].accLo)[0];
-    if (accLo == 0 && gptr<uint64_t>(&d.slots[i].accHi)[0] == 0 &&
+    int64_t hi = gptr<int64_t>(&d.slots[i].accHi)[0];
+    if (hi < 0) atomicOr(d.errorFlag, kErrOverflow);  // negative revenue
+    if ((uint64_t)hi > my) my = (uint64_t)hi;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    uint64_t o = __shfl_down(my, off, 64);
+    if (o > my) my = o;
+  }
+  if ((threadIdx.x & 63) == 0) atomicMax((unsigned long long*)outMaxHi, my);
+}
+
+__global__ void jaMaxKernel(const JoinAggDesc* __restrict__ dp, uint64_t* outMax) {
+  const JoinAggDesc& d = *dp;
+  int64_t n = d.nSlots;
+  uint64_t my = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint64_t skey = gptr<uint64_t>(&d.slots[i].key)[0];
+    if (skey == kEmptyKey) continue;
+    if (gptr<uint64_t>(&d.slots[i].accLo)[0] == 0 &&
+        gptr<uint64_t>(&d.slots[i].accHi)[0] == 0 &&
         gptr<uint64_t>(&d.slots[i].cnt)[0] == 0)
       continue;
-    if (gptr<uint64_t>(&d.slots[i].accHi)[0] != 0)
-      atomicOr(d.errorFlag, kErrOverflow);
-    if (accLo > my) my = accLo;
+    uint64_t k64 = jaKey64(d, i);
+    if (k64 > my) my = k64;
   }
   for (int off = 32; off > 0; off >>= 1) {
     uint64_t o = __shfl_down(my, off, 64);
@@ -1019,7 +1130,7 @@ __global__ void jaMaxKernel(const JoinAggDesc* __restrict__ dp, uint64_t* outMax
 __global__ void jaHistKernel(const JoinAggDesc* __restrict__ dp, uint32_t* hist,
                              int shift) {
   const JoinAggDesc& d = *dp;
-  int64_t n = 1LL << d.slotsLog2;
+  int64_t n = d.nSlots;
   // ~14M occupied slots funneling into 4096 bins: privatize the histogram
   // in LDS per block and flush once (the global-atomic version measured
   // 6.4 ms of pure contention, 25x the scan floor)
@@ -1035,7 +1146,7 @@ __global__ void jaHistKernel(const JoinAggDesc* __restrict__ dp, uint32_t* hist,
     if (accLo == 0 && gptr<uint64_t>(&d.slots[i].accHi)[0] == 0 &&
         gptr<uint64_t>(&d.slots[i].cnt)[0] == 0)
       continue;
-    __hip_atomic_fetch_add(&lh3[(accLo >> shift) & 4095], 1u,
+    __hip_atomic_fetch_add(&lh3[(jaKey64(d, i) >> shift) & 4095], 1u,
                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
   }
   __syncthreads();
@@ -1047,14 +1158,14 @@ __global__ void jaCompactKernel(const JoinAggDesc* __restrict__ dp, TopNOut* out
                                 uint64_t* outCount, uint64_t thresholdBucket,
                                 int shift, uint64_t cap) {
   const JoinAggDesc& d = *dp;
-  int64_t n = 1LL << d.slotsLog2;
+  int64_t n = d.nSlots;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint64_t skey = gptr<uint64_t>(&d.slots[i].key)[0];
     if (skey == kEmptyKey) continue;
     const JoinAggSlot& s = d.slots[i];
     if (s.cnt == 0 && s.accLo == 0 && s.accHi == 0) continue;
-    if ((s.accLo >> shift) < thresholdBucket) continue;
+    if ((jaKey64(d, i) >> shift) < thresholdBucket) continue;
     uint64_t idx = atomicAdd((unsigned long long*)outCount, 1ULL);
     if (idx >= cap) { atomicOr(d.errorFlag, kErrGlobalFull); continue; }
     out[idx] = {s.key, s.payload0, s.payload1, s.accLo, s.accHi};
@@ -1649,19 +1760,34 @@ int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
                          dim3(256), 0, s, devDesc);
       break;
     case 3:
-      hipLaunchKernelGGL(jaBuild1Kernel, dim3(gridFor(h.build1.nRows)),
-                         dim3(256), 0, s, devDesc);
+      if (h.chained)
+        hipLaunchKernelGGL((jaBuild1Kernel<true>),
+                           dim3(gridFor(h.build1.nRows)), dim3(256), 0, s,
+                           devDesc);
+      else
+        hipLaunchKernelGGL((jaBuild1Kernel<false>),
+                           dim3(gridFor(h.build1.nRows)), dim3(256), 0, s,
+                           devDesc);
       break;
     case 4:
-      if (h.wide)
-        hipLaunchKernelGGL((jaProbeKernel<true>), dim3(gridFor(h.probe.nRows)),
+      if (h.chained) {
+        if (h.wide)
+          hipLaunchKernelGGL((jaProbeKernel<true, true>),
+                             dim3(gridFor(h.probe.nRows)), dim3(256), 0, s,
+                             devDesc);
+        else
+          hipLaunchKernelGGL((jaProbeKernel<false, true>),
+                             dim3(gridFor(h.probe.nRows)), dim3(256), 0, s,
+                             devDesc);
+      } else if (h.wide)
+        hipLaunchKernelGGL((jaProbeKernel<true, false>), dim3(gridFor(h.probe.nRows)),
                            dim3(256), 0, s, devDesc);
       else
-        hipLaunchKernelGGL((jaProbeKernel<false>), dim3(gridFor(h.probe.nRows)),
+        hipLaunchKernelGGL((jaProbeKernel<false, false>), dim3(gridFor(h.probe.nRows)),
                            dim3(256), 0, s, devDesc);
       break;
     case 5: {
-      int64_t n = 1LL << h.slotsLog2;
+      int64_t n = h.nSlots > 0 ? h.nSlots : (1LL << h.slotsLog2);
       hipLaunchKernelGGL(jaInitSlotsKernel, dim3((n + 255) / 256), dim3(256), 0, s,
                          h.slots, n);
       break;
@@ -1674,14 +1800,28 @@ int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
 
 int gxJoinAggMax(const JoinAggDesc* devDesc, const JoinAggDesc& h, uint64_t* devMax,
                  void* stream) {
-  hipLaunchKernelGGL(jaMaxKernel, dim3(gridFor(1LL << h.slotsLog2)), dim3(256), 0,
+  hipLaunchKernelGGL(jaMaxKernel, dim3(gridFor(h.nSlots)), dim3(256), 0,
                      (hipStream_t)stream, devDesc, devMax);
+  return (int)hipGetLastError();
+}
+
+int gxJoinAggMergeDups(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                       void* stream) {
+  hipLaunchKernelGGL(jaMergeDupSlotsKernel, dim3(gridFor(h.nSlots)), dim3(256),
+                     0, (hipStream_t)stream, devDesc);
+  return (int)hipGetLastError();
+}
+
+int gxJoinAggMaxHi(const JoinAggDesc* devDesc, const JoinAggDesc& h,
+                   uint64_t* devMaxHi, void* stream) {
+  hipLaunchKernelGGL(jaMaxHiKernel, dim3(gridFor(h.nSlots)), dim3(256), 0,
+                     (hipStream_t)stream, devDesc, devMaxHi);
   return (int)hipGetLastError();
 }
 
 int gxJoinAggHist(const JoinAggDesc* devDesc, const JoinAggDesc& h,
                   uint32_t* devHist, int shift, void* stream) {
-  hipLaunchKernelGGL(jaHistKernel, dim3(gridFor(1LL << h.slotsLog2)), dim3(256), 0,
+  hipLaunchKernelGGL(jaHistKernel, dim3(gridFor(h.nSlots)), dim3(256), 0,
                      (hipStream_t)stream, devDesc, devHist, shift);
   return (int)hipGetLastError();
 }
@@ -1689,7 +1829,7 @@ int gxJoinAggHist(const JoinAggDesc* devDesc, const JoinAggDesc& h,
 int gxJoinAggCompact(const JoinAggDesc* devDesc, const JoinAggDesc& h,
                      TopNOut* out, uint64_t* outCount, uint64_t thresholdBucket,
                      int shift, uint64_t cap, void* stream) {
-  hipLaunchKernelGGL(jaCompactKernel, dim3(gridFor(1LL << h.slotsLog2)), dim3(256),
+  hipLaunchKernelGGL(jaCompactKernel, dim3(gridFor(h.nSlots)), dim3(256),
                      0, (hipStream_t)stream, devDesc, out, outCount,
                      thresholdBucket, shift, cap);
   return (int)hipGetLastError();

@@ -130,7 +130,13 @@ def _run(lib, custs, ords, lis, extra=False, limit=10):
     ex.close()
     ex.free()
     b.free()
-    return out
+    # TopN order among FULL ties (same revenue, same orderdate) is
+    # unspecified in the reference too — compare tie-insensitively, but
+    # check the ordering keys themselves are ordered
+    from fractions import Fraction
+    keys = [(-Fraction(r[3]), r[1]) for r in out]
+    assert keys == sorted(keys)
+    return sorted(out)
 
 
 def _data(dup_payloads="distinct", big=False, n_extra_li=60):

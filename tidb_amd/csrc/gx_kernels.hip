@@ -1788,6 +1788,10 @@ int gxJoinAggPhase(int phase, const JoinAggDesc* devDesc, const JoinAggDesc& h,
       break;
     case 5: {
       int64_t n = h.nSlots > 0 ? h.nSlots : (1LL << h.slotsLog2);
+      // chained mode: a re-run of the build (wide retry, re-open) MUST see
+      // empty chains — stale heads would self-link rows into cycles
+      if (h.chained && h.b1Heads)
+        hipMemsetAsync(h.b1Heads, 0, (1ULL << h.b1HeadsLog2) * 4, s);
       hipLaunchKernelGGL(jaInitSlotsKernel, dim3((n + 255) / 256), dim3(256), 0, s,
                          h.slots, n);
       break;

@@ -3437,7 +3437,7 @@ static int32_t runJoinAgg(gx_exec* ex) {
       return GX_ERR_INTERNAL;
     }
     int bits = 0;
-    while ((maxHi >> bits) != 0) bits++;
+    while (bits < 64 && (maxHi >> bits) != 0) bits++;
     ja.topnShift = bits;
     if (bits) { if ((rc = pushDesc())) return rc; }
   }
@@ -3451,7 +3451,7 @@ static int32_t runJoinAgg(gx_exec* ex) {
   int shift = 0;
   {
     int bits = 0;
-    while ((maxRev >> bits) != 0) bits++;
+    while (bits < 64 && (maxRev >> bits) != 0) bits++;  // bit63 can be set
     shift = bits > 12 ? bits - 12 : 0;
   }
   if (gxp::gxJoinAggHist(ex->devJa, ja, devHist, shift, ex->stream) != 0) {

@@ -145,3 +145,55 @@ def test_outer_join_string_cols_parity():
         want = _run(load_oracle(), jt, brows, prows, **args)
         got = _run(load_product(), jt, brows, prows, **args)
         assert got == want, f"join_type {jt}"
+
+
+def _run_outer_sorted(lib, desc):
+    """ORDER BY a NULLABLE key (the null-extended build payload of a left
+    outer join): sortexec compare semantics — NULL < any value, so NULLs
+    come FIRST ascending and LAST descending."""
+    brows = [[1, 10], [2, 20], [5, 50]]
+    prows = [[1, 100], [2, 200], [3, 300], [None, 400], [5, 500], [7, 700],
+             [2, 201], [9, 900]]
+    b = P.Builder(lib)
+    bsrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    psrc = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)], join_type=1)
+    root = b.topn(j, [b.colref(1, GX_TYPE_I64)], [1 if desc else 0], 100)
+    ex = b.build(root)
+    bch = PyChunk([GX_TYPE_I64] * 2, 8)
+    for r in brows:
+        bch.append_row(r)
+    pch = PyChunk([GX_TYPE_I64] * 2, 16)
+    for r in prows:
+        pch.append_row(r)
+    ex.bind_chunks(bsrc, [bch])
+    ex.bind_chunks(psrc, [pch])
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_I64] * 4)
+    ex.close()
+    ex.free()
+    b.free()
+    keys = [r[1] for r in rows]
+    nn = [k for k in keys if k is not None]
+    if desc:
+        assert nn == sorted(nn, reverse=True)
+        assert keys[len(nn):] == [None] * (len(keys) - len(nn))  # NULLs last
+    else:
+        assert keys[:len(keys) - len(nn)] == [None] * (len(keys) - len(nn))
+        assert nn == sorted(nn)  # NULLs first, then ascending
+    return sorted(rows, key=lambda r: tuple((x is None, x) for x in r))
+
+
+def test_oracle_outer_sort_nullable_key():
+    for desc in (0, 1):
+        rows = _run_outer_sorted(load_oracle(), desc)
+        assert len(rows) == 8
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("desc", [0, 1])
+def test_outer_sort_nullable_key_parity(desc):
+    want = _run_outer_sorted(load_oracle(), desc)
+    got = _run_outer_sorted(load_product(), desc)
+    assert got == want

@@ -579,6 +579,9 @@ int gxSortPairs(uint64_t* keysIn, uint64_t* keysOut, uint32_t* idxIn,
 int gxSortKeyBits(const uint64_t* keys, int64_t n, uint64_t* devOrAnd,
                   void* stream);
 int gxSortIota(uint32_t* idx, int64_t n, void* stream);
+// null-bit key pass for nullable sort keys (NULL < any value)
+int gxSortComposeNullKeys(const uint8_t* bitmap, const uint32_t* idx,
+                          uint64_t* keys, int64_t n, int desc, void* stream);
 // gather one column into out (same layout; elemSize 1, 8 or 40)
 int gxSortGatherCol(const void* in, void* out, const uint32_t* idx, int64_t n,
                     int elemSize, void* stream);

@@ -4745,7 +4745,6 @@ static int32_t runDeviceSort(gx_exec* ex) {
   if (n > 0xFFFFFFFFLL) { ex->err = "sort > 2^32 rows unsupported"; return GX_ERR_INVALID; }
   for (auto& k : ex->devSortKeys) {
     gxp::DevCol& c = tab.cols[k.col];
-    if (c.hasNulls) { ex->err = "NULLs in sort key unsupported this round"; return GX_ERR_INVALID; }
     if (k.kind == 2 && !c.denseOffsets) {
       ex->err = "general varlen sort key unsupported this round";
       return GX_ERR_INVALID;
@@ -4789,16 +4788,36 @@ static int32_t runDeviceSort(gx_exec* ex) {
                               ex->stream));
     HIP_OK(ex, hipStreamSynchronize(ex->stream));
     uint64_t diff = oa[0] ^ oa[1];
-    if (diff == 0) continue;  // all keys equal: order already preserved
-    int beginBit = __builtin_ctzll(diff);
-    int endBit = 64 - __builtin_clzll(diff);
-    size_t tb = tmpBytes;
-    if (gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, tmp, &tb, beginBit,
-                         endBit, ex->stream)) {
-      ex->err = "radix sort failed";
-      return GX_ERR_INTERNAL;
+    // all keys equal -> the value pass is skippable, but a nullable column
+    // still needs its null-bit pass below
+    if (diff != 0) {
+      int beginBit = __builtin_ctzll(diff);
+      int endBit = 64 - __builtin_clzll(diff);
+      size_t tb = tmpBytes;
+      if (gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, tmp, &tb, beginBit,
+                           endBit, ex->stream)) {
+        ex->err = "radix sort failed";
+        return GX_ERR_INTERNAL;
+      }
+      std::swap(idxA, idxB);
     }
-    std::swap(idxA, idxB);
+    // nullable key: a 1-bit stable pass ABOVE the value bits (sortexec
+    // compare semantics: NULL < any value — first on ASC, last on DESC)
+    const gxp::DevCol& kc = tab.cols[ex->devSortKeys[j].col];
+    if (kc.hasNulls && kc.nullBitmap) {
+      if (gxp::gxSortComposeNullKeys(kc.nullBitmap, idxA, keyA, n,
+                                     ex->devSortKeys[j].desc, ex->stream)) {
+        ex->err = "null key compose failed";
+        return GX_ERR_INTERNAL;
+      }
+      size_t tb2 = tmpBytes;
+      if (gxp::gxSortPairs(keyA, keyB, idxA, idxB, n, tmp, &tb2, 0, 1,
+                           ex->stream)) {
+        ex->err = "null radix pass failed";
+        return GX_ERR_INTERNAL;
+      }
+      std::swap(idxA, idxB);
+    }
   }
   uint32_t herr = 0;
   HIP_OK(ex, hipMemcpyAsync(&herr, ex->devErr, 4, hipMemcpyDeviceToHost, ex->stream));

@@ -2899,6 +2899,28 @@ __global__ void sortGatherKernel(const uint8_t* __restrict__ in,
 
 
 
+// null-bit key pass (sortexec compare: NULL < any value — NULLs first on
+// ASC, last on DESC; the 1-bit stable pass runs AFTER the column's value
+// pass, i.e. more significant)
+__global__ void sortComposeNullKeysKernel(const uint8_t* __restrict__ bitmap,
+                                          const uint32_t* __restrict__ idx,
+                                          uint64_t* __restrict__ keys,
+                                          int64_t n, int desc) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t row = idx[i];
+    bool isNull = ((bitmap[row >> 3] >> (row & 7)) & 1) == 0;
+    keys[i] = desc ? (isNull ? 1 : 0) : (isNull ? 0 : 1);
+  }
+}
+
+int gxSortComposeNullKeys(const uint8_t* bitmap, const uint32_t* idx,
+                          uint64_t* keys, int64_t n, int desc, void* stream) {
+  hipLaunchKernelGGL(sortComposeNullKeysKernel, dim3(gridFor(n)), dim3(256), 0,
+                     (hipStream_t)stream, bitmap, idx, keys, n, desc);
+  return (int)hipGetLastError();
+}
+
 int gxSortIota(uint32_t* idx, int64_t n, void* stream) {
   hipLaunchKernelGGL(sortIotaKernel, dim3(gridFor(n)), dim3(256), 0,
                      (hipStream_t)stream, idx, n);

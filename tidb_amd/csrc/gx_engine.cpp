@@ -3917,15 +3917,18 @@ static int64_t hbmBudgetBytes() {
   if (const char* e = getenv("GX_HBM_BUDGET")) return atoll(e);
   size_t freeB = 0, totalB = 0;
   if (hipMemGetInfo(&freeB, &totalB) == hipSuccess)
-    return (int64_t)(freeB / 10 * 9);
+    return (int64_t)(freeB / 100 * 92);
   return INT64_MAX;
 }
 
-// rough resident bytes per row of a source schema (varlen estimated)
+// rough resident bytes per row of a source schema (varlen estimated at the
+// generator's ~1 B payload + the 8 B offsets entry; bound chunks of long
+// strings under-estimate, which only delays the spill until allocation
+// actually fails — never wrong results)
 static int64_t schemaRowBytes(const PNode& srcN) {
   int64_t b = 0;
   for (int t : srcN.colTypes)
-    b += t == GX_TYPE_DECIMAL ? 40 : (t == GX_TYPE_STRING ? 24 : 8);
+    b += t == GX_TYPE_DECIMAL ? 40 : (t == GX_TYPE_STRING ? 9 : 8);
   return b;
 }
 

@@ -170,3 +170,88 @@ def test_sum_length_parity():
     got, _ = _run_sum_length(load_product())
     want, _ = _run_sum_length(load_oracle())
     assert got == want
+
+
+def _run_fused_string_filter(lib, use_like):
+    """String predicates inside the FUSED scan->filter->agg CNF
+    (VectorizedFilter with builtinEQString / LIKE-'prefix%' conjuncts)."""
+    from tests.gxlib import (GX_AGG_MODE_COMPLETE, GX_F_EQ, GX_TPCH_LINEITEM,
+                             GX_TYPE_TIME)
+    b = P.Builder(lib)
+    src = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+    rf = b.colref(P.L_RETFLAG, GX_TYPE_STRING)
+    if use_like:
+        cond = b.call(GX_F_LIKE_PREFIX, GX_TYPE_I64, 0, rf, b.const_str("A"))
+    else:
+        cond = b.call(GX_F_EQ, GX_TYPE_I64, 0, rf, b.const_str("A"))
+    sel = b.selection(src, [cond])
+    qty = b.colref(P.L_QUANTITY, GX_TYPE_DECIMAL, 2)
+    ls = b.colref(P.L_LINESTATUS, GX_TYPE_STRING)
+    agg = b.hashagg(sel, [ls], [(GX_AGG_SUM, qty, 2), (GX_AGG_COUNT, -1, 0)])
+    ex = b.build(agg)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, 100_000)
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_STRING, GX_TYPE_DECIMAL, GX_TYPE_I64],
+                       [0, 2, 0], data_caps=[2048, None, None])
+    ex.close()
+    ex.free()
+    b.free()
+    return sorted(rows)
+
+
+@pytest.mark.parametrize("use_like", [False, True])
+def test_oracle_fused_string_filter(use_like):
+    rows = _run_fused_string_filter(load_oracle(), use_like)
+    assert len(rows) == 2  # linestatus O/F within returnflag 'A'
+    assert sum(r[2] for r in rows) > 20_000
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("use_like", [False, True])
+def test_fused_string_filter_parity(use_like):
+    want = _run_fused_string_filter(load_oracle(), use_like)
+    got = _run_fused_string_filter(load_product(), use_like)
+    assert got == want
+
+
+def _run_fused_like_chunks(lib):
+    """LIKE over longer (non-dense) strings feeding the fused agg, via
+    bound chunks — also exercises the wide-key path (string group key).
+    Matched strings carry no trailing padding: the device emits the TRIMMED
+    group value while the oracle emits the first row's original (both are
+    legitimate under PAD SPACE; see tests/test_wide_groupkeys.py)."""
+    from tests.gxlib import GX_AGG_COUNT
+    chunks, rows = _chunk(4000, seed=9)
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_STRING, GX_TYPE_I64])
+    cond = b.call(GX_F_LIKE_PREFIX, GX_TYPE_I64, 0,
+                  b.colref(0, GX_TYPE_STRING), b.const_str("hello"))
+    sel = b.selection(src, [cond])
+    agg = b.hashagg(sel, [b.colref(0, GX_TYPE_STRING)],
+                    [(GX_AGG_COUNT, -1, 0)])
+    ex = b.build(agg)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out = ex.pull_all([GX_TYPE_STRING, GX_TYPE_I64], [0, 0],
+                      data_caps=[1 << 16, None])
+    ex.close()
+    ex.free()
+    b.free()
+    want = {}
+    for s, _k in rows:
+        if s is not None and s.startswith("hello"):
+            want[s] = want.get(s, 0) + 1
+    return sorted(out), sorted(want.items())
+
+
+def test_oracle_fused_like_chunks():
+    got, want = _run_fused_like_chunks(load_oracle())
+    assert got == want and len(want) >= 1
+
+
+@pytest.mark.gpu
+def test_fused_like_chunks_parity():
+    got, want = _run_fused_like_chunks(load_product())
+    assert got == want
+    got_o, _ = _run_fused_like_chunks(load_oracle())
+    assert got == got_o

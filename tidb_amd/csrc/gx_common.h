@@ -82,6 +82,10 @@ struct PredDesc {
   int32_t cmp;        // GX_F_LT..GX_F_NE
   int32_t slot;       // raw fetch slot (TIME/I64 preds; -1 = load at use)
   uint64_t constU64;  // time value or i64/units bits
+  // string predicates (PRED_STR_EQ_CONST / PRED_STR_LIKE_PREFIX): the
+  // constant lives inline so every conjunct can carry its own
+  uint8_t strC[16] = {0};
+  int32_t strCLen = 0;
 };
 
 constexpr int kMaxPreds = 8;

@@ -1739,6 +1739,32 @@ static int32_t compileFused(gx_exec* ex) {
   if (sel) {
     for (int condId : sel->exprs) {
       const PExpr& e = plan.exprs[condId];
+      if (e.kind == EK_CALL && e.func == GX_F_LIKE_PREFIX &&
+          e.args.size() == 2) {
+        // LIKE 'abc%' in the fused CNF (byte prefix, no PAD trimming)
+        const PExpr& col = plan.exprs[e.args[0]];
+        const PExpr& pat = plan.exprs[e.args[1]];
+        if (col.kind != EK_COLREF || pat.kind != EK_CONST ||
+            col.colIdx < 0 || col.colIdx >= (int)src->colTypes.size() ||
+            src->colTypes[col.colIdx] != GX_TYPE_STRING ||
+            pat.constStr.size() > 16) {
+          ex->err = "device LIKE takes <string column> LIKE <'prefix%' <= 16B>";
+          return GX_ERR_INVALID;
+        }
+        if (ex->desc.nPreds >= gxp::kMaxPreds) {
+          ex->err = "too many filter conjuncts";
+          return GX_ERR_INVALID;
+        }
+        gxp::PredDesc pd{};
+        pd.kind = gxp::PRED_STR_LIKE_PREFIX;
+        pd.col = col.colIdx;
+        pd.cmp = GX_F_EQ;
+        pd.slot = -1;
+        std::memcpy(pd.strC, pat.constStr.data(), pat.constStr.size());
+        pd.strCLen = (int32_t)pat.constStr.size();
+        ex->desc.preds[ex->desc.nPreds++] = pd;
+        continue;
+      }
       if (e.kind != EK_CALL || e.func > GX_F_NE || e.args.size() != 2) {
         ex->err = "unsupported filter expression on device";
         return GX_ERR_INVALID;
@@ -1796,6 +1822,16 @@ static int32_t compileFused(gx_exec* ex) {
         pd.kind = gxp::PRED_DEC_CMP_CONST;
         pd.constU64 = (uint64_t)(int64_t)u;
         pd.slot = fetchSlot(ex, gxp::FETCH_DEC16, lhs->colIdx);
+      } else if (ct == GX_TYPE_STRING && rhs->retType == GX_TYPE_STRING &&
+                 (cmp == GX_F_EQ || cmp == GX_F_NE) &&
+                 rhs->constStr.size() <= 16) {
+        // string EQ/NE const in the fused CNF (PAD SPACE: trim the
+        // constant's trailing spaces like the column's, collate.go:272)
+        pd.kind = gxp::PRED_STR_EQ_CONST;
+        std::string k = rhs->constStr;
+        while (!k.empty() && k.back() == ' ') k.pop_back();
+        std::memcpy(pd.strC, k.data(), k.size());
+        pd.strCLen = (int32_t)k.size();
       } else {
         ex->err = "unsupported filter column/const type combination";
         return GX_ERR_INVALID;

@@ -88,13 +88,17 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
     } else if (pd.kind == gxp::PRED_I64_CMP_CONST) {
       s << "  if (!((int64_t)raw.get(" << pd.slot << ").x " << cmpOp(pd.cmp)
         << " (int64_t)" << (int64_t)pd.constU64 << "LL)) return true;\n";
-    } else {  // PRED_DEC_CMP_CONST
+    } else if (pd.kind == gxp::PRED_DEC_CMP_CONST) {
       s << "  { T u; int sc;\n"
            "    if (!loadDecimalUnits<WIDE>((const uint8_t*)d.table.cols["
         << pd.col << "].data + row * 40, &u, &sc, d.errorFlag)) return false;\n"
            "    if (!(VT<WIDE>::cmp(u, VT<WIDE>::fromI64((int64_t)"
         << (int64_t)pd.constU64 << "LL, nullptr)) " << cmpOp(pd.cmp)
         << " 0)) return true; }\n";
+    } else {  // string EQ/LIKE: shared interpreted helper (cold conjunct)
+      s << "  if (!evalSimplePred(d.table, d.preds[" << p
+        << "], d.preds[" << p << "].strC, d.preds[" << p
+        << "].strCLen, row)) return true;\n";
     }
   }
   s << "  (*mySel)++;\n";

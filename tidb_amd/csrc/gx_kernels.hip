@@ -154,7 +154,7 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
       int64_t k = (int64_t)pd.constU64;
       int cmp = v < k ? -1 : (v > k ? 1 : 0);
       pass = cmpResult(cmp, pd.cmp);
-    } else {  // PRED_DEC_CMP_CONST (engine aligned const to column scale)
+    } else if (pd.kind == PRED_DEC_CMP_CONST) {
       typename VT<WIDE>::T u;
       int sc;
       if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &u, &sc,
@@ -162,6 +162,8 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         return false;
       int cmp = VT<WIDE>::cmp(u, VT<WIDE>::fromI64((int64_t)pd.constU64, nullptr));
       pass = cmpResult(cmp, pd.cmp);
+    } else {  // string EQ/NE const + LIKE-'prefix%' (per-pred inline const)
+      pass = evalSimplePred(d.table, pd, pd.strC, pd.strCLen, row);
     }
   }
   if (!pass) return true;

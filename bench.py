@@ -663,6 +663,126 @@ def bench_strings(args):
     print(json.dumps(out), flush=True)
 
 
+def run_cpu_baseline_groupby(n=2_000_000):
+    """Oracle 3-col group-by over a bounded lineitem sample."""
+    import time as _t
+    from tests.gxlib import load_oracle
+    lib = load_oracle()
+    t0 = _t.perf_counter()
+    _run_groupby_plan(lib, n)
+    dt = _t.perf_counter() - t0
+    return {
+        "value": n / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"3-col group-by over {n} lineitem rows incl. generation "
+                  f"and pull, single thread ({dt:.1f}s)",
+    }
+
+
+def _run_groupby_plan(lib, n, device=0):
+    from tests.gxlib import (GX_AGG_COUNT, GX_AGG_SUM, GX_TPCH_LINEITEM,
+                             GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_STRING)
+    from tidb_amd import plan as P
+    b = P.Builder(lib)
+    src = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+    groups = [b.colref(P.L_RETFLAG, GX_TYPE_STRING),
+              b.colref(P.L_LINESTATUS, GX_TYPE_STRING),
+              b.colref(P.L_ORDERKEY, GX_TYPE_I64)]
+    agg = b.hashagg(src, groups,
+                    [(GX_AGG_SUM, b.colref(P.L_QUANTITY, GX_TYPE_DECIMAL, 2),
+                      2), (GX_AGG_COUNT, -1, 0)])
+    ex = b.build(agg, device=device)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, n)
+    ex.open()
+    out_types = [GX_TYPE_STRING, GX_TYPE_STRING, GX_TYPE_I64,
+                 GX_TYPE_DECIMAL, GX_TYPE_I64]
+    rows = 0
+    import ctypes as C
+    from tidb_amd.chunkpy import PyChunk
+    chunk = PyChunk(out_types, 1024, [0, 0, 0, 2, 0], [4096, 4096, None,
+                                                       None, None])
+    while True:
+        g = chunk.as_gx()
+        nn = C.c_int32(0)
+        rc = lib.gx_next(ex.ex, C.byref(g), C.byref(nn))
+        assert rc == 0, ex.error()
+        if nn.value == 0:
+            break
+        rows += nn.value
+    k = None
+    try:
+        lib.gx_last_kernel_ms.restype = C.c_double
+        lib.gx_last_kernel_ms.argtypes = [C.c_void_p]
+        k = lib.gx_last_kernel_ms(ex.ex)
+    except Exception:
+        pass
+    ex.close()
+    ex.free()
+    b.free()
+    return rows, k
+
+
+def bench_groupby(args):
+    """High-NDV serialized-key group-by (SURVEY §8a rows 9-10 generality):
+    group lineitem by (returnflag, linestatus, orderkey) — 3 key columns
+    force the wide-key path (hash + record-verified, interpreted kernel);
+    NDV = rows/4 x 6. One step = one full pass incl. group decode."""
+    from tests.gxlib import load_product
+    lib = load_product()
+    n = min(args.rows, 59_986_052)  # group decode is host-side O(NDV)
+    for _ in range(args.warmup):
+        _run_groupby_plan(lib, n)
+    t0 = time.perf_counter()
+    kms = []
+    ng = 0
+    for _ in range(args.steps):
+        ng, k = _run_groupby_plan(lib, n)
+        kms.append(k or 0)
+    elapsed = time.perf_counter() - t0
+    avg_kms = sum(kms) / len(kms)
+    value = n / (avg_kms / 1000.0) if avg_kms else 0
+    # algorithmic bytes/row: orderkey 8 + 2 chars 2 + qty 16 (DEC16 fetch) +
+    # key-record traffic ~ (probe reads are cached for repeated groups;
+    # insert writes ~ NDV x 88 B amortized)
+    bpr = 8 + 2 + 16 + 8
+    achieved = n * bpr / (avg_kms / 1000.0) / 1e9 if avg_kms else 0
+    out = {
+        "metric": "wide_groupby_rows_per_sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int128",
+        "data": "synthetic",
+        "config": {
+            "workload": f"lineitem_{n}_groupby_rf_ls_orderkey_wide_keys",
+            "rows": n,
+            "groups": ng,
+            "parallelism": "single-gpu",
+            "bytes_per_row": bpr,
+        },
+        "kernel_ms_avg": avg_kms,
+        "roofline": {
+            "bound": "hbm",
+            "achieved": achieved,
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": achieved / HBM_PEAK_GBS,
+            "traffic": None,
+        },
+        "cpu_baseline": run_cpu_baseline_groupby()
+        if not args.no_cpu_baseline else None,
+    }
+    print(json.dumps(out), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -671,7 +791,8 @@ def main():
     ap.add_argument("--rows", type=int, default=SF100_ROWS,
                     help="rows per GPU (default SF100 — the metric's config)")
     ap.add_argument("--query",
-                    choices=["q1", "q3", "sort", "wide", "join", "strings"],
+                    choices=["q1", "q3", "sort", "wide", "join", "strings",
+                             "groupby"],
                     default="q1")
     ap.add_argument("--sf", type=int, default=100,
                     help="scale factor for --query q3 (lineitem = 6M x SF)")
@@ -690,6 +811,8 @@ def main():
         return bench_join(args)
     if args.query == "strings":
         return bench_strings(args)
+    if args.query == "groupby":
+        return bench_groupby(args)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -2442,7 +2442,11 @@ static int32_t materializeDevice(gx_exec* ex) {
 static int32_t ensureWideKeyStore(gx_exec* ex) {
   gxp::GroupKeyDesc& g = ex->desc.gkey;
   g.recBytes = 16 + 24 * g.nCols;
-  g.recCap = ((int64_t)1 << ex->desc.globalGroupsLog2) + 4096;
+  // the record cursor doubles as the LOAD-FACTOR trigger: growth fires at
+  // ~70% occupancy, long before linear probing degenerates (a near-full
+  // open table costs a whole-table scan per insert)
+  g.recCap = std::max<int64_t>(
+      4096, ((int64_t)1 << ex->desc.globalGroupsLog2) * 7 / 10);
   g.keyStore = (uint8_t*)devAllocP(ex, (size_t)g.recCap * g.recBytes);
   if (!g.recCursor) g.recCursor = (uint64_t*)devAllocP(ex, 8);
   if (!g.keyStore || !g.recCursor) {
@@ -2780,7 +2784,7 @@ restart:
       freeSince(ex, mark);
       goto restart;
     }
-    if ((errFlag & 32u) && ex->desc.globalGroupsLog2 < 25) {
+    if ((errFlag & 32u) && ex->desc.globalGroupsLog2 < 27) {
       ex->desc.globalGroupsLog2 += 3;
       devFreeP(ex, ex->devTable);
       ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAllocP(
@@ -2925,8 +2929,8 @@ static int32_t runFused(gx_exec* ex) {
     if (getenv("GX_DEBUG")) fprintf(stderr, "[gx] LDS table full -> global-direct retry\n");
     return runFused(ex);
   }
-  if ((errFlag & 32u /*kErrGlobalFull*/) && ex->desc.globalGroupsLog2 < 25) {
-    // NDV above the global table: rerun with an 8x table (capped 2^25 groups
+  if ((errFlag & 32u /*kErrGlobalFull*/) && ex->desc.globalGroupsLog2 < 27) {
+    // NDV above the global table: rerun with an 8x table (capped 2^27 groups
     // ~ 10 GB of state; the init kernel resets it, so the rerun is clean)
     ex->desc.globalGroupsLog2 += 3;
     devFreeP(ex, ex->devTable);  // the outgrown table is dead weight

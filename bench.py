@@ -696,12 +696,12 @@ def _run_groupby_plan(lib, n, device=0):
     ex = b.build(agg, device=device)
     ex.bind_tpch(src, GX_TPCH_LINEITEM, n)
     ex.open()
-    out_types = [GX_TYPE_STRING, GX_TYPE_STRING, GX_TYPE_I64,
+    out_types = [GX_TYPE_STRING, GX_TYPE_STRING, GX_TYPE_DECIMAL,
                  GX_TYPE_DECIMAL, GX_TYPE_I64]
     rows = 0
     import ctypes as C
     from tidb_amd.chunkpy import PyChunk
-    chunk = PyChunk(out_types, 1024, [0, 0, 0, 2, 0], [4096, 4096, None,
+    chunk = PyChunk(out_types, 1024, [0, 0, 2, 2, 0], [4096, 4096, None,
                                                        None, None])
     while True:
         g = chunk.as_gx()

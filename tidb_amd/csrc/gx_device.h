@@ -546,13 +546,18 @@ __device__ inline bool wideKeyMatches(const FusedQueryDesc& d,
 
 // probe/insert: returns the global table slot owning this row's group, or
 // false with the error flag set (table or record store full -> the engine
-// grows 8x and reruns, the agg_spill.go partition-growth analog)
+// grows 8x and reruns, the agg_spill.go partition-growth analog).
+// The slow path's per-row ACQUIRE load is the L1-invalidating cost, so the
+// kernel carries a small LDS RESOLUTION CACHE (hash -> verified slot): a
+// cache hit re-verifies the record with plain loads (the bytes entered this
+// CU's L1 under the acquire that filled the entry) — exact, no fence.
+constexpr int kWkCache = 128;
+
 template <bool WIDE, typename VMT, typename RAWT>
-__device__ inline bool makeWideGroupKey(const FusedQueryDesc& d, int64_t row,
-                                        const RAWT& raw, const VMT& vm,
-                                        uint32_t* slotOut) {
-  uint64_t nullBits;
-  uint64_t h = wideKeyHash<WIDE>(d, row, raw, vm, &nullBits);
+__device__ inline bool wideKeyResolve(const FusedQueryDesc& d, int64_t row,
+                                      const RAWT& raw, const VMT& vm,
+                                      uint64_t h, uint64_t nullBits,
+                                      uint32_t* slotOut) {
   uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
   uint32_t slot = (uint32_t)h & gmask;
   uint32_t h32 = (uint32_t)(h >> 32);
@@ -599,6 +604,37 @@ __device__ inline bool makeWideGroupKey(const FusedQueryDesc& d, int64_t row,
     }
     slot = (slot + 1) & gmask;
   }
+}
+
+typedef __attribute__((address_space(3))) uint64_t WkLds64;
+typedef __attribute__((address_space(3))) uint32_t WkLds32;
+
+template <bool WIDE, typename VMT, typename RAWT>
+__device__ inline bool makeWideGroupKey(const FusedQueryDesc& d, int64_t row,
+                                        const RAWT& raw, const VMT& vm,
+                                        WkLds64* cacheH, WkLds32* cacheS,
+                                        uint32_t* slotOut) {
+  uint64_t nullBits;
+  uint64_t h = wideKeyHash<WIDE>(d, row, raw, vm, &nullBits);
+  int ci = (int)((h >> 32) & (kWkCache - 1));
+  if (cacheH[ci] == h) {
+    uint32_t slot = cacheS[ci];
+    uint64_t cur = d.globalTable[slot].key;  // plain: slot keys write once
+    if ((uint32_t)(cur >> 32) == (uint32_t)(h >> 32)) {
+      const uint8_t* rec =
+          d.gkey.keyStore + (cur & 0xFFFFFFFFu) * (uint64_t)d.gkey.recBytes;
+      if (wideKeyMatches<WIDE>(d, rec, row, raw, vm, nullBits)) {
+        *slotOut = slot;
+        return true;
+      }
+    }
+    // hash collision or torn entry: exact verify failed -> full resolve
+  }
+  if (!wideKeyResolve<WIDE>(d, row, raw, vm, h, nullBits, slotOut))
+    return false;
+  cacheH[ci] = h;  // racy plain writes: any torn pair fails the verify above
+  cacheS[ci] = *slotOut;
+  return true;
 }
 
 // simple single-table predicate (direct loads; build phases are cheap scans)

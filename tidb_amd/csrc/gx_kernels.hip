@@ -137,7 +137,9 @@ __global__ void genLineitemKernel(DevTable tab, int64_t rowBegin, int64_t nRows,
 template <bool WIDE, bool DIVOK, typename VMT, bool WK = false, typename RAWT>
 __device__ __attribute__((always_inline)) inline bool processRow(const FusedQueryDesc& d, int64_t row,
                                   const RAWT& raw, Lds3GroupSlot* lds,
-                                  uint64_t* mySel) {
+                                  uint64_t* mySel,
+                                  WkLds64* wkCacheH = nullptr,
+                                  WkLds32* wkCacheS = nullptr) {
   // ---- filter (CNF; NULL rejects — expression.go:507 toBool) ----
   bool pass = true;
   for (int p = 0; p < d.nPreds && pass; p++) {
@@ -376,7 +378,8 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   uint64_t key;
   uint32_t gslot = 0;
   if constexpr (WK) {
-    if (!makeWideGroupKey<WIDE>(d, row, raw, vm, &gslot)) return false;
+    if (!makeWideGroupKey<WIDE>(d, row, raw, vm, wkCacheH, wkCacheS, &gslot))
+      return false;
     key = gslot;
   } else {
     if (!makeGroupKey(d, row, raw, &key, d.errorFlag)) return false;
@@ -503,6 +506,13 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   bool failed = false;
   __shared__ GroupSlot lds[kLdsGroups];
   Lds3GroupSlot* lds3 = (Lds3GroupSlot*)lds;
+  // wide keys: the hash->slot resolution cache (exact; gx_device.h)
+  __shared__ uint64_t wkH[WK ? kWkCache : 1];
+  __shared__ uint32_t wkS[WK ? kWkCache : 1];
+  WkLds64* wkCacheH = (WkLds64*)wkH;
+  WkLds32* wkCacheS = (WkLds32*)wkS;
+  if (WK)
+    for (int i = threadIdx.x; i < kWkCache; i += blockDim.x) wkH[i] = ~0ULL;
   for (int i = threadIdx.x; i < kLdsGroups; i += blockDim.x) {
     lds[i].key = kEmptyKey;
     for (int a = 0; a < kMaxAggs; a++) {
@@ -539,10 +549,12 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
       if (rB < end) fetchRow(d.table, d.fetch, d.nFetch, rB, rawB);
       using VMT = typename std::conditional<NVM <= 12, VmState<WIDE>,
                                             VmState14<WIDE>>::type;
-      if (!processRow<WIDE, DIVOK, VMT, WK>(d, row, rawA, lds3, &mySel)) { failed = true; break; }
+      if (!processRow<WIDE, DIVOK, VMT, WK>(d, row, rawA, lds3, &mySel,
+                                            wkCacheH, wkCacheS)) { failed = true; break; }
       const int64_t rA2 = row + 2 * stride;
       if (rA2 < end) fetchRow(d.table, d.fetch, d.nFetch, rA2, rawA);
-      if (rB < end && !processRow<WIDE, DIVOK, VMT, WK>(d, rB, rawB, lds3, &mySel)) failed = true;
+      if (rB < end && !processRow<WIDE, DIVOK, VMT, WK>(d, rB, rawB, lds3, &mySel,
+                                                        wkCacheH, wkCacheS)) failed = true;
     }
   }
 

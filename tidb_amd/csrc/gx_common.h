@@ -252,6 +252,12 @@ struct FusedQueryDesc {
   // global table (engine retries with this set when a workgroup's LDS table
   // overflows — NDV above kLdsGroups)
   int32_t noLds = 0;
+  // noLds accumulator BANKS: low-NDV tables replicate the accumulator
+  // arrays accBanks x so concurrent blocks hammer different cache lines
+  // (bank = blockIdx & (B-1)); keys/inserts live in bank 0 only and the
+  // host decode merges banks (all supported states are mergeable: int128 /
+  // f64 sums add, biased min/max takes the extreme, counts add)
+  int32_t accBanks = 1;
   // glds (LDS-DMA) staged variant: streams are DMA'd tile-by-tile into LDS
   // (double-buffered, counted vmcnt waits) so compute overlaps the memory
   // stream. Engine enables it when every fetch kind is stageable and no
@@ -559,7 +565,7 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream, int skipInit = 0);
 int gxLaunchMemset(void* p, int v, size_t n, void* stream);
 int gxFusedGrid(int64_t rows);
-int gxLaunchInitTable(GroupSlot* table, int nSlots, void* stream);
+int gxLaunchInitTable(GroupSlot* table, int64_t nSlots, void* stream);
 int gxDumpDesc(const FusedQueryDesc* devDesc, void* stream);
 
 // ---- device full sort (sortexec/sort.go analog): LSD stable radix passes

@@ -2502,9 +2502,12 @@ static int32_t finalizeFusedBind(gx_exec* ex) {
       }
     }
   }
-  // result/error buffers
+  // result/error buffers; low-NDV tables replicate accumulator banks to
+  // spread noLds atomic contention (high-NDV tables have natural spread)
+  ex->desc.accBanks = ex->desc.globalGroupsLog2 <= 16 ? 16 : 1;
   ex->devTable = (gxp::GroupSlot*)devAllocP(
-      ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
+      ex, (sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2) *
+              ex->desc.accBanks);
   ex->devErr = (uint32_t*)devAllocP(ex, 4);
   ex->devSel = (uint64_t*)devAllocP(ex, 8);
   ex->devDesc = (gxp::FusedQueryDesc*)devAllocP(ex, sizeof(gxp::FusedQueryDesc));
@@ -2745,7 +2748,9 @@ restart:
       if (ex->jitProg) {
         lrc = done == 0 ? gxp::gxLaunchInitTable(
                               ex->desc.globalTable,
-                              1 << ex->desc.globalGroupsLog2, ex->stream)
+                              (int64_t)(1 << ex->desc.globalGroupsLog2) *
+                                  ex->desc.accBanks,
+                              ex->stream)
                         : 0;
         if (lrc == 0 && done == 0 && ex->desc.gkey.wideMode &&
             ex->desc.gkey.recCursor)
@@ -2787,8 +2792,10 @@ restart:
     if ((errFlag & 32u) && ex->desc.globalGroupsLog2 < 27) {
       ex->desc.globalGroupsLog2 += 3;
       devFreeP(ex, ex->devTable);
+      ex->desc.accBanks = ex->desc.globalGroupsLog2 <= 16 ? 16 : 1;
       ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAllocP(
-          ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
+          ex, (sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2) *
+                  ex->desc.accBanks);
       if (!ex->devTable) {
         ex->err = "hipMalloc failed (group table)";
         rc = GX_ERR_INTERNAL;
@@ -2891,8 +2898,10 @@ static int32_t runFused(gx_exec* ex) {
   }
   int lrc;
   if (ex->jitProg) {
-    lrc = gxp::gxLaunchInitTable(ex->desc.globalTable,
-                                 1 << ex->desc.globalGroupsLog2, ex->stream);
+    lrc = gxp::gxLaunchInitTable(
+        ex->desc.globalTable,
+        (int64_t)(1 << ex->desc.globalGroupsLog2) * ex->desc.accBanks,
+        ex->stream);
     if (lrc == 0)
       lrc = gxjit::launch(ex->jitProg, ex->desc.wide != 0, ex->devDesc,
                           gxp::gxFusedGrid(ex->desc.table.nRows), ex->stream);
@@ -2934,8 +2943,10 @@ static int32_t runFused(gx_exec* ex) {
     // ~ 10 GB of state; the init kernel resets it, so the rerun is clean)
     ex->desc.globalGroupsLog2 += 3;
     devFreeP(ex, ex->devTable);  // the outgrown table is dead weight
+    ex->desc.accBanks = ex->desc.globalGroupsLog2 <= 16 ? 16 : 1;
     ex->desc.globalTable = ex->devTable = (gxp::GroupSlot*)devAllocP(
-        ex, sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2);
+        ex, (sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2) *
+                ex->desc.accBanks);
     if (!ex->devTable) { ex->err = "hipMalloc failed (group table)"; return GX_ERR_INTERNAL; }
     if (ex->desc.gkey.wideMode) {  // key-record store grows with the table
       devFreeP(ex, ex->desc.gkey.keyStore);
@@ -2976,10 +2987,39 @@ static int32_t runFused(gx_exec* ex) {
 // download the global table + wide-key records and decode the result
 // rows (shared by the one-shot and the out-of-core streaming paths)
 static int32_t fusedDecodeResults(gx_exec* ex) {
-  std::vector<gxp::GroupSlot> table((size_t)1 << ex->desc.globalGroupsLog2);
+  const int64_t nSlots1 = (int64_t)1 << ex->desc.globalGroupsLog2;
+  std::vector<gxp::GroupSlot> table((size_t)nSlots1 * ex->desc.accBanks);
   HIP_OK(ex, hipMemcpy(table.data(), ex->devTable,
-                       sizeof(gxp::GroupSlot) << ex->desc.globalGroupsLog2,
+                       table.size() * sizeof(gxp::GroupSlot),
                        hipMemcpyDeviceToHost));
+  // fold the accumulator banks into bank 0 (sums/counts add, biased
+  // min/max takes the extreme, f64 bits add as doubles)
+  for (int bank = 1; bank < ex->desc.accBanks; bank++) {
+    for (int64_t i = 0; i < nSlots1; i++) {
+      gxp::GroupSlot& dst = table[i];
+      const gxp::GroupSlot& b = table[bank * nSlots1 + i];
+      if (dst.key == gxp::kEmptyKey) continue;
+      for (int s = 0; s < ex->desc.nAccSlots; s++) {
+        if (ex->desc.accKind[s] == 3) {
+          double x, y;
+          std::memcpy(&x, &dst.accLo[s], 8);
+          std::memcpy(&y, &b.accLo[s], 8);
+          x += y;
+          std::memcpy(&dst.accLo[s], &x, 8);
+        } else if (ex->desc.accKind[s] != 0) {
+          if (b.accLo[s] > dst.accLo[s]) dst.accLo[s] = b.accLo[s];
+        } else {
+          __int128 v = (((__int128)dst.accHi[s]) << 64) | dst.accLo[s];
+          v += (((__int128)b.accHi[s]) << 64) | b.accLo[s];
+          dst.accLo[s] = (uint64_t)v;
+          dst.accHi[s] = (int64_t)(v >> 64);
+        }
+      }
+      int nCnt = ex->desc.sharedCnt ? 1 : ex->desc.nAggs;
+      for (int a = 0; a < nCnt; a++) dst.cnt[a] += b.cnt[a];
+    }
+  }
+  table.resize((size_t)nSlots1);
   // collect occupied slots, deterministic order (by key)
   std::vector<const gxp::GroupSlot*> occ;
   for (auto& s : table)

@@ -460,7 +460,9 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
       slot = (slot + 1) & gmask;
     }
   }
-  GroupSlot* target = &d.globalTable[slot];
+  GroupSlot* target =
+      &d.globalTable[slot + ((uint32_t)blockIdx.x & (d.accBanks - 1)) *
+                                (int64_t)(1u << d.globalGroupsLog2)];
   if (d.sharedCnt) accumInto(target, 0, Int128{0, 0}, 1);  // bumps cnt[0] only
   for (int s = 0; s < d.nAccSlots; s++) {
     int reg = d.accReg[s];
@@ -612,8 +614,8 @@ __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   }
 }
 
-__global__ void initGlobalTableKernel(GroupSlot* table, int n) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
+__global__ void initGlobalTableKernel(GroupSlot* table, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   table[i].key = kEmptyKey;
   for (int a = 0; a < kMaxAggs; a++) {
@@ -1665,9 +1667,10 @@ int gxFusedGrid(int64_t rows) {
   return grid;
 }
 
-int gxLaunchInitTable(GroupSlot* table, int nSlots, void* stream) {
-  hipLaunchKernelGGL(initGlobalTableKernel, dim3((nSlots + 255) / 256),
-                     dim3(256), 0, (hipStream_t)stream, table, nSlots);
+int gxLaunchInitTable(GroupSlot* table, int64_t nSlots, void* stream) {
+  hipLaunchKernelGGL(initGlobalTableKernel,
+                     dim3((int)((nSlots + 255) / 256)), dim3(256), 0,
+                     (hipStream_t)stream, table, nSlots);
   return (int)hipGetLastError();
 }
 
@@ -1675,9 +1678,9 @@ int gxLaunchFusedAgg(const FusedQueryDesc& desc, const FusedQueryDesc* devDesc,
                      void* stream, int skipInit) {
   hipStream_t s = (hipStream_t)stream;
   int grid = gxFusedGrid(desc.table.nRows);
-  int nSlots = 1 << desc.globalGroupsLog2;
+  int64_t nSlots = (int64_t)(1 << desc.globalGroupsLog2) * desc.accBanks;
   if (!skipInit) {  // out-of-core slices accumulate into the SAME table
-    hipLaunchKernelGGL(initGlobalTableKernel, dim3((nSlots + 255) / 256),
+    hipLaunchKernelGGL(initGlobalTableKernel, dim3((int)((nSlots + 255) / 256)),
                        dim3(256), 0, s, desc.globalTable, nSlots);
     if (desc.gkey.wideMode && desc.gkey.recCursor)
       hipMemsetAsync(desc.gkey.recCursor, 0, 8, s);

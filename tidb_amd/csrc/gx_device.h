@@ -552,16 +552,28 @@ __device__ inline bool wideKeyMatches(const FusedQueryDesc& d,
 // cache hit re-verifies the record with plain loads (the bytes entered this
 // CU's L1 under the acquire that filled the entry) — exact, no fence.
 constexpr int kWkCache = 128;
+// slot claimed, record still being written. A published pack's low 32 bits
+// are a record index < recCap (<= 2^27), so the all-ones low word can never
+// be a real pack and probers can always tell the states apart.
+constexpr uint64_t kWkLocked = ~1ULL;
 
 template <bool WIDE, typename VMT, typename RAWT>
 __device__ inline bool wideKeyResolve(const FusedQueryDesc& d, int64_t row,
                                       const RAWT& raw, const VMT& vm,
                                       uint64_t h, uint64_t nullBits,
                                       uint32_t* slotOut) {
+  // CLAIM-FIRST insert: CAS the slot to kWkLocked and only the winner
+  // consumes a record index — so recCursor counts exactly the distinct keys
+  // and a startup burst of racers on the same new key cannot leak records
+  // and spuriously trip kErrGlobalFull (reserve-then-publish leaked one
+  // record per CAS loser: ~1e5 on a 60M-row low-NDV table, blowing recCap
+  // and costing full-pass grow retries). Losers re-read the slot; the loop
+  // never spins inside a divergent branch (each re-read is a fresh
+  // iteration, so a same-wave winner's publish completes first) and the
+  // probe cap bounds it absolutely.
   uint32_t gmask = (1u << d.globalGroupsLog2) - 1;
   uint32_t slot = (uint32_t)h & gmask;
   uint32_t h32 = (uint32_t)(h >> 32);
-  uint32_t myRec = 0xFFFFFFFFu;
   for (uint32_t probe = 0;; probe++) {
     if (probe > gmask) {
       atomicOr(d.errorFlag, kErrGlobalFull);
@@ -571,28 +583,39 @@ __device__ inline bool wideKeyResolve(const FusedQueryDesc& d, int64_t row,
         (unsigned long long*)&d.globalTable[slot].key, __ATOMIC_ACQUIRE,
         __HIP_MEMORY_SCOPE_AGENT);
     if (cur == kEmptyKey) {
-      if (myRec == 0xFFFFFFFFu) {
-        uint64_t idx = atomicAdd((unsigned long long*)d.gkey.recCursor, 1ULL);
-        if ((int64_t)idx >= d.gkey.recCap) {
-          atomicOr(d.errorFlag, kErrGlobalFull);
-          return false;
-        }
-        myRec = (uint32_t)idx;
-        wideKeyWrite<WIDE>(d, d.gkey.keyStore + idx * (uint64_t)d.gkey.recBytes,
-                           row, raw, vm, nullBits);
-        __threadfence();  // record visible before the key publishes
-      }
-      uint64_t want = ((uint64_t)h32 << 32) | myRec;
       unsigned long long expect = (unsigned long long)kEmptyKey;
       __hip_atomic_compare_exchange_strong(
           (unsigned long long*)&d.globalTable[slot].key, &expect,
-          (unsigned long long)want, __ATOMIC_ACQ_REL, __ATOMIC_ACQUIRE,
+          (unsigned long long)kWkLocked, __ATOMIC_ACQ_REL, __ATOMIC_ACQUIRE,
           __HIP_MEMORY_SCOPE_AGENT);
-      if (expect == (unsigned long long)kEmptyKey) {  // we won
+      if (expect == (unsigned long long)kEmptyKey) {  // claimed the slot
+        uint64_t idx = atomicAdd((unsigned long long*)d.gkey.recCursor, 1ULL);
+        if ((int64_t)idx >= d.gkey.recCap) {
+          // record store full: flag it so probers parked on kWkLocked bail
+          atomicOr(d.errorFlag, kErrGlobalFull);
+          return false;
+        }
+        wideKeyWrite<WIDE>(d, d.gkey.keyStore + idx * (uint64_t)d.gkey.recBytes,
+                           row, raw, vm, nullBits);
+        __threadfence();  // record visible before the key publishes
+        __hip_atomic_store(
+            (unsigned long long*)&d.globalTable[slot].key,
+            (unsigned long long)(((uint64_t)h32 << 32) | (uint32_t)idx),
+            __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
         *slotOut = slot;
         return true;
       }
-      cur = (uint64_t)expect;  // lost: verify against the winner
+      cur = (uint64_t)expect;  // lost: the winner's state decides
+    }
+    if (cur == kWkLocked) {
+      // the claimant is materializing its record: re-read this slot (it may
+      // turn out to be OUR key). If the claimant overflowed the record
+      // store it set the flag instead of publishing — bail with it.
+      if (__hip_atomic_load(d.errorFlag, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) &
+          kErrGlobalFull)
+        return false;
+      continue;
     }
     if ((uint32_t)(cur >> 32) == h32) {
       const uint8_t* rec =

@@ -378,8 +378,16 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   uint64_t key;
   uint32_t gslot = 0;
   if constexpr (WK) {
-    if (!makeWideGroupKey<WIDE>(d, row, raw, vm, wkCacheH, wkCacheS, &gslot))
+    if (d.ablate == 4) {  // timing ablation: fake slots, skip resolution
+      gslot = (uint32_t)(splitmix64((uint64_t)row) & 255u);
+    } else if (!makeWideGroupKey<WIDE>(d, row, raw, vm, wkCacheH, wkCacheS,
+                                       &gslot)) {
       return false;
+    }
+    if (d.ablate == 3) {  // timing ablation: resolve keys, skip accumulate
+      asm volatile("" ::"v"(gslot));
+      return true;
+    }
     key = gslot;
   } else {
     if (!makeGroupKey(d, row, raw, &key, d.errorFlag)) return false;

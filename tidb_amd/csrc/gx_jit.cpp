@@ -413,6 +413,10 @@ std::string generateSource(const FusedQueryDesc& d) {
 template <bool WIDE>
 __device__ __forceinline__ void kernBody(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
+  // any error flag dooms the pass: late blocks bail before fetching
+  if (__hip_atomic_load(d.errorFlag, __ATOMIC_RELAXED,
+                        __HIP_MEMORY_SCOPE_AGENT))
+    return;
   bool failed = false;
   __shared__ GroupSlot lds[kLdsGroups];
   Lds3GroupSlot* lds3 = (Lds3GroupSlot*)lds;

@@ -513,6 +513,13 @@ template <bool WIDE, int R, bool DIVOK = false, int NVM = 12, bool WK = false>
 __launch_bounds__(256)
 __global__ void fusedAggKernel(const FusedQueryDesc* __restrict__ dp) {
   const FusedQueryDesc& d = *dp;
+  // any error flag dooms the whole pass (the engine discards it and either
+  // retries or fails), so late blocks bail before fetching anything: a
+  // doomed attempt (LdsFull probe, GlobalFull grow, packed->wide) costs
+  // flag-propagation time instead of a full 60M-row pass
+  if (__hip_atomic_load(d.errorFlag, __ATOMIC_RELAXED,
+                        __HIP_MEMORY_SCOPE_AGENT))
+    return;
   bool failed = false;
   __shared__ GroupSlot lds[kLdsGroups];
   Lds3GroupSlot* lds3 = (Lds3GroupSlot*)lds;

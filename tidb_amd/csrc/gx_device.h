@@ -551,7 +551,7 @@ __device__ inline bool wideKeyMatches(const FusedQueryDesc& d,
 // kernel carries a small LDS RESOLUTION CACHE (hash -> verified slot): a
 // cache hit re-verifies the record with plain loads (the bytes entered this
 // CU's L1 under the acquire that filled the entry) — exact, no fence.
-constexpr int kWkCache = 128;
+constexpr int kWkCache = 512;
 // slot claimed, record still being written. A published pack's low 32 bits
 // are a record index < recCap (<= 2^27), so the all-ones low word can never
 // be a real pack and probers can always tell the states apart.

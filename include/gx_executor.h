@@ -93,8 +93,11 @@ enum {
    * LIKE_PREFIX (builtinLikeSig's 'abc%' fast path): args (str, const
    *          prefix WITHOUT the trailing %); 1 iff str starts with prefix
    *          (binary collation, case-sensitive); NULL str -> NULL.
-   * UPPER   (builtinUpperSig): ASCII a-z upcased, other bytes unchanged. */
-  GX_F_LENGTH = 32, GX_F_SUBSTR = 33, GX_F_LIKE_PREFIX = 34, GX_F_UPPER = 35
+   * UPPER   (builtinUpperSig): ASCII a-z upcased, other bytes unchanged.
+   * LOWER   (builtinLowerSig): ASCII A-Z downcased, other bytes unchanged
+   *          (binary/ASCII charset scope, like UPPER). */
+  GX_F_LENGTH = 32, GX_F_SUBSTR = 33, GX_F_LIKE_PREFIX = 34, GX_F_UPPER = 35,
+  GX_F_LOWER = 36
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

@@ -241,7 +241,8 @@ static int32_t evalString(EvalCtx& ctx, const Expr& e, const Chunk& in,
     }
     return GX_OK;
   }
-  if (e.func == GX_F_UPPER) {  // builtinUpperSig (ASCII)
+  if (e.func == GX_F_UPPER || e.func == GX_F_LOWER) {
+    // builtinUpperSig / builtinLowerSig (ASCII byte semantics)
     out.type = GX_TYPE_STRING;
     out.offsets.assign(1, 0);
     for (int i = 0; i < n; i++) {
@@ -249,8 +250,12 @@ static int32_t evalString(EvalCtx& ctx, const Expr& e, const Chunk& in,
       int len;
       const uint8_t* p = a.getBytes(i, &len);
       std::string s((const char*)p, len);
-      for (char& c : s)
-        if (c >= 'a' && c <= 'z') c = (char)(c - 'a' + 'A');
+      for (char& c : s) {
+        if (e.func == GX_F_UPPER && c >= 'a' && c <= 'z')
+          c = (char)(c - 'a' + 'A');
+        if (e.func == GX_F_LOWER && c >= 'A' && c <= 'Z')
+          c = (char)(c - 'A' + 'a');
+      }
       out.appendBytes(s.data(), s.size());
     }
     return GX_OK;
@@ -304,7 +309,7 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       if (e.func <= GX_F_NE) return evalCompare(ctx, e, in, out);
       if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT)
         return evalCast(ctx, e, in, out);
-      if (e.func >= GX_F_LENGTH && e.func <= GX_F_UPPER)
+      if (e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER)
         return evalString(ctx, e, in, out);
       return evalArith(ctx, e, in, out);
   }

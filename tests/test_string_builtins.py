@@ -16,9 +16,9 @@ import numpy as np
 import pytest
 
 from tests.gxlib import (GX_AGG_COUNT, GX_AGG_SUM, GX_F_LENGTH,
-                         GX_F_LIKE_PREFIX, GX_F_SUBSTR, GX_F_UPPER,
-                         GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_STRING,
-                         load_oracle, load_product)
+                         GX_F_LIKE_PREFIX, GX_F_LOWER, GX_F_SUBSTR,
+                         GX_F_UPPER, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                         GX_TYPE_STRING, load_oracle, load_product)
 from tidb_amd import plan as P
 from tidb_amd.chunkpy import PyChunk
 
@@ -96,6 +96,61 @@ def test_oracle_string_projection():
 def test_string_projection_parity(pos, ln):
     want, _ = _run_proj(load_oracle(), pos, ln)
     got, _ = _run_proj(load_product(), pos, ln)
+    assert got == want
+
+
+def _run_lower(lib):
+    """LOWER (builtinLowerSig, ASCII): alone, over SUBSTR, and composed with
+    UPPER both ways — the outermost case op decides for ASCII bytes."""
+    ch, rows = _chunk(n=2000, seed=11)
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_STRING, GX_TYPE_I64])
+    s = b.colref(0, GX_TYPE_STRING)
+    sub = b.call(GX_F_SUBSTR, GX_TYPE_STRING, 0, s, b.const_i64(2),
+                 b.const_i64(7))
+    exprs = [
+        b.call(GX_F_LOWER, GX_TYPE_STRING, 0, s),
+        b.call(GX_F_LOWER, GX_TYPE_STRING, 0, sub),
+        b.call(GX_F_LOWER, GX_TYPE_STRING, 0,
+               b.call(GX_F_UPPER, GX_TYPE_STRING, 0, s)),
+        b.call(GX_F_UPPER, GX_TYPE_STRING, 0,
+               b.call(GX_F_LOWER, GX_TYPE_STRING, 0, s)),
+    ]
+    root = b.projection(src, exprs)
+    ex = b.build(root)
+    ex.bind_chunks(src, ch)
+    ex.open()
+    out = ex.pull_all([GX_TYPE_STRING] * 4, [0] * 4,
+                      data_caps=[1 << 18] * 4)
+    ex.close()
+    ex.free()
+    b.free()
+    return out, rows
+
+
+def _ascii_lower(s):
+    return None if s is None else "".join(
+        chr(ord(c) + 32) if "A" <= c <= "Z" else c for c in s)
+
+
+def _ascii_upper(s):
+    return None if s is None else "".join(
+        chr(ord(c) - 32) if "a" <= c <= "z" else c for c in s)
+
+
+def test_oracle_lower():
+    got, rows = _run_lower(load_oracle())
+    for (o0, o1, o2, o3), (s, _) in zip(got, rows):
+        assert o0 == _ascii_lower(s)
+        assert o1 == _ascii_lower(_py_substr(s, 2, 7))
+        assert o2 == _ascii_lower(s)   # lower(upper(x)) == lower(x) for ASCII
+        assert o3 == _ascii_upper(s)
+
+
+@pytest.mark.gpu
+def test_lower_parity():
+    want, _ = _run_lower(load_oracle())
+    got, _ = _run_lower(load_product())
     assert got == want
 
 

@@ -496,6 +496,7 @@ struct StrProg {
   int64_t winPos[kMaxStrWin];  // MySQL 1-based; negative counts from the end
   int64_t winLen[kMaxStrWin];
   int32_t upper = 0;
+  int32_t lower = 0;  // outermost case op wins (compileStrProg clears the other)
   // per-run device temps (engine-allocated)
   int64_t* starts = nullptr;
   int64_t* lens = nullptr;

@@ -1472,9 +1472,11 @@ static bool compileStrProg(gx_exec* ex, int exprId,
     return true;
   }
   if (e.kind != EK_CALL) return false;
-  if (e.func == GX_F_UPPER && e.args.size() == 1) {
+  if ((e.func == GX_F_UPPER || e.func == GX_F_LOWER) && e.args.size() == 1) {
     if (!compileStrProg(ex, e.args[0], types, sp)) return false;
-    sp->upper = 1;
+    // ASCII case ops: the OUTERMOST call decides (lower(upper(x)) == lower(x))
+    sp->upper = e.func == GX_F_UPPER;
+    sp->lower = e.func == GX_F_LOWER;
     return true;
   }
   if (e.func == GX_F_SUBSTR && e.args.size() == 3) {
@@ -1556,7 +1558,7 @@ static int32_t compileProject(gx_exec* ex) {
       // string outputs: SUBSTR/UPPER chains fold into one windowed view
       gxp::StrProg sp{};
       if (!compileStrProg(ex, eid, *childTypes, &sp)) {
-        ex->err = "unsupported string projection (SUBSTR/UPPER chains over "
+        ex->err = "unsupported string projection (SUBSTR/UPPER/LOWER chains over "
                   "a string column this round)";
         return GX_ERR_INVALID;
       }

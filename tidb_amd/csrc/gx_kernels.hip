@@ -3220,6 +3220,7 @@ __global__ void strEmitKernel(const ProjDesc* __restrict__ dp, int pi,
     for (int64_t j = 0; j < len; j++) {
       uint8_t b = p[s + j];
       if (sp.upper && b >= 'a' && b <= 'z') b = (uint8_t)(b - 'a' + 'A');
+      if (sp.lower && b >= 'A' && b <= 'Z') b = (uint8_t)(b - 'A' + 'a');
       outData[o + j] = b;
     }
   }

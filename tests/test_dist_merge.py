@@ -9,8 +9,11 @@ import os
 
 import pytest
 
-from tests.gxlib import GX_AGG_MODE_PARTIAL, GX_TPCH_LINEITEM, load_oracle
+from tests.gxlib import (GX_AGG_MODE_FINAL, GX_AGG_MODE_PARTIAL,
+                         GX_TPCH_LINEITEM, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                         load_oracle)
 from tidb_amd import plan as P
+from tidb_amd.chunkpy import PyChunk
 
 
 def merge_partials(lib, partial_rows):
@@ -173,6 +176,40 @@ def test_final_merge_minmax_firstrow_both_libs():
     got_o = run(load_oracle())
     got_p = run(load_product())
     assert got_o == got_p == expected
+
+
+def test_final_empty_scalar_default_row():
+    """FINAL with NO group-by over ZERO partial rows must emit the scalar
+    default row (count=0, sum/min NULL) — HashAggExec's empty-input
+    semantics, same as SELECT count(*), sum(x) FROM empty_t. Both libraries'
+    host FINAL paths, CPU."""
+    from tests.gxlib import (GX_AGG_COUNT, GX_AGG_MIN, GX_AGG_SUM,
+                             load_product)
+    part_types = [GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_I64, GX_TYPE_DECIMAL]
+    part_fracs = [2, 0, 0, 2]
+
+    def run(lib):
+        b = P.Builder(lib)
+        src = b.source(part_types, part_fracs)
+        agg = b.hashagg(src, [],
+                        [(GX_AGG_SUM, b.colref(0, GX_TYPE_DECIMAL, 2), 2),
+                         (GX_AGG_COUNT, -1, 0),
+                         (GX_AGG_MIN, b.colref(3, GX_TYPE_DECIMAL, 2), 2)],
+                        GX_AGG_MODE_FINAL)
+        chunk = PyChunk(part_types, 1, part_fracs)  # bound but EMPTY
+        ex = b.build(agg)
+        ex.bind_chunks(src, [chunk])
+        ex.open()
+        rows = ex.pull_all([GX_TYPE_DECIMAL, GX_TYPE_I64, GX_TYPE_DECIMAL],
+                           [2, 0, 2])
+        ex.close()
+        ex.free()
+        b.free()
+        return rows
+
+    want = [(None, 0, None)]
+    assert run(load_oracle()) == want
+    assert run(load_product()) == want
 
 
 def test_shards_final_merge_full_family():

@@ -52,6 +52,47 @@ def test_q3_empty_probe_oracle(oracle_lib):
     assert q3_no_match(oracle_lib) == []
 
 
+def _scalar_agg_rows(lib, nrows):
+    """SELECT count(c0), sum(c1) with NO group-by: zero input rows must
+    still produce ONE row (count=0, sum NULL) — HashAggExec empty-input
+    semantics (executor/aggregate tests; aggregate.result goldens)."""
+    import ctypes
+
+    from tests.gxlib import (GX_AGG_COUNT, GX_AGG_SUM, GX_TYPE_DECIMAL)
+    from tidb_amd.chunkpy import PyChunk
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2])
+    agg = b.hashagg(src, [], [(GX_AGG_COUNT, b.colref(0, GX_TYPE_I64), 0),
+                              (GX_AGG_SUM, b.colref(1, GX_TYPE_DECIMAL, 2),
+                               2)])
+    ex = b.build(agg)
+    ch = PyChunk([GX_TYPE_I64, GX_TYPE_DECIMAL], max(nrows, 1), [0, 2])
+    out = (ctypes.c_uint8 * 40)()
+    assert lib.gx_dec_from_string(b"1.50", 4, out) == 0
+    for i in range(nrows):
+        ch.append_row([i, bytes(out)])
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2])
+    ex.close()
+    ex.free()
+    b.free()
+    return rows
+
+
+def test_scalar_agg_empty_oracle(oracle_lib):
+    assert _scalar_agg_rows(oracle_lib, 0) == [(0, None)]
+    assert _scalar_agg_rows(oracle_lib, 3) == [(3, "4.50")]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("n", [0, 3])
+def test_scalar_agg_empty_parity(n):
+    from tests.gxlib import load_product
+    assert _scalar_agg_rows(load_product(), n) == \
+        _scalar_agg_rows(load_oracle(), n)
+
+
 @pytest.mark.gpu
 @pytest.mark.parametrize("n", [0, 1, 63, 64, 65, 255, 1023])
 def test_q1_tiny_parity(n):

@@ -4769,6 +4769,18 @@ static int32_t runFinalHost(gx_exec* ex) {
     }
   }
   ex->resultRows.clear();
+  // zero partial rows, no group-by => the scalar-agg default row
+  // (count=0, sums/extremes NULL — HashAggExec's empty-input semantics)
+  if (nGroup == 0 && order.empty()) {
+    FState st;
+    st.dec.resize(agg.aggFuncs.size());
+    st.cnt.assign(agg.aggFuncs.size(), 0);
+    st.val.resize(agg.aggFuncs.size());
+    st.has.assign(agg.aggFuncs.size(), 0);
+    st.f64.assign(agg.aggFuncs.size(), 0.0);
+    groups.emplace(std::string(), std::move(st));
+    order.push_back(std::string());
+  }
   for (const std::string& key : order) {
     FState& st = groups[key];
     std::vector<OutRowVal> row = st.groupVals;

@@ -97,7 +97,11 @@ enum {
    * LOWER   (builtinLowerSig): ASCII A-Z downcased, other bytes unchanged
    *          (binary/ASCII charset scope, like UPPER). */
   GX_F_LENGTH = 32, GX_F_SUBSTR = 33, GX_F_LIKE_PREFIX = 34, GX_F_UPPER = 35,
-  GX_F_LOWER = 36
+  GX_F_LOWER = 36,
+  /* IS [NOT] NULL (builtinIntIsNullSig / builtinDecimalIsNullSig /
+   * builtinStringIsNullSig family, builtin_op_vec.go): unary, any column
+   * type; result i64 0/1 and NEVER NULL (the null bit is the value). */
+  GX_F_IS_NULL = 37, GX_F_IS_NOT_NULL = 38
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

@@ -311,6 +311,19 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
         return evalCast(ctx, e, in, out);
       if (e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER)
         return evalString(ctx, e, in, out);
+      if (e.func == GX_F_IS_NULL || e.func == GX_F_IS_NOT_NULL) {
+        // builtin*IsNullSig (builtin_op_vec.go): 0/1, never NULL
+        Column a;
+        int32_t err = evalVec(ctx, e.args[0], in, a);
+        if (err) return err;
+        out.reset();
+        out.type = GX_TYPE_I64;
+        for (int i = 0; i < in.numRows(); i++) {
+          bool isn = a.isNull(i);
+          out.appendI64((e.func == GX_F_IS_NULL) == isn ? 1 : 0);
+        }
+        return GX_OK;
+      }
       return evalArith(ctx, e, in, out);
   }
   return GX_ERR_INVALID;

@@ -1,0 +1,134 @@
+"""IS NULL / IS NOT NULL predicates (builtinIntIsNullSig /
+builtinDecimalIsNullSig / builtinStringIsNullSig family,
+/root/reference/pkg/expression/builtin_op_vec.go): the result is the null
+bit itself — i64 0/1, NEVER NULL — so in a CNF filter a NULL column value
+PASSES `IS NULL` (unlike every comparison predicate, where NULL rejects).
+
+Covered device paths: standalone Selection (both polarities, i64/decimal/
+string columns) and the fused aggregation CNF (count rows per group where a
+nullable column IS NOT NULL, mixed with an ordinary comparison conjunct).
+
+Parity: product (GPU) vs oracle on identical chunks + independent Python
+expectations on the oracle.
+"""
+import ctypes
+
+import numpy as np
+import pytest
+
+from tests.gxlib import (GX_AGG_COUNT, GX_F_GT, GX_F_IS_NOT_NULL,
+                         GX_F_IS_NULL, GX_TYPE_DECIMAL, GX_TYPE_I64,
+                         GX_TYPE_STRING, load_oracle, load_product)
+from tidb_amd import plan as P
+from tidb_amd.chunkpy import PyChunk
+
+TYPES = [GX_TYPE_I64, GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_STRING]
+FRACS = [0, 0, 2, 0]
+
+
+def _dec(lib, s):
+    out = (ctypes.c_uint8 * 40)()
+    assert lib.gx_dec_from_string(s.encode(), len(s.encode()), out) == 0
+    return bytes(out)
+
+
+def _data(lib, n=3000, seed=9):
+    rng = np.random.default_rng(seed)
+    rows = []
+    for i in range(n):
+        k = int(rng.integers(0, 50))
+        v = None if rng.random() < 0.25 else int(rng.integers(-100, 100))
+        d = None if rng.random() < 0.25 else f"{k}.25"
+        s = None if rng.random() < 0.25 else f"s{k % 7}"
+        rows.append([k, v, d, s])
+    chunks = []
+    for base in range(0, n, 1000):
+        m = min(1000, n - base)
+        ch = PyChunk(TYPES, m, FRACS, [None, None, None, m * 8])
+        for r in rows[base:base + m]:
+            ch.append_row([r[0], r[1],
+                           None if r[2] is None else _dec(lib, r[2]), r[3]])
+        chunks.append(ch)
+    return rows, chunks
+
+
+def _run_select(lib, col, ctype, cfrac, func):
+    rows, chunks = _data(lib)
+    b = P.Builder(lib)
+    src = b.source(TYPES, FRACS)
+    cond = b.call(func, GX_TYPE_I64, 0, b.colref(col, ctype, cfrac))
+    root = b.selection(src, [cond])
+    ex = b.build(root)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out = ex.pull_all(TYPES, FRACS, data_caps=[None, None, None, 1 << 16])
+    ex.close()
+    ex.free()
+    b.free()
+    return rows, out
+
+
+CASES = [(1, GX_TYPE_I64, 0), (2, GX_TYPE_DECIMAL, 2), (3, GX_TYPE_STRING, 0)]
+
+
+def test_oracle_isnull_selection():
+    lib = load_oracle()
+    for col, t, f in CASES:
+        for func in (GX_F_IS_NULL, GX_F_IS_NOT_NULL):
+            rows, got = _run_select(lib, col, t, f, func)
+            want = [tuple(r) for r in rows
+                    if (r[col] is None) == (func == GX_F_IS_NULL)]
+            norm = [tuple(r) for r in got]
+            assert norm == want, (col, func)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("col,ctype,cfrac", CASES)
+@pytest.mark.parametrize("func", [GX_F_IS_NULL, GX_F_IS_NOT_NULL])
+def test_isnull_selection_parity(col, ctype, cfrac, func):
+    _, want = _run_select(load_oracle(), col, ctype, cfrac, func)
+    _, got = _run_select(load_product(), col, ctype, cfrac, func)
+    assert got == want
+    assert len(got) > 100  # both polarities select a real subset
+
+
+def _run_fused(lib, func):
+    """count(*) group by k where v IS [NOT] NULL AND k > 5 — IS NULL inside
+    the fused scan→filter→agg CNF next to an ordinary comparison."""
+    rows, chunks = _data(lib)
+    b = P.Builder(lib)
+    src = b.source(TYPES, FRACS)
+    conds = [b.call(func, GX_TYPE_I64, 0, b.colref(1, GX_TYPE_I64)),
+             b.call(GX_F_GT, GX_TYPE_I64, 0, b.colref(0, GX_TYPE_I64),
+                    b.const_i64(5))]
+    sel = b.selection(src, conds)
+    agg = b.hashagg(sel, [b.colref(0, GX_TYPE_I64)],
+                    [(GX_AGG_COUNT, -1, 0)])
+    ex = b.build(agg)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out = ex.pull_all([GX_TYPE_I64, GX_TYPE_I64], [0, 0])
+    ex.close()
+    ex.free()
+    b.free()
+    return rows, sorted(out)
+
+
+def test_oracle_isnull_fused():
+    lib = load_oracle()
+    for func in (GX_F_IS_NULL, GX_F_IS_NOT_NULL):
+        rows, got = _run_fused(lib, func)
+        want = {}
+        for r in rows:
+            if (r[1] is None) == (func == GX_F_IS_NULL) and r[0] > 5:
+                want[r[0]] = want.get(r[0], 0) + 1
+        assert got == sorted(want.items())
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("func", [GX_F_IS_NULL, GX_F_IS_NOT_NULL])
+def test_isnull_fused_parity(func):
+    _, want = _run_fused(load_oracle(), func)
+    _, got = _run_fused(load_product(), func)
+    assert got == want
+    assert len(got) > 10

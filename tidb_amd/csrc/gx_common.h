@@ -350,6 +350,9 @@ struct JoinAggDesc {
 
 enum { PRED_STR_EQ_CONST = 3 };  // extra PredKind for the join path
 enum { PRED_STR_LIKE_PREFIX = 6 };  // LIKE 'abc%' fast path (builtinLikeSig)
+enum { PRED_IS_NULL = 7 };  // IS [NOT] NULL (builtin*IsNullSig: result is
+                            // the null bit itself, never NULL; cmp EQ = IS
+                            // NULL, NE = IS NOT NULL; any column type)
 
 // one conjunct of a post-join filter (NULL operand rejects the row, the
 // VecEvalBool NULL semantics, expression.go:420-504)

@@ -665,6 +665,8 @@ __device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
                                       const uint8_t* strConst, int strConstLen,
                                       int64_t row) {
   const DevCol& c = tab.cols[pd.col];
+  if (pd.kind == PRED_IS_NULL)  // the null bit IS the result (never NULL)
+    return colIsNull(c, row) == (pd.cmp == 4 /*GX_F_EQ*/);
   if (colIsNull(c, row)) return false;
   if (pd.kind == PRED_TIME_CMP_CONST) {
     uint64_t v = gptr<uint64_t>(c.data)[row] & ~0xFULL;

@@ -841,6 +841,24 @@ static bool compileTablePred(gx_exec* ex, const PNode& srcNode, int condId,
     *out = pd;
     return true;
   }
+  if (e.kind == EK_CALL &&
+      (e.func == GX_F_IS_NULL || e.func == GX_F_IS_NOT_NULL) &&
+      e.args.size() == 1) {
+    // builtin*IsNullSig: the null bit is the value (never NULL itself)
+    const PExpr& a0 = ex->plan.exprs[e.args[0]];
+    if (a0.kind != EK_COLREF || a0.colIdx < 0 ||
+        a0.colIdx >= (int)srcNode.colTypes.size()) {
+      ex->err = "IS NULL takes a column";
+      return false;
+    }
+    gxp::PredDesc pd{};
+    pd.kind = gxp::PRED_IS_NULL;
+    pd.col = a0.colIdx;
+    pd.cmp = e.func == GX_F_IS_NULL ? GX_F_EQ : GX_F_NE;
+    pd.slot = -1;
+    *out = pd;
+    return true;
+  }
   if (e.kind != EK_CALL || e.func > GX_F_NE || e.args.size() != 2) {
     ex->err = "unsupported filter expression on device";
     return false;
@@ -1764,6 +1782,27 @@ static int32_t compileFused(gx_exec* ex) {
         pd.slot = -1;
         std::memcpy(pd.strC, pat.constStr.data(), pat.constStr.size());
         pd.strCLen = (int32_t)pat.constStr.size();
+        ex->desc.preds[ex->desc.nPreds++] = pd;
+        continue;
+      }
+      if (e.kind == EK_CALL &&
+          (e.func == GX_F_IS_NULL || e.func == GX_F_IS_NOT_NULL) &&
+          e.args.size() == 1) {
+        const PExpr& a0 = plan.exprs[e.args[0]];
+        if (a0.kind != EK_COLREF || a0.colIdx < 0 ||
+            a0.colIdx >= (int)src->colTypes.size()) {
+          ex->err = "IS NULL takes a column";
+          return GX_ERR_INVALID;
+        }
+        if (ex->desc.nPreds >= gxp::kMaxPreds) {
+          ex->err = "too many filter conjuncts";
+          return GX_ERR_INVALID;
+        }
+        gxp::PredDesc pd{};
+        pd.kind = gxp::PRED_IS_NULL;
+        pd.col = a0.colIdx;
+        pd.cmp = e.func == GX_F_IS_NULL ? GX_F_EQ : GX_F_NE;
+        pd.slot = -1;
         ex->desc.preds[ex->desc.nPreds++] = pd;
         continue;
       }

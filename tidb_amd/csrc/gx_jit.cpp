@@ -80,6 +80,12 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
   for (int p = 0; p < d.nPreds; p++) {
     const gxp::PredDesc& pd = d.preds[p];
     const gxp::DevCol& c = d.table.cols[pd.col];
+    if (pd.kind == gxp::PRED_IS_NULL) {  // no null-reject: null IS the result
+      s << "  if (!evalSimplePred(d.table, d.preds[" << p
+        << "], d.preds[" << p << "].strC, d.preds[" << p
+        << "].strCLen, row)) return true;\n";
+      continue;
+    }
     if (c.hasNulls)
       s << "  if (colIsNull(d.table.cols[" << pd.col << "], row)) return true;\n";
     if (pd.kind == gxp::PRED_TIME_CMP_CONST) {

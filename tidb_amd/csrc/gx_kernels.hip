@@ -145,6 +145,10 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
   for (int p = 0; p < d.nPreds && pass; p++) {
     const PredDesc& pd = d.preds[p];
     const DevCol& c = d.table.cols[pd.col];
+    if (pd.kind == PRED_IS_NULL) {  // null bit is the result; no null-reject
+      pass = colIsNull(c, row) == (pd.cmp == 4 /*GX_F_EQ*/);
+      continue;
+    }
     if (colIsNull(c, row)) { pass = false; break; }
     if (pd.kind == PRED_TIME_CMP_CONST) {
       uint64_t v = raw.get(pd.slot).x & ~0xFULL;

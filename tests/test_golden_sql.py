@@ -100,3 +100,95 @@ def test_minmax_string_null_group():
                       [GX_TYPE_I64, GX_TYPE_STRING, GX_TYPE_STRING],
                       [0, 0, 0], [GX_TYPE_I64, GX_TYPE_STRING], [0, 0]))
     assert got == [(1, "11", "11"), (3, None, None)]
+
+
+def test_count_empty_table_grouped_vs_scalar():
+    """aggregate.result:55-64 — over an EMPTY t(a,b,c):
+    `select count(a) from t group by a` returns ZERO rows, while
+    `select count(a) from t` (no group-by) returns the single default
+    row `0`. Then after `insert t values(0,0,0)`:
+    `select count(b) from t group by a` returns one row `1`."""
+    from tests.gxlib import GX_AGG_COUNT
+    lib = load_oracle()
+
+    def grouped(b, src):
+        a = b.colref(0, GX_TYPE_I64)
+        return b.hashagg(src, [a], [(GX_AGG_COUNT,
+                                     b.colref(1, GX_TYPE_I64), 0)])
+
+    def scalar(b, src):
+        return b.hashagg(src, [], [(GX_AGG_COUNT,
+                                    b.colref(0, GX_TYPE_I64), 0)])
+
+    t3 = [GX_TYPE_I64] * 3
+    assert _run(lib, [], grouped, [GX_TYPE_I64, GX_TYPE_I64], [0, 0],
+                t3, [0, 0, 0]) == []
+    assert _run(lib, [], scalar, [GX_TYPE_I64], [0], t3, [0, 0, 0]) == [(0,)]
+    assert _run(lib, [(0, 0, 0)], grouped, [GX_TYPE_I64, GX_TYPE_I64],
+                [0, 0], t3, [0, 0, 0]) == [(0, 1)]
+
+
+def test_count_filtered_two_key_group_order_limit():
+    """aggregate.result:66-85 — t(a,b,c) with rows (0,0,0),(1,1,1),(3,3,6),
+    (3,2,5),(2,1,4),(1,1,3),(1,1,2):
+    `select count(a) from t where b>0 group by a, b` gives counts
+    {1,1,1,3} (in any order), and with `order by a limit 1` the first
+    group (a=1,b=1) has count 3."""
+    from tests.gxlib import GX_AGG_COUNT, GX_F_GT
+    lib = load_oracle()
+    rows = [(0, 0, 0), (1, 1, 1), (3, 3, 6), (3, 2, 5), (2, 1, 4),
+            (1, 1, 3), (1, 1, 2)]
+    t3 = [GX_TYPE_I64] * 3
+
+    def plan(b, src):
+        a = b.colref(0, GX_TYPE_I64)
+        bb = b.colref(1, GX_TYPE_I64)
+        sel = b.selection(src, [b.call(GX_F_GT, GX_TYPE_I64, 0, bb,
+                                       b.const_i64(0))])
+        return b.hashagg(sel, [a, bb], [(GX_AGG_COUNT, a, 0)])
+
+    got = _run(lib, rows, plan, [GX_TYPE_I64] * 3, [0] * 3, t3, [0] * 3)
+    assert sorted(c for _, _, c in got) == [1, 1, 1, 3]
+
+    def plan_limit(b, src):
+        a = b.colref(0, GX_TYPE_I64)
+        bb = b.colref(1, GX_TYPE_I64)
+        sel = b.selection(src, [b.call(GX_F_GT, GX_TYPE_I64, 0, bb,
+                                       b.const_i64(0))])
+        agg = b.hashagg(sel, [a, bb], [(GX_AGG_COUNT, a, 0)])
+        return b.topn(agg, [b.colref(0, GX_TYPE_I64)], [0], 1)
+
+    got = _run(lib, rows, plan_limit, [GX_TYPE_I64] * 3, [0] * 3, t3, [0] * 3)
+    assert [c for _, _, c in got] == [3]  # group (a=1,b=1)
+
+
+def test_max_both_sides_over_join():
+    """aggregate.result:86-91 — t(1,1,1),(2,1,1); tt(1,2,1);
+    `select max(a.b), max(b.b) from t a join tt b on a.a = b.a group by
+    a.c` returns one row `1 2` (only a=1 joins; group a.c=1)."""
+    from tests.gxlib import GX_AGG_MAX
+    lib = load_oracle()
+    t3 = [GX_TYPE_I64] * 3
+    b = P.Builder(lib)
+    tsrc = b.source(t3, [0] * 3)
+    ttsrc = b.source(t3, [0] * 3)
+    j = b.hashjoin(tsrc, ttsrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)])
+    # joined schema: t.a,t.b,t.c,tt.a,tt.b,tt.c
+    agg = b.hashagg(j, [b.colref(2, GX_TYPE_I64)],
+                    [(GX_AGG_MAX, b.colref(1, GX_TYPE_I64), 0),
+                     (GX_AGG_MAX, b.colref(4, GX_TYPE_I64), 0)])
+    ex = b.build(agg)
+    ch_t = PyChunk(t3, 2)
+    for r in [(1, 1, 1), (2, 1, 1)]:
+        ch_t.append_row(list(r))
+    ch_tt = PyChunk(t3, 1)
+    ch_tt.append_row([1, 2, 1])
+    ex.bind_chunks(tsrc, [ch_t])
+    ex.bind_chunks(ttsrc, [ch_tt])
+    ex.open()
+    got = ex.pull_all([GX_TYPE_I64] * 3, [0] * 3)
+    ex.close()
+    ex.free()
+    b.free()
+    assert got == [(1, 1, 2)]

@@ -111,7 +111,16 @@ enum {
   GX_AGG_AVG = 2,      /* avgOriginal4Decimal: state {sum, count} (func_avg.go:69-135) */
   GX_AGG_MIN = 3,
   GX_AGG_MAX = 4,
-  GX_AGG_FIRSTROW = 5
+  GX_AGG_FIRSTROW = 5,
+  /* DISTINCT variants (aggFuncDesc.HasDistinct, aggregation.go;
+   * executor/aggfuncs distinct wrappers): aggregate over the DISTINCT
+   * non-NULL values of the single arg column. COMPLETE mode only (a
+   * distinct partial state is a set and is not exchanged this round).
+   * SUM/AVG DISTINCT take decimal (or f64) args — int args must arrive
+   * wrapped in CAST_DEC like their non-distinct forms. */
+  GX_AGG_COUNT_DISTINCT = 6,
+  GX_AGG_SUM_DISTINCT = 7,
+  GX_AGG_AVG_DISTINCT = 8
 };
 
 /* ---- hash-agg execution mode (AggFuncDesc partial/final split,

@@ -10,6 +10,7 @@
 
 #include <algorithm>
 #include <cassert>
+#include <set>
 #include <cstring>
 #include <functional>
 
@@ -483,6 +484,7 @@ class ProjectionExec : public Exec {
 //     MIN/MAX         -> value col (null = empty)
 //     FIRSTROW        -> value col
 struct AggState {
+  std::set<std::string> seen;  // DISTINCT: HashGroupKey-encoded values
   MyDecimal dec;       // sum/avg accumulator or min/max/firstrow decimal
   double f64 = 0;
   int64_t i64 = 0;     // count / notNullRowCount
@@ -523,7 +525,9 @@ class HashAggExec : public Exec {
       for (size_t a = 0; a < node.aggFuncs.size(); a++) {
         int f = node.aggFuncs[a];
         switch (f) {
-          case GX_AGG_COUNT: outTypes.push_back(GX_TYPE_I64); outFracs.push_back(0); break;
+          case GX_AGG_COUNT:
+          case GX_AGG_COUNT_DISTINCT:
+            outTypes.push_back(GX_TYPE_I64); outFracs.push_back(0); break;
           default:
             outTypes.push_back(argType(a));
             outFracs.push_back(node.aggFracs[a]);
@@ -533,6 +537,12 @@ class HashAggExec : public Exec {
   }
 
   int32_t open() override {
+    for (int f : node_.aggFuncs)
+      if (f >= GX_AGG_COUNT_DISTINCT &&
+          node_.aggMode != GX_AGG_MODE_COMPLETE) {
+        err = "DISTINCT aggregates support COMPLETE mode only";
+        return GX_ERR_INVALID;
+      }
     done_ = false;
     emitPos_ = 0;
     order_.clear();
@@ -624,9 +634,33 @@ class HashAggExec : public Exec {
   }
 
   // UpdatePartialResult per agg func (func_sum.go:224, func_avg.go:110, ...)
-  int32_t updateState(AggState& s, int func, const Column* argCol, int row, int valueType) {
+  int32_t updateState(AggState& s, int func, const Column* argCol, int row,
+                      int valueType, const std::string* vkey = nullptr) {
     bool isNull = argCol ? argCol->isNull(row) : false;
     switch (func) {
+      case GX_AGG_COUNT_DISTINCT:
+      case GX_AGG_SUM_DISTINCT:
+      case GX_AGG_AVG_DISTINCT: {
+        // distinct wrappers: NULLs excluded; each VALUE updates once per
+        // group (value identity = the codec HashGroupKey encoding, the
+        // same identity the reference's distinct checker uses)
+        if (isNull || !vkey) break;
+        if (!s.seen.insert(*vkey).second) break;
+        if (func == GX_AGG_COUNT_DISTINCT) { s.i64++; break; }
+        if (valueType == GX_TYPE_DECIMAL) {
+          MyDecimal tmp;
+          int32_t ec = DecimalAdd(&s.dec, argCol->getDecimal(row), &tmp);
+          if (ec != E_OK && ec != E_TRUNCATED) return ec;
+          s.dec = tmp;
+          s.i64++;
+        } else if (valueType == GX_TYPE_F64) {
+          s.f64 += argCol->getF64(row);
+          s.i64++;
+        } else {
+          return GX_ERR_INVALID;  // int sum/avg arrives CAST_DEC-wrapped
+        }
+        break;
+      }
       case GX_AGG_COUNT:
         if (!isNull) s.i64++;
         break;
@@ -740,11 +774,17 @@ class HashAggExec : public Exec {
       // eval agg arg exprs once per chunk
       std::vector<Column> argCols(node_.aggFuncs.size());
       std::vector<const Column*> argPtr(node_.aggFuncs.size(), nullptr);
+      std::vector<std::vector<std::string>> valKeys(node_.aggFuncs.size());
       for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
         if (node_.aggArgs[a] >= 0) {
           ec = evalVec(ctx, node_.aggArgs[a], in, argCols[a]);
           if (ec) return ec;
           argPtr[a] = &argCols[a];
+          if (node_.aggFuncs[a] >= GX_AGG_COUNT_DISTINCT) {
+            valKeys[a].assign(n, std::string());
+            ec = HashGroupKeyCol(argCols[a], valKeys[a]);
+            if (ec) return ec;
+          }
         }
       }
       for (int i = 0; i < n; i++) {
@@ -759,7 +799,9 @@ class HashAggExec : public Exec {
         if (node_.kind == PK_STREAMAGG) lastKey_ = keys[i];
         Group& g = getGroup(keys[i], groupCols, i);
         for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
-          ec = updateState(g.states[a], node_.aggFuncs[a], argPtr[a], i, argType(a));
+          ec = updateState(g.states[a], node_.aggFuncs[a], argPtr[a], i,
+                           argType(a),
+                           valKeys[a].empty() ? nullptr : &valKeys[a][i]);
           if (ec) return ec;
         }
       }
@@ -864,8 +906,10 @@ class HashAggExec : public Exec {
       int vt = argType(a);
       switch (f) {
         case GX_AGG_COUNT:
+        case GX_AGG_COUNT_DISTINCT:
           out.cols[col++].appendI64(s.i64);
           break;
+        case GX_AGG_SUM_DISTINCT:  // never partial (rejected at open)
         case GX_AGG_SUM:
           if (partial) {
             if (s.i64 == 0) out.cols[col++].appendNull();
@@ -885,6 +929,7 @@ class HashAggExec : public Exec {
             }
           }
           break;
+        case GX_AGG_AVG_DISTINCT:  // never partial (rejected at open)
         case GX_AGG_AVG:
           if (partial) {
             if (vt == GX_TYPE_DECIMAL) out.cols[col++].appendDecimal(s.dec);

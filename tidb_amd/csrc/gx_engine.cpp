@@ -140,6 +140,12 @@ struct gx_exec {
   bool isBareSource = false;
   bool isJoinAgg = false;
   int aggRoot = -1;  // the HASHAGG node the fused kernel implements
+  // DISTINCT rewrite (aggFuncDesc.HasDistinct): the device kernel groups by
+  // (orig keys..., arg) — the dedup — and the decode folds back to the orig
+  // keys. distinctNKeys >= 0 marks the rewrite; funcs/fracs are the USER's.
+  std::vector<int> distinctFuncs;
+  std::vector<int> distinctFracs;
+  int distinctNKeys = -1;
   int sourceNode = -1;
   gxp::FusedQueryDesc desc;
   std::vector<std::pair<int, int>> projRegs;  // projection idx -> (reg, scale)
@@ -1723,6 +1729,42 @@ static int32_t compileAggOverJoin(gx_exec* ex, int aggNode) {
 
 // compile the fused Source->[Selection]->[Projection]->HashAgg pipeline
 static int32_t compileFused(gx_exec* ex) {
+  // DISTINCT rewrite: aggregate over DISTINCT values == group by
+  // (keys..., arg) on the device (the dedup falls out of the group table),
+  // then fold on the host at decode. All aggs must be DISTINCT over ONE
+  // shared arg this round; the inner device agg becomes count(*) (unused).
+  {
+    PNode& aggN = ex->plan.nodes[ex->root];
+    bool anyD = false;
+    for (int f : aggN.aggFuncs) anyD |= f >= GX_AGG_COUNT_DISTINCT;
+    if (anyD && ex->distinctNKeys < 0) {
+      if (aggN.aggMode != GX_AGG_MODE_COMPLETE) {
+        ex->err = "DISTINCT aggregates support COMPLETE mode only";
+        return GX_ERR_INVALID;
+      }
+      int argE = aggN.aggArgs.empty() ? -1 : aggN.aggArgs[0];
+      for (size_t a = 0; a < aggN.aggFuncs.size(); a++) {
+        if (aggN.aggFuncs[a] < GX_AGG_COUNT_DISTINCT ||
+            aggN.aggArgs[a] != argE || argE < 0) {
+          ex->err = "DISTINCT aggregates must all be DISTINCT over one "
+                    "shared arg column this round";
+          return GX_ERR_INVALID;
+        }
+        if (aggN.aggFuncs[a] != GX_AGG_COUNT_DISTINCT &&
+            ex->plan.exprs[argE].retType != GX_TYPE_DECIMAL) {
+          ex->err = "SUM/AVG DISTINCT takes a decimal arg (cast ints)";
+          return GX_ERR_INVALID;
+        }
+      }
+      ex->distinctFuncs = aggN.aggFuncs;
+      ex->distinctFracs = aggN.aggFracs;
+      ex->distinctNKeys = (int)aggN.exprs.size();
+      aggN.exprs.push_back(argE);
+      aggN.aggFuncs.assign(1, GX_AGG_COUNT);
+      aggN.aggArgs.assign(1, -1);
+      aggN.aggFracs.assign(1, 0);
+    }
+  }
   const PPlan& plan = ex->plan;
   ex->aggRoot = ex->root;
   const PNode* agg = &plan.nodes[ex->root];
@@ -3286,6 +3328,109 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
       }
     }
     ex->resultRows.push_back(std::move(row));
+  }
+  if (ex->distinctNKeys >= 0) {
+    // fold the deduplicated (keys..., value) rows back to the user's
+    // schema: per orig-key group, count/sum/avg the non-NULL values (every
+    // row is already a UNIQUE (keys, value) combination — the device group
+    // table was the distinct check)
+    const int nk = ex->distinctNKeys;
+    auto cmpVal = [](const OutRowVal& a, const OutRowVal& b) -> int {
+      if (a.isNull != b.isNull) return a.isNull ? -1 : 1;
+      if (a.isNull) return 0;
+      switch (a.type) {
+        case GX_TYPE_DECIMAL: return a.dec.Compare(b.dec);
+        case GX_TYPE_STRING:
+          return a.str < b.str ? -1 : (a.str > b.str ? 1 : 0);
+        case GX_TYPE_TIME:
+          return a.u64 < b.u64 ? -1 : (a.u64 > b.u64 ? 1 : 0);
+        case GX_TYPE_F64:
+          return a.f64 < b.f64 ? -1 : (a.f64 > b.f64 ? 1 : 0);
+        default:
+          return a.i64 < b.i64 ? -1 : (a.i64 > b.i64 ? 1 : 0);
+      }
+    };
+    struct DState {
+      std::vector<OutRowVal> keys;
+      int64_t cnt = 0;
+      MyDecimal sum;
+      bool hasSum = false;
+    };
+    auto keyLess = [&](const std::vector<OutRowVal>* a,
+                       const std::vector<OutRowVal>* b) {
+      for (int k = 0; k < nk; k++) {
+        int c = cmpVal((*a)[k], (*b)[k]);
+        if (c) return c < 0;
+      }
+      return false;
+    };
+    std::map<const std::vector<OutRowVal>*, size_t, decltype(keyLess)> idx(
+        keyLess);
+    std::vector<DState> states;
+    for (auto& r : ex->resultRows) {
+      auto it = idx.find(&r);
+      size_t si;
+      if (it == idx.end()) {
+        DState st;
+        st.keys.assign(r.begin(), r.begin() + nk);
+        st.sum.FromInt(0);
+        states.push_back(std::move(st));
+        si = states.size() - 1;
+        idx.emplace(&r, si);
+      } else {
+        si = it->second;
+      }
+      const OutRowVal& v = r[nk];
+      if (v.isNull) continue;  // NULL values never count as distinct
+      DState& st = states[si];
+      st.cnt++;
+      if (v.type == GX_TYPE_DECIMAL) {
+        MyDecimal tmp;
+        int32_t ec = gxp::DecimalAdd(&st.sum, &v.dec, &tmp);
+        if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
+          ex->err = "distinct sum overflow";
+          return GX_ERR_INTERNAL;
+        }
+        st.sum = tmp;
+        st.hasSum = true;
+      }
+    }
+    if (nk == 0 && states.empty()) states.emplace_back();  // scalar default
+    // NOTE: idx keys point into the OLD resultRows — drop before rebuild
+    idx.clear();
+    std::vector<std::vector<OutRowVal>> folded;
+    folded.reserve(states.size());
+    for (DState& st : states) {
+      std::vector<OutRowVal> row = st.keys;
+      for (size_t a = 0; a < ex->distinctFuncs.size(); a++) {
+        OutRowVal v;
+        if (ex->distinctFuncs[a] == GX_AGG_COUNT_DISTINCT) {
+          v.type = GX_TYPE_I64;
+          v.i64 = st.cnt;
+        } else if (st.cnt == 0) {
+          v.type = GX_TYPE_DECIMAL;
+          v.isNull = true;
+        } else if (ex->distinctFuncs[a] == GX_AGG_SUM_DISTINCT) {
+          v.type = GX_TYPE_DECIMAL;
+          v.dec = st.sum;
+          v.dec.Round(&v.dec, ex->distinctFracs[a], gxp::ModeHalfUp);
+        } else {  // AVG_DISTINCT = sum/count with DivPrecisionIncrement
+          v.type = GX_TYPE_DECIMAL;
+          MyDecimal den, res;
+          den.FromInt(st.cnt);
+          int32_t ec = gxp::DecimalDiv(&st.sum, &den, &res, gxp::kDivFracIncr);
+          if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
+            ex->err = "distinct avg division failed";
+            return GX_ERR_INTERNAL;
+          }
+          res.Round(&res, ex->distinctFracs[a], gxp::ModeHalfUp);
+          v.dec = res;
+        }
+        row.push_back(std::move(v));
+      }
+      folded.push_back(std::move(row));
+    }
+    ex->resultRows = std::move(folded);
   }
   applyPostSort(ex);
   return GX_OK;
@@ -5592,6 +5737,11 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
       ex->err = "FINAL agg child must be a bound source";
       return ex;
     }
+    for (int f : rn.aggFuncs)
+      if (f >= GX_AGG_COUNT_DISTINCT) {
+        ex->err = "DISTINCT aggregates support COMPLETE mode only";
+        return ex;
+      }
     ex->isFinalHost = true;
     ex->sourceNode = rn.child;
   } else if (rn.kind == PK_HASHAGG) {

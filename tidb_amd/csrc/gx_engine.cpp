@@ -145,6 +145,7 @@ struct gx_exec {
   // keys. distinctNKeys >= 0 marks the rewrite; funcs/fracs are the USER's.
   std::vector<int> distinctFuncs;
   std::vector<int> distinctFracs;
+  std::vector<int> distinctArgSlot;  // per agg: appended-key index (0-based)
   int distinctNKeys = -1;
   int sourceNode = -1;
   gxp::FusedQueryDesc desc;
@@ -1742,12 +1743,13 @@ static int32_t compileFused(gx_exec* ex) {
         ex->err = "DISTINCT aggregates support COMPLETE mode only";
         return GX_ERR_INVALID;
       }
-      int argE = aggN.aggArgs.empty() ? -1 : aggN.aggArgs[0];
+      std::vector<int> argExprs;  // appended inner keys, deduped
+      std::vector<int> slotOf(aggN.aggFuncs.size(), -1);
       for (size_t a = 0; a < aggN.aggFuncs.size(); a++) {
-        if (aggN.aggFuncs[a] < GX_AGG_COUNT_DISTINCT ||
-            aggN.aggArgs[a] != argE || argE < 0) {
-          ex->err = "DISTINCT aggregates must all be DISTINCT over one "
-                    "shared arg column this round";
+        int argE = aggN.aggArgs[a];
+        if (aggN.aggFuncs[a] < GX_AGG_COUNT_DISTINCT || argE < 0) {
+          ex->err = "DISTINCT aggregates cannot mix with plain aggregates "
+                    "this round";
           return GX_ERR_INVALID;
         }
         if (aggN.aggFuncs[a] != GX_AGG_COUNT_DISTINCT &&
@@ -1755,11 +1757,26 @@ static int32_t compileFused(gx_exec* ex) {
           ex->err = "SUM/AVG DISTINCT takes a decimal arg (cast ints)";
           return GX_ERR_INVALID;
         }
+        // share one inner key per distinct ARG: same expr id, or two
+        // colrefs naming the same column
+        for (size_t p = 0; p < argExprs.size() && slotOf[a] < 0; p++) {
+          const PExpr& x = ex->plan.exprs[argExprs[p]];
+          const PExpr& y = ex->plan.exprs[argE];
+          if (argExprs[p] == argE ||
+              (x.kind == EK_COLREF && y.kind == EK_COLREF &&
+               x.colIdx == y.colIdx && x.retType == y.retType))
+            slotOf[a] = (int)p;
+        }
+        if (slotOf[a] < 0) {
+          slotOf[a] = (int)argExprs.size();
+          argExprs.push_back(argE);
+        }
       }
       ex->distinctFuncs = aggN.aggFuncs;
       ex->distinctFracs = aggN.aggFracs;
+      ex->distinctArgSlot = slotOf;
       ex->distinctNKeys = (int)aggN.exprs.size();
-      aggN.exprs.push_back(argE);
+      for (int e : argExprs) aggN.exprs.push_back(e);
       aggN.aggFuncs.assign(1, GX_AGG_COUNT);
       aggN.aggArgs.assign(1, -1);
       aggN.aggFracs.assign(1, 0);
@@ -3330,11 +3347,12 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
     ex->resultRows.push_back(std::move(row));
   }
   if (ex->distinctNKeys >= 0) {
-    // fold the deduplicated (keys..., value) rows back to the user's
-    // schema: per orig-key group, count/sum/avg the non-NULL values (every
-    // row is already a UNIQUE (keys, value) combination — the device group
-    // table was the distinct check)
+    // fold the deduplicated (keys..., args...) rows back to the user's
+    // schema. Inner rows are unique (keys, arg tuple) combinations; a
+    // single arg's values can still repeat across rows (other args
+    // differ), so each agg dedups ITS arg column with a per-group set.
     const int nk = ex->distinctNKeys;
+    const size_t nA = ex->distinctFuncs.size();
     auto cmpVal = [](const OutRowVal& a, const OutRowVal& b) -> int {
       if (a.isNull != b.isNull) return a.isNull ? -1 : 1;
       if (a.isNull) return 0;
@@ -3350,11 +3368,18 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
           return a.i64 < b.i64 ? -1 : (a.i64 > b.i64 ? 1 : 0);
       }
     };
-    struct DState {
-      std::vector<OutRowVal> keys;
+    auto valLess = [&](const OutRowVal& a, const OutRowVal& b) {
+      return cmpVal(a, b) < 0;
+    };
+    struct AggAcc {
+      std::set<OutRowVal, std::function<bool(const OutRowVal&,
+                                             const OutRowVal&)>> seen;
       int64_t cnt = 0;
       MyDecimal sum;
-      bool hasSum = false;
+    };
+    struct DState {
+      std::vector<OutRowVal> keys;
+      std::vector<AggAcc> accs;
     };
     auto keyLess = [&](const std::vector<OutRowVal>* a,
                        const std::vector<OutRowVal>* b) {
@@ -3373,52 +3398,70 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
       if (it == idx.end()) {
         DState st;
         st.keys.assign(r.begin(), r.begin() + nk);
-        st.sum.FromInt(0);
+        for (size_t a = 0; a < nA; a++) {
+          AggAcc acc{std::set<OutRowVal, std::function<bool(
+                         const OutRowVal&, const OutRowVal&)>>(valLess)};
+          acc.sum.FromInt(0);
+          st.accs.push_back(std::move(acc));
+        }
         states.push_back(std::move(st));
         si = states.size() - 1;
         idx.emplace(&r, si);
       } else {
         si = it->second;
       }
-      const OutRowVal& v = r[nk];
-      if (v.isNull) continue;  // NULL values never count as distinct
       DState& st = states[si];
-      st.cnt++;
-      if (v.type == GX_TYPE_DECIMAL) {
-        MyDecimal tmp;
-        int32_t ec = gxp::DecimalAdd(&st.sum, &v.dec, &tmp);
-        if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
-          ex->err = "distinct sum overflow";
-          return GX_ERR_INTERNAL;
+      for (size_t a = 0; a < nA; a++) {
+        const OutRowVal& v = r[nk + ex->distinctArgSlot[a]];
+        if (v.isNull) continue;  // NULL values never count as distinct
+        AggAcc& acc = st.accs[a];
+        if (!acc.seen.insert(v).second) continue;
+        acc.cnt++;
+        if (v.type == GX_TYPE_DECIMAL &&
+            ex->distinctFuncs[a] != GX_AGG_COUNT_DISTINCT) {
+          MyDecimal tmp;
+          int32_t ec = gxp::DecimalAdd(&acc.sum, &v.dec, &tmp);
+          if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
+            ex->err = "distinct sum overflow";
+            return GX_ERR_INTERNAL;
+          }
+          acc.sum = tmp;
         }
-        st.sum = tmp;
-        st.hasSum = true;
       }
     }
-    if (nk == 0 && states.empty()) states.emplace_back();  // scalar default
-    // NOTE: idx keys point into the OLD resultRows — drop before rebuild
-    idx.clear();
+    if (nk == 0 && states.empty()) {  // scalar default row over zero rows
+      DState st;
+      for (size_t a = 0; a < nA; a++) {
+        AggAcc acc{std::set<OutRowVal, std::function<bool(
+                       const OutRowVal&, const OutRowVal&)>>(valLess)};
+        acc.sum.FromInt(0);
+        st.accs.push_back(std::move(acc));
+      }
+      states.push_back(std::move(st));
+    }
+    idx.clear();  // keys pointed into the OLD resultRows
     std::vector<std::vector<OutRowVal>> folded;
     folded.reserve(states.size());
     for (DState& st : states) {
       std::vector<OutRowVal> row = st.keys;
-      for (size_t a = 0; a < ex->distinctFuncs.size(); a++) {
+      for (size_t a = 0; a < nA; a++) {
+        AggAcc& acc = st.accs[a];
         OutRowVal v;
         if (ex->distinctFuncs[a] == GX_AGG_COUNT_DISTINCT) {
           v.type = GX_TYPE_I64;
-          v.i64 = st.cnt;
-        } else if (st.cnt == 0) {
+          v.i64 = acc.cnt;
+        } else if (acc.cnt == 0) {
           v.type = GX_TYPE_DECIMAL;
           v.isNull = true;
         } else if (ex->distinctFuncs[a] == GX_AGG_SUM_DISTINCT) {
           v.type = GX_TYPE_DECIMAL;
-          v.dec = st.sum;
+          v.dec = acc.sum;
           v.dec.Round(&v.dec, ex->distinctFracs[a], gxp::ModeHalfUp);
         } else {  // AVG_DISTINCT = sum/count with DivPrecisionIncrement
           v.type = GX_TYPE_DECIMAL;
           MyDecimal den, res;
-          den.FromInt(st.cnt);
-          int32_t ec = gxp::DecimalDiv(&st.sum, &den, &res, gxp::kDivFracIncr);
+          den.FromInt(acc.cnt);
+          int32_t ec = gxp::DecimalDiv(&acc.sum, &den, &res, gxp::kDivFracIncr);
           if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
             ex->err = "distinct avg division failed";
             return GX_ERR_INTERNAL;

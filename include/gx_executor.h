@@ -101,7 +101,10 @@ enum {
   /* IS [NOT] NULL (builtinIntIsNullSig / builtinDecimalIsNullSig /
    * builtinStringIsNullSig family, builtin_op_vec.go): unary, any column
    * type; result i64 0/1 and NEVER NULL (the null bit is the value). */
-  GX_F_IS_NULL = 37, GX_F_IS_NOT_NULL = 38
+  GX_F_IS_NULL = 37, GX_F_IS_NOT_NULL = 38,
+  /* TRIM(str) (builtinTrim1ArgSig, builtin_string.go spaceChars = " "):
+   * removes leading AND trailing 0x20 bytes only. */
+  GX_F_TRIM = 39
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

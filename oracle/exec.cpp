@@ -261,6 +261,20 @@ static int32_t evalString(EvalCtx& ctx, const Expr& e, const Chunk& in,
     }
     return GX_OK;
   }
+  if (e.func == GX_F_TRIM) {  // builtinTrim1ArgSig: strip 0x20 both ends
+    out.type = GX_TYPE_STRING;
+    out.offsets.assign(1, 0);
+    for (int i = 0; i < n; i++) {
+      if (a.isNull(i)) { out.appendNull(); continue; }
+      int len;
+      const uint8_t* p = a.getBytes(i, &len);
+      int st = 0, en = len;
+      while (st < en && p[st] == ' ') st++;
+      while (en > st && p[en - 1] == ' ') en--;
+      out.appendBytes((const char*)p + st, en - st);
+    }
+    return GX_OK;
+  }
   if (e.func == GX_F_LIKE_PREFIX) {  // builtinLikeSig 'abc%' fast path
     const Expr& pat = ctx.plan->exprs[e.args[1]];
     out.type = GX_TYPE_I64;
@@ -310,7 +324,8 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       if (e.func <= GX_F_NE) return evalCompare(ctx, e, in, out);
       if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT)
         return evalCast(ctx, e, in, out);
-      if (e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER)
+      if ((e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER) ||
+          e.func == GX_F_TRIM)
         return evalString(ctx, e, in, out);
       if (e.func == GX_F_IS_NULL || e.func == GX_F_IS_NOT_NULL) {
         // builtin*IsNullSig (builtin_op_vec.go): 0/1, never NULL

@@ -154,6 +154,56 @@ def test_lower_parity():
     assert got == want
 
 
+def _run_trim(lib):
+    """TRIM (builtinTrim1ArgSig, spaceChars=' '): alone, composed under and
+    over SUBSTR, and with a case op — a both-ends strip stays a contiguous
+    window, so the whole chain still folds into one windowed view."""
+    from tests.gxlib import GX_F_TRIM
+    ch, rows = _chunk(n=2000, seed=13)
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_STRING, GX_TYPE_I64])
+    s = b.colref(0, GX_TYPE_STRING)
+    exprs = [
+        b.call(GX_F_TRIM, GX_TYPE_STRING, 0, s),
+        b.call(GX_F_TRIM, GX_TYPE_STRING, 0,
+               b.call(GX_F_SUBSTR, GX_TYPE_STRING, 0, s, b.const_i64(2),
+                      b.const_i64(5))),
+        b.call(GX_F_SUBSTR, GX_TYPE_STRING, 0,
+               b.call(GX_F_TRIM, GX_TYPE_STRING, 0, s), b.const_i64(1),
+               b.const_i64(4)),
+        b.call(GX_F_UPPER, GX_TYPE_STRING, 0,
+               b.call(GX_F_TRIM, GX_TYPE_STRING, 0, s)),
+    ]
+    root = b.projection(src, exprs)
+    ex = b.build(root)
+    ex.bind_chunks(src, ch)
+    ex.open()
+    out = ex.pull_all([GX_TYPE_STRING] * 4, [0] * 4, data_caps=[1 << 18] * 4)
+    ex.close()
+    ex.free()
+    b.free()
+    return out, rows
+
+
+def test_oracle_trim():
+    got, rows = _run_trim(load_oracle())
+    for (o0, o1, o2, o3), (s, _) in zip(got, rows):
+        if s is None:
+            assert (o0, o1, o2, o3) == (None, None, None, None)
+            continue
+        assert o0 == s.strip(" ")
+        assert o1 == _py_substr(s, 2, 5).strip(" ")
+        assert o2 == _py_substr(s.strip(" "), 1, 4)
+        assert o3 == _ascii_upper(s.strip(" "))
+
+
+@pytest.mark.gpu
+def test_trim_parity():
+    want, _ = _run_trim(load_oracle())
+    got, _ = _run_trim(load_product())
+    assert got == want
+
+
 def _run_like_selection(lib, prefix):
     ch, rows = _chunk()
     b = P.Builder(lib)

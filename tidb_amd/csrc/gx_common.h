@@ -497,7 +497,7 @@ struct StrProg {
   int32_t col = -1;
   int32_t nWin = 0;
   int64_t winPos[kMaxStrWin];  // MySQL 1-based; negative counts from the end
-  int64_t winLen[kMaxStrWin];
+  int64_t winLen[kMaxStrWin];  // -1 = a TRIM step (scan 0x20 off both ends)
   int32_t upper = 0;
   int32_t lower = 0;  // outermost case op wins (compileStrProg clears the other)
   // per-run device temps (engine-allocated)

@@ -1504,6 +1504,14 @@ static bool compileStrProg(gx_exec* ex, int exprId,
     sp->lower = e.func == GX_F_LOWER;
     return true;
   }
+  if (e.func == GX_F_TRIM && e.args.size() == 1) {
+    if (!compileStrProg(ex, e.args[0], types, sp)) return false;
+    if (sp->nWin >= gxp::kMaxStrWin) return false;
+    sp->winPos[sp->nWin] = 0;
+    sp->winLen[sp->nWin] = -1;  // trim pseudo-window
+    sp->nWin++;
+    return true;
+  }
   if (e.func == GX_F_SUBSTR && e.args.size() == 3) {
     if (!compileStrProg(ex, e.args[0], types, sp)) return false;
     const PExpr& p = ex->plan.exprs[e.args[1]];
@@ -1583,7 +1591,7 @@ static int32_t compileProject(gx_exec* ex) {
       // string outputs: SUBSTR/UPPER chains fold into one windowed view
       gxp::StrProg sp{};
       if (!compileStrProg(ex, eid, *childTypes, &sp)) {
-        ex->err = "unsupported string projection (SUBSTR/UPPER/LOWER chains over "
+        ex->err = "unsupported string projection (SUBSTR/UPPER/LOWER/TRIM chains over "
                   "a string column this round)";
         return GX_ERR_INVALID;
       }

@@ -3192,6 +3192,12 @@ __global__ void strWindowKernel(const ProjDesc* __restrict__ dp, int pi) {
       // 1-based pos, negative from the end, out-of-range/len<=0 -> empty
       for (int w = 0; w < sp.nWin; w++) {
         int64_t pos = sp.winPos[w], L = sp.winLen[w];
+        if (L == -1) {  // TRIM: a both-ends space strip stays a window
+          auto p = gptr<uint8_t>(c.data);
+          while (len > 0 && p[s] == ' ') { s++; len--; }
+          while (len > 0 && p[s + len - 1] == ' ') len--;
+          continue;
+        }
         int64_t start = pos < 0 ? len + pos + 1 : pos;
         if (start < 1 || start > len || L <= 0) {
           len = 0;

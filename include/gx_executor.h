@@ -104,7 +104,11 @@ enum {
   GX_F_IS_NULL = 37, GX_F_IS_NOT_NULL = 38,
   /* TRIM(str) (builtinTrim1ArgSig, builtin_string.go spaceChars = " "):
    * removes leading AND trailing 0x20 bytes only. */
-  GX_F_TRIM = 39
+  GX_F_TRIM = 39,
+  /* IFNULL(a, b) (builtinIfNullSig, builtin_control_vec_generated.go):
+   * first non-NULL operand; NULL only when both are. COALESCE(a,b,c,...)
+   * is the chain IFNULL(a, IFNULL(b, c)). */
+  GX_F_IFNULL = 40
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

@@ -327,6 +327,24 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       if ((e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER) ||
           e.func == GX_F_TRIM)
         return evalString(ctx, e, in, out);
+      if (e.func == GX_F_IFNULL) {
+        // builtinIfNullSig: first non-NULL operand per row
+        Column a, b2;
+        int32_t err = evalVec(ctx, e.args[0], in, a);
+        if (err) return err;
+        err = evalVec(ctx, e.args[1], in, b2);
+        if (err) return err;
+        out.reset();
+        out.type = a.type;
+        out.frac = std::max(a.frac, b2.frac);
+        if (out.isVarlen()) out.offsets.assign(1, 0);
+        for (int i = 0; i < in.numRows(); i++) {
+          const Column& pick = a.isNull(i) ? b2 : a;
+          if (pick.isNull(i)) out.appendNull();
+          else out.appendFrom(pick, i);
+        }
+        return GX_OK;
+      }
       if (e.func == GX_F_IS_NULL || e.func == GX_F_IS_NOT_NULL) {
         // builtin*IsNullSig (builtin_op_vec.go): 0/1, never NULL
         Column a;

@@ -132,3 +132,65 @@ def test_isnull_fused_parity(func):
     _, got = _run_fused(load_product(), func)
     assert got == want
     assert len(got) > 10
+
+
+def _run_ifnull(lib):
+    """IFNULL (builtinIfNullSig): in a projection (value or const default,
+    mixed scales engine-aligned) and as an aggregate arg
+    (sum(IFNULL(d, 0.00)) counts every row)."""
+    from tests.gxlib import GX_AGG_COUNT, GX_AGG_SUM, GX_F_IFNULL
+    rows, chunks = _data(lib)
+    b = P.Builder(lib)
+    src = b.source(TYPES, FRACS)
+    d = b.colref(2, GX_TYPE_DECIMAL, 2)
+    v = b.colref(1, GX_TYPE_I64)
+    zero = b.const_dec(_dec(lib, "0.00"))
+    proj = b.projection(src, [
+        b.colref(0, GX_TYPE_I64),
+        b.call(GX_F_IFNULL, GX_TYPE_DECIMAL, 2, d, zero),
+        b.call(GX_F_IFNULL, GX_TYPE_I64, 0, v, b.const_i64(-1)),
+    ])
+    ex = b.build(proj)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out1 = ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64], [0, 2, 0])
+    ex.close()
+    ex.free()
+
+    agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                    [(GX_AGG_SUM, b.call(GX_F_IFNULL, GX_TYPE_DECIMAL, 2, d,
+                                         zero), 2),
+                     (GX_AGG_COUNT, -1, 0)])
+    ex = b.build(agg)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out2 = sorted(ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64],
+                              [0, 2, 0]))
+    ex.close()
+    ex.free()
+    b.free()
+    return rows, out1, out2
+
+
+def test_oracle_ifnull():
+    from fractions import Fraction
+    rows, out1, out2 = _run_ifnull(load_oracle())
+    for (k, dv, vv), r in zip(out1, rows):
+        assert k == r[0]
+        assert dv == (r[2] if r[2] is not None else "0.00")
+        assert vv == (r[1] if r[1] is not None else -1)
+    want = {}
+    for r in rows:
+        s, c = want.get(r[0], (Fraction(0), 0))
+        want[r[0]] = (s + (Fraction(r[2]) if r[2] is not None else 0), c + 1)
+    assert out2 == sorted((k, f"{s.numerator / s.denominator:.2f}"
+                           if s.denominator == 1 or True else s, c)
+                          for k, (s, c) in want.items())
+
+
+@pytest.mark.gpu
+def test_ifnull_parity():
+    _, w1, w2 = _run_ifnull(load_oracle())
+    _, g1, g2 = _run_ifnull(load_product())
+    assert g1 == w1
+    assert g2 == w2

@@ -55,6 +55,8 @@ enum VmOp : int32_t {
                        // cast family, ProduceDecWithSpecifiedTp/ToInt)
   VM_STRLEN = 9,       // dst <- byte length of string column a
                        // (builtinLengthSig; offsets fetch slot in c)
+  VM_IFNULL = 10,      // dst <- a if a not NULL else b (builtinIfNullSig;
+                       // scales engine-aligned; COALESCE = chained IFNULL)
 };
 
 struct VmIns {

@@ -794,6 +794,16 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
       if (ra < 0) return -1;
       int rb = vmCompile(ex, B, e.args[1], &sb);
       if (rb < 0) return -1;
+      if (e.func == GX_F_IFNULL) {
+        // builtinIfNullSig: operands align to one scale, result keeps it
+        int target = std::max(sa, sb);
+        if (sa < target) ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa, -1);
+        if (sb < target) rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb, -1);
+        if (ra < 0 || rb < 0) break;
+        reg = emit(gxp::VM_IFNULL, allocReg(), ra, rb, -1);
+        *scaleOut = target;
+        break;
+      }
       int op;
       switch (e.func) {
         case GX_F_PLUS: op = gxp::VM_ADD; break;

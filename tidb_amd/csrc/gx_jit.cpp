@@ -152,6 +152,11 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
           << " = *(T*)&c64; } }\n"
           << "  " << nv << " = false;\n";
         break;
+      case gxp::VM_IFNULL:
+        s << "  { bool t3 = n" << ins.a << "; " << v << " = t3 ? v" << ins.b
+          << " : v" << ins.a << "; " << nv << " = t3 && n" << ins.b
+          << "; }\n";
+        break;
       case gxp::VM_ADD:
         s << "  { T t2 = VT<WIDE>::add(v" << ins.a << ", v" << ins.b
           << ", &ovf); bool t3 = n" << ins.a << " || n" << ins.b << "; " << v
@@ -640,6 +645,11 @@ static void emitJaVm(std::ostringstream& s, const JoinAggDesc& d) {
           << "      else { int64_t c64 = " << d.constLo[ins.a] << "LL; " << v
           << " = *(T*)&c64; } }\n"
           << "    const bool " << nv << " = false;\n";
+        break;
+      case gxp::VM_IFNULL:
+        s << "    T " << v << " = n" << ins.a << " ? v" << ins.b << " : v"
+          << ins.a << "; bool " << nv << " = n" << ins.a << " && n" << ins.b
+          << ";\n";
         break;
       case gxp::VM_ADD:
         s << "    T " << v << " = VT<WIDE>::add(v" << ins.a << ", v" << ins.b

@@ -242,6 +242,13 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
         break;
+      case VM_IFNULL: {
+        // builtinIfNullSig: first non-NULL operand; NULL only if both are
+        bool an = vm.isNull(ins.a);
+        vm.set(ins.dst, an ? vm.get(ins.b) : vm.get(ins.a));
+        vm.setNull(ins.dst, an && vm.isNull(ins.b));
+        break;
+      }
       case VM_MUL: {
         bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
         typename VT<WIDE>::T v = VT<WIDE>::zero();
@@ -1011,6 +1018,12 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
           vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
+        case VM_IFNULL: {
+          bool an = vm.isNull(ins.a);
+          vm.set(ins.dst, an ? vm.get(ins.b) : vm.get(ins.a));
+          vm.setNull(ins.dst, an && vm.isNull(ins.b));
+          break;
+        }
         case VM_MUL: {
           bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
           typename VT<WIDE>::T v = VT<WIDE>::zero();
@@ -1422,6 +1435,13 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
       case VM_SUB:
         vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
         break;
+      case VM_IFNULL: {
+        // builtinIfNullSig: first non-NULL operand; NULL only if both are
+        bool an = vm.isNull(ins.a);
+        vm.set(ins.dst, an ? vm.get(ins.b) : vm.get(ins.a));
+        vm.setNull(ins.dst, an && vm.isNull(ins.b));
+        break;
+      }
       case VM_MUL:
         vm.set(ins.dst, VT<WIDE>::mul(vm.get(ins.a), vm.get(ins.b), &ovf));
         break;
@@ -2793,6 +2813,12 @@ __global__ void projectKernel(const ProjDesc* __restrict__ dp) {
           vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
+        case VM_IFNULL: {
+          bool an = vm.isNull(ins.a);
+          vm.set(ins.dst, an ? vm.get(ins.b) : vm.get(ins.a));
+          vm.setNull(ins.dst, an && vm.isNull(ins.b));
+          break;
+        }
         case VM_MUL: {
           bool nul = vm.isNull(ins.a) || vm.isNull(ins.b);
           typename VT<WIDE>::T v = VT<WIDE>::zero();

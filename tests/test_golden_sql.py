@@ -192,3 +192,43 @@ def test_max_both_sides_over_join():
     ex.free()
     b.free()
     assert got == [(1, 1, 2)]
+
+
+def test_count_distinct_with_nulls():
+    """aggregate.result (distinct agg block) — t(a,b) with rows
+    (NULL,NULL),(1,NULL),(NULL,1),(1,2),(3,4):
+    `select count(distinct a) from t` = 2 (NULLs excluded, dup 1 merged),
+    with and without distinct-agg push-down (one execution shape here)."""
+    from tests.gxlib import GX_TYPE_I64
+    lib = load_oracle()
+    rows = [(None, None), (1, None), (None, 1), (1, 2), (3, 4)]
+
+    def plan(b, src):
+        return b.hashagg(src, [], [(6, b.colref(0, GX_TYPE_I64), 0)])
+
+    got = _run(lib, rows, plan, [GX_TYPE_I64], [0],
+               [GX_TYPE_I64, GX_TYPE_I64], [0, 0])
+    assert got == [(2,)]
+
+
+def test_sum_avg_distinct_group_by():
+    """aggregate.result — t1(a,b) = (1,1),(2,2),(3,3),(1,4),(1,1),(3,5),
+    (2,2),(3,5),(3,3): `select avg(distinct b) ... group by a order by a`
+    = 2.5000, 2.0000, 4.0000 and `sum(distinct b)` = 5, 2, 8 (the planner
+    wraps int distinct args in a decimal cast; avg displays at frac 4)."""
+    from tests.gxlib import GX_TYPE_I64
+    lib = load_oracle()
+    rows = [(1, 1), (2, 2), (3, 3), (1, 4), (1, 1), (3, 5), (2, 2), (3, 5),
+            (3, 3)]
+
+    def plan(b, src):
+        bd = b.call(GX_F_CAST_DEC, GX_TYPE_DECIMAL, 0,
+                    b.colref(1, GX_TYPE_I64))
+        return b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                         [(8, bd, 4), (7, bd, 0)])
+
+    got = _run(lib, rows, plan, [GX_TYPE_I64, GX_TYPE_DECIMAL,
+                                 GX_TYPE_DECIMAL], [0, 4, 0],
+               [GX_TYPE_I64, GX_TYPE_I64], [0, 0])
+    assert sorted(got) == [(1, "2.5000", "5"), (2, "2.0000", "2"),
+                           (3, "4.0000", "8")]

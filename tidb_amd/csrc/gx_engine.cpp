@@ -587,6 +587,21 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       if (ra < 0) return -1;
       int rb = compileExpr(ex, e.args[1], &sb);
       if (rb < 0) return -1;
+      if (e.func == GX_F_IFNULL) {
+        // builtinIfNullSig: align operand scales, result keeps the max
+        int target = std::max(sa, sb);
+        int tmpA = -1, tmpB = -1;
+        if (sa < target) tmpA = ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa);
+        if (sb < target) tmpB = rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb);
+        if (ra < 0 || rb < 0) break;
+        reg = emit(gxp::VM_IFNULL, allocReg(), ra, rb);
+        if (tmpA >= 0) ex->vmFreeRegs.push_back(tmpA);
+        if (tmpB >= 0) ex->vmFreeRegs.push_back(tmpB);
+        *scaleOut = target;
+        release(e.args[0]);
+        release(e.args[1]);
+        break;
+      }
       int op;
       switch (e.func) {
         case GX_F_PLUS: op = gxp::VM_ADD; break;

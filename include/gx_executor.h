@@ -108,7 +108,13 @@ enum {
   /* IFNULL(a, b) (builtinIfNullSig, builtin_control_vec_generated.go):
    * first non-NULL operand; NULL only when both are. COALESCE(a,b,c,...)
    * is the chain IFNULL(a, IFNULL(b, c)). */
-  GX_F_IFNULL = 40
+  GX_F_IFNULL = 40,
+  /* argument TUPLE for multi-column DISTINCT aggregates —
+   * count(distinct a, b, ...) (aggregation descriptor with multiple args;
+   * rows with ANY NULL element are excluded, aggregate.result's
+   * count(distinct b,c,d) golden). NOT a value expression: valid only as
+   * a COUNT_DISTINCT arg. */
+  GX_F_TUPLE = 41
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

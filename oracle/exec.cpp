@@ -327,6 +327,10 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       if ((e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER) ||
           e.func == GX_F_TRIM)
         return evalString(ctx, e, in, out);
+      if (e.func == GX_F_TUPLE) {
+        if (ctx.err) *ctx.err = "tuple is only valid as a DISTINCT aggregate argument";
+        return GX_ERR_INVALID;
+      }
       if (e.func == GX_F_IFNULL) {
         // builtinIfNullSig: first non-NULL operand per row
         Column a, b2;
@@ -570,12 +574,20 @@ class HashAggExec : public Exec {
   }
 
   int32_t open() override {
-    for (int f : node_.aggFuncs)
-      if (f >= GX_AGG_COUNT_DISTINCT &&
-          node_.aggMode != GX_AGG_MODE_COMPLETE) {
+    for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
+      int f = node_.aggFuncs[a];
+      if (f < GX_AGG_COUNT_DISTINCT) continue;
+      if (node_.aggMode != GX_AGG_MODE_COMPLETE) {
         err = "DISTINCT aggregates support COMPLETE mode only";
         return GX_ERR_INVALID;
       }
+      if (f != GX_AGG_COUNT_DISTINCT && node_.aggArgs[a] >= 0 &&
+          plan_.exprs[node_.aggArgs[a]].func == GX_F_TUPLE &&
+          plan_.exprs[node_.aggArgs[a]].kind == EK_CALL) {
+        err = "SUM/AVG DISTINCT take a single column";
+        return GX_ERR_INVALID;
+      }
+    }
     done_ = false;
     emitPos_ = 0;
     order_.clear();
@@ -674,10 +686,11 @@ class HashAggExec : public Exec {
       case GX_AGG_COUNT_DISTINCT:
       case GX_AGG_SUM_DISTINCT:
       case GX_AGG_AVG_DISTINCT: {
-        // distinct wrappers: NULLs excluded; each VALUE updates once per
-        // group (value identity = the codec HashGroupKey encoding, the
-        // same identity the reference's distinct checker uses)
-        if (isNull || !vkey) break;
+        // distinct wrappers: NULLs excluded (tuple args: ANY null element
+        // — encoded as an EMPTY key); each VALUE updates once per group
+        // (value identity = the codec HashGroupKey encoding, the same
+        // identity the reference's distinct checker uses)
+        if (!vkey || vkey->empty()) break;
         if (!s.seen.insert(*vkey).second) break;
         if (func == GX_AGG_COUNT_DISTINCT) { s.i64++; break; }
         if (valueType == GX_TYPE_DECIMAL) {
@@ -809,15 +822,38 @@ class HashAggExec : public Exec {
       std::vector<const Column*> argPtr(node_.aggFuncs.size(), nullptr);
       std::vector<std::vector<std::string>> valKeys(node_.aggFuncs.size());
       for (size_t a = 0; a < node_.aggFuncs.size(); a++) {
-        if (node_.aggArgs[a] >= 0) {
-          ec = evalVec(ctx, node_.aggArgs[a], in, argCols[a]);
-          if (ec) return ec;
-          argPtr[a] = &argCols[a];
-          if (node_.aggFuncs[a] >= GX_AGG_COUNT_DISTINCT) {
-            valKeys[a].assign(n, std::string());
-            ec = HashGroupKeyCol(argCols[a], valKeys[a]);
+        if (node_.aggArgs[a] < 0) continue;
+        const Expr& ae = plan_.exprs[node_.aggArgs[a]];
+        if (ae.kind == EK_CALL && ae.func == GX_F_TUPLE) {
+          // multi-column distinct: concatenated element encodings; a row
+          // with ANY NULL element gets an EMPTY key (= excluded)
+          valKeys[a].assign(n, std::string());
+          std::vector<uint8_t> anyNull(n, 0);
+          for (int el : ae.args) {
+            Column c;
+            ec = evalVec(ctx, el, in, c);
             if (ec) return ec;
+            std::vector<std::string> part(n);
+            ec = HashGroupKeyCol(c, part);
+            if (ec) return ec;
+            for (int i = 0; i < n; i++) {
+              if (c.isNull(i)) anyNull[i] = 1;
+              valKeys[a][i] += part[i];
+            }
           }
+          for (int i = 0; i < n; i++)
+            if (anyNull[i]) valKeys[a][i].clear();
+          continue;
+        }
+        ec = evalVec(ctx, node_.aggArgs[a], in, argCols[a]);
+        if (ec) return ec;
+        argPtr[a] = &argCols[a];
+        if (node_.aggFuncs[a] >= GX_AGG_COUNT_DISTINCT) {
+          valKeys[a].assign(n, std::string());
+          ec = HashGroupKeyCol(argCols[a], valKeys[a]);
+          if (ec) return ec;
+          for (int i = 0; i < n; i++)
+            if (argCols[a].isNull(i)) valKeys[a][i].clear();
         }
       }
       for (int i = 0; i < n; i++) {

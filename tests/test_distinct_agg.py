@@ -256,3 +256,60 @@ def test_join_distinct_parity():
     want = _run_join_distinct(load_oracle())
     got = _run_join_distinct(load_product())
     assert got == want
+
+
+def _run_multicol(lib, rows, group=True):
+    """count(distinct v, d, s) — the multi-column distinct: rows with ANY
+    NULL element are excluded; identity is the element tuple."""
+    from tests.gxlib import GX_F_TUPLE
+    b = P.Builder(lib)
+    src = b.source(TYPES, FRACS)
+    tup = b.call(GX_F_TUPLE, GX_TYPE_I64, 0,
+                 b.colref(1, GX_TYPE_I64), b.colref(2, GX_TYPE_DECIMAL, 2),
+                 b.colref(3, GX_TYPE_STRING))
+    keys = [b.colref(0, GX_TYPE_I64)] if group else []
+    agg = b.hashagg(src, keys, [(GX_AGG_COUNT_DISTINCT, tup, 0)])
+    ex = b.build(agg)
+    ex.bind_chunks(src, _chunks(lib, rows))
+    ex.open()
+    out_t = [GX_TYPE_I64, GX_TYPE_I64] if group else [GX_TYPE_I64]
+    got = sorted(ex.pull_all(out_t, [0] * len(out_t)))
+    ex.close()
+    ex.free()
+    b.free()
+    return got
+
+
+def test_oracle_multicol_distinct():
+    """Pins aggregate.result's count(distinct b,c,d) group-by-id pattern
+    (adapted to i64/decimal/string columns): every row with ANY NULL
+    element contributes 0; full rows count distinct tuples."""
+    lib = load_oracle()
+    rows = _data()
+    got = _run_multicol(lib, rows)
+    want = {}
+    for k, v, d, s in rows:
+        want.setdefault(k, set())
+        if v is not None and d is not None and s is not None:
+            want[k].add((v, d, s))
+    assert got == sorted((k, len(t)) for k, t in want.items())
+    # and the aggregate.result golden itself: one row per id; rows with any
+    # NULL -> 0, the single full row -> 1
+    g_rows = [[1, 1, "3.00", None], [2, 1, None, "6"], [3, None, "1.00", "2"],
+              [4, None, None, "1"], [5, None, "2.00", None],
+              [6, 3, None, None], [7, None, None, None],
+              [8, 1, "2.00", "3"]]
+    got = _run_multicol(lib, g_rows)
+    assert got == [(1, 0), (2, 0), (3, 0), (4, 0), (5, 0), (6, 0), (7, 0),
+                   (8, 1)]
+
+
+@pytest.mark.gpu
+def test_multicol_distinct_parity():
+    rows = _data()
+    want = _run_multicol(load_oracle(), rows)
+    got = _run_multicol(load_product(), rows)
+    assert got == want
+    want = _run_multicol(load_oracle(), rows, group=False)
+    got = _run_multicol(load_product(), rows, group=False)
+    assert got == want

@@ -145,7 +145,7 @@ struct gx_exec {
   // keys. distinctNKeys >= 0 marks the rewrite; funcs/fracs are the USER's.
   std::vector<int> distinctFuncs;
   std::vector<int> distinctFracs;
-  std::vector<int> distinctArgSlot;  // per agg: appended-key index (0-based)
+  std::vector<std::vector<int>> distinctArgSlot;  // per agg: appended-key indices (tuple args have several)
   int distinctNKeys = -1;
   int sourceNode = -1;
   gxp::FusedQueryDesc desc;
@@ -1777,7 +1777,21 @@ static int32_t compileFused(gx_exec* ex) {
         return GX_ERR_INVALID;
       }
       std::vector<int> argExprs;  // appended inner keys, deduped
-      std::vector<int> slotOf(aggN.aggFuncs.size(), -1);
+      std::vector<std::vector<int>> slotOf(aggN.aggFuncs.size());
+      auto slotFor = [&](int argE) {
+        // share one inner key per distinct ARG: same expr id, or two
+        // colrefs naming the same column
+        for (size_t p = 0; p < argExprs.size(); p++) {
+          const PExpr& x = ex->plan.exprs[argExprs[p]];
+          const PExpr& y = ex->plan.exprs[argE];
+          if (argExprs[p] == argE ||
+              (x.kind == EK_COLREF && y.kind == EK_COLREF &&
+               x.colIdx == y.colIdx && x.retType == y.retType))
+            return (int)p;
+        }
+        argExprs.push_back(argE);
+        return (int)argExprs.size() - 1;
+      };
       for (size_t a = 0; a < aggN.aggFuncs.size(); a++) {
         int argE = aggN.aggArgs[a];
         if (aggN.aggFuncs[a] < GX_AGG_COUNT_DISTINCT || argE < 0) {
@@ -1785,25 +1799,24 @@ static int32_t compileFused(gx_exec* ex) {
                     "this round";
           return GX_ERR_INVALID;
         }
+        const PExpr& ae = ex->plan.exprs[argE];
+        if (ae.kind == EK_CALL && ae.func == GX_F_TUPLE) {
+          // multi-column distinct (count(distinct a, b, ...)): each tuple
+          // element becomes an inner key; the fold excludes rows with ANY
+          // NULL element
+          if (aggN.aggFuncs[a] != GX_AGG_COUNT_DISTINCT || ae.args.empty()) {
+            ex->err = "tuple args take COUNT DISTINCT only";
+            return GX_ERR_INVALID;
+          }
+          for (int el : ae.args) slotOf[a].push_back(slotFor(el));
+          continue;
+        }
         if (aggN.aggFuncs[a] != GX_AGG_COUNT_DISTINCT &&
-            ex->plan.exprs[argE].retType != GX_TYPE_DECIMAL) {
+            ae.retType != GX_TYPE_DECIMAL) {
           ex->err = "SUM/AVG DISTINCT takes a decimal arg (cast ints)";
           return GX_ERR_INVALID;
         }
-        // share one inner key per distinct ARG: same expr id, or two
-        // colrefs naming the same column
-        for (size_t p = 0; p < argExprs.size() && slotOf[a] < 0; p++) {
-          const PExpr& x = ex->plan.exprs[argExprs[p]];
-          const PExpr& y = ex->plan.exprs[argE];
-          if (argExprs[p] == argE ||
-              (x.kind == EK_COLREF && y.kind == EK_COLREF &&
-               x.colIdx == y.colIdx && x.retType == y.retType))
-            slotOf[a] = (int)p;
-        }
-        if (slotOf[a] < 0) {
-          slotOf[a] = (int)argExprs.size();
-          argExprs.push_back(argE);
-        }
+        slotOf[a].push_back(slotFor(argE));
       }
       ex->distinctFuncs = aggN.aggFuncs;
       ex->distinctFracs = aggN.aggFracs;
@@ -3404,9 +3417,18 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
     auto valLess = [&](const OutRowVal& a, const OutRowVal& b) {
       return cmpVal(a, b) < 0;
     };
+    auto tupLess = [&](const std::vector<OutRowVal>& a,
+                       const std::vector<OutRowVal>& b) {
+      for (size_t k = 0; k < a.size() && k < b.size(); k++) {
+        int c = cmpVal(a[k], b[k]);
+        if (c) return c < 0;
+      }
+      return a.size() < b.size();
+    };
     struct AggAcc {
-      std::set<OutRowVal, std::function<bool(const OutRowVal&,
-                                             const OutRowVal&)>> seen;
+      std::set<std::vector<OutRowVal>,
+               std::function<bool(const std::vector<OutRowVal>&,
+                                  const std::vector<OutRowVal>&)>> seen;
       int64_t cnt = 0;
       MyDecimal sum;
     };
@@ -3432,8 +3454,10 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
         DState st;
         st.keys.assign(r.begin(), r.begin() + nk);
         for (size_t a = 0; a < nA; a++) {
-          AggAcc acc{std::set<OutRowVal, std::function<bool(
-                         const OutRowVal&, const OutRowVal&)>>(valLess)};
+          AggAcc acc{std::set<std::vector<OutRowVal>,
+                              std::function<bool(
+                                  const std::vector<OutRowVal>&,
+                                  const std::vector<OutRowVal>&)>>(tupLess)};
           acc.sum.FromInt(0);
           st.accs.push_back(std::move(acc));
         }
@@ -3445,15 +3469,21 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
       }
       DState& st = states[si];
       for (size_t a = 0; a < nA; a++) {
-        const OutRowVal& v = r[nk + ex->distinctArgSlot[a]];
-        if (v.isNull) continue;  // NULL values never count as distinct
+        std::vector<OutRowVal> tup;
+        bool anyNull = false;
+        for (int sl : ex->distinctArgSlot[a]) {
+          const OutRowVal& v = r[nk + sl];
+          anyNull |= v.isNull;
+          tup.push_back(v);
+        }
+        if (anyNull) continue;  // NULL (any element) never counts distinct
         AggAcc& acc = st.accs[a];
-        if (!acc.seen.insert(v).second) continue;
+        if (!acc.seen.insert(tup).second) continue;
         acc.cnt++;
-        if (v.type == GX_TYPE_DECIMAL &&
+        if (tup[0].type == GX_TYPE_DECIMAL &&
             ex->distinctFuncs[a] != GX_AGG_COUNT_DISTINCT) {
           MyDecimal tmp;
-          int32_t ec = gxp::DecimalAdd(&acc.sum, &v.dec, &tmp);
+          int32_t ec = gxp::DecimalAdd(&acc.sum, &tup[0].dec, &tmp);
           if (ec != gxp::E_OK && ec != gxp::E_TRUNCATED) {
             ex->err = "distinct sum overflow";
             return GX_ERR_INTERNAL;
@@ -3465,8 +3495,10 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
     if (nk == 0 && states.empty()) {  // scalar default row over zero rows
       DState st;
       for (size_t a = 0; a < nA; a++) {
-        AggAcc acc{std::set<OutRowVal, std::function<bool(
-                       const OutRowVal&, const OutRowVal&)>>(valLess)};
+        AggAcc acc{std::set<std::vector<OutRowVal>,
+                            std::function<bool(
+                                const std::vector<OutRowVal>&,
+                                const std::vector<OutRowVal>&)>>(tupLess)};
         acc.sum.FromInt(0);
         st.accs.push_back(std::move(acc));
       }

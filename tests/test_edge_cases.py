@@ -117,3 +117,49 @@ def test_sort_tiny_parity():
         b = run_sort(load_product(), KEYS_2, [0, 1], n_rows=n)
         assert [(r[7], r[0]) for r in a] == [(r[7], r[0]) for r in b]
         assert sorted(map(tuple, a)) == sorted(map(tuple, b))
+
+
+def _run_limit(lib, limit, offset, with_sel):
+    """Keyless TOPN == plain LIMIT/OFFSET (LimitExec, executor/limit.go):
+    child row order preserved, no sort."""
+    from tidb_amd.chunkpy import PyChunk
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    node = src
+    if with_sel:
+        from tests.gxlib import GX_F_LT
+        node = b.selection(src, [b.call(GX_F_LT, GX_TYPE_I64, 0,
+                                        b.colref(1, GX_TYPE_I64),
+                                        b.const_i64(500))])
+    root = b.topn(node, [], [], limit, offset)
+    ex = b.build(root)
+    ch = PyChunk([GX_TYPE_I64] * 2, 1000)
+    for i in range(1000):
+        ch.append_row([i, (i * 7) % 1000])
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    rows = ex.pull_all([GX_TYPE_I64] * 2)
+    ex.close()
+    ex.free()
+    b.free()
+    return rows
+
+
+def test_oracle_keyless_limit():
+    rows = _run_limit(load_oracle(), 7, 0, False)
+    assert rows == [(i, (i * 7) % 1000) for i in range(7)]
+    rows = _run_limit(load_oracle(), 5, 3, False)
+    assert [r[0] for r in rows] == [3, 4, 5, 6, 7]
+    rows = _run_limit(load_oracle(), 4, 2, True)
+    want = [(i, (i * 7) % 1000) for i in range(1000) if (i * 7) % 1000 < 500]
+    assert rows == want[2:6]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("limit,offset,with_sel", [
+    (7, 0, False), (5, 3, False), (4, 2, True), (2000, 0, True)])
+def test_keyless_limit_parity(limit, offset, with_sel):
+    from tests.gxlib import load_product
+    want = _run_limit(load_oracle(), limit, offset, with_sel)
+    got = _run_limit(load_product(), limit, offset, with_sel)
+    assert got == want

@@ -5569,6 +5569,18 @@ static int32_t emitTableChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   return GX_OK;
 }
 
+// LimitExec (executor/limit.go): a keyless TOPN is a plain LIMIT/OFFSET —
+// child row order preserved, offset rows skipped, limit rows kept
+static void applyBareLimit(gx_exec* ex) {
+  if (!ex->devSortKeys.empty() || ex->devSorted) return;
+  if (ex->devSortLimit < 0 && ex->devSortOffset <= 0) return;
+  int64_t n = ex->desc.table.nRows;
+  ex->srcPos = std::min<int64_t>(ex->devSortOffset, n);
+  if (ex->devSortLimit >= 0)
+    ex->desc.table.nRows = std::min<int64_t>(n, ex->srcPos + ex->devSortLimit);
+  ex->devSorted = true;  // applied once
+}
+
 static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
   // out-of-core sort decision comes BEFORE full materialization (spill runs
   // materialize row ranges themselves)
@@ -5583,6 +5595,7 @@ static int32_t emitSourceChunk(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
     rc = runDeviceSort(ex);
     if (rc) return rc;
   }
+  applyBareLimit(ex);
   return emitTableChunk(ex, out, rows_out);
 }
 
@@ -6214,6 +6227,7 @@ int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
         return rc;
       }
     }
+    applyBareLimit(ex);
     return emitTableChunk(ex, out, rows_out);
   }
   if (ex->isHashJoin) {
@@ -6237,6 +6251,13 @@ int32_t gx_next(gx_exec* ex, gx_chunk* out, int32_t* rows_out) {
         return rc;
       }
     }
+    if (ex->devSortKeys.empty() && !ex->devSorted && ex->joinSpill &&
+        (ex->devSortLimit >= 0 || ex->devSortOffset > 0)) {
+      ex->err = "LIMIT over an out-of-core join unsupported this round";
+      *rows_out = 0;
+      return GX_ERR_INVALID;
+    }
+    applyBareLimit(ex);
     int32_t rc = emitTableChunk(ex, out, rows_out);
     // out-of-core: stream the next partitions' outputs
     while (rc == GX_OK && *rows_out == 0 && ex->joinSpill &&

@@ -114,7 +114,12 @@ enum {
    * rows with ANY NULL element are excluded, aggregate.result's
    * count(distinct b,c,d) golden). NOT a value expression: valid only as
    * a COUNT_DISTINCT arg. */
-  GX_F_TUPLE = 41
+  GX_F_TUPLE = 41,
+  /* ROUND(x, d) (builtinRoundWithFracDecSig, builtin_math_vec.go:977):
+   * decimal round half-away-from-zero at scale min(d, ret_frac); const
+   * d >= 0 this round.
+   * ABS(x) (builtinAbsDecSig / builtinAbsIntSig): absolute value. */
+  GX_F_ROUND = 42, GX_F_ABS = 43
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

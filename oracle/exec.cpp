@@ -126,7 +126,24 @@ int32_t evalCast(EvalCtx& ctx, const Expr& e, const Chunk& in, Column& out) {
     if (a.type == GX_TYPE_DECIMAL) d = *a.getDecimal(i);
     else if (a.type == GX_TYPE_I64) d.FromInt(a.getI64(i));
     else { *ctx.err = "unsupported cast argument type"; return GX_ERR_INVALID; }
-    if (e.func == GX_F_CAST_DEC) {
+    if (e.func == GX_F_ROUND) {
+      // builtinRoundWithFracDecSig: Round(min(d, ret frac), ModeHalfUp)
+      const Expr& de = ctx.plan->exprs[e.args[1]];
+      MyDecimal r;
+      int32_t ec = d.Round(
+          &r, (int)std::min<int64_t>(de.constI64, e.retFrac), ModeHalfUp);
+      if (ec != E_OK && ec != E_TRUNCATED) return ec;
+      out.appendDecimal(r);
+    } else if (e.func == GX_F_ABS) {
+      // builtinAbsDecSig / builtinAbsIntSig
+      if (a.type == GX_TYPE_I64) {
+        int64_t v = a.getI64(i);
+        out.appendI64(v < 0 ? -v : v);
+      } else {
+        d.negative = 0;  // |x|: the sign bit IS the sign (mydecimal.go)
+        out.appendDecimal(d);
+      }
+    } else if (e.func == GX_F_CAST_DEC) {
       MyDecimal r;
       int32_t ec = d.Round(&r, e.retFrac, ModeHalfUp);
       if (ec != E_OK && ec != E_TRUNCATED) return ec;
@@ -322,7 +339,8 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       return GX_OK;
     case EK_CALL:
       if (e.func <= GX_F_NE) return evalCompare(ctx, e, in, out);
-      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT)
+      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT ||
+          e.func == GX_F_ROUND || e.func == GX_F_ABS)
         return evalCast(ctx, e, in, out);
       if ((e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER) ||
           e.func == GX_F_TRIM)

@@ -97,3 +97,95 @@ def test_cast_parity(to_int):
     from tests.gxlib import load_product
     assert run_cast(load_oracle(), 50000, to_int) == \
         run_cast(load_product(), 50000, to_int)
+
+
+def _run_round_abs(lib):
+    """ROUND(x, d) (builtinRoundWithFracDecSig: scale min(d, ret frac),
+    half away from zero) and ABS (builtinAbsDecSig/IntSig), standalone
+    projection + as a sum arg in the fused pipeline."""
+    import ctypes
+
+    import numpy as np
+
+    from tests.gxlib import (GX_AGG_SUM, GX_F_ABS, GX_F_ROUND,
+                             GX_TYPE_DECIMAL, GX_TYPE_I64)
+    from tidb_amd import plan as P
+    from tidb_amd.chunkpy import PyChunk
+
+    def dec(s):
+        out = (ctypes.c_uint8 * 40)()
+        assert lib.gx_dec_from_string(s.encode(), len(s.encode()), out) == 0
+        return bytes(out)
+
+    rng = np.random.default_rng(31)
+    rows = []
+    for i in range(3000):
+        sign = "-" if rng.random() < 0.5 else ""
+        rows.append((int(rng.integers(0, 9)),
+                     f"{sign}{int(rng.integers(0, 999))}.{int(rng.integers(0, 1000)):03d}",
+                     int(rng.integers(-(10 ** 6), 10 ** 6))))
+    chunks = []
+    types = [GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64]
+    for base in range(0, len(rows), 1000):
+        part = rows[base:base + 1000]
+        ch = PyChunk(types, len(part), [0, 3, 0])
+        for k, d, v in part:
+            ch.append_row([k, dec(d), v])
+        chunks.append(ch)
+
+    b = P.Builder(lib)
+    src = b.source(types, [0, 3, 0])
+    x = b.colref(1, GX_TYPE_DECIMAL, 3)
+    proj = b.projection(src, [
+        b.call(GX_F_ROUND, GX_TYPE_DECIMAL, 2, x, b.const_i64(2)),
+        b.call(GX_F_ROUND, GX_TYPE_DECIMAL, 3, x, b.const_i64(0)),
+        b.call(GX_F_ABS, GX_TYPE_DECIMAL, 3, x),
+        b.call(GX_F_ABS, GX_TYPE_I64, 0, b.colref(2, GX_TYPE_I64)),
+    ])
+    ex = b.build(proj)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out1 = ex.pull_all([GX_TYPE_DECIMAL, GX_TYPE_DECIMAL, GX_TYPE_DECIMAL,
+                        GX_TYPE_I64], [2, 3, 3, 0])
+    ex.close()
+    ex.free()
+
+    agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                    [(GX_AGG_SUM, b.call(GX_F_ABS, GX_TYPE_DECIMAL, 3, x),
+                      3)])
+    ex = b.build(agg)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out2 = sorted(ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 3]))
+    ex.close()
+    ex.free()
+    b.free()
+    return rows, out1, out2
+
+
+def test_oracle_round_abs():
+    from decimal import ROUND_HALF_UP, Decimal
+    lib = load_oracle()
+    rows, out1, out2 = _run_round_abs(lib)
+    for (r2, r0, ad, ai), (k, d, v) in zip(out1, rows):
+        D = Decimal(d)
+        # mydecimal ModeHalfUp = half AWAY from zero (python ROUND_HALF_UP)
+        assert Decimal(r2) == D.quantize(Decimal("0.01"), ROUND_HALF_UP)
+        # ROUND(x, 0) at ret frac 3 displays 3 decimals of the integer value
+        assert Decimal(r0) == D.quantize(Decimal("1"), ROUND_HALF_UP)
+        assert Decimal(ad) == abs(D)
+        assert ai == abs(v)
+    want = {}
+    for k, d, v in rows:
+        want[k] = want.get(k, Decimal(0)) + abs(Decimal(d))
+    assert out2 == sorted((k, str(s.quantize(Decimal("0.001"))))
+                          for k, s in want.items())
+
+
+@pytest.mark.gpu
+def test_round_abs_parity():
+    from tests.gxlib import load_product
+    _, w1, w2 = _run_round_abs(load_oracle())
+    _, g1, g2 = _run_round_abs(load_product())
+    assert g1 == w1
+    assert g2 == w2

@@ -57,6 +57,8 @@ enum VmOp : int32_t {
                        // (builtinLengthSig; offsets fetch slot in c)
   VM_IFNULL = 10,      // dst <- a if a not NULL else b (builtinIfNullSig;
                        // scales engine-aligned; COALESCE = chained IFNULL)
+  VM_ABS = 11,         // dst <- |a| (builtinAbs*Sig; narrow INT64_MIN
+                       // retries wide)
 };
 
 struct VmIns {

@@ -558,15 +558,26 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
         *scaleOut = 0;
         break;
       }
-      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT) {
-        if (e.args.size() != 1) {
+      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT ||
+          e.func == GX_F_ROUND) {
+        int sr = -1;
+        if (e.func == GX_F_ROUND) {
+          if (e.args.size() != 2 ||
+              ex->plan.exprs[e.args[1]].kind != EK_CONST ||
+              ex->plan.exprs[e.args[1]].constI64 < 0) {
+            ex->err = "ROUND takes (expr, const d >= 0) this round";
+            return -1;
+          }
+          sr = (int)std::min<int64_t>(ex->plan.exprs[e.args[1]].constI64,
+                                      e.retFrac);
+        } else if (e.args.size() != 1) {
           ex->err = "cast takes one argument";
           return -1;
         }
         int sa = 0;
         int ra = compileExpr(ex, e.args[0], &sa);
         if (ra < 0) return -1;
-        int sr = e.func == GX_F_CAST_INT ? 0 : e.retFrac;
+        if (sr < 0) sr = e.func == GX_F_CAST_INT ? 0 : e.retFrac;
         if (sr == sa) {
           reg = ra;  // no-op cast
           *scaleOut = sr;
@@ -576,6 +587,15 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
         if (reg >= 0) d.ins[d.nIns - 1].c = sa;
         release(e.args[0]);
         *scaleOut = sr;
+        break;
+      }
+      if (e.func == GX_F_ABS && e.args.size() == 1) {
+        int sa = 0;
+        int ra = compileExpr(ex, e.args[0], &sa);
+        if (ra < 0) return -1;
+        reg = emit(gxp::VM_ABS, allocReg(), ra, 0);
+        release(e.args[0]);
+        *scaleOut = sa;
         break;
       }
       if (e.args.size() != 2) {
@@ -780,17 +800,30 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
         *scaleOut = 0;
         break;
       }
-      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT) {
-        // cast family (ProduceDecWithSpecifiedTp / ToInt): round half-up
-        // from the arg's scale to the target scale (VM_ROUND_SCALE)
-        if (e.args.size() != 1) {
+      if (e.func == GX_F_CAST_DEC || e.func == GX_F_CAST_INT ||
+          e.func == GX_F_ROUND) {
+        // cast family (ProduceDecWithSpecifiedTp / ToInt) and ROUND(x, d)
+        // (builtinRoundWithFracDecSig: scale min(d, ret frac)): round
+        // half-up from the arg's scale to the target (VM_ROUND_SCALE)
+        int sr;
+        if (e.func == GX_F_ROUND) {
+          if (e.args.size() != 2 ||
+              ex->plan.exprs[e.args[1]].kind != EK_CONST ||
+              ex->plan.exprs[e.args[1]].constI64 < 0) {
+            ex->err = "ROUND takes (expr, const d >= 0) this round";
+            return -1;
+          }
+          sr = (int)std::min<int64_t>(ex->plan.exprs[e.args[1]].constI64,
+                                      e.retFrac);
+        } else if (e.args.size() != 1) {
           ex->err = "cast takes one argument";
           return -1;
+        } else {
+          sr = e.func == GX_F_CAST_INT ? 0 : e.retFrac;
         }
         int sa = 0;
         int ra = vmCompile(ex, B, e.args[0], &sa);
         if (ra < 0) return -1;
-        int sr = e.func == GX_F_CAST_INT ? 0 : e.retFrac;
         if (sr == sa) {
           reg = ra;  // no-op cast
           *scaleOut = sr;
@@ -798,6 +831,14 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
         }
         reg = emit(gxp::VM_ROUND_SCALE, allocReg(), ra, sr, sa);
         *scaleOut = sr;
+        break;
+      }
+      if (e.func == GX_F_ABS && e.args.size() == 1) {
+        int sa = 0;
+        int ra = vmCompile(ex, B, e.args[0], &sa);
+        if (ra < 0) return -1;
+        reg = emit(gxp::VM_ABS, allocReg(), ra, 0, -1);
+        *scaleOut = sa;
         break;
       }
       if (e.args.size() != 2) {

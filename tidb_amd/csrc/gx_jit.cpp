@@ -152,6 +152,12 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
           << " = *(T*)&c64; } }\n"
           << "  " << nv << " = false;\n";
         break;
+      case gxp::VM_ABS:
+        s << "  { T t2 = v" << ins.a << "; if (VT<WIDE>::cmp(t2, "
+             "VT<WIDE>::zero()) < 0) t2 = VT<WIDE>::sub(VT<WIDE>::zero(), "
+             "t2, &ovf); " << v << " = t2; " << nv << " = n" << ins.a
+          << "; }\n";
+        break;
       case gxp::VM_IFNULL:
         s << "  { bool t3 = n" << ins.a << "; " << v << " = t3 ? v" << ins.b
           << " : v" << ins.a << "; " << nv << " = t3 && n" << ins.b
@@ -645,6 +651,12 @@ static void emitJaVm(std::ostringstream& s, const JoinAggDesc& d) {
           << "      else { int64_t c64 = " << d.constLo[ins.a] << "LL; " << v
           << " = *(T*)&c64; } }\n"
           << "    const bool " << nv << " = false;\n";
+        break;
+      case gxp::VM_ABS:
+        s << "    T " << v << " = v" << ins.a << "; if (VT<WIDE>::cmp(" << v
+          << ", VT<WIDE>::zero()) < 0) " << v << " = VT<WIDE>::sub("
+             "VT<WIDE>::zero(), " << v << ", &ovf); bool " << nv << " = n"
+          << ins.a << ";\n";
         break;
       case gxp::VM_IFNULL:
         s << "    T " << v << " = n" << ins.a << " ? v" << ins.b << " : v"

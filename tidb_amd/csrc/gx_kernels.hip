@@ -242,6 +242,15 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
         break;
+      case VM_ABS: {
+        // builtinAbs*Sig; narrow INT64_MIN -> wide retry via mul overflow
+        typename VT<WIDE>::T av = vm.get(ins.a);
+        if (VT<WIDE>::cmp(av, VT<WIDE>::zero()) < 0)
+          av = VT<WIDE>::sub(VT<WIDE>::zero(), av, &ovf);
+        vm.set(ins.dst, av);
+        vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+      }
       case VM_IFNULL: {
         // builtinIfNullSig: first non-NULL operand; NULL only if both are
         bool an = vm.isNull(ins.a);
@@ -1018,6 +1027,14 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
           vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
+        case VM_ABS: {
+          typename VT<WIDE>::T av = vm.get(ins.a);
+          if (VT<WIDE>::cmp(av, VT<WIDE>::zero()) < 0)
+            av = VT<WIDE>::sub(VT<WIDE>::zero(), av, &ovf);
+          vm.set(ins.dst, av);
+          vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        }
         case VM_IFNULL: {
           bool an = vm.isNull(ins.a);
           vm.set(ins.dst, an ? vm.get(ins.b) : vm.get(ins.a));
@@ -1435,6 +1452,15 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
       case VM_SUB:
         vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
         break;
+      case VM_ABS: {
+        // builtinAbs*Sig; narrow INT64_MIN -> wide retry via mul overflow
+        typename VT<WIDE>::T av = vm.get(ins.a);
+        if (VT<WIDE>::cmp(av, VT<WIDE>::zero()) < 0)
+          av = VT<WIDE>::sub(VT<WIDE>::zero(), av, &ovf);
+        vm.set(ins.dst, av);
+        vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+      }
       case VM_IFNULL: {
         // builtinIfNullSig: first non-NULL operand; NULL only if both are
         bool an = vm.isNull(ins.a);
@@ -2813,6 +2839,14 @@ __global__ void projectKernel(const ProjDesc* __restrict__ dp) {
           vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
+        case VM_ABS: {
+          typename VT<WIDE>::T av = vm.get(ins.a);
+          if (VT<WIDE>::cmp(av, VT<WIDE>::zero()) < 0)
+            av = VT<WIDE>::sub(VT<WIDE>::zero(), av, &ovf);
+          vm.set(ins.dst, av);
+          vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        }
         case VM_IFNULL: {
           bool an = vm.isNull(ins.a);
           vm.set(ins.dst, an ? vm.get(ins.b) : vm.get(ins.a));

@@ -119,7 +119,12 @@ enum {
    * decimal round half-away-from-zero at scale min(d, ret_frac); const
    * d >= 0 this round.
    * ABS(x) (builtinAbsDecSig / builtinAbsIntSig): absolute value. */
-  GX_F_ROUND = 42, GX_F_ABS = 43
+  GX_F_ROUND = 42, GX_F_ABS = 43,
+  /* YEAR/MONTH/DAY(t) (builtinYearSig / builtinMonthSig / builtinDaySig,
+   * builtin_time_vec.go): CoreTime bitfield extraction (year@50:14,
+   * month@46:4, day@41:5 — core_time.go); arg must be a TIME column;
+   * NULL propagates. */
+  GX_F_YEAR = 44, GX_F_MONTH = 45, GX_F_DAY = 46
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

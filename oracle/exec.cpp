@@ -345,6 +345,23 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       if ((e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER) ||
           e.func == GX_F_TRIM)
         return evalString(ctx, e, in, out);
+      if (e.func >= GX_F_YEAR && e.func <= GX_F_DAY) {
+        // builtinYear/Month/DaySig: CoreTime bitfield extraction
+        Column a;
+        int32_t err = evalVec(ctx, e.args[0], in, a);
+        if (err) return err;
+        out.reset();
+        out.type = GX_TYPE_I64;
+        for (int i = 0; i < in.numRows(); i++) {
+          if (a.isNull(i)) { out.appendNull(); continue; }
+          uint64_t bits = a.getU64(i);
+          int64_t f = e.func == GX_F_YEAR ? (int64_t)((bits >> 50) & 0x3FFF)
+                    : e.func == GX_F_MONTH ? (int64_t)((bits >> 46) & 0xF)
+                                           : (int64_t)((bits >> 41) & 0x1F);
+          out.appendI64(f);
+        }
+        return GX_OK;
+      }
       if (e.func == GX_F_TUPLE) {
         if (ctx.err) *ctx.err = "tuple is only valid as a DISTINCT aggregate argument";
         return GX_ERR_INVALID;

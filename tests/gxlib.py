@@ -31,6 +31,7 @@ GX_F_TRIM = 39
 GX_F_IFNULL = 40
 GX_F_TUPLE = 41
 GX_F_ROUND, GX_F_ABS = 42, 43
+GX_F_YEAR, GX_F_MONTH, GX_F_DAY = 44, 45, 46
 
 GX_AGG_COUNT, GX_AGG_SUM, GX_AGG_AVG, GX_AGG_MIN, GX_AGG_MAX, GX_AGG_FIRSTROW = range(6)
 GX_AGG_MODE_COMPLETE, GX_AGG_MODE_PARTIAL, GX_AGG_MODE_FINAL = 0, 1, 2

@@ -59,6 +59,8 @@ enum VmOp : int32_t {
                        // scales engine-aligned; COALESCE = chained IFNULL)
   VM_ABS = 11,         // dst <- |a| (builtinAbs*Sig; narrow INT64_MIN
                        // retries wide)
+  VM_TIME_EXTRACT = 12,  // dst <- CoreTime field b of a (0 year@50:14,
+                         // 1 month@46:4, 2 day@41:5 — core_time.go)
 };
 
 struct VmIns {

@@ -598,6 +598,23 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
         *scaleOut = sa;
         break;
       }
+      if (e.func >= GX_F_YEAR && e.func <= GX_F_DAY) {
+        // YEAR/MONTH/DAY: raw CoreTime bits load (nulls tracked by the
+        // load) + bitfield extract; only direct TIME columns this round
+        const PExpr* a0 =
+            e.args.size() == 1 ? &ex->plan.exprs[e.args[0]] : nullptr;
+        if (!a0 || a0->kind != EK_COLREF || a0->colIdx < 0 ||
+            a0->colIdx >= ex->desc.table.nCols ||
+            ex->desc.table.cols[a0->colIdx].type != GX_TYPE_TIME) {
+          ex->err = "YEAR/MONTH/DAY take a time column";
+          return -1;
+        }
+        int lr = emit(gxp::VM_LOAD_I64, allocRegFresh(), a0->colIdx, 0);
+        if (lr < 0) break;
+        reg = emit(gxp::VM_TIME_EXTRACT, allocReg(), lr, e.func - GX_F_YEAR);
+        *scaleOut = 0;
+        break;
+      }
       if (e.args.size() != 2) {
         ex->err = "unsupported call arity on device";
         return -1;
@@ -839,6 +856,24 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
         if (ra < 0) return -1;
         reg = emit(gxp::VM_ABS, allocReg(), ra, 0, -1);
         *scaleOut = sa;
+        break;
+      }
+      if (e.func >= GX_F_YEAR && e.func <= GX_F_DAY) {
+        const PExpr* a0 =
+            e.args.size() == 1 ? &ex->plan.exprs[e.args[0]] : nullptr;
+        int col = a0 && a0->kind == EK_COLREF ? a0->colIdx - B.colBase : -1;
+        if (col < 0 || col >= (int)B.colTypes->size() ||
+            (*B.colTypes)[col] != GX_TYPE_TIME) {
+          ex->err = "YEAR/MONTH/DAY take a time column";
+          return -1;
+        }
+        int slot = vmFetchSlot(B, gxp::FETCH_8B, col);
+        if (slot < 0) { ex->err = "fetch plan full"; return -1; }
+        int lr = emit(gxp::VM_LOAD_I64, allocReg(), col, 0, slot);
+        if (lr < 0) break;
+        reg = emit(gxp::VM_TIME_EXTRACT, allocReg(), lr, e.func - GX_F_YEAR,
+                   -1);
+        *scaleOut = 0;
         break;
       }
       if (e.args.size() != 2) {

@@ -152,6 +152,14 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
           << " = *(T*)&c64; } }\n"
           << "  " << nv << " = false;\n";
         break;
+      case gxp::VM_TIME_EXTRACT: {
+        int sh = ins.b == 0 ? 50 : (ins.b == 1 ? 46 : 41);
+        uint64_t mask = ins.b == 0 ? 0x3FFF : (ins.b == 1 ? 0xF : 0x1F);
+        s << "  { uint64_t bits = VT<WIDE>::toAcc(v" << ins.a << ").lo; "
+          << v << " = VT<WIDE>::fromI64((int64_t)((bits >> " << sh << ") & "
+          << mask << "ULL), &ovf); " << nv << " = n" << ins.a << "; }\n";
+        break;
+      }
       case gxp::VM_ABS:
         s << "  { T t2 = v" << ins.a << "; if (VT<WIDE>::cmp(t2, "
              "VT<WIDE>::zero()) < 0) t2 = VT<WIDE>::sub(VT<WIDE>::zero(), "
@@ -652,6 +660,14 @@ static void emitJaVm(std::ostringstream& s, const JoinAggDesc& d) {
           << " = *(T*)&c64; } }\n"
           << "    const bool " << nv << " = false;\n";
         break;
+      case gxp::VM_TIME_EXTRACT: {
+        int sh = ins.b == 0 ? 50 : (ins.b == 1 ? 46 : 41);
+        uint64_t mask = ins.b == 0 ? 0x3FFF : (ins.b == 1 ? 0xF : 0x1F);
+        s << "    T " << v << " = VT<WIDE>::fromI64((int64_t)((VT<WIDE>::toAcc(v"
+          << ins.a << ").lo >> " << sh << ") & " << mask
+          << "ULL), &ovf); bool " << nv << " = n" << ins.a << ";\n";
+        break;
+      }
       case gxp::VM_ABS:
         s << "    T " << v << " = v" << ins.a << "; if (VT<WIDE>::cmp(" << v
           << ", VT<WIDE>::zero()) < 0) " << v << " = VT<WIDE>::sub("

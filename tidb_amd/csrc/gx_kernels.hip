@@ -242,6 +242,16 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
         break;
+      case VM_TIME_EXTRACT: {
+        // CoreTime bitfield (core_time.go): year@50:14 month@46:4 day@41:5
+        uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
+        int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
+                  : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
+                               : (int64_t)((bits >> 41) & 0x1F);
+        vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
+        vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+      }
       case VM_ABS: {
         // builtinAbs*Sig; narrow INT64_MIN -> wide retry via mul overflow
         typename VT<WIDE>::T av = vm.get(ins.a);
@@ -1027,6 +1037,15 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
           vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
+        case VM_TIME_EXTRACT: {
+          uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
+          int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
+                    : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
+                                 : (int64_t)((bits >> 41) & 0x1F);
+          vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        }
         case VM_ABS: {
           typename VT<WIDE>::T av = vm.get(ins.a);
           if (VT<WIDE>::cmp(av, VT<WIDE>::zero()) < 0)
@@ -1452,6 +1471,16 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
       case VM_SUB:
         vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
         break;
+      case VM_TIME_EXTRACT: {
+        // CoreTime bitfield (core_time.go): year@50:14 month@46:4 day@41:5
+        uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
+        int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
+                  : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
+                               : (int64_t)((bits >> 41) & 0x1F);
+        vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
+        vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+      }
       case VM_ABS: {
         // builtinAbs*Sig; narrow INT64_MIN -> wide retry via mul overflow
         typename VT<WIDE>::T av = vm.get(ins.a);
@@ -2839,6 +2868,15 @@ __global__ void projectKernel(const ProjDesc* __restrict__ dp) {
           vm.set(ins.dst, VT<WIDE>::sub(vm.get(ins.a), vm.get(ins.b), &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
+        case VM_TIME_EXTRACT: {
+          uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
+          int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
+                    : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
+                                 : (int64_t)((bits >> 41) & 0x1F);
+          vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        }
         case VM_ABS: {
           typename VT<WIDE>::T av = vm.get(ins.a);
           if (VT<WIDE>::cmp(av, VT<WIDE>::zero()) < 0)

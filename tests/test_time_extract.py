@@ -103,3 +103,46 @@ def test_time_extract_parity():
     _, w2 = _run_agg(load_oracle())
     _, g2 = _run_agg(load_product())
     assert g2 == w2
+
+
+def _run_groupby_year(lib):
+    """GROUP BY YEAR(t), MONTH(t) — computed i64 group keys ride the wide
+    serialized-key path via stable VM registers (kind 4)."""
+    from tests.gxlib import GX_AGG_COUNT
+    rows = _data(lib, n=4000, seed=47)
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64, GX_TYPE_TIME])
+    t = b.colref(1, GX_TYPE_TIME)
+    y = b.call(GX_F_YEAR, GX_TYPE_I64, 0, t)
+    m = b.call(GX_F_MONTH, GX_TYPE_I64, 0, t)
+    proj = b.projection(src, [y, m, b.colref(0, GX_TYPE_I64)])
+    agg = b.hashagg(proj, [b.colref(0, GX_TYPE_I64),
+                           b.colref(1, GX_TYPE_I64)],
+                    [(GX_AGG_COUNT, b.colref(2, GX_TYPE_I64), 0)])
+    ex = b.build(agg)
+    ex.bind_chunks(src, _chunks(rows))
+    ex.open()
+    got = ex.pull_all([GX_TYPE_I64] * 3)
+    ex.close()
+    ex.free()
+    b.free()
+    want = {}
+    for k, ymd, _ in rows:
+        key = (None, None) if ymd is None else (ymd[0], ymd[1])
+        want[key] = want.get(key, 0) + 1
+    wl = sorted(((y, m, c) for (y, m), c in want.items()),
+                key=lambda r: tuple((x is None, x) for x in r))
+    gl = sorted(got, key=lambda r: tuple((x is None, x) for x in r))
+    assert gl == wl, (gl[:4], wl[:4])
+    return gl
+
+
+def test_oracle_groupby_computed_year():
+    _run_groupby_year(load_oracle())
+
+
+@pytest.mark.gpu
+def test_groupby_computed_year_parity():
+    want = _run_groupby_year(load_oracle())
+    got = _run_groupby_year(load_product())
+    assert got == want

@@ -380,7 +380,8 @@ __device__ __attribute__((always_inline)) inline bool makeGroupKey(const FusedQu
                                     uint32_t* err) {
   uint64_t key = 0;
   for (int k = 0; k < d.gkey.nCols; k++) {
-    const DevCol& c = d.table.cols[d.gkey.col[k]];
+    // computed keys (col < 0) never touch the column; clamp the index
+    const DevCol& c = d.table.cols[d.gkey.col[k] < 0 ? 0 : d.gkey.col[k]];
     uint32_t lane;
     if (colIsNull(c, row)) {
       lane = 0xFF000000u;
@@ -452,7 +453,8 @@ __device__ inline uint64_t wideKeyHash(const FusedQueryDesc& d, int64_t row,
   uint64_t h = 0x243F6A8885A308D3ULL;
   uint64_t nb = 0;
   for (int k = 0; k < d.gkey.nCols; k++) {
-    const DevCol& c = d.table.cols[d.gkey.col[k]];
+    // computed keys (col < 0) never touch the column; clamp the index
+    const DevCol& c = d.table.cols[d.gkey.col[k] < 0 ? 0 : d.gkey.col[k]];
     int kind = d.gkey.kind[k];
     bool nul = kind == 4 ? vm.isNull(d.gkey.slot[k]) : colIsNull(c, row);
     if (nul) {
@@ -503,7 +505,7 @@ __device__ inline void wideKeyWrite(const FusedQueryDesc& d, uint8_t* rec,
       f[0] = u.lo;
       f[1] = (uint64_t)u.hi;
     } else {
-      const DevCol& c = d.table.cols[d.gkey.col[k]];
+      const DevCol& c = d.table.cols[d.gkey.col[k] < 0 ? 0 : d.gkey.col[k]];
       int64_t s;
       wideStrMeta(c, raw.get(d.gkey.slot[k]), &s, &f[0], &f[1], &f[2]);
     }
@@ -527,7 +529,7 @@ __device__ inline bool wideKeyMatches(const FusedQueryDesc& d,
       Int128 u = VT<WIDE>::toAcc(vm.get(d.gkey.slot[k]));
       if (f[0] != u.lo || f[1] != (uint64_t)u.hi) return false;
     } else {
-      const DevCol& c = d.table.cols[d.gkey.col[k]];
+      const DevCol& c = d.table.cols[d.gkey.col[k] < 0 ? 0 : d.gkey.col[k]];
       int64_t s;
       uint64_t len, w0, w1;
       wideStrMeta(c, raw.get(d.gkey.slot[k]), &s, &len, &w0, &w1);

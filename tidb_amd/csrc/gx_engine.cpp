@@ -2125,22 +2125,48 @@ static int32_t compileFused(gx_exec* ex) {
   for (int ge : agg->exprs) {
     const PExpr& e = plan.exprs[ge];
     int srcCol = -1;
+    bool computed = false;  // projection-computed key (e.g. YEAR(t))
     if (e.kind != EK_COLREF) {
       ex->err = "group-by expression must be a column";
       return GX_ERR_INVALID;
     }
     if (proj) {
-      if (e.colIdx >= (int)projSrcCol.size() || projSrcCol[e.colIdx] < 0) {
-        ex->err = "group-by must reference a passthrough column";
+      if (e.colIdx >= (int)projSrcCol.size()) {
+        ex->err = "group-by column out of range";
         return GX_ERR_INVALID;
       }
-      srcCol = projSrcCol[e.colIdx];
+      if (projSrcCol[e.colIdx] >= 0) srcCol = projSrcCol[e.colIdx];
+      else computed = true;
     } else {
       srcCol = e.colIdx;
     }
     if (gk.nCols >= gxp::kMaxGroupKeyCols) {
       ex->err = "too many group-by key columns";
       return GX_ERR_INVALID;
+    }
+    if (computed) {
+      // computed i64/decimal group key (GROUP BY YEAR(t), a+b, ...): the
+      // value lives in a stable VM register; wide keys hash/record it at
+      // its compile-time scale (kind 4)
+      const PExpr& pe = plan.exprs[proj->exprs[e.colIdx]];
+      if (pe.retType != GX_TYPE_I64 && pe.retType != GX_TYPE_DECIMAL) {
+        ex->err = "computed group keys are i64/decimal this round";
+        return GX_ERR_INVALID;
+      }
+      forceWideKeys = true;
+      int sc = 0;
+      int reg = compileExpr(ex, proj->exprs[e.colIdx], &sc);
+      if (reg < 0) {
+        if (ex->err.empty()) ex->err = "group key compile failed";
+        return GX_ERR_INVALID;
+      }
+      // col < 0 marks a computed key; -2 = i64 result, -1 = decimal
+      gk.col[gk.nCols] = pe.retType == GX_TYPE_I64 ? -2 : -1;
+      gk.kind[gk.nCols] = 4;
+      gk.slot[gk.nCols] = reg;
+      gk.kscale[gk.nCols] = sc;
+      gk.nCols++;
+      continue;
     }
     int t = src->colTypes[srcCol];
     gk.col[gk.nCols] = srcCol;
@@ -3277,7 +3303,8 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
       const uint8_t* rec = recs.data() + (size_t)recIdx * g.recBytes;
       const uint64_t* r64 = (const uint64_t*)rec;
       int srcCol = g.col[k];
-      v->type = ex->desc.table.cols[srcCol].type;
+      if (srcCol >= 0) v->type = ex->desc.table.cols[srcCol].type;
+      else v->type = srcCol == -2 ? GX_TYPE_I64 : GX_TYPE_DECIMAL;
       if ((r64[0] >> k) & 1) {
         v->isNull = true;
         canon->push_back('\x01');
@@ -3291,8 +3318,14 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
         canon->append((const char*)&f[0], 8);
       } else if (g.kind[k] == 4) {
         __int128 units = ((__int128)(int64_t)f[1] << 64) | f[0];
-        v->type = GX_TYPE_DECIMAL;
-        v->dec = decFromUnits(units, g.kscale[k]);
+        if (g.col[k] == -2) {  // computed i64 key (e.g. YEAR(t))
+          v->type = GX_TYPE_I64;
+          v->i64 = (int64_t)units;
+          v->u64 = (uint64_t)v->i64;
+        } else {
+          v->type = GX_TYPE_DECIMAL;
+          v->dec = decFromUnits(units, g.kscale[k]);
+        }
         canon->append((const char*)&f[0], 16);
       } else {  // kind 5: trimmed string (prefix inline, tail via owner row)
         uint64_t len = f[0];

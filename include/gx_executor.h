@@ -124,7 +124,12 @@ enum {
    * builtin_time_vec.go): CoreTime bitfield extraction (year@50:14,
    * month@46:4, day@41:5 — core_time.go); arg must be a TIME column;
    * NULL propagates. */
-  GX_F_YEAR = 44, GX_F_MONTH = 45, GX_F_DAY = 46
+  GX_F_YEAR = 44, GX_F_MONTH = 45, GX_F_DAY = 46,
+  GX_F_HOUR = 47, GX_F_MINUTE = 48, GX_F_SECOND = 49,
+  /* GREATEST/LEAST (builtinGreatest*Sig / builtinLeast*Sig,
+   * builtin_compare_vec.go:27: MergeNulls => NULL if ANY arg is NULL).
+   * Two args on the device path; n-ary composes as a chain. */
+  GX_F_GREATEST = 50, GX_F_LEAST = 51
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

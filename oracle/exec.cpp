@@ -345,21 +345,50 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
       if ((e.func >= GX_F_LENGTH && e.func <= GX_F_LOWER) ||
           e.func == GX_F_TRIM)
         return evalString(ctx, e, in, out);
-      if (e.func >= GX_F_YEAR && e.func <= GX_F_DAY) {
-        // builtinYear/Month/DaySig: CoreTime bitfield extraction
+      if (e.func >= GX_F_YEAR && e.func <= GX_F_SECOND) {
+        // builtinYear/Month/Day/Hour/Minute/SecondSig: CoreTime bitfields
+        static const int kSh[6] = {50, 46, 41, 36, 30, 24};
+        static const uint64_t kMk[6] = {0x3FFF, 0xF, 0x1F, 0x1F, 0x3F, 0x3F};
         Column a;
         int32_t err = evalVec(ctx, e.args[0], in, a);
         if (err) return err;
         out.reset();
         out.type = GX_TYPE_I64;
+        int fi = e.func - GX_F_YEAR;
         for (int i = 0; i < in.numRows(); i++) {
           if (a.isNull(i)) { out.appendNull(); continue; }
-          uint64_t bits = a.getU64(i);
-          int64_t f = e.func == GX_F_YEAR ? (int64_t)((bits >> 50) & 0x3FFF)
-                    : e.func == GX_F_MONTH ? (int64_t)((bits >> 46) & 0xF)
-                                           : (int64_t)((bits >> 41) & 0x1F);
-          out.appendI64(f);
+          out.appendI64((int64_t)((a.getU64(i) >> kSh[fi]) & kMk[fi]));
         }
+        return GX_OK;
+      }
+      if (e.func == GX_F_GREATEST || e.func == GX_F_LEAST) {
+        // builtinGreatest/Least*Sig (n-ary; NULL if ANY arg is NULL)
+        Column acc;
+        int32_t err = evalVec(ctx, e.args[0], in, acc);
+        if (err) return err;
+        for (size_t j = 1; j < e.args.size(); j++) {
+          Column b2, merged;
+          err = evalVec(ctx, e.args[j], in, b2);
+          if (err) return err;
+          merged.type = acc.type;
+          merged.frac = std::max(acc.frac, b2.frac);
+          merged.reset();
+          if (merged.isVarlen()) merged.offsets.assign(1, 0);
+          for (int i = 0; i < in.numRows(); i++) {
+            if (acc.isNull(i) || b2.isNull(i)) { merged.appendNull(); continue; }
+            int c;
+            if (acc.type == GX_TYPE_DECIMAL)
+              c = acc.getDecimal(i)->Compare(*b2.getDecimal(i));
+            else {
+              int64_t x = acc.getI64(i), y = b2.getI64(i);
+              c = x < y ? -1 : (x > y ? 1 : 0);
+            }
+            bool takeA = (e.func == GX_F_GREATEST) == (c >= 0);
+            merged.appendFrom(takeA ? acc : b2, i);
+          }
+          acc = std::move(merged);
+        }
+        out = std::move(acc);
         return GX_OK;
       }
       if (e.func == GX_F_TUPLE) {

@@ -146,3 +146,81 @@ def test_groupby_computed_year_parity():
     want = _run_groupby_year(load_oracle())
     got = _run_groupby_year(load_product())
     assert got == want
+
+
+def _run_hms_greatest(lib):
+    """HOUR/MINUTE/SECOND over datetimes + GREATEST/LEAST over decimals
+    and ints (NULL if any arg NULL, builtin_compare_vec.go MergeNulls)."""
+    import ctypes
+
+    from tests.gxlib import (GX_F_GREATEST, GX_F_HOUR, GX_F_LEAST,
+                             GX_F_MINUTE, GX_F_SECOND, GX_TYPE_DECIMAL)
+    rng = np.random.default_rng(53)
+    lib.gx_time_from_datetime.restype = __import__("ctypes").c_uint64
+
+    def dec(s):
+        out = (ctypes.c_uint8 * 40)()
+        assert lib.gx_dec_from_string(s.encode(), len(s.encode()), out) == 0
+        return bytes(out)
+
+    rows = []
+    for i in range(2000):
+        if rng.random() < 0.15:
+            rows.append((None, None, None, None))
+        else:
+            hms = (int(rng.integers(0, 24)), int(rng.integers(0, 60)),
+                   int(rng.integers(0, 60)))
+            t = lib.gx_time_from_datetime(1995, 6, 7, *hms, 0, 0)
+            a = f"{int(rng.integers(-50, 50))}.{int(rng.integers(0, 100)):02d}"
+            bb = f"{int(rng.integers(-50, 50))}.{int(rng.integers(0, 100)):02d}"
+            rows.append((hms, t, a, bb))
+    chunks = []
+    types = [GX_TYPE_TIME, GX_TYPE_DECIMAL, GX_TYPE_DECIMAL]
+    for base in range(0, len(rows), 1000):
+        part = rows[base:base + 1000]
+        ch = PyChunk(types, len(part), [0, 2, 2])
+        for _, t, a, bb in part:
+            ch.append_row([t, None if a is None else dec(a),
+                           None if bb is None else dec(bb)])
+        chunks.append(ch)
+    b = P.Builder(lib)
+    src = b.source(types, [0, 2, 2])
+    t = b.colref(0, GX_TYPE_TIME)
+    da = b.colref(1, GX_TYPE_DECIMAL, 2)
+    db = b.colref(2, GX_TYPE_DECIMAL, 2)
+    proj = b.projection(src, [
+        b.call(GX_F_HOUR, GX_TYPE_I64, 0, t),
+        b.call(GX_F_MINUTE, GX_TYPE_I64, 0, t),
+        b.call(GX_F_SECOND, GX_TYPE_I64, 0, t),
+        b.call(GX_F_GREATEST, GX_TYPE_DECIMAL, 2, da, db),
+        b.call(GX_F_LEAST, GX_TYPE_DECIMAL, 2, da, db),
+    ])
+    ex = b.build(proj)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    got = ex.pull_all([GX_TYPE_I64] * 3 + [GX_TYPE_DECIMAL] * 2,
+                      [0, 0, 0, 2, 2])
+    ex.close()
+    ex.free()
+    b.free()
+    return rows, got
+
+
+def test_oracle_hms_greatest():
+    from decimal import Decimal
+    lib = load_oracle()
+    rows, got = _run_hms_greatest(lib)
+    for (h, m, s, g, l), (hms, _, a, bb) in zip(got, rows):
+        if hms is None:
+            assert (h, m, s, g, l) == (None,) * 5
+            continue
+        assert (h, m, s) == hms
+        assert Decimal(g) == max(Decimal(a), Decimal(bb))
+        assert Decimal(l) == min(Decimal(a), Decimal(bb))
+
+
+@pytest.mark.gpu
+def test_hms_greatest_parity():
+    _, want = _run_hms_greatest(load_oracle())
+    _, got = _run_hms_greatest(load_product())
+    assert got == want

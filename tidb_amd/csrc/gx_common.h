@@ -60,7 +60,11 @@ enum VmOp : int32_t {
   VM_ABS = 11,         // dst <- |a| (builtinAbs*Sig; narrow INT64_MIN
                        // retries wide)
   VM_TIME_EXTRACT = 12,  // dst <- CoreTime field b of a (0 year@50:14,
-                         // 1 month@46:4, 2 day@41:5 — core_time.go)
+                         // 1 month@46:4, 2 day@41:5, 3 hour@36:5,
+                         // 4 minute@30:6, 5 second@24:6 — core_time.go)
+  VM_MAX2 = 13,          // dst <- max(a, b) (builtinGreatest*Sig; NULL if
+                         // either is NULL; scales engine-aligned)
+  VM_MIN2 = 14,          // dst <- min(a, b) (builtinLeast*Sig)
 };
 
 struct VmIns {

@@ -598,7 +598,7 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
         *scaleOut = sa;
         break;
       }
-      if (e.func >= GX_F_YEAR && e.func <= GX_F_DAY) {
+      if (e.func >= GX_F_YEAR && e.func <= GX_F_SECOND) {
         // YEAR/MONTH/DAY: raw CoreTime bits load (nulls tracked by the
         // load) + bitfield extract; only direct TIME columns this round
         const PExpr* a0 =
@@ -624,14 +624,20 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       if (ra < 0) return -1;
       int rb = compileExpr(ex, e.args[1], &sb);
       if (rb < 0) return -1;
-      if (e.func == GX_F_IFNULL) {
-        // builtinIfNullSig: align operand scales, result keeps the max
+      if (e.func == GX_F_IFNULL || e.func == GX_F_GREATEST ||
+          e.func == GX_F_LEAST) {
+        // builtinIfNullSig / builtinGreatest*Sig / builtinLeast*Sig:
+        // align operand scales, result keeps the max scale
+        int op2 = e.func == GX_F_IFNULL
+                      ? gxp::VM_IFNULL
+                      : (e.func == GX_F_GREATEST ? gxp::VM_MAX2
+                                                 : gxp::VM_MIN2);
         int target = std::max(sa, sb);
         int tmpA = -1, tmpB = -1;
         if (sa < target) tmpA = ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa);
         if (sb < target) tmpB = rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb);
         if (ra < 0 || rb < 0) break;
-        reg = emit(gxp::VM_IFNULL, allocReg(), ra, rb);
+        reg = emit(op2, allocReg(), ra, rb);
         if (tmpA >= 0) ex->vmFreeRegs.push_back(tmpA);
         if (tmpB >= 0) ex->vmFreeRegs.push_back(tmpB);
         *scaleOut = target;
@@ -858,7 +864,7 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
         *scaleOut = sa;
         break;
       }
-      if (e.func >= GX_F_YEAR && e.func <= GX_F_DAY) {
+      if (e.func >= GX_F_YEAR && e.func <= GX_F_SECOND) {
         const PExpr* a0 =
             e.args.size() == 1 ? &ex->plan.exprs[e.args[0]] : nullptr;
         int col = a0 && a0->kind == EK_COLREF ? a0->colIdx - B.colBase : -1;
@@ -885,13 +891,17 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
       if (ra < 0) return -1;
       int rb = vmCompile(ex, B, e.args[1], &sb);
       if (rb < 0) return -1;
-      if (e.func == GX_F_IFNULL) {
-        // builtinIfNullSig: operands align to one scale, result keeps it
+      if (e.func == GX_F_IFNULL || e.func == GX_F_GREATEST ||
+          e.func == GX_F_LEAST) {
+        int op2 = e.func == GX_F_IFNULL
+                      ? gxp::VM_IFNULL
+                      : (e.func == GX_F_GREATEST ? gxp::VM_MAX2
+                                                 : gxp::VM_MIN2);
         int target = std::max(sa, sb);
         if (sa < target) ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa, -1);
         if (sb < target) rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb, -1);
         if (ra < 0 || rb < 0) break;
-        reg = emit(gxp::VM_IFNULL, allocReg(), ra, rb, -1);
+        reg = emit(op2, allocReg(), ra, rb, -1);
         *scaleOut = target;
         break;
       }

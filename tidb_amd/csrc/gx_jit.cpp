@@ -153,13 +153,22 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
           << "  " << nv << " = false;\n";
         break;
       case gxp::VM_TIME_EXTRACT: {
-        int sh = ins.b == 0 ? 50 : (ins.b == 1 ? 46 : 41);
-        uint64_t mask = ins.b == 0 ? 0x3FFF : (ins.b == 1 ? 0xF : 0x1F);
+        static const int kSh[6] = {50, 46, 41, 36, 30, 24};
+        static const uint64_t kMk[6] = {0x3FFF, 0xF, 0x1F, 0x1F, 0x3F, 0x3F};
         s << "  { uint64_t bits = VT<WIDE>::toAcc(v" << ins.a << ").lo; "
-          << v << " = VT<WIDE>::fromI64((int64_t)((bits >> " << sh << ") & "
-          << mask << "ULL), &ovf); " << nv << " = n" << ins.a << "; }\n";
+          << v << " = VT<WIDE>::fromI64((int64_t)((bits >> " << kSh[ins.b]
+          << ") & " << kMk[ins.b] << "ULL), &ovf); " << nv << " = n" << ins.a
+          << "; }\n";
         break;
       }
+      case gxp::VM_MAX2:
+      case gxp::VM_MIN2:
+        s << "  { int c = VT<WIDE>::cmp(v" << ins.a << ", v" << ins.b
+          << "); " << v << " = " << (ins.op == gxp::VM_MAX2 ? "c >= 0"
+                                                            : "c <= 0")
+          << " ? v" << ins.a << " : v" << ins.b << "; " << nv << " = n"
+          << ins.a << " || n" << ins.b << "; }\n";
+        break;
       case gxp::VM_ABS:
         s << "  { T t2 = v" << ins.a << "; if (VT<WIDE>::cmp(t2, "
              "VT<WIDE>::zero()) < 0) t2 = VT<WIDE>::sub(VT<WIDE>::zero(), "
@@ -661,13 +670,20 @@ static void emitJaVm(std::ostringstream& s, const JoinAggDesc& d) {
           << "    const bool " << nv << " = false;\n";
         break;
       case gxp::VM_TIME_EXTRACT: {
-        int sh = ins.b == 0 ? 50 : (ins.b == 1 ? 46 : 41);
-        uint64_t mask = ins.b == 0 ? 0x3FFF : (ins.b == 1 ? 0xF : 0x1F);
+        static const int kSh[6] = {50, 46, 41, 36, 30, 24};
+        static const uint64_t kMk[6] = {0x3FFF, 0xF, 0x1F, 0x1F, 0x3F, 0x3F};
         s << "    T " << v << " = VT<WIDE>::fromI64((int64_t)((VT<WIDE>::toAcc(v"
-          << ins.a << ").lo >> " << sh << ") & " << mask
+          << ins.a << ").lo >> " << kSh[ins.b] << ") & " << kMk[ins.b]
           << "ULL), &ovf); bool " << nv << " = n" << ins.a << ";\n";
         break;
       }
+      case gxp::VM_MAX2:
+      case gxp::VM_MIN2:
+        s << "    T " << v << " = VT<WIDE>::cmp(v" << ins.a << ", v" << ins.b
+          << ") " << (ins.op == gxp::VM_MAX2 ? ">= 0" : "<= 0") << " ? v"
+          << ins.a << " : v" << ins.b << "; bool " << nv << " = n" << ins.a
+          << " || n" << ins.b << ";\n";
+        break;
       case gxp::VM_ABS:
         s << "    T " << v << " = v" << ins.a << "; if (VT<WIDE>::cmp(" << v
           << ", VT<WIDE>::zero()) < 0) " << v << " = VT<WIDE>::sub("

@@ -244,12 +244,22 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         break;
       case VM_TIME_EXTRACT: {
         // CoreTime bitfield (core_time.go): year@50:14 month@46:4 day@41:5
+        // hour@36:5 minute@30:6 second@24:6
+        constexpr int kShift[6] = {50, 46, 41, 36, 30, 24};
+        constexpr uint64_t kMask[6] = {0x3FFF, 0xF, 0x1F, 0x1F, 0x3F, 0x3F};
         uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
-        int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
-                  : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
-                               : (int64_t)((bits >> 41) & 0x1F);
+        int64_t f = (int64_t)((bits >> kShift[ins.b]) & kMask[ins.b]);
         vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+      }
+      case VM_MAX2:
+      case VM_MIN2: {
+        // builtinGreatest/Least*Sig: NULL if either operand is NULL
+        typename VT<WIDE>::T va = vm.get(ins.a), vb = vm.get(ins.b);
+        int c = VT<WIDE>::cmp(va, vb);
+        vm.set(ins.dst, (ins.op == VM_MAX2) == (c >= 0) ? va : vb);
+        vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
         break;
       }
       case VM_ABS: {
@@ -1038,12 +1048,20 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         case VM_TIME_EXTRACT: {
+          constexpr int kShift[6] = {50, 46, 41, 36, 30, 24};
+          constexpr uint64_t kMask[6] = {0x3FFF, 0xF, 0x1F, 0x1F, 0x3F, 0x3F};
           uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
-          int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
-                    : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
-                                 : (int64_t)((bits >> 41) & 0x1F);
+          int64_t f = (int64_t)((bits >> kShift[ins.b]) & kMask[ins.b]);
           vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        }
+        case VM_MAX2:
+        case VM_MIN2: {
+          typename VT<WIDE>::T va = vm.get(ins.a), vb = vm.get(ins.b);
+          int c = VT<WIDE>::cmp(va, vb);
+          vm.set(ins.dst, (ins.op == VM_MAX2) == (c >= 0) ? va : vb);
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         }
         case VM_ABS: {
@@ -1473,12 +1491,22 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
         break;
       case VM_TIME_EXTRACT: {
         // CoreTime bitfield (core_time.go): year@50:14 month@46:4 day@41:5
+        // hour@36:5 minute@30:6 second@24:6
+        constexpr int kShift[6] = {50, 46, 41, 36, 30, 24};
+        constexpr uint64_t kMask[6] = {0x3FFF, 0xF, 0x1F, 0x1F, 0x3F, 0x3F};
         uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
-        int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
-                  : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
-                               : (int64_t)((bits >> 41) & 0x1F);
+        int64_t f = (int64_t)((bits >> kShift[ins.b]) & kMask[ins.b]);
         vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+      }
+      case VM_MAX2:
+      case VM_MIN2: {
+        // builtinGreatest/Least*Sig: NULL if either operand is NULL
+        typename VT<WIDE>::T va = vm.get(ins.a), vb = vm.get(ins.b);
+        int c = VT<WIDE>::cmp(va, vb);
+        vm.set(ins.dst, (ins.op == VM_MAX2) == (c >= 0) ? va : vb);
+        vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
         break;
       }
       case VM_ABS: {
@@ -2869,12 +2897,20 @@ __global__ void projectKernel(const ProjDesc* __restrict__ dp) {
           vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         case VM_TIME_EXTRACT: {
+          constexpr int kShift[6] = {50, 46, 41, 36, 30, 24};
+          constexpr uint64_t kMask[6] = {0x3FFF, 0xF, 0x1F, 0x1F, 0x3F, 0x3F};
           uint64_t bits = VT<WIDE>::toAcc(vm.get(ins.a)).lo;
-          int64_t f = ins.b == 0 ? (int64_t)((bits >> 50) & 0x3FFF)
-                    : ins.b == 1 ? (int64_t)((bits >> 46) & 0xF)
-                                 : (int64_t)((bits >> 41) & 0x1F);
+          int64_t f = (int64_t)((bits >> kShift[ins.b]) & kMask[ins.b]);
           vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        }
+        case VM_MAX2:
+        case VM_MIN2: {
+          typename VT<WIDE>::T va = vm.get(ins.a), vb = vm.get(ins.b);
+          int c = VT<WIDE>::cmp(va, vb);
+          vm.set(ins.dst, (ins.op == VM_MAX2) == (c >= 0) ? va : vb);
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
           break;
         }
         case VM_ABS: {

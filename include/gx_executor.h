@@ -217,7 +217,11 @@ int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
  *    (base_semi_join.go)
  *  4 anti semi: each probe row once on NO match (NULL-key probe rows match
  *    nothing and emit); output = probe columns only
- *    (anti_semi_join_probe.go, non-null-aware variant) */
+ *    (anti_semi_join_probe.go, non-null-aware variant)
+ *  5 null-aware anti semi — the x NOT IN (y set) shape (null_aware NAASJ,
+ *    hash_join_v1.go:599): an EMPTY (post-filter) build side accepts every
+ *    probe row incl. NULL keys; any NULL build key accepts none; otherwise
+ *    anti semi that also rejects NULL-key probe rows. Probe columns only. */
 int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
                        const int32_t* build_keys, const int32_t* probe_keys,
                        int32_t n_keys, int32_t join_type);

@@ -1257,7 +1257,7 @@ class HashJoinExec : public Exec {
                std::unique_ptr<Exec> probe)
       : plan_(plan), node_(node), build_(std::move(build)), probe_(std::move(probe)) {
     int jt = node_.joinType;
-    if (jt != 3 && jt != 4) {
+    if (jt < 3) {
       for (size_t i = 0; i < build_->outTypes.size(); i++) {
         outTypes.push_back(build_->outTypes[i]);
         outFracs.push_back(build_->outFracs[i]);
@@ -1270,6 +1270,8 @@ class HashJoinExec : public Exec {
   }
   int32_t open() override {
     built_ = false;
+    buildRows_ = 0;
+    buildHasNullKey_ = false;
     pending_.reset();
     pendEmit_ = 0;
     buildData_.clear();
@@ -1296,7 +1298,7 @@ class HashJoinExec : public Exec {
       built_ = true;
     }
     int jt = node_.joinType;
-    size_t nb = (jt == 3 || jt == 4) ? 0 : build_->outTypes.size();
+    size_t nb = jt >= 3 ? 0 : build_->outTypes.size();
     size_t np = probe_->outTypes.size();
     for (;;) {
       // drain matches pending from the previous probe chunk first: one Next
@@ -1364,6 +1366,16 @@ class HashJoinExec : public Exec {
           continue;
         }
         if (jt == 4) {  // anti semi
+          if (!any) appendProbeOnly(i);
+          continue;
+        }
+        if (jt == 5) {
+          // null-aware anti semi — x NOT IN (y set), null_aware NAASJ
+          // (hash_join_v1.go:599): empty y accepts EVERY x (incl. NULL);
+          // a NULL y (or a NULL x against nonempty y) can never be TRUE;
+          // else plain anti semi
+          if (buildRows_ == 0) { appendProbeOnly(i); continue; }
+          if (hasNullKey[i] || buildHasNullKey_) continue;
           if (!any) appendProbeOnly(i);
           continue;
         }
@@ -1450,8 +1462,9 @@ class HashJoinExec : public Exec {
       std::vector<uint8_t> hasNull(n, 0);
       ec = serializeJoinKeys(ctx, node_.buildKeys, in, keys, hasNull);
       if (ec) return ec;
+      buildRows_ += n;
       for (int i = 0; i < n; i++) {
-        if (hasNull[i]) continue;
+        if (hasNull[i]) { buildHasNullKey_ = true; continue; }
         table_[keys[i]].push_back({ci, i});
       }
       matched_.push_back(std::vector<uint8_t>(n, 0));
@@ -1470,6 +1483,8 @@ class HashJoinExec : public Exec {
   std::vector<std::vector<uint8_t>> matched_;  // right outer: per build row
   bool probeDone_ = false;
   size_t drainChunk_ = 0, drainRow_ = 0;  // right-outer drain cursor
+  int64_t buildRows_ = 0;        // null-aware anti semi build scalars
+  bool buildHasNullKey_ = false;
 };
 
 }  // namespace

@@ -25,7 +25,7 @@ def _run(lib, jt, brows, prows, btypes=None, ptypes=None, out_types=None):
     btypes = btypes or [GX_TYPE_I64, GX_TYPE_I64]
     ptypes = ptypes or [GX_TYPE_I64, GX_TYPE_I64]
     if out_types is None:
-        out_types = ptypes if jt in (3, 4) else btypes + ptypes
+        out_types = ptypes if jt in (3, 4, 5) else btypes + ptypes
     b = P.Builder(lib)
     bsrc = b.source(btypes)
     psrc = b.source(ptypes)
@@ -197,3 +197,53 @@ def test_outer_sort_nullable_key_parity(desc):
     want = _run_outer_sorted(load_oracle(), desc)
     got = _run_outer_sorted(load_product(), desc)
     assert got == want
+
+
+def _naaj_cases():
+    """Null-aware anti semi (x NOT IN ...) shapes: plain, NULL probe key,
+    NULL build key, empty build, pred-filtered build (the 'valid set' is
+    POST-filter)."""
+    build_plain = [[1, 10], [2, 20], [2, 21]]
+    build_null = [[1, 10], [None, 30]]
+    probes = [[1, 100], [3, 300], [None, 400], [2, 200]]
+    return build_plain, build_null, probes
+
+
+def test_oracle_null_aware_anti_semi():
+    lib = load_oracle()
+    bp, bn, pr = _naaj_cases()
+    k = lambda rows: sorted(rows, key=lambda r: tuple((x is None, x) for x in r))
+    # plain: 3 and NULL?? -> NULL probe rejected; 1,2 match -> reject
+    assert _run(lib, 5, bp, pr) == k([(3, 300)])
+    # NULL build key: nothing qualifies
+    assert _run(lib, 5, bn, pr) == []
+    # empty build: EVERYTHING qualifies incl. the NULL probe row
+    assert _run(lib, 5, [], pr) == k([(1, 100), (3, 300), (None, 400),
+                                      (2, 200)])
+    # contrast with plain anti semi (jt=4): NULL probe row EMITS there
+    assert _run(lib, 4, bp, pr) == k([(3, 300), (None, 400)])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("case", ["plain", "nullbuild", "empty"])
+def test_null_aware_anti_semi_parity(case):
+    bp, bn, pr = _naaj_cases()
+    brows = {"plain": bp, "nullbuild": bn, "empty": []}[case]
+    want = _run(load_oracle(), 5, brows, pr)
+    got = _run(load_product(), 5, brows, pr)
+    assert got == want
+
+
+@pytest.mark.gpu
+def test_null_aware_anti_semi_random_parity():
+    brows, prows = _random_data(seed=7)
+    want = _run(load_oracle(), 5, brows, prows)
+    got = _run(load_product(), 5, brows, prows)
+    assert got == want
+    # build contains NULL keys (k==0 -> None) so the result must be empty
+    assert want == []
+    brows2 = [[k, v] for k, v in brows if k is not None]
+    want2 = _run(load_oracle(), 5, brows2, prows)
+    got2 = _run(load_product(), 5, brows2, prows)
+    assert got2 == want2
+    assert len(want2) > 50  # non-matching non-NULL probe rows

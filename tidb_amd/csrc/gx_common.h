@@ -476,6 +476,8 @@ int gxPackRows(const RowPackDesc& d, void* stream);
 int gxUnpackRows(const RowPackDesc& d, void* stream);
 
 // out-of-core join: per-row partition ids for side 0 (build) / 1 (probe)
+int gxHjBuildStats(const HashJoinDesc* devDesc, const HashJoinDesc& h,
+                   uint64_t* out2, void* stream);
 int gxHjPartIds(const HashJoinDesc* devDesc, const HashJoinDesc& h, int side,
                 int nParts, uint32_t* out, void* stream);
 

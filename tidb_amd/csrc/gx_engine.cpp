@@ -1432,9 +1432,9 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
                                 const PNode* postSel, int* stageOut) {
   const PPlan& plan = ex->plan;
   const PNode& jn = plan.nodes[joinNode];
-  if (jn.joinType < 0 || jn.joinType > 4) {
+  if (jn.joinType < 0 || jn.joinType > 5) {
     ex->err = "unsupported join type (0=inner, 1=left outer, 2=right outer, "
-              "3=semi, 4=anti semi)";
+              "3=semi, 4=anti semi, 5=null-aware anti semi)";
     return GX_ERR_INVALID;
   }
   if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
@@ -1516,8 +1516,8 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
   st.srcP = P.srcNode;
   st.buildStage = B.stage;
   st.probeStage = P.stage;
-  if (jn.joinType == 3 || jn.joinType == 4) {
-    // semi / anti semi emit the probe (outer) side only
+  if (jn.joinType >= 3) {
+    // semi / anti semi / null-aware anti semi emit the probe side only
     st.types = P.types;
     st.fracs = P.fracs;
   } else {
@@ -4265,6 +4265,32 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
     ex->err = "join kernel launch failed";
     return GX_ERR_INTERNAL;
   }
+  if (hj.joinType == 5) {
+    // null-aware anti semi (x NOT IN ...): two build scalars decide the
+    // shape (null_aware NAASJ, hash_join_v1.go:599):
+    //   no passing build rows  -> NOT IN (empty) is TRUE for EVERY probe
+    //                             row incl. NULL keys == plain anti semi
+    //   any NULL build key     -> NOT IN is never TRUE -> empty output
+    //   else                   -> anti semi that also rejects NULL probe
+    //                             keys (the kernels' joinType==5 paths)
+    uint64_t* bstats = (uint64_t*)devAlloc(ex, 16);
+    if (!bstats) { ex->err = "hipMalloc failed"; return GX_ERR_INTERNAL; }
+    HIP_OK(ex, hipMemsetAsync(bstats, 0, 16, ex->stream));
+    if (gxp::gxHjBuildStats(ex->devHj, hj, bstats, ex->stream) != 0) {
+      ex->err = "join build-stats launch failed";
+      return GX_ERR_INTERNAL;
+    }
+    uint64_t hstats[2] = {0, 0};
+    HIP_OK(ex, hipStreamSynchronize(ex->stream));
+    HIP_OK(ex, hipMemcpy(hstats, bstats, 16, hipMemcpyDeviceToHost));
+    if (hstats[0] == 0) {
+      hj.joinType = 4;  // empty valid set: every probe row qualifies
+    } else if (hstats[1] > 0) {
+      hj.probe.nRows = 0;  // a NULL y: no probe row can qualify
+    }
+    HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj),
+                              hipMemcpyHostToDevice, ex->stream));
+  }
   HIP_OK(ex, hipEventRecord(evB, ex->stream));
   if (gxp::gxHashJoinPhase(1, ex->devHj, hj, ex->stream) != 0) {
     ex->err = "join kernel launch failed";
@@ -4360,8 +4386,8 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
       gxp::DevTable bsub = hj.build;
       gxp::DevTable psub = hj.probe;
       int32_t rc;
-      if (hj.joinType == 3 || hj.joinType == 4) {
-        // semi / anti semi: probe (outer) columns only
+      if (hj.joinType >= 3) {
+        // semi / anti semi / null-aware: probe (outer) columns only
         rc = gatherCols(ex, psub, gatherP, total, &st.out, 0, false);
       } else {
         // outer joins carry null-extended rows on the inner side; the build

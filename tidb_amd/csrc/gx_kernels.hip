@@ -2321,7 +2321,9 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
       }
       if (needWalk) {
         uint64_t h;
-        if (hjLoad(d, d.probe, d.pKeyCol, row, K, &h)) {
+        if (!hjLoad(d, d.probe, d.pKeyCol, row, K, &h)) {
+          if (d.joinType == 5) pass = false;  // NAAJ: NULL key rejects
+        } else {
           head = gptr<uint32_t>(d.heads)[(uint32_t)(h & mask)];
           for (uint32_t cur = head; cur != 0;) {
             uint32_t brow = cur - 1;
@@ -2350,7 +2352,7 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
         if (pass && cnt == 0) { emit = 1; nullExt = true; }
       } else if (d.joinType == 3) {  // semi: once on any match
         emit = cnt ? 1 : 0;
-      } else if (d.joinType == 4) {  // anti semi: once on no match
+      } else if (d.joinType >= 4) {  // anti semi / null-aware: no match
         emit = (pass && cnt == 0) ? 1 : 0;
         nullExt = emit != 0;
       }
@@ -2472,6 +2474,50 @@ __global__ void hjFilterPairsKernel(const HashJoinDesc* __restrict__ dp) {
   }
 }
 
+// null-aware anti semi (x NOT IN (...)): the probe outcome depends on two
+// build-side scalars — how many rows pass the build filter, and how many of
+// those have a NULL key (null_aware NAASJ, hash_join_v1.go:599). Count both.
+template <bool G>
+__global__ void hjBuildStatsKernel(const HashJoinDesc* __restrict__ dp,
+                                   uint64_t* __restrict__ out2) {
+  const HashJoinDesc& d = *dp;
+  int64_t n = d.build.nRows;
+  uint64_t myPass = 0, myNull = 0;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    bool pass = d.nPredB == 0 ||
+                evalSimplePred(d.build, d.predB, d.strConstB, d.strConstBLen, row);
+    if (!pass) continue;
+    myPass++;
+    HjKeys<G> K;
+    uint64_t h;
+    if (!hjLoad(d, d.build, d.bKeyCol, row, K, &h)) myNull++;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    myPass += __shfl_down(myPass, off, 64);
+    myNull += __shfl_down(myNull, off, 64);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    if (myPass) atomicAdd((unsigned long long*)&out2[0],
+                          (unsigned long long)myPass);
+    if (myNull) atomicAdd((unsigned long long*)&out2[1],
+                          (unsigned long long)myNull);
+  }
+}
+
+int gxHjBuildStats(const HashJoinDesc* devDesc, const HashJoinDesc& h,
+                   uint64_t* out2, void* stream) {
+  int64_t n = h.build.nRows;
+  if (n == 0) return 0;
+  if (h.generalKeys)
+    hipLaunchKernelGGL(hjBuildStatsKernel<true>, dim3(gridFor(n)), dim3(256),
+                       0, (hipStream_t)stream, devDesc, out2);
+  else
+    hipLaunchKernelGGL(hjBuildStatsKernel<false>, dim3(gridFor(n)), dim3(256),
+                       0, (hipStream_t)stream, devDesc, out2);
+  return (int)hipGetLastError();
+}
+
 // out-of-core join (hash_join_spill.go analog): per-row partition id from
 // the join-key hash, using bits independent of the chain-table index so the
 // per-partition tables hash freely. NULL-key rows round-robin by row index —
@@ -2524,6 +2570,9 @@ __device__ inline void hjCountProlog(const HashJoinDesc& d, int64_t row,
   uint64_t h;
   if (*pass && hjLoad(d, d.probe, d.pKeyCol, row, *K, &h))
     *cur = gptr<uint32_t>(d.heads)[(uint32_t)(h & mask)];
+  else if (*pass && d.joinType == 5)
+    *pass = false;  // null-aware anti semi: NULL probe key -> NOT IN is
+                    // NULL -> reject (null_aware NAASJ step, hash_join_v1.go)
 }
 
 __device__ inline uint32_t hjCountEmit(const HashJoinDesc& d, bool active,
@@ -2531,7 +2580,7 @@ __device__ inline uint32_t hjCountEmit(const HashJoinDesc& d, bool active,
   if (!active) return 0;
   if (d.joinType == 1) return pass && cnt == 0 ? 1 : cnt;  // left outer
   if (d.joinType == 3) return cnt ? 1 : 0;                 // semi
-  if (d.joinType == 4) return pass && cnt == 0 ? 1 : 0;    // anti semi
+  if (d.joinType >= 4) return pass && cnt == 0 ? 1 : 0;    // anti semi / NAAJ
   return cnt;
 }
 

@@ -237,13 +237,13 @@ def test_null_aware_anti_semi_parity(case):
 @pytest.mark.gpu
 def test_null_aware_anti_semi_random_parity():
     brows, prows = _random_data(seed=7)
-    want = _run(load_oracle(), 5, brows, prows)
-    got = _run(load_product(), 5, brows, prows)
-    assert got == want
-    # build contains NULL keys (k==0 -> None) so the result must be empty
-    assert want == []
+    brows_null = brows + [[None, -1]]  # force a NULL build key
+    want = _run(load_oracle(), 5, brows_null, prows)
+    got = _run(load_product(), 5, brows_null, prows)
+    assert got == want == []  # a NULL y: NOT IN can never be TRUE
     brows2 = [[k, v] for k, v in brows if k is not None]
     want2 = _run(load_oracle(), 5, brows2, prows)
     got2 = _run(load_product(), 5, brows2, prows)
     assert got2 == want2
     assert len(want2) > 50  # non-matching non-NULL probe rows
+    assert all(k is not None for k, _ in want2)  # NULL probes rejected

@@ -221,7 +221,13 @@ int32_t gx_pb_topn(gx_pb* pb, int32_t child, const int32_t* key_exprs,
  *  5 null-aware anti semi — the x NOT IN (y set) shape (null_aware NAASJ,
  *    hash_join_v1.go:599): an EMPTY (post-filter) build side accepts every
  *    probe row incl. NULL keys; any NULL build key accepts none; otherwise
- *    anti semi that also rejects NULL-key probe rows. Probe columns only. */
+ *    anti semi that also rejects NULL-key probe rows. Probe columns only.
+ *  6 left outer semi (LeftOuterSemiJoin): EVERY probe row once, with a
+ *    trailing i64 scalar column = 1 if any build match else 0 (the
+ *    IN-subquery-in-select-list shape, non-null-aware).
+ *  7 null-aware left outer semi: scalar = x IN (y set) with NULL
+ *    semantics — 1 on match; NULL when no match but x is NULL (nonempty y)
+ *    or y contains NULL; else 0 (joinNAALOSJ, hash_join_v1.go). */
 int32_t gx_pb_hashjoin(gx_pb* pb, int32_t build_child, int32_t probe_child,
                        const int32_t* build_keys, const int32_t* probe_keys,
                        int32_t n_keys, int32_t join_type);

@@ -1267,6 +1267,10 @@ class HashJoinExec : public Exec {
       outTypes.push_back(probe_->outTypes[i]);
       outFracs.push_back(probe_->outFracs[i]);
     }
+    if (jt >= 6) {  // left outer semi: the x IN (...) scalar column
+      outTypes.push_back(GX_TYPE_I64);
+      outFracs.push_back(0);
+    }
   }
   int32_t open() override {
     built_ = false;
@@ -1377,6 +1381,20 @@ class HashJoinExec : public Exec {
           if (buildRows_ == 0) { appendProbeOnly(i); continue; }
           if (hasNullKey[i] || buildHasNullKey_) continue;
           if (!any) appendProbeOnly(i);
+          continue;
+        }
+        if (jt >= 6) {
+          // left outer semi: probe row + the x IN (y set) scalar.
+          // Null-aware (7): NULL when no match but the evidence is
+          // inconclusive (NULL x against nonempty y, or a NULL y)
+          appendProbeOnly(i);
+          Column& fc = pending_.cols[np];
+          if (any) fc.appendI64(1);
+          else if (jt == 7 && ((hasNullKey[i] && buildRows_ > 0) ||
+                               buildHasNullKey_))
+            fc.appendNull();
+          else
+            fc.appendI64(0);
           continue;
         }
         if (any) {

@@ -25,7 +25,12 @@ def _run(lib, jt, brows, prows, btypes=None, ptypes=None, out_types=None):
     btypes = btypes or [GX_TYPE_I64, GX_TYPE_I64]
     ptypes = ptypes or [GX_TYPE_I64, GX_TYPE_I64]
     if out_types is None:
-        out_types = ptypes if jt in (3, 4, 5) else btypes + ptypes
+        if jt in (3, 4, 5):
+            out_types = ptypes
+        elif jt in (6, 7):
+            out_types = ptypes + [GX_TYPE_I64]
+        else:
+            out_types = btypes + ptypes
     b = P.Builder(lib)
     bsrc = b.source(btypes)
     psrc = b.source(ptypes)
@@ -247,3 +252,43 @@ def test_null_aware_anti_semi_random_parity():
     assert got2 == want2
     assert len(want2) > 50  # non-matching non-NULL probe rows
     assert all(k is not None for k, _ in want2)  # NULL probes rejected
+
+
+def test_oracle_left_outer_semi():
+    """jt 6/7: every probe row once + the x IN (y set) scalar
+    (LeftOuterSemiJoin / its null-aware form, hash_join_v1.go)."""
+    lib = load_oracle()
+    k = lambda rows: sorted(rows, key=lambda r: tuple((x is None, x) for x in r))
+    # BROWS keys {1,2,2,None,5}; PROWS keys {1,2,3,None,2}
+    assert _run(lib, 6, BROWS, PROWS) == k(
+        [(1, 100, 1), (2, 200, 1), (2, 201, 1), (3, 300, 0), (None, 400, 0)])
+    # null-aware: build HAS a NULL key -> every no-match flag is NULL
+    assert _run(lib, 7, BROWS, PROWS) == k(
+        [(1, 100, 1), (2, 200, 1), (2, 201, 1), (3, 300, None),
+         (None, 400, None)])
+    # null-aware, build without NULL keys: only the NULL probe is NULL
+    bnn = [r for r in BROWS if r[0] is not None]
+    assert _run(lib, 7, bnn, PROWS) == k(
+        [(1, 100, 1), (2, 200, 1), (2, 201, 1), (3, 300, 0), (None, 400, None)])
+    # empty build: x IN (empty) is plain FALSE for everyone, even NULL x
+    assert _run(lib, 7, [], PROWS) == k(
+        [(1, 100, 0), (2, 200, 0), (3, 300, 0), (None, 400, 0), (2, 201, 0)])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("jt", [6, 7])
+def test_left_outer_semi_parity(jt):
+    for brows in (BROWS, [r for r in BROWS if r[0] is not None], []):
+        want = _run(load_oracle(), jt, brows, PROWS)
+        got = _run(load_product(), jt, brows, PROWS)
+        assert got == want, brows
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("jt", [6, 7])
+def test_left_outer_semi_random_parity(jt):
+    brows, prows = _random_data()
+    want = _run(load_oracle(), jt, brows, prows)
+    got = _run(load_product(), jt, brows, prows)
+    assert got == want
+    assert len(got) == len(prows)  # every probe row exactly once

@@ -422,6 +422,12 @@ struct HashJoinDesc {
   uint64_t* counters = nullptr;  // [0] count-phase total, [1] fill cursor,
                                  // [2] post-filter cursor
   uint32_t* errorFlag = nullptr;
+  // left outer semi scalar (joinType 6/7): the appended i64 flag is NULL
+  // when the no-match case cannot be decided (x IN S with NULL evidence):
+  // flag2 iff (probe key NULL && naNullIfKeyNull) || naNullAlways
+  // (naNullIfKeyNull = build nonempty, naNullAlways = build has a NULL key)
+  int32_t naNullIfKeyNull = 0;
+  int32_t naNullAlways = 0;
   // post-join filter (the join's "other conditions": VectorizedFilter over
   // the joined chunk, inner_join_probe.go:75 — expressed as a Selection
   // above the join). CNF; evaluated on the match pairs BEFORE the gather so
@@ -478,6 +484,8 @@ int gxUnpackRows(const RowPackDesc& d, void* stream);
 // out-of-core join: per-row partition ids for side 0 (build) / 1 (probe)
 int gxHjBuildStats(const HashJoinDesc* devDesc, const HashJoinDesc& h,
                    uint64_t* out2, void* stream);
+int gxHjFlagCol(const uint32_t* enc, int64_t n, int64_t* data,
+                uint8_t* nullBitmap, void* stream);
 int gxHjPartIds(const HashJoinDesc* devDesc, const HashJoinDesc& h, int side,
                 int nParts, uint32_t* out, void* stream);
 

@@ -1432,9 +1432,10 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
                                 const PNode* postSel, int* stageOut) {
   const PPlan& plan = ex->plan;
   const PNode& jn = plan.nodes[joinNode];
-  if (jn.joinType < 0 || jn.joinType > 5) {
+  if (jn.joinType < 0 || jn.joinType > 7) {
     ex->err = "unsupported join type (0=inner, 1=left outer, 2=right outer, "
-              "3=semi, 4=anti semi, 5=null-aware anti semi)";
+              "3=semi, 4=anti semi, 5=null-aware anti semi, 6=left outer "
+              "semi, 7=null-aware left outer semi)";
     return GX_ERR_INVALID;
   }
   if (jn.buildKeys.size() != jn.probeKeys.size() || jn.buildKeys.empty() ||
@@ -1517,9 +1518,14 @@ static int32_t compileJoinStage(gx_exec* ex, int joinNode,
   st.buildStage = B.stage;
   st.probeStage = P.stage;
   if (jn.joinType >= 3) {
-    // semi / anti semi / null-aware anti semi emit the probe side only
+    // semi family emits the probe side only; the left-outer-semi forms
+    // (6/7) append the x IN (...) scalar as a trailing i64 column
     st.types = P.types;
     st.fracs = P.fracs;
+    if (jn.joinType >= 6) {
+      st.types.push_back(GX_TYPE_I64);
+      st.fracs.push_back(0);
+    }
   } else {
     st.types = B.types;
     st.fracs = B.fracs;
@@ -4265,7 +4271,7 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
     ex->err = "join kernel launch failed";
     return GX_ERR_INTERNAL;
   }
-  if (hj.joinType == 5) {
+  if (hj.joinType == 5 || hj.joinType == 7) {
     // null-aware anti semi (x NOT IN ...): two build scalars decide the
     // shape (null_aware NAASJ, hash_join_v1.go:599):
     //   no passing build rows  -> NOT IN (empty) is TRUE for EVERY probe
@@ -4283,10 +4289,15 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
     uint64_t hstats[2] = {0, 0};
     HIP_OK(ex, hipStreamSynchronize(ex->stream));
     HIP_OK(ex, hipMemcpy(hstats, bstats, 16, hipMemcpyDeviceToHost));
-    if (hstats[0] == 0) {
-      hj.joinType = 4;  // empty valid set: every probe row qualifies
-    } else if (hstats[1] > 0) {
-      hj.probe.nRows = 0;  // a NULL y: no probe row can qualify
+    if (hj.joinType == 5) {
+      if (hstats[0] == 0) {
+        hj.joinType = 4;  // empty valid set: every probe row qualifies
+      } else if (hstats[1] > 0) {
+        hj.probe.nRows = 0;  // a NULL y: no probe row can qualify
+      }
+    } else {  // null-aware left outer semi: no-match flag NULL conditions
+      hj.naNullIfKeyNull = hstats[0] > 0 ? 1 : 0;
+      hj.naNullAlways = hstats[1] > 0 ? 1 : 0;
     }
     HIP_OK(ex, hipMemcpyAsync(ex->devHj, &hj, sizeof(hj),
                               hipMemcpyHostToDevice, ex->stream));
@@ -4387,8 +4398,23 @@ static int32_t runJoinStage(gx_exec* ex, gx_exec::JoinStage& st, bool isRoot) {
       gxp::DevTable psub = hj.probe;
       int32_t rc;
       if (hj.joinType >= 3) {
-        // semi / anti semi / null-aware: probe (outer) columns only
+        // semi family: probe (outer) columns only; 6/7 add the scalar col
         rc = gatherCols(ex, psub, gatherP, total, &st.out, 0, false);
+        if (rc == GX_OK && hj.joinType >= 6) {
+          gxp::DevCol& fc = st.out.cols[psub.nCols];
+          fc.data = devAlloc(ex, (size_t)total * 8 + 16);
+          fc.nullBitmap = (uint8_t*)devAlloc(ex, (total + 7) / 8 + 1);
+          if (!fc.data || !fc.nullBitmap) {
+            ex->err = "hipMalloc failed";
+            return GX_ERR_INTERNAL;
+          }
+          fc.hasNulls = 1;  // flag 2 rows (undecidable x IN S) are NULL
+          if (gxp::gxHjFlagCol(gatherB, (int64_t)total, (int64_t*)fc.data,
+                               fc.nullBitmap, ex->stream) != 0) {
+            ex->err = "flag column launch failed";
+            return GX_ERR_INTERNAL;
+          }
+        }
       } else {
         // outer joins carry null-extended rows on the inner side; the build
         // side (the random-access gather) goes through the row-pack path

@@ -2289,6 +2289,25 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
        row += stride) {
     bool active = row < n;
     if (__ballot(active) == 0) break;
+    if (FILL && d.joinType >= 6) {
+      // left outer semi: hits already IS the scalar; one output row per
+      // eligible probe row, outBuild carries the flag (0/1/2=NULL)
+      uint32_t enc = active ? gptr<uint32_t>(d.hits)[row] : kHjIneligible;
+      uint64_t emit = enc != kHjIneligible ? 1 : 0;
+      uint64_t pre = emit;
+      for (int off = 1; off < 64; off <<= 1) {
+        uint64_t t = __shfl_up(pre, off, 64);
+        if (lane >= off) pre += t;
+      }
+      if (emit) {
+        uint64_t base =
+            (uint64_t)gptr<int64_t>(d.tileBases)[(row - lane) >> 6] +
+            (pre - emit);
+        d.outBuild[base] = enc;
+        d.outProbe[base] = (uint32_t)row;
+      }
+      continue;
+    }
     uint32_t cnt = 0;
     uint32_t head = 0;
     bool pass = false;
@@ -2505,6 +2524,34 @@ __global__ void hjBuildStatsKernel(const HashJoinDesc* __restrict__ dp,
   }
 }
 
+// materialize the left-outer-semi scalar column from the flag-encoded
+// match array (one byte of null bitmap per 8 output rows; LSB-first 1=valid)
+__global__ void hjFlagColKernel(const uint32_t* __restrict__ enc, int64_t n,
+                                int64_t* __restrict__ data,
+                                uint8_t* __restrict__ nullBitmap) {
+  int64_t nBytes = (n + 7) / 8;
+  for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < nBytes;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t bits = 0;
+    for (int j = 0; j < 8; j++) {
+      int64_t i = b * 8 + j;
+      if (i >= n) break;
+      uint32_t e = gptr<uint32_t>(enc)[i];
+      data[i] = e == 1 ? 1 : 0;
+      if (e != 2) bits |= (uint8_t)(1u << j);
+    }
+    nullBitmap[b] = bits;
+  }
+}
+
+int gxHjFlagCol(const uint32_t* enc, int64_t n, int64_t* data,
+                uint8_t* nullBitmap, void* stream) {
+  if (n == 0) return 0;
+  hipLaunchKernelGGL(hjFlagColKernel, dim3(gridFor((n + 7) / 8)), dim3(256),
+                     0, (hipStream_t)stream, enc, n, data, nullBitmap);
+  return (int)hipGetLastError();
+}
+
 int gxHjBuildStats(const HashJoinDesc* devDesc, const HashJoinDesc& h,
                    uint64_t* out2, void* stream) {
   int64_t n = h.build.nRows;
@@ -2563,16 +2610,21 @@ int gxHjPartIds(const HashJoinDesc* devDesc, const HashJoinDesc& h, int side,
 template <bool G>
 __device__ inline void hjCountProlog(const HashJoinDesc& d, int64_t row,
                                      uint32_t mask, bool* pass,
-                                     uint32_t* cur, HjKeys<G>* K) {
+                                     uint32_t* cur, HjKeys<G>* K,
+                                     bool* keyNull) {
   *pass = d.nPredP == 0 || evalSimplePred(d.probe, d.predP, d.strConstP,
                                           d.strConstPLen, row);
   *cur = 0;
+  *keyNull = false;
   uint64_t h;
-  if (*pass && hjLoad(d, d.probe, d.pKeyCol, row, *K, &h))
+  if (*pass && hjLoad(d, d.probe, d.pKeyCol, row, *K, &h)) {
     *cur = gptr<uint32_t>(d.heads)[(uint32_t)(h & mask)];
-  else if (*pass && d.joinType == 5)
-    *pass = false;  // null-aware anti semi: NULL probe key -> NOT IN is
-                    // NULL -> reject (null_aware NAASJ step, hash_join_v1.go)
+  } else if (*pass) {
+    *keyNull = true;
+    if (d.joinType == 5)
+      *pass = false;  // null-aware anti semi: NULL probe key -> NOT IN is
+                      // NULL -> reject (null_aware NAASJ, hash_join_v1.go)
+  }
 }
 
 __device__ inline uint32_t hjCountEmit(const HashJoinDesc& d, bool active,
@@ -2580,7 +2632,9 @@ __device__ inline uint32_t hjCountEmit(const HashJoinDesc& d, bool active,
   if (!active) return 0;
   if (d.joinType == 1) return pass && cnt == 0 ? 1 : cnt;  // left outer
   if (d.joinType == 3) return cnt ? 1 : 0;                 // semi
-  if (d.joinType >= 4) return pass && cnt == 0 ? 1 : 0;    // anti semi / NAAJ
+  if (d.joinType == 4 || d.joinType == 5)
+    return pass && cnt == 0 ? 1 : 0;  // anti semi / null-aware anti semi
+  if (d.joinType >= 6) return pass ? 1 : 0;  // left outer semi: the scalar
   return cnt;
 }
 
@@ -2600,8 +2654,9 @@ __global__ void hjCountKernel(const HashJoinDesc* __restrict__ dp) {
     bool passA = false, passB = false;
     uint32_t curA = 0, curB = 0;
     HjKeys<G> KA, KB;
-    if (actA) hjCountProlog(d, rowA, mask, &passA, &curA, &KA);
-    if (actB) hjCountProlog(d, rowB, mask, &passB, &curB, &KB);
+    bool knA = false, knB = false;
+    if (actA) hjCountProlog(d, rowA, mask, &passA, &curA, &KA, &knA);
+    if (actB) hjCountProlog(d, rowB, mask, &passB, &curB, &KB, &knB);
     uint32_t cntA = 0, cntB = 0, hit0A = 0, hit0B = 0;
     while (curA != 0 || curB != 0) {
       if (curA != 0) {
@@ -2625,16 +2680,29 @@ __global__ void hjCountKernel(const HashJoinDesc* __restrict__ dp) {
         curB = gptr<uint32_t>(d.next)[br];
       }
     }
-    if (actA)
-      d.hits[rowA] = !passA      ? kHjIneligible
-                     : cntA == 0 ? 0u
-                     : cntA == 1 ? hit0A + 1
-                                 : kHjMulti;
-    if (actB)
-      d.hits[rowB] = !passB      ? kHjIneligible
-                     : cntB == 0 ? 0u
-                     : cntB == 1 ? hit0B + 1
-                                 : kHjMulti;
+    if (d.joinType >= 6) {
+      // left outer semi: hits IS the output scalar (0 / 1 / 2 = NULL);
+      // the fill pass streams it straight into the flag column
+      auto flag = [&](bool pass, uint32_t cnt, bool kn) -> uint32_t {
+        if (!pass) return kHjIneligible;
+        if (cnt) return 1u;
+        if ((kn && d.naNullIfKeyNull) || d.naNullAlways) return 2u;
+        return 0u;
+      };
+      if (actA) d.hits[rowA] = flag(passA, cntA, knA);
+      if (actB) d.hits[rowB] = flag(passB, cntB, knB);
+    } else {
+      if (actA)
+        d.hits[rowA] = !passA      ? kHjIneligible
+                       : cntA == 0 ? 0u
+                       : cntA == 1 ? hit0A + 1
+                                   : kHjMulti;
+      if (actB)
+        d.hits[rowB] = !passB      ? kHjIneligible
+                       : cntB == 0 ? 0u
+                       : cntB == 1 ? hit0B + 1
+                                   : kHjMulti;
+    }
     uint32_t emitA = hjCountEmit(d, actA, passA, cntA);
     uint32_t emitB = hjCountEmit(d, actB, passB, cntB);
     my += emitA + emitB;

@@ -232,3 +232,53 @@ def test_sum_avg_distinct_group_by():
                [GX_TYPE_I64, GX_TYPE_I64], [0, 0])
     assert sorted(got) == [(1, "2.5000", "5"), (2, "2.0000", "2"),
                            (3, "4.0000", "8")]
+
+
+def test_not_in_null_aware_join_goldens():
+    """jointest/join.result:1460-1500 — t1={1}, t2(b)={1,NULL}:
+    `where 1 not in (select b from t2)` and `where 2 not in ...` both
+    return NO rows (a NULL y defeats NOT IN); `select 1 in (select b from
+    t2)` = 1; after deleting b=1 (t2={NULL}): `1 in (...)` = NULL; with
+    t2 empty: NOT IN accepts and the IN scalar is 0."""
+    from tests.gxlib import GX_TYPE_I64
+    from tidb_amd.chunkpy import PyChunk
+    lib = load_oracle()
+
+    def run(jt, build_vals, probe_vals, out_types):
+        b = P.Builder(lib)
+        bsrc = b.source([GX_TYPE_I64])
+        psrc = b.source([GX_TYPE_I64])
+        j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                       [b.colref(0, GX_TYPE_I64)], join_type=jt)
+        ex = b.build(j)
+        bch = PyChunk([GX_TYPE_I64], max(len(build_vals), 1))
+        for v in build_vals:
+            bch.append_row([v])
+        pch = PyChunk([GX_TYPE_I64], max(len(probe_vals), 1))
+        for v in probe_vals:
+            pch.append_row([v])
+        ex.bind_chunks(bsrc, [bch])
+        ex.bind_chunks(psrc, [pch])
+        ex.open()
+        rows = ex.pull_all(out_types)
+        ex.close()
+        ex.free()
+        b.free()
+        return rows
+
+    t2 = [1, None]
+    # where 1 not in (select b from t2)  -> empty ; where 2 not in -> empty
+    assert run(5, t2, [1], [GX_TYPE_I64]) == []
+    assert run(5, t2, [2], [GX_TYPE_I64]) == []
+    # select 1 in (select b from t2) -> 1
+    assert run(7, t2, [1], [GX_TYPE_I64] * 2) == [(1, 1)]
+    # select 2 in (select b from t2) -> NULL (no match, NULL evidence)
+    assert run(7, t2, [2], [GX_TYPE_I64] * 2) == [(2, None)]
+    # delete b=1: t2 = {NULL}
+    assert run(7, [None], [1], [GX_TYPE_I64] * 2) == [(1, None)]
+    assert run(5, [None], [1], [GX_TYPE_I64]) == []
+    # t2 empty: NOT IN accepts; IN scalar is plain 0
+    assert run(5, [], [1], [GX_TYPE_I64]) == [(1,)]
+    assert run(7, [], [1], [GX_TYPE_I64] * 2) == [(1, 0)]
+    # select 1 in (select 1 from t2={NULL}) -> 1 (the subquery projects 1)
+    assert run(7, [1], [1], [GX_TYPE_I64] * 2) == [(1, 1)]

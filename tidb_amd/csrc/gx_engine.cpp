@@ -3252,13 +3252,18 @@ static int32_t runFused(gx_exec* ex) {
 // rows (shared by the one-shot and the out-of-core streaming paths)
 static int32_t fusedDecodeResults(gx_exec* ex) {
   const int64_t nSlots1 = (int64_t)1 << ex->desc.globalGroupsLog2;
-  std::vector<gxp::GroupSlot> table((size_t)nSlots1 * ex->desc.accBanks);
+  // banks beyond 0 are written ONLY by the AOT noLds direct-accumulate
+  // path; the LDS flush and the hipRTC kernels land in bank 0 — download
+  // and fold just bank 0 otherwise (a 16x table D2H every step cost the
+  // Q1 decode ~28 ms of pure host overhead)
+  const int usedBanks = ex->desc.noLds ? ex->desc.accBanks : 1;
+  std::vector<gxp::GroupSlot> table((size_t)nSlots1 * usedBanks);
   HIP_OK(ex, hipMemcpy(table.data(), ex->devTable,
                        table.size() * sizeof(gxp::GroupSlot),
                        hipMemcpyDeviceToHost));
   // fold the accumulator banks into bank 0 (sums/counts add, biased
   // min/max takes the extreme, f64 bits add as doubles)
-  for (int bank = 1; bank < ex->desc.accBanks; bank++) {
+  for (int bank = 1; bank < usedBanks; bank++) {
     for (int64_t i = 0; i < nSlots1; i++) {
       gxp::GroupSlot& dst = table[i];
       const gxp::GroupSlot& b = table[bank * nSlots1 + i];

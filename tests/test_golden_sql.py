@@ -282,3 +282,31 @@ def test_not_in_null_aware_join_goldens():
     assert run(7, [], [1], [GX_TYPE_I64] * 2) == [(1, 0)]
     # select 1 in (select 1 from t2={NULL}) -> 1 (the subquery projects 1)
     assert run(7, [1], [1], [GX_TYPE_I64] * 2) == [(1, 1)]
+
+
+def test_ifnull_with_div_by_zero_golden():
+    """expression/builtin.result:1342-1369 — t(b int) rows (0),(NULL),(4):
+    `select ifnull(b, b/0) from t` = 0.0000, NULL, 4.0000 — division by
+    zero yields NULL (never an error) and IFNULL falls back per row; the
+    display frac comes from the unified decimal type."""
+    from tests.gxlib import GX_F_IFNULL, GX_TYPE_I64
+    from tidb_amd.chunkpy import PyChunk
+    lib = load_oracle()
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64])
+    col = b.colref(0, GX_TYPE_I64)
+    bd = b.call(GX_F_CAST_DEC, GX_TYPE_DECIMAL, 4, col)
+    div = b.call(GX_F_DIV, GX_TYPE_DECIMAL, 4, bd, b.const_i64(0))
+    proj = b.projection(src, [b.call(GX_F_IFNULL, GX_TYPE_DECIMAL, 4, bd,
+                                     div)])
+    ex = b.build(proj)
+    ch = PyChunk([GX_TYPE_I64], 3)
+    for v in (0, None, 4):
+        ch.append_row([v])
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    got = ex.pull_all([GX_TYPE_DECIMAL], [4])
+    ex.close()
+    ex.free()
+    b.free()
+    assert got == [("0.0000",), (None,), ("4.0000",)]

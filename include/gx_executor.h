@@ -129,7 +129,14 @@ enum {
   /* GREATEST/LEAST (builtinGreatest*Sig / builtinLeast*Sig,
    * builtin_compare_vec.go:27: MergeNulls => NULL if ANY arg is NULL).
    * Two args on the device path; n-ary composes as a chain. */
-  GX_F_GREATEST = 50, GX_F_LEAST = 51
+  GX_F_GREATEST = 50, GX_F_LEAST = 51,
+  /* IF(cond, a, b) (builtinIfSig, builtin_control_vec_generated.go):
+   * a when cond is non-NULL and != 0, else b; the result is the chosen
+   * branch's value/NULL. CASE WHEN c1 THEN v1 WHEN c2 THEN v2 ELSE e END
+   * is the chain IF(c1, v1, IF(c2, v2, e)). In value context the compare
+   * family (GX_F_LT..GX_F_NE) evaluates to i64 0/1 (NULL if either
+   * operand is NULL), so conditions compose from comparisons. */
+  GX_F_IF = 52
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

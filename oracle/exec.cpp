@@ -395,6 +395,27 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
         if (ctx.err) *ctx.err = "tuple is only valid as a DISTINCT aggregate argument";
         return GX_ERR_INVALID;
       }
+      if (e.func == GX_F_IF) {
+        // builtinIfSig: NULL/0 cond -> else branch; result = that branch
+        Column c0, a1, b2;
+        int32_t err = evalVec(ctx, e.args[0], in, c0);
+        if (err) return err;
+        err = evalVec(ctx, e.args[1], in, a1);
+        if (err) return err;
+        err = evalVec(ctx, e.args[2], in, b2);
+        if (err) return err;
+        out.reset();
+        out.type = a1.type;
+        out.frac = std::max(a1.frac, b2.frac);
+        if (out.isVarlen()) out.offsets.assign(1, 0);
+        for (int i = 0; i < in.numRows(); i++) {
+          bool t = !c0.isNull(i) && c0.getI64(i) != 0;
+          const Column& pick = t ? a1 : b2;
+          if (pick.isNull(i)) out.appendNull();
+          else out.appendFrom(pick, i);
+        }
+        return GX_OK;
+      }
       if (e.func == GX_F_IFNULL) {
         // builtinIfNullSig: first non-NULL operand per row
         Column a, b2;

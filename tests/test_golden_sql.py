@@ -310,3 +310,36 @@ def test_ifnull_with_div_by_zero_golden():
     ex.free()
     b.free()
     assert got == [("0.0000",), (None,), ("4.0000",)]
+
+
+def test_if_golden():
+    """expression/builtin.result:1337-1341 — t(b int) rows (0),(NULL),(4):
+    `select if(b=0, 1, 1/b) from t` = 1.0000, NULL, 0.2500 — a NULL
+    condition picks the else branch, and 1/NULL is NULL; display frac 4
+    from the division type."""
+    from tests.gxlib import GX_F_EQ, GX_F_IF, GX_TYPE_I64
+    from tidb_amd.chunkpy import PyChunk
+    lib = load_oracle()
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64])
+    col = b.colref(0, GX_TYPE_I64)
+    bd = b.call(GX_F_CAST_DEC, GX_TYPE_DECIMAL, 4, col)
+    cond = b.call(GX_F_EQ, GX_TYPE_I64, 0, col, b.const_i64(0))
+    one = b.call(GX_F_CAST_DEC, GX_TYPE_DECIMAL, 4, b.const_i64(1))
+    div = b.call(GX_F_DIV, GX_TYPE_DECIMAL, 4, one, bd)
+    # the planner rounds the unified IF type to the display frac
+    # (ProduceDecWithSpecifiedTp) — expressed as the cast wrapper here
+    ife = b.call(GX_F_IF, GX_TYPE_DECIMAL, 4, cond, one, div)
+    proj = b.projection(src, [b.call(GX_F_CAST_DEC, GX_TYPE_DECIMAL, 4,
+                                     ife)])
+    ex = b.build(proj)
+    ch = PyChunk([GX_TYPE_I64], 3)
+    for v in (0, None, 4):
+        ch.append_row([v])
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    got = ex.pull_all([GX_TYPE_DECIMAL], [4])
+    ex.close()
+    ex.free()
+    b.free()
+    assert got == [("1.0000",), (None,), ("0.2500",)]

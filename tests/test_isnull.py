@@ -194,3 +194,75 @@ def test_ifnull_parity():
     _, g1, g2 = _run_ifnull(load_product())
     assert g1 == w1
     assert g2 == w2
+
+
+def _run_if(lib):
+    """IF(cond, a, b) with a value-context comparison condition: in a
+    projection and as a fused sum arg (sum(if(v > 3, d, 0)))."""
+    from tests.gxlib import GX_AGG_SUM, GX_F_GT, GX_F_IF
+    rows, chunks = _data(lib)
+    b = P.Builder(lib)
+    src = b.source(TYPES, FRACS)
+    v = b.colref(1, GX_TYPE_I64)
+    d = b.colref(2, GX_TYPE_DECIMAL, 2)
+    zero = b.const_dec(_dec(lib, "0.00"))
+    cond = b.call(GX_F_GT, GX_TYPE_I64, 0, v, b.const_i64(3))
+    proj = b.projection(src, [
+        b.colref(0, GX_TYPE_I64),
+        b.call(GX_F_IF, GX_TYPE_DECIMAL, 2, cond, d, zero),
+        b.call(GX_F_IF, GX_TYPE_I64, 0, cond, v, b.const_i64(-9)),
+    ])
+    ex = b.build(proj)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out1 = ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64], [0, 2, 0])
+    ex.close()
+    ex.free()
+
+    agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                    [(GX_AGG_SUM,
+                      b.call(GX_F_IF, GX_TYPE_DECIMAL, 2, cond, d, zero),
+                      2)])
+    ex = b.build(agg)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    out2 = sorted(ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2]))
+    ex.close()
+    ex.free()
+    b.free()
+    return rows, out1, out2
+
+
+def test_oracle_if():
+    from fractions import Fraction
+    rows, out1, out2 = _run_if(load_oracle())
+    for (k, dv, vv), r in zip(out1, rows):
+        t = r[1] is not None and r[1] > 3
+        assert k == r[0]
+        if t:
+            assert dv == r[2] and vv == r[1]
+        else:
+            assert dv == "0.00" and vv == -9
+    want = {}
+    for r in rows:
+        t = r[1] is not None and r[1] > 3
+        add = Fraction(r[2]) if (t and r[2] is not None) else \
+            (Fraction(0) if not t else None)
+        s, seen = want.get(r[0], (Fraction(0), False))
+        if add is not None:
+            s, seen = s + add, True
+        want[r[0]] = (s, seen)
+    got = {k: v for k, v in out2}
+    for k, (s, seen) in want.items():
+        if not seen:
+            assert got[k] is None
+        else:
+            assert Fraction(got[k]) == s
+
+
+@pytest.mark.gpu
+def test_if_parity():
+    _, w1, w2 = _run_if(load_oracle())
+    _, g1, g2 = _run_if(load_product())
+    assert g1 == w1
+    assert g2 == w2

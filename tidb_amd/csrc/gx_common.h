@@ -65,6 +65,10 @@ enum VmOp : int32_t {
   VM_MAX2 = 13,          // dst <- max(a, b) (builtinGreatest*Sig; NULL if
                          // either is NULL; scales engine-aligned)
   VM_MIN2 = 14,          // dst <- min(a, b) (builtinLeast*Sig)
+  VM_CMP = 15,           // dst <- (a <op> b) as i64 0/1, op in ins.c
+                         // (GX_F_LT..GX_F_NE); NULL if either is NULL
+  VM_IF = 16,            // dst <- cond(a) truthy ? b : c (builtinIfSig;
+                         // NULL cond counts false; branch scales aligned)
 };
 
 struct VmIns {

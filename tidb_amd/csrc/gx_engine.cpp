@@ -598,6 +598,30 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
         *scaleOut = sa;
         break;
       }
+      if (e.func == GX_F_IF && e.args.size() == 3) {
+        // builtinIfSig: branches scale-align; the cond keeps its own scale
+        int sc2 = 0, sa = 0, sb = 0;
+        int rc2 = compileExpr(ex, e.args[0], &sc2);
+        if (rc2 < 0) return -1;
+        int ra = compileExpr(ex, e.args[1], &sa);
+        if (ra < 0) return -1;
+        int rb = compileExpr(ex, e.args[2], &sb);
+        if (rb < 0) return -1;
+        int target = std::max(sa, sb);
+        int tmpA = -1, tmpB = -1;
+        if (sa < target) tmpA = ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa);
+        if (sb < target) tmpB = rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb);
+        if (ra < 0 || rb < 0) break;
+        reg = emit(gxp::VM_IF, allocReg(), rc2, ra);
+        if (reg >= 0) d.ins[d.nIns - 1].c = rb;
+        if (tmpA >= 0) ex->vmFreeRegs.push_back(tmpA);
+        if (tmpB >= 0) ex->vmFreeRegs.push_back(tmpB);
+        *scaleOut = target;
+        release(e.args[0]);
+        release(e.args[1]);
+        release(e.args[2]);
+        break;
+      }
       if (e.func >= GX_F_YEAR && e.func <= GX_F_SECOND) {
         // YEAR/MONTH/DAY: raw CoreTime bits load (nulls tracked by the
         // load) + bitfield extract; only direct TIME columns this round
@@ -625,22 +649,26 @@ static int compileExpr(gx_exec* ex, int exprId, int* scaleOut) {
       int rb = compileExpr(ex, e.args[1], &sb);
       if (rb < 0) return -1;
       if (e.func == GX_F_IFNULL || e.func == GX_F_GREATEST ||
-          e.func == GX_F_LEAST) {
-        // builtinIfNullSig / builtinGreatest*Sig / builtinLeast*Sig:
-        // align operand scales, result keeps the max scale
-        int op2 = e.func == GX_F_IFNULL
-                      ? gxp::VM_IFNULL
-                      : (e.func == GX_F_GREATEST ? gxp::VM_MAX2
-                                                 : gxp::VM_MIN2);
+          e.func == GX_F_LEAST || e.func <= GX_F_NE) {
+        // builtinIfNullSig / builtinGreatest*Sig / builtinLeast*Sig, and
+        // the compare family in VALUE context (i64 0/1, NULL on NULL):
+        // align operand scales; compares emit VM_CMP with the op in .c
+        int op2 = e.func <= GX_F_NE
+                      ? gxp::VM_CMP
+                      : e.func == GX_F_IFNULL
+                            ? gxp::VM_IFNULL
+                            : (e.func == GX_F_GREATEST ? gxp::VM_MAX2
+                                                       : gxp::VM_MIN2);
         int target = std::max(sa, sb);
         int tmpA = -1, tmpB = -1;
         if (sa < target) tmpA = ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa);
         if (sb < target) tmpB = rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb);
         if (ra < 0 || rb < 0) break;
         reg = emit(op2, allocReg(), ra, rb);
+        if (reg >= 0 && op2 == gxp::VM_CMP) d.ins[d.nIns - 1].c = e.func;
         if (tmpA >= 0) ex->vmFreeRegs.push_back(tmpA);
         if (tmpB >= 0) ex->vmFreeRegs.push_back(tmpB);
-        *scaleOut = target;
+        *scaleOut = op2 == gxp::VM_CMP ? 0 : target;
         release(e.args[0]);
         release(e.args[1]);
         break;
@@ -864,6 +892,22 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
         *scaleOut = sa;
         break;
       }
+      if (e.func == GX_F_IF && e.args.size() == 3) {
+        int sc2 = 0, sa = 0, sb = 0;
+        int rc2 = vmCompile(ex, B, e.args[0], &sc2);
+        if (rc2 < 0) return -1;
+        int ra = vmCompile(ex, B, e.args[1], &sa);
+        if (ra < 0) return -1;
+        int rb = vmCompile(ex, B, e.args[2], &sb);
+        if (rb < 0) return -1;
+        int target = std::max(sa, sb);
+        if (sa < target) ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa, -1);
+        if (sb < target) rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb, -1);
+        if (ra < 0 || rb < 0) break;
+        reg = emit(gxp::VM_IF, allocReg(), rc2, ra, rb);
+        *scaleOut = target;
+        break;
+      }
       if (e.func >= GX_F_YEAR && e.func <= GX_F_SECOND) {
         const PExpr* a0 =
             e.args.size() == 1 ? &ex->plan.exprs[e.args[0]] : nullptr;
@@ -892,17 +936,20 @@ static int vmCompile(gx_exec* ex, VmBuild& B, int exprId, int* scaleOut) {
       int rb = vmCompile(ex, B, e.args[1], &sb);
       if (rb < 0) return -1;
       if (e.func == GX_F_IFNULL || e.func == GX_F_GREATEST ||
-          e.func == GX_F_LEAST) {
-        int op2 = e.func == GX_F_IFNULL
-                      ? gxp::VM_IFNULL
-                      : (e.func == GX_F_GREATEST ? gxp::VM_MAX2
-                                                 : gxp::VM_MIN2);
+          e.func == GX_F_LEAST || e.func <= GX_F_NE) {
+        int op2 = e.func <= GX_F_NE
+                      ? gxp::VM_CMP
+                      : e.func == GX_F_IFNULL
+                            ? gxp::VM_IFNULL
+                            : (e.func == GX_F_GREATEST ? gxp::VM_MAX2
+                                                       : gxp::VM_MIN2);
         int target = std::max(sa, sb);
         if (sa < target) ra = emit(gxp::VM_SCALE_UP, allocReg(), ra, target - sa, -1);
         if (sb < target) rb = emit(gxp::VM_SCALE_UP, allocReg(), rb, target - sb, -1);
         if (ra < 0 || rb < 0) break;
-        reg = emit(op2, allocReg(), ra, rb, -1);
-        *scaleOut = target;
+        reg = emit(op2, allocReg(), ra, rb,
+                   op2 == gxp::VM_CMP ? e.func : -1);
+        *scaleOut = op2 == gxp::VM_CMP ? 0 : target;
         break;
       }
       int op;

@@ -161,6 +161,18 @@ static void emitRowPipe(std::ostringstream& s, const FusedQueryDesc& d) {
           << "; }\n";
         break;
       }
+      case gxp::VM_CMP:
+        s << "  { int c = VT<WIDE>::cmp(v" << ins.a << ", v" << ins.b
+          << "); " << v << " = VT<WIDE>::fromI64(cmpResult(c, " << ins.c
+          << ") ? 1 : 0, &ovf); " << nv << " = n" << ins.a << " || n"
+          << ins.b << "; }\n";
+        break;
+      case gxp::VM_IF:
+        s << "  { bool t = !n" << ins.a << " && VT<WIDE>::cmp(v" << ins.a
+          << ", VT<WIDE>::zero()) != 0; " << v << " = t ? v" << ins.b
+          << " : v" << ins.c << "; " << nv << " = t ? n" << ins.b << " : n"
+          << ins.c << "; }\n";
+        break;
       case gxp::VM_MAX2:
       case gxp::VM_MIN2:
         s << "  { int c = VT<WIDE>::cmp(v" << ins.a << ", v" << ins.b
@@ -677,6 +689,19 @@ static void emitJaVm(std::ostringstream& s, const JoinAggDesc& d) {
           << "ULL), &ovf); bool " << nv << " = n" << ins.a << ";\n";
         break;
       }
+      case gxp::VM_CMP:
+        s << "    T " << v << " = VT<WIDE>::fromI64(cmpResult(VT<WIDE>::cmp(v"
+          << ins.a << ", v" << ins.b << "), " << ins.c
+          << ") ? 1 : 0, &ovf); bool " << nv << " = n" << ins.a << " || n"
+          << ins.b << ";\n";
+        break;
+      case gxp::VM_IF:
+        s << "    bool t" << ins.dst << " = !n" << ins.a
+          << " && VT<WIDE>::cmp(v" << ins.a << ", VT<WIDE>::zero()) != 0; T "
+          << v << " = t" << ins.dst << " ? v" << ins.b << " : v" << ins.c
+          << "; bool " << nv << " = t" << ins.dst << " ? n" << ins.b
+          << " : n" << ins.c << ";\n";
+        break;
       case gxp::VM_MAX2:
       case gxp::VM_MIN2:
         s << "    T " << v << " = VT<WIDE>::cmp(v" << ins.a << ", v" << ins.b

@@ -253,6 +253,22 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
         vm.setNull(ins.dst, vm.isNull(ins.a));
         break;
       }
+      case VM_CMP: {
+        // comparisons in VALUE context (builtin_compare_vec family): i64
+        // 0/1, NULL if either operand is NULL
+        int c = VT<WIDE>::cmp(vm.get(ins.a), vm.get(ins.b));
+        vm.set(ins.dst, VT<WIDE>::fromI64(cmpResult(c, ins.c) ? 1 : 0, &ovf));
+        vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+        break;
+      }
+      case VM_IF: {
+        // builtinIfSig: NULL/0 cond -> else branch; result = chosen branch
+        bool t = !vm.isNull(ins.a) &&
+                 VT<WIDE>::cmp(vm.get(ins.a), VT<WIDE>::zero()) != 0;
+        vm.set(ins.dst, t ? vm.get(ins.b) : vm.get(ins.c));
+        vm.setNull(ins.dst, t ? vm.isNull(ins.b) : vm.isNull(ins.c));
+        break;
+      }
       case VM_MAX2:
       case VM_MIN2: {
         // builtinGreatest/Least*Sig: NULL if either operand is NULL
@@ -1056,6 +1072,20 @@ __global__ void jaProbeKernel(const JoinAggDesc* __restrict__ dp) {
           vm.setNull(ins.dst, vm.isNull(ins.a));
           break;
         }
+        case VM_CMP: {
+          int c = VT<WIDE>::cmp(vm.get(ins.a), vm.get(ins.b));
+          vm.set(ins.dst,
+                 VT<WIDE>::fromI64(cmpResult(c, ins.c) ? 1 : 0, &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+          break;
+        }
+        case VM_IF: {
+          bool t = !vm.isNull(ins.a) &&
+                   VT<WIDE>::cmp(vm.get(ins.a), VT<WIDE>::zero()) != 0;
+          vm.set(ins.dst, t ? vm.get(ins.b) : vm.get(ins.c));
+          vm.setNull(ins.dst, t ? vm.isNull(ins.b) : vm.isNull(ins.c));
+          break;
+        }
         case VM_MAX2:
         case VM_MIN2: {
           typename VT<WIDE>::T va = vm.get(ins.a), vb = vm.get(ins.b);
@@ -1498,6 +1528,22 @@ __device__ __attribute__((always_inline)) inline bool processRowStaged(
         int64_t f = (int64_t)((bits >> kShift[ins.b]) & kMask[ins.b]);
         vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
         vm.setNull(ins.dst, vm.isNull(ins.a));
+        break;
+      }
+      case VM_CMP: {
+        // comparisons in VALUE context (builtin_compare_vec family): i64
+        // 0/1, NULL if either operand is NULL
+        int c = VT<WIDE>::cmp(vm.get(ins.a), vm.get(ins.b));
+        vm.set(ins.dst, VT<WIDE>::fromI64(cmpResult(c, ins.c) ? 1 : 0, &ovf));
+        vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+        break;
+      }
+      case VM_IF: {
+        // builtinIfSig: NULL/0 cond -> else branch; result = chosen branch
+        bool t = !vm.isNull(ins.a) &&
+                 VT<WIDE>::cmp(vm.get(ins.a), VT<WIDE>::zero()) != 0;
+        vm.set(ins.dst, t ? vm.get(ins.b) : vm.get(ins.c));
+        vm.setNull(ins.dst, t ? vm.isNull(ins.b) : vm.isNull(ins.c));
         break;
       }
       case VM_MAX2:
@@ -3020,6 +3066,20 @@ __global__ void projectKernel(const ProjDesc* __restrict__ dp) {
           int64_t f = (int64_t)((bits >> kShift[ins.b]) & kMask[ins.b]);
           vm.set(ins.dst, VT<WIDE>::fromI64(f, &ovf));
           vm.setNull(ins.dst, vm.isNull(ins.a));
+          break;
+        }
+        case VM_CMP: {
+          int c = VT<WIDE>::cmp(vm.get(ins.a), vm.get(ins.b));
+          vm.set(ins.dst,
+                 VT<WIDE>::fromI64(cmpResult(c, ins.c) ? 1 : 0, &ovf));
+          vm.setNull(ins.dst, vm.isNull(ins.a) || vm.isNull(ins.b));
+          break;
+        }
+        case VM_IF: {
+          bool t = !vm.isNull(ins.a) &&
+                   VT<WIDE>::cmp(vm.get(ins.a), VT<WIDE>::zero()) != 0;
+          vm.set(ins.dst, t ? vm.get(ins.b) : vm.get(ins.c));
+          vm.setNull(ins.dst, t ? vm.isNull(ins.b) : vm.isNull(ins.c));
           break;
         }
         case VM_MAX2:

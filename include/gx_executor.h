@@ -136,7 +136,14 @@ enum {
    * is the chain IF(c1, v1, IF(c2, v2, e)). In value context the compare
    * family (GX_F_LT..GX_F_NE) evaluates to i64 0/1 (NULL if either
    * operand is NULL), so conditions compose from comparisons. */
-  GX_F_IF = 52
+  GX_F_IF = 52,
+  /* logical OR (builtinLogicOrSig) INSIDE a Selection conjunct: each
+   * top-level condition may be a disjunction of simple predicates
+   * (col cmp const / col cmp col / LIKE / IS NULL); nested ORs flatten.
+   * Supported on the standalone Selection and inner-join other-condition
+   * paths this round (a NULL leaf behaves as FALSE, which matches
+   * VecEvalBool's NULL-rejects within both OR and AND). */
+  GX_F_OR = 53
 };
 
 /* ---- aggregate function codes (pkg/executor/aggfuncs) ---- */

@@ -395,6 +395,25 @@ int32_t evalVec(EvalCtx& ctx, int exprId, const Chunk& in, Column& out) {
         if (ctx.err) *ctx.err = "tuple is only valid as a DISTINCT aggregate argument";
         return GX_ERR_INVALID;
       }
+      if (e.func == GX_F_OR) {
+        // builtinLogicOrSig: TRUE if either side is truthy, NULL when
+        // undecidable (NULL vs 0/NULL), else FALSE
+        Column a, b2;
+        int32_t err = evalVec(ctx, e.args[0], in, a);
+        if (err) return err;
+        err = evalVec(ctx, e.args[1], in, b2);
+        if (err) return err;
+        out.reset();
+        out.type = GX_TYPE_I64;
+        for (int i = 0; i < in.numRows(); i++) {
+          bool aT = !a.isNull(i) && a.getI64(i) != 0;
+          bool bT = !b2.isNull(i) && b2.getI64(i) != 0;
+          if (aT || bT) out.appendI64(1);
+          else if (a.isNull(i) || b2.isNull(i)) out.appendNull();
+          else out.appendI64(0);
+        }
+        return GX_OK;
+      }
       if (e.func == GX_F_IF) {
         // builtinIfSig: NULL/0 cond -> else branch; result = that branch
         Column c0, a1, b2;

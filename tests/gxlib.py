@@ -35,6 +35,7 @@ GX_F_YEAR, GX_F_MONTH, GX_F_DAY = 44, 45, 46
 GX_F_HOUR, GX_F_MINUTE, GX_F_SECOND = 47, 48, 49
 GX_F_GREATEST, GX_F_LEAST = 50, 51
 GX_F_IF = 52
+GX_F_OR = 53
 
 GX_AGG_COUNT, GX_AGG_SUM, GX_AGG_AVG, GX_AGG_MIN, GX_AGG_MAX, GX_AGG_FIRSTROW = range(6)
 GX_AGG_MODE_COMPLETE, GX_AGG_MODE_PARTIAL, GX_AGG_MODE_FINAL = 0, 1, 2

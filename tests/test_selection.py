@@ -95,3 +95,63 @@ def test_selection_varlen_parity():
     b = run(load_product())
     assert len(a) == len(b) > 100
     assert a == b
+
+
+def _run_or_selection(lib, n=3000):
+    """Disjunctive conjuncts (LogicOr inside the CNF): WHERE (qty < 10.00
+    OR retflag = 'R' OR qty > 45.00) AND shipdate > 1995-03-15."""
+    import ctypes
+
+    from tests.gxlib import GX_F_OR
+
+    def dec(s):
+        out = (ctypes.c_uint8 * 40)()
+        assert lib.gx_dec_from_string(s.encode(), len(s.encode()), out) == 0
+        return bytes(out)
+
+    b = P.Builder(lib)
+    src = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+    qty = b.colref(P.L_QUANTITY, GX_TYPE_DECIMAL, 2)
+    c_or = b.call(GX_F_OR, GX_TYPE_I64, 0,
+                  b.call(GX_F_OR, GX_TYPE_I64, 0,
+                         b.call(GX_F_LT, GX_TYPE_I64, 0, qty,
+                                b.const_dec(dec("10.00"))),
+                         b.call(GX_F_EQ, GX_TYPE_I64, 0,
+                                b.colref(P.L_RETFLAG, GX_TYPE_STRING),
+                                lib.gx_pb_const_str(b.pb, b"R", 1))),
+                  b.call(GX_F_GT, GX_TYPE_I64, 0, qty,
+                         b.const_dec(dec("45.00"))))
+    c_date = b.call(GX_F_GT, GX_TYPE_I64, 0,
+                    b.colref(P.L_SHIPDATE, GX_TYPE_TIME),
+                    b.const_time(lib.gx_time_from_date(1995, 3, 15)))
+    root = b.selection(src, [c_or, c_date])
+    ex = b.build(root)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, n)
+    ex.open()
+    rows = ex.pull_all(P.LINEITEM_TYPES, P.LINEITEM_FRACS,
+                       data_caps=[None] * 5 + [2048, 2048] + [None])
+    ex.close()
+    ex.free()
+    b.free()
+    return rows
+
+
+def test_oracle_or_selection():
+    from decimal import Decimal
+    lib = load_oracle()
+    rows = _run_or_selection(lib)
+    from tests.test_oracle_q1 import pull_lineitem
+    raw = pull_lineitem(lib, 3000)
+    cut = (1995 << 50) | (3 << 46) | (15 << 41)
+    want = [r for r in raw
+            if (Decimal(r[1]) < 10 or r[5] == "R" or Decimal(r[1]) > 45)
+            and (r[7] & ~0xF) > cut]
+    assert rows == want
+    assert 0 < len(rows) < len(raw)
+
+
+@pytest.mark.gpu
+def test_or_selection_parity():
+    want = _run_or_selection(load_oracle())
+    got = _run_or_selection(load_product())
+    assert got == want

@@ -371,6 +371,8 @@ enum { PRED_IS_NULL = 7 };  // IS [NOT] NULL (builtin*IsNullSig: result is
 // one conjunct of a post-join filter (NULL operand rejects the row, the
 // VecEvalBool NULL semantics, expression.go:420-504)
 struct JoinPostPred {
+  int32_t orWith = 0;  // this pred ORs with the NEXT orWith entries (a
+                       // disjunctive conjunct: LogicOr inside the CNF)
   int32_t kind = 0;    // 0 = side-local <col cmp const>; 1 = <col cmp col>
   int32_t side = 0;    // kind 0: 0 = build, 1 = probe (pd.col is side-local)
   PredDesc pd{};       // kind 0
@@ -436,7 +438,7 @@ struct HashJoinDesc {
   // the joined chunk, inner_join_probe.go:75 — expressed as a Selection
   // above the join). CNF; evaluated on the match pairs BEFORE the gather so
   // rejected rows never touch HBM output.
-  JoinPostPred post[4];
+  JoinPostPred post[8];  // CNF conjuncts; OR groups flatten into entries
   int32_t nPost = 0;
   int64_t nPairs = 0;            // fill-phase total (filter input size)
   uint32_t* outBuild2 = nullptr; // filter-surviving pairs

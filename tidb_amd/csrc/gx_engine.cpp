@@ -1435,7 +1435,25 @@ static int32_t compilePostJoinPreds(gx_exec* ex, const PNode& sel,
   PNode outN;
   outN.colTypes = outTypes;
   outN.colFracs = outFracs;
-  for (int cid : sel.exprs) {
+  // each top-level conjunct may be a disjunction: flatten OR trees into
+  // leaf predicates; the group head records how many follow (orWith)
+  std::vector<int> leaves;
+  std::function<bool(int)> flatten = [&](int cid) {
+    const PExpr& ee = plan.exprs[cid];
+    if (ee.kind == EK_CALL && ee.func == GX_F_OR && ee.args.size() == 2)
+      return flatten(ee.args[0]) && flatten(ee.args[1]);
+    leaves.push_back(cid);
+    return true;
+  };
+  for (int condId : sel.exprs) {
+    leaves.clear();
+    if (!flatten(condId)) return GX_ERR_INVALID;
+    if (hj.nPost + (int)leaves.size() > 8) {
+      ex->err = "too many filter predicates after OR flattening";
+      return GX_ERR_INVALID;
+    }
+    int groupHead = hj.nPost;
+  for (int cid : leaves) {
     const PExpr& e = plan.exprs[cid];
     gxp::JoinPostPred q{};
     bool colcol = e.kind == EK_CALL && e.args.size() == 2 &&
@@ -1471,6 +1489,8 @@ static int32_t compilePostJoinPreds(gx_exec* ex, const PNode& sel,
       if (q.side == 1) q.pd.col -= nb;
     }
     hj.post[hj.nPost++] = q;
+  }
+    hj.post[groupHead].orWith = (int32_t)(hj.nPost - groupHead - 1);
   }
   return GX_OK;
 }

@@ -2473,10 +2473,9 @@ __global__ void hjProbeKernel(const HashJoinDesc* __restrict__ dp) {
 
 // evaluate the post-join CNF on one match pair; NULL operands reject
 // (VecEvalBool semantics)
-__device__ inline bool hjPostPass(const HashJoinDesc& d, uint32_t brow,
-                                  uint32_t prow) {
-  for (int p = 0; p < d.nPost; p++) {
-    const JoinPostPred& q = d.post[p];
+__device__ inline bool hjPostOne(const HashJoinDesc& d,
+                                 const JoinPostPred& q, uint32_t brow,
+                                 uint32_t prow) {
     if (q.kind == 0) {
       const DevTable& t = q.side == 0 ? d.build : d.probe;
       int64_t row = q.side == 0 ? (int64_t)brow : (int64_t)prow;
@@ -2502,6 +2501,20 @@ __device__ inline bool hjPostPass(const HashJoinDesc& d, uint32_t brow,
       }
       if (!cmpResult(c, q.cmp)) return false;
     }
+  return true;
+}
+
+__device__ inline bool hjPostPass(const HashJoinDesc& d, uint32_t brow,
+                                  uint32_t prow) {
+  // CNF of disjunction groups: a conjunct passes when ANY of its OR'd
+  // entries passes (LogicOr; a NULL leaf counts false on both sides)
+  for (int p = 0; p < d.nPost;) {
+    int n = d.post[p].orWith + 1;
+    bool ok = false;
+    for (int j = 0; j < n && !ok; j++)
+      ok = hjPostOne(d, d.post[p + j], brow, prow);
+    if (!ok) return false;
+    p += n;
   }
   return true;
 }

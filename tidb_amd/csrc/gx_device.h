@@ -665,11 +665,23 @@ __device__ inline bool makeWideGroupKey(const FusedQueryDesc& d, int64_t row,
 // simple single-table predicate (direct loads; build phases are cheap scans)
 __device__ inline bool evalSimplePred(const DevTable& tab, const PredDesc& pd,
                                       const uint8_t* strConst, int strConstLen,
-                                      int64_t row) {
+                                      int64_t row, uint32_t* err = nullptr) {
   const DevCol& c = tab.cols[pd.col];
   if (pd.kind == PRED_IS_NULL)  // the null bit IS the result (never NULL)
     return colIsNull(c, row) == (pd.cmp == 4 /*GX_F_EQ*/);
   if (colIsNull(c, row)) return false;
+  if (pd.kind == PRED_DEC_CMP_CONST) {
+    // units compare at the column's declared frac (narrow path; >18-digit
+    // rows flag RetryWide on the caller's error flag and reject)
+    if (!err) return false;
+    int64_t u;
+    int sc;
+    if (!loadDecimalUnits<false>((const uint8_t*)c.data + row * 40, &u, &sc,
+                                 err))
+      return false;
+    int64_t k = (int64_t)pd.constU64;
+    return cmpResult(u < k ? -1 : (u > k ? 1 : 0), pd.cmp);
+  }
   if (pd.kind == PRED_TIME_CMP_CONST) {
     uint64_t v = gptr<uint64_t>(c.data)[row] & ~0xFULL;
     uint64_t k = pd.constU64 & ~0xFULL;

@@ -2479,7 +2479,8 @@ __device__ inline bool hjPostOne(const HashJoinDesc& d,
     if (q.kind == 0) {
       const DevTable& t = q.side == 0 ? d.build : d.probe;
       int64_t row = q.side == 0 ? (int64_t)brow : (int64_t)prow;
-      if (!evalSimplePred(t, q.pd, q.strC, q.strCLen, row)) return false;
+      if (!evalSimplePred(t, q.pd, q.strC, q.strCLen, row, d.errorFlag))
+        return false;
     } else {
       int nbc = d.build.nCols;
       const DevCol& lc = q.lcol < nbc ? d.build.cols[q.lcol]

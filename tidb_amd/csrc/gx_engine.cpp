@@ -1056,6 +1056,26 @@ static bool compileTablePred(gx_exec* ex, const PNode& srcNode, int condId,
   } else if (ct == GX_TYPE_I64 && rhs->retType == GX_TYPE_I64) {
     pd.kind = gxp::PRED_I64_CMP_CONST;
     pd.constU64 = (uint64_t)rhs->constI64;
+  } else if (ct == GX_TYPE_DECIMAL && rhs->retType == GX_TYPE_DECIMAL) {
+    // units compare at the column's declared frac (evalSimplePred loads
+    // the narrow decimal; wider-than-i64 consts are out of device range)
+    __int128 u;
+    int sc;
+    if (!decToUnits(rhs->constDec, &u, &sc)) {
+      ex->err = "filter const decimal too wide";
+      return false;
+    }
+    int colFrac = srcNode.colFracs[pd.col];
+    while (sc < colFrac) {
+      u *= 10;
+      sc++;
+    }
+    if (sc != colFrac || u > INT64_MAX || u < INT64_MIN) {
+      ex->err = "filter decimal const/scale unsupported";
+      return false;
+    }
+    pd.kind = gxp::PRED_DEC_CMP_CONST;
+    pd.constU64 = (uint64_t)(int64_t)u;
   } else if (ct == GX_TYPE_STRING && rhs->retType == GX_TYPE_STRING &&
              (cmp == GX_F_EQ || cmp == GX_F_NE)) {
     if (rhs->constStr.size() > 16) {

@@ -155,3 +155,63 @@ def test_or_selection_parity():
     want = _run_or_selection(load_oracle())
     got = _run_or_selection(load_product())
     assert got == want
+
+
+def _run_or_agg(lib, n=3000):
+    """OR inside the FUSED CNF: count(*) group by retflag where
+    (qty < 10.00 OR qty > 45.00) AND shipdate > 1995-03-15 — the hipRTC
+    generator declines disjunctions, the interpreted kernel groups them."""
+    import ctypes
+
+    from tests.gxlib import GX_AGG_COUNT, GX_F_OR
+
+    def dec(s):
+        out = (ctypes.c_uint8 * 40)()
+        assert lib.gx_dec_from_string(s.encode(), len(s.encode()), out) == 0
+        return bytes(out)
+
+    b = P.Builder(lib)
+    src = b.source(P.LINEITEM_TYPES, P.LINEITEM_FRACS)
+    qty = b.colref(P.L_QUANTITY, GX_TYPE_DECIMAL, 2)
+    c_or = b.call(GX_F_OR, GX_TYPE_I64, 0,
+                  b.call(GX_F_LT, GX_TYPE_I64, 0, qty,
+                         b.const_dec(dec("10.00"))),
+                  b.call(GX_F_GT, GX_TYPE_I64, 0, qty,
+                         b.const_dec(dec("45.00"))))
+    c_date = b.call(GX_F_GT, GX_TYPE_I64, 0,
+                    b.colref(P.L_SHIPDATE, GX_TYPE_TIME),
+                    b.const_time(lib.gx_time_from_date(1995, 3, 15)))
+    sel = b.selection(src, [c_or, c_date])
+    agg = b.hashagg(sel, [b.colref(P.L_RETFLAG, GX_TYPE_STRING)],
+                    [(GX_AGG_COUNT, -1, 0)])
+    ex = b.build(agg)
+    ex.bind_tpch(src, GX_TPCH_LINEITEM, n)
+    ex.open()
+    rows = sorted(ex.pull_all([GX_TYPE_STRING, GX_TYPE_I64], [0, 0],
+                              data_caps=[1 << 12, None]))
+    ex.close()
+    ex.free()
+    b.free()
+    return rows
+
+
+def test_oracle_or_agg():
+    from decimal import Decimal
+    lib = load_oracle()
+    rows = _run_or_agg(lib)
+    from tests.test_oracle_q1 import pull_lineitem
+    raw = pull_lineitem(lib, 3000)
+    cut = (1995 << 50) | (3 << 46) | (15 << 41)
+    want = {}
+    for r in raw:
+        if (Decimal(r[1]) < 10 or Decimal(r[1]) > 45) and (r[7] & ~0xF) > cut:
+            want[r[5]] = want.get(r[5], 0) + 1
+    assert rows == sorted(want.items())
+    assert len(rows) >= 2
+
+
+@pytest.mark.gpu
+def test_or_agg_parity():
+    want = _run_or_agg(load_oracle())
+    got = _run_or_agg(load_product())
+    assert got == want

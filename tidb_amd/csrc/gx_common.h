@@ -91,6 +91,7 @@ enum PredKind : int32_t {
 };
 
 struct PredDesc {
+  int32_t orWith = 0;  // this pred ORs with the NEXT orWith entries
   int32_t kind;
   int32_t col;
   int32_t cmp;        // GX_F_LT..GX_F_NE

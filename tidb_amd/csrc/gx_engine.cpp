@@ -2039,9 +2039,24 @@ static int32_t compileFused(gx_exec* ex) {
   for (size_t c = 0; c < src->colTypes.size(); c++)
     setDevColMeta(&ex->desc.table.cols[c], src->colTypes[c], src->colFracs[c]);
 
-  // selection -> PredDescs
+  // selection -> PredDescs (CNF of OR groups: each conjunct may be a
+  // LogicOr tree of simple predicates, flattened with orWith group heads)
   if (sel) {
-    for (int condId : sel->exprs) {
+    std::vector<int> fusedLeaves;
+    std::function<void(int)> fusedFlatten = [&](int cid) {
+      const PExpr& fe = plan.exprs[cid];
+      if (fe.kind == EK_CALL && fe.func == GX_F_OR && fe.args.size() == 2) {
+        fusedFlatten(fe.args[0]);
+        fusedFlatten(fe.args[1]);
+      } else {
+        fusedLeaves.push_back(cid);
+      }
+    };
+    for (int topCond : sel->exprs) {
+      fusedLeaves.clear();
+      fusedFlatten(topCond);
+      int groupHead = ex->desc.nPreds;
+    for (int condId : fusedLeaves) {
       const PExpr& e = plan.exprs[condId];
       if (e.kind == EK_CALL && e.func == GX_F_LIKE_PREFIX &&
           e.args.size() == 2) {
@@ -2162,6 +2177,10 @@ static int32_t compileFused(gx_exec* ex) {
         return GX_ERR_INVALID;
       }
       ex->desc.preds[ex->desc.nPreds++] = pd;
+    }
+      if (ex->desc.nPreds > groupHead)
+        ex->desc.preds[groupHead].orWith =
+            (int32_t)(ex->desc.nPreds - groupHead - 1);
     }
   }
 

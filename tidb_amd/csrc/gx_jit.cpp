@@ -604,6 +604,11 @@ const JitProg* compile(const FusedQueryDesc& d, std::string* whyNot) {
       if (whyNot) *whyNot = "string builtin not specialized";
       return nullptr;
     }
+  for (int p = 0; p < d.nPreds; p++)
+    if (d.preds[p].orWith > 0) {
+      if (whyNot) *whyNot = "disjunctive filter not specialized";
+      return nullptr;
+    }
   return compileSource(generateSource(d), "genq_narrow", "genq_wide", whyNot);
 }
 

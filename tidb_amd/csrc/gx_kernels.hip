@@ -140,37 +140,46 @@ __device__ __attribute__((always_inline)) inline bool processRow(const FusedQuer
                                   uint64_t* mySel,
                                   WkLds64* wkCacheH = nullptr,
                                   WkLds32* wkCacheS = nullptr) {
-  // ---- filter (CNF; NULL rejects — expression.go:507 toBool) ----
-  bool pass = true;
-  for (int p = 0; p < d.nPreds && pass; p++) {
-    const PredDesc& pd = d.preds[p];
+  // ---- filter (CNF of OR groups; NULL rejects — expression.go:507
+  // toBool; a disjunctive conjunct passes when ANY member does) ----
+  bool hardFail = false;
+  auto evalOne = [&](const PredDesc& pd) -> bool {
     const DevCol& c = d.table.cols[pd.col];
-    if (pd.kind == PRED_IS_NULL) {  // null bit is the result; no null-reject
-      pass = colIsNull(c, row) == (pd.cmp == 4 /*GX_F_EQ*/);
-      continue;
-    }
-    if (colIsNull(c, row)) { pass = false; break; }
+    if (pd.kind == PRED_IS_NULL)  // null bit is the result; no null-reject
+      return colIsNull(c, row) == (pd.cmp == 4 /*GX_F_EQ*/);
+    if (colIsNull(c, row)) return false;
     if (pd.kind == PRED_TIME_CMP_CONST) {
       uint64_t v = raw.get(pd.slot).x & ~0xFULL;
       uint64_t k = pd.constU64 & ~0xFULL;
-      int cmp = v < k ? -1 : (v > k ? 1 : 0);
-      pass = cmpResult(cmp, pd.cmp);
-    } else if (pd.kind == PRED_I64_CMP_CONST) {
+      return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+    }
+    if (pd.kind == PRED_I64_CMP_CONST) {
       int64_t v = (int64_t)raw.get(pd.slot).x;
       int64_t k = (int64_t)pd.constU64;
-      int cmp = v < k ? -1 : (v > k ? 1 : 0);
-      pass = cmpResult(cmp, pd.cmp);
-    } else if (pd.kind == PRED_DEC_CMP_CONST) {
+      return cmpResult(v < k ? -1 : (v > k ? 1 : 0), pd.cmp);
+    }
+    if (pd.kind == PRED_DEC_CMP_CONST) {
       typename VT<WIDE>::T u;
       int sc;
       if (!loadDecimalUnits<WIDE>((const uint8_t*)c.data + row * 40, &u, &sc,
-                                  d.errorFlag))
+                                  d.errorFlag)) {
+        hardFail = true;
         return false;
+      }
       int cmp = VT<WIDE>::cmp(u, VT<WIDE>::fromI64((int64_t)pd.constU64, nullptr));
-      pass = cmpResult(cmp, pd.cmp);
-    } else {  // string EQ/NE const + LIKE-'prefix%' (per-pred inline const)
-      pass = evalSimplePred(d.table, pd, pd.strC, pd.strCLen, row);
+      return cmpResult(cmp, pd.cmp);
     }
+    // string EQ/NE const + LIKE-'prefix%' (per-pred inline const)
+    return evalSimplePred(d.table, pd, pd.strC, pd.strCLen, row);
+  };
+  bool pass = true;
+  for (int p = 0; p < d.nPreds && pass;) {
+    int ng = d.preds[p].orWith + 1;
+    bool ok = false;
+    for (int j = 0; j < ng && !ok; j++) ok = evalOne(d.preds[p + j]);
+    if (hardFail) return false;
+    pass = ok;
+    p += ng;
   }
   if (!pass) return true;
   (*mySel)++;

@@ -1169,7 +1169,6 @@ static int32_t compileJoinAgg(gx_exec* ex) {
   const PNode& liN = plan.nodes[srcL];
   int nc = (int)custN.colTypes.size();
   int no = (int)ordN.colTypes.size();
-  int nl = (int)liN.colTypes.size();
   gxp::JoinAggDesc& ja = ex->ja;
 
   // join keys (single-column int64)
@@ -3650,9 +3649,6 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
           return a.i64 < b.i64 ? -1 : (a.i64 > b.i64 ? 1 : 0);
       }
     };
-    auto valLess = [&](const OutRowVal& a, const OutRowVal& b) {
-      return cmpVal(a, b) < 0;
-    };
     auto tupLess = [&](const std::vector<OutRowVal>& a,
                        const std::vector<OutRowVal>& b) {
       for (size_t k = 0; k < a.size() && k < b.size(); k++) {
@@ -5182,7 +5178,6 @@ static int32_t runProject(gx_exec* ex) {
 
 static int32_t runFinalHost(gx_exec* ex) {
   const PNode& agg = ex->plan.nodes[ex->root];
-  const PNode& src = ex->plan.nodes[agg.child];
   auto it = ex->bindings.find(agg.child);
   if (it == ex->bindings.end() || !it->second.haveChunks) {
     ex->err = "FINAL mode needs bound partial chunks";

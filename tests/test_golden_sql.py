@@ -343,3 +343,43 @@ def test_if_golden():
     ex.free()
     b.free()
     assert got == [("1.0000",), (None,), ("0.2500",)]
+
+
+def test_order_by_limit_offset_goldens():
+    """executor.result:388-410 — t(a,b) = (1,1),(2,2),(3,30),(4,40),(5,5),
+    (6,6): `order by a limit 1, k` (offset 1) for k=1..4, and the keyless
+    `where a > 0 limit 1, 1` = row a=2 (child order preserved)."""
+    from tests.gxlib import GX_F_GT, GX_TYPE_I64
+    from tidb_amd.chunkpy import PyChunk
+    lib = load_oracle()
+    rows = [(1, 1), (2, 2), (3, 30), (4, 40), (5, 5), (6, 6)]
+
+    def run(limit, offset, keyless=False):
+        b = P.Builder(lib)
+        src = b.source([GX_TYPE_I64] * 2)
+        node = src
+        if keyless:
+            node = b.selection(src, [b.call(GX_F_GT, GX_TYPE_I64, 0,
+                                            b.colref(0, GX_TYPE_I64),
+                                            b.const_i64(0))])
+            root = b.topn(node, [], [], limit, offset)
+        else:
+            root = b.topn(src, [b.colref(0, GX_TYPE_I64)], [0], limit,
+                          offset)
+        ex = b.build(root)
+        ch = PyChunk([GX_TYPE_I64] * 2, len(rows))
+        for r in rows:
+            ch.append_row(list(r))
+        ex.bind_chunks(src, [ch])
+        ex.open()
+        out = ex.pull_all([GX_TYPE_I64] * 2)
+        ex.close()
+        ex.free()
+        b.free()
+        return out
+
+    assert run(1, 1) == [(2, 2)]
+    assert run(2, 1) == [(2, 2), (3, 30)]
+    assert run(3, 1) == [(2, 2), (3, 30), (4, 40)]
+    assert run(4, 1) == [(2, 2), (3, 30), (4, 40), (5, 5)]
+    assert run(1, 1, keyless=True) == [(2, 2)]

@@ -143,6 +143,8 @@ struct gx_exec {
   // DISTINCT rewrite (aggFuncDesc.HasDistinct): the device kernel groups by
   // (orig keys..., arg) — the dedup — and the decode folds back to the orig
   // keys. distinctNKeys >= 0 marks the rewrite; funcs/fracs are the USER's.
+  std::vector<int> havingConds;  // HAVING: Selection over the agg output,
+                                 // filtered on host over decoded group rows
   std::vector<int> distinctFuncs;
   std::vector<int> distinctFracs;
   std::vector<std::vector<int>> distinctArgSlot;  // per agg: appended-key indices (tuple args have several)
@@ -3772,6 +3774,88 @@ static int32_t fusedDecodeResults(gx_exec* ex) {
     }
     ex->resultRows = std::move(folded);
   }
+  if (!ex->havingConds.empty()) {
+    // HAVING over the decoded group rows (host; O(#groups)). Leaves:
+    // <output col cmp const>, IS [NOT] NULL, and LogicOr trees of those —
+    // VecEvalBool semantics (NULL rejects within both AND and OR)
+    auto leafPass = [&](const PExpr& e, const std::vector<OutRowVal>& r,
+                        bool* ok) -> bool {
+      if (e.kind == EK_CALL &&
+          (e.func == GX_F_IS_NULL || e.func == GX_F_IS_NOT_NULL) &&
+          e.args.size() == 1) {
+        const PExpr& a0 = ex->plan.exprs[e.args[0]];
+        if (a0.kind != EK_COLREF || a0.colIdx < 0 ||
+            a0.colIdx >= (int)r.size())
+          return *ok = false;
+        return r[a0.colIdx].isNull == (e.func == GX_F_IS_NULL);
+      }
+      if (e.kind != EK_CALL || e.func > GX_F_NE || e.args.size() != 2)
+        return *ok = false;
+      const PExpr* l = &ex->plan.exprs[e.args[0]];
+      const PExpr* rr = &ex->plan.exprs[e.args[1]];
+      int cmp = e.func;
+      if (l->kind == EK_CONST && rr->kind == EK_COLREF) {
+        std::swap(l, rr);
+        static const int mirror[6] = {GX_F_GT, GX_F_GE, GX_F_LT, GX_F_LE,
+                                      GX_F_EQ, GX_F_NE};
+        cmp = mirror[cmp];
+      }
+      if (l->kind != EK_COLREF || rr->kind != EK_CONST || l->colIdx < 0 ||
+          l->colIdx >= (int)r.size())
+        return *ok = false;
+      const OutRowVal& v = r[l->colIdx];
+      if (v.isNull) return false;  // NULL rejects
+      int c;
+      if (v.type == GX_TYPE_DECIMAL && rr->retType == GX_TYPE_DECIMAL) {
+        c = v.dec.Compare(rr->constDec);
+      } else if (v.type == GX_TYPE_I64 && rr->retType == GX_TYPE_I64) {
+        c = v.i64 < rr->constI64 ? -1 : (v.i64 > rr->constI64 ? 1 : 0);
+      } else if (v.type == GX_TYPE_TIME && rr->retType == GX_TYPE_TIME) {
+        uint64_t a = v.u64 & ~0xFULL, b = rr->constTime & ~0xFULL;
+        c = a < b ? -1 : (a > b ? 1 : 0);
+      } else if (v.type == GX_TYPE_STRING &&
+                 rr->retType == GX_TYPE_STRING &&
+                 (cmp == GX_F_EQ || cmp == GX_F_NE)) {
+        std::string a = v.str, b = rr->constStr;  // PAD SPACE
+        while (!a.empty() && a.back() == ' ') a.pop_back();
+        while (!b.empty() && b.back() == ' ') b.pop_back();
+        c = a < b ? -1 : (a > b ? 1 : 0);
+      } else {
+        return *ok = false;
+      }
+      switch (cmp) {
+        case GX_F_LT: return c < 0;
+        case GX_F_LE: return c <= 0;
+        case GX_F_GT: return c > 0;
+        case GX_F_GE: return c >= 0;
+        case GX_F_EQ: return c == 0;
+        default: return c != 0;
+      }
+    };
+    std::function<bool(int, const std::vector<OutRowVal>&, bool*)> condPass =
+        [&](int cid, const std::vector<OutRowVal>& r, bool* ok) -> bool {
+      const PExpr& e = ex->plan.exprs[cid];
+      if (e.kind == EK_CALL && e.func == GX_F_OR && e.args.size() == 2)
+        return condPass(e.args[0], r, ok) || condPass(e.args[1], r, ok);
+      return leafPass(e, r, ok);
+    };
+    bool ok = true;
+    std::vector<std::vector<OutRowVal>> kept;
+    kept.reserve(ex->resultRows.size());
+    for (auto& r : ex->resultRows) {
+      bool pass = true;
+      for (int cid : ex->havingConds) {
+        if (!condPass(cid, r, &ok)) { pass = false; break; }
+      }
+      if (!ok) {
+        ex->err = "unsupported HAVING condition (col cmp const, IS NULL, "
+                  "OR of those)";
+        return GX_ERR_INVALID;
+      }
+      if (pass) kept.push_back(std::move(r));
+    }
+    ex->resultRows = std::move(kept);
+  }
   applyPostSort(ex);
   return GX_OK;
 }
@@ -6142,6 +6226,27 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
       }
     ex->isFinalHost = true;
     ex->sourceNode = rn.child;
+  } else if (rn.kind == PK_SELECTION &&
+             ex->plan.nodes[rn.child].kind == PK_HASHAGG &&
+             ex->plan.nodes[rn.child].aggMode != GX_AGG_MODE_FINAL) {
+    // HAVING: a Selection over the aggregate's output — run the fused
+    // aggregation and filter the (small) decoded group rows on the host
+    // (the reference evaluates HAVING the same way, above the agg)
+    ex->havingConds = rn.exprs;
+    int aggRoot = rn.child;
+    int node = aggRoot;
+    while (ex->plan.nodes[node].kind == PK_HASHAGG ||
+           ex->plan.nodes[node].kind == PK_PROJECTION ||
+           ex->plan.nodes[node].kind == PK_SELECTION)
+      node = ex->plan.nodes[node].child;
+    int saved = ex->root;
+    ex->root = aggRoot;
+    int32_t rc = ex->plan.nodes[node].kind == PK_HASHJOIN
+                     ? compileAggOverJoin(ex, aggRoot)
+                     : compileFused(ex);
+    ex->root = saved;
+    if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
+    (void)rc;
   } else if (rn.kind == PK_HASHAGG) {
     // bottom of the subtree: a Source (fused pipeline) or a HashJoin
     // (aggregation over materialized joined rows)

@@ -103,3 +103,48 @@ def test_having_parity(case):
     got = _run(load_product(), rows, _cases(load_product())[case])
     assert got == want
     assert 0 < len(got)
+
+
+def _run_topn_having(lib, rows):
+    """ORDER BY count DESC, k LIMIT 5 over a HAVING filter over the agg."""
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2])
+    agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                    [(GX_AGG_SUM, b.colref(1, GX_TYPE_DECIMAL, 2), 2),
+                     (GX_AGG_COUNT, -1, 0)])
+    hav = b.selection(agg, [b.call(GX_F_GT, GX_TYPE_I64, 0,
+                                   b.colref(2, GX_TYPE_I64),
+                                   b.const_i64(120))])
+    root = b.topn(hav, [b.colref(2, GX_TYPE_I64),
+                        b.colref(0, GX_TYPE_I64)], [1, 0], 5)
+    ex = b.build(root)
+    chunks = []
+    for base in range(0, len(rows), 1000):
+        part = rows[base:base + 1000]
+        ch = PyChunk([GX_TYPE_I64, GX_TYPE_DECIMAL], len(part), [0, 2])
+        for k, d in part:
+            ch.append_row([k, None if d is None else _dec(lib, d)])
+        chunks.append(ch)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    got = ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64], [0, 2, 0])
+    ex.close()
+    ex.free()
+    b.free()
+    cnts = [c for _, _, c in got]
+    assert cnts == sorted(cnts, reverse=True) and all(c > 120 for c in cnts)
+    return got
+
+
+def test_oracle_topn_having():
+    rows = _data()
+    got = _run_topn_having(load_oracle(), rows)
+    assert 0 < len(got) <= 5
+
+
+@pytest.mark.gpu
+def test_topn_having_parity():
+    rows = _data()
+    want = _run_topn_having(load_oracle(), rows)
+    got = _run_topn_having(load_product(), rows)
+    assert got == want

@@ -6261,12 +6261,22 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     if (rc != GX_OK && ex->err.empty()) ex->err = "plan compilation failed";
     (void)rc;
   } else if (rn.kind == PK_TOPN &&
-             ex->plan.nodes[rn.child].kind == PK_HASHAGG &&
-             ex->plan.nodes[rn.child].aggMode != GX_AGG_MODE_FINAL) {
-    // ORDER BY / TopN over a fusable aggregation: run the fused kernel, sort
-    // the (small) group output on the host — the reference sorts the agg
-    // output the same way (Q1's final Sort, sortexec/sort.go)
+             ((ex->plan.nodes[rn.child].kind == PK_HASHAGG &&
+               ex->plan.nodes[rn.child].aggMode != GX_AGG_MODE_FINAL) ||
+              (ex->plan.nodes[rn.child].kind == PK_SELECTION &&
+               ex->plan.nodes[ex->plan.nodes[rn.child].child].kind ==
+                   PK_HASHAGG &&
+               ex->plan.nodes[ex->plan.nodes[rn.child].child].aggMode !=
+                   GX_AGG_MODE_FINAL))) {
+    // ORDER BY / TopN over a fusable aggregation, optionally through a
+    // HAVING Selection: run the fused kernel, filter + sort the (small)
+    // group output on the host (the reference sorts the agg output the
+    // same way — Q1's final Sort, sortexec/sort.go)
     int aggRoot = rn.child;
+    if (ex->plan.nodes[aggRoot].kind == PK_SELECTION) {
+      ex->havingConds = ex->plan.nodes[aggRoot].exprs;
+      aggRoot = ex->plan.nodes[aggRoot].child;
+    }
     int aggWidth = (int)ex->plan.nodes[aggRoot].exprs.size() +
                    (int)ex->plan.nodes[aggRoot].aggFuncs.size();
     bool ok = true;

@@ -148,3 +148,51 @@ def test_topn_having_parity():
     want = _run_topn_having(load_oracle(), rows)
     got = _run_topn_having(load_product(), rows)
     assert got == want
+
+
+def _run_having_distinct(lib, rows):
+    """HAVING over a DISTINCT aggregate: the decode folds the distinct
+    rewrite FIRST, then filters the user-schema rows."""
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2])
+    agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                    [(6, b.colref(1, GX_TYPE_DECIMAL, 2), 0)])
+    hav = b.selection(agg, [b.call(GX_F_GT, GX_TYPE_I64, 0,
+                                   b.colref(1, GX_TYPE_I64),
+                                   b.const_i64(20))])
+    ex = b.build(hav)
+    chunks = []
+    for base in range(0, len(rows), 1000):
+        part = rows[base:base + 1000]
+        ch = PyChunk([GX_TYPE_I64, GX_TYPE_DECIMAL], len(part), [0, 2])
+        for k, d in part:
+            ch.append_row([k, None if d is None else _dec(lib, d)])
+        chunks.append(ch)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    got = sorted(ex.pull_all([GX_TYPE_I64, GX_TYPE_I64]))
+    ex.close()
+    ex.free()
+    b.free()
+    assert all(c > 20 for _, c in got)
+    return got
+
+
+def test_oracle_having_distinct():
+    rows = _data()
+    got = _run_having_distinct(load_oracle(), rows)
+    want = {}
+    for k, d in rows:
+        want.setdefault(k, set())
+        if d is not None:
+            want[k].add(d)
+    assert got == sorted((k, len(s)) for k, s in want.items() if len(s) > 20)
+    assert len(got) > 3
+
+
+@pytest.mark.gpu
+def test_having_distinct_parity():
+    rows = _data()
+    want = _run_having_distinct(load_oracle(), rows)
+    got = _run_having_distinct(load_product(), rows)
+    assert got == want

@@ -1892,6 +1892,30 @@ static int32_t compileProject(gx_exec* ex) {
 
 static int32_t compileFused(gx_exec* ex);
 
+// structural validation of HAVING conditions at compile time (the value
+// checks run at decode): leaves are <agg output col cmp const> or
+// IS [NOT] NULL(col), composed with LogicOr; col indexes bound by the
+// aggregate's USER output width
+static bool validHavingCond(gx_exec* ex, int cid, int width) {
+  const PExpr& e = ex->plan.exprs[cid];
+  if (e.kind != EK_CALL) return false;
+  if (e.func == GX_F_OR && e.args.size() == 2)
+    return validHavingCond(ex, e.args[0], width) &&
+           validHavingCond(ex, e.args[1], width);
+  if ((e.func == GX_F_IS_NULL || e.func == GX_F_IS_NOT_NULL) &&
+      e.args.size() == 1) {
+    const PExpr& a0 = ex->plan.exprs[e.args[0]];
+    return a0.kind == EK_COLREF && a0.colIdx >= 0 && a0.colIdx < width;
+  }
+  if (e.func > GX_F_NE || e.args.size() != 2) return false;
+  const PExpr& l = ex->plan.exprs[e.args[0]];
+  const PExpr& r = ex->plan.exprs[e.args[1]];
+  const PExpr* col = l.kind == EK_COLREF ? &l : &r;
+  const PExpr* cst = l.kind == EK_COLREF ? &r : &l;
+  return col->kind == EK_COLREF && cst->kind == EK_CONST &&
+         col->colIdx >= 0 && col->colIdx < width;
+}
+
 // general aggregation over joined rows: HashAgg <- [Projection] <-
 // [Selection] <- HashJoin. The join materializes its output table on device
 // (runHashJoin), then the fused aggregation kernel runs over that table —
@@ -6233,6 +6257,16 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     // aggregation and filter the (small) decoded group rows on the host
     // (the reference evaluates HAVING the same way, above the agg)
     ex->havingConds = rn.exprs;
+    {
+      const PNode& an = ex->plan.nodes[rn.child];
+      int width = (int)(an.exprs.size() + an.aggFuncs.size());
+      for (int cid : ex->havingConds)
+        if (!validHavingCond(ex, cid, width)) {
+          ex->err = "unsupported HAVING condition (col cmp const, IS NULL, "
+                    "OR of those)";
+          return ex;
+        }
+    }
     int aggRoot = rn.child;
     int node = aggRoot;
     while (ex->plan.nodes[node].kind == PK_HASHAGG ||
@@ -6276,6 +6310,14 @@ gx_exec* gx_build(gx_pb* pb, int32_t root, int32_t device) {
     if (ex->plan.nodes[aggRoot].kind == PK_SELECTION) {
       ex->havingConds = ex->plan.nodes[aggRoot].exprs;
       aggRoot = ex->plan.nodes[aggRoot].child;
+      const PNode& an = ex->plan.nodes[aggRoot];
+      int width = (int)(an.exprs.size() + an.aggFuncs.size());
+      for (int cid : ex->havingConds)
+        if (!validHavingCond(ex, cid, width)) {
+          ex->err = "unsupported HAVING condition (col cmp const, IS NULL, "
+                    "OR of those)";
+          return ex;
+        }
     }
     int aggWidth = (int)ex->plan.nodes[aggRoot].exprs.size() +
                    (int)ex->plan.nodes[aggRoot].aggFuncs.size();

@@ -92,3 +92,69 @@ def test_too_many_group_keys_rejected():
         ch = PyChunk(types, 1)
         return agg, src, [ch]
     _expect_error(plan, "group", "key")
+
+
+def test_bad_join_type_rejected():
+    def plan(b, lib):
+        b1 = b.source([GX_TYPE_I64])
+        p1 = b.source([GX_TYPE_I64])
+        j = b.hashjoin(b1, p1, [b.colref(0, GX_TYPE_I64)],
+                       [b.colref(0, GX_TYPE_I64)], join_type=8)
+        ch1 = PyChunk([GX_TYPE_I64], 1)
+        ch2 = PyChunk([GX_TYPE_I64], 1)
+        return j, None, None  # bind manually below
+    lib = load_product()
+    b = P.Builder(lib)
+    b1 = b.source([GX_TYPE_I64])
+    p1 = b.source([GX_TYPE_I64])
+    j = b.hashjoin(b1, p1, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)], join_type=8)
+    ex = b.build(j)
+    rc = lib.gx_open(ex.ex)
+    err = ex.error()
+    ex.free()
+    b.free()
+    assert rc != 0 and "join type" in err
+
+
+def test_too_many_join_keys_rejected():
+    lib = load_product()
+    b = P.Builder(lib)
+    types = [GX_TYPE_I64] * 5
+    b1 = b.source(types)
+    p1 = b.source(types)
+    keys_b = [b.colref(i, GX_TYPE_I64) for i in range(5)]
+    keys_p = [b.colref(i, GX_TYPE_I64) for i in range(5)]
+    j = b.hashjoin(b1, p1, keys_b, keys_p)
+    ex = b.build(j)
+    rc = lib.gx_open(ex.ex)
+    err = ex.error()
+    ex.free()
+    b.free()
+    assert rc != 0 and "key" in err
+
+
+def test_long_string_const_rejected():
+    def plan(b, lib):
+        from tests.gxlib import GX_F_EQ
+        src = b.source([GX_TYPE_STRING, GX_TYPE_I64])
+        cond = b.call(GX_F_EQ, GX_TYPE_I64, 0,
+                      b.colref(0, GX_TYPE_STRING),
+                      lib.gx_pb_const_str(b.pb, b"x" * 20, 20))
+        sel = b.selection(src, [cond])
+        ch = PyChunk([GX_TYPE_STRING, GX_TYPE_I64], 1, None, [64, None])
+        return sel, src, [ch]
+    _expect_error(plan, "string const", "16")
+
+
+def test_tuple_outside_distinct_rejected():
+    def plan(b, lib):
+        from tests.gxlib import GX_F_TUPLE
+        src = b.source([GX_TYPE_I64, GX_TYPE_I64])
+        tup = b.call(GX_F_TUPLE, GX_TYPE_I64, 0,
+                     b.colref(0, GX_TYPE_I64), b.colref(1, GX_TYPE_I64))
+        agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                        [(GX_AGG_SUM, tup, 0)])
+        return agg, src, _i64_chunk()
+    _expect_error(plan, "tuple", "DISTINCT", "decimal",
+                  "unsupported function")

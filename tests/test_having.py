@@ -196,3 +196,58 @@ def test_having_distinct_parity():
     want = _run_having_distinct(load_oracle(), rows)
     got = _run_having_distinct(load_product(), rows)
     assert got == want
+
+
+def _run_having_over_join(lib):
+    """HAVING above an aggregation over a JOIN: the agg-over-join pipeline
+    materializes the joined table, the fused agg runs over it, and the
+    decode filters groups."""
+    rng = np.random.default_rng(71)
+    brows = [[i, int(rng.integers(0, 6))] for i in range(150)]
+    prows = [[int(rng.integers(0, 200)), int(rng.integers(1, 9))]
+             for _ in range(4000)]
+    b = P.Builder(lib)
+    t2 = [GX_TYPE_I64, GX_TYPE_I64]
+    bsrc = b.source(t2)
+    psrc = b.source(t2)
+    j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)])
+    # joined: b.key, b.grp, p.key, p.val — group by b.grp, count(*)
+    agg = b.hashagg(j, [b.colref(1, GX_TYPE_I64)],
+                    [(GX_AGG_COUNT, -1, 0)])
+    hav = b.selection(agg, [b.call(GX_F_GT, GX_TYPE_I64, 0,
+                                   b.colref(1, GX_TYPE_I64),
+                                   b.const_i64(400))])
+    ex = b.build(hav)
+    bch = PyChunk(t2, len(brows))
+    for r in brows:
+        bch.append_row(r)
+    pch = PyChunk(t2, len(prows))
+    for r in prows:
+        pch.append_row(r)
+    ex.bind_chunks(bsrc, [bch])
+    ex.bind_chunks(psrc, [pch])
+    ex.open()
+    got = sorted(ex.pull_all(t2))
+    ex.close()
+    ex.free()
+    b.free()
+    bmap = {k: g for k, g in brows}
+    want = {}
+    for k, _ in prows:
+        if k in bmap:
+            want[bmap[k]] = want.get(bmap[k], 0) + 1
+    assert got == sorted((g, c) for g, c in want.items() if c > 400)
+    assert 0 < len(got) < 6
+    return got
+
+
+def test_oracle_having_over_join():
+    _run_having_over_join(load_oracle())
+
+
+@pytest.mark.gpu
+def test_having_over_join_parity():
+    want = _run_having_over_join(load_oracle())
+    got = _run_having_over_join(load_product())
+    assert got == want

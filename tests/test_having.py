@@ -217,7 +217,7 @@ def _run_having_over_join(lib):
                     [(GX_AGG_COUNT, -1, 0)])
     hav = b.selection(agg, [b.call(GX_F_GT, GX_TYPE_I64, 0,
                                    b.colref(1, GX_TYPE_I64),
-                                   b.const_i64(400))])
+                                   b.const_i64(450))])
     ex = b.build(hav)
     bch = PyChunk(t2, len(brows))
     for r in brows:
@@ -237,8 +237,8 @@ def _run_having_over_join(lib):
     for k, _ in prows:
         if k in bmap:
             want[bmap[k]] = want.get(bmap[k], 0) + 1
-    assert got == sorted((g, c) for g, c in want.items() if c > 400)
-    assert 0 < len(got) < 6
+    assert got == sorted((g, c) for g, c in want.items() if c > 450)
+    assert 0 < len(got) < 6  # the cut removes some groups (seeded data)
     return got
 
 

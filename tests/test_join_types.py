@@ -292,3 +292,30 @@ def test_left_outer_semi_random_parity(jt):
     got = _run(load_product(), jt, brows, prows)
     assert got == want
     assert len(got) == len(prows)  # every probe row exactly once
+
+
+def test_oracle_right_outer_drain_chunked():
+    """Right-outer unmatched-build drain across MULTIPLE Next calls
+    (>1024 unmatched rows must stream out in executor-contract chunks)."""
+    lib = load_oracle()
+    brows = [[i, i * 10] for i in range(3000)]       # keys 0..2999
+    prows = [[i, -i] for i in range(100)]            # matches 0..99 only
+    rows = _run(lib, 2, brows, prows)
+    assert len(rows) == 100 + 2900                   # matched + drained
+    drained = [r for r in rows if r[2] is None]
+    assert len(drained) == 2900
+    assert all(r[3] is None for r in drained)
+    matched = [r for r in rows if r[2] is not None]
+    assert sorted(r[0] for r in matched) == list(range(100))
+
+
+def test_oracle_semi_dedup_large():
+    """Semi join emits each probe row ONCE regardless of duplicate build
+    matches, across chunk boundaries."""
+    lib = load_oracle()
+    brows = [[k, j] for k in range(50) for j in range(40)]  # 40 dups/key
+    prows = [[i % 80, i] for i in range(2000)]
+    rows = _run(lib, 3, brows, prows)
+    want = sorted([tuple(r) for r in prows if r[0] < 50],
+                  key=lambda r: tuple((x is None, x) for x in r))
+    assert rows == want

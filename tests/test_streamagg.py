@@ -81,3 +81,29 @@ def test_streamagg_parity(desc):
     a = run_sa(load_oracle(), 50000, desc=desc)
     b = run_sa(load_product(), 50000, desc=desc)
     assert a == b  # ordered comparison: stream order must match too
+
+
+def test_streamagg_distinct_oracle(oracle_lib):
+    """StreamAgg with a DISTINCT aggregate over grouped input: the
+    per-group value sets reset with each new group (the stream order
+    guarantees contiguity)."""
+    from tests.gxlib import GX_TYPE_I64
+    from tidb_amd.chunkpy import PyChunk
+    from tidb_amd import plan as P
+    b = P.Builder(oracle_lib)
+    src = b.source([GX_TYPE_I64, GX_TYPE_I64])
+    agg = b.streamagg(src, [b.colref(0, GX_TYPE_I64)],
+                      [(6, b.colref(1, GX_TYPE_I64), 0)])
+    ex = b.build(agg)
+    rows = [(1, 5), (1, 5), (1, 7), (2, 5), (2, None), (3, 9), (3, 9),
+            (3, 8), (3, 5)]
+    ch = PyChunk([GX_TYPE_I64] * 2, len(rows))
+    for r in rows:
+        ch.append_row(list(r))
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    got = ex.pull_all([GX_TYPE_I64] * 2)
+    ex.close()
+    ex.free()
+    b.free()
+    assert got == [(1, 2), (2, 1), (3, 3)]  # stream order preserved

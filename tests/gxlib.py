@@ -38,6 +38,7 @@ GX_F_IF = 52
 GX_F_OR = 53
 
 GX_AGG_COUNT, GX_AGG_SUM, GX_AGG_AVG, GX_AGG_MIN, GX_AGG_MAX, GX_AGG_FIRSTROW = range(6)
+GX_AGG_COUNT_DISTINCT, GX_AGG_SUM_DISTINCT, GX_AGG_AVG_DISTINCT = 6, 7, 8
 GX_AGG_MODE_COMPLETE, GX_AGG_MODE_PARTIAL, GX_AGG_MODE_FINAL = 0, 1, 2
 
 GX_TPCH_LINEITEM, GX_TPCH_ORDERS, GX_TPCH_CUSTOMER = 0, 1, 2

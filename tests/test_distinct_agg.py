@@ -23,9 +23,9 @@ from tests.gxlib import (GX_AGG_MODE_PARTIAL, GX_F_GT, GX_TYPE_DECIMAL,
                          GX_TYPE_I64, GX_TYPE_STRING, load_oracle,
                          load_product)
 from tidb_amd import plan as P
+from tests.gxlib import (GX_AGG_AVG_DISTINCT, GX_AGG_COUNT_DISTINCT,
+                         GX_AGG_SUM_DISTINCT)
 from tidb_amd.chunkpy import PyChunk
-
-GX_AGG_COUNT_DISTINCT, GX_AGG_SUM_DISTINCT, GX_AGG_AVG_DISTINCT = 6, 7, 8
 
 TYPES = [GX_TYPE_I64, GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_STRING]
 FRACS = [0, 0, 2, 0]

@@ -319,3 +319,16 @@ def test_oracle_semi_dedup_large():
     want = sorted([tuple(r) for r in prows if r[0] < 50],
                   key=lambda r: tuple((x is None, x) for x in r))
     assert rows == want
+
+
+def test_oracle_left_outer_semi_chunked():
+    """jt 6/7 at >1024 probe rows: one output row per probe row across
+    chunk boundaries, flags stable."""
+    lib = load_oracle()
+    brows = [[k, k] for k in range(0, 100, 2)]  # even keys present
+    prows = [[i % 100, i] for i in range(3000)]
+    for jt in (6, 7):
+        rows = _run(lib, jt, brows, prows)
+        assert len(rows) == 3000
+        for k, _, flag in rows:
+            assert flag == (1 if k % 2 == 0 else 0)

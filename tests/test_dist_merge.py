@@ -295,3 +295,45 @@ def test_shards_final_merge_full_family():
     as_map = lambda rows: {(r[0], r[1]): tuple(r[2:]) for r in rows}
     assert as_map(got) == as_map(want)
     assert len(got) >= 4
+
+
+def test_final_merge_string_extremes_both_libs():
+    """FINAL merge of MIN/MAX over VARCHAR partials: binary collation with
+    PAD SPACE ('abc ' == 'abc' for compare), NULL shard partials skipped.
+    Both libraries' host FINAL paths, CPU."""
+    from tests.gxlib import (GX_AGG_MAX, GX_AGG_MIN, GX_TYPE_STRING,
+                             load_product)
+    part_types = [GX_TYPE_I64, GX_TYPE_STRING, GX_TYPE_STRING]
+    partials = [
+        (1, "apple", "pear"),
+        (1, "apple  ", "zebra"),   # PAD SPACE: ties 'apple'
+        (1, None, None),           # all-NULL shard partial
+        (2, "kiwi", "kiwi"),
+    ]
+
+    def run(lib):
+        b = P.Builder(lib)
+        src = b.source(part_types)
+        agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                        [(GX_AGG_MIN, b.colref(1, GX_TYPE_STRING), 0),
+                         (GX_AGG_MAX, b.colref(2, GX_TYPE_STRING), 0)],
+                        GX_AGG_MODE_FINAL)
+        chunk = PyChunk(part_types, len(partials), None, [None, 256, 256])
+        for r in partials:
+            chunk.append_row(list(r))
+        ex = b.build(agg)
+        ex.bind_chunks(src, [chunk])
+        ex.open()
+        rows = sorted(ex.pull_all([GX_TYPE_I64, GX_TYPE_STRING,
+                                   GX_TYPE_STRING]))
+        ex.close()
+        ex.free()
+        b.free()
+        return rows
+
+    got_o = run(load_oracle())
+    got_p = run(load_product())
+    assert got_o == got_p
+    assert got_o[0][0] == 1 and got_o[0][2] == "zebra"
+    assert got_o[0][1] in ("apple", "apple  ")  # PAD-SPACE tie, either rep
+    assert got_o[1] == (2, "kiwi", "kiwi")

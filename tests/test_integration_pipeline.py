@@ -126,3 +126,48 @@ def test_composed_pipeline_parity(seed):
     got = _run_oracle(load_product(), rows)
     assert got == want
     assert len(got) > 5
+
+
+def test_oracle_in_scalar_aggregated():
+    """IN-scalar join feeding an aggregation (oracle, CPU): count how many
+    probe rows per group are IN the build set — agg over the jt-6 flag."""
+    from tests.gxlib import GX_AGG_COUNT, GX_AGG_SUM, GX_F_CAST_DEC
+    lib = load_oracle()
+    rng = np.random.default_rng(97)
+    brows = [[int(k), 0] for k in rng.choice(200, size=60, replace=False)]
+    prows = [[int(rng.integers(0, 200)), int(rng.integers(0, 8))]
+             for _ in range(3000)]
+    b = P.Builder(lib)
+    t2 = [GX_TYPE_I64, GX_TYPE_I64]
+    bsrc = b.source(t2)
+    psrc = b.source(t2)
+    j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                   [b.colref(0, GX_TYPE_I64)], join_type=6)
+    # jt-6 output: p.key, p.grp, flag — group by p.grp,
+    # sum(cast(flag)) counts the IN rows
+    agg = b.hashagg(j, [b.colref(1, GX_TYPE_I64)],
+                    [(GX_AGG_SUM, b.call(GX_F_CAST_DEC, GX_TYPE_DECIMAL, 0,
+                                         b.colref(2, GX_TYPE_I64)), 0),
+                     (GX_AGG_COUNT, -1, 0)])
+    ex = b.build(agg)
+    bch = PyChunk(t2, len(brows))
+    for r in brows:
+        bch.append_row(r)
+    pch = PyChunk(t2, len(prows))
+    for r in prows:
+        pch.append_row(r)
+    ex.bind_chunks(bsrc, [bch])
+    ex.bind_chunks(psrc, [pch])
+    ex.open()
+    got = sorted(ex.pull_all([GX_TYPE_I64, GX_TYPE_DECIMAL, GX_TYPE_I64],
+                             [0, 0, 0]))
+    ex.close()
+    ex.free()
+    b.free()
+    bset = {k for k, _ in brows}
+    want = {}
+    for k, g in prows:
+        s, c = want.get(g, (0, 0))
+        want[g] = (s + (1 if k in bset else 0), c + 1)
+    assert got == sorted((g, str(s), c) for g, (s, c) in want.items())
+    assert len(got) == 8

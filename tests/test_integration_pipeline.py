@@ -171,3 +171,49 @@ def test_oracle_in_scalar_aggregated():
         want[g] = (s + (1 if k in bset else 0), c + 1)
     assert got == sorted((g, str(s), c) for g, (s, c) in want.items())
     assert len(got) == 8
+
+
+def test_oracle_not_in_filter_then_agg():
+    """NOT IN (jt 5) as a filter feeding aggregation (oracle, CPU):
+    count the rows whose key is NOT in the build set, per group —
+    with and without a NULL poisoning the build side."""
+    from tests.gxlib import GX_AGG_COUNT
+    lib = load_oracle()
+    rng = np.random.default_rng(101)
+    clean = [[int(k), 0] for k in rng.choice(150, size=40, replace=False)]
+    prows = [[int(rng.integers(0, 150)), int(rng.integers(0, 5))]
+             for _ in range(2000)]
+
+    def run(brows):
+        b = P.Builder(lib)
+        t2 = [GX_TYPE_I64, GX_TYPE_I64]
+        bsrc = b.source(t2)
+        psrc = b.source(t2)
+        j = b.hashjoin(bsrc, psrc, [b.colref(0, GX_TYPE_I64)],
+                       [b.colref(0, GX_TYPE_I64)], join_type=5)
+        agg = b.hashagg(j, [b.colref(1, GX_TYPE_I64)],
+                        [(GX_AGG_COUNT, -1, 0)])
+        ex = b.build(agg)
+        bch = PyChunk(t2, max(len(brows), 1))
+        for r in brows:
+            bch.append_row(r)
+        pch = PyChunk(t2, len(prows))
+        for r in prows:
+            pch.append_row(r)
+        ex.bind_chunks(bsrc, [bch])
+        ex.bind_chunks(psrc, [pch])
+        ex.open()
+        got = sorted(ex.pull_all([GX_TYPE_I64, GX_TYPE_I64]))
+        ex.close()
+        ex.free()
+        b.free()
+        return got
+
+    bset = {k for k, _ in clean}
+    want = {}
+    for k, g in prows:
+        if k not in bset:
+            want[g] = want.get(g, 0) + 1
+    assert run(clean) == sorted(want.items())
+    # one NULL build key: NOT IN never true -> zero rows -> zero groups
+    assert run(clean + [[None, 0]]) == []

@@ -251,3 +251,44 @@ def test_having_over_join_parity():
     want = _run_having_over_join(load_oracle())
     got = _run_having_over_join(load_product())
     assert got == want
+
+
+def test_oracle_topn_having_distinct():
+    """Full stack: ORDER BY count-distinct DESC LIMIT 3 over HAVING over
+    COUNT(DISTINCT) (oracle, CPU)."""
+    from tests.gxlib import GX_AGG_COUNT_DISTINCT
+    lib = load_oracle()
+    rows = _data()
+    b = P.Builder(lib)
+    src = b.source([GX_TYPE_I64, GX_TYPE_DECIMAL], [0, 2])
+    agg = b.hashagg(src, [b.colref(0, GX_TYPE_I64)],
+                    [(GX_AGG_COUNT_DISTINCT, b.colref(1, GX_TYPE_DECIMAL, 2),
+                      0)])
+    hav = b.selection(agg, [b.call(GX_F_GT, GX_TYPE_I64, 0,
+                                   b.colref(1, GX_TYPE_I64),
+                                   b.const_i64(10))])
+    root = b.topn(hav, [b.colref(1, GX_TYPE_I64),
+                        b.colref(0, GX_TYPE_I64)], [1, 0], 3)
+    ex = b.build(root)
+    chunks = []
+    for base in range(0, len(rows), 1000):
+        part = rows[base:base + 1000]
+        ch = PyChunk([GX_TYPE_I64, GX_TYPE_DECIMAL], len(part), [0, 2])
+        for k, d in part:
+            ch.append_row([k, None if d is None else _dec(lib, d)])
+        chunks.append(ch)
+    ex.bind_chunks(src, chunks)
+    ex.open()
+    got = ex.pull_all([GX_TYPE_I64, GX_TYPE_I64])
+    ex.close()
+    ex.free()
+    b.free()
+    want = {}
+    for k, d in rows:
+        want.setdefault(k, set())
+        if d is not None:
+            want[k].add(d)
+    exp = sorted(((k, len(s)) for k, s in want.items() if len(s) > 10),
+                 key=lambda r: (-r[1], r[0]))[:3]
+    assert got == exp
+    assert len(got) == 3

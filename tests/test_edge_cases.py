@@ -163,3 +163,18 @@ def test_keyless_limit_parity(limit, offset, with_sel):
     want = _run_limit(load_oracle(), limit, offset, with_sel)
     got = _run_limit(load_product(), limit, offset, with_sel)
     assert got == want
+
+
+def test_oracle_limit_offset_beyond_input():
+    """OFFSET past the input: empty result, no error (LimitExec)."""
+    rows = _run_limit(load_oracle(), 5, 5000, False)
+    assert rows == []
+    rows = _run_limit(load_oracle(), 0, 0, False)
+    assert rows == []  # LIMIT 0
+
+
+def test_oracle_topn_offset_beyond_groups():
+    from tests.test_full_sort import KEYS_2, run_sort
+    rows = run_sort(load_oracle(), KEYS_2, [0, 0], n_rows=7, limit=10,
+                    offset=20)
+    assert rows == []

@@ -117,3 +117,98 @@ def test_fuzz_decimal_exprs(seed):
                 assert v is None, (i, j, v)
             else:
                 assert Fraction(v) == want, (i, j, v, want)
+
+
+@pytest.mark.parametrize("seed", [11, 53, 97])
+def test_fuzz_groupby_aggs(seed):
+    """Random GROUP BY plans: 1-2 int keys, a random mix of
+    sum/avg/count/min/max/firstrow over decimal/int args with NULLs,
+    oracle vs exact Fraction accounting."""
+    from tests.gxlib import (GX_AGG_AVG, GX_AGG_COUNT, GX_AGG_FIRSTROW,
+                             GX_AGG_MAX, GX_AGG_MIN, GX_AGG_SUM,
+                             GX_F_CAST_DEC)
+    lib = load_oracle()
+    rng = np.random.default_rng(seed)
+    n = 1500
+    nkeys = int(rng.integers(1, 3))
+    sc = int(rng.integers(1, 4))
+    types = [GX_TYPE_I64] * nkeys + [GX_TYPE_DECIMAL, GX_TYPE_I64]
+    fracs = [0] * nkeys + [sc, 0]
+    data = []
+    for _ in range(n):
+        keys = [int(rng.integers(0, 8)) for _ in range(nkeys)]
+        d = None if rng.random() < 0.2 else \
+            Fraction(int(rng.integers(-10**4, 10**4)), 10**sc)
+        v = None if rng.random() < 0.2 else int(rng.integers(-500, 500))
+        data.append((tuple(keys), d, v))
+
+    b = P.Builder(lib)
+    src = b.source(types, fracs)
+    kexprs = [b.colref(i, GX_TYPE_I64) for i in range(nkeys)]
+    dref = b.colref(nkeys, GX_TYPE_DECIMAL, sc)
+    vref = b.colref(nkeys + 1, GX_TYPE_I64)
+    vdec = b.call(GX_F_CAST_DEC, GX_TYPE_DECIMAL, 0, vref)
+    aggs = [(GX_AGG_SUM, dref, sc), (GX_AGG_COUNT, dref, 0),
+            (GX_AGG_MIN, vref, 0), (GX_AGG_MAX, vref, 0),
+            (GX_AGG_AVG, vdec, 4), (GX_AGG_COUNT, -1, 0)]
+    agg = b.hashagg(src, kexprs, aggs)
+    ex = b.build(agg)
+    ch = PyChunk(types, n, fracs)
+    for keys, d, v in data:
+        row = list(keys)
+        if d is None:
+            row.append(None)
+        else:
+            q = d * 10**sc
+            s = f"{'-' if q < 0 else ''}{abs(q.numerator)//10**sc}"
+            s += f".{abs(q.numerator) % 10**sc:0{sc}d}"
+            row.append(_dec_cached(lib, s))
+        row.append(v)
+        ch.append_row(row)
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    out_t = [GX_TYPE_I64] * nkeys + [GX_TYPE_DECIMAL, GX_TYPE_I64,
+                                     GX_TYPE_I64, GX_TYPE_I64,
+                                     GX_TYPE_DECIMAL, GX_TYPE_I64]
+    got = sorted(ex.pull_all(out_t, [0] * nkeys + [sc, 0, 0, 0, 4, 0]))
+    ex.close()
+    ex.free()
+    b.free()
+
+    groups = {}
+    for keys, d, v in data:
+        g = groups.setdefault(keys, {"s": Fraction(0), "dc": 0, "mn": None,
+                                     "mx": None, "vs": Fraction(0), "vc": 0,
+                                     "n": 0})
+        g["n"] += 1
+        if d is not None:
+            g["s"] += d
+            g["dc"] += 1
+        if v is not None:
+            g["mn"] = v if g["mn"] is None else min(g["mn"], v)
+            g["mx"] = v if g["mx"] is None else max(g["mx"], v)
+            g["vs"] += v
+            g["vc"] += 1
+    want = []
+    for keys, g in groups.items():
+        s = None if g["dc"] == 0 else g["s"]
+        avg = None if g["vc"] == 0 else g["vs"] / g["vc"]
+        want.append(tuple(keys) + (s, g["dc"], g["mn"], g["mx"], avg,
+                                   g["n"]))
+    want.sort(key=lambda r: r[:nkeys])
+    assert len(got) == len(want)
+    for gr, wr in zip(got, want):
+        assert gr[:nkeys] == wr[:nkeys]
+        s = gr[nkeys]
+        assert (s is None) == (wr[nkeys] is None)
+        if s is not None:
+            assert Fraction(s) == wr[nkeys]
+        assert gr[nkeys + 1] == wr[nkeys + 1]
+        assert gr[nkeys + 2] == wr[nkeys + 2]
+        assert gr[nkeys + 3] == wr[nkeys + 3]
+        a = gr[nkeys + 4]
+        assert (a is None) == (wr[nkeys + 4] is None)
+        if a is not None:  # avg rounds half-up to frac 4
+            exact = wr[nkeys + 4]
+            assert abs(Fraction(a) - exact) <= Fraction(1, 2 * 10**4)
+        assert gr[nkeys + 5] == wr[nkeys + 5]

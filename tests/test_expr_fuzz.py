@@ -212,3 +212,71 @@ def test_fuzz_groupby_aggs(seed):
             exact = wr[nkeys + 4]
             assert abs(Fraction(a) - exact) <= Fraction(1, 2 * 10**4)
         assert gr[nkeys + 5] == wr[nkeys + 5]
+
+
+@pytest.mark.parametrize("seed", [7, 29])
+def test_fuzz_join_types_model(seed):
+    """All 8 join types vs one independent Python model on random nullable
+    data (oracle, CPU) — one generator, eight semantics."""
+    from tests.test_join_types import _run
+    lib = load_oracle()
+    rng = np.random.default_rng(seed)
+    brows = [[None if rng.random() < 0.06 else int(rng.integers(0, 40)),
+              int(i)] for i in range(300)]
+    prows = [[None if rng.random() < 0.06 else int(rng.integers(0, 60)),
+              -int(i)] for i in range(900)]
+    bkeys = [k for k, _ in brows if k is not None]
+    bset = set(bkeys)
+    bnull = any(k is None for k, _ in brows)
+    bmatches = {}
+    for k, v in brows:
+        if k is not None:
+            bmatches.setdefault(k, []).append(v)
+
+    def model(jt):
+        out = []
+        for k, v in prows:
+            ms = bmatches.get(k, []) if k is not None else []
+            if jt == 0:
+                out += [(k, bv, k, v) for bv in ms]
+            elif jt == 1:
+                out += [(k, bv, k, v) for bv in ms] if ms else \
+                    [(None, None, k, v)]
+            elif jt == 3:
+                if ms:
+                    out.append((k, v))
+            elif jt == 4:
+                if not ms:
+                    out.append((k, v))
+            elif jt == 5:
+                if len(brows) == 0:
+                    out.append((k, v))
+                elif k is None or bnull:
+                    pass
+                elif not ms:
+                    out.append((k, v))
+            elif jt == 6:
+                out.append((k, v, 1 if ms else 0))
+            elif jt == 7:
+                if ms:
+                    out.append((k, v, 1))
+                elif (k is None and len(brows) > 0) or bnull:
+                    out.append((k, v, None))
+                else:
+                    out.append((k, v, 0))
+        if jt == 2:
+            matched_b = set()
+            for k, v in prows:
+                if k is not None and k in bmatches:
+                    matched_b.add(k)
+                    out += [(k, bv, k, v) for bv in bmatches[k]]
+            for k, v in brows:
+                if k is None or k not in {kk for kk, _ in prows
+                                          if kk is not None}:
+                    out.append((k, v, None, None))
+        key = lambda r: tuple((x is None, x) for x in r)
+        return sorted(out, key=key)
+
+    for jt in range(8):
+        got = _run(lib, jt, brows, prows)
+        assert got == model(jt), f"join_type {jt}"

@@ -280,3 +280,43 @@ def test_fuzz_join_types_model(seed):
     for jt in range(8):
         got = _run(lib, jt, brows, prows)
         assert got == model(jt), f"join_type {jt}"
+
+
+@pytest.mark.parametrize("seed", [13, 61])
+def test_fuzz_sort_nullable_multikey(seed):
+    """TopN over 2 nullable keys with random asc/desc vs Python's sort
+    with sortexec NULL-first-asc / NULL-last-desc semantics (oracle)."""
+    lib = load_oracle()
+    rng = np.random.default_rng(seed)
+    desc = [int(rng.integers(0, 2)), int(rng.integers(0, 2))]
+    rows = []
+    for i in range(1200):
+        a = None if rng.random() < 0.12 else int(rng.integers(-50, 50))
+        bb = None if rng.random() < 0.12 else int(rng.integers(-9, 9))
+        rows.append((a, bb, i))
+    b = P.Builder(lib)
+    types = [GX_TYPE_I64] * 3
+    src = b.source(types)
+    root = b.topn(src, [b.colref(0, GX_TYPE_I64), b.colref(1, GX_TYPE_I64)],
+                  desc, 200)
+    ex = b.build(root)
+    ch = PyChunk(types, len(rows))
+    for r in rows:
+        ch.append_row(list(r))
+    ex.bind_chunks(src, [ch])
+    ex.open()
+    got = ex.pull_all(types)
+    ex.close()
+    ex.free()
+    b.free()
+
+    def keyf(r):
+        out = []
+        for v, d in zip(r[:2], desc):
+            # NULL < any value; desc reverses
+            rank = (0 if v is None else 1, 0 if v is None else v)
+            out.append((-rank[0], -rank[1]) if d else rank)
+        return tuple(out)
+
+    want_keys = [keyf(r) for r in sorted(rows, key=keyf)][:200]
+    assert [keyf(r) for r in got] == want_keys

@@ -320,3 +320,17 @@ def test_fuzz_sort_nullable_multikey(seed):
 
     want_keys = [keyf(r) for r in sorted(rows, key=keyf)][:200]
     assert [keyf(r) for r in got] == want_keys
+
+
+def test_fuzz_substr_windows():
+    """SUBSTR byte semantics fuzz: 60 random (pos, len) pairs incl.
+    negatives and out-of-range vs the Python model (oracle)."""
+    from tests.test_string_builtins import _py_substr, _run_proj
+    rng = np.random.default_rng(37)
+    lib = load_oracle()
+    for _ in range(20):
+        pos = int(rng.integers(-12, 13))
+        ln = int(rng.integers(-2, 15))
+        got, rows = _run_proj(lib, pos, ln)
+        for (o0, _, _, _, _), (s, _) in zip(got, rows):
+            assert o0 == _py_substr(s, pos, ln), (pos, ln, s)
